@@ -1,0 +1,63 @@
+# brpc_amd build: C++ core -> brpc_amd/_core.so (pybind11, g++),
+# HIP gfx950 kernels -> brpc_amd/libbrpc_hip.so (hipcc, no torch dep).
+PY        := python3
+PY_INc    := $(shell $(PY) -c "import sysconfig; print(sysconfig.get_paths()['include'])")
+PYBIND_INC:= $(shell $(PY) -c "import pybind11; print(pybind11.get_include())")
+EXT_SUFFIX:= $(shell $(PY) -c "import sysconfig; print(sysconfig.get_config_var('EXT_SUFFIX'))")
+
+CXX       := g++
+HIPCC     := /opt/rocm/bin/hipcc
+GPU_ARCH  := gfx950
+
+CXXFLAGS  := -O2 -g -std=c++17 -fPIC -pthread -Wall -Wno-unused-function \
+             -Isrc -msse4.2 -fvisibility=hidden
+LDFLAGS   := -shared -pthread -ldl
+
+CORE_SRCS := $(wildcard src/base/*.cc) $(wildcard src/fiber/*.cc) $(wildcard src/rpc/*.cc) \
+             $(wildcard src/rpc/policy/*.cc) $(wildcard src/rpc/builtin/*.cc) \
+             $(wildcard src/var/*.cc)
+CORE_ASM  := $(wildcard src/fiber/*.S)
+BIND_SRCS := $(wildcard src/bindings/*.cc)
+CORE_OBJS := $(CORE_SRCS:%.cc=build/%.o) $(CORE_ASM:%.S=build/%.o)
+BIND_OBJS := $(BIND_SRCS:%.cc=build/%.bo)
+
+HIP_SRCS  := $(wildcard hip/*.hip)
+HIP_OBJS  := $(HIP_SRCS:%.hip=build/%.o)
+
+CORE_SO   := brpc_amd/_core$(EXT_SUFFIX)
+HIP_SO    := brpc_amd/libbrpc_hip.so
+
+all: $(CORE_SO) $(HIP_SO)
+
+core: $(CORE_SO)
+
+build/%.o: %.cc
+	@mkdir -p $(dir $@)
+	$(CXX) $(CXXFLAGS) -c $< -o $@
+
+build/%.o: %.S
+	@mkdir -p $(dir $@)
+	$(CXX) $(CXXFLAGS) -c $< -o $@
+
+build/%.bo: %.cc
+	@mkdir -p $(dir $@)
+	$(CXX) $(CXXFLAGS) -I$(PY_INc) -I$(PYBIND_INC) -c $< -o $@
+
+$(CORE_SO): $(CORE_OBJS) $(BIND_OBJS)
+	$(CXX) $(CORE_OBJS) $(BIND_OBJS) $(LDFLAGS) -o $@
+
+build/hip/%.o: hip/%.hip
+	@mkdir -p $(dir $@)
+	$(HIPCC) --offload-arch=$(GPU_ARCH) -O3 -std=c++17 -fPIC -Isrc -Ihip -c $< -o $@
+
+$(HIP_SO): $(HIP_OBJS)
+	@if [ -n "$(HIP_OBJS)" ]; then \
+	  $(HIPCC) --offload-arch=$(GPU_ARCH) -shared -fPIC $(HIP_OBJS) -o $@ ; \
+	else \
+	  echo "no hip sources yet"; \
+	fi
+
+clean:
+	rm -rf build $(CORE_SO) $(HIP_SO)
+
+.PHONY: all core clean

@@ -1,0 +1,8 @@
+#pragma once
+#include <pybind11/pybind11.h>
+namespace py = pybind11;
+
+void bind_base(py::module_& m);
+void bind_fiber(py::module_& m);
+void bind_rpc(py::module_& m);
+void bind_var(py::module_& m);

@@ -1,0 +1,92 @@
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "base/crc32c.h"
+#include "base/endpoint.h"
+#include "base/fast_rand.h"
+#include "base/iobuf.h"
+#include "bindings/bind.h"
+
+namespace {
+
+using bam::IOBuf;
+
+py::bytes iobuf_to_bytes(const IOBuf& b) {
+  std::string s = b.to_string();
+  return py::bytes(s);
+}
+
+}  // namespace
+
+void bind_base(py::module_& m) {
+  py::enum_<bam::Residency>(m, "Residency")
+      .value("HOST", bam::RES_HOST)
+      .value("PINNED", bam::RES_PINNED)
+      .value("HBM", bam::RES_HBM);
+
+  py::class_<IOBuf>(m, "IOBuf")
+      .def(py::init<>())
+      .def("append", [](IOBuf& b, py::bytes data) {
+        char* ptr;
+        Py_ssize_t len;
+        PyBytes_AsStringAndSize(data.ptr(), &ptr, &len);
+        b.append(ptr, (size_t)len);
+      })
+      .def("append_iobuf", [](IOBuf& b, const IOBuf& other) { b.append(other); })
+      .def("append_with_residency",
+           [](IOBuf& b, py::bytes data, bam::Residency res, int dev, uint32_t block_payload) {
+             char* ptr;
+             Py_ssize_t len;
+             PyBytes_AsStringAndSize(data.ptr(), &ptr, &len);
+             return b.append_with_residency(ptr, (size_t)len, res, dev, block_payload);
+           },
+           py::arg("data"), py::arg("res"), py::arg("dev") = 0, py::arg("block_payload") = 0)
+      .def("cutn_to_iobuf", [](IOBuf& b, IOBuf& out, size_t n) { return b.cutn(&out, n); })
+      .def("cutn", [](IOBuf& b, size_t n) {
+        std::string s;
+        b.cutn(&s, n);
+        return py::bytes(s);
+      })
+      .def("pop_front", &IOBuf::pop_front)
+      .def("pop_back", &IOBuf::pop_back)
+      .def("copy_to",
+           [](const IOBuf& b, size_t n, size_t pos) {
+             std::string s;
+             b.copy_to(&s, n, pos);
+             return py::bytes(s);
+           },
+           py::arg("n") = (size_t)-1L, py::arg("pos") = 0)
+      .def("to_bytes", &iobuf_to_bytes)
+      .def("size", &IOBuf::size)
+      .def("empty", &IOBuf::empty)
+      .def("clear", &IOBuf::clear)
+      .def("backing_block_num", &IOBuf::backing_block_num)
+      .def("cpu_addressable", &IOBuf::cpu_addressable)
+      .def("hbm_bytes", &IOBuf::hbm_bytes)
+      .def("__len__", &IOBuf::size);
+
+  m.def("iobuf_block_count", &IOBuf::block_count);
+  m.def("iobuf_block_memory", &IOBuf::block_memory);
+  m.def("iobuf_flush_tls_cache", &bam::iobuf_flush_tls_cache);
+
+  m.def("crc32c", [](py::bytes data, uint32_t init) {
+    char* ptr;
+    Py_ssize_t len;
+    PyBytes_AsStringAndSize(data.ptr(), &ptr, &len);
+    return bam::crc32c::Extend(init, ptr, (size_t)len);
+  }, py::arg("data"), py::arg("init") = 0);
+  m.def("crc32c_combine", &bam::crc32c::Combine);
+  m.def("crc32c_hw", &bam::crc32c::IsFastCrc32Supported);
+
+  m.def("fast_rand", []() { return bam::fast_rand(); });
+
+  py::class_<bam::EndPoint>(m, "EndPoint")
+      .def(py::init<>())
+      .def_property_readonly("port", [](const bam::EndPoint& ep) { return ep.port; })
+      .def("__str__", [](const bam::EndPoint& ep) { return bam::endpoint2str(ep); });
+  m.def("str2endpoint", [](const std::string& s) {
+    bam::EndPoint ep;
+    if (bam::str2endpoint(s.c_str(), &ep) != 0) throw std::runtime_error("bad endpoint: " + s);
+    return ep;
+  });
+}

@@ -1,0 +1,9 @@
+#include "bindings/bind.h"
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "brpc_amd core: MI355X-native RPC runtime (bRPC capability rebuild)";
+  bind_base(m);
+  bind_fiber(m);
+  bind_rpc(m);
+  bind_var(m);
+}
