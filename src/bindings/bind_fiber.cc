@@ -1,0 +1,38 @@
+#include <pybind11/pybind11.h>
+
+#include "fiber/fiber.h"
+#include "bindings/bind.h"
+
+namespace bam {
+namespace selftest {
+int64_t start_join_test(int nfibers, int iters);
+bool urgent_test();
+int64_t usleep_test(int64_t us);
+bool butex_wake_test();
+bool butex_timeout_test();
+int64_t mutex_test(int nfibers, int iters);
+bool countdown_test(int n);
+bool timer_test();
+}  // namespace selftest
+}  // namespace bam
+
+void bind_fiber(py::module_& m) {
+  auto f = m.def_submodule("fiber");
+  // Release the GIL: these block on fiber completion.
+  f.def("start_join_test", &bam::selftest::start_join_test,
+        py::call_guard<py::gil_scoped_release>());
+  f.def("urgent_test", &bam::selftest::urgent_test, py::call_guard<py::gil_scoped_release>());
+  f.def("usleep_test", &bam::selftest::usleep_test, py::call_guard<py::gil_scoped_release>());
+  f.def("butex_wake_test", &bam::selftest::butex_wake_test,
+        py::call_guard<py::gil_scoped_release>());
+  f.def("butex_timeout_test", &bam::selftest::butex_timeout_test,
+        py::call_guard<py::gil_scoped_release>());
+  f.def("mutex_test", &bam::selftest::mutex_test, py::call_guard<py::gil_scoped_release>());
+  f.def("countdown_test", &bam::selftest::countdown_test,
+        py::call_guard<py::gil_scoped_release>());
+  f.def("timer_test", &bam::selftest::timer_test, py::call_guard<py::gil_scoped_release>());
+  f.def("concurrency", &bam::fiber_get_concurrency);
+  f.def("set_concurrency", &bam::fiber_set_concurrency);
+  f.def("count_created", &bam::fiber_count_created);
+  f.def("count_active", &bam::fiber_count_active);
+}

@@ -1,5 +1,4 @@
 #include "bindings/bind.h"
 // Filled in as the fiber / rpc / var layers land.
-void bind_fiber(py::module_& m) { (void)m; }
 void bind_rpc(py::module_& m) { (void)m; }
 void bind_var(py::module_& m) { (void)m; }

@@ -1,0 +1,45 @@
+// brpc_amd: public fiber API — M:N user-space threads.
+// Capability parity with reference bthread/bthread.h (start_urgent /
+// start_background / join / yield / usleep / self), rebuilt for the MI355X
+// runtime: the same scheduler that runs RPC fibers also integrates HIP
+// stream/event waits (a fiber blocking on a GPU op parks on a butex that a
+// hipEvent poller wakes — see hip/gpu_api.h + fiber/gpu_wait.h).
+#pragma once
+
+#include <stdint.h>
+
+namespace bam {
+
+typedef uint64_t fiber_t;  // versioned id; 0 = invalid
+
+struct FiberAttr {
+  uint32_t stack_size = 0;  // 0 = default (256 KiB)
+};
+
+// Starts a fiber. *tid receives its id. "urgent" runs the new fiber
+// immediately on the calling worker (the caller is re-queued) — the
+// latency trick used on the RPC dispatch path; "background" enqueues.
+int fiber_start_urgent(fiber_t* tid, void (*fn)(void*), void* arg,
+                       const FiberAttr* attr = nullptr);
+int fiber_start_background(fiber_t* tid, void (*fn)(void*), void* arg,
+                           const FiberAttr* attr = nullptr);
+
+// Waits for fiber termination. Returns 0; joining an ended/invalid id
+// returns 0 immediately.
+int fiber_join(fiber_t tid);
+// True if the fiber still runs (diagnostics).
+bool fiber_exists(fiber_t tid);
+
+int fiber_yield();
+int fiber_usleep(uint64_t us);
+fiber_t fiber_self();
+bool is_running_on_fiber();
+
+// Worker pool control.
+void fiber_set_concurrency(int n);  // only effective before first start
+int fiber_get_concurrency();
+// #fibers created / active (diagnostics, /status page).
+int64_t fiber_count_created();
+int64_t fiber_count_active();
+
+}  // namespace bam

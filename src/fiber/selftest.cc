@@ -1,0 +1,212 @@
+// C++-side fiber test scenarios, driven from pytest through the bindings.
+// (Fibers never execute Python code — the GIL and stack switching do not
+// mix — so the scenario bodies live here.)
+#include <errno.h>
+
+#include <atomic>
+#include <vector>
+
+#include "base/fast_rand.h"
+#include "base/time.h"
+#include "fiber/butex.h"
+#include "fiber/fiber.h"
+#include "fiber/sync.h"
+#include "fiber/timer_thread.h"
+
+namespace bam {
+namespace selftest {
+
+// ---- start/join ----
+namespace {
+struct AddArg {
+  std::atomic<int64_t>* counter;
+  int iters;
+};
+void add_fn(void* raw) {
+  AddArg* a = (AddArg*)raw;
+  for (int i = 0; i < a->iters; ++i) {
+    a->counter->fetch_add(1, std::memory_order_relaxed);
+    if ((i & 63) == 0) fiber_yield();
+  }
+}
+}  // namespace
+
+int64_t start_join_test(int nfibers, int iters) {
+  std::atomic<int64_t> counter{0};
+  AddArg arg{&counter, iters};
+  std::vector<fiber_t> tids(nfibers);
+  for (int i = 0; i < nfibers; ++i) {
+    if (fiber_start_background(&tids[i], add_fn, &arg) != 0) return -1;
+  }
+  for (int i = 0; i < nfibers; ++i) fiber_join(tids[i]);
+  return counter.load();
+}
+
+// ---- urgent start preempts ----
+namespace {
+std::atomic<int> g_order_token{0};
+void urgent_child(void*) { g_order_token.store(1, std::memory_order_release); }
+struct UrgentArg {
+  bool child_ran_first;
+};
+void urgent_parent(void* raw) {
+  UrgentArg* a = (UrgentArg*)raw;
+  g_order_token.store(0, std::memory_order_release);
+  fiber_t t;
+  fiber_start_urgent(&t, urgent_child, nullptr);
+  // With urgent start from a worker, the child runs before we resume.
+  a->child_ran_first = g_order_token.load(std::memory_order_acquire) == 1;
+  fiber_join(t);
+}
+}  // namespace
+
+bool urgent_test() {
+  UrgentArg a{false};
+  fiber_t t;
+  fiber_start_background(&t, urgent_parent, &a);
+  fiber_join(t);
+  return a.child_ran_first;
+}
+
+// ---- usleep accuracy ----
+namespace {
+struct SleepArg {
+  int64_t us;
+  int64_t measured;
+};
+void sleep_fn(void* raw) {
+  SleepArg* a = (SleepArg*)raw;
+  int64_t t0 = monotonic_time_us();
+  fiber_usleep(a->us);
+  a->measured = monotonic_time_us() - t0;
+}
+}  // namespace
+
+int64_t usleep_test(int64_t us) {
+  SleepArg a{us, 0};
+  fiber_t t;
+  fiber_start_background(&t, sleep_fn, &a);
+  fiber_join(t);
+  return a.measured;
+}
+
+// ---- butex wake/wait + timeout ----
+namespace {
+struct ButexArg {
+  std::atomic<int>* b;
+  int rc;
+  int saved_errno;
+  int64_t waited_us;
+  int64_t timeout_us;  // 0 = none
+};
+void butex_wait_fn(void* raw) {
+  ButexArg* a = (ButexArg*)raw;
+  int64_t t0 = monotonic_time_us();
+  if (a->timeout_us > 0) {
+    int64_t abst = t0 + a->timeout_us;
+    a->rc = butex_wait(a->b, 0, &abst);
+  } else {
+    a->rc = butex_wait(a->b, 0, nullptr);
+  }
+  a->saved_errno = errno;
+  a->waited_us = monotonic_time_us() - t0;
+}
+}  // namespace
+
+bool butex_wake_test() {
+  std::atomic<int>* b = butex_create();
+  b->store(0);
+  ButexArg a{b, -99, 0, 0, 0};
+  fiber_t t;
+  fiber_start_background(&t, butex_wait_fn, &a);
+  usleep(30000);  // let it park
+  b->store(1, std::memory_order_release);
+  butex_wake_all(b);
+  fiber_join(t);
+  butex_destroy(b);
+  return a.rc == 0 && a.waited_us >= 20000;
+}
+
+bool butex_timeout_test() {
+  std::atomic<int>* b = butex_create();
+  b->store(0);
+  ButexArg a{b, -99, 0, 0, 50000 /*50ms*/};
+  fiber_t t;
+  fiber_start_background(&t, butex_wait_fn, &a);
+  fiber_join(t);
+  bool ok = a.rc == -1 && a.saved_errno == ETIMEDOUT && a.waited_us >= 45000 &&
+            a.waited_us < 500000;
+  butex_destroy(b);
+  return ok;
+}
+
+// ---- mutex + condition stress ----
+namespace {
+struct MutexArg {
+  FiberMutex* mu;
+  int64_t* shared;  // unsynchronized; mutex must protect it
+  int iters;
+};
+void mutex_fn(void* raw) {
+  MutexArg* a = (MutexArg*)raw;
+  for (int i = 0; i < a->iters; ++i) {
+    FiberMutexGuard g(*a->mu);
+    int64_t v = *a->shared;
+    if ((i & 15) == 0) fiber_yield();  // force contention across workers
+    *a->shared = v + 1;
+  }
+}
+}  // namespace
+
+int64_t mutex_test(int nfibers, int iters) {
+  FiberMutex mu;
+  int64_t shared = 0;
+  MutexArg a{&mu, &shared, iters};
+  std::vector<fiber_t> tids(nfibers);
+  for (int i = 0; i < nfibers; ++i) fiber_start_background(&tids[i], mutex_fn, &a);
+  for (int i = 0; i < nfibers; ++i) fiber_join(tids[i]);
+  return shared;
+}
+
+// ---- countdown event ----
+namespace {
+struct CdArg {
+  CountdownEvent* ev;
+};
+void cd_fn(void* raw) {
+  CdArg* a = (CdArg*)raw;
+  fiber_usleep(1000 + fast_rand_less_than(5000));
+  a->ev->signal();
+}
+}  // namespace
+
+bool countdown_test(int n) {
+  CountdownEvent ev(n);
+  CdArg a{&ev};
+  for (int i = 0; i < n; ++i) {
+    fiber_t t;
+    fiber_start_background(&t, cd_fn, &a);
+  }
+  return ev.timed_wait(monotonic_time_us() + 2000000);
+}
+
+// ---- timer thread ----
+namespace {
+void timer_note(void* a, void*) { ((std::atomic<int>*)a)->fetch_add(1); }
+}  // namespace
+
+bool timer_test() {
+  std::atomic<int> fired{0};
+  int64_t now = monotonic_time_us();
+  TimerId t1 = timer_add(now + 20000, timer_note, &fired, nullptr);
+  TimerId t2 = timer_add(now + 30000, timer_note, &fired, nullptr);
+  (void)t1;
+  if (timer_delete(t2) != 0) return false;  // cancel before run
+  usleep(100000);
+  if (fired.load() != 1) return false;
+  if (timer_delete(t1) != -1) return false;  // already ran
+  return true;
+}
+
+}  // namespace selftest
+}  // namespace bam

@@ -1,0 +1,59 @@
+"""Fiber runtime tests (≙ reference bthread unittests).
+
+The scenario bodies are C++ (fibers must never run Python code); pytest
+drives them through the bindings and checks the results.
+"""
+import brpc_amd as b
+
+f = b.core.fiber
+
+
+def test_workers_started():
+    assert f.concurrency() >= 1
+
+
+def test_start_join_many():
+    assert f.start_join_test(100, 1000) == 100_000
+
+
+def test_start_join_single():
+    assert f.start_join_test(1, 1) == 1
+
+
+def test_urgent_start_preempts():
+    # start_urgent from a worker runs the child before the parent resumes
+    # (the reference's latency trick on the RPC dispatch path).
+    assert f.urgent_test()
+
+
+def test_usleep():
+    measured = f.usleep_test(20_000)
+    assert 18_000 <= measured < 1_000_000
+
+
+def test_butex_wake():
+    assert f.butex_wake_test()
+
+
+def test_butex_timeout():
+    assert f.butex_timeout_test()
+
+
+def test_mutex_mutual_exclusion():
+    # 16 fibers x 2000 increments of an unsynchronized counter under a
+    # FiberMutex with forced yields inside the critical section.
+    assert f.mutex_test(16, 2000) == 32_000
+
+
+def test_countdown_event():
+    assert f.countdown_test(50)
+
+
+def test_timer_add_delete():
+    assert f.timer_test()
+
+
+def test_counters_move():
+    created = f.count_created()
+    f.start_join_test(10, 1)
+    assert f.count_created() >= created + 10
