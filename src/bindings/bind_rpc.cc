@@ -1,0 +1,88 @@
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "bindings/bind.h"
+
+namespace bam {
+namespace rpctest {
+struct BenchResult {
+  double qps;
+  double mbps;
+  int64_t p50_us, p90_us, p99_us, p999_us, max_us, avg_us;
+  int64_t errors;
+  int64_t total;
+};
+int start_echo_server(int port);
+int echo_once(const std::string& addr, const std::string& payload, int timeout_ms,
+              std::string* response_out, int64_t* latency_us);
+int call_method_once(const std::string& addr, const std::string& method,
+                     const std::string& payload, int timeout_ms, int max_retry,
+                     std::string* response_out, std::string* error_text);
+bool attachment_test(const std::string& addr);
+BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
+                       int64_t total_calls, int timeout_ms);
+}  // namespace rpctest
+}  // namespace bam
+
+void bind_rpc(py::module_& m) {
+  auto r = m.def_submodule("rpc");
+  r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("echo_once",
+        [](const std::string& addr, py::bytes payload, int timeout_ms) {
+          char* ptr;
+          Py_ssize_t len;
+          PyBytes_AsStringAndSize(payload.ptr(), &ptr, &len);
+          std::string p(ptr, len), resp;
+          int64_t lat = 0;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::echo_once(addr, p, timeout_ms, &resp, &lat);
+          }
+          return py::make_tuple(rc, py::bytes(resp), lat);
+        },
+        py::arg("addr"), py::arg("payload"), py::arg("timeout_ms") = 1000);
+  r.def("call_method_once",
+        [](const std::string& addr, const std::string& method, py::bytes payload,
+           int timeout_ms, int max_retry) {
+          char* ptr;
+          Py_ssize_t len;
+          PyBytes_AsStringAndSize(payload.ptr(), &ptr, &len);
+          std::string p(ptr, len), resp, err;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::call_method_once(addr, method, p, timeout_ms, max_retry,
+                                                &resp, &err);
+          }
+          return py::make_tuple(rc, py::bytes(resp), err);
+        },
+        py::arg("addr"), py::arg("method"), py::arg("payload"), py::arg("timeout_ms") = 1000,
+        py::arg("max_retry") = 3);
+  r.def("attachment_test", &bam::rpctest::attachment_test,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("echo_bench",
+        [](const std::string& addr, int payload_size, int concurrency, int64_t total,
+           int timeout_ms) {
+          bam::rpctest::BenchResult b;
+          {
+            py::gil_scoped_release rel;
+            b = bam::rpctest::echo_bench(addr, payload_size, concurrency, total, timeout_ms);
+          }
+          py::dict d;
+          d["qps"] = b.qps;
+          d["mbps"] = b.mbps;
+          d["p50_us"] = b.p50_us;
+          d["p90_us"] = b.p90_us;
+          d["p99_us"] = b.p99_us;
+          d["p999_us"] = b.p999_us;
+          d["max_us"] = b.max_us;
+          d["avg_us"] = b.avg_us;
+          d["errors"] = b.errors;
+          d["total"] = b.total;
+          return d;
+        },
+        py::arg("addr"), py::arg("payload_size") = 64, py::arg("concurrency") = 8,
+        py::arg("total") = 10000, py::arg("timeout_ms") = 5000);
+}

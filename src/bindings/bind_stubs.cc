@@ -1,4 +1,3 @@
 #include "bindings/bind.h"
 // Filled in as the fiber / rpc / var layers land.
-void bind_rpc(py::module_& m) { (void)m; }
 void bind_var(py::module_& m) { (void)m; }

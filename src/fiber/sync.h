@@ -4,6 +4,8 @@
 // futex for plain pthreads), never blocks the worker.
 #pragma once
 
+#include <cerrno>
+
 #include "base/time.h"
 #include "fiber/butex.h"
 

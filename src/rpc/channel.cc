@@ -1,0 +1,165 @@
+#include "rpc/channel.h"
+
+#include <string.h>
+
+#include "base/logging.h"
+#include "base/time.h"
+#include "rpc/load_balancer.h"
+#include "rpc/policy/std_protocol.h"
+#include "rpc/socket_map.h"
+
+namespace bam {
+
+Channel::~Channel() {}
+
+int Channel::Init(EndPoint ep, const ChannelOptions* options) {
+  if (options != nullptr) options_ = *options;
+  server_ep_ = ep;
+  single_server_ = true;
+  policy::RegisterStdProtocol();
+  return 0;
+}
+
+int Channel::Init(const char* server_addr, const ChannelOptions* options) {
+  EndPoint ep;
+  if (str2endpoint(server_addr, &ep) != 0) {
+    // maybe a naming URL used with the 2-arg Init by mistake
+    return -1;
+  }
+  return Init(ep, options);
+}
+
+int Channel::Init(const char* naming_url, const char* lb_name, const ChannelOptions* options) {
+  if (options != nullptr) options_ = *options;
+  single_server_ = false;
+  policy::RegisterStdProtocol();
+  lb_ = LoadBalancerWithNaming::Create(naming_url, lb_name);
+  if (lb_ == nullptr) return -1;
+  return 0;
+}
+
+// ---------------- call machinery ----------------
+
+static void split_full_method(const std::string& full, std::string* service,
+                              std::string* method) {
+  size_t pos = full.find_last_of("./");
+  if (pos == std::string::npos) {
+    *service = "";
+    *method = full;
+  } else {
+    *service = full.substr(0, pos);
+    *method = full.substr(pos + 1);
+  }
+}
+
+// Completes the RPC: cancels the timeout timer, destroys the (locked)
+// session and runs done.
+void EndRPC(Controller* cntl, SessionId locked_id) {
+  cntl->end_us_ = monotonic_time_us();
+  if (cntl->call.timeout_timer != 0) {
+    timer_delete(cntl->call.timeout_timer);
+    cntl->call.timeout_timer = 0;
+  }
+  if (cntl->call.lb != nullptr) {
+    cntl->call.lb->Feedback(cntl->call.server_ep, cntl->error_code_,
+                            cntl->end_us_ - cntl->start_us_);
+  }
+  Closure* done = cntl->call.done;
+  session_unlock_and_destroy(locked_id);
+  if (done != nullptr) done->Run();
+}
+
+// on_error handler: runs LOCKED; must unlock or destroy.
+static int HandleSessionError(SessionId id, void* data, int error_code) {
+  Controller* cntl = (Controller*)data;
+  if (!session_is_current(id)) {
+    // failure of a stale attempt (e.g. old socket died after retry)
+    session_unlock(id);
+    return 0;
+  }
+  if (error_code != ERPCTIMEDOUT && cntl->retry_count_ < cntl->max_retry_) {
+    ++cntl->retry_count_;
+    session_bump_slot(id);
+    if (cntl->call.lb != nullptr && cntl->call.server_ep.port != 0) {
+      cntl->call.lb->Feedback(cntl->call.server_ep, error_code, 0);
+    }
+    IssueRPC(cntl);  // synchronous failures enqueue on the session (no deadlock)
+    session_unlock(id);
+    return 0;
+  }
+  cntl->SetFailed(error_code, error_code == ERPCTIMEDOUT
+                                  ? "RPC deadline exceeded"
+                                  : std::string("socket error: ") + strerror(error_code));
+  EndRPC(cntl, id);
+  return 0;
+}
+
+static void TimeoutCb(void* a, void* /*b*/) {
+  SessionId cid = (SessionId)(uintptr_t)a;
+  session_error(cid, ERPCTIMEDOUT);
+}
+
+void IssueRPC(Controller* cntl) {
+  // Select a server.
+  EndPoint ep = cntl->call.server_ep;
+  if (cntl->call.lb != nullptr) {
+    if (cntl->call.lb->SelectServer(&ep) != 0) {
+      session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
+      return;
+    }
+    cntl->call.server_ep = ep;
+  }
+  SocketUniquePtr sock;
+  if (GetClientSocket(ep, &sock) != 0) {
+    // Conduct the failure through the session so retry/ending logic runs.
+    session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
+    return;
+  }
+  cntl->remote_side_ = ep;
+  SessionId current = session_current_id(cntl->call.cid);
+  IOBuf packet;
+  policy::PackStdRequest(&packet, cntl, current);
+  sock->add_pending_session(current);
+  cntl->call.pending_socket = sock->id();
+  Socket::WriteOptions wo;
+  wo.id_wait = current;
+  sock->Write(&packet, &wo);  // failure conducts through the session
+}
+
+void Channel::CallMethod(const std::string& full_method, Controller* cntl,
+                         const IOBuf* request, IOBuf* response, Closure* done) {
+  cntl->start_us_ = monotonic_time_us();
+  if (cntl->timeout_ms_ == -1) cntl->timeout_ms_ = options_.timeout_ms;
+  if (cntl->max_retry_ == 3 /*default*/) cntl->max_retry_ = options_.max_retry;
+  split_full_method(full_method, &cntl->call.service_name, &cntl->call.method_name);
+  if (request != nullptr) cntl->call.request_buf = *request;  // zero-copy ref share
+  cntl->call.response = response;
+  cntl->call.done = done;
+  if (single_server_) {
+    cntl->call.server_ep = server_ep_;
+    cntl->call.lb = nullptr;
+  } else {
+    cntl->call.lb = lb_.get();
+  }
+
+  SessionId cid;
+  int rc = session_create(&cid, cntl, HandleSessionError, cntl->max_retry_ + 1);
+  if (rc != 0) {
+    cntl->SetFailed(EINTERNAL, "session_create failed");
+    if (done) done->Run();
+    return;
+  }
+  cntl->cid_ = cid;
+  cntl->call.cid = cid;
+
+  if (cntl->timeout_ms_ > 0) {
+    cntl->call.timeout_timer = timer_add(cntl->start_us_ + cntl->timeout_ms_ * 1000,
+                                         TimeoutCb, (void*)(uintptr_t)cid, nullptr);
+  }
+  IssueRPC(cntl);
+  if (done == nullptr) {
+    session_join(cid);
+  }
+}
+
+}  // namespace bam

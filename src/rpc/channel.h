@@ -1,0 +1,62 @@
+// brpc_amd: Channel — the client-side call endpoint.
+// API parity: reference brpc/channel.h (Init with "ip:port" or naming URL +
+// load-balancer name, ChannelOptions{timeout_ms, connect_timeout_ms,
+// max_retry, protocol}, CallMethod). The combo channels (Parallel /
+// Selective / Partition) layer on the same CallMethod contract.
+#pragma once
+
+#include <memory>
+#include <string>
+
+#include "rpc/controller.h"
+
+namespace bam {
+
+class LoadBalancerWithNaming;
+
+struct ChannelOptions {
+  int32_t connect_timeout_ms = 200;
+  int32_t timeout_ms = 500;
+  int32_t backup_request_ms = -1;
+  int max_retry = 3;
+  std::string protocol = "std";
+  std::string connection_group;
+};
+
+class ChannelBase {
+ public:
+  virtual ~ChannelBase() {}
+  // full_method = "ServiceName.MethodName" (or "ServiceName/Method").
+  virtual void CallMethod(const std::string& full_method, Controller* cntl,
+                          const IOBuf* request, IOBuf* response, Closure* done) = 0;
+};
+
+class Channel : public ChannelBase {
+ public:
+  Channel() {}
+  ~Channel() override;
+
+  // "ip:port" / "host:port" single server, or "list://h1:p1,h2:p2" /
+  // "file://path" with a load balancer name ("rr", "random", "c_hash",
+  // "la", "p2c").
+  int Init(const char* server_addr, const ChannelOptions* options);
+  int Init(const char* naming_url, const char* lb_name, const ChannelOptions* options);
+  int Init(EndPoint ep, const ChannelOptions* options);
+
+  void CallMethod(const std::string& full_method, Controller* cntl, const IOBuf* request,
+                  IOBuf* response, Closure* done) override;
+
+  const ChannelOptions& options() const { return options_; }
+
+ private:
+  ChannelOptions options_;
+  EndPoint server_ep_;
+  bool single_server_ = false;
+  std::shared_ptr<LoadBalancerWithNaming> lb_;
+};
+
+// Starts the RPC described by cntl->call (already filled). Used by Channel
+// and by the retry path in HandleSessionError.
+void IssueRPC(Controller* cntl);
+
+}  // namespace bam

@@ -1,0 +1,116 @@
+// brpc_amd: Controller — per-RPC context for both client and server side.
+// API parity: reference brpc/controller.h (set_timeout_ms / set_max_retry /
+// request_attachment / response_attachment / Failed / ErrorCode/ErrorText /
+// call_id / log_id / remote_side), payload-centric: request/response bodies
+// are IOBufs (opaque bytes or serialized messages); attachments bypass
+// serialization exactly like baidu_std's attachment field.
+#pragma once
+
+#include <stdint.h>
+
+#include <string>
+
+#include "base/endpoint.h"
+#include "base/iobuf.h"
+#include "fiber/session.h"
+#include "fiber/timer_thread.h"
+#include "rpc/closure.h"
+#include "rpc/rpc_errno.h"
+#include "rpc/socket.h"
+
+namespace bam {
+
+enum CompressType {
+  COMPRESS_TYPE_NONE = 0,
+  COMPRESS_TYPE_SNAPPY = 1,
+  COMPRESS_TYPE_GZIP = 2,
+};
+
+class Channel;
+class Server;
+
+class Controller {
+ public:
+  Controller() { Reset(); }
+  ~Controller();
+
+  void Reset();
+
+  // ---- client-side knobs ----
+  void set_timeout_ms(int64_t ms) { timeout_ms_ = ms; }
+  int64_t timeout_ms() const { return timeout_ms_; }
+  void set_max_retry(int n) { max_retry_ = n; }
+  int max_retry() const { return max_retry_; }
+  void set_backup_request_ms(int64_t ms) { backup_request_ms_ = ms; }
+  int64_t backup_request_ms() const { return backup_request_ms_; }
+  void set_log_id(uint64_t id) { log_id_ = id; }
+  uint64_t log_id() const { return log_id_; }
+  void set_request_compress_type(CompressType t) { request_compress_ = t; }
+  CompressType request_compress_type() const { return request_compress_; }
+
+  IOBuf& request_attachment() { return request_attachment_; }
+  IOBuf& response_attachment() { return response_attachment_; }
+
+  // ---- status ----
+  bool Failed() const { return error_code_ != 0; }
+  int ErrorCode() const { return error_code_; }
+  const std::string& ErrorText() const { return error_text_; }
+  void SetFailed(int code, const std::string& reason) {
+    error_code_ = code != 0 ? code : EINTERNAL;
+    error_text_ = reason;
+  }
+  void SetFailed(const std::string& reason) { SetFailed(EINTERNAL, reason); }
+
+  SessionId call_id() const { return cid_; }
+  int64_t latency_us() const { return end_us_ - start_us_; }
+  int retried_count() const { return retry_count_; }
+
+  EndPoint remote_side() const { return remote_side_; }
+  EndPoint local_side() const { return local_side_; }
+
+  bool is_server_side() const { return server_ != nullptr; }
+
+  // ---- internals (channel / protocol / server plumbing) ----
+  struct Call {
+    SessionId cid = 0;              // whole-call session
+    IOBuf request_buf;              // packed user payload (pre-header)
+    IOBuf* response = nullptr;      // user's output buffer
+    Closure* done = nullptr;        // nullptr = synchronous
+    std::string service_name;
+    std::string method_name;
+    EndPoint server_ep;
+    TimerId timeout_timer = 0;
+    SocketId pending_socket = 0;    // socket holding the pending session
+    class LoadBalancer* lb = nullptr;
+    class SubChannelCtx* sub_ctx = nullptr;  // ParallelChannel bookkeeping
+  };
+  Call call;
+
+  // server-side context
+  Server* server_ = nullptr;
+  SocketId server_socket_ = 0;
+  int64_t server_cid_ = 0;  // correlation id to echo back
+  CompressType response_compress_ = COMPRESS_TYPE_NONE;
+  void set_response_compress_type(CompressType t) { response_compress_ = t; }
+
+  int64_t start_us_ = 0;
+  int64_t end_us_ = 0;
+  int error_code_ = 0;
+  std::string error_text_;
+  int64_t timeout_ms_ = -1;
+  int64_t backup_request_ms_ = -1;
+  int max_retry_ = 3;
+  int retry_count_ = 0;
+  uint64_t log_id_ = 0;
+  CompressType request_compress_ = COMPRESS_TYPE_NONE;
+  IOBuf request_attachment_;
+  IOBuf response_attachment_;
+  EndPoint remote_side_;
+  EndPoint local_side_;
+  SessionId cid_ = 0;
+};
+
+// Waits until the RPC identified by `id` completes (client-side sync call).
+inline int Join(SessionId id) { return session_join(id); }
+
+}  // namespace bam

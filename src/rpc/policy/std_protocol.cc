@@ -1,0 +1,292 @@
+#include "rpc/policy/std_protocol.h"
+
+#include <mutex>
+
+#include "base/logging.h"
+#include "base/time.h"
+#include "rpc/channel.h"
+#include "rpc/controller.h"
+#include "rpc/server.h"
+#include "rpc/wire.h"
+
+namespace bam {
+
+void EndRPC(Controller* cntl, SessionId locked_id);  // channel.cc
+
+namespace policy {
+
+static const char kMagic[4] = {'P', 'R', 'P', 'C'};
+static const size_t kHeaderLen = 12;
+
+// ---------------- meta codec ----------------
+
+void SerializeRpcMeta(const RpcMeta& meta, std::string* out) {
+  if (meta.has_request) {
+    std::string sub;
+    wire::put_str_field(&sub, 1, meta.service_name);
+    wire::put_str_field(&sub, 2, meta.method_name);
+    if (meta.log_id != 0) wire::put_int_field(&sub, 3, (int64_t)meta.log_id);
+    wire::put_msg_field(out, 1, sub);
+  }
+  if (meta.has_response) {
+    std::string sub;
+    if (meta.error_code != 0) wire::put_int_field(&sub, 1, meta.error_code);
+    if (!meta.error_text.empty()) wire::put_str_field(&sub, 2, meta.error_text);
+    wire::put_msg_field(out, 2, sub);
+  }
+  if (meta.compress_type != 0) wire::put_int_field(out, 3, meta.compress_type);
+  wire::put_int_field(out, 4, meta.correlation_id);
+  if (meta.attachment_size != 0) wire::put_int_field(out, 5, meta.attachment_size);
+}
+
+bool ParseRpcMeta(const char* data, size_t n, RpcMeta* out) {
+  wire::Reader r(data, n);
+  int wtype;
+  for (int field; (field = r.read_tag(&wtype)) != 0;) {
+    switch (field) {
+      case 1: {  // RpcRequestMeta
+        std::string sub = r.read_string();
+        if (!r.ok()) return false;
+        out->has_request = true;
+        wire::Reader rr(sub.data(), sub.size());
+        int wt2;
+        for (int f2; (f2 = rr.read_tag(&wt2)) != 0;) {
+          if (f2 == 1) out->service_name = rr.read_string();
+          else if (f2 == 2) out->method_name = rr.read_string();
+          else if (f2 == 3) out->log_id = rr.varint();
+          else rr.skip(wt2);
+          if (!rr.ok()) return false;
+        }
+        break;
+      }
+      case 2: {  // RpcResponseMeta
+        std::string sub = r.read_string();
+        if (!r.ok()) return false;
+        out->has_response = true;
+        wire::Reader rr(sub.data(), sub.size());
+        int wt2;
+        for (int f2; (f2 = rr.read_tag(&wt2)) != 0;) {
+          if (f2 == 1) out->error_code = (int)rr.varint();
+          else if (f2 == 2) out->error_text = rr.read_string();
+          else rr.skip(wt2);
+          if (!rr.ok()) return false;
+        }
+        break;
+      }
+      case 3:
+        out->compress_type = (int)r.varint();
+        break;
+      case 4:
+        out->correlation_id = (int64_t)r.varint();
+        break;
+      case 5:
+        out->attachment_size = (int32_t)r.varint();
+        break;
+      default:
+        r.skip(wtype);
+    }
+    if (!r.ok()) return false;
+  }
+  return r.ok();
+}
+
+// ---------------- framing ----------------
+
+struct StdMessage : public InputMessageBase {
+  RpcMeta meta;
+  IOBuf payload;  // user data (+ attachment at tail)
+};
+
+static ParseResult ParseStdMessage(IOBuf* source, Socket* /*sock*/, bool /*read_eof*/) {
+  char aux[kHeaderLen];
+  if (source->size() < kHeaderLen) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  const char* h = (const char*)source->fetch(aux, kHeaderLen);
+  if (h == nullptr || memcmp(h, kMagic, 4) != 0)
+    return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  uint32_t body_size = wire::get_u32_be(h + 4);
+  uint32_t meta_size = wire::get_u32_be(h + 8);
+  if (meta_size > body_size || body_size > (256u << 20))
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  if (source->size() < kHeaderLen + body_size)
+    return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  source->pop_front(kHeaderLen);
+  std::string meta_bytes;
+  source->cutn(&meta_bytes, meta_size);
+  StdMessage* msg = new StdMessage;
+  if (!ParseRpcMeta(meta_bytes.data(), meta_bytes.size(), &msg->meta)) {
+    delete msg;
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  }
+  source->cutn(&msg->payload, body_size - meta_size);
+  return ParseResult::make_ok(msg);
+}
+
+void PackStdRequest(IOBuf* out, Controller* cntl, SessionId correlation_id) {
+  RpcMeta meta;
+  meta.has_request = true;
+  meta.service_name = cntl->call.service_name;
+  meta.method_name = cntl->call.method_name;
+  meta.log_id = cntl->log_id();
+  meta.compress_type = (int)cntl->request_compress_type();
+  meta.correlation_id = (int64_t)correlation_id;
+  meta.attachment_size = (int32_t)cntl->request_attachment().size();
+  std::string meta_bytes;
+  SerializeRpcMeta(meta, &meta_bytes);
+  size_t body = meta_bytes.size() + cntl->call.request_buf.size() +
+                cntl->request_attachment().size();
+  char header[kHeaderLen];
+  memcpy(header, kMagic, 4);
+  wire::put_u32_be(header + 4, (uint32_t)body);
+  wire::put_u32_be(header + 8, (uint32_t)meta_bytes.size());
+  out->append(header, kHeaderLen);
+  out->append(meta_bytes);
+  out->append(cntl->call.request_buf);       // zero-copy ref share
+  out->append(cntl->request_attachment());   // zero-copy ref share
+}
+
+static void PackStdResponse(IOBuf* out, int64_t correlation_id, int error_code,
+                            const std::string& error_text, const IOBuf& payload,
+                            const IOBuf& attachment) {
+  RpcMeta meta;
+  meta.has_response = true;
+  meta.error_code = error_code;
+  meta.error_text = error_text;
+  meta.correlation_id = correlation_id;
+  meta.attachment_size = (int32_t)attachment.size();
+  std::string meta_bytes;
+  SerializeRpcMeta(meta, &meta_bytes);
+  size_t body = meta_bytes.size() + payload.size() + attachment.size();
+  char header[kHeaderLen];
+  memcpy(header, kMagic, 4);
+  wire::put_u32_be(header + 4, (uint32_t)body);
+  wire::put_u32_be(header + 8, (uint32_t)meta_bytes.size());
+  out->append(header, kHeaderLen);
+  out->append(meta_bytes);
+  out->append(payload);
+  out->append(attachment);
+}
+
+// ---------------- server side ----------------
+
+static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* resp) {
+  SocketUniquePtr sock;
+  if (Socket::Address(sid, &sock) == 0) {
+    IOBuf packet;
+    PackStdResponse(&packet, cid, cntl->ErrorCode(), cntl->ErrorText(), *resp,
+                    cntl->response_attachment());
+    sock->Write(&packet);
+  }
+  if (cntl->server_ != nullptr) {
+    cntl->server_->concurrency.fetch_sub(1, std::memory_order_relaxed);
+    cntl->server_->nprocessed.fetch_add(1, std::memory_order_relaxed);
+  }
+  delete resp;
+  delete cntl;
+}
+
+static void ProcessStdRequest(InputMessageBase* msg_base) {
+  StdMessage* msg = (StdMessage*)msg_base;
+  SocketUniquePtr sock;
+  if (Socket::Address(msg->socket_id, &sock) != 0) {
+    delete msg;
+    return;
+  }
+  Server* server = (Server*)sock->user();
+  Controller* cntl = new Controller;
+  cntl->server_ = server;
+  cntl->server_socket_ = sock->id();
+  cntl->server_cid_ = msg->meta.correlation_id;
+  cntl->log_id_ = msg->meta.log_id;
+  cntl->remote_side_ = sock->remote_side();
+  cntl->local_side_ = sock->local_side();
+  IOBuf* resp = new IOBuf;
+  SocketId sid = sock->id();
+  int64_t cid = msg->meta.correlation_id;
+  Closure* done = NewCallback([sid, cid, cntl, resp] { SendStdResponse(sid, cid, cntl, resp); });
+
+  // attachment split
+  IOBuf req_data;
+  size_t att = (size_t)msg->meta.attachment_size;
+  size_t data_len = msg->payload.size() >= att ? msg->payload.size() - att : 0;
+  msg->payload.cutn(&req_data, data_len);
+  cntl->request_attachment().swap(msg->payload);
+
+  Service* svc = nullptr;
+  const MethodFn* fn =
+      server != nullptr
+          ? server->FindMethod(msg->meta.service_name, msg->meta.method_name, &svc)
+          : nullptr;
+  if (fn == nullptr) {
+    cntl->SetFailed(msg->meta.service_name.empty() || svc == nullptr ? ENOSERVICE : ENOMETHOD,
+                    "unknown service/method " + msg->meta.service_name + "." +
+                        msg->meta.method_name);
+    delete msg;
+    done->Run();
+    return;
+  }
+  if (server->max_concurrency() > 0 &&
+      server->concurrency.load(std::memory_order_relaxed) >= server->max_concurrency()) {
+    cntl->SetFailed(ELIMIT, "reached server max_concurrency");
+    delete msg;
+    done->Run();
+    return;
+  }
+  server->concurrency.fetch_add(1, std::memory_order_relaxed);
+  (*fn)(cntl, req_data, resp, done);
+  delete msg;
+}
+
+// ---------------- client side ----------------
+
+static void ProcessStdResponse(InputMessageBase* msg_base) {
+  StdMessage* msg = (StdMessage*)msg_base;
+  SessionId cid = (SessionId)msg->meta.correlation_id;
+  void* data = nullptr;
+  if (session_lock(cid, &data) != 0) {
+    delete msg;  // late/duplicate response
+    return;
+  }
+  Controller* cntl = (Controller*)data;
+  if (!session_is_current(cid)) {
+    session_unlock(cid);  // response for a stale retry attempt
+    delete msg;
+    return;
+  }
+  {
+    SocketUniquePtr sock;
+    if (Socket::Address(msg->socket_id, &sock) == 0) sock->remove_pending_session(cid);
+  }
+  if (msg->meta.error_code != 0) {
+    cntl->SetFailed(msg->meta.error_code, msg->meta.error_text);
+  } else {
+    size_t att = (size_t)msg->meta.attachment_size;
+    size_t data_len = msg->payload.size() >= att ? msg->payload.size() - att : 0;
+    if (cntl->call.response != nullptr) {
+      cntl->call.response->clear();
+      msg->payload.cutn(cntl->call.response, data_len);
+    } else {
+      msg->payload.pop_front(data_len);
+    }
+    cntl->response_attachment().clear();
+    cntl->response_attachment().swap(msg->payload);
+  }
+  delete msg;
+  EndRPC(cntl, cid);
+}
+
+void RegisterStdProtocol() {
+  static std::once_flag flag;
+  std::call_once(flag, [] {
+    Protocol p;
+    p.parse = ParseStdMessage;
+    p.process_request = ProcessStdRequest;
+    p.process_response = ProcessStdResponse;
+    p.support_server = true;
+    p.support_client = true;
+    p.name = "std";
+    RegisterProtocol(p);
+  });
+}
+
+}  // namespace policy
+}  // namespace bam

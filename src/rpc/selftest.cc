@@ -1,0 +1,188 @@
+// C++ RPC test scenarios + benchmark loops, driven from pytest / bench.py.
+// Models the reference's single-process loopback fixtures
+// (test/brpc_channel_unittest.cpp ChannelTest): server and client in one
+// process over 127.0.0.1.
+#include <algorithm>
+#include <atomic>
+#include <memory>
+#include <vector>
+
+#include "base/fast_rand.h"
+#include "base/time.h"
+#include "fiber/fiber.h"
+#include "fiber/sync.h"
+#include "rpc/channel.h"
+#include "rpc/controller.h"
+#include "rpc/server.h"
+
+namespace bam {
+namespace rpctest {
+
+// ---- the universal fixture: EchoService (parity: test/echo.proto) ----
+
+Service* NewEchoService() {
+  Service* svc = new Service("EchoService");
+  svc->AddMethod("Echo", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    resp->append(req);
+    cntl->response_attachment().append(cntl->request_attachment());
+    done->Run();
+  });
+  svc->AddMethod("Sleep", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    char buf[16] = {0};
+    req.copy_to(buf, sizeof(buf) - 1, 0);
+    int ms = atoi(buf);
+    fiber_usleep((uint64_t)ms * 1000);
+    resp->append("slept");
+    done->Run();
+  });
+  svc->AddMethod("Fail", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    cntl->SetFailed(EINTERNAL, "you asked for it");
+    done->Run();
+  });
+  return svc;
+}
+
+// Starts an echo server on 127.0.0.1:port (0 = auto). Returns port or -1.
+// Servers leak by design in tests (process-lifetime).
+int start_echo_server(int port) {
+  Server* server = new Server;
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  if (server->Start(port, nullptr) != 0) return -1;
+  return server->listen_address().port;
+}
+
+// One sync echo; returns 0 on success and fills latency_us, else error code.
+int echo_once(const std::string& addr, const std::string& payload, int timeout_ms,
+              std::string* response_out, int64_t* latency_us) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = timeout_ms;
+  if (channel.Init(addr.c_str(), &opts) != 0) return -1;
+  Controller cntl;
+  IOBuf request, response;
+  request.append(payload);
+  channel.CallMethod("EchoService.Echo", &cntl, &request, &response, nullptr);
+  if (latency_us != nullptr) *latency_us = cntl.latency_us();
+  if (cntl.Failed()) return cntl.ErrorCode();
+  if (response_out != nullptr) *response_out = response.to_string();
+  return 0;
+}
+
+int call_method_once(const std::string& addr, const std::string& method,
+                     const std::string& payload, int timeout_ms, int max_retry,
+                     std::string* response_out, std::string* error_text) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = timeout_ms;
+  opts.max_retry = max_retry;
+  if (channel.Init(addr.c_str(), &opts) != 0) return -1;
+  Controller cntl;
+  cntl.set_max_retry(max_retry);
+  IOBuf request, response;
+  request.append(payload);
+  channel.CallMethod(method, &cntl, &request, &response, nullptr);
+  if (error_text != nullptr) *error_text = cntl.ErrorText();
+  if (cntl.Failed()) return cntl.ErrorCode();
+  if (response_out != nullptr) *response_out = response.to_string();
+  return 0;
+}
+
+// Attachment round-trip.
+bool attachment_test(const std::string& addr) {
+  Channel channel;
+  if (channel.Init(addr.c_str(), nullptr) != 0) return false;
+  Controller cntl;
+  IOBuf request, response;
+  request.append("body");
+  std::string att(10000, 'A');
+  cntl.request_attachment().append(att);
+  channel.CallMethod("EchoService.Echo", &cntl, &request, &response, nullptr);
+  if (cntl.Failed()) return false;
+  return response.to_string() == "body" && cntl.response_attachment().to_string() == att;
+}
+
+// ---- benchmark: concurrent echo QPS + latency percentiles ----
+
+struct BenchResult {
+  double qps = 0;
+  double mbps = 0;  // payload throughput (both directions counted once)
+  int64_t p50_us = 0, p90_us = 0, p99_us = 0, p999_us = 0, max_us = 0, avg_us = 0;
+  int64_t errors = 0;
+  int64_t total = 0;
+};
+
+namespace {
+
+struct BenchWorkerArg {
+  Channel* channel;
+  std::atomic<int64_t>* remaining;
+  std::vector<int64_t>* latencies;  // pre-sized; indexed by call #
+  std::atomic<int64_t>* errors;
+  std::string payload;
+  CountdownEvent* done_event;
+};
+
+void bench_worker(void* raw) {
+  BenchWorkerArg* a = (BenchWorkerArg*)raw;
+  for (;;) {
+    int64_t idx = a->remaining->fetch_sub(1, std::memory_order_relaxed);
+    if (idx <= 0) break;
+    Controller cntl;
+    IOBuf request, response;
+    request.append(a->payload);
+    a->channel->CallMethod("EchoService.Echo", &cntl, &request, &response, nullptr);
+    if (cntl.Failed() || response.size() != a->payload.size()) {
+      a->errors->fetch_add(1, std::memory_order_relaxed);
+    }
+    (*a->latencies)[idx - 1] = cntl.latency_us();
+  }
+  a->done_event->signal();
+}
+
+}  // namespace
+
+BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
+                       int64_t total_calls, int timeout_ms) {
+  BenchResult res;
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = timeout_ms;
+  if (channel.Init(addr.c_str(), &opts) != 0) return res;
+  std::string payload(payload_size, 'x');
+  for (int i = 0; i < payload_size; ++i) payload[i] = (char)fast_rand();
+  std::atomic<int64_t> remaining{total_calls};
+  std::atomic<int64_t> errors{0};
+  std::vector<int64_t> latencies(total_calls, 0);
+  CountdownEvent done_event(concurrency);
+  BenchWorkerArg arg{&channel, &remaining, &latencies, &errors, payload, &done_event};
+  int64_t t0 = monotonic_time_us();
+  for (int i = 0; i < concurrency; ++i) {
+    fiber_t th;
+    fiber_start_background(&th, bench_worker, &arg);
+  }
+  done_event.wait();
+  int64_t elapsed = monotonic_time_us() - t0;
+  std::sort(latencies.begin(), latencies.end());
+  res.total = total_calls;
+  res.errors = errors.load();
+  res.qps = total_calls * 1e6 / (double)elapsed;
+  res.mbps = res.qps * payload_size / 1e6;
+  int64_t sum = 0;
+  for (int64_t v : latencies) sum += v;
+  res.avg_us = total_calls > 0 ? sum / total_calls : 0;
+  auto pct = [&](double p) {
+    size_t i = (size_t)(p * (latencies.size() - 1));
+    return latencies[i];
+  };
+  if (!latencies.empty()) {
+    res.p50_us = pct(0.50);
+    res.p90_us = pct(0.90);
+    res.p99_us = pct(0.99);
+    res.p999_us = pct(0.999);
+    res.max_us = latencies.back();
+  }
+  return res;
+}
+
+}  // namespace rpctest
+}  // namespace bam

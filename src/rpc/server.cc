@@ -1,0 +1,109 @@
+#include "rpc/server.h"
+
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include "base/logging.h"
+#include "rpc/event_dispatcher.h"
+#include "rpc/policy/std_protocol.h"
+
+namespace bam {
+
+Server::Server() : messenger_(/*server_side=*/true) {}
+
+Server::~Server() {
+  Stop(0);
+  for (auto& kv : ownership_) {
+    if (kv.second == SERVER_OWNS_SERVICE) delete kv.first;
+  }
+}
+
+int Server::AddService(Service* service, ServiceOwnership ownership) {
+  if (service == nullptr || IsRunning()) return -1;
+  if (services_.count(service->name()) != 0) return -1;
+  services_[service->name()] = service;
+  ownership_[service] = ownership;
+  return 0;
+}
+
+const MethodFn* Server::FindMethod(const std::string& service, const std::string& method,
+                                   Service** svc_out) const {
+  auto it = services_.find(service);
+  if (it == services_.end()) {
+    // Single-service convenience: empty service name matches the only one.
+    if (service.empty() && services_.size() == 1) it = services_.begin();
+    else return nullptr;
+  }
+  if (svc_out != nullptr) *svc_out = it->second;
+  return it->second->FindMethod(method);
+}
+
+void Server::OnNewConnections(Socket* listen_socket) {
+  Server* server = (Server*)listen_socket->user();
+  for (;;) {
+    int fd = accept4(listen_socket->fd(), nullptr, nullptr, SOCK_NONBLOCK | SOCK_CLOEXEC);
+    if (fd < 0) {
+      if (errno == EAGAIN || errno == EWOULDBLOCK) return;
+      if (errno == EINTR) continue;
+      if (errno == EMFILE || errno == ENFILE) {
+        LOG(WARNING) << "accept failed: out of fds";
+        return;
+      }
+      return;
+    }
+    InputMessenger* messenger = server->messenger();
+    SocketOptions opts;
+    opts.fd = fd;
+    opts.user = server;
+    opts.on_edge_triggered_events = [messenger](Socket* s) { messenger->OnNewMessages(s); };
+    SocketId sid;
+    if (Socket::Create(opts, &sid) != 0) {
+      ::close(fd);
+    }
+  }
+}
+
+int Server::Start(int port, const ServerOptions* opt) {
+  EndPoint ep;
+  hostname2endpoint("0.0.0.0", port, &ep);
+  return Start(ep, opt);
+}
+
+int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
+  if (IsRunning()) return -1;
+  if (opt != nullptr) options_ = *opt;
+  policy::RegisterStdProtocol();
+  int listen_fd = tcp_listen(ep);
+  if (listen_fd < 0) {
+    LOG(ERROR) << "tcp_listen on " << endpoint2str(ep) << " failed";
+    return -1;
+  }
+  get_local_side(listen_fd, &listen_ep_);
+  if (listen_ep_.ip.s_addr == 0) hostname2endpoint("127.0.0.1", listen_ep_.port, &listen_ep_);
+  SocketOptions opts;
+  opts.fd = listen_fd;
+  opts.user = this;
+  opts.on_edge_triggered_events = OnNewConnections;
+  if (Socket::Create(opts, &listen_socket_) != 0) {
+    ::close(listen_fd);
+    return -1;
+  }
+  running_.store(true, std::memory_order_release);
+  return 0;
+}
+
+int Server::Stop(int /*wait_ms*/) {
+  if (!running_.exchange(false, std::memory_order_acq_rel)) return 0;
+  SocketUniquePtr s;
+  if (Socket::Address(listen_socket_, &s) == 0) {
+    s->SetFailed(ELOGOFF, "server stopped");
+  }
+  return 0;
+}
+
+int Server::Join() {
+  while (concurrency.load(std::memory_order_acquire) > 0) usleep(1000);
+  return 0;
+}
+
+}  // namespace bam

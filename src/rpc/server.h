@@ -1,0 +1,91 @@
+// brpc_amd: Server — accepts connections, routes parsed requests to
+// registered service methods running in fibers.
+// API parity: reference brpc/server.h (AddService, Start(port, opts),
+// Stop/Join, ServerOptions{num_threads, max_concurrency, idle_timeout}).
+#pragma once
+
+#include <atomic>
+#include <functional>
+#include <map>
+#include <memory>
+#include <string>
+
+#include "base/endpoint.h"
+#include "rpc/controller.h"
+#include "rpc/input_messenger.h"
+
+namespace bam {
+
+// A service method: fills *response (and/or cntl error state) then MUST
+// call done->Run() exactly once (may be after returning, for async).
+typedef std::function<void(Controller* cntl, const IOBuf& request, IOBuf* response,
+                           Closure* done)>
+    MethodFn;
+
+class Service {
+ public:
+  explicit Service(std::string name) : name_(std::move(name)) {}
+  virtual ~Service() {}
+  const std::string& name() const { return name_; }
+  void AddMethod(const std::string& method, MethodFn fn) { methods_[method] = std::move(fn); }
+  const MethodFn* FindMethod(const std::string& method) const {
+    auto it = methods_.find(method);
+    return it == methods_.end() ? nullptr : &it->second;
+  }
+  const std::map<std::string, MethodFn>& methods() const { return methods_; }
+
+ private:
+  std::string name_;
+  std::map<std::string, MethodFn> methods_;
+};
+
+enum ServiceOwnership { SERVER_OWNS_SERVICE, SERVER_DOESNT_OWN_SERVICE };
+
+struct ServerOptions {
+  int idle_timeout_sec = -1;
+  int max_concurrency = 0;          // 0 = unlimited
+  bool has_builtin_services = true;
+};
+
+class MethodStatusRecorder;  // var/latency recorder per method (var layer)
+
+class Server {
+ public:
+  Server();
+  ~Server();
+
+  int AddService(Service* service, ServiceOwnership ownership);
+  int Start(int port, const ServerOptions* opt);  // port 0 = pick free port
+  int Start(const EndPoint& ep, const ServerOptions* opt);
+  int Stop(int wait_ms = 0);
+  int Join();
+
+  bool IsRunning() const { return running_.load(std::memory_order_acquire); }
+  EndPoint listen_address() const { return listen_ep_; }
+
+  // Request routing (called by protocol ProcessRequest).
+  const MethodFn* FindMethod(const std::string& service, const std::string& method,
+                             Service** svc_out = nullptr) const;
+
+  // stats
+  std::atomic<int64_t> nprocessed{0};
+  std::atomic<int32_t> concurrency{0};
+  int max_concurrency() const { return options_.max_concurrency; }
+  const ServerOptions& options() const { return options_; }
+  InputMessenger* messenger() { return &messenger_; }
+
+  const std::map<std::string, Service*>& services() const { return services_; }
+
+ private:
+  static void OnNewConnections(Socket* listen_socket);
+
+  std::map<std::string, Service*> services_;
+  std::map<Service*, ServiceOwnership> ownership_;
+  ServerOptions options_;
+  EndPoint listen_ep_;
+  SocketId listen_socket_ = 0;
+  std::atomic<bool> running_{false};
+  InputMessenger messenger_;
+};
+
+}  // namespace bam

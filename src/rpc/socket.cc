@@ -1,0 +1,418 @@
+#include "rpc/socket.h"
+
+#include <errno.h>
+#include <string.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include "base/logging.h"
+#include "base/object_pool.h"
+#include "base/resource_pool.h"
+#include "base/time.h"
+#include "fiber/butex.h"
+#include "fiber/fiber.h"
+#include "fiber/session.h"
+#include "rpc/event_dispatcher.h"
+
+namespace bam {
+
+static Socket::WriteRequest* const kWriteSentinel = (Socket::WriteRequest*)1;
+
+// ---------------- versioned ref pool ----------------
+
+namespace {
+inline uint32_t rid_of_sock(SocketId id) { return (uint32_t)(id & 0xffffffffu) - 1; }
+inline uint32_t ver_of_sock(SocketId id) { return (uint32_t)(id >> 32); }
+
+std::mutex g_socket_list_mu;
+std::vector<SocketId> g_socket_list;  // live ids (diagnostics)
+
+void track_socket(SocketId id) {
+  std::lock_guard<std::mutex> lk(g_socket_list_mu);
+  g_socket_list.push_back(id);
+}
+void untrack_socket(SocketId id) {
+  std::lock_guard<std::mutex> lk(g_socket_list_mu);
+  for (size_t i = 0; i < g_socket_list.size(); ++i) {
+    if (g_socket_list[i] == id) {
+      g_socket_list[i] = g_socket_list.back();
+      g_socket_list.pop_back();
+      return;
+    }
+  }
+}
+}  // namespace
+
+void ListSockets(std::vector<SocketId>* out) {
+  std::lock_guard<std::mutex> lk(g_socket_list_mu);
+  *out = g_socket_list;
+}
+
+void SocketUniquePtr::reset(Socket* s) {
+  if (s_ != nullptr) s_->ReleaseRef();
+  s_ = s;
+}
+
+int Socket::Create(const SocketOptions& options, SocketId* id) {
+  ResourceId rid;
+  Socket* s = get_resource<Socket>(&rid);
+  if (s == nullptr) return -1;
+  s->pool_index_ = rid;
+  if (s->epollout_butex_ == nullptr) {
+    s->epollout_butex_ = butex_create();
+    s->epollout_butex_->store(0, std::memory_order_relaxed);
+  }
+  uint64_t vr = s->versioned_ref_.load(std::memory_order_relaxed);
+  uint32_t ver = (uint32_t)(vr >> 32);
+  CHECK((ver & 1) == 0) << "creating from a dying socket";
+  s->id_ = ((uint64_t)ver << 32) | (rid + 1);
+  s->failed_.store(false, std::memory_order_relaxed);
+  s->error_code_ = 0;
+  s->error_text_.clear();
+  s->user_ = options.user;
+  s->on_edge_triggered_events_ = options.on_edge_triggered_events;
+  s->on_failed_ = options.on_failed;
+  s->remote_side_ = options.remote_side;
+  s->read_buf_.clear();
+  s->preferred_protocol_index = -1;
+  s->input_events_.store(0, std::memory_order_relaxed);
+  s->write_head_.store(nullptr, std::memory_order_relaxed);
+  s->in_bytes = 0;
+  s->out_bytes = 0;
+  s->in_messages = 0;
+  s->out_messages = 0;
+
+  int fd = options.fd;
+  bool connecting = false;
+  if (fd < 0 && options.connect_on_create) {
+    fd = tcp_connect(options.remote_side, &connecting);
+    if (fd < 0) {
+      // mark free again (nref stays 0, version even)
+      return_resource<Socket>(rid);
+      return -1;
+    }
+  }
+  if (fd >= 0) {
+    make_non_blocking(fd);
+    make_no_delay(fd);
+    get_local_side(fd, &s->local_side_);
+    if (options.remote_side.port == 0) get_remote_side(fd, &s->remote_side_);
+  }
+  s->fd_.store(fd, std::memory_order_release);
+  s->connecting_.store(connecting, std::memory_order_release);
+  // creation reference
+  s->versioned_ref_.fetch_add(1, std::memory_order_acq_rel);
+  *id = s->id_;
+  track_socket(s->id_);
+  if (fd >= 0) {
+    if (EventDispatcher::singleton()->add_consumer(s->id_, fd) != 0) {
+      s->SetFailed(errno, "epoll add failed");
+      return -1;
+    }
+  }
+  return 0;
+}
+
+int Socket::Address(SocketId id, SocketUniquePtr* ptr) {
+  Socket* s = address_resource<Socket>(rid_of_sock(id));
+  if (s == nullptr) return -1;
+  uint64_t vr = s->versioned_ref_.load(std::memory_order_acquire);
+  for (;;) {
+    if ((uint32_t)(vr >> 32) != ver_of_sock(id)) return -1;
+    if (s->versioned_ref_.compare_exchange_weak(vr, vr + 1, std::memory_order_acq_rel)) break;
+  }
+  ptr->reset(s);
+  return 0;
+}
+
+void Socket::ReleaseRef() {
+  uint64_t vr = versioned_ref_.fetch_sub(1, std::memory_order_acq_rel);
+  if ((vr & 0xffffffffULL) == 1 && ((vr >> 32) & 1)) Recycle();
+}
+
+int Socket::SetFailed(int error_code, const char* error_text) {
+  uint64_t vr = versioned_ref_.load(std::memory_order_acquire);
+  for (;;) {
+    if ((vr >> 32) & 1) return -1;  // already failed
+    if (versioned_ref_.compare_exchange_weak(vr, vr + (1ULL << 32),
+                                             std::memory_order_acq_rel)) {
+      break;
+    }
+  }
+  failed_.store(true, std::memory_order_release);
+  error_code_ = error_code;
+  error_text_ = error_text != nullptr ? error_text : "";
+  untrack_socket(id_);
+  int fd = fd_.load(std::memory_order_acquire);
+  if (fd >= 0) EventDispatcher::singleton()->remove_consumer(fd);
+  // Wake writers parked on epollout.
+  epollout_butex_->fetch_add(1, std::memory_order_release);
+  butex_wake_all(epollout_butex_);
+  // Fail pending RPC sessions.
+  std::vector<uint64_t> pending;
+  {
+    std::lock_guard<std::mutex> lk(pending_mu_);
+    pending.swap(pending_sessions_);
+  }
+  for (uint64_t sid : pending) session_error(sid, error_code != 0 ? error_code : ECONNRESET);
+  // NOTE: the write queue is NOT drained here — only the current write
+  // owner frees WriteRequests (see ReleaseAllWriteRequests), avoiding a
+  // double-free race between SetFailed and KeepWrite.
+  if (on_failed_) on_failed_(id_);
+  ReleaseRef();  // creation ref
+  return 0;
+}
+
+void Socket::Recycle() {
+  int fd = fd_.load(std::memory_order_acquire);
+  if (fd >= 0) {
+    ::close(fd);
+    fd_.store(-1, std::memory_order_release);
+  }
+  read_buf_.clear();
+  on_edge_triggered_events_ = nullptr;
+  on_failed_ = nullptr;
+  // version: odd -> next even (free state)
+  versioned_ref_.fetch_add(1ULL << 32, std::memory_order_acq_rel);
+  return_resource<Socket>(pool_index_);
+}
+
+void Socket::add_pending_session(uint64_t sid) {
+  std::lock_guard<std::mutex> lk(pending_mu_);
+  pending_sessions_.push_back(sid);
+}
+
+void Socket::remove_pending_session(uint64_t sid) {
+  std::lock_guard<std::mutex> lk(pending_mu_);
+  for (size_t i = 0; i < pending_sessions_.size(); ++i) {
+    if (pending_sessions_[i] == sid) {
+      pending_sessions_[i] = pending_sessions_.back();
+      pending_sessions_.pop_back();
+      return;
+    }
+  }
+}
+
+// ---------------- write path ----------------
+
+void Socket::NotifyWriteFailure(WriteRequest* head, int err) {
+  // The chain may contain sentinel next pointers being published; spin.
+  WriteRequest* p = head;
+  while (p != nullptr) {
+    WriteRequest* nx;
+    while ((nx = p->next.load(std::memory_order_acquire)) == kWriteSentinel) sched_yield();
+    if (p->id_wait != 0) session_error(p->id_wait, err != 0 ? err : EPIPE);
+    p->data.clear();
+    return_object(p);
+    p = nx;
+  }
+}
+
+void Socket::ReleaseAllWriteRequests(WriteRequest* fifo_head, int err) {
+  if (err == 0) err = EPIPE;
+  // Find the tail of our FIFO chain — the node write_head_ may point at.
+  WriteRequest* last = fifo_head;
+  for (;;) {
+    WriteRequest* nx = last->next.load(std::memory_order_acquire);
+    if (nx == nullptr) break;
+    CHECK(nx != kWriteSentinel);
+    last = nx;
+  }
+  WriteRequest* h = write_head_.exchange(nullptr, std::memory_order_acq_rel);
+  // Free anything newer than `last` (chain h -> ... -> last).
+  if (h != nullptr && h != last) {
+    WriteRequest* p = h;
+    while (p != last) {
+      WriteRequest* nx;
+      while ((nx = p->next.load(std::memory_order_acquire)) == kWriteSentinel) sched_yield();
+      if (p->id_wait != 0) session_error(p->id_wait, err);
+      p->data.clear();
+      return_object(p);
+      p = nx;
+    }
+  }
+  // Free our own FIFO chain (includes `last`).
+  WriteRequest* p = fifo_head;
+  while (p != nullptr) {
+    WriteRequest* nx = p->next.load(std::memory_order_acquire);
+    if (p->id_wait != 0) session_error(p->id_wait, err);
+    p->data.clear();
+    return_object(p);
+    p = nx;
+  }
+}
+
+int Socket::wait_epoll_out(int64_t abstime_us) {
+  int v = epollout_butex_->load(std::memory_order_acquire);
+  if (Failed()) return -1;
+  butex_wait(epollout_butex_, v, abstime_us > 0 ? &abstime_us : nullptr);
+  return Failed() ? -1 : 0;
+}
+
+void Socket::on_output_event() {
+  epollout_butex_->fetch_add(1, std::memory_order_release);
+  butex_wake_all(epollout_butex_);
+}
+
+Socket::WriteRequest* Socket::PopNextRequest(WriteRequest* done) {
+  WriteRequest* expected = done;
+  if (write_head_.compare_exchange_strong(expected, nullptr, std::memory_order_acq_rel)) {
+    done->data.clear();
+    return_object(done);
+    return nullptr;
+  }
+  // Newer requests were pushed: expected = newest. Walk newest->older until
+  // `done`, reversing into FIFO order.
+  WriteRequest* p = expected;
+  WriteRequest* fifo = nullptr;
+  while (p != done) {
+    WriteRequest* nx;
+    while ((nx = p->next.load(std::memory_order_acquire)) == kWriteSentinel) sched_yield();
+    p->next.store(fifo, std::memory_order_relaxed);
+    fifo = p;
+    p = nx;
+  }
+  done->data.clear();
+  return_object(done);
+  return fifo;
+}
+
+int Socket::DoWrite(WriteRequest* req) {
+  // Returns when everything is written + queue drained, or error.
+  WriteRequest* cur = req;
+  while (cur != nullptr) {
+    if (Failed()) {
+      ReleaseAllWriteRequests(cur, error_code_);
+      return -1;
+    }
+    if (connecting_.load(std::memory_order_acquire)) {
+      if (wait_epoll_out(monotonic_time_us() + 30 * 1000000) != 0) continue;  // ->Failed
+      int err = 0;
+      socklen_t len = sizeof(err);
+      getsockopt(fd(), SOL_SOCKET, SO_ERROR, &err, &len);
+      if (err != 0) {
+        SetFailed(err, "connect failed");
+        continue;
+      }
+      get_local_side(fd(), &local_side_);
+      connecting_.store(false, std::memory_order_release);
+    }
+    ssize_t nw = cur->data.cut_into_file_descriptor(fd());
+    if (nw < 0) {
+      if (errno == EAGAIN || errno == EWOULDBLOCK) {
+        wait_epoll_out(monotonic_time_us() + 1000000);  // 1s backstop, then retry
+        continue;
+      }
+      if (errno == EINTR) continue;
+      int err = errno;
+      SetFailed(err, strerror(err));
+      continue;  // loop top runs ReleaseAllWriteRequests
+    }
+    out_bytes.fetch_add(nw, std::memory_order_relaxed);
+    if (!cur->data.empty()) continue;  // partial write; try again
+    out_messages.fetch_add(1, std::memory_order_relaxed);
+    WriteRequest* nx = cur->next.load(std::memory_order_acquire);
+    if (nx != nullptr && nx != kWriteSentinel) {
+      cur->data.clear();
+      return_object(cur);
+      cur = nx;
+    } else {
+      cur = PopNextRequest(cur);  // nullptr when drained
+    }
+  }
+  return 0;
+}
+
+struct KeepWriteArg {
+  Socket* socket;
+  Socket::WriteRequest* req;
+};
+
+void Socket::KeepWriteFiber(void* arg) {
+  KeepWriteArg* kw = (KeepWriteArg*)arg;
+  Socket* s = kw->socket;
+  s->DoWrite(kw->req);
+  s->ReleaseRef();  // the manual ref taken when spawning
+  delete kw;
+}
+
+int Socket::Write(IOBuf* data, const WriteOptions* opt) {
+  WriteOptions dummy;
+  if (opt == nullptr) opt = &dummy;
+  if (Failed()) {
+    if (opt->id_wait != 0) session_error(opt->id_wait, error_code_ != 0 ? error_code_ : EPIPE);
+    errno = EPIPE;
+    return -1;
+  }
+  WriteRequest* req = get_object<WriteRequest>();
+  req->data.clear();
+  req->data.swap(*data);
+  req->id_wait = opt->id_wait;
+  req->socket = this;
+  req->next.store(kWriteSentinel, std::memory_order_relaxed);
+  WriteRequest* prev = write_head_.exchange(req, std::memory_order_acq_rel);
+  if (prev != nullptr) {
+    req->next.store(prev, std::memory_order_release);  // publish link
+    return 0;
+  }
+  req->next.store(nullptr, std::memory_order_release);
+  // We own the queue. Inline attempt only when connected (never block the
+  // caller); otherwise hand to a KeepWrite fiber.
+  if (!connecting_.load(std::memory_order_acquire)) {
+    ssize_t nw = req->data.cut_into_file_descriptor(fd());
+    if (nw < 0 && errno != EAGAIN && errno != EWOULDBLOCK && errno != EINTR) {
+      int err = errno;
+      SetFailed(err, strerror(err));
+      ReleaseAllWriteRequests(req, err);  // we are the owner
+      errno = err;
+      return -1;
+    }
+    if (nw > 0) out_bytes.fetch_add(nw, std::memory_order_relaxed);
+    if (req->data.empty()) {
+      out_messages.fetch_add(1, std::memory_order_relaxed);
+      WriteRequest* next = PopNextRequest(req);
+      if (next == nullptr) return 0;
+      req = next;
+    }
+  }
+  // Not done: spawn KeepWrite holding a manual ref.
+  versioned_ref_.fetch_add(1, std::memory_order_acq_rel);
+  KeepWriteArg* kw = new KeepWriteArg{this, req};
+  fiber_t th;
+  if (fiber_start_background(&th, KeepWriteFiber, kw) != 0) {
+    KeepWriteFiber(kw);  // degrade: run inline
+  }
+  return 0;
+}
+
+// ---------------- read-side event entry ----------------
+
+void Socket::RunInputEventsFiber(void* arg) {
+  Socket* s = (Socket*)arg;  // carries one manual ref
+  for (;;) {
+    if (s->Failed()) {
+      s->input_events_.store(0, std::memory_order_release);
+      break;
+    }
+    s->run_edge_callback();  // drains until EAGAIN
+    int v = s->input_events_.load(std::memory_order_acquire);
+    if (s->input_events_.compare_exchange_strong(v, 0, std::memory_order_acq_rel)) break;
+  }
+  s->ReleaseRef();
+}
+
+void Socket::run_edge_callback() {
+  if (on_edge_triggered_events_) on_edge_triggered_events_(this);
+}
+
+void Socket::on_input_event() {
+  if (input_events_.fetch_add(1, std::memory_order_acq_rel) == 0) {
+    versioned_ref_.fetch_add(1, std::memory_order_acq_rel);  // ref for fiber
+    fiber_t th;
+    if (fiber_start_background(&th, RunInputEventsFiber, this) != 0) {
+      RunInputEventsFiber(this);
+    }
+  }
+}
+
+}  // namespace bam

@@ -1,0 +1,176 @@
+// brpc_amd: Socket — the central connection object.
+// Parity: reference brpc/socket.h redesigned for this runtime:
+//  * addressed by versioned 64-bit SocketId through a resource pool (ABA-safe)
+//  * wait-free multi-producer Write(): producers atomically push WriteRequests;
+//    the winner writes inline once, then a KeepWrite fiber drains
+//  * read side: InputMessenger drains via edge-triggered epoll + fibers
+//  * residency-aware: outgoing IOBufs may reference HBM blocks; the write
+//    path stages them through the pinned ring (base/iobuf.cc + hip/)
+#pragma once
+
+#include <atomic>
+#include <functional>
+#include <mutex>
+#include <vector>
+
+#include "base/endpoint.h"
+#include "base/iobuf.h"
+#include "rpc/protocol.h"
+
+namespace bam {
+
+class Socket;
+
+// RAII reference to a socket; releasing decrements the versioned refcount.
+class SocketUniquePtr {
+ public:
+  SocketUniquePtr() : s_(nullptr) {}
+  explicit SocketUniquePtr(Socket* s) : s_(s) {}
+  ~SocketUniquePtr() { reset(nullptr); }
+  SocketUniquePtr(const SocketUniquePtr&) = delete;
+  SocketUniquePtr& operator=(const SocketUniquePtr&) = delete;
+  SocketUniquePtr(SocketUniquePtr&& o) noexcept : s_(o.s_) { o.s_ = nullptr; }
+  SocketUniquePtr& operator=(SocketUniquePtr&& o) noexcept {
+    if (this != &o) {
+      reset(o.s_);
+      o.s_ = nullptr;
+    }
+    return *this;
+  }
+  Socket* get() const { return s_; }
+  Socket* operator->() const { return s_; }
+  Socket& operator*() const { return *s_; }
+  explicit operator bool() const { return s_ != nullptr; }
+  void reset(Socket* s);
+  Socket* release() {
+    Socket* s = s_;
+    s_ = nullptr;
+    return s;
+  }
+
+ private:
+  Socket* s_;
+};
+
+struct SocketOptions {
+  int fd = -1;                    // already-established fd (server side)
+  EndPoint remote_side;
+  bool connect_on_create = false; // client side: connect to remote_side
+  void* user = nullptr;           // owner cookie (InputMessenger, Acceptor)
+  std::function<void(Socket*)> on_edge_triggered_events;  // readable callback
+  std::function<void(SocketId)> on_failed;                // teardown hook
+};
+
+class Socket {
+ public:
+  struct WriteRequest {
+    IOBuf data;
+    std::atomic<WriteRequest*> next{nullptr};
+    uint64_t id_wait = 0;  // session to error on write failure
+    Socket* socket = nullptr;
+  };
+
+  struct WriteOptions {
+    uint64_t id_wait = 0;        // correlation session notified on failure
+    bool ignore_eovercrowded = false;
+  };
+
+  // Creates a socket, returns 0 and its id.
+  static int Create(const SocketOptions& options, SocketId* id);
+  // Re-addresses an id; returns 0 and a referenced ptr, or -1 if recycled.
+  static int Address(SocketId id, SocketUniquePtr* ptr);
+
+  // Wait-free write: ownership of *data is taken (moved-from on return).
+  // Returns 0 on success (queued or written), -1 with errno otherwise.
+  int Write(IOBuf* data, const WriteOptions* opt = nullptr);
+
+  // Marks failed: new Address() fail, pending writes error out, epoll
+  // deregistered, fd closed when the last ref drops.
+  int SetFailed(int error_code, const char* error_text);
+  bool Failed() const { return failed_.load(std::memory_order_acquire); }
+
+  int fd() const { return fd_.load(std::memory_order_acquire); }
+  SocketId id() const { return id_; }
+  const EndPoint& remote_side() const { return remote_side_; }
+  const EndPoint& local_side() const { return local_side_; }
+  void* user() const { return user_; }
+
+  // Read buffer for InputMessenger.
+  IOBuf& read_buf() { return read_buf_; }
+  int preferred_protocol_index = -1;
+
+  // Correlation sessions waiting for responses on this socket; failed
+  // when the connection breaks. (Parity: reference conducts errors to
+  // ids; we keep an explicit registry.)
+  void add_pending_session(uint64_t sid);
+  void remove_pending_session(uint64_t sid);
+
+  // Called by the event dispatcher.
+  void on_input_event();    // edge-triggered readable
+  void on_output_event();   // edge-triggered writable (wakes epollout waiters)
+  void run_edge_callback();
+
+  // Blocks current fiber until the fd is writable (or failed/timeout).
+  int wait_epoll_out(int64_t abstime_us);
+
+  // Per-connection stats (builtin /connections page).
+  std::atomic<int64_t> in_bytes{0};
+  std::atomic<int64_t> out_bytes{0};
+  std::atomic<int64_t> in_messages{0};
+  std::atomic<int64_t> out_messages{0};
+
+  // internal: versioned-ref bookkeeping
+  void ReleaseRef();
+
+  Socket() {}  // public for ResourcePool; use Create()
+
+ private:
+  friend class SocketUniquePtr;
+
+  static void KeepWriteFiber(void* arg);
+  static void RunInputEventsFiber(void* arg);
+  // Writes req->data (+ successors) until EAGAIN/empty. Returns 0, or -1.
+  int DoWrite(WriteRequest* req);
+  // After finishing `done`, returns the next request (FIFO) or nullptr if
+  // the queue drained (CAS head -> nullptr succeeded).
+  WriteRequest* PopNextRequest(WriteRequest* done);
+  void NotifyWriteFailure(WriteRequest* head_chain, int err);
+  // Owner-side cleanup on failure: frees its own FIFO chain plus anything
+  // newer in the queue (SetFailed never touches the queue; the owner does).
+  void ReleaseAllWriteRequests(WriteRequest* fifo_head, int err);
+  void Recycle();  // last ref of a failed socket dropped
+
+  SocketId id_ = 0;
+  uint32_t pool_index_ = 0;
+  // packs (version << 32) | nref
+  std::atomic<uint64_t> versioned_ref_{0};
+  std::atomic<int> fd_{-1};
+  std::atomic<bool> failed_{false};
+  std::atomic<bool> connecting_{false};
+  int error_code_ = 0;
+  std::string error_text_;
+  EndPoint remote_side_;
+  EndPoint local_side_;
+  void* user_ = nullptr;
+  std::function<void(Socket*)> on_edge_triggered_events_;
+  std::function<void(SocketId)> on_failed_;
+  IOBuf read_buf_;
+
+ public:
+  std::atomic<int> input_events_{0};
+
+ private:
+
+  std::atomic<WriteRequest*> write_head_{nullptr};
+  std::atomic<int>* epollout_butex_ = nullptr;
+
+  std::mutex pending_mu_;
+  std::vector<uint64_t> pending_sessions_;
+
+  friend class EventDispatcher;
+};
+
+// Iterates live sockets (builtin /connections).
+void ListSockets(std::vector<SocketId>* out);
+
+}  // namespace bam

@@ -1,0 +1,95 @@
+"""End-to-end RPC tests over loopback TCP in one process.
+
+Models the reference's ChannelTest fixture (test/brpc_channel_unittest.cpp):
+in-process server + client, sync calls, errors, timeouts, attachments.
+BASELINE config 1 (echo over loopback on CPU) is exercised here.
+"""
+import pytest
+
+import brpc_amd as b
+
+r = b.core.rpc
+
+# error-code parity with reference brpc/errno.proto
+ENOSERVICE = 1001
+ENOMETHOD = 1002
+ERPCTIMEDOUT = 1008
+EINTERNAL = 2001
+
+
+@pytest.fixture(scope="module")
+def server_addr():
+    port = r.start_echo_server(0)
+    assert port > 0
+    return f"127.0.0.1:{port}"
+
+
+def test_echo_roundtrip(server_addr):
+    rc, resp, lat = r.echo_once(server_addr, b"hello rpc world", 2000)
+    assert rc == 0
+    assert resp == b"hello rpc world"
+    assert 0 < lat < 2_000_000
+
+
+def test_echo_empty_payload(server_addr):
+    rc, resp, _ = r.echo_once(server_addr, b"", 2000)
+    assert rc == 0
+    assert resp == b""
+
+
+def test_echo_large_payload(server_addr):
+    import os
+
+    payload = os.urandom(1 << 20)  # 1 MiB spans many IOBuf blocks
+    rc, resp, _ = r.echo_once(server_addr, payload, 10000)
+    assert rc == 0
+    assert resp == payload
+
+
+def test_attachment_roundtrip(server_addr):
+    assert r.attachment_test(server_addr)
+
+
+def test_unknown_method(server_addr):
+    rc, _, err = r.call_method_once(server_addr, "EchoService.NoSuch", b"x", 1000, 0)
+    assert rc == ENOMETHOD
+    assert "NoSuch" in err
+
+
+def test_unknown_service(server_addr):
+    rc, _, err = r.call_method_once(server_addr, "Nope.Echo", b"x", 1000, 0)
+    assert rc in (ENOSERVICE, ENOMETHOD)
+
+
+def test_server_side_failure(server_addr):
+    rc, _, err = r.call_method_once(server_addr, "EchoService.Fail", b"x", 1000, 0)
+    assert rc == EINTERNAL
+    assert "asked for it" in err
+
+
+def test_timeout(server_addr):
+    rc, _, err = r.call_method_once(server_addr, "EchoService.Sleep", b"300", 100, 0)
+    assert rc == ERPCTIMEDOUT
+
+
+def test_slow_call_within_deadline(server_addr):
+    rc, resp, _ = r.call_method_once(server_addr, "EchoService.Sleep", b"50", 2000, 0)
+    assert rc == 0
+    assert resp == b"slept"
+
+
+def test_connection_refused_with_retries():
+    rc, _, err = r.call_method_once("127.0.0.1:1", "EchoService.Echo", b"x", 2000, 2)
+    assert rc != 0 and rc != ERPCTIMEDOUT
+
+
+def test_concurrent_echo_bench_small(server_addr):
+    res = r.echo_bench(server_addr, 64, 8, 2000, 5000)
+    assert res["errors"] == 0
+    assert res["total"] == 2000
+    assert res["qps"] > 100
+
+
+def test_concurrent_echo_16k(server_addr):
+    res = r.echo_bench(server_addr, 16384, 8, 500, 10000)
+    assert res["errors"] == 0
