@@ -2,6 +2,7 @@
 
 #include <errno.h>
 #include <string.h>
+#include <poll.h>
 #include <sys/socket.h>
 #include <unistd.h>
 
@@ -245,6 +246,14 @@ void Socket::ReleaseAllWriteRequests(WriteRequest* fifo_head, int err) {
 int Socket::wait_epoll_out(int64_t abstime_us) {
   int v = epollout_butex_->load(std::memory_order_acquire);
   if (Failed()) return -1;
+  // Edge-triggered EPOLLOUT may have fired before we captured v (e.g. the
+  // connect completed between Create and the first Write) — never park when
+  // the fd is already writable.
+  struct pollfd pfd;
+  pfd.fd = fd_.load(std::memory_order_acquire);
+  pfd.events = POLLOUT;
+  pfd.revents = 0;
+  if (::poll(&pfd, 1, 0) > 0 && (pfd.revents & (POLLOUT | POLLERR | POLLHUP))) return Failed() ? -1 : 0;
   butex_wait(epollout_butex_, v, abstime_us > 0 ? &abstime_us : nullptr);
   return Failed() ? -1 : 0;
 }
@@ -300,7 +309,7 @@ int Socket::DoWrite(WriteRequest* req) {
     ssize_t nw = cur->data.cut_into_file_descriptor(fd());
     if (nw < 0) {
       if (errno == EAGAIN || errno == EWOULDBLOCK) {
-        wait_epoll_out(monotonic_time_us() + 1000000);  // 1s backstop, then retry
+        wait_epoll_out(monotonic_time_us() + 100000);  // 100ms backstop, then retry
         continue;
       }
       if (errno == EINTR) continue;
