@@ -1,0 +1,140 @@
+#!/usr/bin/env python3
+"""Flagship benchmark (driver contract): echo RPC QPS with IOBuf-in-HBM.
+
+BASELINE.json metric: "echo QPS + p99 latency (64B & 16KB payload)".
+This measures config 2 (multi_threaded_echo, 64 B payload, IOBuf blocks in
+HBM3E, baidu_std-compatible wire protocol) on N GPUs of one node: each rank
+runs an in-process echo server + multi-fiber client over loopback TCP with
+request payloads staged into HBM blocks and responses staged back out of
+HBM by the server (weak scaling: per-GPU work fixed).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+       (N>1 is launched by the driver via torch.distributed.run, one rank
+        per GPU over RCCL; RANK/LOCAL_RANK/WORLD_SIZE read from env.)
+One JSON line is printed by rank 0.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+# Pin this process to its GPU BEFORE importing torch / brpc_amd.
+_local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+if "WORLD_SIZE" in os.environ and int(os.environ.get("WORLD_SIZE", "1")) > 1:
+    os.environ.setdefault("HIP_VISIBLE_DEVICES", str(_local_rank))
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+import brpc_amd as b  # noqa: E402
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--payload", type=int, default=64)
+    ap.add_argument("--concurrency", type=int, default=32)
+    ap.add_argument("--calls-per-step", type=int, default=5000)
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+
+    dist = None
+    torch = None
+    if world_size > 1:
+        import torch  # noqa: F401
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend=backend)
+
+    ndev = b.core.gpu.initialize()
+    use_gpu = ndev > 0
+    if use_gpu:
+        method, hbm_req = "EchoService.EchoHbm", True
+    else:
+        # CPU-only container: same benchmark on the host path.
+        method, hbm_req = "EchoService.Echo", False
+
+    port = b.core.rpc.start_echo_server(0)
+    addr = "127.0.0.1:%d" % port
+
+    def run_step():
+        res = b.core.rpc.echo_bench(addr, args.payload, args.concurrency,
+                                    args.calls_per_step, 30000, method, hbm_req)
+        if res["errors"]:
+            raise RuntimeError("bench errors: %s" % res)
+        return res
+
+    for _ in range(args.warmup):
+        run_step()
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+            if torch is not None and torch.cuda.is_available():
+                torch.cuda.synchronize()
+
+    barrier_sync()
+    t0 = time.monotonic()
+    p99s, qps_acc = [], []
+    total_calls = 0
+    for _ in range(args.steps):
+        res = run_step()
+        p99s.append(res["p99_us"])
+        qps_acc.append(res["qps"])
+        total_calls += res["total"]
+    barrier_sync()
+    elapsed = time.monotonic() - t0
+
+    # MAX elapsed over ranks; SUM of calls over ranks.
+    if dist is not None:
+        import torch
+
+        te = torch.tensor([elapsed], dtype=torch.float64)
+        tc = torch.tensor([float(total_calls)], dtype=torch.float64)
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        dist.all_reduce(tc, op=dist.ReduceOp.SUM)
+        elapsed = te.item()
+        total_calls = int(tc.item())
+
+    qps = total_calls / elapsed
+    ms_per_step = elapsed * 1000.0 / args.steps
+    p99_us = max(p99s)
+
+    if rank == 0:
+        out = {
+            "metric": "echo_qps",
+            "value": qps,
+            "unit": "requests/s",
+            "n_gpus": world_size if world_size > 1 else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bytes",
+            "data": "synthetic",
+            "config": {
+                "model": "multi_threaded_echo (baidu_std wire, IOBuf-in-HBM)"
+                if use_gpu else "multi_threaded_echo (baidu_std wire, host IOBuf)",
+                "payload_bytes": args.payload,
+                "concurrency_per_gpu": args.concurrency,
+                "calls_per_step": args.calls_per_step,
+                "p99_us": p99_us,
+                "parallelism": "dp%d" % (world_size if world_size > 1 else 1),
+                "gpu_payload_residency": "HBM" if use_gpu else "host",
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

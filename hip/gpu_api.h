@@ -1,0 +1,47 @@
+// brpc_amd: C ABI of libbrpc_hip.so (hipcc-built, gfx950-only).
+// The core runtime dlopens this library when a GPU is present and wires it
+// into IOBuf (block allocators + byte movers) and the checksum/compress
+// registries. No torch dependency — plain HIP runtime.
+#pragma once
+
+#include <stddef.h>
+#include <stdint.h>
+
+extern "C" {
+
+// Returns #GPUs (0 = none / init failed). Safe to call repeatedly.
+int bam_gpu_device_count(void);
+
+// ---- block memory (IOBuf residency backends) ----
+// HBM pools use size classes {8 KiB, 64 KiB, 2 MiB} carved from big slabs
+// (parity: reference rdma/block_pool.cpp registered-memory pool design).
+void* bam_gpu_alloc_hbm(uint32_t cap, int dev);
+void bam_gpu_free_hbm(void* p, uint32_t cap, int dev);
+void* bam_gpu_alloc_pinned(uint32_t cap, int dev);
+void bam_gpu_free_pinned(void* p, uint32_t cap, int dev);
+
+// ---- byte movement ----
+// res codes match bam::Residency: 0 host, 1 pinned, 2 hbm.
+void bam_gpu_memcpy(void* dst, int dst_res, int dst_dev, const void* src, int src_res,
+                    int src_dev, size_t n);
+
+// ---- gfx950 kernels ----
+// CRC32-C of device memory; init follows the host convention (0 for fresh).
+uint32_t bam_gpu_crc32c(const void* dev_ptr, size_t n, uint32_t init, int dev);
+
+// Gather scattered device spans into one contiguous device buffer.
+// srcs/lens are host arrays of device pointers/lengths.
+int bam_gpu_gather(void* dst_dev, const void* const* srcs, const size_t* lens, int nspans,
+                   int dev);
+// Scatter a contiguous device buffer into device spans.
+int bam_gpu_scatter(void* const* dsts, const size_t* lens, int nspans, const void* src_dev,
+                    int dev);
+
+// Fill device memory with a repeating 64-bit pattern (tests / synthetic
+// payload generation without H2D traffic).
+int bam_gpu_fill(void* dst_dev, size_t n, uint64_t pattern, int dev);
+
+// Last error string (static buffer).
+const char* bam_gpu_last_error(void);
+
+}  // extern "C"

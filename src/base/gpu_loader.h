@@ -1,0 +1,38 @@
+// brpc_amd: runtime loader for libbrpc_hip.so (the gfx950 kernel library).
+// The core stays torch/HIP-free; when a GPU is present the library is
+// dlopened and wired into IOBuf (PINNED/HBM allocators + byte mover) and
+// the checksum registry. On a GPU machine a missing library is a HARD
+// error — the HIP path must never silently fall back to host code.
+#pragma once
+
+#include <stddef.h>
+#include <stdint.h>
+
+namespace bam {
+namespace gpu {
+
+struct GpuApi {
+  int (*device_count)(void);
+  void* (*alloc_hbm)(uint32_t, int);
+  void (*free_hbm)(void*, uint32_t, int);
+  void* (*alloc_pinned)(uint32_t, int);
+  void (*free_pinned)(void*, uint32_t, int);
+  void (*memcpy_res)(void*, int, int, const void*, int, int, size_t);
+  uint32_t (*crc32c)(const void*, size_t, uint32_t, int);
+  int (*gather)(void*, const void* const*, const size_t*, int, int);
+  int (*scatter)(void* const*, const size_t*, int, const void*, int);
+  int (*fill)(void*, size_t, uint64_t, int);
+  const char* (*last_error)(void);
+};
+
+// Loads the library (idempotent). Returns device count (0 = no GPU or no
+// library; check loaded() / error()). Registers IOBuf hooks on success.
+int initialize();
+
+bool loaded();
+const GpuApi* api();          // nullptr until loaded
+const char* load_error();     // empty if ok
+int device_count();           // 0 when not loaded
+
+}  // namespace gpu
+}  // namespace bam

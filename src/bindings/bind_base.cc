@@ -90,3 +90,34 @@ void bind_base(py::module_& m) {
     return ep;
   });
 }
+
+// ---- GPU submodule (wired to libbrpc_hip.so via base/gpu_loader) ----
+#include "base/gpu_loader.h"
+
+namespace bam {
+namespace gputest {
+bool hbm_iobuf_roundtrip(size_t n, int dev);
+bool crc_matches(size_t n, int dev);
+bool crc_extend_matches(size_t n1, size_t n2, int dev);
+bool gather_matches(size_t total, uint32_t block, int dev);
+bool pinned_roundtrip(size_t n);
+}  // namespace gputest
+}  // namespace bam
+
+void bind_gpu(py::module_& m) {
+  auto g = m.def_submodule("gpu");
+  g.def("initialize", &bam::gpu::initialize, py::call_guard<py::gil_scoped_release>());
+  g.def("device_count", &bam::gpu::device_count, py::call_guard<py::gil_scoped_release>());
+  g.def("loaded", &bam::gpu::loaded);
+  g.def("load_error", &bam::gpu::load_error);
+  g.def("hbm_iobuf_roundtrip", &bam::gputest::hbm_iobuf_roundtrip, py::arg("n"),
+        py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
+  g.def("crc_matches", &bam::gputest::crc_matches, py::arg("n"), py::arg("dev") = 0,
+        py::call_guard<py::gil_scoped_release>());
+  g.def("crc_extend_matches", &bam::gputest::crc_extend_matches, py::arg("n1"), py::arg("n2"),
+        py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
+  g.def("gather_matches", &bam::gputest::gather_matches, py::arg("total"), py::arg("block"),
+        py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
+  g.def("pinned_roundtrip", &bam::gputest::pinned_roundtrip, py::arg("n"),
+        py::call_guard<py::gil_scoped_release>());
+}

@@ -20,7 +20,8 @@ int call_method_once(const std::string& addr, const std::string& method,
                      std::string* response_out, std::string* error_text);
 bool attachment_test(const std::string& addr);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
-                       int64_t total_calls, int timeout_ms);
+                       int64_t total_calls, int timeout_ms, const std::string& method,
+                       bool hbm_request);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -64,11 +65,12 @@ void bind_rpc(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   r.def("echo_bench",
         [](const std::string& addr, int payload_size, int concurrency, int64_t total,
-           int timeout_ms) {
+           int timeout_ms, const std::string& method, bool hbm_request) {
           bam::rpctest::BenchResult b;
           {
             py::gil_scoped_release rel;
-            b = bam::rpctest::echo_bench(addr, payload_size, concurrency, total, timeout_ms);
+            b = bam::rpctest::echo_bench(addr, payload_size, concurrency, total, timeout_ms,
+                                         method, hbm_request);
           }
           py::dict d;
           d["qps"] = b.qps;
@@ -84,5 +86,6 @@ void bind_rpc(py::module_& m) {
           return d;
         },
         py::arg("addr"), py::arg("payload_size") = 64, py::arg("concurrency") = 8,
-        py::arg("total") = 10000, py::arg("timeout_ms") = 5000);
+        py::arg("total") = 10000, py::arg("timeout_ms") = 5000,
+        py::arg("method") = "EchoService.Echo", py::arg("hbm_request") = false);
 }

@@ -6,4 +6,5 @@ PYBIND11_MODULE(_core, m) {
   bind_fiber(m);
   bind_rpc(m);
   bind_var(m);
+  bind_gpu(m);
 }

@@ -39,6 +39,21 @@ Service* NewEchoService() {
     cntl->SetFailed(EINTERNAL, "you asked for it");
     done->Run();
   });
+  // BASELINE config 2 (IOBuf-in-HBM echo): the response payload is staged
+  // into HBM-resident blocks; the socket write path stages it back through
+  // the GPU byte mover. Falls back with an explicit error if no GPU.
+  svc->AddMethod("EchoHbm", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    if (!has_block_allocator(RES_HBM)) {
+      cntl->SetFailed(EINTERNAL, "no HBM allocator (GPU library not loaded)");
+      done->Run();
+      return;
+    }
+    std::string bytes = req.to_string();
+    if (resp->append_with_residency(bytes.data(), bytes.size(), RES_HBM, 0, 0) != 0) {
+      cntl->SetFailed(EINTERNAL, "HBM append failed");
+    }
+    done->Run();
+  });
   return svc;
 }
 
@@ -119,6 +134,8 @@ struct BenchWorkerArg {
   std::vector<int64_t>* latencies;  // pre-sized; indexed by call #
   std::atomic<int64_t>* errors;
   std::string payload;
+  std::string method;
+  bool hbm_request;
   CountdownEvent* done_event;
 };
 
@@ -129,8 +146,12 @@ void bench_worker(void* raw) {
     if (idx <= 0) break;
     Controller cntl;
     IOBuf request, response;
-    request.append(a->payload);
-    a->channel->CallMethod("EchoService.Echo", &cntl, &request, &response, nullptr);
+    if (a->hbm_request) {
+      request.append_with_residency(a->payload.data(), a->payload.size(), RES_HBM, 0, 0);
+    } else {
+      request.append(a->payload);
+    }
+    a->channel->CallMethod(a->method, &cntl, &request, &response, nullptr);
     if (cntl.Failed() || response.size() != a->payload.size()) {
       a->errors->fetch_add(1, std::memory_order_relaxed);
     }
@@ -142,7 +163,8 @@ void bench_worker(void* raw) {
 }  // namespace
 
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
-                       int64_t total_calls, int timeout_ms) {
+                       int64_t total_calls, int timeout_ms, const std::string& method,
+                       bool hbm_request) {
   BenchResult res;
   Channel channel;
   ChannelOptions opts;
@@ -154,7 +176,8 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
   std::atomic<int64_t> errors{0};
   std::vector<int64_t> latencies(total_calls, 0);
   CountdownEvent done_event(concurrency);
-  BenchWorkerArg arg{&channel, &remaining, &latencies, &errors, payload, &done_event};
+  BenchWorkerArg arg{&channel,  &remaining,  &latencies, &errors,
+                     payload,   method,      hbm_request, &done_event};
   int64_t t0 = monotonic_time_us();
   for (int i = 0; i < concurrency; ++i) {
     fiber_t th;

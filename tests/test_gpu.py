@@ -1,0 +1,70 @@
+"""GPU-path tests (MI355X, gfx950). Run via gpurun: pytest -m gpu.
+
+Each test checks a gfx950 kernel / residency path against the host
+reference (numerics oracle = src/base/crc32c.cc and plain byte compares).
+"""
+import pytest
+
+import brpc_amd as b
+
+g = b.core.gpu
+r = b.core.rpc
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def gpu_available():
+    n = g.initialize()
+    if n == 0:
+        pytest.fail(f"GPU expected but HIP library not usable: {g.load_error()}")
+    return n
+
+
+def test_hip_lib_loaded():
+    assert g.loaded(), g.load_error()
+    assert g.device_count() >= 1
+
+
+@pytest.mark.parametrize("n", [1, 64, 4096, 8192, 65536, 1 << 20, (1 << 22) + 12345])
+def test_hbm_iobuf_roundtrip(n):
+    assert g.hbm_iobuf_roundtrip(n, 0)
+
+
+@pytest.mark.parametrize("n", [1, 7, 64, 4095, 4096, 32768, 32769, 65536,
+                               (1 << 21) - 1, 1 << 22, (1 << 24) + 999])
+def test_gpu_crc32c_matches_host(n):
+    # Exercises both the LDS-tiled full-group path and the boundary path.
+    assert g.crc_matches(n, 0)
+
+
+@pytest.mark.parametrize("n1,n2", [(10, 100), (4096, 65536), (123, 1 << 20)])
+def test_gpu_crc32c_extend(n1, n2):
+    assert g.crc_extend_matches(n1, n2, 0)
+
+
+@pytest.mark.parametrize("total,block", [(100000, 8192), (1 << 21, 65536), (333333, 16384)])
+def test_gpu_gather(total, block):
+    assert g.gather_matches(total, block, 0)
+
+
+def test_pinned_roundtrip():
+    assert g.pinned_roundtrip(1 << 20)
+
+
+def test_hbm_echo_rpc():
+    """BASELINE config 2 path: echo with server response staged in HBM."""
+    port = r.start_echo_server(0)
+    addr = f"127.0.0.1:{port}"
+    rc, resp, err = r.call_method_once(addr, "EchoService.EchoHbm", b"gpu payload" * 100,
+                                       10000, 0)
+    assert rc == 0, err
+    assert resp == b"gpu payload" * 100
+
+
+def test_hbm_echo_bench_smoke():
+    port = r.start_echo_server(0)
+    addr = f"127.0.0.1:{port}"
+    res = r.echo_bench(addr, 16384, 8, 500, 20000, "EchoService.EchoHbm", True)
+    assert res["errors"] == 0, res
+    assert res["qps"] > 10
