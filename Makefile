@@ -10,7 +10,7 @@ HIPCC     := /opt/rocm/bin/hipcc
 GPU_ARCH  := gfx950
 
 CXXFLAGS  := -O2 -g -std=c++17 -fPIC -pthread -Wall -Wno-unused-function \
-             -Isrc -msse4.2 -fvisibility=hidden
+             -Isrc -msse4.2 -fvisibility=hidden -MMD -MP
 LDFLAGS   := -shared -pthread -ldl
 
 CORE_SRCS := $(wildcard src/base/*.cc) $(wildcard src/fiber/*.cc) $(wildcard src/rpc/*.cc) \
@@ -59,5 +59,7 @@ $(HIP_SO): $(HIP_OBJS)
 
 clean:
 	rm -rf build $(CORE_SO) $(HIP_SO)
+
+-include $(shell find build -name '*.d' 2>/dev/null)
 
 .PHONY: all core clean

@@ -168,7 +168,8 @@ static void PackStdResponse(IOBuf* out, int64_t correlation_id, int error_code,
 
 // ---------------- server side ----------------
 
-static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* resp) {
+static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* resp,
+                            var::LatencyRecorder* status, int64_t start_us) {
   SocketUniquePtr sock;
   if (Socket::Address(sid, &sock) == 0) {
     IOBuf packet;
@@ -180,6 +181,7 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
     cntl->server_->concurrency.fetch_sub(1, std::memory_order_relaxed);
     cntl->server_->nprocessed.fetch_add(1, std::memory_order_relaxed);
   }
+  if (status != nullptr) *status << (monotonic_time_us() - start_us);
   delete resp;
   delete cntl;
 }
@@ -202,7 +204,12 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   IOBuf* resp = new IOBuf;
   SocketId sid = sock->id();
   int64_t cid = msg->meta.correlation_id;
-  Closure* done = NewCallback([sid, cid, cntl, resp] { SendStdResponse(sid, cid, cntl, resp); });
+  int64_t start_us = monotonic_time_us();
+  var::LatencyRecorder* status =
+      server != nullptr ? server->method_status(msg->meta.service_name, msg->meta.method_name)
+                        : nullptr;
+  Closure* done = NewCallback(
+      [sid, cid, cntl, resp, status, start_us] { SendStdResponse(sid, cid, cntl, resp, status, start_us); });
 
   // attachment split
   IOBuf req_data;

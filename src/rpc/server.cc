@@ -101,6 +101,18 @@ int Server::Stop(int /*wait_ms*/) {
   return 0;
 }
 
+var::LatencyRecorder* Server::method_status(const std::string& service,
+                                            const std::string& method) {
+  std::string key = service + "." + method;
+  std::lock_guard<std::mutex> lk(status_mu_);
+  auto it = method_status_.find(key);
+  if (it != method_status_.end()) return it->second;
+  var::LatencyRecorder* rec = new var::LatencyRecorder;
+  rec->expose("rpc_server_" + std::to_string(listen_ep_.port) + "_" + key);
+  method_status_[key] = rec;
+  return rec;
+}
+
 int Server::Join() {
   while (concurrency.load(std::memory_order_acquire) > 0) usleep(1000);
   return 0;

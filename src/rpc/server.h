@@ -13,6 +13,7 @@
 #include "base/endpoint.h"
 #include "rpc/controller.h"
 #include "rpc/input_messenger.h"
+#include "var/variable.h"
 
 namespace bam {
 
@@ -67,6 +68,9 @@ class Server {
   const MethodFn* FindMethod(const std::string& service, const std::string& method,
                              Service** svc_out = nullptr) const;
 
+  // Per-method latency recorder (parity: reference details/method_status.h).
+  var::LatencyRecorder* method_status(const std::string& service, const std::string& method);
+
   // stats
   std::atomic<int64_t> nprocessed{0};
   std::atomic<int32_t> concurrency{0};
@@ -81,6 +85,8 @@ class Server {
 
   std::map<std::string, Service*> services_;
   std::map<Service*, ServiceOwnership> ownership_;
+  std::mutex status_mu_;
+  std::map<std::string, var::LatencyRecorder*> method_status_;
   ServerOptions options_;
   EndPoint listen_ep_;
   SocketId listen_socket_ = 0;

@@ -1,0 +1,218 @@
+#include "var/variable.h"
+
+#include <unistd.h>
+
+#include <algorithm>
+#include <condition_variable>
+#include <set>
+#include <thread>
+
+#include "base/time.h"
+
+namespace bam {
+namespace var {
+
+// ---------------- registry ----------------
+
+namespace {
+struct Registry {
+  std::mutex mu;
+  std::map<std::string, Variable*> vars;
+};
+Registry& registry() {
+  static Registry* r = new Registry;
+  return *r;
+}
+}  // namespace
+
+Variable::~Variable() {}
+
+int Variable::expose(const std::string& name) {
+  hide();
+  Registry& r = registry();
+  std::lock_guard<std::mutex> lk(r.mu);
+  name_ = name;
+  r.vars[name] = this;
+  return 0;
+}
+
+void Variable::hide() {
+  if (name_.empty()) return;
+  Registry& r = registry();
+  std::lock_guard<std::mutex> lk(r.mu);
+  auto it = r.vars.find(name_);
+  if (it != r.vars.end() && it->second == this) r.vars.erase(it);
+  name_.clear();
+}
+
+void Variable::dump_exposed(std::ostream& os, const std::string& filter) {
+  Registry& r = registry();
+  std::lock_guard<std::mutex> lk(r.mu);
+  for (const auto& kv : r.vars) {
+    if (!filter.empty() && kv.first.find(filter) == std::string::npos) continue;
+    os << kv.first << " : ";
+    kv.second->describe(os);
+    os << "\n";
+  }
+}
+
+Variable* Variable::find_exposed(const std::string& name) {
+  Registry& r = registry();
+  std::lock_guard<std::mutex> lk(r.mu);
+  auto it = r.vars.find(name);
+  return it == r.vars.end() ? nullptr : it->second;
+}
+
+size_t Variable::count_exposed() {
+  Registry& r = registry();
+  std::lock_guard<std::mutex> lk(r.mu);
+  return r.vars.size();
+}
+
+// ---------------- sampler thread ----------------
+
+namespace {
+struct Sampler {
+  std::mutex mu;
+  std::set<WindowedInt*> targets;
+  bool started = false;
+
+  void ensure_started() {
+    if (started) return;
+    started = true;
+    std::thread([this] { run(); }).detach();
+  }
+
+  void run() {
+    for (;;) {
+      usleep(1000000);
+      std::vector<WindowedInt*> snapshot;
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        snapshot.assign(targets.begin(), targets.end());
+      }
+      for (WindowedInt* w : snapshot) w->take_sample();
+    }
+  }
+};
+Sampler& sampler() {
+  static Sampler* s = new Sampler;
+  return *s;
+}
+}  // namespace
+
+void register_sampler(WindowedInt* w) {
+  Sampler& s = sampler();
+  std::lock_guard<std::mutex> lk(s.mu);
+  s.targets.insert(w);
+  s.ensure_started();
+}
+
+void unregister_sampler(WindowedInt* w) {
+  Sampler& s = sampler();
+  std::lock_guard<std::mutex> lk(s.mu);
+  s.targets.erase(w);
+}
+
+// ---------------- WindowedInt ----------------
+
+WindowedInt::WindowedInt(SourceFn src, int window_sec, bool per_second)
+    : src_(std::move(src)), window_(window_sec), per_second_(per_second), latest_(0),
+      nsamples_(0) {
+  for (auto& v : ring_) v = 0;
+  register_sampler(this);
+}
+
+WindowedInt::~WindowedInt() { unregister_sampler(this); }
+
+void WindowedInt::take_sample() {
+  int64_t v = src_();
+  std::lock_guard<std::mutex> lk(mu_);
+  ring_[nsamples_ % 64] = v;
+  latest_ = v;
+  ++nsamples_;
+}
+
+int64_t WindowedInt::get_value() const {
+  std::lock_guard<std::mutex> lk(mu_);
+  if (nsamples_ == 0) return 0;
+  int w = std::min(window_, nsamples_ - 1);
+  if (w <= 0) {
+    return per_second_ ? latest_ : latest_;  // best effort before 2 samples
+  }
+  int64_t old = ring_[(nsamples_ - 1 - w) % 64];
+  int64_t diff = latest_ - old;
+  return per_second_ ? diff / w : diff;
+}
+
+// ---------------- LatencyRecorder ----------------
+
+LatencyRecorder::LatencyRecorder() : ring_(kRingSize) {
+  qps_window_.reset(new WindowedInt([this] { return count_.get_value(); }, 1, true));
+}
+
+LatencyRecorder::LatencyRecorder(const std::string& prefix) : LatencyRecorder() {
+  expose(prefix);
+}
+
+LatencyRecorder::~LatencyRecorder() {
+  for (Variable* v : exposed_) delete v;
+}
+
+LatencyRecorder& LatencyRecorder::operator<<(int64_t latency_us) {
+  count_ << 1;
+  sum_us_ << latency_us;
+  int64_t cur = window_max_.load(std::memory_order_relaxed);
+  while (latency_us > cur &&
+         !window_max_.compare_exchange_weak(cur, latency_us, std::memory_order_relaxed)) {
+  }
+  size_t i = (size_t)ring_idx_.fetch_add(1, std::memory_order_relaxed) % kRingSize;
+  ring_[i].store((uint32_t)std::min<int64_t>(latency_us, UINT32_MAX),
+                 std::memory_order_relaxed);
+  return *this;
+}
+
+int64_t LatencyRecorder::qps() const { return qps_window_->get_value(); }
+
+int64_t LatencyRecorder::latency_avg() const {
+  int64_t c = count_.get_value();
+  return c > 0 ? sum_us_.get_value() / c : 0;
+}
+
+int64_t LatencyRecorder::latency_percentile(double p) const {
+  size_t n = std::min<size_t>((size_t)ring_idx_.load(std::memory_order_relaxed), kRingSize);
+  if (n == 0) return 0;
+  std::vector<uint32_t> copy(n);
+  for (size_t i = 0; i < n; ++i) copy[i] = ring_[i].load(std::memory_order_relaxed);
+  std::sort(copy.begin(), copy.end());
+  size_t idx = (size_t)(p * (n - 1));
+  return copy[idx];
+}
+
+void LatencyRecorder::expose(const std::string& prefix) {
+  prefix_ = prefix;
+  exposed_.push_back(new PassiveStatus(
+      prefix + "_latency", [this] { return std::to_string(latency_avg()); }));
+  exposed_.push_back(new PassiveStatus(
+      prefix + "_latency_9999", [this] { return std::to_string(latency_percentile(0.9999)); }));
+  exposed_.push_back(new PassiveStatus(
+      prefix + "_latency_99", [this] { return std::to_string(latency_percentile(0.99)); }));
+  exposed_.push_back(new PassiveStatus(
+      prefix + "_latency_50", [this] { return std::to_string(latency_percentile(0.50)); }));
+  exposed_.push_back(
+      new PassiveStatus(prefix + "_qps", [this] { return std::to_string(qps()); }));
+  exposed_.push_back(
+      new PassiveStatus(prefix + "_count", [this] { return std::to_string(count()); }));
+  for (size_t i = 0; i < exposed_.size(); ++i) {
+    // names set in PassiveStatus ctor via expose? No: expose explicitly
+  }
+  // PassiveStatus above were constructed without names; expose them now.
+  const char* suffixes[] = {"_latency", "_latency_9999", "_latency_99",
+                            "_latency_50", "_qps", "_count"};
+  for (size_t i = 0; i < exposed_.size() && i < 6; ++i) {
+    exposed_[i]->expose(prefix + suffixes[i]);
+  }
+}
+
+}  // namespace var
+}  // namespace bam

@@ -1,0 +1,285 @@
+// brpc_amd: var — lock-cheap metrics (≙ reference bvar L2).
+// Variable base + global name registry; Adder/Maxer/Miner reducers with
+// per-thread cells (writes are a relaxed atomic op on a thread-local cell,
+// no contention); Window/PerSecond via a 1 Hz sampler thread;
+// LatencyRecorder = qps + avg + max + percentiles. Parity: bvar/variable.h,
+// bvar/reducer.h, bvar/window.h, bvar/latency_recorder.h.
+#pragma once
+
+#include <stdint.h>
+
+#include <atomic>
+#include <functional>
+#include <limits>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <sstream>
+#include <string>
+#include <type_traits>
+#include <vector>
+
+namespace bam {
+namespace var {
+
+class Variable {
+ public:
+  virtual ~Variable();
+  virtual void describe(std::ostream& os) const = 0;
+  std::string get_description() const {
+    std::ostringstream os;
+    describe(os);
+    return os.str();
+  }
+  // Registers under `name` (replaces previous owner of the name).
+  int expose(const std::string& name);
+  void hide();
+  const std::string& name() const { return name_; }
+
+  static void dump_exposed(std::ostream& os, const std::string& filter = "");
+  static Variable* find_exposed(const std::string& name);
+  static size_t count_exposed();
+
+ protected:
+  std::string name_;
+};
+
+// ---------------- reducers ----------------
+
+namespace detail {
+
+// Per-thread cell list; read = combine over all cells (+ sum of cells from
+// dead threads). Writes touch only the caller's cell.
+template <typename T, typename Op>
+class AgentCombiner {
+ public:
+  struct Cell {
+    std::atomic<T> value{T()};
+  };
+
+  T combine(T identity) const {
+    T acc = identity;
+    std::lock_guard<std::mutex> lk(mu_);
+    acc = Op()(acc, terminated_);
+    for (const auto& c : cells_) acc = Op()(acc, c->value.load(std::memory_order_relaxed));
+    return acc;
+  }
+
+  void reset_all(T identity) {
+    std::lock_guard<std::mutex> lk(mu_);
+    terminated_ = identity;
+    for (auto& c : cells_) c->value.store(identity, std::memory_order_relaxed);
+  }
+
+  Cell* local_cell() {
+    // tls map: combiner -> cell
+    static thread_local std::map<const void*, Cell*> tls;
+    auto it = tls.find(this);
+    if (it != tls.end()) return it->second;
+    auto cell = std::make_shared<Cell>();
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      cells_.push_back(cell);
+    }
+    tls[this] = cell.get();
+    return cell.get();
+  }
+
+ private:
+  mutable std::mutex mu_;
+  std::vector<std::shared_ptr<Cell>> cells_;
+  T terminated_ = T();
+};
+
+struct AddOp {
+  template <typename T>
+  T operator()(T a, T b) const {
+    return a + b;
+  }
+};
+struct MaxOp {
+  template <typename T>
+  T operator()(T a, T b) const {
+    return a > b ? a : b;
+  }
+};
+struct MinOp {
+  template <typename T>
+  T operator()(T a, T b) const {
+    return a < b ? a : b;
+  }
+};
+
+}  // namespace detail
+
+template <typename T>
+class Adder : public Variable {
+ public:
+  Adder() {}
+  explicit Adder(const std::string& name) { expose(name); }
+  ~Adder() override { hide(); }
+
+  Adder& operator<<(T v) {
+    auto* cell = combiner_.local_cell();
+    // fetch_add for integral; CAS loop otherwise
+    atomic_add(cell->value, v);
+    return *this;
+  }
+  T get_value() const { return combiner_.combine(T()); }
+  void reset() { combiner_.reset_all(T()); }
+  void describe(std::ostream& os) const override { os << get_value(); }
+
+ private:
+  static void atomic_add(std::atomic<T>& a, T v) {
+    if constexpr (std::is_integral<T>::value) {
+      a.fetch_add(v, std::memory_order_relaxed);
+    } else {
+      T cur = a.load(std::memory_order_relaxed);
+      while (!a.compare_exchange_weak(cur, cur + v, std::memory_order_relaxed)) {
+      }
+    }
+  }
+  detail::AgentCombiner<T, detail::AddOp> combiner_;
+};
+
+template <typename T>
+class Maxer : public Variable {
+ public:
+  Maxer() {}
+  explicit Maxer(const std::string& name) { expose(name); }
+  ~Maxer() override { hide(); }
+  Maxer& operator<<(T v) {
+    auto* cell = combiner_.local_cell();
+    T cur = cell->value.load(std::memory_order_relaxed);
+    while (v > cur &&
+           !cell->value.compare_exchange_weak(cur, v, std::memory_order_relaxed)) {
+    }
+    return *this;
+  }
+  T get_value() const { return combiner_.combine(std::numeric_limits<T>::lowest()); }
+  void reset() { combiner_.reset_all(std::numeric_limits<T>::lowest()); }
+  void describe(std::ostream& os) const override { os << get_value(); }
+
+ private:
+  detail::AgentCombiner<T, detail::MaxOp> combiner_;
+};
+
+template <typename T>
+class Miner : public Variable {
+ public:
+  Miner() {}
+  explicit Miner(const std::string& name) { expose(name); }
+  ~Miner() override { hide(); }
+  Miner& operator<<(T v) {
+    auto* cell = combiner_.local_cell();
+    T cur = cell->value.load(std::memory_order_relaxed);
+    while (v < cur &&
+           !cell->value.compare_exchange_weak(cur, v, std::memory_order_relaxed)) {
+    }
+    return *this;
+  }
+  T get_value() const { return combiner_.combine(std::numeric_limits<T>::max()); }
+  void describe(std::ostream& os) const override { os << get_value(); }
+
+ private:
+  detail::AgentCombiner<T, detail::MinOp> combiner_;
+};
+
+// ---------------- status ----------------
+
+template <typename T>
+class Status : public Variable {
+ public:
+  explicit Status(T v = T()) : value_(v) {}
+  Status(const std::string& name, T v) : value_(v) { expose(name); }
+  ~Status() override { hide(); }
+  void set_value(T v) {
+    std::lock_guard<std::mutex> lk(mu_);
+    value_ = v;
+  }
+  T get_value() const {
+    std::lock_guard<std::mutex> lk(mu_);
+    return value_;
+  }
+  void describe(std::ostream& os) const override { os << get_value(); }
+
+ private:
+  mutable std::mutex mu_;
+  T value_;
+};
+
+class PassiveStatus : public Variable {
+ public:
+  typedef std::function<std::string()> Fn;
+  explicit PassiveStatus(Fn fn) : fn_(std::move(fn)) {}
+  PassiveStatus(const std::string& name, Fn fn) : fn_(std::move(fn)) { expose(name); }
+  ~PassiveStatus() override { hide(); }
+  void describe(std::ostream& os) const override { os << fn_(); }
+
+ private:
+  Fn fn_;
+};
+
+// ---------------- windowed values (1 Hz sampler) ----------------
+
+// Samples an int64 source once per second into a 61-slot ring; value(w) =
+// latest - sample[w seconds ago]; per_second(w) = value(w)/w.
+class WindowedInt : public Variable {
+ public:
+  typedef std::function<int64_t()> SourceFn;
+  WindowedInt(SourceFn src, int window_sec, bool per_second);
+  ~WindowedInt() override;
+  int64_t get_value() const;
+  void describe(std::ostream& os) const override { os << get_value(); }
+  void take_sample();  // called by the sampler thread
+
+ private:
+  SourceFn src_;
+  int window_;
+  bool per_second_;
+  mutable std::mutex mu_;
+  int64_t ring_[64];
+  int64_t latest_;
+  int nsamples_;
+};
+
+// Registers fn to run at 1 Hz on the global sampler thread.
+void register_sampler(WindowedInt* w);
+void unregister_sampler(WindowedInt* w);
+
+// ---------------- latency recorder ----------------
+
+// qps + avg + max + p50/p90/p99/p999 over a sliding sample reservoir.
+// (Percentiles: ring of the most recent 8192 samples — an approximation in
+// the same spirit as the reference's per-interval sampling.)
+class LatencyRecorder {
+ public:
+  LatencyRecorder();
+  explicit LatencyRecorder(const std::string& prefix);
+  ~LatencyRecorder();
+
+  LatencyRecorder& operator<<(int64_t latency_us);
+
+  int64_t count() const { return count_.get_value(); }
+  int64_t qps() const;  // over the last 1s window
+  int64_t latency_avg() const;
+  int64_t latency_max() const { return window_max_.load(std::memory_order_relaxed); }
+  int64_t latency_percentile(double p) const;
+
+  void expose(const std::string& prefix);
+
+ private:
+  friend class LatencyDumper;
+  Adder<int64_t> count_;
+  Adder<int64_t> sum_us_;
+  std::atomic<int64_t> window_max_{0};
+  static const size_t kRingSize = 8192;
+  std::atomic<int64_t> ring_idx_{0};
+  std::vector<std::atomic<uint32_t>> ring_;
+  std::unique_ptr<WindowedInt> qps_window_;
+  std::vector<Variable*> exposed_;
+  std::string prefix_;
+};
+
+}  // namespace var
+}  // namespace bam

@@ -1,0 +1,25 @@
+"""var layer tests (≙ reference bvar unittests)."""
+import brpc_amd as b
+
+v = b.core.var
+r = b.core.rpc
+
+
+def test_adder_across_fibers():
+    assert v.adder_selftest(16, 1000) == 16_000
+
+
+def test_latency_recorder():
+    assert v.latency_recorder_selftest()
+
+
+def test_server_method_status_exposed():
+    port = r.start_echo_server(0)
+    addr = f"127.0.0.1:{port}"
+    for _ in range(10):
+        rc, _, _ = r.echo_once(addr, b"x" * 100, 2000)
+        assert rc == 0
+    dump = v.dump_exposed(f"rpc_server_{port}_EchoService.Echo")
+    assert "_count" in dump
+    count = v.describe(f"rpc_server_{port}_EchoService.Echo_count")
+    assert count is not None and int(count) >= 10
