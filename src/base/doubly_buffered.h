@@ -36,7 +36,10 @@ class DoublyBufferedData {
     std::mutex* lock_;
   };
 
-  DoublyBufferedData() : index_(0) {}
+  DoublyBufferedData() : index_(0) {
+    static std::atomic<uint64_t> next_id{1};
+    id_ = next_id.fetch_add(1, std::memory_order_relaxed);
+  }
 
   // Returns 0 on success. Holds the calling thread's wrapper lock until
   // `ptr` goes out of scope.
@@ -74,19 +77,24 @@ class DoublyBufferedData {
   };
 
   Wrapper* local_wrapper() {
-    static thread_local std::unordered_map<DoublyBufferedData*, Wrapper*> tls_map;
-    auto it = tls_map.find(this);
-    if (it != tls_map.end()) return it->second;
+    // Keyed by a process-unique instance id (NOT `this`): a destroyed
+    // instance's address can be reused, and handing back the old wrapper
+    // would lock a destroyed mutex. The TLS map holds shared_ptrs so stale
+    // entries stay valid memory until the thread exits.
+    static thread_local std::unordered_map<uint64_t, std::shared_ptr<Wrapper>> tls_map;
+    auto it = tls_map.find(id_);
+    if (it != tls_map.end()) return it->second.get();
     auto w = std::make_shared<Wrapper>();
     {
       std::lock_guard<std::mutex> lk(wrappers_mu_);
       wrappers_.push_back(w);
     }
-    tls_map[this] = w.get();
+    tls_map[id_] = w;
     return w.get();
   }
 
   T data_[2];
+  uint64_t id_;
   std::atomic<int> index_;
   std::mutex modify_mu_;
   std::mutex wrappers_mu_;

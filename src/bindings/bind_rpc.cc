@@ -89,3 +89,54 @@ void bind_rpc(py::module_& m) {
         py::arg("total") = 10000, py::arg("timeout_ms") = 5000,
         py::arg("method") = "EchoService.Echo", py::arg("hbm_request") = false);
 }
+
+// ---- combo channels & LBs ----
+namespace bam {
+namespace rpctest {
+int parallel_echo_test(const std::vector<int>& ports, const std::string& payload,
+                       int fail_limit, std::string* merged, std::string* err);
+int selective_test(int dead_port, int live_port, std::string* resp_out);
+int partition_test(const std::vector<int>& ports, std::string* merged);
+int lb_spread_test(const std::string& lb_name, const std::vector<int>& ports, int ncalls);
+}  // namespace rpctest
+}  // namespace bam
+
+void bind_rpc_combo(py::module_& m) {
+  auto r = m.def_submodule("combo");
+  r.def("parallel_echo",
+        [](const std::vector<int>& ports, py::bytes payload, int fail_limit) {
+          char* ptr;
+          Py_ssize_t len;
+          PyBytes_AsStringAndSize(payload.ptr(), &ptr, &len);
+          std::string p(ptr, len), merged, err;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::parallel_echo_test(ports, p, fail_limit, &merged, &err);
+          }
+          return py::make_tuple(rc, py::bytes(merged), err);
+        },
+        py::arg("ports"), py::arg("payload"), py::arg("fail_limit") = -1);
+  r.def("selective",
+        [](int dead_port, int live_port) {
+          std::string resp;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::selective_test(dead_port, live_port, &resp);
+          }
+          return py::make_tuple(rc, py::bytes(resp));
+        });
+  r.def("partition",
+        [](const std::vector<int>& ports) {
+          std::string merged;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::partition_test(ports, &merged);
+          }
+          return py::make_tuple(rc, py::bytes(merged));
+        });
+  r.def("lb_spread", &bam::rpctest::lb_spread_test,
+        py::call_guard<py::gil_scoped_release>());
+}

@@ -7,4 +7,5 @@ PYBIND11_MODULE(_core, m) {
   bind_rpc(m);
   bind_var(m);
   bind_gpu(m);
+  bind_rpc_combo(m);
 }

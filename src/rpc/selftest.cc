@@ -39,6 +39,10 @@ Service* NewEchoService() {
     cntl->SetFailed(EINTERNAL, "you asked for it");
     done->Run();
   });
+  svc->AddMethod("Port", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    resp->append(std::to_string(cntl->server_->listen_address().port));
+    done->Run();
+  });
   // BASELINE config 2 (IOBuf-in-HBM echo): the response payload is staged
   // into HBM-resident blocks; the socket write path stages it back through
   // the GPU byte mover. Falls back with an explicit error if no GPU.
