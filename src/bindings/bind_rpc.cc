@@ -140,3 +140,30 @@ void bind_rpc_combo(py::module_& m) {
   r.def("lb_spread", &bam::rpctest::lb_spread_test,
         py::call_guard<py::gil_scoped_release>());
 }
+
+// ---- streaming ----
+namespace bam {
+namespace rpctest {
+int start_stream_echo_server();
+int stream_echo_test(int port, int nframes, int frame_size, std::string* err);
+double stream_throughput_test(int port, int nframes, int frame_size);
+}  // namespace rpctest
+}  // namespace bam
+
+void bind_rpc_stream(py::module_& m) {
+  auto s = m.def_submodule("stream");
+  s.def("start_server", &bam::rpctest::start_stream_echo_server,
+        py::call_guard<py::gil_scoped_release>());
+  s.def("echo_test",
+        [](int port, int nframes, int frame_size) {
+          std::string err;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::stream_echo_test(port, nframes, frame_size, &err);
+          }
+          return py::make_tuple(rc, err);
+        });
+  s.def("throughput", &bam::rpctest::stream_throughput_test,
+        py::call_guard<py::gil_scoped_release>());
+}

@@ -8,4 +8,5 @@ PYBIND11_MODULE(_core, m) {
   bind_var(m);
   bind_gpu(m);
   bind_rpc_combo(m);
+  bind_rpc_stream(m);
 }

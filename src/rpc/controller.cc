@@ -19,6 +19,8 @@ void Controller::Reset() {
   server_ = nullptr;
   server_socket_ = 0;
   server_cid_ = 0;
+  remote_stream_id_ = 0;
+  response_stream_id_ = 0;
   call = Call();
 }
 

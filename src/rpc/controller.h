@@ -83,12 +83,15 @@ class Controller {
     SocketId pending_socket = 0;    // socket holding the pending session
     class LoadBalancer* lb = nullptr;
     class SubChannelCtx* sub_ctx = nullptr;  // ParallelChannel bookkeeping
+    uint64_t stream_id = 0;                  // client-created stream (StreamCreate)
   };
   Call call;
 
   // server-side context
   Server* server_ = nullptr;
   SocketId server_socket_ = 0;
+  uint64_t remote_stream_id_ = 0;    // stream id carried by the peer's meta
+  uint64_t response_stream_id_ = 0;  // stream accepted by this server (StreamAccept)
   int64_t server_cid_ = 0;  // correlation id to echo back
   CompressType response_compress_ = COMPRESS_TYPE_NONE;
   void set_response_compress_type(CompressType t) { response_compress_ = t; }

@@ -7,6 +7,7 @@
 #include "rpc/channel.h"
 #include "rpc/controller.h"
 #include "rpc/server.h"
+#include "rpc/stream.h"
 #include "rpc/wire.h"
 
 namespace bam {
@@ -37,6 +38,11 @@ void SerializeRpcMeta(const RpcMeta& meta, std::string* out) {
   if (meta.compress_type != 0) wire::put_int_field(out, 3, meta.compress_type);
   wire::put_int_field(out, 4, meta.correlation_id);
   if (meta.attachment_size != 0) wire::put_int_field(out, 5, meta.attachment_size);
+  if (meta.stream_id != 0) {
+    std::string sub;
+    wire::put_int_field(&sub, 1, (int64_t)meta.stream_id);
+    wire::put_msg_field(out, 7, sub);
+  }
 }
 
 bool ParseRpcMeta(const char* data, size_t n, RpcMeta* out) {
@@ -82,6 +88,18 @@ bool ParseRpcMeta(const char* data, size_t n, RpcMeta* out) {
       case 5:
         out->attachment_size = (int32_t)r.varint();
         break;
+      case 7: {
+        std::string sub = r.read_string();
+        if (!r.ok()) return false;
+        wire::Reader rr(sub.data(), sub.size());
+        int wt2;
+        for (int f2; (f2 = rr.read_tag(&wt2)) != 0;) {
+          if (f2 == 1) out->stream_id = rr.varint();
+          else rr.skip(wt2);
+          if (!rr.ok()) return false;
+        }
+        break;
+      }
       default:
         r.skip(wtype);
     }
@@ -130,6 +148,7 @@ void PackStdRequest(IOBuf* out, Controller* cntl, SessionId correlation_id) {
   meta.compress_type = (int)cntl->request_compress_type();
   meta.correlation_id = (int64_t)correlation_id;
   meta.attachment_size = (int32_t)cntl->request_attachment().size();
+  meta.stream_id = cntl->call.stream_id;
   std::string meta_bytes;
   SerializeRpcMeta(meta, &meta_bytes);
   size_t body = meta_bytes.size() + cntl->call.request_buf.size() +
@@ -146,13 +165,14 @@ void PackStdRequest(IOBuf* out, Controller* cntl, SessionId correlation_id) {
 
 static void PackStdResponse(IOBuf* out, int64_t correlation_id, int error_code,
                             const std::string& error_text, const IOBuf& payload,
-                            const IOBuf& attachment) {
+                            const IOBuf& attachment, uint64_t stream_id) {
   RpcMeta meta;
   meta.has_response = true;
   meta.error_code = error_code;
   meta.error_text = error_text;
   meta.correlation_id = correlation_id;
   meta.attachment_size = (int32_t)attachment.size();
+  meta.stream_id = stream_id;
   std::string meta_bytes;
   SerializeRpcMeta(meta, &meta_bytes);
   size_t body = meta_bytes.size() + payload.size() + attachment.size();
@@ -174,7 +194,7 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
   if (Socket::Address(sid, &sock) == 0) {
     IOBuf packet;
     PackStdResponse(&packet, cid, cntl->ErrorCode(), cntl->ErrorText(), *resp,
-                    cntl->response_attachment());
+                    cntl->response_attachment(), cntl->response_stream_id_);
     sock->Write(&packet);
   }
   if (cntl->server_ != nullptr) {
@@ -199,6 +219,7 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   cntl->server_socket_ = sock->id();
   cntl->server_cid_ = msg->meta.correlation_id;
   cntl->log_id_ = msg->meta.log_id;
+  cntl->remote_stream_id_ = msg->meta.stream_id;
   cntl->remote_side_ = sock->remote_side();
   cntl->local_side_ = sock->local_side();
   IOBuf* resp = new IOBuf;
@@ -266,6 +287,10 @@ static void ProcessStdResponse(InputMessageBase* msg_base) {
   if (msg->meta.error_code != 0) {
     cntl->SetFailed(msg->meta.error_code, msg->meta.error_text);
   } else {
+    if (msg->meta.stream_id != 0 && cntl->call.stream_id != 0) {
+      stream_internal::ConnectLocalStream(cntl->call.stream_id, msg->meta.stream_id,
+                                          msg->socket_id);
+    }
     size_t att = (size_t)msg->meta.attachment_size;
     size_t data_len = msg->payload.size() >= att ? msg->payload.size() - att : 0;
     if (cntl->call.response != nullptr) {
@@ -292,6 +317,7 @@ void RegisterStdProtocol() {
     p.support_client = true;
     p.name = "std";
     RegisterProtocol(p);
+    stream_internal::RegisterStreamProtocol();
   });
 }
 

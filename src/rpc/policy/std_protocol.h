@@ -29,6 +29,7 @@ struct RpcMeta {
   int compress_type = 0;
   int64_t correlation_id = 0;
   int32_t attachment_size = 0;
+  uint64_t stream_id = 0;  // StreamSettings (field 7 sub-message)
 };
 
 void SerializeRpcMeta(const RpcMeta& meta, std::string* out);
