@@ -1,0 +1,204 @@
+// brpc_amd: builtin HTTP services.
+// Parity: reference brpc/builtin/ (27 services; registration at
+// server.cpp:501-603). This file serves the core set: /index, /status,
+// /vars, /flags, /health, /version, /connections, /protobufs, /fibers
+// (≙ /bthreads), /memory, /brpc_metrics (Prometheus), /rpcz (spans).
+#include <sstream>
+
+#include "base/flags.h"
+#include "base/gpu_loader.h"
+#include "base/iobuf.h"
+#include "base/time.h"
+#include "fiber/fiber.h"
+#include "rpc/policy/http_protocol.h"
+#include "rpc/rpcz.h"
+#include "rpc/server.h"
+#include "rpc/socket.h"
+#include "var/variable.h"
+
+namespace bam {
+namespace policy {
+
+namespace {
+
+int64_t g_start_time_us = monotonic_time_us();
+
+void page_index(HttpResponse* resp) {
+  resp->content_type = "text/html";
+  std::ostringstream os;
+  os << "<html><head><title>brpc_amd</title></head><body><h1>brpc_amd server</h1><ul>";
+  const char* pages[] = {"status", "vars",   "flags",  "health",       "version",
+                         "connections", "protobufs", "fibers", "memory", "rpcz",
+                         "brpc_metrics"};
+  for (const char* p : pages) os << "<li><a href=\"/" << p << "\">/" << p << "</a></li>";
+  os << "</ul></body></html>";
+  resp->body.append(os.str());
+}
+
+void page_status(Server* server, HttpResponse* resp) {
+  std::ostringstream os;
+  os << "version: brpc_amd/1.0 (MI355X-native)\n";
+  os << "uptime_s: " << (monotonic_time_us() - g_start_time_us) / 1000000 << "\n";
+  os << "fiber_workers: " << fiber_get_concurrency() << "\n";
+  os << "fibers_created: " << fiber_count_created() << "\n";
+  os << "fibers_active: " << fiber_count_active() << "\n";
+  os << "gpu_devices: " << gpu::device_count() << "\n";
+  if (server != nullptr) {
+    os << "listen: " << endpoint2str(server->listen_address()) << "\n";
+    os << "processed_requests: " << server->nprocessed.load() << "\n";
+    os << "concurrency: " << server->concurrency.load() << "\n";
+    for (const auto& kv : server->services()) {
+      os << "service: " << kv.first << "\n";
+      for (const auto& m : kv.second->methods()) {
+        var::LatencyRecorder* rec = server->method_status(kv.first, m.first);
+        os << "  " << m.first << "  count=" << rec->count() << " qps=" << rec->qps()
+           << " latency_avg_us=" << rec->latency_avg()
+           << " p99_us=" << rec->latency_percentile(0.99) << "\n";
+      }
+    }
+  }
+  resp->body.append(os.str());
+}
+
+void page_vars(const HttpRequest& req, HttpResponse* resp) {
+  std::string filter;
+  auto it = req.query.find("filter");
+  if (it != req.query.end()) filter = it->second;
+  // /vars/some_prefix form
+  if (req.path.size() > 6) filter = req.path.substr(6);
+  std::ostringstream os;
+  var::Variable::dump_exposed(os, filter);
+  resp->body.append(os.str());
+}
+
+void page_flags(const HttpRequest& req, HttpResponse* resp) {
+  // /flags/NAME?setvalue=V modifies (parity: reloadable gflags at /flags)
+  if (req.path.size() > 7) {
+    std::string name = req.path.substr(7);
+    auto it = req.query.find("setvalue");
+    if (it != req.query.end()) {
+      int rc = flags::SetFlagValue(name, it->second);
+      if (rc == 0) {
+        resp->body.append("set " + name + "=" + it->second + "\n");
+      } else {
+        resp->status = rc == -1 ? 404 : 403;
+        resp->body.append(rc == -1 ? "unknown flag\n" : "validation failed\n");
+      }
+      return;
+    }
+    resp->body.append(name + " : " + flags::GetFlagValue(name) + "\n");
+    return;
+  }
+  std::vector<flags::FlagInfo> all;
+  flags::ListFlags(&all);
+  std::ostringstream os;
+  for (const auto& f : all) {
+    os << f.name << " : " << flags::GetFlagValue(f.name) << "  (default: " << f.default_value
+       << ")  # " << f.description << "\n";
+  }
+  resp->body.append(os.str());
+}
+
+void page_connections(HttpResponse* resp) {
+  std::vector<SocketId> ids;
+  ListSockets(&ids);
+  std::ostringstream os;
+  os << "socket_count: " << ids.size() << "\n";
+  os << "id | remote | in_bytes | out_bytes | in_msgs | out_msgs\n";
+  for (SocketId id : ids) {
+    SocketUniquePtr s;
+    if (Socket::Address(id, &s) != 0) continue;
+    os << id << " | " << endpoint2str(s->remote_side()) << " | " << s->in_bytes.load()
+       << " | " << s->out_bytes.load() << " | " << s->in_messages.load() << " | "
+       << s->out_messages.load() << "\n";
+  }
+  resp->body.append(os.str());
+}
+
+void page_protobufs(Server* server, HttpResponse* resp) {
+  std::ostringstream os;
+  if (server != nullptr) {
+    for (const auto& kv : server->services()) {
+      for (const auto& m : kv.second->methods()) {
+        os << kv.first << "." << m.first << "\n";
+      }
+    }
+  }
+  resp->body.append(os.str());
+}
+
+void page_fibers(HttpResponse* resp) {
+  std::ostringstream os;
+  os << "workers: " << fiber_get_concurrency() << "\n";
+  os << "fibers_created: " << fiber_count_created() << "\n";
+  os << "fibers_active: " << fiber_count_active() << "\n";
+  resp->body.append(os.str());
+}
+
+void page_memory(HttpResponse* resp) {
+  std::ostringstream os;
+  os << "iobuf_block_count: " << IOBuf::block_count() << "\n";
+  os << "iobuf_block_memory: " << IOBuf::block_memory() << "\n";
+  resp->body.append(os.str());
+}
+
+void page_metrics(HttpResponse* resp) {
+  // Prometheus exposition format from the var registry.
+  std::ostringstream raw;
+  var::Variable::dump_exposed(raw, "");
+  std::istringstream in(raw.str());
+  std::ostringstream os;
+  std::string line;
+  while (std::getline(in, line)) {
+    size_t sep = line.find(" : ");
+    if (sep == std::string::npos) continue;
+    std::string name = line.substr(0, sep);
+    std::string value = line.substr(sep + 3);
+    // prometheus names: [a-zA-Z0-9_]
+    for (char& c : name)
+      if (!isalnum((unsigned char)c)) c = '_';
+    // only numeric values are representable
+    char* end = nullptr;
+    strtod(value.c_str(), &end);
+    if (end == value.c_str()) continue;
+    os << "# TYPE " << name << " gauge\n" << name << " " << value << "\n";
+  }
+  resp->body.append(os.str());
+}
+
+}  // namespace
+
+bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse* resp) {
+  const std::string& p = req.path;
+  if (p == "/" || p == "/index") {
+    page_index(resp);
+  } else if (p == "/health") {
+    resp->body.append("OK\n");
+  } else if (p == "/status") {
+    page_status(server, resp);
+  } else if (p == "/vars" || p.rfind("/vars/", 0) == 0) {
+    page_vars(req, resp);
+  } else if (p == "/flags" || p.rfind("/flags/", 0) == 0) {
+    page_flags(req, resp);
+  } else if (p == "/version") {
+    resp->body.append("brpc_amd/1.0 gfx950\n");
+  } else if (p == "/connections") {
+    page_connections(resp);
+  } else if (p == "/protobufs") {
+    page_protobufs(server, resp);
+  } else if (p == "/fibers" || p == "/bthreads") {
+    page_fibers(resp);
+  } else if (p == "/memory") {
+    page_memory(resp);
+  } else if (p == "/brpc_metrics" || p == "/metrics") {
+    page_metrics(resp);
+  } else if (p == "/rpcz") {
+    rpcz::DumpRecentSpans(&resp->body, req.query.count("verbose") != 0);
+  } else {
+    return false;
+  }
+  return true;
+}
+
+}  // namespace policy
+}  // namespace bam

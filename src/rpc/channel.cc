@@ -6,6 +6,7 @@
 #include "base/time.h"
 #include "rpc/load_balancer.h"
 #include "rpc/policy/std_protocol.h"
+#include "rpc/rpcz.h"
 #include "rpc/socket_map.h"
 
 namespace bam {
@@ -56,6 +57,18 @@ static void split_full_method(const std::string& full, std::string* service,
 // session and runs done.
 void EndRPC(Controller* cntl, SessionId locked_id) {
   cntl->end_us_ = monotonic_time_us();
+  if (rpcz::enabled()) {
+    rpcz::Span span;
+    span.start_us = cntl->start_us_;
+    span.end_us = cntl->end_us_;
+    span.full_method = cntl->call.service_name + "." + cntl->call.method_name;
+    span.remote = cntl->remote_side_;
+    span.error_code = cntl->error_code_;
+    span.log_id = cntl->log_id_;
+    span.server_side = false;
+    span.request_size = cntl->call.request_buf.size();
+    rpcz::RecordSpan(span);
+  }
   if (cntl->call.timeout_timer != 0) {
     timer_delete(cntl->call.timeout_timer);
     cntl->call.timeout_timer = 0;

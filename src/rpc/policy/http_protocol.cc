@@ -1,0 +1,227 @@
+// brpc_amd: HTTP/1.1 server-side protocol.
+// Parity: reference policy/http_rpc_protocol.cpp + details/http_parser —
+// serves (a) the builtin service pages (/status /vars /flags /health
+// /connections /index ... see rpc/builtin/) and (b) RPC-over-HTTP:
+// POST/GET /ServiceName/MethodName with the body as request payload.
+// Hand-written incremental parser (request line + headers + content-length
+// body); keep-alive + pipelining via the standard parse loop.
+#include "rpc/policy/http_protocol.h"
+
+#include <unistd.h>
+
+#include <algorithm>
+#include <mutex>
+
+#include "base/flat_map.h"
+#include "base/logging.h"
+#include "base/time.h"
+#include "rpc/controller.h"
+#include "rpc/server.h"
+#include "rpc/wire.h"
+
+namespace bam {
+namespace policy {
+
+namespace {
+
+struct HttpMessage : public InputMessageBase {
+  HttpRequest req;
+};
+
+// Finds "\r\n\r\n" within the first `limit` bytes of buf. Returns offset of
+// header end (past the blank line) or npos.
+size_t find_header_end(const std::string& head) {
+  size_t p = head.find("\r\n\r\n");
+  return p == std::string::npos ? std::string::npos : p + 4;
+}
+
+bool parse_headers(const std::string& head, HttpRequest* out) {
+  size_t line_end = head.find("\r\n");
+  if (line_end == std::string::npos) return false;
+  // request line: METHOD SP PATH SP VERSION
+  const std::string line = head.substr(0, line_end);
+  size_t sp1 = line.find(' ');
+  size_t sp2 = line.rfind(' ');
+  if (sp1 == std::string::npos || sp2 == sp1) return false;
+  out->method = line.substr(0, sp1);
+  std::string uri = line.substr(sp1 + 1, sp2 - sp1 - 1);
+  size_t q = uri.find('?');
+  if (q == std::string::npos) {
+    out->path = uri;
+  } else {
+    out->path = uri.substr(0, q);
+    std::string query = uri.substr(q + 1);
+    size_t pos = 0;
+    while (pos < query.size()) {
+      size_t amp = query.find('&', pos);
+      std::string kv = query.substr(pos, amp == std::string::npos ? amp : amp - pos);
+      size_t eq = kv.find('=');
+      if (eq == std::string::npos) {
+        out->query[kv] = "";
+      } else {
+        out->query[kv.substr(0, eq)] = kv.substr(eq + 1);
+      }
+      if (amp == std::string::npos) break;
+      pos = amp + 1;
+    }
+  }
+  size_t pos = line_end + 2;
+  while (pos < head.size()) {
+    size_t eol = head.find("\r\n", pos);
+    if (eol == std::string::npos || eol == pos) break;
+    std::string hline = head.substr(pos, eol - pos);
+    size_t colon = hline.find(':');
+    if (colon != std::string::npos) {
+      std::string key = hline.substr(0, colon);
+      size_t vstart = hline.find_first_not_of(' ', colon + 1);
+      std::string val = vstart == std::string::npos ? "" : hline.substr(vstart);
+      // lower-case keys for lookup
+      std::transform(key.begin(), key.end(), key.begin(), ::tolower);
+      out->headers[key] = val;
+    }
+    pos = eol + 2;
+  }
+  return true;
+}
+
+ParseResult ParseHttpMessage(IOBuf* source, Socket* /*sock*/, bool /*eof*/) {
+  // Cheap probe for an HTTP method prefix.
+  char probe[8];
+  size_t n = std::min<size_t>(source->size(), 8);
+  if (n < 4) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  const char* p = (const char*)source->fetch(probe, n);
+  static const char* kMethods[] = {"GET ", "POST", "PUT ", "DELE", "HEAD", "OPTI", "PATC"};
+  bool maybe = false;
+  for (const char* m : kMethods) {
+    if (memcmp(p, m, 4) == 0) {
+      maybe = true;
+      break;
+    }
+  }
+  if (!maybe) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  // Copy up to 64 KiB looking for header end.
+  std::string head;
+  size_t scan = std::min<size_t>(source->size(), 64 << 10);
+  source->copy_to(&head, scan, 0);
+  size_t hend = find_header_end(head);
+  if (hend == std::string::npos) {
+    if (scan >= (64 << 10)) return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+    return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  }
+  HttpMessage* msg = new HttpMessage;
+  if (!parse_headers(head.substr(0, hend), &msg->req)) {
+    delete msg;
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  }
+  size_t content_len = 0;
+  auto it = msg->req.headers.find("content-length");
+  if (it != msg->req.headers.end()) content_len = (size_t)atoll(it->second.c_str());
+  if (source->size() < hend + content_len) {
+    delete msg;
+    return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  }
+  source->pop_front(hend);
+  source->cutn(&msg->req.body, content_len);
+  auto conn = msg->req.headers.find("connection");
+  msg->req.keep_alive = !(conn != msg->req.headers.end() && conn->second == "close");
+  return ParseResult::make_ok(msg);
+}
+
+const char* status_reason(int code) {
+  switch (code) {
+    case 200: return "OK";
+    case 400: return "Bad Request";
+    case 403: return "Forbidden";
+    case 404: return "Not Found";
+    case 500: return "Internal Server Error";
+    case 503: return "Service Unavailable";
+    default: return "Unknown";
+  }
+}
+
+void ProcessHttpRequest(InputMessageBase* msg_base) {
+  HttpMessage* msg = (HttpMessage*)msg_base;
+  SocketUniquePtr sock;
+  if (Socket::Address(msg->socket_id, &sock) != 0) {
+    delete msg;
+    return;
+  }
+  Server* server = (Server*)sock->user();
+  HttpResponse resp;
+  resp.status = 200;
+  resp.content_type = "text/plain";
+
+  bool handled = DispatchBuiltinService(server, msg->req, &resp);
+  if (!handled && server != nullptr) {
+    // RPC over HTTP: /Service/Method
+    std::string path = msg->req.path;
+    if (!path.empty() && path[0] == '/') path = path.substr(1);
+    size_t slash = path.find('/');
+    std::string svc = slash == std::string::npos ? "" : path.substr(0, slash);
+    std::string method = slash == std::string::npos ? path : path.substr(slash + 1);
+    const MethodFn* fn = server->FindMethod(svc, method);
+    if (fn != nullptr) {
+      Controller cntl;
+      cntl.server_ = server;
+      cntl.server_socket_ = sock->id();
+      cntl.remote_side_ = sock->remote_side();
+      IOBuf http_resp_body;
+      // synchronous closure: handlers that run inline complete before return
+      std::atomic<bool> done_flag{false};
+      Closure* done = NewCallback([&done_flag] { done_flag.store(true); });
+      (*fn)(&cntl, msg->req.body, &http_resp_body, done);
+      // NOTE: async handlers over HTTP are not yet supported; spin briefly.
+      for (int i = 0; i < 100000 && !done_flag.load(std::memory_order_acquire); ++i) {
+        usleep(100);
+      }
+      if (cntl.Failed()) {
+        resp.status = 500;
+        resp.body.append(cntl.ErrorText());
+      } else {
+        resp.body.append(std::move(http_resp_body));
+        resp.content_type = "application/octet-stream";
+      }
+      handled = true;
+    }
+  }
+  if (!handled) {
+    resp.status = 404;
+    resp.body.append("no such page/method: " + msg->req.path + "\n");
+  }
+
+  // serialize response
+  std::string head;
+  head.reserve(256);
+  head += "HTTP/1.1 " + std::to_string(resp.status) + " " + status_reason(resp.status) +
+          "\r\n";
+  head += "Content-Type: " + resp.content_type + "\r\n";
+  head += "Content-Length: " + std::to_string(resp.body.size()) + "\r\n";
+  for (const auto& kv : resp.headers) head += kv.first + ": " + kv.second + "\r\n";
+  if (!msg->req.keep_alive) head += "Connection: close\r\n";
+  head += "\r\n";
+  IOBuf out;
+  out.append(head);
+  out.append(std::move(resp.body));
+  sock->Write(&out);
+  if (!msg->req.keep_alive) sock->SetFailed(0, "connection: close");
+  delete msg;
+}
+
+}  // namespace
+
+void RegisterHttpProtocol() {
+  static std::once_flag flag;
+  std::call_once(flag, [] {
+    Protocol p;
+    p.parse = ParseHttpMessage;
+    p.process_request = ProcessHttpRequest;
+    p.process_response = nullptr;  // client-side HTTP: later phase
+    p.support_server = true;
+    p.support_client = false;
+    p.name = "http";
+    RegisterProtocol(p);
+  });
+}
+
+}  // namespace policy
+}  // namespace bam

@@ -1,0 +1,38 @@
+// brpc_amd: HTTP/1.1 types + registration (see http_protocol.cc).
+#pragma once
+
+#include <map>
+#include <string>
+
+#include "base/iobuf.h"
+
+namespace bam {
+
+class Server;
+
+namespace policy {
+
+struct HttpRequest {
+  std::string method;
+  std::string path;
+  std::map<std::string, std::string> query;
+  std::map<std::string, std::string> headers;  // lower-cased keys
+  IOBuf body;
+  bool keep_alive = true;
+};
+
+struct HttpResponse {
+  int status = 200;
+  std::string content_type = "text/plain";
+  std::map<std::string, std::string> headers;
+  IOBuf body;
+};
+
+void RegisterHttpProtocol();
+
+// Implemented in rpc/builtin/builtin_services.cc; returns true if the path
+// matched a builtin page.
+bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse* resp);
+
+}  // namespace policy
+}  // namespace bam

@@ -7,6 +7,8 @@
 #include "rpc/channel.h"
 #include "rpc/controller.h"
 #include "rpc/server.h"
+#include "rpc/policy/http_protocol.h"
+#include "rpc/rpcz.h"
 #include "rpc/stream.h"
 #include "rpc/wire.h"
 
@@ -202,6 +204,18 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
     cntl->server_->nprocessed.fetch_add(1, std::memory_order_relaxed);
   }
   if (status != nullptr) *status << (monotonic_time_us() - start_us);
+  if (rpcz::enabled()) {
+    rpcz::Span span;
+    span.start_us = start_us;
+    span.end_us = monotonic_time_us();
+    span.full_method = cntl->call.service_name + "." + cntl->call.method_name;
+    span.remote = cntl->remote_side();
+    span.error_code = cntl->ErrorCode();
+    span.log_id = cntl->log_id();
+    span.server_side = true;
+    span.response_size = resp->size();
+    rpcz::RecordSpan(span);
+  }
   delete resp;
   delete cntl;
 }
@@ -219,6 +233,8 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   cntl->server_socket_ = sock->id();
   cntl->server_cid_ = msg->meta.correlation_id;
   cntl->log_id_ = msg->meta.log_id;
+  cntl->call.service_name = msg->meta.service_name;
+  cntl->call.method_name = msg->meta.method_name;
   cntl->remote_stream_id_ = msg->meta.stream_id;
   cntl->remote_side_ = sock->remote_side();
   cntl->local_side_ = sock->local_side();
@@ -318,6 +334,7 @@ void RegisterStdProtocol() {
     p.name = "std";
     RegisterProtocol(p);
     stream_internal::RegisterStreamProtocol();
+    RegisterHttpProtocol();
   });
 }
 

@@ -1,0 +1,106 @@
+"""HTTP/1.1 protocol + builtin services tests.
+
+Models reference brpc_builtin_service_unittest.cpp: hit the builtin pages
+over a raw HTTP client against the same port that serves std-protocol RPC
+(protocols multiplex on one listener, like the reference).
+"""
+import urllib.request
+
+import pytest
+
+import brpc_amd as b
+
+r = b.core.rpc
+
+
+@pytest.fixture(scope="module")
+def base():
+    port = r.start_echo_server(0)
+    # prime an RPC so /status /rpcz have content
+    rc, _, _ = r.echo_once(f"127.0.0.1:{port}", b"prime", 2000)
+    assert rc == 0
+    return f"http://127.0.0.1:{port}"
+
+
+def get(url, timeout=5):
+    with urllib.request.urlopen(url, timeout=timeout) as resp:
+        return resp.status, resp.read().decode()
+
+
+def test_health(base):
+    st, body = get(base + "/health")
+    assert st == 200 and body == "OK\n"
+
+
+def test_index(base):
+    st, body = get(base + "/index")
+    assert st == 200 and "/vars" in body
+
+
+def test_status_lists_methods(base):
+    st, body = get(base + "/status")
+    assert st == 200
+    assert "EchoService" in body
+    assert "processed_requests" in body
+
+
+def test_vars(base):
+    st, body = get(base + "/vars")
+    assert st == 200
+    assert "_count" in body or len(body) > 0
+
+
+def test_flags_get_and_set(base):
+    st, body = get(base + "/flags")
+    assert st == 200 and "enable_rpcz" in body
+    st, body = get(base + "/flags/rpcz_max_spans?setvalue=1024")
+    assert st == 200
+    st, body = get(base + "/flags/rpcz_max_spans")
+    assert "1024" in body
+
+
+def test_connections(base):
+    st, body = get(base + "/connections")
+    assert st == 200 and "socket_count" in body
+
+
+def test_protobufs(base):
+    st, body = get(base + "/protobufs")
+    assert st == 200 and "EchoService.Echo" in body
+
+
+def test_fibers_page(base):
+    st, body = get(base + "/fibers")
+    assert st == 200 and "workers" in body
+
+
+def test_memory_page(base):
+    st, body = get(base + "/memory")
+    assert st == 200 and "iobuf_block_count" in body
+
+
+def test_rpcz_records_spans(base):
+    st, body = get(base + "/rpcz")
+    assert st == 200
+    assert "EchoService.Echo" in body  # both client+server spans recorded
+
+
+def test_prometheus_metrics(base):
+    st, body = get(base + "/brpc_metrics")
+    assert st == 200
+    assert "# TYPE" in body
+
+
+def test_404(base):
+    import urllib.error
+    with pytest.raises(urllib.error.HTTPError) as ei:
+        get(base + "/nope")
+    assert ei.value.code == 404
+
+
+def test_rpc_over_http(base):
+    req = urllib.request.Request(base + "/EchoService/Echo", data=b"http payload",
+                                 method="POST")
+    with urllib.request.urlopen(req, timeout=5) as resp:
+        assert resp.status == 200
+        assert resp.read() == b"http payload"
