@@ -100,3 +100,70 @@ bool pinned_roundtrip(size_t n) {
 
 }  // namespace gputest
 }  // namespace bam
+
+// ---- perf probes (GB/s numbers for profiles/) ----
+
+#include "base/time.h"
+
+namespace bam {
+namespace gputest {
+
+// GPU crc32c throughput over a resident device buffer.
+double crc_gbps(size_t n, int iters, int dev) {
+  const gpu::GpuApi* api = gpu::api();
+  if (api == nullptr) return -1;
+  void* d = api->alloc_hbm((uint32_t)n, dev);
+  if (d == nullptr) return -1;
+  api->fill(d, n, 0x0123456789abcdefULL, dev);
+  api->crc32c(d, n, 0, dev);  // warm
+  int64_t t0 = monotonic_time_us();
+  for (int i = 0; i < iters; ++i) api->crc32c(d, n, 0, dev);
+  int64_t el = monotonic_time_us() - t0;
+  api->free_hbm(d, (uint32_t)n, dev);
+  return (double)n * iters / el / 1000.0;  // GB/s
+}
+
+// gather kernel throughput (spans of `block` bytes into contiguous).
+double gather_gbps(size_t total, uint32_t block, int iters, int dev) {
+  const gpu::GpuApi* api = gpu::api();
+  if (api == nullptr) return -1;
+  int nspans = (int)((total + block - 1) / block);
+  std::vector<void*> blocks(nspans);
+  std::vector<const void*> srcs(nspans);
+  std::vector<size_t> lens(nspans);
+  for (int i = 0; i < nspans; ++i) {
+    blocks[i] = api->alloc_hbm(block, dev);
+    if (blocks[i] == nullptr) return -1;
+    srcs[i] = blocks[i];
+    lens[i] = (i == nspans - 1) ? total - (size_t)(nspans - 1) * block : block;
+  }
+  void* dst = api->alloc_hbm((uint32_t)total, dev);
+  if (dst == nullptr) return -1;
+  api->gather(dst, srcs.data(), lens.data(), nspans, dev);  // warm
+  int64_t t0 = monotonic_time_us();
+  for (int i = 0; i < iters; ++i) api->gather(dst, srcs.data(), lens.data(), nspans, dev);
+  int64_t el = monotonic_time_us() - t0;
+  for (int i = 0; i < nspans; ++i) api->free_hbm(blocks[i], block, dev);
+  api->free_hbm(dst, (uint32_t)total, dev);
+  // count read+write bytes
+  return 2.0 * total * iters / el / 1000.0;  // GB/s
+}
+
+// D2H staging throughput through the byte mover (the HBM->wire path).
+double d2h_gbps(size_t n, int iters, int dev) {
+  const gpu::GpuApi* api = gpu::api();
+  if (api == nullptr) return -1;
+  void* d = api->alloc_hbm((uint32_t)n, dev);
+  void* h = malloc(n);
+  if (d == nullptr || h == nullptr) return -1;
+  api->memcpy_res(h, 0, -1, d, 2, dev, n);  // warm
+  int64_t t0 = monotonic_time_us();
+  for (int i = 0; i < iters; ++i) api->memcpy_res(h, 0, -1, d, 2, dev, n);
+  int64_t el = monotonic_time_us() - t0;
+  free(h);
+  api->free_hbm(d, (uint32_t)n, dev);
+  return (double)n * iters / el / 1000.0;
+}
+
+}  // namespace gputest
+}  // namespace bam

@@ -101,6 +101,9 @@ bool crc_matches(size_t n, int dev);
 bool crc_extend_matches(size_t n1, size_t n2, int dev);
 bool gather_matches(size_t total, uint32_t block, int dev);
 bool pinned_roundtrip(size_t n);
+double crc_gbps(size_t n, int iters, int dev);
+double gather_gbps(size_t total, uint32_t block, int iters, int dev);
+double d2h_gbps(size_t n, int iters, int dev);
 }  // namespace gputest
 }  // namespace bam
 
@@ -120,4 +123,10 @@ void bind_gpu(py::module_& m) {
         py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
   g.def("pinned_roundtrip", &bam::gputest::pinned_roundtrip, py::arg("n"),
         py::call_guard<py::gil_scoped_release>());
+  g.def("crc_gbps", &bam::gputest::crc_gbps, py::arg("n"), py::arg("iters") = 10,
+        py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
+  g.def("gather_gbps", &bam::gputest::gather_gbps, py::arg("total"), py::arg("block"),
+        py::arg("iters") = 10, py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
+  g.def("d2h_gbps", &bam::gputest::d2h_gbps, py::arg("n"), py::arg("iters") = 10,
+        py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
 }
