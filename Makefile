@@ -11,7 +11,7 @@ GPU_ARCH  := gfx950
 
 CXXFLAGS  := -O2 -g -std=c++17 -fPIC -pthread -Wall -Wno-unused-function \
              -Isrc -msse4.2 -fvisibility=hidden -MMD -MP
-LDFLAGS   := -shared -pthread -ldl
+LDFLAGS   := -shared -pthread -ldl -lz
 
 CORE_SRCS := $(wildcard src/base/*.cc) $(wildcard src/fiber/*.cc) $(wildcard src/rpc/*.cc) \
              $(wildcard src/rpc/policy/*.cc) $(wildcard src/rpc/builtin/*.cc) \

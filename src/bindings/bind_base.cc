@@ -130,3 +130,26 @@ void bind_gpu(py::module_& m) {
   g.def("d2h_gbps", &bam::gputest::d2h_gbps, py::arg("n"), py::arg("iters") = 10,
         py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
 }
+
+// ---- snappy host codec bindings (oracle for the gfx950 kernel) ----
+#include "base/snappy.h"
+
+void bind_snappy(py::module_& m) {
+  auto sn = m.def_submodule("snappy");
+  sn.def("compress", [](py::bytes data) {
+    char* ptr;
+    Py_ssize_t len;
+    PyBytes_AsStringAndSize(data.ptr(), &ptr, &len);
+    std::string out;
+    bam::snappy::Compress(ptr, (size_t)len, &out);
+    return py::bytes(out);
+  });
+  sn.def("uncompress", [](py::bytes data) -> py::object {
+    char* ptr;
+    Py_ssize_t len;
+    PyBytes_AsStringAndSize(data.ptr(), &ptr, &len);
+    std::string out;
+    if (!bam::snappy::Uncompress(ptr, (size_t)len, &out)) return py::none();
+    return py::bytes(out);
+  });
+}

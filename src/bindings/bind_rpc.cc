@@ -98,6 +98,8 @@ int parallel_echo_test(const std::vector<int>& ports, const std::string& payload
 int selective_test(int dead_port, int live_port, std::string* resp_out);
 int partition_test(const std::vector<int>& ports, std::string* merged);
 int lb_spread_test(const std::string& lb_name, const std::vector<int>& ports, int ncalls);
+int compressed_echo_test(const std::string& addr, const std::string& payload,
+                         int compress_type, std::string* response_out);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -139,6 +141,19 @@ void bind_rpc_combo(py::module_& m) {
         });
   r.def("lb_spread", &bam::rpctest::lb_spread_test,
         py::call_guard<py::gil_scoped_release>());
+  r.def("compressed_echo",
+        [](const std::string& addr, py::bytes payload, int ctype) {
+          char* ptr;
+          Py_ssize_t len;
+          PyBytes_AsStringAndSize(payload.ptr(), &ptr, &len);
+          std::string p(ptr, len), resp;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::compressed_echo_test(addr, p, ctype, &resp);
+          }
+          return py::make_tuple(rc, py::bytes(resp));
+        });
 }
 
 // ---- streaming ----

@@ -9,4 +9,5 @@ PYBIND11_MODULE(_core, m) {
   bind_gpu(m);
   bind_rpc_combo(m);
   bind_rpc_stream(m);
+  bind_snappy(m);
 }

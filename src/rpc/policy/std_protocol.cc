@@ -7,6 +7,7 @@
 #include "rpc/channel.h"
 #include "rpc/controller.h"
 #include "rpc/server.h"
+#include "rpc/compress.h"
 #include "rpc/policy/http_protocol.h"
 #include "rpc/rpcz.h"
 #include "rpc/stream.h"
@@ -151,23 +152,35 @@ void PackStdRequest(IOBuf* out, Controller* cntl, SessionId correlation_id) {
   meta.correlation_id = (int64_t)correlation_id;
   meta.attachment_size = (int32_t)cntl->request_attachment().size();
   meta.stream_id = cntl->call.stream_id;
+  // Compress the serialized request (NOT the attachment — it is a raw
+  // pass-through by contract, like the reference's baidu_std attachment).
+  IOBuf body_buf;
+  if (cntl->request_compress_type() != COMPRESS_TYPE_NONE) {
+    RegisterBuiltinCompressHandlers();
+    if (!ApplyCompress(cntl->request_compress_type(), cntl->call.request_buf, &body_buf)) {
+      meta.compress_type = 0;
+      body_buf = cntl->call.request_buf;
+    }
+  } else {
+    body_buf = cntl->call.request_buf;
+  }
   std::string meta_bytes;
   SerializeRpcMeta(meta, &meta_bytes);
-  size_t body = meta_bytes.size() + cntl->call.request_buf.size() +
-                cntl->request_attachment().size();
+  size_t body = meta_bytes.size() + body_buf.size() + cntl->request_attachment().size();
   char header[kHeaderLen];
   memcpy(header, kMagic, 4);
   wire::put_u32_be(header + 4, (uint32_t)body);
   wire::put_u32_be(header + 8, (uint32_t)meta_bytes.size());
   out->append(header, kHeaderLen);
   out->append(meta_bytes);
-  out->append(cntl->call.request_buf);       // zero-copy ref share
+  out->append(std::move(body_buf));
   out->append(cntl->request_attachment());   // zero-copy ref share
 }
 
 static void PackStdResponse(IOBuf* out, int64_t correlation_id, int error_code,
                             const std::string& error_text, const IOBuf& payload,
-                            const IOBuf& attachment, uint64_t stream_id) {
+                            const IOBuf& attachment, uint64_t stream_id,
+                            int compress_type = 0) {
   RpcMeta meta;
   meta.has_response = true;
   meta.error_code = error_code;
@@ -175,6 +188,7 @@ static void PackStdResponse(IOBuf* out, int64_t correlation_id, int error_code,
   meta.correlation_id = correlation_id;
   meta.attachment_size = (int32_t)attachment.size();
   meta.stream_id = stream_id;
+  meta.compress_type = compress_type;
   std::string meta_bytes;
   SerializeRpcMeta(meta, &meta_bytes);
   size_t body = meta_bytes.size() + payload.size() + attachment.size();
@@ -194,9 +208,19 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
                             var::LatencyRecorder* status, int64_t start_us) {
   SocketUniquePtr sock;
   if (Socket::Address(sid, &sock) == 0) {
+    IOBuf body = *resp;
+    int resp_compress = 0;
+    if (cntl->response_compress_ != COMPRESS_TYPE_NONE && !cntl->Failed()) {
+      RegisterBuiltinCompressHandlers();
+      IOBuf compressed;
+      if (ApplyCompress(cntl->response_compress_, body, &compressed)) {
+        body.swap(compressed);
+        resp_compress = (int)cntl->response_compress_;
+      }
+    }
     IOBuf packet;
-    PackStdResponse(&packet, cid, cntl->ErrorCode(), cntl->ErrorText(), *resp,
-                    cntl->response_attachment(), cntl->response_stream_id_);
+    PackStdResponse(&packet, cid, cntl->ErrorCode(), cntl->ErrorText(), body,
+                    cntl->response_attachment(), cntl->response_stream_id_, resp_compress);
     sock->Write(&packet);
   }
   if (cntl->server_ != nullptr) {
@@ -254,6 +278,17 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   size_t data_len = msg->payload.size() >= att ? msg->payload.size() - att : 0;
   msg->payload.cutn(&req_data, data_len);
   cntl->request_attachment().swap(msg->payload);
+  if (msg->meta.compress_type != 0) {
+    RegisterBuiltinCompressHandlers();
+    IOBuf plain;
+    if (!ApplyDecompress((CompressType)msg->meta.compress_type, req_data, &plain)) {
+      cntl->SetFailed(EREQUEST, "request decompression failed");
+      delete msg;
+      done->Run();
+      return;
+    }
+    req_data.swap(plain);
+  }
 
   Service* svc = nullptr;
   const MethodFn* fn =
@@ -312,6 +347,16 @@ static void ProcessStdResponse(InputMessageBase* msg_base) {
     if (cntl->call.response != nullptr) {
       cntl->call.response->clear();
       msg->payload.cutn(cntl->call.response, data_len);
+      if (msg->meta.compress_type != 0) {
+        RegisterBuiltinCompressHandlers();
+        IOBuf plain;
+        if (ApplyDecompress((CompressType)msg->meta.compress_type, *cntl->call.response,
+                            &plain)) {
+          cntl->call.response->swap(plain);
+        } else {
+          cntl->SetFailed(ERESPONSE, "response decompression failed");
+        }
+      }
     } else {
       msg->payload.pop_front(data_len);
     }

@@ -39,6 +39,12 @@ Service* NewEchoService() {
     cntl->SetFailed(EINTERNAL, "you asked for it");
     done->Run();
   });
+  svc->AddMethod("EchoCompressed",
+                 [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    resp->append(req);
+    cntl->set_response_compress_type(COMPRESS_TYPE_SNAPPY);
+    done->Run();
+  });
   svc->AddMethod("Port", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
     resp->append(std::to_string(cntl->server_->listen_address().port));
     done->Run();
@@ -209,6 +215,31 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
     res.max_us = latencies.back();
   }
   return res;
+}
+
+}  // namespace rpctest
+}  // namespace bam
+
+// ---- compression round trips (snappy/gzip over the wire) ----
+#include "rpc/compress.h"
+
+namespace bam {
+namespace rpctest {
+
+int compressed_echo_test(const std::string& addr, const std::string& payload,
+                         int compress_type, std::string* response_out) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = 5000;
+  if (channel.Init(addr.c_str(), &opts) != 0) return -1;
+  Controller cntl;
+  cntl.set_request_compress_type((CompressType)compress_type);
+  IOBuf request, response;
+  request.append(payload);
+  channel.CallMethod("EchoService.EchoCompressed", &cntl, &request, &response, nullptr);
+  if (cntl.Failed()) return cntl.ErrorCode();
+  if (response_out != nullptr) *response_out = response.to_string();
+  return 0;
 }
 
 }  // namespace rpctest
