@@ -1,11 +1,17 @@
 // brpc_amd: CRC32-C on gfx950.
-// Design (MI355X-first, see /opt guide §2): the message is split into
-// 32 KiB chunks; one wave (64 lanes) owns 64 consecutive chunks. HBM reads
-// are fully coalesced row-wise into an LDS tile (64x64 words, padded), then
-// each lane walks ITS chunk's words out of LDS serially (CRC is a serial
-// recurrence per chunk) using slice-by-8 tables staged in LDS. Per-chunk
-// CRCs are merged on the host with one precomputed GF(2) shift operator
-// (x^(8*chunk)) — O(1) matrix-vector per chunk.
+//
+// Two-kernel design (MI355X-first; see /opt/skills/guides):
+//  kernel 1 (crc_chunks_kernel): the message is split into 2 KiB chunks;
+//    one wave (64 lanes) owns 64 consecutive chunks. HBM reads are fully
+//    coalesced row-wise into an LDS tile (64x64 words, padded), then each
+//    lane walks ITS chunk's words serially out of LDS (CRC is a serial
+//    recurrence per chunk) with slice-by-8 tables staged in LDS.
+//    2 KiB chunks -> 2048 workgroups at 256 MiB: fills 256 CUs (≫256 rule).
+//  kernel 2 (crc_combine_kernel): each lane shifts its chunk's CRC by the
+//    byte-distance to the end of the message using precomputed GF(2)
+//    matrices for x^(8*chunk*2^k) (+ one per-call tail operator), then
+//    XOR-reduces with device-scope atomics. Only 4 bytes return to host.
+//
 // Host reference / test oracle: src/base/crc32c.cc.
 #include <hip/hip_runtime.h>
 
@@ -16,12 +22,14 @@
 namespace {
 
 constexpr uint32_t kPoly = 0x82F63B78u;
-constexpr int kChunkWords = 4096;  // 32 KiB per chunk
+constexpr int kChunkWords = 256;  // 2 KiB per chunk
 constexpr size_t kChunkBytes = (size_t)kChunkWords * 8;
+constexpr int kMaxPow = 28;  // supports up to 2^28 full chunks (512 TB)
 
 __device__ uint32_t d_tab[8][256];
+__device__ uint32_t d_pow[kMaxPow][32];  // shift-by-(chunk*2^k) operators
 
-// ---- host-side table + GF(2) combine helpers ----
+// ---- host-side table + GF(2) helpers ----
 
 uint32_t h_tab[8][256];
 
@@ -54,8 +62,13 @@ inline void gf2_square(uint32_t* sq, const uint32_t* mat) {
   for (int n = 0; n < 32; ++n) sq[n] = gf2_times(mat, mat[n]);
 }
 
-// Builds the operator for len zero-bytes appended (as in zlib crc32_combine).
-void build_shift_operator(uint32_t* op /*32*/, size_t len) {
+inline void gf2_matmul(uint32_t* out, const uint32_t* a, const uint32_t* b) {
+  // out = a∘b (apply b then a); GF(2) poly-mod matrices commute anyway.
+  for (int n = 0; n < 32; ++n) out[n] = gf2_times(a, b[n]);
+}
+
+// operator for `len` zero BYTES appended (zlib crc32_combine technique)
+void build_shift_operator(uint32_t* op, size_t len) {
   uint32_t even[32], odd[32];
   odd[0] = kPoly;
   uint32_t row = 1;
@@ -63,23 +76,21 @@ void build_shift_operator(uint32_t* op /*32*/, size_t len) {
     odd[n] = row;
     row <<= 1;
   }
-  gf2_square(even, odd);
-  gf2_square(odd, even);
-  // identity
-  for (int n = 0; n < 32; ++n) op[n] = 1u << n;
+  gf2_square(even, odd);  // x^2
+  gf2_square(odd, even);  // x^4
+  for (int n = 0; n < 32; ++n) op[n] = 1u << n;  // identity
   uint32_t tmp[32];
   bool use_even = true;
-  do {
+  while (len != 0) {
     gf2_square(use_even ? even : odd, use_even ? odd : even);
     const uint32_t* cur = use_even ? even : odd;
     if (len & 1) {
-      for (int n = 0; n < 32; ++n) tmp[n] = gf2_times(op, cur[n]);
-      // op = cur * op  (apply cur after op)
-      for (int n = 0; n < 32; ++n) op[n] = tmp[n];
+      gf2_matmul(tmp, cur, op);
+      memcpy(op, tmp, sizeof(tmp));
     }
     len >>= 1;
     use_even = !use_even;
-  } while (len != 0);
+  }
 }
 
 uint32_t combine_with_op(const uint32_t* op, uint32_t crc_a, uint32_t crc_b) {
@@ -87,17 +98,20 @@ uint32_t combine_with_op(const uint32_t* op, uint32_t crc_a, uint32_t crc_b) {
 }
 
 std::once_flag g_init_flag;
-uint32_t g_chunk_op[32];  // shift by kChunkBytes
 
 void ensure_init() {
   std::call_once(g_init_flag, [] {
     build_tables();
     hipMemcpyToSymbol(HIP_SYMBOL(d_tab), h_tab, sizeof(h_tab));
-    build_shift_operator(g_chunk_op, kChunkBytes);
+    // power operators: pow[k] = shift by chunk*2^k bytes
+    static uint32_t pows[kMaxPow][32];
+    build_shift_operator(pows[0], kChunkBytes);
+    for (int k = 1; k < kMaxPow; ++k) gf2_matmul(pows[k], pows[k - 1], pows[k - 1]);
+    hipMemcpyToSymbol(HIP_SYMBOL(d_pow), pows, sizeof(pows));
   });
 }
 
-// ---- kernel ----
+// ---- kernels ----
 
 __global__ __launch_bounds__(64) void crc_chunks_kernel(const uint8_t* data, size_t n,
                                                         uint32_t* out, int nchunks) {
@@ -113,17 +127,15 @@ __global__ __launch_bounds__(64) void crc_chunks_kernel(const uint8_t* data, siz
   const size_t my_end = my_start + kChunkBytes < n ? my_start + kChunkBytes : n;
   uint32_t crc = 0xFFFFFFFFu;
 
-  const bool group_full = ((size_t)(cbase + 64) * kChunkBytes) <= n &&
-                          (((uintptr_t)data & 7) == 0);
+  const bool group_full =
+      ((size_t)(cbase + 64) * kChunkBytes) <= n && (((uintptr_t)data & 7) == 0);
   if (group_full) {
     const uint64_t* wdata = (const uint64_t*)data + (size_t)cbase * kChunkWords;
     for (int t = 0; t < kChunkWords; t += 64) {
-      // Coalesced stage: row r = chunk (cbase+r), words [t, t+64).
       for (int r = 0; r < 64; ++r) {
         tile[r][lane] = wdata[(size_t)r * kChunkWords + t + lane];
       }
       __syncthreads();
-      // Each lane consumes its own chunk's words from LDS.
       for (int w = 0; w < 64; ++w) {
         uint64_t v = tile[lane][w] ^ crc;
         crc = tab[7][v & 0xff] ^ tab[6][(v >> 8) & 0xff] ^ tab[5][(v >> 16) & 0xff] ^
@@ -133,13 +145,52 @@ __global__ __launch_bounds__(64) void crc_chunks_kernel(const uint8_t* data, siz
       __syncthreads();
     }
   } else if (my_start < n) {
-    // Boundary group: byte-serial from global (last <16 MiB of the input).
+    // Boundary group: byte-serial from global (≤128 KiB tail of the input).
     for (size_t i = my_start; i < my_end; ++i) {
       crc = tab[0][(crc ^ data[i]) & 0xff] ^ (crc >> 8);
     }
   }
   if (my_chunk < nchunks) out[my_chunk] = ~crc;
 }
+
+__device__ __forceinline__ uint32_t dev_gf2_times(const uint32_t* mat, uint32_t vec) {
+  uint32_t sum = 0;
+  while (vec) {
+    if (vec & 1) sum ^= *mat;
+    vec >>= 1;
+    ++mat;
+  }
+  return sum;
+}
+
+// tail_op: operator for the LAST chunk's length (identity if it is full).
+// Every chunk i < nchunks-1 is shifted by (nchunks-2-i) full chunks + tail;
+// the last chunk is unshifted. XOR-reduce into *result (pre-zeroed).
+__global__ void crc_combine_kernel(const uint32_t* chunk_crcs, int nchunks,
+                                   const uint32_t* tail_op, uint32_t* result) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= nchunks) return;
+  uint32_t crc = chunk_crcs[i];
+  if (i < nchunks - 1) {
+    uint32_t m = (uint32_t)(nchunks - 2 - i);  // full-chunk shifts
+    for (int k = 0; k < kMaxPow && m != 0; ++k, m >>= 1) {
+      if (m & 1) crc = dev_gf2_times(d_pow[k], crc);
+    }
+    crc = dev_gf2_times(tail_op, crc);
+  }
+  atomicXor(result, crc);
+}
+
+// persistent per-device scratch
+constexpr int kMaxDevCrc = 16;
+struct CrcScratch {
+  uint32_t* chunk_crcs = nullptr;
+  size_t cap = 0;
+  uint32_t* result = nullptr;
+  uint32_t* tail_op = nullptr;
+};
+CrcScratch g_scratch[kMaxDevCrc];
+std::mutex g_scratch_mu;
 
 }  // namespace
 
@@ -151,32 +202,43 @@ extern "C" uint32_t bam_gpu_crc32c(const void* dev_ptr, size_t n, uint32_t init,
   if (dev != old_dev) hipSetDevice(dev);
   const int nchunks = (int)((n + kChunkBytes - 1) / kChunkBytes);
   const int nblocks = (nchunks + 63) / 64;
-  uint32_t* d_out = nullptr;
-  hipMalloc(&d_out, sizeof(uint32_t) * nchunks);
-  hipLaunchKernelGGL(crc_chunks_kernel, dim3(nblocks), dim3(64), 0, 0,
-                     (const uint8_t*)dev_ptr, n, d_out, nchunks);
-  uint32_t* h_out = (uint32_t*)malloc(sizeof(uint32_t) * nchunks);
-  hipMemcpy(h_out, d_out, sizeof(uint32_t) * nchunks, hipMemcpyDeviceToHost);
-  hipFree(d_out);
-  // Merge: full chunks use the cached fixed-shift operator; the final
-  // (possibly short) chunk gets a one-off operator.
-  uint32_t crc = h_out[0];
-  for (int i = 1; i < nchunks; ++i) {
-    size_t clen = (i == nchunks - 1) ? n - (size_t)i * kChunkBytes : kChunkBytes;
-    if (clen == kChunkBytes) {
-      crc = combine_with_op(g_chunk_op, crc, h_out[i]);
-    } else {
-      uint32_t op[32];
-      build_shift_operator(op, clen);
-      crc = combine_with_op(op, crc, h_out[i]);
+
+  CrcScratch* sc;
+  {
+    std::lock_guard<std::mutex> lk(g_scratch_mu);
+    sc = &g_scratch[dev < kMaxDevCrc ? dev : 0];
+    if (sc->cap < (size_t)nchunks) {
+      if (sc->chunk_crcs != nullptr) hipFree(sc->chunk_crcs);
+      sc->cap = (size_t)nchunks * 2;
+      hipMalloc(&sc->chunk_crcs, sc->cap * sizeof(uint32_t));
+    }
+    if (sc->result == nullptr) {
+      hipMalloc(&sc->result, sizeof(uint32_t));
+      hipMalloc(&sc->tail_op, 32 * sizeof(uint32_t));
     }
   }
+
+  hipLaunchKernelGGL(crc_chunks_kernel, dim3(nblocks), dim3(64), 0, 0,
+                     (const uint8_t*)dev_ptr, n, sc->chunk_crcs, nchunks);
+
+  // tail operator (identity when the last chunk is full)
+  size_t tail_len = n - (size_t)(nchunks - 1) * kChunkBytes;
+  uint32_t tail_op[32];
+  build_shift_operator(tail_op, tail_len);
+  hipMemcpy(sc->tail_op, tail_op, sizeof(tail_op), hipMemcpyHostToDevice);
+  uint32_t zero = 0;
+  hipMemcpy(sc->result, &zero, sizeof(zero), hipMemcpyHostToDevice);
+  int cblocks = (nchunks + 255) / 256;
+  hipLaunchKernelGGL(crc_combine_kernel, dim3(cblocks), dim3(256), 0, 0, sc->chunk_crcs,
+                     nchunks, sc->tail_op, sc->result);
+  uint32_t crc = 0;
+  hipMemcpy(&crc, sc->result, sizeof(crc), hipMemcpyDeviceToHost);
+
   if (init != 0) {
     uint32_t op[32];
     build_shift_operator(op, n);
     crc = combine_with_op(op, init, crc);
   }
-  free(h_out);
   if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
   return crc;
 }

@@ -55,42 +55,58 @@ int launch_spans(Span* spans, int nspans, int dev) {
 
 }  // namespace
 
+// Large spans are chopped into ≤256 KiB sub-spans so the grid has ≫256
+// workgroups (one block per sub-span) even for a few multi-MiB blocks.
+constexpr size_t kSubSpan = 256u << 10;
+
+template <typename NextFn>
+static int run_spans(int nspans, NextFn next, int dev) {
+  Span spans[kMaxSpansPerLaunch];
+  int batch = 0;
+  int rc = 0;
+  for (int i = 0; i < nspans; ++i) {
+    Span s = next(i);
+    size_t off = 0;
+    while (off < s.len) {
+      size_t piece = s.len - off < kSubSpan ? s.len - off : kSubSpan;
+      spans[batch].src = s.src + off;
+      spans[batch].dst = s.dst + off;
+      spans[batch].len = piece;
+      off += piece;
+      if (++batch == kMaxSpansPerLaunch) {
+        rc |= launch_spans(spans, batch, dev);
+        batch = 0;
+      }
+    }
+  }
+  if (batch > 0) rc |= launch_spans(spans, batch, dev);
+  return rc;
+}
+
 extern "C" int bam_gpu_gather(void* dst_dev, const void* const* srcs, const size_t* lens,
                               int nspans, int dev) {
   if (nspans <= 0) return 0;
-  Span spans[kMaxSpansPerLaunch];
   char* out = (char*)dst_dev;
-  int rc = 0;
-  int batch = 0;
-  for (int i = 0; i < nspans; ++i) {
-    spans[batch].src = (const char*)srcs[i];
-    spans[batch].dst = out;
-    spans[batch].len = lens[i];
-    out += lens[i];
-    if (++batch == kMaxSpansPerLaunch || i == nspans - 1) {
-      rc |= launch_spans(spans, batch, dev);
-      batch = 0;
-    }
-  }
-  return rc;
+  size_t acc = 0;
+  return run_spans(nspans,
+                   [&](int i) {
+                     Span s{(const char*)srcs[i], out + acc, lens[i]};
+                     acc += lens[i];
+                     return s;
+                   },
+                   dev);
 }
 
 extern "C" int bam_gpu_scatter(void* const* dsts, const size_t* lens, int nspans,
                                const void* src_dev, int dev) {
   if (nspans <= 0) return 0;
-  Span spans[kMaxSpansPerLaunch];
   const char* in = (const char*)src_dev;
-  int rc = 0;
-  int batch = 0;
-  for (int i = 0; i < nspans; ++i) {
-    spans[batch].src = in;
-    spans[batch].dst = (char*)dsts[i];
-    spans[batch].len = lens[i];
-    in += lens[i];
-    if (++batch == kMaxSpansPerLaunch || i == nspans - 1) {
-      rc |= launch_spans(spans, batch, dev);
-      batch = 0;
-    }
-  }
-  return rc;
+  size_t acc = 0;
+  return run_spans(nspans,
+                   [&](int i) {
+                     Span s{in + acc, (char*)dsts[i], lens[i]};
+                     acc += lens[i];
+                     return s;
+                   },
+                   dev);
 }
