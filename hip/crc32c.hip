@@ -15,6 +15,8 @@
 // Host reference / test oracle: src/base/crc32c.cc.
 #include <hip/hip_runtime.h>
 
+#include <string.h>
+
 #include <mutex>
 
 #include "gpu_api.h"
@@ -86,7 +88,7 @@ void build_shift_operator(uint32_t* op, size_t len) {
     const uint32_t* cur = use_even ? even : odd;
     if (len & 1) {
       gf2_matmul(tmp, cur, op);
-      memcpy(op, tmp, sizeof(tmp));
+      ::memcpy(op, tmp, sizeof(tmp));
     }
     len >>= 1;
     use_even = !use_even;
