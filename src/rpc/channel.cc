@@ -77,6 +77,11 @@ void EndRPC(Controller* cntl, SessionId locked_id) {
     cntl->call.lb->Feedback(cntl->call.server_ep, cntl->error_code_,
                             cntl->end_us_ - cntl->start_us_);
   }
+  if (cntl->remote_side_.port != 0) {
+    const int ec = cntl->error_code_;
+    const bool network_error = ec != 0 && (ec < 1000 || ec == EFAILEDSOCKET);
+    ReportClientCallResult(cntl->remote_side_, network_error);
+  }
   Closure* done = cntl->call.done;
   session_unlock_and_destroy(locked_id);
   if (done != nullptr) done->Run();

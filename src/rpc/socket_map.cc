@@ -1,11 +1,24 @@
 #include "rpc/socket_map.h"
 
+#include <poll.h>
+#include <unistd.h>
+
 #include <map>
 #include <mutex>
 
+#include "base/flags.h"
+#include "base/time.h"
+#include "fiber/fiber.h"
 #include "rpc/input_messenger.h"
 
 namespace bam {
+
+BAM_DEFINE_bool(enable_circuit_breaker, true,
+                "isolate endpoints after repeated connection failures");
+BAM_DEFINE_int64(circuit_breaker_max_failures, 3,
+                 "consecutive network failures before isolating an endpoint");
+BAM_DEFINE_int64(health_check_interval_ms, 500,
+                 "re-probe interval for isolated endpoints");
 
 namespace {
 
@@ -24,9 +37,97 @@ ClientSocketMap& the_map() {
   return *m;
 }
 
+struct EndpointHealth {
+  std::atomic<int> consecutive_errors{0};
+  std::atomic<bool> isolated{false};
+};
+
+struct HealthRegistry {
+  std::mutex mu;
+  std::map<EndPoint, EndpointHealth*> map;
+};
+HealthRegistry& health_registry() {
+  static HealthRegistry* r = new HealthRegistry;
+  return *r;
+}
+
+EndpointHealth* health_of(const EndPoint& ep, bool create) {
+  HealthRegistry& r = health_registry();
+  std::lock_guard<std::mutex> lk(r.mu);
+  auto it = r.map.find(ep);
+  if (it != r.map.end()) return it->second;
+  if (!create) return nullptr;
+  EndpointHealth* h = new EndpointHealth;
+  r.map[ep] = h;
+  return h;
+}
+
+struct HealthCheckArg {
+  EndPoint ep;
+  EndpointHealth* h;
+};
+
+// Probes the endpoint until a connect succeeds, then revives it.
+void health_check_fiber(void* raw) {
+  HealthCheckArg* a = (HealthCheckArg*)raw;
+  for (;;) {
+    fiber_usleep((uint64_t)FLAG_health_check_interval_ms * 1000);
+    bool in_progress = false;
+    int fd = tcp_connect(a->ep, &in_progress);
+    if (fd >= 0) {
+      bool ok = !in_progress;
+      if (in_progress) {
+        struct pollfd pfd{fd, POLLOUT, 0};
+        if (::poll(&pfd, 1, 300) > 0 && (pfd.revents & POLLOUT)) {
+          int err = 0;
+          socklen_t len = sizeof(err);
+          getsockopt(fd, SOL_SOCKET, SO_ERROR, &err, &len);
+          ok = err == 0;
+        }
+      }
+      ::close(fd);
+      if (ok) {
+        a->h->consecutive_errors.store(0, std::memory_order_relaxed);
+        a->h->isolated.store(false, std::memory_order_release);
+        delete a;
+        return;
+      }
+    }
+  }
+}
+
 }  // namespace
 
+void ReportClientCallResult(const EndPoint& ep, bool network_error) {
+  if (!FLAG_enable_circuit_breaker) return;
+  if (!network_error) {
+    EndpointHealth* h = health_of(ep, false);
+    if (h != nullptr) h->consecutive_errors.store(0, std::memory_order_relaxed);
+    return;
+  }
+  EndpointHealth* h = health_of(ep, true);
+  int n = h->consecutive_errors.fetch_add(1, std::memory_order_relaxed) + 1;
+  if (n >= FLAG_circuit_breaker_max_failures &&
+      !h->isolated.exchange(true, std::memory_order_acq_rel)) {
+    HealthCheckArg* a = new HealthCheckArg{ep, h};
+    fiber_t th;
+    if (fiber_start_background(&th, health_check_fiber, a) != 0) {
+      h->isolated.store(false, std::memory_order_release);
+      delete a;
+    }
+  }
+}
+
+bool IsEndpointIsolated(const EndPoint& ep) {
+  EndpointHealth* h = health_of(ep, false);
+  return h != nullptr && h->isolated.load(std::memory_order_acquire);
+}
+
 int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out) {
+  if (IsEndpointIsolated(ep)) {
+    errno = EHOSTDOWN;
+    return -1;
+  }
   ClientSocketMap& m = the_map();
   {
     std::lock_guard<std::mutex> lk(m.mu);

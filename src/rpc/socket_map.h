@@ -14,4 +14,12 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out);
 // Drops the cached socket for ep (e.g. after failure).
 void RemoveClientSocket(const EndPoint& ep, SocketId expected);
 
+// ---- circuit breaker + health check ----
+// Parity: reference brpc/circuit_breaker.{h,cpp} + details/health_check.cpp:
+// consecutive network failures isolate the endpoint (GetClientSocket fails
+// fast with EHOSTDOWN); a background fiber re-probes every
+// -health_check_interval_ms and revives on a successful connect.
+void ReportClientCallResult(const EndPoint& ep, bool network_error);
+bool IsEndpointIsolated(const EndPoint& ep);
+
 }  // namespace bam

@@ -1,0 +1,40 @@
+"""Distributed bench-path test: 2 processes over gloo on CPU (the same
+rendezvous/aggregation path the driver uses with RCCL on GPUs)."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_two_ranks_gloo():
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--master-addr", "127.0.0.1", "--master-port", "29511",
+        "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
+        "--calls-per-step", "500", "--concurrency", "8",
+    ]
+    out = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True, text=True,
+                         timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["metric"] == "echo_qps"
+    assert res["n_gpus"] == 2
+    assert res["value"] > 0
+    assert res["scaling"] == "weak"
+
+
+def test_bench_single_process():
+    out = subprocess.run([sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+                          "--calls-per-step", "500"],
+                         cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["value"] > 0
+    assert res["config"]["p99_us"] > 0
