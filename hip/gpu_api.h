@@ -37,6 +37,12 @@ int bam_gpu_gather(void* dst_dev, const void* const* srcs, const size_t* lens, i
 int bam_gpu_scatter(void* const* dsts, const size_t* lens, int nspans, const void* src_dev,
                     int dev);
 
+// Snappy (standard wire format) on device buffers. Returns 0 on success.
+int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_dev, size_t dst_cap,
+                            size_t* out_len, int dev);
+int bam_gpu_snappy_decompress(const void* src_dev, size_t n, void* dst_dev, size_t dst_cap,
+                              size_t* out_len, int dev);
+
 // Fill device memory with a repeating 64-bit pattern (tests / synthetic
 // payload generation without H2D traffic).
 int bam_gpu_fill(void* dst_dev, size_t n, uint64_t pattern, int dev);

@@ -59,6 +59,10 @@ void do_load() {
   g_api.scatter =
       (int (*)(void* const*, const size_t*, int, const void*, int))must_sym(h, "bam_gpu_scatter");
   g_api.fill = (int (*)(void*, size_t, uint64_t, int))must_sym(h, "bam_gpu_fill");
+  g_api.snappy_compress = (int (*)(const void*, size_t, void*, size_t, size_t*, int))must_sym(
+      h, "bam_gpu_snappy_compress");
+  g_api.snappy_decompress = (int (*)(const void*, size_t, void*, size_t, size_t*,
+                                     int))must_sym(h, "bam_gpu_snappy_decompress");
   g_api.last_error = (const char* (*)(void))must_sym(h, "bam_gpu_last_error");
   if (!g_error.empty()) return;
 

@@ -167,3 +167,80 @@ double d2h_gbps(size_t n, int iters, int dev) {
 
 }  // namespace gputest
 }  // namespace bam
+
+// ---- GPU snappy vs host codec (both directions) ----
+
+#include "base/snappy.h"
+
+namespace bam {
+namespace gputest {
+
+// GPU-compressed stream must decompress with the HOST codec, and a
+// HOST-compressed stream must decompress on the GPU.
+bool snappy_cross_check(size_t n, int mode /*0 compressible, 1 random*/, int dev) {
+  const gpu::GpuApi* api = gpu::api();
+  if (api == nullptr) return false;
+  std::string data(n, 0);
+  if (mode == 0) {
+    for (size_t i = 0; i < n; ++i) data[i] = (char)("abcabcab"[i % 8] + (i / 1024) % 3);
+  } else {
+    for (size_t i = 0; i < n; ++i) data[i] = (char)fast_rand();
+  }
+  void* d_in = api->alloc_hbm((uint32_t)n, dev);
+  size_t comp_cap = n + n / 3 + 64;
+  void* d_comp = api->alloc_hbm((uint32_t)comp_cap, dev);
+  if (d_in == nullptr || d_comp == nullptr) return false;
+  api->memcpy_res(d_in, 2, dev, data.data(), 0, -1, n);
+  size_t comp_len = 0;
+  if (api->snappy_compress(d_in, n, d_comp, comp_cap, &comp_len, dev) != 0) return false;
+  // host decompress of the GPU stream
+  std::string comp(comp_len, 0);
+  api->memcpy_res(&comp[0], 0, -1, d_comp, 2, dev, comp_len);
+  std::string back;
+  if (!snappy::Uncompress(comp.data(), comp.size(), &back)) return false;
+  if (back != data) return false;
+  // GPU decompress of a HOST-compressed stream
+  std::string host_comp;
+  snappy::Compress(data.data(), data.size(), &host_comp);
+  void* d_hcomp = api->alloc_hbm((uint32_t)host_comp.size(), dev);
+  void* d_out = api->alloc_hbm((uint32_t)n, dev);
+  if (d_hcomp == nullptr || d_out == nullptr) return false;
+  api->memcpy_res(d_hcomp, 2, dev, host_comp.data(), 0, -1, host_comp.size());
+  size_t out_len = 0;
+  if (api->snappy_decompress(d_hcomp, host_comp.size(), d_out, n, &out_len, dev) != 0)
+    return false;
+  if (out_len != n) return false;
+  std::string back2(n, 0);
+  api->memcpy_res(&back2[0], 0, -1, d_out, 2, dev, n);
+  // also decompress the GPU's own stream on the GPU
+  size_t out_len2 = 0;
+  if (api->snappy_decompress(d_comp, comp_len, d_out, n, &out_len2, dev) != 0) return false;
+  std::string back3(n, 0);
+  api->memcpy_res(&back3[0], 0, -1, d_out, 2, dev, n);
+  api->free_hbm(d_in, (uint32_t)n, dev);
+  api->free_hbm(d_comp, (uint32_t)comp_cap, dev);
+  api->free_hbm(d_hcomp, (uint32_t)host_comp.size(), dev);
+  api->free_hbm(d_out, (uint32_t)n, dev);
+  return back2 == data && back3 == data;
+}
+
+double snappy_compress_gbps(size_t n, int iters, int dev) {
+  const gpu::GpuApi* api = gpu::api();
+  if (api == nullptr) return -1;
+  void* d_in = api->alloc_hbm((uint32_t)n, dev);
+  size_t cap = n + n / 3 + 64;
+  void* d_comp = api->alloc_hbm((uint32_t)cap, dev);
+  if (d_in == nullptr || d_comp == nullptr) return -1;
+  api->fill(d_in, n, 0x6162636461626364ULL, dev);  // compressible
+  size_t comp_len;
+  api->snappy_compress(d_in, n, d_comp, cap, &comp_len, dev);
+  int64_t t0 = monotonic_time_us();
+  for (int i = 0; i < iters; ++i) api->snappy_compress(d_in, n, d_comp, cap, &comp_len, dev);
+  int64_t el = monotonic_time_us() - t0;
+  api->free_hbm(d_in, (uint32_t)n, dev);
+  api->free_hbm(d_comp, (uint32_t)cap, dev);
+  return (double)n * iters / el / 1000.0;
+}
+
+}  // namespace gputest
+}  // namespace bam

@@ -104,6 +104,8 @@ bool pinned_roundtrip(size_t n);
 double crc_gbps(size_t n, int iters, int dev);
 double gather_gbps(size_t total, uint32_t block, int iters, int dev);
 double d2h_gbps(size_t n, int iters, int dev);
+bool snappy_cross_check(size_t n, int mode, int dev);
+double snappy_compress_gbps(size_t n, int iters, int dev);
 }  // namespace gputest
 }  // namespace bam
 
@@ -129,6 +131,10 @@ void bind_gpu(py::module_& m) {
         py::arg("iters") = 10, py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
   g.def("d2h_gbps", &bam::gputest::d2h_gbps, py::arg("n"), py::arg("iters") = 10,
         py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
+  g.def("snappy_cross_check", &bam::gputest::snappy_cross_check, py::arg("n"),
+        py::arg("mode") = 0, py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
+  g.def("snappy_compress_gbps", &bam::gputest::snappy_compress_gbps, py::arg("n"),
+        py::arg("iters") = 5, py::arg("dev") = 0, py::call_guard<py::gil_scoped_release>());
 }
 
 // ---- snappy host codec bindings (oracle for the gfx950 kernel) ----

@@ -68,3 +68,10 @@ def test_hbm_echo_bench_smoke():
     res = r.echo_bench(addr, 16384, 8, 500, 20000, "EchoService.EchoHbm", True)
     assert res["errors"] == 0, res
     assert res["qps"] > 10
+
+
+@pytest.mark.parametrize("n,mode", [(1000, 0), (65536, 0), (1 << 20, 0),
+                                     ((1 << 22) + 777, 0), (1 << 20, 1)])
+def test_gpu_snappy_cross_check(n, mode):
+    """GPU-compressed streams decompress with the host codec and vice versa."""
+    assert g.snappy_cross_check(n, mode, 0)
