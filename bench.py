@@ -66,7 +66,7 @@ def main():
         res = b.core.rpc.echo_bench(addr, args.payload, args.concurrency,
                                     args.calls_per_step, 30000, method, hbm_req)
         if res["errors"]:
-            raise RuntimeError("bench errors: %s" % res)
+            raise RuntimeError("bench errors: n=%s first=%s" % (res["errors"], res.get("first_error")))
         return res
 
     for _ in range(args.warmup):

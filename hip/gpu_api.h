@@ -37,6 +37,11 @@ int bam_gpu_gather(void* dst_dev, const void* const* srcs, const size_t* lens, i
 int bam_gpu_scatter(void* const* dsts, const size_t* lens, int nspans, const void* src_dev,
                     int dev);
 
+// Stages N device spans into one contiguous host buffer (device gather +
+// single D2H). The write path's staging ring.
+int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs, const size_t* lens,
+                           int nspans, int dev);
+
 // Snappy (standard wire format) on device buffers. Returns 0 on success.
 int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_dev, size_t dst_cap,
                             size_t* out_len, int dev);

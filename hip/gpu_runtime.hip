@@ -216,3 +216,49 @@ extern "C" int bam_gpu_fill(void* dst_dev, size_t n, uint64_t pattern, int dev) 
   }
   return 0;
 }
+
+// ---------------- scattered-span D2H staging ----------------
+// Stages N device spans into one contiguous HOST buffer: device-side
+// gather into a persistent scratch (≈4 TB/s span-copy kernel), then ONE
+// D2H transfer. This is the pinned-staging-ring leg of the IOBuf write
+// path (cut_into_file_descriptor with HBM-resident blocks).
+extern "C" int bam_gpu_gather(void*, const void* const*, const size_t*, int, int);
+
+namespace {
+struct StageScratch {
+  void* dev = nullptr;
+  size_t cap = 0;
+};
+StageScratch g_stage[kMaxDev];
+std::mutex g_stage_mu;
+}  // namespace
+
+extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
+                                      const size_t* lens, int nspans, int dev) {
+  size_t total = 0;
+  for (int i = 0; i < nspans; ++i) total += lens[i];
+  if (total == 0) return 0;
+  ScopedDevice sd(dev);
+  if (nspans == 1) {
+    hipError_t e = hipMemcpy(host_dst, srcs[0], total, hipMemcpyDeviceToHost);
+    return e == hipSuccess ? 0 : -1;
+  }
+  void* scratch;
+  {
+    std::lock_guard<std::mutex> lk(g_stage_mu);
+    StageScratch& s = g_stage[dev < kMaxDev ? dev : 0];
+    if (s.cap < total) {
+      if (s.dev != nullptr) hipFree(s.dev);
+      s.cap = total * 2;
+      if (hipMalloc(&s.dev, s.cap) != hipSuccess) {
+        s.dev = nullptr;
+        s.cap = 0;
+        return -1;
+      }
+    }
+    scratch = s.dev;
+  }
+  if (bam_gpu_gather(scratch, srcs, lens, nspans, dev) != 0) return -1;
+  hipError_t e = hipMemcpy(host_dst, scratch, total, hipMemcpyDeviceToHost);
+  return e == hipSuccess ? 0 : -1;
+}

@@ -59,6 +59,8 @@ void do_load() {
   g_api.scatter =
       (int (*)(void* const*, const size_t*, int, const void*, int))must_sym(h, "bam_gpu_scatter");
   g_api.fill = (int (*)(void*, size_t, uint64_t, int))must_sym(h, "bam_gpu_fill");
+  g_api.gather_to_host = (int (*)(void*, const void* const*, const size_t*, int,
+                                  int))must_sym(h, "bam_gpu_gather_to_host");
   g_api.snappy_compress = (int (*)(const void*, size_t, void*, size_t, size_t*, int))must_sym(
       h, "bam_gpu_snappy_compress");
   g_api.snappy_decompress = (int (*)(const void*, size_t, void*, size_t, size_t*,
@@ -79,6 +81,7 @@ void do_load() {
       g_api.memcpy_res(dst, (int)dres, ddev, src, (int)sres, sdev, n);
     }};
     set_byte_mover(mover);
+    set_gather_to_host(g_api.gather_to_host);
     LOG(INFO) << "brpc_amd HIP runtime loaded: " << g_ndev << " GPU(s)";
   }
 }

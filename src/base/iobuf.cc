@@ -17,7 +17,7 @@ namespace bam {
 static std::atomic<size_t> g_nblock{0};
 static std::atomic<size_t> g_blockmem{0};
 
-static BlockMemFns g_allocators[3] = {{nullptr, nullptr}, {nullptr, nullptr}, {nullptr, nullptr}};
+BlockMemFns g_allocators[3] = {{nullptr, nullptr}, {nullptr, nullptr}, {nullptr, nullptr}};
 static ByteMoverFns g_mover = {nullptr};
 
 void set_block_allocator(Residency res, BlockMemFns fns) { g_allocators[res] = fns; }
@@ -544,15 +544,59 @@ bool IOBuf::equals(const std::string& s) const {
 
 static const size_t kMaxIov = 64;
 
+// Staging-ring hook installed by the HIP loader: gathers scattered device
+// spans into one contiguous host buffer (device gather + single D2H).
+static GatherToHostFn g_gather_to_host = nullptr;
+void set_gather_to_host(GatherToHostFn fn) { g_gather_to_host = fn; }
+
+namespace {
+// Per-thread pinned bounce buffer for the HBM->wire staging path. Pinned
+// memory doubles D2H bandwidth vs pageable and is reused across calls.
+struct TlsBounce {
+  char* buf = nullptr;
+  size_t cap = 0;
+  bool pinned = false;
+  char* get(size_t need) {
+    if (cap >= need) return buf;
+    release();
+    cap = need < (1u << 20) ? (1u << 20) : need;
+    if (has_block_allocator(RES_PINNED)) {
+      buf = (char*)g_allocators[RES_PINNED].alloc((uint32_t)cap, 0);
+      pinned = buf != nullptr;
+    }
+    if (buf == nullptr) {
+      buf = (char*)malloc(cap);
+      pinned = false;
+    }
+    return buf;
+  }
+  void release() {
+    if (buf == nullptr) return;
+    if (pinned) {
+      g_allocators[RES_PINNED].dealloc(buf, (uint32_t)cap, 0);
+    } else {
+      free(buf);
+    }
+    buf = nullptr;
+    cap = 0;
+  }
+  ~TlsBounce() { release(); }
+};
+thread_local TlsBounce tls_bounce;
+}  // namespace
+
 ssize_t IOBuf::cut_into_file_descriptor(int fd, size_t size_hint) {
   if (empty()) return 0;
   struct iovec iov[kMaxIov];
   size_t niov = 0;
   size_t queued = 0;
-  // Host-addressable fast path; HBM blocks require staging (installed by
-  // the HIP library as a byte mover -> bounce buffer path, see gpu_api).
-  char* bounce = nullptr;
-  size_t bounce_cap = 0, bounce_used = 0;
+  // Collect spans; HBM spans are staged in ONE device-gather + D2H below.
+  const void* hbm_srcs[kMaxIov];
+  size_t hbm_lens[kMaxIov];
+  size_t hbm_iov_idx[kMaxIov];
+  int nhbm = 0;
+  size_t hbm_total = 0;
+  int hbm_dev = 0;
   for (uint32_t i = 0; i < count_ && niov < kMaxIov && queued < size_hint; ++i) {
     const BlockRef& r = ref_at(i);
     size_t take = std::min<size_t>(r.length, size_hint - queued);
@@ -560,29 +604,44 @@ ssize_t IOBuf::cut_into_file_descriptor(int fd, size_t size_hint) {
       iov[niov].iov_base = r.block->data + r.offset;
       iov[niov].iov_len = take;
     } else {
-      if (bounce == nullptr) {
-        bounce_cap = 1u << 20;
-        bounce = (char*)malloc(bounce_cap);
-        if (!bounce) {
-          errno = ENOMEM;
-          return -1;
-        }
-      }
-      if (bounce_used + take > bounce_cap) break;
-      move_bytes(bounce + bounce_used, RES_HOST, -1, r.block->data + r.offset, RES_HBM,
-                 r.block->dev, take);
-      iov[niov].iov_base = bounce + bounce_used;
+      hbm_srcs[nhbm] = r.block->data + r.offset;
+      hbm_lens[nhbm] = take;
+      hbm_iov_idx[nhbm] = niov;
+      hbm_dev = r.block->dev;
+      ++nhbm;
+      hbm_total += take;
+      iov[niov].iov_base = nullptr;  // patched after staging
       iov[niov].iov_len = take;
-      bounce_used += take;
     }
     ++niov;
     queued += take;
   }
+  if (nhbm > 0) {
+    char* bounce = tls_bounce.get(hbm_total);
+    if (bounce == nullptr) {
+      errno = ENOMEM;
+      return -1;
+    }
+    if (g_gather_to_host != nullptr) {
+      if (g_gather_to_host(bounce, hbm_srcs, hbm_lens, nhbm, hbm_dev) != 0) {
+        errno = EIO;
+        return -1;
+      }
+    } else {
+      size_t off = 0;
+      for (int k = 0; k < nhbm; ++k) {
+        move_bytes(bounce + off, RES_HOST, -1, hbm_srcs[k], RES_HBM, hbm_dev, hbm_lens[k]);
+        off += hbm_lens[k];
+      }
+    }
+    size_t off = 0;
+    for (int k = 0; k < nhbm; ++k) {
+      iov[hbm_iov_idx[k]].iov_base = bounce + off;
+      off += hbm_lens[k];
+    }
+  }
   ssize_t nw = ::writev(fd, iov, (int)niov);
-  int saved_errno = errno;
-  if (bounce) free(bounce);
   if (nw > 0) pop_front((size_t)nw);
-  errno = saved_errno;
   return nw;
 }
 

@@ -36,6 +36,10 @@ bool has_block_allocator(Residency res);
 
 // Byte movers between residencies; defaults to memcpy for host<->host and
 // aborts for device until the HIP library installs real ones.
+typedef int (*GatherToHostFn)(void* host_dst, const void* const* srcs, const size_t* lens,
+                              int nspans, int dev);
+void set_gather_to_host(GatherToHostFn fn);
+
 struct ByteMoverFns {
   // dst_res/src_res in {HOST,PINNED,HBM}; devices are -1 for host memory.
   void (*copy)(void* dst, Residency dst_res, int dst_dev, const void* src, Residency src_res,

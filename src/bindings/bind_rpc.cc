@@ -11,6 +11,7 @@ struct BenchResult {
   int64_t p50_us, p90_us, p99_us, p999_us, max_us, avg_us;
   int64_t errors;
   int64_t total;
+  std::string first_error;
 };
 int start_echo_server(int port);
 int echo_once(const std::string& addr, const std::string& payload, int timeout_ms,
@@ -82,6 +83,7 @@ void bind_rpc(py::module_& m) {
           d["max_us"] = b.max_us;
           d["avg_us"] = b.avg_us;
           d["errors"] = b.errors;
+          d["first_error"] = b.first_error;
           d["total"] = b.total;
           return d;
         },

@@ -5,6 +5,7 @@
 #include <algorithm>
 #include <atomic>
 #include <memory>
+#include <mutex>
 #include <vector>
 
 #include "base/fast_rand.h"
@@ -134,6 +135,7 @@ struct BenchResult {
   int64_t p50_us = 0, p90_us = 0, p99_us = 0, p999_us = 0, max_us = 0, avg_us = 0;
   int64_t errors = 0;
   int64_t total = 0;
+  std::string first_error;
 };
 
 namespace {
@@ -147,6 +149,8 @@ struct BenchWorkerArg {
   std::string method;
   bool hbm_request;
   CountdownEvent* done_event;
+  std::mutex* err_mu;
+  std::string* first_error;
 };
 
 void bench_worker(void* raw) {
@@ -164,6 +168,12 @@ void bench_worker(void* raw) {
     a->channel->CallMethod(a->method, &cntl, &request, &response, nullptr);
     if (cntl.Failed() || response.size() != a->payload.size()) {
       a->errors->fetch_add(1, std::memory_order_relaxed);
+      std::lock_guard<std::mutex> lk(*a->err_mu);
+      if (a->first_error->empty()) {
+        *a->first_error = cntl.Failed()
+                              ? std::to_string(cntl.ErrorCode()) + ":" + cntl.ErrorText()
+                              : "short response " + std::to_string(response.size());
+      }
     }
     (*a->latencies)[idx - 1] = cntl.latency_us();
   }
@@ -186,8 +196,9 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
   std::atomic<int64_t> errors{0};
   std::vector<int64_t> latencies(total_calls, 0);
   CountdownEvent done_event(concurrency);
-  BenchWorkerArg arg{&channel,  &remaining,  &latencies, &errors,
-                     payload,   method,      hbm_request, &done_event};
+  std::mutex err_mu;
+  BenchWorkerArg arg{&channel,  &remaining,  &latencies,  &errors,   payload,
+                     method,    hbm_request, &done_event, &err_mu,   &res.first_error};
   int64_t t0 = monotonic_time_us();
   for (int i = 0; i < concurrency; ++i) {
     fiber_t th;

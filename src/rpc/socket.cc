@@ -196,15 +196,25 @@ void Socket::remove_pending_session(uint64_t sid) {
 
 // ---------------- write path ----------------
 
+static void release_write_request(Socket::WriteRequest* p, int err) {
+  if (err != 0) {
+    if (p->id_wait != 0) session_error(p->id_wait, err);
+    for (uint64_t id : p->merged_ids) session_error(id, err);
+  }
+  p->id_wait = 0;
+  p->merged_ids.clear();
+  p->data.clear();
+  return_object(p);
+}
+
 void Socket::NotifyWriteFailure(WriteRequest* head, int err) {
   // The chain may contain sentinel next pointers being published; spin.
+  if (err == 0) err = EPIPE;
   WriteRequest* p = head;
   while (p != nullptr) {
     WriteRequest* nx;
     while ((nx = p->next.load(std::memory_order_acquire)) == kWriteSentinel) sched_yield();
-    if (p->id_wait != 0) session_error(p->id_wait, err != 0 ? err : EPIPE);
-    p->data.clear();
-    return_object(p);
+    release_write_request(p, err);
     p = nx;
   }
 }
@@ -226,9 +236,7 @@ void Socket::ReleaseAllWriteRequests(WriteRequest* fifo_head, int err) {
     while (p != last) {
       WriteRequest* nx;
       while ((nx = p->next.load(std::memory_order_acquire)) == kWriteSentinel) sched_yield();
-      if (p->id_wait != 0) session_error(p->id_wait, err);
-      p->data.clear();
-      return_object(p);
+      release_write_request(p, err);
       p = nx;
     }
   }
@@ -236,9 +244,7 @@ void Socket::ReleaseAllWriteRequests(WriteRequest* fifo_head, int err) {
   WriteRequest* p = fifo_head;
   while (p != nullptr) {
     WriteRequest* nx = p->next.load(std::memory_order_acquire);
-    if (p->id_wait != 0) session_error(p->id_wait, err);
-    p->data.clear();
-    return_object(p);
+    release_write_request(p, err);
     p = nx;
   }
 }
@@ -281,8 +287,7 @@ Socket::WriteRequest* Socket::PopNextRequest(WriteRequest* done) {
     fifo = p;
     p = nx;
   }
-  done->data.clear();
-  return_object(done);
+  release_write_request(done, 0);
   return fifo;
 }
 
@@ -295,7 +300,23 @@ int Socket::DoWrite(WriteRequest* req) {
       return -1;
     }
     if (connecting_.load(std::memory_order_acquire)) {
-      if (wait_epoll_out(monotonic_time_us() + 30 * 1000000) != 0) continue;  // ->Failed
+      // Short re-polling waits: wait_epoll_out's poll() pre-check makes a
+      // lost EPOLLOUT edge cost at most one backstop interval.
+      int64_t deadline = monotonic_time_us() + 30 * 1000000;
+      struct pollfd pfd;
+      pfd.fd = fd();
+      pfd.events = POLLOUT;
+      pfd.revents = 0;
+      while (::poll(&pfd, 1, 0) == 0) {
+        if (Failed()) break;
+        if (monotonic_time_us() > deadline) {
+          SetFailed(ETIMEDOUT, "connect timed out");
+          break;
+        }
+        wait_epoll_out(monotonic_time_us() + 50000);  // 50ms backstop
+        pfd.revents = 0;
+      }
+      if (Failed()) continue;
       int err = 0;
       socklen_t len = sizeof(err);
       getsockopt(fd(), SOL_SOCKET, SO_ERROR, &err, &len);
@@ -305,6 +326,20 @@ int Socket::DoWrite(WriteRequest* req) {
       }
       get_local_side(fd(), &local_side_);
       connecting_.store(false, std::memory_order_release);
+    }
+    // Coalesce queued successors into one writev/staging batch (parity:
+    // reference DoWrite cutting up to NWMAX iovecs across requests). The
+    // FIFO tail (the node write_head_ may point at) is never merged.
+    for (;;) {
+      WriteRequest* nx = cur->next.load(std::memory_order_acquire);
+      if (nx == nullptr || nx == kWriteSentinel) break;
+      if (nx->next.load(std::memory_order_acquire) == nullptr) break;  // tail
+      if (cur->data.size() + nx->data.size() > (1u << 20)) break;
+      cur->data.append(std::move(nx->data));
+      if (nx->id_wait != 0) cur->merged_ids.push_back(nx->id_wait);
+      for (uint64_t id : nx->merged_ids) cur->merged_ids.push_back(id);
+      cur->next.store(nx->next.load(std::memory_order_acquire), std::memory_order_relaxed);
+      release_write_request(nx, 0);
     }
     ssize_t nw = cur->data.cut_into_file_descriptor(fd());
     if (nw < 0) {
@@ -322,8 +357,7 @@ int Socket::DoWrite(WriteRequest* req) {
     out_messages.fetch_add(1, std::memory_order_relaxed);
     WriteRequest* nx = cur->next.load(std::memory_order_acquire);
     if (nx != nullptr && nx != kWriteSentinel) {
-      cur->data.clear();
-      return_object(cur);
+      release_write_request(cur, 0);
       cur = nx;
     } else {
       cur = PopNextRequest(cur);  // nullptr when drained
@@ -355,6 +389,7 @@ int Socket::Write(IOBuf* data, const WriteOptions* opt) {
   }
   WriteRequest* req = get_object<WriteRequest>();
   req->data.clear();
+  req->merged_ids.clear();
   req->data.swap(*data);
   req->id_wait = opt->id_wait;
   req->socket = this;
@@ -403,9 +438,13 @@ void Socket::RunInputEventsFiber(void* arg) {
       s->input_events_.store(0, std::memory_order_release);
       break;
     }
+    // Capture the event count BEFORE draining: if more edges arrive during
+    // the drain the CAS below fails and we drain again. Capturing after
+    // the drain would fold a mid-drain edge into the reset and lose its
+    // data forever (edge-triggered epoll never re-notifies).
+    int before = s->input_events_.load(std::memory_order_acquire);
     s->run_edge_callback();  // drains until EAGAIN
-    int v = s->input_events_.load(std::memory_order_acquire);
-    if (s->input_events_.compare_exchange_strong(v, 0, std::memory_order_acq_rel)) break;
+    if (s->input_events_.compare_exchange_strong(before, 0, std::memory_order_acq_rel)) break;
   }
   s->ReleaseRef();
 }

@@ -68,6 +68,9 @@ class Socket {
     std::atomic<WriteRequest*> next{nullptr};
     uint64_t id_wait = 0;  // session to error on write failure
     Socket* socket = nullptr;
+    // id_waits of requests coalesced into this one by KeepWrite (failure
+    // conduction for merged writes).
+    std::vector<uint64_t> merged_ids;
   };
 
   struct WriteOptions {

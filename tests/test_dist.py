@@ -2,6 +2,7 @@
 rendezvous/aggregation path the driver uses with RCCL on GPUs)."""
 import json
 import os
+import socket
 import subprocess
 import sys
 
@@ -11,16 +12,20 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 def test_bench_two_ranks_gloo():
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", "--nproc-per-node=2",
-        "--master-addr", "127.0.0.1", "--master-port", "29511",
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
         "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
         "--calls-per-step", "500", "--concurrency", "8",
     ]
     out = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True, text=True,
                          timeout=300)
-    assert out.returncode == 0, out.stderr[-2000:]
+    assert out.returncode == 0, out.stderr[-3000:] + out.stdout[-1000:]
     line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
     res = json.loads(line)
     assert res["metric"] == "echo_qps"
