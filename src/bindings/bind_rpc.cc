@@ -102,6 +102,7 @@ int partition_test(const std::vector<int>& ports, std::string* merged);
 int lb_spread_test(const std::string& lb_name, const std::vector<int>& ports, int ncalls);
 int compressed_echo_test(const std::string& addr, const std::string& payload,
                          int compress_type, std::string* response_out);
+int64_t backup_request_test(int slow_port, int fast_port, int backup_ms, int calls);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -142,6 +143,8 @@ void bind_rpc_combo(py::module_& m) {
           return py::make_tuple(rc, py::bytes(merged));
         });
   r.def("lb_spread", &bam::rpctest::lb_spread_test,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("backup_request", &bam::rpctest::backup_request_test,
         py::call_guard<py::gil_scoped_release>());
   r.def("compressed_echo",
         [](const std::string& addr, py::bytes payload, int ctype) {

@@ -90,6 +90,17 @@ void EndRPC(Controller* cntl, SessionId locked_id) {
 // on_error handler: runs LOCKED; must unlock or destroy.
 static int HandleSessionError(SessionId id, void* data, int error_code) {
   Controller* cntl = (Controller*)data;
+  if (error_code == EBACKUPREQUEST) {
+    // Backup-request trigger: launch a second attempt; the earlier one
+    // stays pending and whichever response arrives first wins.
+    if (cntl->retry_count_ < cntl->max_retry_) {
+      ++cntl->retry_count_;
+      session_bump_slot(id);
+      IssueRPC(cntl);
+    }
+    session_unlock(id);
+    return 0;
+  }
   if (!session_is_current(id)) {
     // failure of a stale attempt (e.g. old socket died after retry)
     session_unlock(id);
@@ -115,6 +126,11 @@ static int HandleSessionError(SessionId id, void* data, int error_code) {
 static void TimeoutCb(void* a, void* /*b*/) {
   SessionId cid = (SessionId)(uintptr_t)a;
   session_error(cid, ERPCTIMEDOUT);
+}
+
+static void BackupRequestCb(void* a, void* /*b*/) {
+  SessionId cid = (SessionId)(uintptr_t)a;
+  session_error(cid, EBACKUPREQUEST);
 }
 
 void IssueRPC(Controller* cntl) {
@@ -170,9 +186,15 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->cid_ = cid;
   cntl->call.cid = cid;
 
+  if (cntl->backup_request_ms_ == -1) cntl->backup_request_ms_ = options_.backup_request_ms;
   if (cntl->timeout_ms_ > 0) {
     cntl->call.timeout_timer = timer_add(cntl->start_us_ + cntl->timeout_ms_ * 1000,
                                          TimeoutCb, (void*)(uintptr_t)cid, nullptr);
+  }
+  if (cntl->backup_request_ms_ > 0 &&
+      (cntl->timeout_ms_ <= 0 || cntl->backup_request_ms_ < cntl->timeout_ms_)) {
+    timer_add(cntl->start_us_ + cntl->backup_request_ms_ * 1000, BackupRequestCb,
+              (void*)(uintptr_t)cid, nullptr);
   }
   IssueRPC(cntl);
   if (done == nullptr) {

@@ -326,11 +326,9 @@ static void ProcessStdResponse(InputMessageBase* msg_base) {
     return;
   }
   Controller* cntl = (Controller*)data;
-  if (!session_is_current(cid)) {
-    session_unlock(cid);  // response for a stale retry attempt
-    delete msg;
-    return;
-  }
+  // Responses from ANY live attempt complete the call (backup requests /
+  // retries race; first response wins — parity with the reference's
+  // versioned-correlation semantics).
   {
     SocketUniquePtr sock;
     if (Socket::Address(msg->socket_id, &sock) == 0) sock->remove_pending_session(cid);

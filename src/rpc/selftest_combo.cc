@@ -114,3 +114,32 @@ int lb_spread_test(const std::string& lb_name, const std::vector<int>& ports, in
 
 }  // namespace rpctest
 }  // namespace bam
+
+namespace bam {
+namespace rpctest {
+
+// Backup request: first attempt lands on a slow server; the backup fires
+// after backup_ms and (via rr) reaches the fast server. Returns max
+// latency over `calls` sync calls.
+int64_t backup_request_test(int slow_port, int fast_port, int backup_ms, int calls) {
+  std::string url = "list://127.0.0.1:" + std::to_string(slow_port) + ",127.0.0.1:" +
+                    std::to_string(fast_port);
+  Channel chan;
+  ChannelOptions copt;
+  copt.timeout_ms = 5000;
+  copt.backup_request_ms = backup_ms;
+  if (chan.Init(url.c_str(), "rr", &copt) != 0) return -1;
+  int64_t max_lat = 0;
+  for (int i = 0; i < calls; ++i) {
+    Controller cntl;
+    IOBuf request, response;
+    request.append("1000");  // slow server sleeps 1000 ms
+    chan.CallMethod("EchoService.Sleep", &cntl, &request, &response, nullptr);
+    if (cntl.Failed()) return -cntl.ErrorCode();
+    if (cntl.latency_us() > max_lat) max_lat = cntl.latency_us();
+  }
+  return max_lat;
+}
+
+}  // namespace rpctest
+}  // namespace bam
