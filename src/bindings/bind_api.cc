@@ -1,0 +1,163 @@
+// Public Python API: Server with Python-defined services (handlers run on
+// the usercode pthread pool with the GIL — never on fiber workers) and a
+// Channel for client calls. This is the user-facing surface mirroring the
+// reference's Channel/Server/Controller API in Python.
+#include <pybind11/functional.h>
+#include <pybind11/pybind11.h>
+
+#include <memory>
+
+#include "bindings/bind.h"
+#include "rpc/channel.h"
+#include "rpc/controller.h"
+#include "rpc/parallel_channel.h"
+#include "rpc/server.h"
+#include "rpc/usercode_pool.h"
+
+namespace {
+
+using namespace bam;
+
+struct PyRpcError : public std::runtime_error {
+  int code;
+  PyRpcError(int c, const std::string& msg) : std::runtime_error(msg), code(c) {}
+};
+
+// Wraps a Python callable as a MethodFn: dispatched to the usercode pool,
+// runs under the GIL there, then done->Run() sends the response.
+MethodFn wrap_py_handler(py::function fn) {
+  // keep the callable alive via shared_ptr captured in the lambda
+  auto holder = std::make_shared<py::object>(std::move(fn));
+  return [holder](Controller* cntl, const IOBuf& request, IOBuf* response, Closure* done) {
+    std::string req_bytes = request.to_string();
+    std::string att_bytes = cntl->request_attachment().to_string();
+    SubmitUserCode([holder, cntl, req_bytes, att_bytes, response, done] {
+      py::gil_scoped_acquire gil;
+      try {
+        py::object result =
+            (*holder)(py::bytes(req_bytes), py::bytes(att_bytes));
+        if (py::isinstance<py::tuple>(result)) {
+          py::tuple t = result.cast<py::tuple>();
+          std::string body = t[0].cast<std::string>();
+          std::string att = t[1].cast<std::string>();
+          response->append(body);
+          cntl->response_attachment().append(att);
+        } else if (!result.is_none()) {
+          std::string body = result.cast<std::string>();
+          response->append(body);
+        }
+      } catch (const std::exception& e) {
+        cntl->SetFailed(EINTERNAL, std::string("python handler: ") + e.what());
+      }
+      {
+        py::gil_scoped_release rel;
+        done->Run();
+      }
+    });
+  };
+}
+
+class PyServer {
+ public:
+  PyServer() : server_(new Server) {}
+
+  void add_method(const std::string& service, const std::string& method, py::function fn) {
+    Service* svc;
+    auto it = services_.find(service);
+    if (it == services_.end()) {
+      svc = new Service(service);
+      services_[service] = svc;
+      server_->AddService(svc, SERVER_OWNS_SERVICE);
+    } else {
+      svc = it->second;
+    }
+    svc->AddMethod(method, wrap_py_handler(std::move(fn)));
+  }
+
+  int start(int port, int max_concurrency) {
+    ServerOptions opts;
+    opts.max_concurrency = max_concurrency;
+    if (server_->Start(port, &opts) != 0) throw std::runtime_error("Server.start failed");
+    return server_->listen_address().port;
+  }
+
+  void stop() { server_->Stop(0); }
+  bool running() const { return server_->IsRunning(); }
+  int port() const { return server_->listen_address().port; }
+  int64_t processed() const { return server_->nprocessed.load(); }
+
+ private:
+  Server* server_;  // leaked deliberately: sockets may still reference it
+  std::map<std::string, Service*> services_;
+};
+
+class PyChannel {
+ public:
+  PyChannel(const std::string& addr, const std::string& lb, int timeout_ms, int max_retry,
+            int backup_request_ms, int compress) {
+    ChannelOptions opts;
+    opts.timeout_ms = timeout_ms;
+    opts.max_retry = max_retry;
+    opts.backup_request_ms = backup_request_ms;
+    compress_ = (CompressType)compress;
+    int rc = lb.empty() ? channel_.Init(addr.c_str(), &opts)
+                        : channel_.Init(addr.c_str(), lb.c_str(), &opts);
+    if (rc != 0) throw std::runtime_error("Channel.Init failed for " + addr);
+  }
+
+  py::tuple call(const std::string& full_method, const std::string& request,
+                 const std::string& attachment, int timeout_ms, uint64_t log_id) {
+    Controller cntl;
+    if (timeout_ms > 0) cntl.set_timeout_ms(timeout_ms);
+    if (log_id != 0) cntl.set_log_id(log_id);
+    cntl.set_request_compress_type(compress_);
+    IOBuf req, resp;
+    req.append(request);
+    if (!attachment.empty()) cntl.request_attachment().append(attachment);
+    {
+      py::gil_scoped_release rel;
+      channel_.CallMethod(full_method, &cntl, &req, &resp, nullptr);
+    }
+    if (cntl.Failed()) throw PyRpcError(cntl.ErrorCode(), cntl.ErrorText());
+    return py::make_tuple(py::bytes(resp.to_string()),
+                          py::bytes(cntl.response_attachment().to_string()),
+                          cntl.latency_us());
+  }
+
+ private:
+  Channel channel_;
+  CompressType compress_ = COMPRESS_TYPE_NONE;
+};
+
+}  // namespace
+
+void bind_api(py::module_& m) {
+  static py::exception<PyRpcError> exc(m, "RpcError");
+  py::register_exception_translator([](std::exception_ptr p) {
+    try {
+      if (p) std::rethrow_exception(p);
+    } catch (const PyRpcError& e) {
+      // RpcError(code, message): args = (code, message)
+      PyErr_SetObject(exc.ptr(), py::make_tuple(e.code, e.what()).ptr());
+    }
+  });
+
+  py::class_<PyServer>(m, "Server")
+      .def(py::init<>())
+      .def("add_method", &PyServer::add_method, py::arg("service"), py::arg("method"),
+           py::arg("handler"))
+      .def("start", &PyServer::start, py::arg("port") = 0, py::arg("max_concurrency") = 0)
+      .def("stop", &PyServer::stop)
+      .def("running", &PyServer::running)
+      .def("port", &PyServer::port)
+      .def("processed", &PyServer::processed);
+
+  py::class_<PyChannel>(m, "Channel")
+      .def(py::init<const std::string&, const std::string&, int, int, int, int>(),
+           py::arg("addr"), py::arg("lb") = "", py::arg("timeout_ms") = 500,
+           py::arg("max_retry") = 3, py::arg("backup_request_ms") = -1,
+           py::arg("compress") = 0)
+      .def("call", &PyChannel::call, py::arg("method"), py::arg("request"),
+           py::arg("attachment") = std::string(), py::arg("timeout_ms") = 0,
+           py::arg("log_id") = 0);
+}

@@ -10,4 +10,5 @@ PYBIND11_MODULE(_core, m) {
   bind_rpc_combo(m);
   bind_rpc_stream(m);
   bind_snappy(m);
+  bind_api(m);
 }

@@ -1,0 +1,99 @@
+"""Python-facing Server/Channel API: Python-defined service handlers run on
+the usercode pthread pool (≙ reference usercode_in_pthread), never on
+fiber workers."""
+import threading
+
+import pytest
+
+import brpc_amd as b
+
+
+@pytest.fixture(scope="module")
+def py_server():
+    srv = b.Server()
+
+    def upper(request, attachment):
+        return request.upper()
+
+    def with_attachment(request, attachment):
+        return request[::-1], attachment + b"!"
+
+    def boom(request, attachment):
+        raise ValueError("kaboom")
+
+    threads_seen = set()
+
+    def which_thread(request, attachment):
+        threads_seen.add(threading.get_ident())
+        return b"ok"
+
+    srv.add_method("PySvc", "Upper", upper)
+    srv.add_method("PySvc", "WithAtt", with_attachment)
+    srv.add_method("PySvc", "Boom", boom)
+    srv.add_method("PySvc", "WhichThread", which_thread)
+    port = srv.start(0)
+    assert port > 0
+    return srv, port, threads_seen
+
+
+def test_python_handler_roundtrip(py_server):
+    _, port, _ = py_server
+    ch = b.Channel(f"127.0.0.1:{port}", timeout_ms=3000)
+    resp, att, lat = ch.call("PySvc.Upper", b"hello")
+    assert resp == b"HELLO"
+    assert lat > 0
+
+
+def test_python_handler_attachment(py_server):
+    _, port, _ = py_server
+    ch = b.Channel(f"127.0.0.1:{port}", timeout_ms=3000)
+    resp, att, _ = ch.call("PySvc.WithAtt", b"abc", attachment=b"bin")
+    assert resp == b"cba"
+    assert att == b"bin!"
+
+
+def test_python_handler_exception_propagates(py_server):
+    _, port, _ = py_server
+    ch = b.Channel(f"127.0.0.1:{port}", timeout_ms=3000)
+    with pytest.raises(b.RpcError) as ei:
+        ch.call("PySvc.Boom", b"x")
+    code, msg = ei.value.args
+    assert code == 2001  # EINTERNAL
+    assert "kaboom" in msg
+
+
+def test_handlers_not_on_fiber_workers(py_server):
+    _, port, threads_seen = py_server
+    ch = b.Channel(f"127.0.0.1:{port}", timeout_ms=3000)
+    for _ in range(10):
+        ch.call("PySvc.WhichThread", b"")
+    assert threading.get_ident() not in threads_seen  # ran on pool threads
+
+
+def test_concurrent_python_calls(py_server):
+    _, port, _ = py_server
+    ch = b.Channel(f"127.0.0.1:{port}", timeout_ms=5000)
+    errs = []
+
+    def worker():
+        try:
+            for i in range(50):
+                resp, _, _ = ch.call("PySvc.Upper", b"x%d" % i)
+                assert resp == b"X%d" % i
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    ts = [threading.Thread(target=worker) for _ in range(4)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert not errs
+
+
+def test_unknown_method_raises(py_server):
+    _, port, _ = py_server
+    ch = b.Channel(f"127.0.0.1:{port}", timeout_ms=2000)
+    with pytest.raises(b.RpcError) as ei:
+        ch.call("PySvc.Nope", b"x")
+    assert ei.value.args[0] == 1002  # ENOMETHOD
