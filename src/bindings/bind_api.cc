@@ -4,6 +4,7 @@
 // reference's Channel/Server/Controller API in Python.
 #include <pybind11/functional.h>
 #include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
 
 #include <memory>
 
@@ -12,6 +13,7 @@
 #include "rpc/controller.h"
 #include "rpc/parallel_channel.h"
 #include "rpc/server.h"
+#include "fiber/fiber.h"
 #include "rpc/usercode_pool.h"
 
 namespace {
@@ -160,4 +162,122 @@ void bind_api(py::module_& m) {
       .def("call", &PyChannel::call, py::arg("method"), py::arg("request"),
            py::arg("attachment") = std::string(), py::arg("timeout_ms") = 0,
            py::arg("log_id") = 0);
+}
+
+// ---- redis bindings ----
+#include "rpc/redis.h"
+
+namespace {
+
+py::object reply_to_py(const bam::RedisReply& r) {
+  using bam::RedisReply;
+  switch (r.type) {
+    case RedisReply::NIL:
+      return py::none();
+    case RedisReply::STATUS:
+      return py::str(r.str);
+    case RedisReply::ERROR:
+      throw PyRpcError(2002, r.str);
+    case RedisReply::INTEGER:
+      return py::int_(r.integer);
+    case RedisReply::STRING:
+      return py::bytes(r.str);
+    case RedisReply::ARRAY: {
+      py::list lst;
+      for (const auto& e : r.elements) lst.append(reply_to_py(e));
+      return lst;
+    }
+  }
+  return py::none();
+}
+
+bam::RedisReply py_to_reply(py::handle obj) {
+  using bam::RedisReply;
+  if (obj.is_none()) return RedisReply::Nil();
+  if (py::isinstance<py::bool_>(obj)) return RedisReply::Integer(obj.cast<bool>() ? 1 : 0);
+  if (py::isinstance<py::int_>(obj)) return RedisReply::Integer(obj.cast<int64_t>());
+  if (py::isinstance<py::bytes>(obj)) return RedisReply::Bulk(obj.cast<std::string>());
+  if (py::isinstance<py::str>(obj)) return RedisReply::Status(obj.cast<std::string>());
+  if (py::isinstance<py::list>(obj) || py::isinstance<py::tuple>(obj)) {
+    RedisReply arr;
+    arr.type = RedisReply::ARRAY;
+    for (py::handle e : obj) arr.elements.push_back(py_to_reply(e));
+    return arr;
+  }
+  return RedisReply::Error("ERR unconvertible python reply");
+}
+
+class PyRedisServer {
+ public:
+  PyRedisServer() : server_(new Server), service_(new RedisService) {}
+
+  void add_handler(const std::string& command, py::function fn) {
+    auto holder = std::make_shared<py::object>(std::move(fn));
+    service_->AddCommandHandler(
+        command, [holder](const std::vector<std::string>& args) -> RedisReply {
+          // Handlers run in fibers for redis (fast path) — but Python needs
+          // the GIL and must not run on a fiber: hop to the usercode pool
+          // and wait.
+          RedisReply out;
+          std::atomic<bool> done{false};
+          SubmitUserCode([&] {
+            py::gil_scoped_acquire gil;
+            try {
+              py::list pyargs;
+              for (const auto& a : args) pyargs.append(py::bytes(a));
+              out = py_to_reply((*holder)(pyargs));
+            } catch (const std::exception& e) {
+              out = RedisReply::Error(std::string("ERR ") + e.what());
+            }
+            done.store(true, std::memory_order_release);
+          });
+          while (!done.load(std::memory_order_acquire)) fiber_yield();
+          return out;
+        });
+  }
+
+  int start(int port) {
+    ServerOptions opts;
+    opts.redis_service = service_;
+    if (server_->Start(port, &opts) != 0) throw std::runtime_error("redis server start failed");
+    return server_->listen_address().port;
+  }
+
+ private:
+  Server* server_;
+  RedisService* service_;
+};
+
+}  // namespace
+
+void bind_redis(py::module_& m) {
+  py::class_<PyRedisServer>(m, "RedisServer")
+      .def(py::init<>())
+      .def("add_handler", &PyRedisServer::add_handler)
+      .def("start", &PyRedisServer::start, py::arg("port") = 0);
+
+  m.def("redis_call", [](const std::string& addr, const std::vector<std::string>& args,
+                         int timeout_ms) {
+    bam::ChannelOptions opts;
+    opts.timeout_ms = timeout_ms;
+    opts.protocol = "redis";
+    bam::policy::RegisterRedisProtocol();
+    bam::Channel ch;
+    if (ch.Init(addr.c_str(), &opts) != 0) throw std::runtime_error("redis channel init");
+    std::string cmd;
+    bam::EncodeRedisCommand(args, &cmd);
+    bam::Controller cntl;
+    bam::IOBuf req, resp;
+    req.append(cmd);
+    {
+      py::gil_scoped_release rel;
+      ch.CallMethod("redis.command", &cntl, &req, &resp, nullptr);
+    }
+    if (cntl.Failed()) throw PyRpcError(cntl.ErrorCode(), cntl.ErrorText());
+    std::string raw = resp.to_string();
+    bam::RedisReply reply;
+    ssize_t c = bam::ParseRedisValue(raw.data(), raw.size(), &reply);
+    if (c <= 0) throw std::runtime_error("bad redis reply");
+    return reply_to_py(reply);
+  }, py::arg("addr"), py::arg("args"), py::arg("timeout_ms") = 1000);
 }

@@ -11,4 +11,5 @@ PYBIND11_MODULE(_core, m) {
   bind_rpc_stream(m);
   bind_snappy(m);
   bind_api(m);
+  bind_redis(m);
 }

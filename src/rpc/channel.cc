@@ -6,6 +6,7 @@
 #include "base/time.h"
 #include "rpc/load_balancer.h"
 #include "rpc/policy/std_protocol.h"
+#include "rpc/redis.h"
 #include "rpc/rpcz.h"
 #include "rpc/socket_map.h"
 
@@ -18,6 +19,9 @@ int Channel::Init(EndPoint ep, const ChannelOptions* options) {
   server_ep_ = ep;
   single_server_ = true;
   policy::RegisterStdProtocol();
+  if (options_.protocol == "redis") policy::RegisterRedisProtocol();
+  protocol_index_ = FindProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
+  if (protocol_index_ < 0) return -1;
   return 0;
 }
 
@@ -34,6 +38,8 @@ int Channel::Init(const char* naming_url, const char* lb_name, const ChannelOpti
   if (options != nullptr) options_ = *options;
   single_server_ = false;
   policy::RegisterStdProtocol();
+  protocol_index_ = FindProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
+  if (protocol_index_ < 0) return -1;
   lb_ = LoadBalancerWithNaming::Create(naming_url, lb_name);
   if (lb_ == nullptr) return -1;
   return 0;
@@ -152,8 +158,14 @@ void IssueRPC(Controller* cntl) {
   cntl->remote_side_ = ep;
   SessionId current = session_current_id(cntl->call.cid);
   IOBuf packet;
-  policy::PackStdRequest(&packet, cntl, current);
+  const Protocol* proto = GetProtocol(cntl->call.protocol_index);
+  if (proto != nullptr && proto->pack_request != nullptr) {
+    proto->pack_request(&packet, cntl, current);
+  } else {
+    policy::PackStdRequest(&packet, cntl, current);
+  }
   sock->add_pending_session(current);
+  if (proto != nullptr && proto->client_pipelined) sock->push_pipeline(current);
   cntl->call.pending_socket = sock->id();
   Socket::WriteOptions wo;
   wo.id_wait = current;
@@ -169,6 +181,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   if (request != nullptr) cntl->call.request_buf = *request;  // zero-copy ref share
   cntl->call.response = response;
   cntl->call.done = done;
+  cntl->call.protocol_index = protocol_index_;
   if (single_server_) {
     cntl->call.server_ep = server_ep_;
     cntl->call.lb = nullptr;

@@ -52,6 +52,7 @@ class Channel : public ChannelBase {
   ChannelOptions options_;
   EndPoint server_ep_;
   bool single_server_ = false;
+  int protocol_index_ = -1;  // resolved from options_.protocol
   std::shared_ptr<LoadBalancerWithNaming> lb_;
 };
 

@@ -84,6 +84,7 @@ class Controller {
     class LoadBalancer* lb = nullptr;
     class SubChannelCtx* sub_ctx = nullptr;  // ParallelChannel bookkeeping
     uint64_t stream_id = 0;                  // client-created stream (StreamCreate)
+    int protocol_index = -1;                 // wire protocol for this call
   };
   Call call;
 

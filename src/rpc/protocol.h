@@ -38,6 +38,9 @@ struct ParseResult {
   static ParseResult make_error(ParseErrorCode e) { return ParseResult{e, nullptr}; }
 };
 
+class Controller;
+typedef uint64_t SessionId2;  // mirrors fiber SessionId without the include
+
 struct Protocol {
   // Cuts one complete message from `source` (the socket's read buffer).
   ParseResult (*parse)(IOBuf* source, Socket* sock, bool read_eof) = nullptr;
@@ -45,6 +48,11 @@ struct Protocol {
   void (*process_request)(InputMessageBase* msg) = nullptr;
   // Runs in a fiber; must delete/recycle msg. Client side.
   void (*process_response)(InputMessageBase* msg) = nullptr;
+  // Client-side request packing (header+meta+payload). nullptr = std only.
+  void (*pack_request)(IOBuf* out, Controller* cntl, uint64_t correlation_id) = nullptr;
+  // True for protocols without correlation ids: responses match requests
+  // FIFO on the connection (redis/memcache pipelining).
+  bool client_pipelined = false;
   // True if this protocol can appear on server connections.
   bool support_server = false;
   bool support_client = false;

@@ -6,6 +6,7 @@
 #include "base/logging.h"
 #include "rpc/event_dispatcher.h"
 #include "rpc/policy/std_protocol.h"
+#include "rpc/redis.h"
 
 namespace bam {
 
@@ -73,6 +74,7 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   if (IsRunning()) return -1;
   if (opt != nullptr) options_ = *opt;
   policy::RegisterStdProtocol();
+  if (options_.redis_service != nullptr) policy::RegisterRedisProtocol();
   int listen_fd = tcp_listen(ep);
   if (listen_fd < 0) {
     LOG(ERROR) << "tcp_listen on " << endpoint2str(ep) << " failed";

@@ -42,10 +42,13 @@ class Service {
 
 enum ServiceOwnership { SERVER_OWNS_SERVICE, SERVER_DOESNT_OWN_SERVICE };
 
+class RedisService;
+
 struct ServerOptions {
   int idle_timeout_sec = -1;
   int max_concurrency = 0;          // 0 = unlimited
   bool has_builtin_services = true;
+  RedisService* redis_service = nullptr;  // serve RESP on the same port
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)
@@ -77,6 +80,7 @@ class Server {
   int max_concurrency() const { return options_.max_concurrency; }
   const ServerOptions& options() const { return options_; }
   InputMessenger* messenger() { return &messenger_; }
+  RedisService* redis_service() const { return options_.redis_service; }
 
   const std::map<std::string, Service*>& services() const { return services_; }
 

@@ -9,6 +9,7 @@
 #pragma once
 
 #include <atomic>
+#include <deque>
 #include <functional>
 #include <mutex>
 #include <vector>
@@ -169,6 +170,24 @@ class Socket {
 
   std::mutex pending_mu_;
   std::vector<uint64_t> pending_sessions_;
+
+ public:
+  // FIFO correlation queue for pipelined protocols (redis/memcache):
+  // requests push, responses pop in order.
+  void push_pipeline(uint64_t sid) {
+    std::lock_guard<std::mutex> lk(pending_mu_);
+    pipeline_q_.push_back(sid);
+  }
+  uint64_t pop_pipeline() {
+    std::lock_guard<std::mutex> lk(pending_mu_);
+    if (pipeline_q_.empty()) return 0;
+    uint64_t sid = pipeline_q_.front();
+    pipeline_q_.pop_front();
+    return sid;
+  }
+
+ private:
+  std::deque<uint64_t> pipeline_q_;
 
   friend class EventDispatcher;
 };
