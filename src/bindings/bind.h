@@ -12,3 +12,4 @@ void bind_rpc_stream(py::module_& m);
 void bind_snappy(py::module_& m);
 void bind_api(py::module_& m);
 void bind_redis(py::module_& m);
+void bind_util(py::module_& m);

@@ -159,3 +159,37 @@ void bind_snappy(py::module_& m) {
     return py::bytes(out);
   });
 }
+
+// ---- recordio + rpc_dump + flags bindings ----
+#include "base/flags.h"
+#include "base/recordio.h"
+#include "rpc/rpc_dump.h"
+
+void bind_util(py::module_& m) {
+  auto u = m.def_submodule("util");
+  py::class_<bam::RecordWriter>(u, "RecordWriter")
+      .def(py::init<const std::string&>())
+      .def("ok", &bam::RecordWriter::ok)
+      .def("write", [](bam::RecordWriter& w, py::bytes data) {
+        return w.Write(data.cast<std::string>());
+      })
+      .def("flush", &bam::RecordWriter::Flush);
+  py::class_<bam::RecordReader>(u, "RecordReader")
+      .def(py::init<const std::string&>())
+      .def("ok", &bam::RecordReader::ok)
+      .def("next", [](bam::RecordReader& r) -> py::object {
+        std::string payload;
+        if (!r.Next(&payload)) return py::none();
+        return py::bytes(payload);
+      })
+      .def("last_error", &bam::RecordReader::last_error);
+  u.def("set_flag", &bam::flags::SetFlagValue);
+  u.def("get_flag", &bam::flags::GetFlagValue);
+  u.def("rpc_dump_count", &bam::rpc_dump::sampled_count);
+  u.def("decode_dump_record", [](py::bytes rec) -> py::object {
+    std::string service, method, body;
+    if (!bam::rpc_dump::DecodeSample(rec.cast<std::string>(), &service, &method, &body))
+      return py::none();
+    return py::make_tuple(service, method, py::bytes(body));
+  });
+}

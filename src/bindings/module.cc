@@ -12,4 +12,5 @@ PYBIND11_MODULE(_core, m) {
   bind_snappy(m);
   bind_api(m);
   bind_redis(m);
+  bind_util(m);
 }

@@ -9,6 +9,7 @@
 #include "rpc/server.h"
 #include "rpc/compress.h"
 #include "rpc/policy/http_protocol.h"
+#include "rpc/rpc_dump.h"
 #include "rpc/rpcz.h"
 #include "rpc/stream.h"
 #include "rpc/wire.h"
@@ -311,6 +312,7 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
     return;
   }
   server->concurrency.fetch_add(1, std::memory_order_relaxed);
+  rpc_dump::SampleRequest(msg->meta.service_name, msg->meta.method_name, req_data);
   (*fn)(cntl, req_data, resp, done);
   delete msg;
 }
