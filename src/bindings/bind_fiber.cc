@@ -13,6 +13,7 @@ bool butex_timeout_test();
 int64_t mutex_test(int nfibers, int iters);
 bool countdown_test(int n);
 bool timer_test();
+bool fiber_key_test();
 }  // namespace selftest
 }  // namespace bam
 
@@ -31,6 +32,7 @@ void bind_fiber(py::module_& m) {
   f.def("countdown_test", &bam::selftest::countdown_test,
         py::call_guard<py::gil_scoped_release>());
   f.def("timer_test", &bam::selftest::timer_test, py::call_guard<py::gil_scoped_release>());
+  f.def("key_test", &bam::selftest::fiber_key_test, py::call_guard<py::gil_scoped_release>());
   f.def("concurrency", &bam::fiber_get_concurrency);
   f.def("set_concurrency", &bam::fiber_set_concurrency);
   f.def("count_created", &bam::fiber_count_created);

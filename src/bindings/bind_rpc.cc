@@ -103,6 +103,8 @@ int lb_spread_test(const std::string& lb_name, const std::vector<int>& ports, in
 int compressed_echo_test(const std::string& addr, const std::string& payload,
                          int compress_type, std::string* response_out);
 int64_t backup_request_test(int slow_port, int fast_port, int backup_ms, int calls);
+int start_intercepted_echo_server(const std::string& magic_logid);
+int call_with_logid(const std::string& addr, uint64_t log_id, std::string* err);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -146,6 +148,17 @@ void bind_rpc_combo(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   r.def("backup_request", &bam::rpctest::backup_request_test,
         py::call_guard<py::gil_scoped_release>());
+  r.def("start_intercepted_server", &bam::rpctest::start_intercepted_echo_server,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("call_with_logid", [](const std::string& addr, uint64_t log_id) {
+    std::string err;
+    int rc;
+    {
+      py::gil_scoped_release rel;
+      rc = bam::rpctest::call_with_logid(addr, log_id, &err);
+    }
+    return py::make_tuple(rc, err);
+  });
   r.def("compressed_echo",
         [](const std::string& addr, py::bytes payload, int ctype) {
           char* ptr;

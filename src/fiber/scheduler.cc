@@ -11,6 +11,7 @@
 #include "base/logging.h"
 #include "base/resource_pool.h"
 #include "fiber/butex.h"
+#include "fiber/key.h"
 
 namespace bam {
 
@@ -131,6 +132,7 @@ void fiber_entry_fn(void* raw) {
     TaskGroup* g = current_task_group();
     FiberMeta* m = g->cur();
     m->fn(m->arg);
+    destroy_current_keytable();  // run fiber-local destructors
     // Fiber finished: invalidate the id, wake joiners, then free resources
     // from the next context (we cannot free the stack we stand on).
     g = current_task_group();  // may have migrated

@@ -26,6 +26,7 @@ struct FiberMeta {
   uint32_t index = 0;                  // ResourcePool id
   std::atomic<uint32_t> version{1};    // bumped at exit; fiber_t carries it
   std::atomic<int>* version_butex = nullptr;  // mirrors version; joiners wait here
+  void* keytable = nullptr;                   // fiber-local storage (fiber/key.h)
   bool is_main = false;
 };
 

@@ -210,3 +210,53 @@ bool timer_test() {
 
 }  // namespace selftest
 }  // namespace bam
+
+// ---- fiber-local storage ----
+
+#include "fiber/key.h"
+
+namespace bam {
+namespace selftest {
+
+namespace {
+std::atomic<int> g_dtor_runs{0};
+void key_dtor(void* p) {
+  g_dtor_runs.fetch_add(1);
+  delete (int*)p;
+}
+struct KeyArg {
+  fiber_key_t key;
+  bool ok;
+};
+void key_fiber(void* raw) {
+  KeyArg* a = (KeyArg*)raw;
+  a->ok = fiber_getspecific(a->key) == nullptr;  // fresh fiber: empty
+  int* v = new int(42);
+  fiber_setspecific(a->key, v);
+  fiber_yield();
+  a->ok = a->ok && fiber_getspecific(a->key) == v;  // survives reschedule
+}
+}  // namespace
+
+bool fiber_key_test() {
+  fiber_key_t key;
+  if (fiber_key_create(&key, key_dtor) != 0) return false;
+  g_dtor_runs.store(0);
+  KeyArg args[8];
+  fiber_t tids[8];
+  for (int i = 0; i < 8; ++i) {
+    args[i].key = key;
+    args[i].ok = false;
+    fiber_start_background(&tids[i], key_fiber, &args[i]);
+  }
+  for (int i = 0; i < 8; ++i) fiber_join(tids[i]);
+  bool all_ok = true;
+  for (int i = 0; i < 8; ++i) all_ok = all_ok && args[i].ok;
+  // destructors ran at each fiber's exit
+  all_ok = all_ok && g_dtor_runs.load() == 8;
+  fiber_key_delete(key);
+  return all_ok;
+}
+
+}  // namespace selftest
+}  // namespace bam

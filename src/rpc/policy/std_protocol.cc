@@ -311,6 +311,16 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
     done->Run();
     return;
   }
+  if (server->options().interceptor) {
+    int ec = 0;
+    std::string etext;
+    if (!server->options().interceptor(cntl, &ec, &etext)) {
+      cntl->SetFailed(ec != 0 ? ec : EREQUEST, etext.empty() ? "rejected by interceptor" : etext);
+      delete msg;
+      done->Run();
+      return;
+    }
+  }
   server->concurrency.fetch_add(1, std::memory_order_relaxed);
   rpc_dump::SampleRequest(msg->meta.service_name, msg->meta.method_name, req_data);
   (*fn)(cntl, req_data, resp, done);

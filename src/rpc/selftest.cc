@@ -255,3 +255,42 @@ int compressed_echo_test(const std::string& addr, const std::string& payload,
 
 }  // namespace rpctest
 }  // namespace bam
+
+// ---- interceptor (request admission hook) ----
+
+namespace bam {
+namespace rpctest {
+
+int start_intercepted_echo_server(const std::string& magic_logid) {
+  Server* server = new Server;
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  uint64_t magic = strtoull(magic_logid.c_str(), nullptr, 10);
+  opts.interceptor = [magic](Controller* cntl, int* ec, std::string* et) {
+    if (cntl->log_id() != magic) {
+      *ec = ERPCAUTH;
+      *et = "bad log_id credential";
+      return false;
+    }
+    return true;
+  };
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
+int call_with_logid(const std::string& addr, uint64_t log_id, std::string* err) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = 2000;
+  if (channel.Init(addr.c_str(), &opts) != 0) return -1;
+  Controller cntl;
+  cntl.set_log_id(log_id);
+  IOBuf request, response;
+  request.append("x");
+  channel.CallMethod("EchoService.Echo", &cntl, &request, &response, nullptr);
+  if (err != nullptr) *err = cntl.ErrorText();
+  return cntl.ErrorCode();
+}
+
+}  // namespace rpctest
+}  // namespace bam

@@ -44,11 +44,17 @@ enum ServiceOwnership { SERVER_OWNS_SERVICE, SERVER_DOESNT_OWN_SERVICE };
 
 class RedisService;
 
+// Runs before every request handler (parity: reference brpc/interceptor.h).
+// Return false to reject; fill *error_code/*error_text.
+typedef std::function<bool(Controller* cntl, int* error_code, std::string* error_text)>
+    InterceptorFn;
+
 struct ServerOptions {
   int idle_timeout_sec = -1;
   int max_concurrency = 0;          // 0 = unlimited
   bool has_builtin_services = true;
   RedisService* redis_service = nullptr;  // serve RESP on the same port
+  InterceptorFn interceptor;              // request admission hook
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)

@@ -57,3 +57,8 @@ def test_counters_move():
     created = f.count_created()
     f.start_join_test(10, 1)
     assert f.count_created() >= created + 10
+
+
+def test_fiber_local_storage():
+    # fiber_key (≙ bthread_key): per-fiber slots + exit-time destructors
+    assert f.key_test()

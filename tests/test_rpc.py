@@ -93,3 +93,16 @@ def test_concurrent_echo_bench_small(server_addr):
 def test_concurrent_echo_16k(server_addr):
     res = r.echo_bench(server_addr, 16384, 8, 500, 10000)
     assert res["errors"] == 0
+
+
+def test_interceptor_admission():
+    """ServerOptions.interceptor rejects requests lacking the credential
+    (parity: reference brpc/interceptor.h)."""
+    port = b.core.combo.start_intercepted_server("777") if hasattr(
+        b.core.combo, "start_intercepted_server") else r.start_intercepted_server("777")
+    addr = f"127.0.0.1:{port}"
+    rc, err = r.call_with_logid(addr, 777)
+    assert rc == 0, err
+    rc, err = r.call_with_logid(addr, 123)
+    assert rc == 1004  # EAUTH
+    assert "credential" in err
