@@ -98,11 +98,10 @@ def test_concurrent_echo_16k(server_addr):
 def test_interceptor_admission():
     """ServerOptions.interceptor rejects requests lacking the credential
     (parity: reference brpc/interceptor.h)."""
-    port = b.core.combo.start_intercepted_server("777") if hasattr(
-        b.core.combo, "start_intercepted_server") else r.start_intercepted_server("777")
+    port = b.core.combo.start_intercepted_server("777")
     addr = f"127.0.0.1:{port}"
-    rc, err = r.call_with_logid(addr, 777)
+    rc, err = b.core.combo.call_with_logid(addr, 777)
     assert rc == 0, err
-    rc, err = r.call_with_logid(addr, 123)
+    rc, err = b.core.combo.call_with_logid(addr, 123)
     assert rc == 1004  # EAUTH
     assert "credential" in err
