@@ -37,6 +37,8 @@ __global__ void copy_spans_kernel(int nspans) {
   }
 }
 
+// NOTE: launches + synchronizes. Batches share the d_spans symbol, so a
+// sync is required between batches; the common case is a single batch.
 int launch_spans(Span* spans, int nspans, int dev) {
   int old_dev = -1;
   hipGetDevice(&old_dev);
@@ -44,6 +46,8 @@ int launch_spans(Span* spans, int nspans, int dev) {
   int rc = 0;
   for (int off = 0; off < nspans; off += kMaxSpansPerLaunch) {
     int batch = nspans - off < kMaxSpansPerLaunch ? nspans - off : kMaxSpansPerLaunch;
+    // hipMemcpyToSymbol is stream-ordered with the prior kernel on the
+    // null stream, so no explicit inter-batch sync is needed.
     hipMemcpyToSymbol(HIP_SYMBOL(d_spans), spans + off, sizeof(Span) * batch);
     hipLaunchKernelGGL(copy_spans_kernel, dim3(batch), dim3(256), 0, 0, batch);
   }
