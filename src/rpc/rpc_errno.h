@@ -27,6 +27,9 @@ enum RpcError {
   ELIMIT = 2004,        // concurrency limit reached
 };
 
+// alias used in errno positions
+const int EOVERCROWDED_ERRNO = EOVERCROWDED;
+
 const char* rpc_strerror(int code);
 
 }  // namespace bam

@@ -6,6 +6,7 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include "base/flags.h"
 #include "base/logging.h"
 #include "base/object_pool.h"
 #include "base/resource_pool.h"
@@ -14,10 +15,14 @@
 #include "fiber/fiber.h"
 #include "fiber/session.h"
 #include "rpc/event_dispatcher.h"
+#include "rpc/rpc_errno.h"
 
 namespace bam {
 
 static Socket::WriteRequest* const kWriteSentinel = (Socket::WriteRequest*)1;
+
+BAM_DEFINE_int64(socket_max_unwritten_bytes, 64 << 20,
+                 "Write() fails with EOVERCROWDED beyond this backlog");
 
 // ---------------- versioned ref pool ----------------
 
@@ -198,6 +203,8 @@ void Socket::remove_pending_session(uint64_t sid) {
 
 static void release_write_request(Socket::WriteRequest* p, int err) {
   if (err != 0) {
+    if (p->socket != nullptr)
+      p->socket->unwritten_bytes.fetch_sub((int64_t)p->data.size(), std::memory_order_relaxed);
     if (p->id_wait != 0) session_error(p->id_wait, err);
     for (uint64_t id : p->merged_ids) session_error(id, err);
   }
@@ -353,6 +360,7 @@ int Socket::DoWrite(WriteRequest* req) {
       continue;  // loop top runs ReleaseAllWriteRequests
     }
     out_bytes.fetch_add(nw, std::memory_order_relaxed);
+    unwritten_bytes.fetch_sub(nw, std::memory_order_relaxed);
     if (!cur->data.empty()) continue;  // partial write; try again
     out_messages.fetch_add(1, std::memory_order_relaxed);
     WriteRequest* nx = cur->next.load(std::memory_order_acquire);
@@ -387,10 +395,17 @@ int Socket::Write(IOBuf* data, const WriteOptions* opt) {
     errno = EPIPE;
     return -1;
   }
+  const int64_t queued = unwritten_bytes.load(std::memory_order_relaxed);
+  if (queued > FLAG_socket_max_unwritten_bytes && !opt->ignore_eovercrowded) {
+    if (opt->id_wait != 0) session_error(opt->id_wait, EOVERCROWDED);
+    errno = EOVERCROWDED_ERRNO;
+    return -1;
+  }
   WriteRequest* req = get_object<WriteRequest>();
   req->data.clear();
   req->merged_ids.clear();
   req->data.swap(*data);
+  unwritten_bytes.fetch_add((int64_t)req->data.size(), std::memory_order_relaxed);
   req->id_wait = opt->id_wait;
   req->socket = this;
   req->next.store(kWriteSentinel, std::memory_order_relaxed);
@@ -411,7 +426,10 @@ int Socket::Write(IOBuf* data, const WriteOptions* opt) {
       errno = err;
       return -1;
     }
-    if (nw > 0) out_bytes.fetch_add(nw, std::memory_order_relaxed);
+    if (nw > 0) {
+      out_bytes.fetch_add(nw, std::memory_order_relaxed);
+      unwritten_bytes.fetch_sub(nw, std::memory_order_relaxed);
+    }
     if (req->data.empty()) {
       out_messages.fetch_add(1, std::memory_order_relaxed);
       WriteRequest* next = PopNextRequest(req);

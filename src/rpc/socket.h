@@ -117,6 +117,10 @@ class Socket {
   // Blocks current fiber until the fd is writable (or failed/timeout).
   int wait_epoll_out(int64_t abstime_us);
 
+  // Unwritten bytes queued on this socket; Write fails with EOVERCROWDED
+  // above -socket_max_unwritten_bytes (parity: reference socket.cpp:1640).
+  std::atomic<int64_t> unwritten_bytes{0};
+
   // Per-connection stats (builtin /connections page).
   std::atomic<int64_t> in_bytes{0};
   std::atomic<int64_t> out_bytes{0};
