@@ -43,3 +43,23 @@ def test_bench_single_process():
     res = json.loads(line)
     assert res["value"] > 0
     assert res["config"]["p99_us"] > 0
+
+
+def test_stream_p2p_two_ranks_gloo():
+    """Config-3 harness on CPU: 2 ranks stream frames over gloo (the same
+    script rides RCCL/xGMI on GPUs)."""
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
+        "tools/stream_xgmi_bench.py", "--frames", "20", "--warmup", "2",
+    ]
+    out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["value"] > 0
