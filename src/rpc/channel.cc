@@ -150,10 +150,21 @@ void IssueRPC(Controller* cntl) {
     cntl->call.server_ep = ep;
   }
   SocketUniquePtr sock;
-  if (GetClientSocket(ep, &sock) != 0) {
-    // Conduct the failure through the session so retry/ending logic runs.
-    session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
-    return;
+  // Fast path: the channel's cached socket (skips the global map mutex).
+  std::atomic<uint64_t>* cache = cntl->call.socket_cache;
+  if (cache != nullptr) {
+    uint64_t sid = cache->load(std::memory_order_acquire);
+    if (sid != 0 && (Socket::Address(sid, &sock) != 0 || sock->Failed())) {
+      sock.reset(nullptr);
+    }
+  }
+  if (!sock) {
+    if (GetClientSocket(ep, &sock) != 0) {
+      // Conduct the failure through the session so retry/ending logic runs.
+      session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
+      return;
+    }
+    if (cache != nullptr) cache->store(sock->id(), std::memory_order_release);
   }
   cntl->remote_side_ = ep;
   SessionId current = session_current_id(cntl->call.cid);
@@ -185,6 +196,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   if (single_server_) {
     cntl->call.server_ep = server_ep_;
     cntl->call.lb = nullptr;
+    cntl->call.socket_cache = &cached_socket_;
   } else {
     cntl->call.lb = lb_.get();
   }

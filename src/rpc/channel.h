@@ -53,7 +53,9 @@ class Channel : public ChannelBase {
   EndPoint server_ep_;
   bool single_server_ = false;
   int protocol_index_ = -1;  // resolved from options_.protocol
+  std::atomic<uint64_t> cached_socket_{0};  // single-server fast path
   std::shared_ptr<LoadBalancerWithNaming> lb_;
+  friend void IssueRPC(Controller*);
 };
 
 // Starts the RPC described by cntl->call (already filled). Used by Channel

@@ -85,6 +85,7 @@ class Controller {
     class SubChannelCtx* sub_ctx = nullptr;  // ParallelChannel bookkeeping
     uint64_t stream_id = 0;                  // client-created stream (StreamCreate)
     int protocol_index = -1;                 // wire protocol for this call
+    std::atomic<uint64_t>* socket_cache = nullptr;  // channel's cached socket id
   };
   Call call;
 

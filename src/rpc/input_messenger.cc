@@ -2,10 +2,21 @@
 
 #include <errno.h>
 
+#include "base/flags.h"
 #include "base/logging.h"
 #include "fiber/fiber.h"
 
 namespace bam {
+
+// Processing responses inline in the parse fiber skips one fiber spawn per
+// call (response processing is framework code: it completes a session and
+// wakes the joiner). Requests keep fiber-per-message isolation by default
+// (user handlers may block) — parity with the reference's ProcessInplace
+// option.
+BAM_DEFINE_bool(process_response_in_place, true,
+                "run client response processing inline in the parse fiber");
+BAM_DEFINE_bool(process_request_in_place, false,
+                "run server request handlers inline in the parse fiber");
 
 namespace {
 
@@ -30,6 +41,13 @@ void ProcessMessageFiber(void* raw) {
 void InputMessenger::DispatchMessage(InputMessageBase* msg, int protocol_index) {
   const Protocol* proto = GetProtocol(protocol_index);
   msg->protocol_index = protocol_index;
+  const bool in_place =
+      server_side_ ? FLAG_process_request_in_place : FLAG_process_response_in_place;
+  if (in_place) {
+    if (server_side_) proto->process_request(msg);
+    else proto->process_response(msg);
+    return;
+  }
   ProcessArg* a = new ProcessArg{msg, proto, server_side_};
   fiber_t th;
   // Urgent start: the parsing fiber is requeued, the message handler runs
