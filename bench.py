@@ -64,7 +64,8 @@ def main():
 
     def run_step():
         res = b.core.rpc.echo_bench(addr, args.payload, args.concurrency,
-                                    args.calls_per_step, 30000, method, hbm_req)
+                                    args.calls_per_step, 30000, method, hbm_req,
+                                    True)  # pooled connections (reference headline mode)
         if res["errors"]:
             raise RuntimeError("bench errors: n=%s first=%s" % (res["errors"], res.get("first_error")))
         return res
@@ -126,6 +127,7 @@ def main():
                 "concurrency_per_gpu": args.concurrency,
                 "calls_per_step": args.calls_per_step,
                 "p99_us": p99_us,
+                "connection_type": "pooled",
                 "parallelism": "dp%d" % (world_size if world_size > 1 else 1),
                 "gpu_payload_residency": "HBM" if use_gpu else "host",
             },

@@ -22,7 +22,7 @@ int call_method_once(const std::string& addr, const std::string& method,
 bool attachment_test(const std::string& addr);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
-                       bool hbm_request);
+                       bool hbm_request, bool pooled);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -66,12 +66,12 @@ void bind_rpc(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   r.def("echo_bench",
         [](const std::string& addr, int payload_size, int concurrency, int64_t total,
-           int timeout_ms, const std::string& method, bool hbm_request) {
+           int timeout_ms, const std::string& method, bool hbm_request, bool pooled) {
           bam::rpctest::BenchResult b;
           {
             py::gil_scoped_release rel;
             b = bam::rpctest::echo_bench(addr, payload_size, concurrency, total, timeout_ms,
-                                         method, hbm_request);
+                                         method, hbm_request, pooled);
           }
           py::dict d;
           d["qps"] = b.qps;
@@ -89,7 +89,8 @@ void bind_rpc(py::module_& m) {
         },
         py::arg("addr"), py::arg("payload_size") = 64, py::arg("concurrency") = 8,
         py::arg("total") = 10000, py::arg("timeout_ms") = 5000,
-        py::arg("method") = "EchoService.Echo", py::arg("hbm_request") = false);
+        py::arg("method") = "EchoService.Echo", py::arg("hbm_request") = false,
+        py::arg("pooled") = false);
 }
 
 // ---- combo channels & LBs ----

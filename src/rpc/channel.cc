@@ -159,7 +159,7 @@ void IssueRPC(Controller* cntl) {
     }
   }
   if (!sock) {
-    if (GetClientSocket(ep, &sock) != 0) {
+    if (GetClientSocket(ep, &sock, cntl->call.connection_shard) != 0) {
       // Conduct the failure through the session so retry/ending logic runs.
       session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
       return;
@@ -193,10 +193,17 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->call.response = response;
   cntl->call.done = done;
   cntl->call.protocol_index = protocol_index_;
+  if (options_.connection_type == "pooled") {
+    static std::atomic<uint32_t> rr{0};
+    cntl->call.connection_shard =
+        1 + (int)(rr.fetch_add(1, std::memory_order_relaxed) %
+                  (uint32_t)(options_.connection_pool_size > 0 ? options_.connection_pool_size
+                                                               : 8));
+  }
   if (single_server_) {
     cntl->call.server_ep = server_ep_;
     cntl->call.lb = nullptr;
-    cntl->call.socket_cache = &cached_socket_;
+    if (cntl->call.connection_shard == 0) cntl->call.socket_cache = &cached_socket_;
   } else {
     cntl->call.lb = lb_.get();
   }

@@ -20,6 +20,10 @@ struct ChannelOptions {
   int32_t backup_request_ms = -1;
   int max_retry = 3;
   std::string protocol = "std";
+  // "single" (default) or "pooled" (N connections round-robined; the
+  // reference's multi-connection mode — its highest-throughput config).
+  std::string connection_type = "single";
+  int connection_pool_size = 8;
   std::string connection_group;
 };
 

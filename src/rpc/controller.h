@@ -86,6 +86,7 @@ class Controller {
     uint64_t stream_id = 0;                  // client-created stream (StreamCreate)
     int protocol_index = -1;                 // wire protocol for this call
     std::atomic<uint64_t>* socket_cache = nullptr;  // channel's cached socket id
+    int connection_shard = 0;                        // pooled connection index
   };
   Call call;
 

@@ -184,11 +184,12 @@ void bench_worker(void* raw) {
 
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
-                       bool hbm_request) {
+                       bool hbm_request, bool pooled) {
   BenchResult res;
   Channel channel;
   ChannelOptions opts;
   opts.timeout_ms = timeout_ms;
+  if (pooled) opts.connection_type = "pooled";
   if (channel.Init(addr.c_str(), &opts) != 0) return res;
   std::string payload(payload_size, 'x');
   for (int i = 0; i < payload_size; ++i) payload[i] = (char)fast_rand();

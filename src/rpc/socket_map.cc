@@ -29,7 +29,7 @@ InputMessenger* client_messenger() {
 
 struct ClientSocketMap {
   std::mutex mu;
-  std::map<EndPoint, SocketId> sockets;
+  std::map<std::pair<EndPoint, int>, SocketId> sockets;
 };
 
 ClientSocketMap& the_map() {
@@ -123,15 +123,16 @@ bool IsEndpointIsolated(const EndPoint& ep) {
   return h != nullptr && h->isolated.load(std::memory_order_acquire);
 }
 
-int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out) {
+int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard) {
   if (IsEndpointIsolated(ep)) {
     errno = EHOSTDOWN;
     return -1;
   }
   ClientSocketMap& m = the_map();
+  const auto key = std::make_pair(ep, shard);
   {
     std::lock_guard<std::mutex> lk(m.mu);
-    auto it = m.sockets.find(ep);
+    auto it = m.sockets.find(key);
     if (it != m.sockets.end()) {
       if (Socket::Address(it->second, out) == 0 && !(*out)->Failed()) return 0;
       m.sockets.erase(it);
@@ -149,7 +150,7 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out) {
   if (Socket::Address(sid, out) != 0) return -1;
   {
     std::lock_guard<std::mutex> lk(m.mu);
-    auto it = m.sockets.find(ep);
+    auto it = m.sockets.find(key);
     if (it != m.sockets.end()) {
       // Raced with another creator: prefer the existing healthy one.
       SocketUniquePtr existing;
@@ -161,7 +162,7 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out) {
       it->second = sid;
       return 0;
     }
-    m.sockets[ep] = sid;
+    m.sockets[key] = sid;
   }
   return 0;
 }
@@ -169,8 +170,12 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out) {
 void RemoveClientSocket(const EndPoint& ep, SocketId expected) {
   ClientSocketMap& m = the_map();
   std::lock_guard<std::mutex> lk(m.mu);
-  auto it = m.sockets.find(ep);
-  if (it != m.sockets.end() && it->second == expected) m.sockets.erase(it);
+  for (auto it = m.sockets.begin(); it != m.sockets.end(); ++it) {
+    if (it->first.first == ep && it->second == expected) {
+      m.sockets.erase(it);
+      return;
+    }
+  }
 }
 
 }  // namespace bam

@@ -8,8 +8,10 @@
 namespace bam {
 
 // Returns a referenced, healthy client socket to `ep`, creating/connecting
-// if needed. 0 on success.
-int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out);
+// if needed. 0 on success. `shard` > 0 selects a distinct pooled
+// connection (parity: reference pooled connection_type — multiple
+// connections to one server spread parse/write parallelism).
+int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard = 0);
 
 // Drops the cached socket for ep (e.g. after failure).
 void RemoveClientSocket(const EndPoint& ep, SocketId expected);
