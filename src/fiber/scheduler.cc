@@ -233,7 +233,9 @@ TaskControl::TaskControl() {
   }
   if (n <= 0) n = (int)std::thread::hardware_concurrency();
   if (n <= 0) n = 4;
-  if (n > 64) n = 64;
+  // More workers than ~16 hurts: the steal scan is O(workers) and the
+  // spin-before-park burn multiplies (reference default is 8+1 workers).
+  if (n > 16) n = 16;
   start_workers(n);
 }
 
