@@ -54,7 +54,10 @@ def main():
     ndev = b.core.gpu.initialize()
     use_gpu = ndev > 0
     if use_gpu:
-        method, hbm_req = "EchoService.EchoHbm", True
+        # IOBuf-in-HBM: the server stores every response payload in
+        # HBM-resident IOBuf blocks; the socket write path stages them out
+        # through the device-gather + pinned-ring leg (hip/).
+        method, hbm_req = "EchoService.EchoHbm", False
     else:
         # CPU-only container: same benchmark on the host path.
         method, hbm_req = "EchoService.Echo", False
@@ -133,7 +136,7 @@ def main():
                 "p99_us": p99_us,
                 "connection_type": "pooled" if pooled else "single",
                 "parallelism": "dp%d" % (world_size if world_size > 1 else 1),
-                "gpu_payload_residency": "HBM" if use_gpu else "host",
+                "gpu_payload_residency": "HBM (server response payloads)" if use_gpu else "host",
             },
         }
         print(json.dumps(out), flush=True)
