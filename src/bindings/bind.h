@@ -13,3 +13,4 @@ void bind_snappy(py::module_& m);
 void bind_api(py::module_& m);
 void bind_redis(py::module_& m);
 void bind_util(py::module_& m);
+void bind_memcache(py::module_& m);

@@ -281,3 +281,39 @@ void bind_redis(py::module_& m) {
     return reply_to_py(reply);
   }, py::arg("addr"), py::arg("args"), py::arg("timeout_ms") = 1000);
 }
+
+// ---- memcache client ----
+#include "rpc/memcache.h"
+
+void bind_memcache(py::module_& m) {
+  py::class_<bam::MemcacheClient>(m, "MemcacheClient")
+      .def(py::init<const std::string&, int>(), py::arg("addr"), py::arg("timeout_ms") = 1000)
+      .def("ok", &bam::MemcacheClient::ok)
+      .def("set", [](bam::MemcacheClient& c, const std::string& key, const std::string& value,
+                     uint32_t flags, uint32_t exptime) {
+        return c.Set(key, value, flags, exptime);
+      }, py::arg("key"), py::arg("value"), py::arg("flags") = 0, py::arg("exptime") = 0,
+         py::call_guard<py::gil_scoped_release>())
+      .def("get", [](bam::MemcacheClient& c, const std::string& key) -> py::object {
+        std::string value;
+        int rc;
+        {
+          py::gil_scoped_release rel;
+          rc = c.Get(key, &value);
+        }
+        if (rc == 10001) return py::none();  // key not found
+        if (rc != 0) throw std::runtime_error("memcache get rc=" + std::to_string(rc));
+        return py::bytes(value);
+      })
+      .def("delete", &bam::MemcacheClient::Delete, py::call_guard<py::gil_scoped_release>())
+      .def("version", [](bam::MemcacheClient& c) {
+        std::string v;
+        int rc;
+        {
+          py::gil_scoped_release rel;
+          rc = c.Version(&v);
+        }
+        if (rc != 0) throw std::runtime_error("memcache version rc=" + std::to_string(rc));
+        return v;
+      });
+}

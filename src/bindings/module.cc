@@ -13,4 +13,5 @@ PYBIND11_MODULE(_core, m) {
   bind_api(m);
   bind_redis(m);
   bind_util(m);
+  bind_memcache(m);
 }
