@@ -14,3 +14,4 @@ void bind_api(py::module_& m);
 void bind_redis(py::module_& m);
 void bind_util(py::module_& m);
 void bind_memcache(py::module_& m);
+void bind_json2pb(py::module_& m);

@@ -317,3 +317,66 @@ void bind_memcache(py::module_& m) {
         return v;
       });
 }
+
+// ---- json2pb ----
+#include "rpc/json2pb.h"
+
+namespace {
+
+bam::json2pb::Schema schema_from_py(py::dict d);
+
+bam::json2pb::FieldDesc field_from_py(py::handle spec) {
+  using bam::json2pb::FieldDesc;
+  py::tuple t = spec.cast<py::tuple>();
+  FieldDesc f;
+  f.number = t[0].cast<int>();
+  std::string ty = t[1].cast<std::string>();
+  if (ty.rfind("repeated ", 0) == 0) {
+    f.repeated = true;
+    ty = ty.substr(9);
+  }
+  if (ty == "int64") f.type = FieldDesc::INT64;
+  else if (ty == "int32") f.type = FieldDesc::INT32;
+  else if (ty == "uint64") f.type = FieldDesc::UINT64;
+  else if (ty == "uint32") f.type = FieldDesc::UINT32;
+  else if (ty == "bool") f.type = FieldDesc::BOOL;
+  else if (ty == "double") f.type = FieldDesc::DOUBLE;
+  else if (ty == "float") f.type = FieldDesc::FLOAT;
+  else if (ty == "string") f.type = FieldDesc::STRING;
+  else if (ty == "bytes") f.type = FieldDesc::BYTES;
+  else if (ty == "message") {
+    f.type = FieldDesc::MESSAGE;
+    f.message_fields = std::make_shared<bam::json2pb::Schema>(schema_from_py(t[2].cast<py::dict>()));
+  } else {
+    throw std::runtime_error("unknown field type: " + ty);
+  }
+  return f;
+}
+
+bam::json2pb::Schema schema_from_py(py::dict d) {
+  bam::json2pb::Schema s;
+  for (auto item : d) {
+    s[item.first.cast<std::string>()] = field_from_py(item.second);
+  }
+  return s;
+}
+
+}  // namespace
+
+void bind_json2pb(py::module_& m) {
+  auto j = m.def_submodule("json2pb");
+  j.def("json_to_pb", [](py::dict schema, const std::string& json_text) {
+    auto s = schema_from_py(schema);
+    std::string wire, err;
+    if (!bam::json2pb::JsonToPb(s, json_text, &wire, &err))
+      throw std::runtime_error("json_to_pb: " + err);
+    return py::bytes(wire);
+  });
+  j.def("pb_to_json", [](py::dict schema, py::bytes wire) {
+    auto s = schema_from_py(schema);
+    std::string text, err;
+    if (!bam::json2pb::PbToJson(s, wire.cast<std::string>(), &text, &err))
+      throw std::runtime_error("pb_to_json: " + err);
+    return text;
+  });
+}

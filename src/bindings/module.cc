@@ -14,4 +14,5 @@ PYBIND11_MODULE(_core, m) {
   bind_redis(m);
   bind_util(m);
   bind_memcache(m);
+  bind_json2pb(m);
 }

@@ -84,6 +84,17 @@ class Reader {
     return s;
   }
 
+  // Reads n raw bytes (fixed32/fixed64 payloads). Empty string on underrun.
+  std::string read_fixed(size_t n) {
+    if ((size_t)(end_ - p_) < n) {
+      ok_ = false;
+      return std::string();
+    }
+    std::string s(p_, n);
+    p_ += n;
+    return s;
+  }
+
   void skip(int wtype) {
     switch (wtype) {
       case 0:
