@@ -6,6 +6,7 @@
 #include "base/logging.h"
 #include "rpc/event_dispatcher.h"
 #include "rpc/policy/std_protocol.h"
+#include "rpc/policy/http_protocol.h"
 #include "rpc/redis.h"
 
 namespace bam {
@@ -74,6 +75,7 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   if (IsRunning()) return -1;
   if (opt != nullptr) options_ = *opt;
   policy::RegisterStdProtocol();
+  policy::RegisterH2Protocol();
   if (options_.redis_service != nullptr) policy::RegisterRedisProtocol();
   int listen_fd = tcp_listen(ep);
   if (listen_fd < 0) {

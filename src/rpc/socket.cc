@@ -81,6 +81,8 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   s->remote_side_ = options.remote_side;
   s->read_buf_.clear();
   s->preferred_protocol_index = -1;
+  s->protocol_ctx = nullptr;
+  s->protocol_ctx_deleter = nullptr;
   s->input_events_.store(0, std::memory_order_relaxed);
   s->write_head_.store(nullptr, std::memory_order_relaxed);
   s->in_bytes = 0;
@@ -170,6 +172,11 @@ int Socket::SetFailed(int error_code, const char* error_text) {
 }
 
 void Socket::Recycle() {
+  if (protocol_ctx != nullptr && protocol_ctx_deleter != nullptr) {
+    protocol_ctx_deleter(protocol_ctx);
+  }
+  protocol_ctx = nullptr;
+  protocol_ctx_deleter = nullptr;
   int fd = fd_.load(std::memory_order_acquire);
   if (fd >= 0) {
     ::close(fd);

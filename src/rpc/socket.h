@@ -102,6 +102,10 @@ class Socket {
   // Read buffer for InputMessenger.
   IOBuf& read_buf() { return read_buf_; }
   int preferred_protocol_index = -1;
+  // Per-connection protocol state (e.g. the h2 session); freed by the
+  // deleter when the socket is recycled.
+  void* protocol_ctx = nullptr;
+  void (*protocol_ctx_deleter)(void*) = nullptr;
 
   // Correlation sessions waiting for responses on this socket; failed
   // when the connection breaks. (Parity: reference conducts errors to
