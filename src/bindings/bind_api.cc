@@ -14,6 +14,7 @@
 #include "rpc/parallel_channel.h"
 #include "rpc/server.h"
 #include "fiber/fiber.h"
+#include "rpc/policy/http_protocol.h"
 #include "rpc/usercode_pool.h"
 
 namespace {
@@ -379,4 +380,27 @@ void bind_json2pb(py::module_& m) {
       throw std::runtime_error("pb_to_json: " + err);
     return text;
   });
+}
+
+// ---- thrift passthrough client ----
+void bind_thrift(py::module_& m) {
+  m.def("thrift_call", [](const std::string& addr, const std::string& method,
+                          const std::string& payload, int timeout_ms) {
+    bam::ChannelOptions opts;
+    opts.timeout_ms = timeout_ms;
+    opts.protocol = "thrift";
+    bam::policy::RegisterThriftProtocol();
+    bam::Channel ch;
+    if (ch.Init(addr.c_str(), &opts) != 0) throw std::runtime_error("thrift channel init");
+    bam::Controller cntl;
+    bam::IOBuf req, resp;
+    req.append(payload);
+    cntl.call.method_name = method;
+    {
+      py::gil_scoped_release rel;
+      ch.CallMethod("thrift." + method, &cntl, &req, &resp, nullptr);
+    }
+    if (cntl.Failed()) throw PyRpcError(cntl.ErrorCode(), cntl.ErrorText());
+    return py::bytes(resp.to_string());
+  }, py::arg("addr"), py::arg("method"), py::arg("payload"), py::arg("timeout_ms") = 1000);
 }

@@ -15,4 +15,5 @@ PYBIND11_MODULE(_core, m) {
   bind_util(m);
   bind_memcache(m);
   bind_json2pb(m);
+  bind_thrift(m);
 }

@@ -5,6 +5,7 @@
 #include "base/logging.h"
 #include "base/time.h"
 #include "rpc/load_balancer.h"
+#include "rpc/policy/http_protocol.h"
 #include "rpc/policy/std_protocol.h"
 #include "rpc/redis.h"
 #include "rpc/rpcz.h"
@@ -20,6 +21,7 @@ int Channel::Init(EndPoint ep, const ChannelOptions* options) {
   single_server_ = true;
   policy::RegisterStdProtocol();
   if (options_.protocol == "redis") policy::RegisterRedisProtocol();
+  if (options_.protocol == "thrift") policy::RegisterThriftProtocol();
   protocol_index_ = FindProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
   return 0;

@@ -76,6 +76,7 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   if (opt != nullptr) options_ = *opt;
   policy::RegisterStdProtocol();
   policy::RegisterH2Protocol();
+  policy::RegisterThriftProtocol();
   if (options_.redis_service != nullptr) policy::RegisterRedisProtocol();
   int listen_fd = tcp_listen(ep);
   if (listen_fd < 0) {
