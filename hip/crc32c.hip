@@ -1,16 +1,14 @@
-// brpc_amd: CRC32-C on gfx950.
+// brpc_amd: CRC32-C on gfx950 — single-kernel design.
 //
-// Two-kernel design (MI355X-first; see /opt/skills/guides):
-//  kernel 1 (crc_chunks_kernel): the message is split into 2 KiB chunks;
-//    one wave (64 lanes) owns 64 consecutive chunks. HBM reads are fully
-//    coalesced row-wise into an LDS tile (64x64 words, padded), then each
-//    lane walks ITS chunk's words serially out of LDS (CRC is a serial
-//    recurrence per chunk) with slice-by-8 tables staged in LDS.
-//    2 KiB chunks -> 2048 workgroups at 256 MiB: fills 256 CUs (≫256 rule).
-//  kernel 2 (crc_combine_kernel): each lane shifts its chunk's CRC by the
-//    byte-distance to the end of the message using precomputed GF(2)
-//    matrices for x^(8*chunk*2^k) (+ one per-call tail operator), then
-//    XOR-reduces with device-scope atomics. Only 4 bytes return to host.
+// The message is split into 2 KiB chunks; one wave (64 lanes) owns 64
+// consecutive chunks. HBM reads are fully coalesced row-wise into an LDS
+// tile (64x64 words, padded), then each lane walks ITS chunk's words
+// serially out of LDS (CRC is a serial recurrence per chunk) with
+// slice-by-8 tables staged in LDS. Each lane then shifts its chunk CRC by
+// the byte-distance to the end of the message using precomputed GF(2)
+// byte-power operator matrices x^(8·2^k) (device constant memory) and
+// XOR-reduces with a device-scope atomic — ONE kernel, 4 bytes D2H.
+// 2 KiB chunks -> 2048 workgroups at 256 MiB (≫256 CUs, guide §1).
 //
 // Host reference / test oracle: src/base/crc32c.cc.
 #include <hip/hip_runtime.h>
@@ -26,10 +24,10 @@ namespace {
 constexpr uint32_t kPoly = 0x82F63B78u;
 constexpr int kChunkWords = 256;  // 2 KiB per chunk
 constexpr size_t kChunkBytes = (size_t)kChunkWords * 8;
-constexpr int kMaxPow = 28;  // supports up to 2^28 full chunks (512 TB)
+constexpr int kMaxBytePow = 40;  // shifts up to 2^40 bytes (1 TB)
 
 __device__ uint32_t d_tab[8][256];
-__device__ uint32_t d_pow[kMaxPow][32];  // shift-by-(chunk*2^k) operators
+__device__ uint32_t d_bpow[kMaxBytePow][32];  // shift-by-2^k-BYTES operators
 
 // ---- host-side table + GF(2) helpers ----
 
@@ -65,7 +63,6 @@ inline void gf2_square(uint32_t* sq, const uint32_t* mat) {
 }
 
 inline void gf2_matmul(uint32_t* out, const uint32_t* a, const uint32_t* b) {
-  // out = a∘b (apply b then a); GF(2) poly-mod matrices commute anyway.
   for (int n = 0; n < 32; ++n) out[n] = gf2_times(a, b[n]);
 }
 
@@ -105,18 +102,27 @@ void ensure_init() {
   std::call_once(g_init_flag, [] {
     build_tables();
     hipMemcpyToSymbol(HIP_SYMBOL(d_tab), h_tab, sizeof(h_tab));
-    // power operators: pow[k] = shift by chunk*2^k bytes
-    static uint32_t pows[kMaxPow][32];
-    build_shift_operator(pows[0], kChunkBytes);
-    for (int k = 1; k < kMaxPow; ++k) gf2_matmul(pows[k], pows[k - 1], pows[k - 1]);
-    hipMemcpyToSymbol(HIP_SYMBOL(d_pow), pows, sizeof(pows));
+    static uint32_t bpows[kMaxBytePow][32];
+    build_shift_operator(bpows[0], 1);  // one byte
+    for (int k = 1; k < kMaxBytePow; ++k) gf2_matmul(bpows[k], bpows[k - 1], bpows[k - 1]);
+    hipMemcpyToSymbol(HIP_SYMBOL(d_bpow), bpows, sizeof(bpows));
   });
 }
 
-// ---- kernels ----
+// ---- kernel ----
+
+__device__ __forceinline__ uint32_t dev_gf2_times(const uint32_t* mat, uint32_t vec) {
+  uint32_t sum = 0;
+  while (vec) {
+    if (vec & 1) sum ^= *mat;
+    vec >>= 1;
+    ++mat;
+  }
+  return sum;
+}
 
 __global__ __launch_bounds__(64) void crc_chunks_kernel(const uint8_t* data, size_t n,
-                                                        uint32_t* out, int nchunks) {
+                                                        uint32_t* result, int nchunks) {
   __shared__ uint32_t tab[8][256];   // 8 KiB
   __shared__ uint64_t tile[64][65];  // 32.5 KiB, padded against bank conflicts
   const int lane = threadIdx.x;
@@ -152,46 +158,21 @@ __global__ __launch_bounds__(64) void crc_chunks_kernel(const uint8_t* data, siz
       crc = tab[0][(crc ^ data[i]) & 0xff] ^ (crc >> 8);
     }
   }
-  if (my_chunk < nchunks) out[my_chunk] = ~crc;
-}
-
-__device__ __forceinline__ uint32_t dev_gf2_times(const uint32_t* mat, uint32_t vec) {
-  uint32_t sum = 0;
-  while (vec) {
-    if (vec & 1) sum ^= *mat;
-    vec >>= 1;
-    ++mat;
-  }
-  return sum;
-}
-
-// tail_op: operator for the LAST chunk's length (identity if it is full).
-// Every chunk i < nchunks-1 is shifted by (nchunks-2-i) full chunks + tail;
-// the last chunk is unshifted. XOR-reduce into *result (pre-zeroed).
-__global__ void crc_combine_kernel(const uint32_t* chunk_crcs, int nchunks,
-                                   const uint32_t* tail_op, uint32_t* result) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= nchunks) return;
-  uint32_t crc = chunk_crcs[i];
-  if (i < nchunks - 1) {
-    uint32_t m = (uint32_t)(nchunks - 2 - i);  // full-chunk shifts
-    for (int k = 0; k < kMaxPow && m != 0; ++k, m >>= 1) {
-      if (m & 1) crc = dev_gf2_times(d_pow[k], crc);
-    }
-    crc = dev_gf2_times(tail_op, crc);
+  if (my_chunk >= nchunks) return;
+  crc = ~crc;
+  // Shift by the bytes after this chunk and fold into the result.
+  uint64_t dist = n - my_end;
+  for (int k = 0; k < kMaxBytePow && dist != 0; ++k, dist >>= 1) {
+    if (dist & 1) crc = dev_gf2_times(d_bpow[k], crc);
   }
   atomicXor(result, crc);
 }
 
 // persistent per-device scratch
-constexpr int kMaxDevCrc = 16;
 struct CrcScratch {
-  uint32_t* chunk_crcs = nullptr;
-  size_t cap = 0;
   uint32_t* result = nullptr;
-  uint32_t* tail_op = nullptr;
 };
-CrcScratch g_scratch[kMaxDevCrc];
+CrcScratch g_scratch[16];
 std::mutex g_scratch_mu;
 
 }  // namespace
@@ -208,31 +189,12 @@ extern "C" uint32_t bam_gpu_crc32c(const void* dev_ptr, size_t n, uint32_t init,
   CrcScratch* sc;
   {
     std::lock_guard<std::mutex> lk(g_scratch_mu);
-    sc = &g_scratch[dev < kMaxDevCrc ? dev : 0];
-    if (sc->cap < (size_t)nchunks) {
-      if (sc->chunk_crcs != nullptr) hipFree(sc->chunk_crcs);
-      sc->cap = (size_t)nchunks * 2;
-      hipMalloc(&sc->chunk_crcs, sc->cap * sizeof(uint32_t));
-    }
-    if (sc->result == nullptr) {
-      hipMalloc(&sc->result, sizeof(uint32_t));
-      hipMalloc(&sc->tail_op, 32 * sizeof(uint32_t));
-    }
+    sc = &g_scratch[dev < 16 ? dev : 0];
+    if (sc->result == nullptr) hipMalloc(&sc->result, sizeof(uint32_t));
   }
-
+  hipMemsetAsync(sc->result, 0, sizeof(uint32_t), 0);
   hipLaunchKernelGGL(crc_chunks_kernel, dim3(nblocks), dim3(64), 0, 0,
-                     (const uint8_t*)dev_ptr, n, sc->chunk_crcs, nchunks);
-
-  // tail operator (identity when the last chunk is full)
-  size_t tail_len = n - (size_t)(nchunks - 1) * kChunkBytes;
-  uint32_t tail_op[32];
-  build_shift_operator(tail_op, tail_len);
-  hipMemcpy(sc->tail_op, tail_op, sizeof(tail_op), hipMemcpyHostToDevice);
-  uint32_t zero = 0;
-  hipMemcpy(sc->result, &zero, sizeof(zero), hipMemcpyHostToDevice);
-  int cblocks = (nchunks + 255) / 256;
-  hipLaunchKernelGGL(crc_combine_kernel, dim3(cblocks), dim3(256), 0, 0, sc->chunk_crcs,
-                     nchunks, sc->tail_op, sc->result);
+                     (const uint8_t*)dev_ptr, n, sc->result, nchunks);
   uint32_t crc = 0;
   hipMemcpy(&crc, sc->result, sizeof(crc), hipMemcpyDeviceToHost);
 

@@ -221,12 +221,61 @@ __global__ __launch_bounds__(64) void snappy_decompress_kernel(const uint8_t* co
   if (lane == 0) *result = (ok && op == expected) ? (uint64_t)op : ~0ULL;
 }
 
+// ---- device-side compaction (prefix sums + scatter; zero host loops) ----
+
+__global__ void snappy_block_sums(const uint32_t* sizes, int nchunks, uint64_t* bsums) {
+  __shared__ uint64_t sdata[256];
+  int t = threadIdx.x;
+  int i = blockIdx.x * 256 + t;
+  sdata[t] = i < nchunks ? sizes[i] : 0;
+  __syncthreads();
+  for (int s = 128; s > 0; s >>= 1) {
+    if (t < s) sdata[t] += sdata[t + s];
+    __syncthreads();
+  }
+  if (t == 0) bsums[blockIdx.x] = sdata[0];
+}
+
+__global__ void snappy_scan_bsums(uint64_t* bsums, int nb, uint64_t* total) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    uint64_t acc = 0;
+    for (int i = 0; i < nb; ++i) {
+      uint64_t v = bsums[i];
+      bsums[i] = acc;
+      acc += v;
+    }
+    *total = acc;
+  }
+}
+
+__global__ void snappy_compact(const uint8_t* scratch, const uint32_t* sizes,
+                               const uint64_t* bsums, uint8_t* dst, int nchunks,
+                               int max_per_chunk) {
+  __shared__ uint64_t offset_sh;
+  int c = blockIdx.x;
+  if (c >= nchunks) return;
+  int t = threadIdx.x;
+  if (t == 0) {
+    uint64_t off = bsums[c / 256];
+    int start = (c / 256) * 256;
+    for (int i = start; i < c; ++i) off += sizes[i];
+    offset_sh = off;
+  }
+  __syncthreads();
+  uint64_t off = offset_sh;
+  uint32_t len = sizes[c];
+  const uint8_t* src = scratch + (size_t)c * max_per_chunk;
+  for (uint32_t i = t; i < len; i += blockDim.x) dst[off + i] = src[i];
+}
+
 std::mutex g_snappy_mu;
 struct SnappyScratch {
   uint8_t* per_chunk = nullptr;
   uint32_t* sizes = nullptr;
+  uint64_t* bsums = nullptr;
   size_t chunk_cap = 0;
   uint64_t* result = nullptr;
+  uint64_t* total = nullptr;
 };
 SnappyScratch g_snappy[16];
 
@@ -257,40 +306,34 @@ extern "C" int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_
     if (sc->chunk_cap < (size_t)nchunks) {
       if (sc->per_chunk) hipFree(sc->per_chunk);
       if (sc->sizes) hipFree(sc->sizes);
+      if (sc->bsums) hipFree(sc->bsums);
       sc->chunk_cap = (size_t)nchunks * 2;
       hipMalloc(&sc->per_chunk, sc->chunk_cap * kMaxPerChunk);
       hipMalloc(&sc->sizes, sc->chunk_cap * sizeof(uint32_t));
+      hipMalloc(&sc->bsums, ((sc->chunk_cap + 255) / 256 + 1) * sizeof(uint64_t));
     }
+    if (sc->total == nullptr) hipMalloc(&sc->total, sizeof(uint64_t));
   }
   hipLaunchKernelGGL(snappy_compress_kernel, dim3(nblocks), dim3(64), 0, 0,
                      (const uint8_t*)src_dev, n, sc->per_chunk, sc->sizes, nchunks);
-  uint32_t* h_sizes = (uint32_t*)malloc(sizeof(uint32_t) * nchunks);
-  hipMemcpy(h_sizes, sc->sizes, sizeof(uint32_t) * nchunks, hipMemcpyDeviceToHost);
-  // preamble
+  // Device-side compaction: block sums -> exclusive scan -> scatter.
+  const int nb = (nchunks + 255) / 256;
+  hipLaunchKernelGGL(snappy_block_sums, dim3(nb), dim3(256), 0, 0, sc->sizes, nchunks,
+                     sc->bsums);
+  hipLaunchKernelGGL(snappy_scan_bsums, dim3(1), dim3(64), 0, 0, sc->bsums, nb, sc->total);
+  uint64_t payload_total = 0;
+  hipMemcpy(&payload_total, sc->total, sizeof(payload_total), hipMemcpyDeviceToHost);
   uint8_t pre[8];
   int pre_n;
   emit_varint_host(pre, n, &pre_n);
-  size_t total = (size_t)pre_n;
-  for (int i = 0; i < nchunks; ++i) total += h_sizes[i];
-  if (total > dst_cap) {
-    free(h_sizes);
-    return -1;
-  }
+  if (payload_total + pre_n > dst_cap) return -1;
   hipMemcpy(dst_dev, pre, pre_n, hipMemcpyHostToDevice);
-  // compact: span copy per chunk
-  const void** srcs = (const void**)malloc(sizeof(void*) * nchunks);
-  size_t* lens = (size_t*)malloc(sizeof(size_t) * nchunks);
-  for (int i = 0; i < nchunks; ++i) {
-    srcs[i] = sc->per_chunk + (size_t)i * kMaxPerChunk;
-    lens[i] = h_sizes[i];
-  }
-  int rc = bam_gpu_gather((uint8_t*)dst_dev + pre_n, srcs, lens, nchunks, dev);
-  free(srcs);
-  free(lens);
-  free(h_sizes);
-  *out_len = total;
+  hipLaunchKernelGGL(snappy_compact, dim3(nchunks), dim3(256), 0, 0, sc->per_chunk,
+                     sc->sizes, sc->bsums, (uint8_t*)dst_dev + pre_n, nchunks, kMaxPerChunk);
+  hipError_t e = hipDeviceSynchronize();
+  *out_len = (size_t)payload_total + pre_n;
   if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
-  return rc;
+  return e == hipSuccess ? 0 : -1;
 }
 
 extern "C" int bam_gpu_snappy_decompress(const void* src_dev, size_t n, void* dst_dev,
