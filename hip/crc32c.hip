@@ -24,10 +24,11 @@ namespace {
 constexpr uint32_t kPoly = 0x82F63B78u;
 constexpr int kChunkWords = 256;  // 2 KiB per chunk
 constexpr size_t kChunkBytes = (size_t)kChunkWords * 8;
-constexpr int kMaxBytePow = 40;  // shifts up to 2^40 bytes (1 TB)
+constexpr int kPowLevels = 5;  // base-256 digits: shifts up to 256^5 = 1 TB
 
 __device__ uint32_t d_tab[8][256];
-__device__ uint32_t d_bpow[kMaxBytePow][32];  // shift-by-2^k-BYTES operators
+// d_pow256[k][b] = operator for shifting by b * 256^k bytes.
+__device__ uint32_t d_pow256[kPowLevels][256][32];
 
 // ---- host-side table + GF(2) helpers ----
 
@@ -102,10 +103,18 @@ void ensure_init() {
   std::call_once(g_init_flag, [] {
     build_tables();
     hipMemcpyToSymbol(HIP_SYMBOL(d_tab), h_tab, sizeof(h_tab));
-    static uint32_t bpows[kMaxBytePow][32];
-    build_shift_operator(bpows[0], 1);  // one byte
-    for (int k = 1; k < kMaxBytePow; ++k) gf2_matmul(bpows[k], bpows[k - 1], bpows[k - 1]);
-    hipMemcpyToSymbol(HIP_SYMBOL(d_bpow), bpows, sizeof(bpows));
+    // base-256 digit tables: pows[k][b] = shift by b*256^k bytes
+    static uint32_t pows[kPowLevels][256][32];
+    for (int n = 0; n < 32; ++n) pows[0][0][n] = 1u << n;  // identity
+    build_shift_operator(pows[0][1], 1);
+    for (int b = 2; b < 256; ++b) gf2_matmul(pows[0][b], pows[0][b - 1], pows[0][1]);
+    for (int k = 1; k < kPowLevels; ++k) {
+      for (int n = 0; n < 32; ++n) pows[k][0][n] = 1u << n;
+      // pows[k][1] = (pows[k-1][255] ∘ pows[k-1][1]) = shift by 256^k
+      gf2_matmul(pows[k][1], pows[k - 1][255], pows[k - 1][1]);
+      for (int b = 2; b < 256; ++b) gf2_matmul(pows[k][b], pows[k][b - 1], pows[k][1]);
+    }
+    hipMemcpyToSymbol(HIP_SYMBOL(d_pow256), pows, sizeof(pows));
   });
 }
 
@@ -160,10 +169,12 @@ __global__ __launch_bounds__(64) void crc_chunks_kernel(const uint8_t* data, siz
   }
   if (my_chunk >= nchunks) return;
   crc = ~crc;
-  // Shift by the bytes after this chunk and fold into the result.
+  // Shift by the bytes after this chunk (base-256 digit decomposition:
+  // at most kPowLevels matrix applications) and fold into the result.
   uint64_t dist = n - my_end;
-  for (int k = 0; k < kMaxBytePow && dist != 0; ++k, dist >>= 1) {
-    if (dist & 1) crc = dev_gf2_times(d_bpow[k], crc);
+  for (int k = 0; k < kPowLevels && dist != 0; ++k, dist >>= 8) {
+    uint32_t digit = (uint32_t)(dist & 0xff);
+    if (digit) crc = dev_gf2_times(d_pow256[k][digit], crc);
   }
   atomicXor(result, crc);
 }
