@@ -10,7 +10,7 @@ HIPCC     := /opt/rocm/bin/hipcc
 GPU_ARCH  := gfx950
 
 CXXFLAGS  := -O2 -g -std=c++17 -fPIC -pthread -Wall -Wno-unused-function \
-             -Isrc -I/opt/conda/include -msse4.2 -fvisibility=hidden -MMD -MP
+             -Isrc -I/opt/conda/include -msse4.2 -fno-omit-frame-pointer -MMD -MP
 LDFLAGS   := -shared -pthread -ldl -lz
 
 CORE_SRCS := $(wildcard src/base/*.cc) $(wildcard src/fiber/*.cc) $(wildcard src/rpc/*.cc) \

@@ -17,6 +17,7 @@
 #include "var/variable.h"
 
 namespace bam {
+std::string dump_fiber_stacks(int max_fibers);  // fiber/tracer.cc
 namespace policy {
 
 namespace {
@@ -127,11 +128,15 @@ void page_protobufs(Server* server, HttpResponse* resp) {
   resp->body.append(os.str());
 }
 
-void page_fibers(HttpResponse* resp) {
+void page_fibers(const HttpRequest& req, HttpResponse* resp) {
   std::ostringstream os;
   os << "workers: " << fiber_get_concurrency() << "\n";
   os << "fibers_created: " << fiber_count_created() << "\n";
   os << "fibers_active: " << fiber_count_active() << "\n";
+  if (req.query.count("st") != 0) {
+    // stack capture of suspended fibers (parity: /bthreads/<tid>?st=1)
+    os << bam::dump_fiber_stacks(64);
+  }
   resp->body.append(os.str());
 }
 
@@ -187,7 +192,7 @@ bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse
   } else if (p == "/protobufs") {
     page_protobufs(server, resp);
   } else if (p == "/fibers" || p == "/bthreads") {
-    page_fibers(resp);
+    page_fibers(req, resp);
   } else if (p == "/memory") {
     page_memory(resp);
   } else if (p == "/brpc_metrics" || p == "/metrics") {

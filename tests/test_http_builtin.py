@@ -104,3 +104,20 @@ def test_rpc_over_http(base):
     with urllib.request.urlopen(req, timeout=5) as resp:
         assert resp.status == 200
         assert resp.read() == b"http payload"
+
+
+def test_fiber_stack_tracer(base):
+    """/fibers?st=1 captures suspended fibers' stacks (≙ bthread tracer)."""
+    import brpc_amd as bb
+    # park some fibers so there is something to trace
+    import threading
+    t = threading.Thread(target=lambda: bb.core.fiber.usleep_test(600_000))
+    t.start()
+    import time
+    time.sleep(0.15)  # let the fiber park in fiber_usleep
+    st, body = get(base + "/fibers?st=1")
+    t.join()
+    assert st == 200
+    assert "fiber #" in body
+    # the parked fiber's stack should include the butex/usleep path
+    assert "butex" in body or "usleep" in body or "sched" in body, body[:2000]
