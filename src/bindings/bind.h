@@ -16,3 +16,4 @@ void bind_util(py::module_& m);
 void bind_memcache(py::module_& m);
 void bind_json2pb(py::module_& m);
 void bind_thrift(py::module_& m);
+void bind_codecs(py::module_& m);

@@ -193,3 +193,54 @@ void bind_util(py::module_& m) {
     return py::make_tuple(service, method, py::bytes(body));
   });
 }
+
+// ---- codecs + containers selftests ----
+#include "base/codecs.h"
+#include "base/containers.h"
+#include "var/variable.h"
+
+void bind_codecs(py::module_& m) {
+  auto c = m.def_submodule("codecs");
+  c.def("base64_encode", [](py::bytes data) {
+    std::string out;
+    bam::Base64Encode(data.cast<std::string>(), &out);
+    return out;
+  });
+  c.def("base64_decode", [](const std::string& data) -> py::object {
+    std::string out;
+    if (!bam::Base64Decode(data, &out)) return py::none();
+    return py::bytes(out);
+  });
+  c.def("sha1_hex", [](py::bytes data) { return bam::SHA1HexDigest(data.cast<std::string>()); });
+  c.def("murmur3_32", [](py::bytes data, uint32_t seed) {
+    std::string s = data.cast<std::string>();
+    return bam::MurmurHash3_32(s.data(), s.size(), seed);
+  }, py::arg("data"), py::arg("seed") = 0);
+  c.def("containers_selftest", [] {
+    bam::BoundedQueue<int> q(3);
+    if (!q.push(1) || !q.push(2) || !q.push(3) || q.push(4)) return false;
+    int v;
+    if (!q.pop(&v) || v != 1) return false;
+    if (!q.push(4)) return false;
+    bam::MPSCQueue<int> mq;
+    for (int i = 0; i < 100; ++i) mq.push(i);
+    for (int i = 0; i < 100; ++i) {
+      int x;
+      if (!mq.pop(&x) || x != i) return false;
+    }
+    bam::MRUCache<std::string, int> cache(2);
+    cache.Put("a", 1);
+    cache.Put("b", 2);
+    cache.Get("a");       // a is now MRU
+    cache.Put("c", 3);    // evicts b
+    return cache.Get("b") == nullptr && *cache.Get("a") == 1 && *cache.Get("c") == 3;
+  });
+  c.def("multidim_selftest", [] {
+    static bam::var::MultiDimension<bam::var::Adder<int64_t>> mdim(
+        "selftest_mdim", {"method"});
+    *mdim.get_stats({"echo"}) << 5;
+    *mdim.get_stats({"echo"}) << 2;
+    *mdim.get_stats({"sleep"}) << 1;
+    return mdim.count_stats() == 2 && mdim.get_stats({"echo"})->get_value() == 7;
+  });
+}

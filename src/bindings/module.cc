@@ -16,4 +16,5 @@ PYBIND11_MODULE(_core, m) {
   bind_memcache(m);
   bind_json2pb(m);
   bind_thrift(m);
+  bind_codecs(m);
 }

@@ -220,6 +220,61 @@ class PassiveStatus : public Variable {
   Fn fn_;
 };
 
+// ---------------- multi-dimension (labeled) vars ----------------
+
+// Parity: reference bvar MultiDimension (mvar): one logical metric with
+// per-label-value child reducers, dumped as name{label="v",...}.
+template <typename VarType>
+class MultiDimension : public Variable {
+ public:
+  MultiDimension(const std::string& name, std::vector<std::string> label_names)
+      : labels_(std::move(label_names)) {
+    expose(name);
+  }
+  ~MultiDimension() override {
+    hide();
+    for (auto& kv : children_) delete kv.second;
+  }
+
+  VarType* get_stats(const std::vector<std::string>& label_values) {
+    std::string key = join(label_values);
+    std::lock_guard<std::mutex> lk(mu_);
+    auto it = children_.find(key);
+    if (it != children_.end()) return it->second;
+    VarType* child = new VarType;
+    children_[key] = child;
+    return child;
+  }
+
+  size_t count_stats() const {
+    std::lock_guard<std::mutex> lk(mu_);
+    return children_.size();
+  }
+
+  void describe(std::ostream& os) const override {
+    std::lock_guard<std::mutex> lk(mu_);
+    bool first = true;
+    for (const auto& kv : children_) {
+      if (!first) os << " ";
+      first = false;
+      os << "{" << kv.first << "}=" << kv.second->get_value();
+    }
+  }
+
+ private:
+  static std::string join(const std::vector<std::string>& vals) {
+    std::string k;
+    for (size_t i = 0; i < vals.size(); ++i) {
+      if (i) k += ",";
+      k += vals[i];
+    }
+    return k;
+  }
+  std::vector<std::string> labels_;
+  mutable std::mutex mu_;
+  std::map<std::string, VarType*> children_;
+};
+
 // ---------------- windowed values (1 Hz sampler) ----------------
 
 // Samples an int64 source once per second into a 61-slot ring; value(w) =
