@@ -127,6 +127,7 @@ void run_remained(Transfer* tr) {
 
 // Entry point of every fresh fiber context. Never returns.
 void fiber_entry_fn(void* raw) {
+  BAM_ASAN_FINISH_SWITCH(nullptr, nullptr, nullptr);
   run_remained((Transfer*)raw);
   for (;;) {
     TaskGroup* g = current_task_group();
@@ -157,7 +158,16 @@ void TaskGroup::sched_to(FiberMeta* next, RemainedFn remained, void* remained_ar
   FiberMeta* cur = cur_;
   cur_ = next;
   Transfer tr{remained, remained_arg};
+#ifdef BAM_ASAN_ENABLED
+  const void* next_bottom = next->is_main ? nullptr : next->stack_base;
+  size_t next_size = next->is_main ? 0 : next->stack_size;
+  BAM_ASAN_START_SWITCH(&cur->asan_fake_stack, next_bottom, next_size);
+#endif
   Transfer* ret = (Transfer*)bam_jump_context(&cur->ctx_sp, next->ctx_sp, &tr);
+#ifdef BAM_ASAN_ENABLED
+  BAM_ASAN_FINISH_SWITCH(cur->asan_fake_stack, &cur->asan_stack_bottom,
+                         &cur->asan_stack_size);
+#endif
   run_remained(ret);
   // NOTE: `this` may be stale here if the fiber migrated; callers must
   // re-read tls_task_group after sched()/sched_to() returns.
@@ -211,7 +221,13 @@ void TaskGroup::run_main_loop() {
     if (m == nullptr) continue;  // stale (should not happen for queued fibers)
     Transfer tr{nullptr, nullptr};
     cur_ = m;
+#ifdef BAM_ASAN_ENABLED
+    BAM_ASAN_START_SWITCH(&main_meta_.asan_fake_stack, m->stack_base, m->stack_size);
+#endif
     Transfer* ret = (Transfer*)bam_jump_context(&main_meta_.ctx_sp, m->ctx_sp, &tr);
+#ifdef BAM_ASAN_ENABLED
+    BAM_ASAN_FINISH_SWITCH(main_meta_.asan_fake_stack, nullptr, nullptr);
+#endif
     run_remained(ret);
     cur_ = &main_meta_;
   }

@@ -4,6 +4,30 @@
 #pragma once
 
 #include <atomic>
+
+// ASan fiber support (parity: reference bthread/stack_inl.h ASan hooks):
+// every context switch notifies the sanitizer of the destination stack so
+// fake-stack bookkeeping and use-after-return detection stay correct
+// across fibers. No-ops without -fsanitize=address.
+#if defined(__SANITIZE_ADDRESS__)
+#define BAM_ASAN_ENABLED 1
+#elif defined(__has_feature)
+#if __has_feature(address_sanitizer)
+#define BAM_ASAN_ENABLED 1
+#endif
+#endif
+#ifdef BAM_ASAN_ENABLED
+extern "C" {
+void __sanitizer_start_switch_fiber(void** fake_stack_save, const void* bottom, size_t size);
+void __sanitizer_finish_switch_fiber(void* fake_stack_save, const void** bottom_old,
+                                     size_t* size_old);
+}
+#define BAM_ASAN_START_SWITCH(fss, bottom, size)   __sanitizer_start_switch_fiber(fss, bottom, size)
+#define BAM_ASAN_FINISH_SWITCH(fss, bottom_old, size_old)   __sanitizer_finish_switch_fiber(fss, bottom_old, size_old)
+#else
+#define BAM_ASAN_START_SWITCH(fss, bottom, size) ((void)0)
+#define BAM_ASAN_FINISH_SWITCH(fss, bottom_old, size_old) ((void)0)
+#endif
 #include <deque>
 #include <mutex>
 #include <vector>
@@ -19,6 +43,11 @@ typedef void (*RemainedFn)(void*);
 
 struct FiberMeta {
   void* ctx_sp = nullptr;
+#ifdef BAM_ASAN_ENABLED
+  void* asan_fake_stack = nullptr;
+  const void* asan_stack_bottom = nullptr;
+  size_t asan_stack_size = 0;
+#endif
   void (*fn)(void*) = nullptr;
   void* arg = nullptr;
   char* stack_base = nullptr;

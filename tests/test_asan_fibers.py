@@ -1,0 +1,56 @@
+"""Race/memory detection parity: the fiber runtime builds and
+passes its stress scenarios under AddressSanitizer with fiber-switch
+annotations (≙ reference ASan fiber support in bthread/stack_inl.h)."""
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.mark.slow
+def test_fiber_suite_under_asan(tmp_path):
+    main = tmp_path / "asan_main.cc"
+    main.write_text("""
+#include <cstdio>
+#include <cstdlib>
+#include "fiber/fiber.h"
+namespace bam { namespace selftest {
+int64_t start_join_test(int, int);
+bool urgent_test();
+int64_t usleep_test(int64_t);
+bool butex_wake_test();
+bool butex_timeout_test();
+int64_t mutex_test(int, int);
+bool countdown_test(int);
+bool fiber_key_test();
+}}
+using namespace bam::selftest;
+int main() {
+  if (start_join_test(50, 500) != 25000) return 1;
+  if (!urgent_test()) return 2;
+  if (usleep_test(10000) < 9000) return 3;
+  if (!butex_wake_test()) return 4;
+  if (!butex_timeout_test()) return 5;
+  if (mutex_test(8, 500) != 4000) return 6;
+  if (!countdown_test(20)) return 7;
+  if (!fiber_key_test()) return 8;
+  printf("asan fiber suite OK\\n");
+  return 0;
+}
+""")
+    exe = tmp_path / "asan_fibers"
+    srcs = [f"src/fiber/{f}" for f in os.listdir(f"{REPO}/src/fiber")
+            if f.endswith((".cc", ".S"))]
+    cmd = ["g++", "-O1", "-g", "-std=c++17", "-fsanitize=address",
+           "-fno-omit-frame-pointer", "-pthread", f"-I{REPO}/src", str(main),
+           *srcs, "src/base/logging.cc", "-o", str(exe)]
+    build = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert build.returncode == 0, build.stderr[-2000:]
+    env = dict(os.environ)
+    env["ASAN_OPTIONS"] = "detect_leaks=0"
+    run = subprocess.run([str(exe)], capture_output=True, text=True, timeout=120, env=env)
+    assert run.returncode == 0, run.stdout[-500:] + run.stderr[-2000:]
+    assert "OK" in run.stdout
