@@ -17,13 +17,11 @@
 #endif
 #endif
 #ifdef BAM_ASAN_ENABLED
-extern "C" {
-void __sanitizer_start_switch_fiber(void** fake_stack_save, const void* bottom, size_t size);
-void __sanitizer_finish_switch_fiber(void* fake_stack_save, const void** bottom_old,
-                                     size_t* size_old);
-}
-#define BAM_ASAN_START_SWITCH(fss, bottom, size)   __sanitizer_start_switch_fiber(fss, bottom, size)
-#define BAM_ASAN_FINISH_SWITCH(fss, bottom_old, size_old)   __sanitizer_finish_switch_fiber(fss, bottom_old, size_old)
+#include <sanitizer/common_interface_defs.h>
+#define BAM_ASAN_START_SWITCH(fss, bottom, size) \
+  __sanitizer_start_switch_fiber(fss, bottom, size)
+#define BAM_ASAN_FINISH_SWITCH(fss, bottom_old, size_old) \
+  __sanitizer_finish_switch_fiber(fss, bottom_old, size_old)
 #else
 #define BAM_ASAN_START_SWITCH(fss, bottom, size) ((void)0)
 #define BAM_ASAN_FINISH_SWITCH(fss, bottom_old, size_old) ((void)0)
