@@ -106,6 +106,9 @@ int compressed_echo_test(const std::string& addr, const std::string& payload,
 int64_t backup_request_test(int slow_port, int fast_port, int backup_ms, int calls);
 int start_intercepted_echo_server(const std::string& magic_logid);
 int call_with_logid(const std::string& addr, uint64_t log_id, std::string* err);
+int grpc_client_call(const std::string& addr, const std::string& full_method,
+                     const std::string& payload, int timeout_ms, std::string* response_out,
+                     std::string* err);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -151,6 +154,16 @@ void bind_rpc_combo(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   r.def("start_intercepted_server", &bam::rpctest::start_intercepted_echo_server,
         py::call_guard<py::gil_scoped_release>());
+  r.def("grpc_call", [](const std::string& addr, const std::string& method,
+                        const std::string& payload, int timeout_ms) {
+    std::string resp, err;
+    int rc;
+    {
+      py::gil_scoped_release rel;
+      rc = bam::rpctest::grpc_client_call(addr, method, payload, timeout_ms, &resp, &err);
+    }
+    return py::make_tuple(rc, py::bytes(resp), err);
+  }, py::arg("addr"), py::arg("method"), py::arg("payload"), py::arg("timeout_ms") = 3000);
   r.def("call_with_logid", [](const std::string& addr, uint64_t log_id) {
     std::string err;
     int rc;

@@ -22,6 +22,7 @@ int Channel::Init(EndPoint ep, const ChannelOptions* options) {
   policy::RegisterStdProtocol();
   if (options_.protocol == "redis") policy::RegisterRedisProtocol();
   if (options_.protocol == "thrift") policy::RegisterThriftProtocol();
+  if (options_.protocol == "grpc" || options_.protocol == "h2") policy::RegisterGrpcClientProtocol();
   protocol_index_ = FindProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
   return 0;
@@ -170,8 +171,16 @@ void IssueRPC(Controller* cntl) {
   }
   cntl->remote_side_ = ep;
   SessionId current = session_current_id(cntl->call.cid);
-  IOBuf packet;
   const Protocol* proto = GetProtocol(cntl->call.protocol_index);
+  if (proto != nullptr && proto->issue_request != nullptr) {
+    sock->add_pending_session(current);
+    cntl->call.pending_socket = sock->id();
+    if (proto->issue_request(sock.get(), cntl, current) != 0) {
+      session_error(current, EFAILEDSOCKET);
+    }
+    return;
+  }
+  IOBuf packet;
   if (proto != nullptr && proto->pack_request != nullptr) {
     proto->pack_request(&packet, cntl, current);
   } else {

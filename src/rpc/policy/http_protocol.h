@@ -31,6 +31,7 @@ struct HttpResponse {
 void RegisterHttpProtocol();
 void RegisterH2Protocol();  // h2 + gRPC (policy/h2_protocol.cc)
 void RegisterThriftProtocol();  // framed TBinary (policy/thrift_protocol.cc)
+void RegisterGrpcClientProtocol();  // h2/gRPC client (policy/h2_client.cc)
 
 // Implemented in rpc/builtin/builtin_services.cc; returns true if the path
 // matched a builtin page.

@@ -50,6 +50,9 @@ struct Protocol {
   void (*process_response)(InputMessageBase* msg) = nullptr;
   // Client-side request packing (header+meta+payload). nullptr = std only.
   void (*pack_request)(IOBuf* out, Controller* cntl, uint64_t correlation_id) = nullptr;
+  // Stateful client protocols (h2): take over the whole issue step
+  // (session bookkeeping + socket write). Returns 0 on success.
+  int (*issue_request)(Socket* sock, Controller* cntl, uint64_t correlation_id) = nullptr;
   // True for protocols without correlation ids: responses match requests
   // FIFO on the connection (redis/memcache pipelining).
   bool client_pipelined = false;

@@ -295,3 +295,29 @@ int call_with_logid(const std::string& addr, uint64_t log_id, std::string* err) 
 
 }  // namespace rpctest
 }  // namespace bam
+
+// ---- gRPC client (h2) ----
+
+namespace bam {
+namespace rpctest {
+
+int grpc_client_call(const std::string& addr, const std::string& full_method,
+                     const std::string& payload, int timeout_ms, std::string* response_out,
+                     std::string* err) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = timeout_ms;
+  opts.protocol = "grpc";
+  if (channel.Init(addr.c_str(), &opts) != 0) return -1;
+  Controller cntl;
+  IOBuf request, response;
+  request.append(payload);
+  channel.CallMethod(full_method, &cntl, &request, &response, nullptr);
+  if (err != nullptr) *err = cntl.ErrorText();
+  if (cntl.Failed()) return cntl.ErrorCode();
+  if (response_out != nullptr) *response_out = response.to_string();
+  return 0;
+}
+
+}  // namespace rpctest
+}  // namespace bam

@@ -57,3 +57,44 @@ def test_grpc_large_payload(addr):
     with grpc.insecure_channel(addr) as ch:
         call = _stub(ch, "/EchoService/Echo")
         assert call(data, timeout=20) == data
+
+
+# ---- OUR gRPC client ----
+
+def test_our_grpc_client_to_our_server(addr):
+    rc, resp, err = b.core.combo.grpc_call(addr, "EchoService/Echo", b"c2s payload")
+    assert rc == 0, err
+    assert resp == b"c2s payload"
+
+
+def test_our_grpc_client_to_real_grpc_server():
+    """Ultimate interop: OUR h2/gRPC client against the OFFICIAL grpc
+    python server."""
+    from concurrent import futures
+
+    class Handler(grpc.GenericRpcHandler):
+        def service(self, handler_call_details):
+            if handler_call_details.method == "/Real/Upper":
+                return grpc.unary_unary_rpc_method_handler(
+                    lambda req, ctx: req.upper(),
+                    request_deserializer=lambda x: x,
+                    response_serializer=lambda x: x)
+            return None
+
+    server = grpc.server(futures.ThreadPoolExecutor(max_workers=4))
+    server.add_generic_rpc_handlers((Handler(),))
+    port = server.add_insecure_port("127.0.0.1:0")
+    server.start()
+    try:
+        rc, resp, err = b.core.combo.grpc_call(f"127.0.0.1:{port}", "Real/Upper",
+                                               b"interop!", 5000)
+        assert rc == 0, err
+        assert resp == b"INTEROP!"
+    finally:
+        server.stop(0)
+
+
+def test_our_grpc_client_unknown_method(addr):
+    rc, resp, err = b.core.combo.grpc_call(addr, "EchoService/Missing", b"x")
+    assert rc != 0
+    assert "12" in err or "grpc-status" in err
