@@ -8,6 +8,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: test requires an MI355X GPU (run via gpurun)")
+    config.addinivalue_line("markers", "slow: long-running test (sanitizer builds, stress loops)")
 
 
 def pytest_collection_modifyitems(config, items):
