@@ -189,6 +189,7 @@ std::mutex g_scratch_mu;
 }  // namespace
 
 extern "C" uint32_t bam_gpu_crc32c(const void* dev_ptr, size_t n, uint32_t init, int dev) {
+  bam_gpu_quiesce(dev);  // order after any in-flight async HBM uploads
   ensure_init();
   if (n == 0) return init;
   int old_dev = -1;

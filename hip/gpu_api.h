@@ -55,4 +55,12 @@ int bam_gpu_fill(void* dst_dev, size_t n, uint64_t pattern, int dev);
 // Last error string (static buffer).
 const char* bam_gpu_last_error(void);
 
+// Fire-and-forget small H2D into HBM via the pinned staging ring + copy
+// kernel on the per-device staging stream (no host synchronization).
+// Returns nonzero if unavailable — caller must fall back to bam_gpu_memcpy.
+// Ordering: later direct gathers are stream-ordered after it; every other
+// HBM entry point drains pending uploads via bam_gpu_quiesce first.
+int bam_gpu_upload_async(void* dst_dev, const void* src_host, size_t n, int dev);
+void bam_gpu_quiesce(int dev);
+
 }  // extern "C"

@@ -6,6 +6,7 @@
 // carve into size-class blocks, per-class freelists. 288 GB HBM3E per GPU
 // means we bias toward big slabs and never return memory to the runtime.
 #include <hip/hip_runtime.h>
+#include <stdlib.h>
 #include <string.h>
 
 #include <mutex>
@@ -161,6 +162,11 @@ extern "C" void bam_gpu_free_pinned(void* p, uint32_t cap, int /*dev*/) {
 
 // ---------------- memcpy ----------------
 
+// Drains any in-flight async HBM uploads on dev's staging stream (defined
+// with the direct-gather state below). Every entry point that touches HBM
+// outside that stream must call this first for write→read ordering.
+extern "C" void bam_gpu_quiesce(int dev);
+
 extern "C" void bam_gpu_memcpy(void* dst, int dst_res, int dst_dev, const void* src,
                                int src_res, int src_dev, size_t n) {
   const bool dst_dev_mem = dst_res == 2;
@@ -169,6 +175,7 @@ extern "C" void bam_gpu_memcpy(void* dst, int dst_res, int dst_dev, const void* 
     ::memcpy(dst, src, n);
     return;
   }
+  bam_gpu_quiesce(dst_dev_mem ? dst_dev : src_dev);
   hipMemcpyKind kind;
   int dev = 0;
   if (dst_dev_mem && src_dev_mem) {
@@ -195,6 +202,7 @@ __global__ void fill_kernel(uint64_t* dst, size_t nwords, uint64_t pattern) {
 }
 
 extern "C" int bam_gpu_fill(void* dst_dev, size_t n, uint64_t pattern, int dev) {
+  bam_gpu_quiesce(dev);
   ScopedDevice sd(dev);
   size_t nwords = n / 8;
   if (nwords > 0) {
@@ -233,12 +241,315 @@ StageScratch g_stage[kMaxDev];
 std::mutex g_stage_mu;
 }  // namespace
 
+// --- direct-to-pinned fast path (latency-bound small batches) ---
+// A blocking hipMemcpy / hipDeviceSynchronize costs 10-20 µs each in host
+// wakeup latency; at ~100k batches/s that IS the staging ceiling. For small
+// batch totals a single kernel instead writes the scattered HBM spans
+// straight into the pinned bounce buffer (hipHostMalloc memory is
+// fine-grained coherent on ROCm: the device dereferences the host VA) and
+// then raises a pinned completion flag that the host spin-waits on — no
+// runtime synchronization call at all.
+namespace {
+
+struct DirectSpan {
+  const char* src;
+  unsigned int len;
+};
+constexpr int kDirectMaxSpans = 64;  // matches IOBuf kMaxIov
+
+struct DirectArgs {
+  DirectSpan spans[kDirectMaxSpans];
+  char* dst;                            // pinned host buffer (device-visible VA)
+  unsigned long long* counter;          // persistent device counter, monotonic
+  volatile unsigned long long* flag;    // pinned host flag (device-visible VA)
+  unsigned long long expect;            // counter value once ALL blocks of this call ran
+  unsigned long long ticket;            // value the last block publishes to *flag
+  int nspans;
+};
+
+__global__ void gather_direct_kernel(DirectArgs a) {
+  const int i = blockIdx.x;
+  const DirectSpan s = a.spans[i];
+  size_t off = 0;
+  for (int k = 0; k < i; ++k) off += a.spans[k].len;
+  const char* src = s.src;
+  char* dst = a.dst + off;
+  const unsigned int tid = threadIdx.x;
+  const unsigned int nt = blockDim.x;
+  if ((((uintptr_t)src ^ (uintptr_t)dst) & 15) == 0) {
+    unsigned int head = (16 - ((uintptr_t)src & 15)) & 15;
+    if (head > s.len) head = s.len;
+    for (unsigned int j = tid; j < head; j += nt) dst[j] = src[j];
+    const unsigned int nvec = (s.len - head) / 16;
+    const uint4* vs = (const uint4*)(src + head);
+    uint4* vd = (uint4*)(dst + head);
+    for (unsigned int j = tid; j < nvec; j += nt) vd[j] = vs[j];
+    for (unsigned int j = head + nvec * 16 + tid; j < s.len; j += nt) dst[j] = src[j];
+  } else {
+    for (unsigned int j = tid; j < s.len; j += nt) dst[j] = src[j];
+  }
+  __syncthreads();
+  if (tid == 0) {
+    __threadfence_system();  // publish this block's host writes before counting it done
+    unsigned long long done = atomicAdd(a.counter, 1ull) + 1;
+    if (done == a.expect) {
+      *a.flag = a.ticket;
+      __threadfence_system();
+    }
+  }
+}
+
+// Slim single-span variant: used for 1-span gathers and for the async
+// HBM upload path (pinned ring slot -> HBM block) — 1 KiB less kernarg
+// traffic per launch than the 64-span struct.
+__global__ void copy1_kernel(const char* src, char* dst, unsigned int len,
+                             unsigned long long* counter,
+                             volatile unsigned long long* flag,
+                             unsigned long long expect, unsigned long long ticket) {
+  const unsigned int tid = threadIdx.x;
+  const unsigned int nt = blockDim.x;
+  if ((((uintptr_t)src ^ (uintptr_t)dst) & 15) == 0) {
+    unsigned int head = (16 - ((uintptr_t)src & 15)) & 15;
+    if (head > len) head = len;
+    for (unsigned int j = tid; j < head; j += nt) dst[j] = src[j];
+    const unsigned int nvec = (len - head) / 16;
+    const uint4* vs = (const uint4*)(src + head);
+    uint4* vd = (uint4*)(dst + head);
+    for (unsigned int j = tid; j < nvec; j += nt) vd[j] = vs[j];
+    for (unsigned int j = head + nvec * 16 + tid; j < len; j += nt) dst[j] = src[j];
+  } else {
+    for (unsigned int j = tid; j < len; j += nt) dst[j] = src[j];
+  }
+  __syncthreads();
+  if (tid == 0) {
+    __threadfence_system();
+    unsigned long long done = atomicAdd(counter, 1ull) + 1;
+    if (done == expect) {
+      *flag = ticket;
+      __threadfence_system();
+    }
+  }
+}
+
+constexpr int kUploadSlots = 256;
+constexpr size_t kUploadSlotBytes = 32u << 10;
+
+struct DirectState {
+  std::mutex mu;
+  hipStream_t stream = nullptr;
+  unsigned long long* counter_dev = nullptr;
+  volatile unsigned long long* flag = nullptr;  // pinned
+  unsigned long long launched_blocks = 0;
+  unsigned long long ticket = 0;
+  int status = 0;  // 0 = untried, 1 = ok, -1 = unavailable
+  // async-upload staging ring (pinned): slot i reusable once *flag has
+  // passed the ticket of its previous occupant.
+  char* ring = nullptr;
+  unsigned long long slot_ticket[kUploadSlots] = {0};
+  unsigned long long nslots_used = 0;
+};
+DirectState g_direct[kMaxDev];
+
+// Spin until the device has drained every launch up to `want`; bounded,
+// with a hard hipStreamSynchronize fallback. Call with st.mu held.
+bool direct_wait(DirectState& st, unsigned long long want) {
+  for (uint64_t spin = 0; *st.flag < want; ++spin) {
+    if (spin > 4000000) {
+      if (hipStreamSynchronize(st.stream) != hipSuccess || *st.flag < want) return false;
+      break;
+    }
+#if defined(__x86_64__)
+    __builtin_ia32_pause();
+#endif
+  }
+  return true;
+}
+
+size_t direct_max_bytes() {
+  static size_t v = [] {
+    const char* e = getenv("BAM_GATHER_DIRECT_MAX");
+    return e != nullptr ? (size_t)strtoull(e, nullptr, 10) : (size_t)(128u << 10);
+  }();
+  return v;
+}
+
+// One-time per-device self-test: if the device cannot dereference pinned
+// host VAs on this platform/config, disable the path for good.
+bool direct_init(DirectState& st, int dev) {
+  if (hipStreamCreateWithFlags(&st.stream, hipStreamNonBlocking) != hipSuccess) return false;
+  if (hipMalloc(&st.counter_dev, 8) != hipSuccess) return false;
+  if (hipMemset(st.counter_dev, 0, 8) != hipSuccess) return false;
+  void* f = nullptr;
+  if (hipHostMalloc(&f, 64, hipHostMallocDefault) != hipSuccess) return false;
+  st.flag = (volatile unsigned long long*)f;
+  *st.flag = 0;
+  void* ring = nullptr;
+  if (hipHostMalloc(&ring, kUploadSlots * kUploadSlotBytes, hipHostMallocDefault) != hipSuccess)
+    return false;
+  st.ring = (char*)ring;
+  // self-test: gather 64 known bytes device -> pinned host
+  char* src_dev = nullptr;
+  if (hipMalloc(&src_dev, 64) != hipSuccess) return false;
+  char pattern[64];
+  for (int i = 0; i < 64; ++i) pattern[i] = (char)(i * 7 + 3);
+  hipMemcpy(src_dev, pattern, 64, hipMemcpyHostToDevice);
+  char* probe = nullptr;
+  if (hipHostMalloc((void**)&probe, 64, hipHostMallocDefault) != hipSuccess) {
+    hipFree(src_dev);
+    return false;
+  }
+  memset(probe, 0, 64);
+  DirectArgs a{};
+  a.spans[0].src = src_dev;
+  a.spans[0].len = 64;
+  a.dst = probe;
+  a.counter = st.counter_dev;
+  a.flag = st.flag;
+  a.expect = 1;
+  a.ticket = 1;
+  a.nspans = 1;
+  hipLaunchKernelGGL(gather_direct_kernel, dim3(1), dim3(256), 0, st.stream, a);
+  bool ok = hipGetLastError() == hipSuccess &&
+            hipStreamSynchronize(st.stream) == hipSuccess && *st.flag == 1 &&
+            memcmp(probe, pattern, 64) == 0;
+  st.launched_blocks = 1;
+  st.ticket = 1;
+  hipFree(src_dev);
+  hipHostFree(probe);
+  return ok;
+}
+
+// Returns 0 on success, nonzero to fall back to the scratch+memcpy path.
+int gather_direct(void* host_dst, const void* const* srcs, const size_t* lens,
+                  int nspans, int dev) {
+  // host_dst must be pinned (device-dereferenceable): the IOBuf bounce is
+  // normally RES_PINNED but falls back to malloc if hipHostMalloc failed.
+  // The bounce base is thread-stable, so cache the attribute lookup.
+  static thread_local struct {
+    const void* p;
+    bool ok;
+  } tls_chk = {nullptr, false};
+  if (tls_chk.p != host_dst) {
+    hipPointerAttribute_t attr;
+    tls_chk.ok = hipPointerGetAttributes(&attr, host_dst) == hipSuccess &&
+                 attr.type == hipMemoryTypeHost;
+    (void)hipGetLastError();
+    tls_chk.p = host_dst;
+  }
+  if (!tls_chk.ok) return 1;
+  DirectState& st = g_direct[dev < kMaxDev ? dev : 0];
+  unsigned long long ticket;
+  {
+    // Critical section covers ONLY ticket allocation + launch (so enqueue
+    // order matches ticket order on the stream). The completion spin runs
+    // outside the lock: holding st.mu while waiting would block every
+    // concurrent handler upload — and with M:N fibers, a blocked pthread
+    // mutex parks the whole worker thread.
+    std::lock_guard<std::mutex> lk(st.mu);
+    if (st.status == 0) st.status = direct_init(st, dev) ? 1 : -1;
+    if (st.status < 0) return 1;
+    st.launched_blocks += (unsigned long long)nspans;
+    const unsigned long long expect = st.launched_blocks;
+    ticket = ++st.ticket;
+    if (nspans == 1) {
+      hipLaunchKernelGGL(copy1_kernel, dim3(1), dim3(256), 0, st.stream, (const char*)srcs[0],
+                         (char*)host_dst, (unsigned int)lens[0], st.counter_dev, st.flag, expect,
+                         ticket);
+    } else {
+      DirectArgs a{};
+      for (int i = 0; i < nspans; ++i) {
+        a.spans[i].src = (const char*)srcs[i];
+        a.spans[i].len = (unsigned int)lens[i];
+      }
+      a.dst = (char*)host_dst;
+      a.counter = st.counter_dev;
+      a.flag = st.flag;
+      a.nspans = nspans;
+      a.expect = expect;
+      a.ticket = ticket;
+      hipLaunchKernelGGL(gather_direct_kernel, dim3(nspans), dim3(256), 0, st.stream, a);
+    }
+    if (hipGetLastError() != hipSuccess) {
+      st.status = -1;  // keep ticket/launched consistent by never using them again
+      return 1;
+    }
+  }
+  if (!direct_wait(st, ticket)) {
+    std::lock_guard<std::mutex> lk(st.mu);
+    st.status = -1;
+    return 1;
+  }
+  return 0;
+}
+
+// Fire-and-forget H2D: copies src into a pinned ring slot and enqueues a
+// slot->HBM copy kernel on the staging stream WITHOUT waiting. Ordering
+// with later reads comes from stream order (direct gather) or from
+// bam_gpu_quiesce (every other HBM entry point). Returns nonzero to tell
+// the caller to fall back to a synchronous copy.
+int upload_direct(void* dst_dev, const void* src, size_t n, int dev) {
+  if (n == 0) return 0;
+  if (n > kUploadSlotBytes) return 1;
+  DirectState& st = g_direct[dev < kMaxDev ? dev : 0];
+  std::lock_guard<std::mutex> lk(st.mu);
+  if (st.status == 0) st.status = direct_init(st, dev) ? 1 : -1;
+  if (st.status < 0) return 1;
+  const int slot = (int)(st.nslots_used++ % kUploadSlots);
+  // ring full wrap: the previous occupant's kernel must have drained
+  if (!direct_wait(st, st.slot_ticket[slot])) {
+    st.status = -1;
+    return 1;
+  }
+  char* sp = st.ring + (size_t)slot * kUploadSlotBytes;
+  ::memcpy(sp, src, n);
+  st.launched_blocks += 1;
+  const unsigned long long expect = st.launched_blocks;
+  const unsigned long long ticket = ++st.ticket;
+  hipLaunchKernelGGL(copy1_kernel, dim3(1), dim3(256), 0, st.stream, sp, (char*)dst_dev,
+                     (unsigned int)n, st.counter_dev, st.flag, expect, ticket);
+  if (hipGetLastError() != hipSuccess) {
+    st.status = -1;
+    return 1;
+  }
+  st.slot_ticket[slot] = ticket;
+  return 0;
+}
+
+}  // namespace
+
+extern "C" void bam_gpu_quiesce(int dev) {
+  if (dev < 0 || dev >= kMaxDev) dev = 0;
+  DirectState& st = g_direct[dev];
+  unsigned long long want;
+  {
+    std::lock_guard<std::mutex> lk(st.mu);
+    if (st.status != 1) return;
+    want = st.ticket;
+  }
+  if (*st.flag >= want) return;
+  if (!direct_wait(st, want)) {
+    std::lock_guard<std::mutex> lk(st.mu);
+    st.status = -1;
+  }
+}
+
+extern "C" int bam_gpu_upload_async(void* dst_dev, const void* src, size_t n, int dev) {
+  ScopedDevice sd(dev);
+  return upload_direct(dst_dev, src, n, dev);
+}
+
 extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
                                       const size_t* lens, int nspans, int dev) {
   size_t total = 0;
   for (int i = 0; i < nspans; ++i) total += lens[i];
   if (total == 0) return 0;
   ScopedDevice sd(dev);
+  if (total <= direct_max_bytes() && nspans <= kDirectMaxSpans &&
+      gather_direct(host_dst, srcs, lens, nspans, dev) == 0) {
+    return 0;
+  }
+  // Fallback paths read HBM outside the staging stream: drain uploads first.
+  bam_gpu_quiesce(dev);
   if (nspans == 1) {
     hipError_t e = hipMemcpy(host_dst, srcs[0], total, hipMemcpyDeviceToHost);
     return e == hipSuccess ? 0 : -1;

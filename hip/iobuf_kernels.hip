@@ -40,6 +40,7 @@ __global__ void copy_spans_kernel(int nspans) {
 // NOTE: launches + synchronizes. Batches share the d_spans symbol, so a
 // sync is required between batches; the common case is a single batch.
 int launch_spans(Span* spans, int nspans, int dev) {
+  bam_gpu_quiesce(dev);  // order after any in-flight async HBM uploads
   int old_dev = -1;
   hipGetDevice(&old_dev);
   if (dev != old_dev) hipSetDevice(dev);

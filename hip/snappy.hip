@@ -293,6 +293,7 @@ void emit_varint_host(uint8_t* dst, size_t v, int* n) {
 
 extern "C" int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_dev,
                                        size_t dst_cap, size_t* out_len, int dev) {
+  bam_gpu_quiesce(dev);  // order after any in-flight async HBM uploads
   if (n == 0) return -1;
   int old_dev = -1;
   hipGetDevice(&old_dev);
@@ -338,6 +339,7 @@ extern "C" int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_
 
 extern "C" int bam_gpu_snappy_decompress(const void* src_dev, size_t n, void* dst_dev,
                                          size_t dst_cap, size_t* out_len, int dev) {
+  bam_gpu_quiesce(dev);  // order after any in-flight async HBM uploads
   int old_dev = -1;
   hipGetDevice(&old_dev);
   if (dev != old_dev) hipSetDevice(dev);

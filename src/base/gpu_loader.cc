@@ -61,6 +61,9 @@ void do_load() {
   g_api.fill = (int (*)(void*, size_t, uint64_t, int))must_sym(h, "bam_gpu_fill");
   g_api.gather_to_host = (int (*)(void*, const void* const*, const size_t*, int,
                                   int))must_sym(h, "bam_gpu_gather_to_host");
+  g_api.upload_async =
+      (int (*)(void*, const void*, size_t, int))must_sym(h, "bam_gpu_upload_async");
+  g_api.quiesce = (void (*)(int))must_sym(h, "bam_gpu_quiesce");
   g_api.snappy_compress = (int (*)(const void*, size_t, void*, size_t, size_t*, int))must_sym(
       h, "bam_gpu_snappy_compress");
   g_api.snappy_decompress = (int (*)(const void*, size_t, void*, size_t, size_t*,
@@ -82,6 +85,7 @@ void do_load() {
     }};
     set_byte_mover(mover);
     set_gather_to_host(g_api.gather_to_host);
+    set_upload_async(g_api.upload_async);
     LOG(INFO) << "brpc_amd HIP runtime loaded: " << g_ndev << " GPU(s)";
   }
 }

@@ -23,6 +23,8 @@ struct GpuApi {
   int (*scatter)(void* const*, const size_t*, int, const void*, int);
   int (*fill)(void*, size_t, uint64_t, int);
   int (*gather_to_host)(void*, const void* const*, const size_t*, int, int);
+  int (*upload_async)(void*, const void*, size_t, int);
+  void (*quiesce)(int);
   int (*snappy_compress)(const void*, size_t, void*, size_t, size_t*, int);
   int (*snappy_decompress)(const void*, size_t, void*, size_t, size_t*, int);
   const char* (*last_error)(void);

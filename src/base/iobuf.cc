@@ -26,6 +26,17 @@ bool has_block_allocator(Residency res) {
 }
 void set_byte_mover(ByteMoverFns fns) { g_mover = fns; }
 
+static UploadAsyncFn g_upload_async = nullptr;
+void set_upload_async(UploadAsyncFn fn) {
+  // Default OFF: same-box A/B showed per-response upload kernels serialize
+  // on the staging stream and the batch gather must drain them, losing to
+  // the runtime's internal SDMA parallelism (137k vs 94k QPS at 64B/c32).
+  // BAM_UPLOAD_ASYNC=1 re-enables for experiments.
+  const char* e = getenv("BAM_UPLOAD_ASYNC");
+  if (e == nullptr || e[0] != '1') return;
+  g_upload_async = fn;
+}
+
 static void move_bytes(void* dst, Residency dres, int ddev, const void* src, Residency sres,
                        int sdev, size_t n) {
   if (dres != RES_HBM && sres != RES_HBM) {
@@ -369,7 +380,12 @@ int IOBuf::append_with_residency(const void* host_data, size_t n, Residency res,
     uint32_t cap = (uint32_t)std::min<size_t>(n, block_payload);
     Block* b = create_block(cap, res, dev);
     if (b == nullptr) return -1;
-    move_bytes(b->data, res, dev, p, RES_HOST, -1, cap);
+    // HBM: prefer the async upload leg (no blocking hipMemcpy per block —
+    // the response-side gather is stream-ordered after it).
+    if (res != RES_HBM || g_upload_async == nullptr ||
+        g_upload_async(b->data, p, cap, dev) != 0) {
+      move_bytes(b->data, res, dev, p, RES_HOST, -1, cap);
+    }
     b->size = cap;
     push_ref_back(BlockRef{0, cap, b});
     p += cap;

@@ -47,6 +47,12 @@ struct ByteMoverFns {
 };
 void set_byte_mover(ByteMoverFns fns);
 
+// Optional fire-and-forget host->HBM upload (pinned staging ring + copy
+// kernel, no host sync). Returns nonzero when the caller must fall back to
+// the synchronous byte mover. Installed by the HIP loader.
+typedef int (*UploadAsyncFn)(void* dst_dev, const void* src_host, size_t n, int dev);
+void set_upload_async(UploadAsyncFn fn);
+
 class IOBuf {
  public:
   static const uint32_t kDefaultBlockPayload = 8192;  // multiple of 4096
