@@ -9,6 +9,7 @@
 #include <memory>
 
 #include "bindings/bind.h"
+#include "rpc/authenticator.h"
 #include "rpc/channel.h"
 #include "rpc/controller.h"
 #include "rpc/parallel_channel.h"
@@ -77,9 +78,14 @@ class PyServer {
     svc->AddMethod(method, wrap_py_handler(std::move(fn)));
   }
 
-  int start(int port, int max_concurrency) {
+  int start(int port, int max_concurrency, const std::string& auth_user,
+            const std::string& auth_password) {
     ServerOptions opts;
     opts.max_concurrency = max_concurrency;
+    if (!auth_user.empty()) {
+      auth_.reset(new PasswordAuthenticator(auth_user, auth_password));
+      opts.auth = auth_.get();
+    }
     if (server_->Start(port, &opts) != 0) throw std::runtime_error("Server.start failed");
     return server_->listen_address().port;
   }
@@ -92,16 +98,22 @@ class PyServer {
  private:
   Server* server_;  // leaked deliberately: sockets may still reference it
   std::map<std::string, Service*> services_;
+  std::unique_ptr<PasswordAuthenticator> auth_;
 };
 
 class PyChannel {
  public:
   PyChannel(const std::string& addr, const std::string& lb, int timeout_ms, int max_retry,
-            int backup_request_ms, int compress) {
+            int backup_request_ms, int compress, const std::string& auth_user,
+            const std::string& auth_password) {
     ChannelOptions opts;
     opts.timeout_ms = timeout_ms;
     opts.max_retry = max_retry;
     opts.backup_request_ms = backup_request_ms;
+    if (!auth_user.empty()) {
+      auth_.reset(new PasswordAuthenticator(auth_user, auth_password));
+      opts.auth = auth_.get();
+    }
     compress_ = (CompressType)compress;
     int rc = lb.empty() ? channel_.Init(addr.c_str(), &opts)
                         : channel_.Init(addr.c_str(), lb.c_str(), &opts);
@@ -129,6 +141,7 @@ class PyChannel {
 
  private:
   Channel channel_;
+  std::unique_ptr<PasswordAuthenticator> auth_;  // outlives channel_ (declared first... kept until PyChannel dies)
   CompressType compress_ = COMPRESS_TYPE_NONE;
 };
 
@@ -149,17 +162,19 @@ void bind_api(py::module_& m) {
       .def(py::init<>())
       .def("add_method", &PyServer::add_method, py::arg("service"), py::arg("method"),
            py::arg("handler"))
-      .def("start", &PyServer::start, py::arg("port") = 0, py::arg("max_concurrency") = 0)
+      .def("start", &PyServer::start, py::arg("port") = 0, py::arg("max_concurrency") = 0,
+           py::arg("auth_user") = "", py::arg("auth_password") = "")
       .def("stop", &PyServer::stop)
       .def("running", &PyServer::running)
       .def("port", &PyServer::port)
       .def("processed", &PyServer::processed);
 
   py::class_<PyChannel>(m, "Channel")
-      .def(py::init<const std::string&, const std::string&, int, int, int, int>(),
+      .def(py::init<const std::string&, const std::string&, int, int, int, int,
+                    const std::string&, const std::string&>(),
            py::arg("addr"), py::arg("lb") = "", py::arg("timeout_ms") = 500,
            py::arg("max_retry") = 3, py::arg("backup_request_ms") = -1,
-           py::arg("compress") = 0)
+           py::arg("compress") = 0, py::arg("auth_user") = "", py::arg("auth_password") = "")
       .def("call", &PyChannel::call, py::arg("method"), py::arg("request"),
            py::arg("attachment") = std::string(), py::arg("timeout_ms") = 0,
            py::arg("log_id") = 0);

@@ -1,5 +1,7 @@
 #include "rpc/channel.h"
 
+#include "rpc/authenticator.h"
+
 #include <string.h>
 
 #include "base/logging.h"
@@ -170,6 +172,13 @@ void IssueRPC(Controller* cntl) {
     if (cache != nullptr) cache->store(sock->id(), std::memory_order_release);
   }
   cntl->remote_side_ = ep;
+  cntl->call.auth_data.clear();
+  if (cntl->call.auth != nullptr) {
+    if (cntl->call.auth->GenerateCredential(&cntl->call.auth_data) != 0) {
+      session_error(session_current_id(cntl->call.cid), ERPCAUTH);
+      return;
+    }
+  }
   SessionId current = session_current_id(cntl->call.cid);
   const Protocol* proto = GetProtocol(cntl->call.protocol_index);
   if (proto != nullptr && proto->issue_request != nullptr) {
@@ -204,6 +213,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->call.response = response;
   cntl->call.done = done;
   cntl->call.protocol_index = protocol_index_;
+  cntl->call.auth = options_.auth;
   if (options_.connection_type == "pooled") {
     static std::atomic<uint32_t> rr{0};
     cntl->call.connection_shard =

@@ -25,6 +25,9 @@ struct ChannelOptions {
   std::string connection_type = "single";
   int connection_pool_size = 8;
   std::string connection_group;
+  // Client credential source (rpc/authenticator.h); not owned. When set,
+  // every request carries RpcMeta.authentication_data.
+  const class Authenticator* auth = nullptr;
 };
 
 class ChannelBase {

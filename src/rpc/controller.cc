@@ -21,6 +21,8 @@ void Controller::Reset() {
   server_cid_ = 0;
   remote_stream_id_ = 0;
   response_stream_id_ = 0;
+  auth_context_ = nullptr;
+  concurrency_counted_ = false;
   call = Call();
 }
 

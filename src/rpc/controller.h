@@ -87,6 +87,8 @@ class Controller {
     int protocol_index = -1;                 // wire protocol for this call
     std::atomic<uint64_t>* socket_cache = nullptr;  // channel's cached socket id
     int connection_shard = 0;                        // pooled connection index
+    const class Authenticator* auth = nullptr;       // from ChannelOptions
+    std::string auth_data;                           // credential for this attempt
   };
   Call call;
 
@@ -96,6 +98,11 @@ class Controller {
   uint64_t remote_stream_id_ = 0;    // stream id carried by the peer's meta
   uint64_t response_stream_id_ = 0;  // stream accepted by this server (StreamAccept)
   int64_t server_cid_ = 0;  // correlation id to echo back
+  bool concurrency_counted_ = false;  // incremented server->concurrency (must decrement)
+  // Connection's verified auth result (owned by the server Socket);
+  // nullptr when the server has no authenticator. (rpc/authenticator.h)
+  const class AuthContext* auth_context_ = nullptr;
+  const class AuthContext* auth_context() const { return auth_context_; }
   CompressType response_compress_ = COMPRESS_TYPE_NONE;
   void set_response_compress_type(CompressType t) { response_compress_ = t; }
 

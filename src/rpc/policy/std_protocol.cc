@@ -4,6 +4,7 @@
 
 #include "base/logging.h"
 #include "base/time.h"
+#include "rpc/authenticator.h"
 #include "rpc/channel.h"
 #include "rpc/controller.h"
 #include "rpc/server.h"
@@ -42,10 +43,11 @@ void SerializeRpcMeta(const RpcMeta& meta, std::string* out) {
   if (meta.compress_type != 0) wire::put_int_field(out, 3, meta.compress_type);
   wire::put_int_field(out, 4, meta.correlation_id);
   if (meta.attachment_size != 0) wire::put_int_field(out, 5, meta.attachment_size);
+  if (!meta.auth_data.empty()) wire::put_str_field(out, 7, meta.auth_data);
   if (meta.stream_id != 0) {
     std::string sub;
     wire::put_int_field(&sub, 1, (int64_t)meta.stream_id);
-    wire::put_msg_field(out, 7, sub);
+    wire::put_msg_field(out, 8, sub);
   }
 }
 
@@ -92,7 +94,10 @@ bool ParseRpcMeta(const char* data, size_t n, RpcMeta* out) {
       case 5:
         out->attachment_size = (int32_t)r.varint();
         break;
-      case 7: {
+      case 7:
+        out->auth_data = r.read_string();
+        break;
+      case 8: {
         std::string sub = r.read_string();
         if (!r.ok()) return false;
         wire::Reader rr(sub.data(), sub.size());
@@ -153,6 +158,7 @@ void PackStdRequest(IOBuf* out, Controller* cntl, SessionId correlation_id) {
   meta.correlation_id = (int64_t)correlation_id;
   meta.attachment_size = (int32_t)cntl->request_attachment().size();
   meta.stream_id = cntl->call.stream_id;
+  meta.auth_data = cntl->call.auth_data;
   // Compress the serialized request (NOT the attachment — it is a raw
   // pass-through by contract, like the reference's baidu_std attachment).
   IOBuf body_buf;
@@ -225,7 +231,11 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
     sock->Write(&packet);
   }
   if (cntl->server_ != nullptr) {
-    cntl->server_->concurrency.fetch_sub(1, std::memory_order_relaxed);
+    // Only requests that reached the handler were counted in (early
+    // failures — unknown method, ELIMIT, auth — never incremented).
+    if (cntl->concurrency_counted_) {
+      cntl->server_->concurrency.fetch_sub(1, std::memory_order_relaxed);
+    }
     cntl->server_->nprocessed.fetch_add(1, std::memory_order_relaxed);
   }
   if (status != nullptr) *status << (monotonic_time_us() - start_us);
@@ -263,6 +273,28 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   cntl->remote_stream_id_ = msg->meta.stream_id;
   cntl->remote_side_ = sock->remote_side();
   cntl->local_side_ = sock->local_side();
+  // Per-connection authentication (verified once, cached on the socket).
+  if (server != nullptr && server->options().auth != nullptr) {
+    AuthContext* ctx = sock->auth_context();
+    if (ctx == nullptr) {
+      AuthContext* fresh = new AuthContext;
+      if (server->options().auth->VerifyCredential(msg->meta.auth_data, sock->remote_side(),
+                                                   fresh) != 0) {
+        delete fresh;
+        IOBuf* resp = new IOBuf;
+        cntl->SetFailed(ERPCAUTH, "authentication failed: " +
+                                      server->options().auth->GetUnauthorizedErrorText());
+        SocketId sid0 = sock->id();
+        int64_t cid0 = msg->meta.correlation_id;
+        delete msg;
+        SendStdResponse(sid0, cid0, cntl, resp, nullptr, monotonic_time_us());
+        return;
+      }
+      if (!sock->set_auth_context(fresh)) delete fresh;  // lost install race
+      ctx = sock->auth_context();
+    }
+    cntl->auth_context_ = ctx;
+  }
   IOBuf* resp = new IOBuf;
   SocketId sid = sock->id();
   int64_t cid = msg->meta.correlation_id;
@@ -322,6 +354,7 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
     }
   }
   server->concurrency.fetch_add(1, std::memory_order_relaxed);
+  cntl->concurrency_counted_ = true;
   rpc_dump::SampleRequest(msg->meta.service_name, msg->meta.method_name, req_data);
   (*fn)(cntl, req_data, resp, done);
   delete msg;

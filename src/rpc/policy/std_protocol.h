@@ -29,7 +29,8 @@ struct RpcMeta {
   int compress_type = 0;
   int64_t correlation_id = 0;
   int32_t attachment_size = 0;
-  uint64_t stream_id = 0;  // StreamSettings (field 7 sub-message)
+  std::string auth_data;   // authentication_data (field 7, first request on a connection)
+  uint64_t stream_id = 0;  // StreamSettings (field 8 sub-message)
 };
 
 void SerializeRpcMeta(const RpcMeta& meta, std::string* out);

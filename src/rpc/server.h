@@ -55,6 +55,9 @@ struct ServerOptions {
   bool has_builtin_services = true;
   RedisService* redis_service = nullptr;  // serve RESP on the same port
   InterceptorFn interceptor;              // request admission hook
+  // Verifies RpcMeta.authentication_data once per connection
+  // (rpc/authenticator.h); not owned. nullptr = no auth.
+  const class Authenticator* auth = nullptr;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)

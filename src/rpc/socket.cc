@@ -1,5 +1,7 @@
 #include "rpc/socket.h"
 
+#include "rpc/authenticator.h"
+
 #include <errno.h>
 #include <string.h>
 #include <poll.h>
@@ -183,6 +185,7 @@ void Socket::Recycle() {
     fd_.store(-1, std::memory_order_release);
   }
   read_buf_.clear();
+  delete (AuthContext*)auth_ctx_.exchange(nullptr, std::memory_order_acq_rel);
   on_edge_triggered_events_ = nullptr;
   on_failed_ = nullptr;
   // version: odd -> next even (free state)

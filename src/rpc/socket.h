@@ -107,6 +107,18 @@ class Socket {
   void* protocol_ctx = nullptr;
   void (*protocol_ctx_deleter)(void*) = nullptr;
 
+  // Server-side auth result, set on the first verified request of the
+  // connection (rpc/authenticator.h); owned by the socket, freed at
+  // recycle. nullptr = not yet authenticated.
+  class AuthContext* auth_context() const {
+    return (class AuthContext*)auth_ctx_.load(std::memory_order_acquire);
+  }
+  // Returns false if another thread won the race (ctx not installed).
+  bool set_auth_context(class AuthContext* ctx) {
+    void* expected = nullptr;
+    return auth_ctx_.compare_exchange_strong(expected, ctx, std::memory_order_acq_rel);
+  }
+
   // Correlation sessions waiting for responses on this socket; failed
   // when the connection breaks. (Parity: reference conducts errors to
   // ids; we keep an explicit registry.)
@@ -164,6 +176,7 @@ class Socket {
   EndPoint remote_side_;
   EndPoint local_side_;
   void* user_ = nullptr;
+  std::atomic<void*> auth_ctx_{nullptr};
   std::function<void(Socket*)> on_edge_triggered_events_;
   std::function<void(SocketId)> on_failed_;
   IOBuf read_buf_;
