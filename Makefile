@@ -10,8 +10,8 @@ HIPCC     := /opt/rocm/bin/hipcc
 GPU_ARCH  := gfx950
 
 CXXFLAGS  := -O2 -g -std=c++17 -fPIC -pthread -Wall -Wno-unused-function \
-             -Isrc -I/opt/conda/include -msse4.2 -fno-omit-frame-pointer -MMD -MP
-LDFLAGS   := -shared -pthread -ldl -lz
+             -Isrc -I/usr/include -I/opt/conda/include -msse4.2 -fno-omit-frame-pointer -MMD -MP
+LDFLAGS   := -shared -pthread -ldl -lz -lssl -lcrypto
 
 CORE_SRCS := $(wildcard src/base/*.cc) $(wildcard src/fiber/*.cc) $(wildcard src/rpc/*.cc) \
              $(wildcard src/rpc/policy/*.cc) $(wildcard src/rpc/builtin/*.cc) \

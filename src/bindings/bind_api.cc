@@ -10,6 +10,7 @@
 
 #include "bindings/bind.h"
 #include "rpc/authenticator.h"
+#include "rpc/ssl_util.h"
 #include "rpc/channel.h"
 #include "rpc/controller.h"
 #include "rpc/parallel_channel.h"
@@ -79,13 +80,16 @@ class PyServer {
   }
 
   int start(int port, int max_concurrency, const std::string& auth_user,
-            const std::string& auth_password) {
+            const std::string& auth_password, const std::string& ssl_cert,
+            const std::string& ssl_key) {
     ServerOptions opts;
     opts.max_concurrency = max_concurrency;
     if (!auth_user.empty()) {
       auth_.reset(new PasswordAuthenticator(auth_user, auth_password));
       opts.auth = auth_.get();
     }
+    opts.ssl_cert = ssl_cert;
+    opts.ssl_key = ssl_key;
     if (server_->Start(port, &opts) != 0) throw std::runtime_error("Server.start failed");
     return server_->listen_address().port;
   }
@@ -105,8 +109,9 @@ class PyChannel {
  public:
   PyChannel(const std::string& addr, const std::string& lb, int timeout_ms, int max_retry,
             int backup_request_ms, int compress, const std::string& auth_user,
-            const std::string& auth_password) {
+            const std::string& auth_password, bool ssl) {
     ChannelOptions opts;
+    opts.ssl = ssl;
     opts.timeout_ms = timeout_ms;
     opts.max_retry = max_retry;
     opts.backup_request_ms = backup_request_ms;
@@ -163,7 +168,8 @@ void bind_api(py::module_& m) {
       .def("add_method", &PyServer::add_method, py::arg("service"), py::arg("method"),
            py::arg("handler"))
       .def("start", &PyServer::start, py::arg("port") = 0, py::arg("max_concurrency") = 0,
-           py::arg("auth_user") = "", py::arg("auth_password") = "")
+           py::arg("auth_user") = "", py::arg("auth_password") = "",
+           py::arg("ssl_cert") = "", py::arg("ssl_key") = "")
       .def("stop", &PyServer::stop)
       .def("running", &PyServer::running)
       .def("port", &PyServer::port)
@@ -171,13 +177,22 @@ void bind_api(py::module_& m) {
 
   py::class_<PyChannel>(m, "Channel")
       .def(py::init<const std::string&, const std::string&, int, int, int, int,
-                    const std::string&, const std::string&>(),
+                    const std::string&, const std::string&, bool>(),
            py::arg("addr"), py::arg("lb") = "", py::arg("timeout_ms") = 500,
            py::arg("max_retry") = 3, py::arg("backup_request_ms") = -1,
-           py::arg("compress") = 0, py::arg("auth_user") = "", py::arg("auth_password") = "")
+           py::arg("compress") = 0, py::arg("auth_user") = "", py::arg("auth_password") = "",
+           py::arg("ssl") = false)
       .def("call", &PyChannel::call, py::arg("method"), py::arg("request"),
            py::arg("attachment") = std::string(), py::arg("timeout_ms") = 0,
            py::arg("log_id") = 0);
+
+  m.def("gen_self_signed_cert", [](const std::string& cn) {
+    std::string cert, key;
+    if (ssl::GenerateSelfSignedCert(&cert, &key, cn) != 0) {
+      throw std::runtime_error(std::string("cert generation failed: ") + ssl::LastError());
+    }
+    return py::make_tuple(cert, key);
+  }, py::arg("cn") = "localhost");
 }
 
 // ---- redis bindings ----

@@ -164,7 +164,7 @@ void IssueRPC(Controller* cntl) {
     }
   }
   if (!sock) {
-    if (GetClientSocket(ep, &sock, cntl->call.connection_shard) != 0) {
+    if (GetClientSocket(ep, &sock, cntl->call.connection_shard, cntl->call.ssl) != 0) {
       // Conduct the failure through the session so retry/ending logic runs.
       session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
       return;
@@ -214,6 +214,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->call.done = done;
   cntl->call.protocol_index = protocol_index_;
   cntl->call.auth = options_.auth;
+  cntl->call.ssl = options_.ssl;
   if (options_.connection_type == "pooled") {
     static std::atomic<uint32_t> rr{0};
     cntl->call.connection_shard =

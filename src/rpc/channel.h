@@ -28,6 +28,9 @@ struct ChannelOptions {
   // Client credential source (rpc/authenticator.h); not owned. When set,
   // every request carries RpcMeta.authentication_data.
   const class Authenticator* auth = nullptr;
+  // TLS to the server (certificate verification off, like the reference's
+  // default ChannelSSLOptions).
+  bool ssl = false;
 };
 
 class ChannelBase {

@@ -89,6 +89,7 @@ class Controller {
     int connection_shard = 0;                        // pooled connection index
     const class Authenticator* auth = nullptr;       // from ChannelOptions
     std::string auth_data;                           // credential for this attempt
+    bool ssl = false;                                // TLS client connection
   };
   Call call;
 

@@ -59,9 +59,15 @@ void InputMessenger::DispatchMessage(InputMessageBase* msg, int protocol_index) 
 
 void InputMessenger::OnNewMessages(Socket* s) {
   const size_t kOnceRead = 256 * 1024;
+  // TLS: input edges drive the (non-blocking) handshake until it is done;
+  // no application bytes exist before that.
+  if (s->ssl_state() == 1) {
+    if (s->ssl_continue_handshake() != 0) return;  // failed -> SetFailed
+    if (s->ssl_state() == 1) return;               // wants more transport data
+  }
   bool eof = false;
   while (!s->Failed()) {
-    ssize_t nr = s->read_buf().append_from_file_descriptor(s->fd(), kOnceRead);
+    ssize_t nr = s->read_bytes(&s->read_buf(), kOnceRead);
     if (nr < 0) {
       if (errno == EAGAIN || errno == EWOULDBLOCK) {
         // drained; fall through to parse what we have, then return

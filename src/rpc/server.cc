@@ -1,5 +1,7 @@
 #include "rpc/server.h"
 
+#include "rpc/ssl_util.h"
+
 #include <sys/socket.h>
 #include <unistd.h>
 
@@ -61,6 +63,18 @@ void Server::OnNewConnections(Socket* listen_socket) {
     SocketId sid;
     if (Socket::Create(opts, &sid) != 0) {
       ::close(fd);
+      continue;
+    }
+    if (server->ssl_ctx_ != nullptr) {
+      SocketUniquePtr sp;
+      if (Socket::Address(sid, &sp) == 0) {
+        void* ssl = ssl::NewSsl(server->ssl_ctx_, fd, /*client=*/false);
+        if (ssl != nullptr) {
+          sp->set_ssl(ssl);
+        } else {
+          sp->SetFailed(ECONNRESET, "SSL_new failed");
+        }
+      }
     }
   }
 }
@@ -74,6 +88,13 @@ int Server::Start(int port, const ServerOptions* opt) {
 int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   if (IsRunning()) return -1;
   if (opt != nullptr) options_ = *opt;
+  if (!options_.ssl_cert.empty() || !options_.ssl_key.empty()) {
+    ssl_ctx_ = ssl::NewServerCtx(options_.ssl_cert, options_.ssl_key);
+    if (ssl_ctx_ == nullptr) {
+      LOG(ERROR) << "server TLS context failed: " << ssl::LastError();
+      return -1;
+    }
+  }
   policy::RegisterStdProtocol();
   policy::RegisterH2Protocol();
   policy::RegisterThriftProtocol();

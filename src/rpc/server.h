@@ -58,6 +58,10 @@ struct ServerOptions {
   // Verifies RpcMeta.authentication_data once per connection
   // (rpc/authenticator.h); not owned. nullptr = no auth.
   const class Authenticator* auth = nullptr;
+  // TLS: PEM content (starting with "-----BEGIN") or file paths. Both
+  // set => every accepted connection speaks TLS (rpc/ssl_util.h).
+  std::string ssl_cert;
+  std::string ssl_key;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)
@@ -105,6 +109,7 @@ class Server {
   SocketId listen_socket_ = 0;
   std::atomic<bool> running_{false};
   InputMessenger messenger_;
+  void* ssl_ctx_ = nullptr;  // SSL_CTX* when TLS enabled (never freed: sockets may outlive Stop)
 };
 
 }  // namespace bam

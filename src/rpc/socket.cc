@@ -1,6 +1,7 @@
 #include "rpc/socket.h"
 
 #include "rpc/authenticator.h"
+#include "rpc/ssl_util.h"
 
 #include <errno.h>
 #include <string.h>
@@ -86,6 +87,8 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   s->protocol_ctx = nullptr;
   s->protocol_ctx_deleter = nullptr;
   s->input_events_.store(0, std::memory_order_relaxed);
+  s->ssl_ = nullptr;
+  s->ssl_state_.store(0, std::memory_order_relaxed);
   s->write_head_.store(nullptr, std::memory_order_relaxed);
   s->in_bytes = 0;
   s->out_bytes = 0;
@@ -186,6 +189,9 @@ void Socket::Recycle() {
   }
   read_buf_.clear();
   delete (AuthContext*)auth_ctx_.exchange(nullptr, std::memory_order_acq_rel);
+  ssl::FreeSsl(ssl_);
+  ssl_ = nullptr;
+  ssl_state_.store(0, std::memory_order_relaxed);
   on_edge_triggered_events_ = nullptr;
   on_failed_ = nullptr;
   // version: odd -> next even (free state)
@@ -266,6 +272,71 @@ void Socket::ReleaseAllWriteRequests(WriteRequest* fifo_head, int err) {
   }
 }
 
+// ---------------- TLS byte paths ----------------
+
+int Socket::ssl_continue_handshake() {
+  std::lock_guard<std::mutex> lk(ssl_hs_mu_);
+  if (ssl_state_.load(std::memory_order_acquire) != 1) return 0;
+  int rc = ssl::HandshakeStep(ssl_);
+  if (rc < 0) {
+    SetFailed(ECONNRESET, (std::string("TLS handshake failed: ") + ssl::LastError()).c_str());
+    return -1;
+  }
+  if (rc == 1) ssl_state_.store(2, std::memory_order_release);
+  return 0;
+}
+
+int Socket::ssl_handshake_wait() {
+  const int64_t deadline = monotonic_time_us() + 30LL * 1000000;
+  while (ssl_state_.load(std::memory_order_acquire) == 1) {
+    if (Failed()) return -1;
+    if (ssl_continue_handshake() != 0) return -1;
+    if (ssl_state_.load(std::memory_order_acquire) == 2) break;
+    if (monotonic_time_us() > deadline) {
+      SetFailed(ETIMEDOUT, "TLS handshake timed out");
+      return -1;
+    }
+    // The step wants transport IO; wait briefly for readability (the
+    // common WANT_READ case; WANT_WRITE resolves on the next poll too
+    // since the socket buffer drains in the background).
+    struct pollfd pfd;
+    pfd.fd = fd();
+    pfd.events = POLLIN | POLLOUT;
+    pfd.revents = 0;
+    ::poll(&pfd, 1, 20);
+  }
+  return 0;
+}
+
+ssize_t Socket::write_bytes(IOBuf* data, bool may_block) {
+  if (ssl_ == nullptr) return data->cut_into_file_descriptor(fd());
+  if (ssl_state_.load(std::memory_order_acquire) != 2) {
+    if (!may_block) {
+      errno = EAGAIN;  // KeepWrite fiber will drive the handshake
+      return -1;
+    }
+    if (ssl_handshake_wait() != 0) {
+      errno = EPIPE;
+      return -1;
+    }
+  }
+  char tmp[16384];
+  size_t n = data->copy_to(tmp, sizeof(tmp), 0);
+  if (n == 0) return 0;
+  ssize_t rc = ssl::Write(ssl_, tmp, n);
+  if (rc > 0) data->pop_front((size_t)rc);
+  return rc;
+}
+
+ssize_t Socket::read_bytes(IOBuf* out, size_t max) {
+  if (ssl_ == nullptr) return out->append_from_file_descriptor(fd(), max);
+  char tmp[16384];
+  size_t want = max < sizeof(tmp) ? max : sizeof(tmp);
+  ssize_t rc = ssl::Read(ssl_, tmp, want);
+  if (rc > 0) out->append(tmp, (size_t)rc);
+  return rc;
+}
+
 int Socket::wait_epoll_out(int64_t abstime_us) {
   int v = epollout_butex_->load(std::memory_order_acquire);
   if (Failed()) return -1;
@@ -344,6 +415,9 @@ int Socket::DoWrite(WriteRequest* req) {
       get_local_side(fd(), &local_side_);
       connecting_.store(false, std::memory_order_release);
     }
+    if (ssl_ != nullptr && ssl_state_.load(std::memory_order_acquire) == 1) {
+      if (ssl_handshake_wait() != 0) continue;  // SetFailed inside
+    }
     // Coalesce queued successors into one writev/staging batch (parity:
     // reference DoWrite cutting up to NWMAX iovecs across requests). The
     // FIFO tail (the node write_head_ may point at) is never merged.
@@ -358,7 +432,7 @@ int Socket::DoWrite(WriteRequest* req) {
       cur->next.store(nx->next.load(std::memory_order_acquire), std::memory_order_relaxed);
       release_write_request(nx, 0);
     }
-    ssize_t nw = cur->data.cut_into_file_descriptor(fd());
+    ssize_t nw = write_bytes(&cur->data, /*may_block=*/true);
     if (nw < 0) {
       if (errno == EAGAIN || errno == EWOULDBLOCK) {
         wait_epoll_out(monotonic_time_us() + 100000);  // 100ms backstop, then retry
@@ -428,7 +502,7 @@ int Socket::Write(IOBuf* data, const WriteOptions* opt) {
   // We own the queue. Inline attempt only when connected (never block the
   // caller); otherwise hand to a KeepWrite fiber.
   if (!connecting_.load(std::memory_order_acquire)) {
-    ssize_t nw = req->data.cut_into_file_descriptor(fd());
+    ssize_t nw = write_bytes(&req->data, /*may_block=*/false);
     if (nw < 0 && errno != EAGAIN && errno != EWOULDBLOCK && errno != EINTR) {
       int err = errno;
       SetFailed(err, strerror(err));

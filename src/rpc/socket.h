@@ -133,6 +133,25 @@ class Socket {
   // Blocks current fiber until the fd is writable (or failed/timeout).
   int wait_epoll_out(int64_t abstime_us);
 
+  // --- TLS (rpc/ssl_util.h; handles are opaque so this header stays
+  // OpenSSL-free). The handshake is driven lazily: from write paths via
+  // write_bytes() and from input edges via ssl_continue_handshake().
+  void set_ssl(void* ssl) {
+    ssl_ = ssl;
+    ssl_state_.store(1, std::memory_order_release);
+  }
+  bool ssl_enabled() const { return ssl_ != nullptr; }
+  int ssl_state() const { return ssl_state_.load(std::memory_order_acquire); }
+  // One non-blocking handshake step; -1 = failed (socket SetFailed).
+  int ssl_continue_handshake();
+  // Poll-loop handshake (client side / write path); 0 done, -1 failed.
+  int ssl_handshake_wait();
+  // TLS-aware byte movement; plain fd readv/writev (with HBM staging)
+  // when TLS is off. may_block=false never waits for the handshake
+  // (returns EAGAIN so the caller hands off to KeepWrite).
+  ssize_t write_bytes(IOBuf* data, bool may_block);
+  ssize_t read_bytes(IOBuf* out, size_t max);
+
   // Unwritten bytes queued on this socket; Write fails with EOVERCROWDED
   // above -socket_max_unwritten_bytes (parity: reference socket.cpp:1640).
   std::atomic<int64_t> unwritten_bytes{0};
@@ -177,6 +196,9 @@ class Socket {
   EndPoint local_side_;
   void* user_ = nullptr;
   std::atomic<void*> auth_ctx_{nullptr};
+  void* ssl_ = nullptr;                 // SSL* when TLS is enabled
+  std::atomic<int> ssl_state_{0};       // 0 off, 1 handshaking, 2 ready
+  std::mutex ssl_hs_mu_;                // serializes handshake stepping
   std::function<void(Socket*)> on_edge_triggered_events_;
   std::function<void(SocketId)> on_failed_;
   IOBuf read_buf_;

@@ -1,5 +1,6 @@
 #include "rpc/socket_map.h"
 
+#include <errno.h>
 #include <poll.h>
 #include <unistd.h>
 
@@ -10,6 +11,7 @@
 #include "base/time.h"
 #include "fiber/fiber.h"
 #include "rpc/input_messenger.h"
+#include "rpc/ssl_util.h"
 
 namespace bam {
 
@@ -123,13 +125,15 @@ bool IsEndpointIsolated(const EndPoint& ep) {
   return h != nullptr && h->isolated.load(std::memory_order_acquire);
 }
 
-int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard) {
+int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ssl) {
   if (IsEndpointIsolated(ep)) {
     errno = EHOSTDOWN;
     return -1;
   }
   ClientSocketMap& m = the_map();
-  const auto key = std::make_pair(ep, shard);
+  // TLS connections never share plaintext sockets: fold the flag into the
+  // map key (shards are small, bit 30 is free).
+  const auto key = std::make_pair(ep, shard | (ssl ? (1 << 30) : 0));
   {
     std::lock_guard<std::mutex> lk(m.mu);
     auto it = m.sockets.find(key);
@@ -148,6 +152,18 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard) {
   SocketId sid;
   if (Socket::Create(opts, &sid) != 0) return -1;
   if (Socket::Address(sid, out) != 0) return -1;
+  if (ssl) {
+    static void* g_client_ctx = ssl::NewClientCtx();  // process-lifetime
+    void* h = g_client_ctx != nullptr
+                  ? ssl::NewSsl(g_client_ctx, (*out)->fd(), /*client=*/true)
+                  : nullptr;
+    if (h == nullptr) {
+      (*out)->SetFailed(ECONNRESET, "client SSL_new failed");
+      out->reset(nullptr);
+      return -1;
+    }
+    (*out)->set_ssl(h);
+  }
   {
     std::lock_guard<std::mutex> lk(m.mu);
     auto it = m.sockets.find(key);

@@ -11,7 +11,8 @@ namespace bam {
 // if needed. 0 on success. `shard` > 0 selects a distinct pooled
 // connection (parity: reference pooled connection_type — multiple
 // connections to one server spread parse/write parallelism).
-int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard = 0);
+int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard = 0,
+                    bool ssl = false);
 
 // Drops the cached socket for ep (e.g. after failure).
 void RemoveClientSocket(const EndPoint& ep, SocketId expected);
