@@ -63,3 +63,31 @@ clean:
 -include $(shell find build -name '*.d' 2>/dev/null)
 
 .PHONY: all core clean
+
+# ---------------- fuzzing (SURVEY §4 parity: reference test/fuzzing/) ----------------
+# Whole core compiled by ROCm clang with ASan + fuzzer coverage; each
+# harness in tests/fuzz/ links against it. `make fuzz` builds all targets
+# into build/fuzz/bin/.
+FUZZ_CLANG   := /opt/rocm/lib/llvm/bin/clang++
+FUZZ_FLAGS   := -O1 -g -std=c++17 -fPIC -pthread -fsanitize=address,fuzzer-no-link \
+                -fno-omit-frame-pointer -Isrc -I/usr/include -I/opt/conda/include -MMD -MP
+FUZZ_OBJDIR  := build/fuzz
+FUZZ_OBJS    := $(patsubst src/%.cc,$(FUZZ_OBJDIR)/%.o,$(CORE_SRCS)) $(FUZZ_OBJDIR)/fiber/context.o
+FUZZ_BINS    := $(patsubst tests/fuzz/%.cc,$(FUZZ_OBJDIR)/bin/%,$(wildcard tests/fuzz/*.cc))
+
+$(FUZZ_OBJDIR)/%.o: src/%.cc
+	@mkdir -p $(dir $@)
+	$(FUZZ_CLANG) $(FUZZ_FLAGS) -c $< -o $@
+
+$(FUZZ_OBJDIR)/fiber/context.o: src/fiber/context.S
+	@mkdir -p $(dir $@)
+	$(FUZZ_CLANG) -c $< -o $@
+
+$(FUZZ_OBJDIR)/bin/%: tests/fuzz/%.cc $(FUZZ_OBJS)
+	@mkdir -p $(dir $@)
+	$(FUZZ_CLANG) $(FUZZ_FLAGS:fuzzer-no-link=fuzzer) $< $(FUZZ_OBJS) -ldl -lz -lssl -lcrypto -o $@
+
+fuzz: $(FUZZ_BINS)
+
+-include $(FUZZ_OBJS:.o=.d)
+.PHONY: fuzz

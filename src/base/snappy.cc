@@ -222,6 +222,11 @@ void Compress(const char* src, size_t n, std::string* out) {
 bool Uncompress(const char* compressed, size_t n, std::string* out) {
   size_t len;
   if (!GetUncompressedLength(compressed, n, &len)) return false;
+  // Malloc-bomb guard (found by tests/fuzz/fuzz_snappy.cc): the preamble
+  // can claim any length; real snappy streams expand at most 64x per
+  // 2-byte copy tag, so a claim beyond 64*n is certainly corrupt — reject
+  // BEFORE allocating.
+  if (len > 64 * n + 1024) return false;
   out->resize(len);
   return RawUncompress(compressed, n, len > 0 ? &(*out)[0] : (char*)"");
 }

@@ -223,5 +223,9 @@ void RegisterHttpProtocol() {
   });
 }
 
+bool ParseHttpHead(const std::string& head, HttpRequest* out) {
+  return parse_headers(head, out);
+}
+
 }  // namespace policy
 }  // namespace bam

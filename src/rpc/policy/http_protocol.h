@@ -37,5 +37,9 @@ void RegisterGrpcClientProtocol();  // h2/gRPC client (policy/h2_client.cc)
 // matched a builtin page.
 bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse* resp);
 
+// Parses a request head ("METHOD /path HTTP/1.1\r\nHeader: v\r\n...").
+// Public for the fuzz harness (tests/fuzz/fuzz_http.cc).
+bool ParseHttpHead(const std::string& head, HttpRequest* out);
+
 }  // namespace policy
 }  // namespace bam

@@ -328,7 +328,7 @@ class LatencyRecorder {
   Adder<int64_t> count_;
   Adder<int64_t> sum_us_;
   std::atomic<int64_t> window_max_{0};
-  static const size_t kRingSize = 8192;
+  static constexpr size_t kRingSize = 8192;  // implicitly inline (C++17): no out-of-line def needed
   std::atomic<int64_t> ring_idx_{0};
   std::vector<std::atomic<uint32_t>> ring_;
   std::unique_ptr<WindowedInt> qps_window_;
