@@ -100,6 +100,7 @@ int parallel_echo_test(const std::vector<int>& ports, const std::string& payload
                        int fail_limit, std::string* merged, std::string* err);
 int selective_test(int dead_port, int live_port, std::string* resp_out);
 int partition_test(const std::vector<int>& ports, std::string* merged);
+int dynpart_test(const std::vector<int>& ports2, const std::vector<int>& ports3, int calls);
 int lb_spread_test(const std::string& lb_name, const std::vector<int>& ports, int ncalls);
 int compressed_echo_test(const std::string& addr, const std::string& payload,
                          int compress_type, std::string* response_out);
@@ -148,6 +149,8 @@ void bind_rpc_combo(py::module_& m) {
           }
           return py::make_tuple(rc, py::bytes(merged));
         });
+  r.def("dynamic_partition", &bam::rpctest::dynpart_test,
+        py::call_guard<py::gil_scoped_release>());
   r.def("lb_spread", &bam::rpctest::lb_spread_test,
         py::call_guard<py::gil_scoped_release>());
   r.def("backup_request", &bam::rpctest::backup_request_test,

@@ -1,5 +1,9 @@
 #include "rpc/parallel_channel.h"
 
+#include <map>
+
+#include "base/fast_rand.h"
+
 #include <atomic>
 
 #include "base/logging.h"
@@ -294,6 +298,73 @@ int PartitionChannel::Init(int num_partitions, const char* naming_url,
     parallel_.AddChannel(ch, /*owned=*/true);
   }
   return 0;
+}
+
+// ---------------- DynamicPartitionChannel ----------------
+
+DynamicPartitionChannel::~DynamicPartitionChannel() {
+  for (Scheme& sc : schemes_) delete sc.chan;
+}
+
+int DynamicPartitionChannel::Init(const char* naming_url, const PartitionChannelOptions* opt) {
+  // Group "i/N addr" entries by N; untagged entries are rejected here
+  // (they belong to every scheme in PartitionChannel, which is ambiguous
+  // across schemes).
+  std::string url = naming_url;
+  std::string body = url;
+  std::string prefix = "list://";
+  auto scheme_sep = url.find("://");
+  if (scheme_sep != std::string::npos) body = url.substr(scheme_sep + 3);
+  std::map<int, std::vector<std::string>> groups;  // N -> ["i/N addr", ...]
+  size_t pos = 0;
+  while (pos < body.size()) {
+    size_t comma = body.find(',', pos);
+    std::string item =
+        body.substr(pos, comma == std::string::npos ? comma : comma - pos);
+    pos = comma == std::string::npos ? body.size() : comma + 1;
+    if (item.empty()) continue;
+    size_t slash = item.find('/');
+    size_t space = item.find(' ');
+    if (slash == std::string::npos || space == std::string::npos || slash >= space) continue;
+    int nparts = atoi(item.substr(slash + 1, space - slash - 1).c_str());
+    if (nparts <= 0) continue;
+    groups[nparts].push_back(item);
+  }
+  if (groups.empty()) return -1;
+  for (auto& kv : groups) {
+    std::string sub_url = prefix;
+    for (size_t i = 0; i < kv.second.size(); ++i) {
+      if (i) sub_url += ",";
+      sub_url += kv.second[i];
+    }
+    PartitionChannel* pc = new PartitionChannel;
+    if (pc->Init(kv.first, sub_url.c_str(), opt) != 0) {
+      delete pc;  // scheme with an empty partition: skip it entirely
+      continue;
+    }
+    Scheme sc;
+    sc.num_partitions = kv.first;
+    sc.capacity = (int)kv.second.size();
+    sc.chan = pc;
+    schemes_.push_back(sc);
+    total_capacity_ += sc.capacity;
+  }
+  return schemes_.empty() ? -1 : 0;
+}
+
+void DynamicPartitionChannel::CallMethod(const std::string& full_method, Controller* cntl,
+                                         const IOBuf* request, IOBuf* response, Closure* done) {
+  // Capacity-weighted scheme pick (traffic splits by server counts, so a
+  // growing N-partition deployment takes over proportionally).
+  uint64_t r = fast_rand() % (uint64_t)total_capacity_;
+  for (Scheme& sc : schemes_) {
+    if (r < (uint64_t)sc.capacity) {
+      sc.chan->CallMethod(full_method, cntl, request, response, done);
+      return;
+    }
+    r -= sc.capacity;
+  }
+  schemes_.back().chan->CallMethod(full_method, cntl, request, response, done);
 }
 
 void PartitionChannel::CallMethod(const std::string& full_method, Controller* cntl,

@@ -108,4 +108,33 @@ class PartitionChannel : public ChannelBase {
 int ResolvePartitionedNaming(const std::string& url, int num_partitions,
                              std::vector<std::vector<EndPoint>>* partitions);
 
+// Discovers servers partitioned under DIFFERENT schemes from one naming
+// URL ("0/2 a,1/2 b,0/3 c,1/3 d,2/3 e" = a 2-partition group and a
+// 3-partition group), builds a PartitionChannel per scheme and splits
+// traffic between schemes by capacity (= number of servers in the
+// scheme). Purpose (parity: reference brpc/partition_channel.h
+// DynamicPartitionChannel): migrate a sharded service from M to N
+// partitions without touching client code. Delta: partition groups are
+// resolved at Init from the naming URL (static list naming), not
+// re-discovered live.
+class DynamicPartitionChannel : public ChannelBase {
+ public:
+  ~DynamicPartitionChannel();
+  int Init(const char* naming_url, const PartitionChannelOptions* opt);
+  size_t scheme_count() const { return schemes_.size(); }
+  int scheme_capacity(size_t i) const { return schemes_[i].capacity; }
+
+  void CallMethod(const std::string& full_method, Controller* cntl, const IOBuf* request,
+                  IOBuf* response, Closure* done) override;
+
+ private:
+  struct Scheme {
+    int num_partitions = 0;
+    int capacity = 0;  // number of servers carrying this scheme
+    PartitionChannel* chan = nullptr;
+  };
+  std::vector<Scheme> schemes_;
+  int total_capacity_ = 0;
+};
+
 }  // namespace bam

@@ -88,6 +88,39 @@ int partition_test(const std::vector<int>& ports, std::string* merged) {
   return 0;
 }
 
+// DynamicPartitionChannel: two co-existing schemes (ports2 = 2-partition
+// group, ports3 = 3-partition group). Runs `calls` echoes; returns the
+// number that were served by the 3-partition scheme (its merged response
+// has 3 fragments vs 2), so the caller can assert the 3/5-2/5 capacity
+// split. -1 on setup error, -2 on call failure.
+int dynpart_test(const std::vector<int>& ports2, const std::vector<int>& ports3, int calls) {
+  std::string url = "list://";
+  for (size_t i = 0; i < ports2.size(); ++i) {
+    url += std::to_string(i) + "/" + std::to_string(ports2.size()) + " 127.0.0.1:" +
+           std::to_string(ports2[i]) + ",";
+  }
+  for (size_t i = 0; i < ports3.size(); ++i) {
+    if (i) url += ",";
+    url += std::to_string(i) + "/" + std::to_string(ports3.size()) + " 127.0.0.1:" +
+           std::to_string(ports3[i]);
+  }
+  DynamicPartitionChannel dc;
+  PartitionChannelOptions opt;
+  opt.parallel.timeout_ms = 3000;
+  if (dc.Init(url.c_str(), &opt) != 0) return -1;
+  if (dc.scheme_count() != 2) return -1;
+  int hits3 = 0;
+  for (int i = 0; i < calls; ++i) {
+    Controller cntl;
+    IOBuf request, response;
+    request.append("D");
+    dc.CallMethod("EchoService.Echo", &cntl, &request, &response, nullptr);
+    if (cntl.Failed()) return -2;
+    if (response.size() == 3) ++hits3;  // each partition echoes 1 byte
+  }
+  return hits3;
+}
+
 // LB distribution check: run n calls via naming URL + lb; count distinct
 // servers hit (server returns its port via a Port method).
 int lb_spread_test(const std::string& lb_name, const std::vector<int>& ports, int ncalls) {

@@ -45,6 +45,17 @@ def test_partition_channel(three_ports):
     assert merged == b"P" * 3
 
 
+def test_dynamic_partition_channel(three_ports):
+    """DynamicPartitionChannel (reference partition_channel.h:127-169):
+    a 2-partition group and a 3-partition group co-exist; traffic splits
+    by capacity 2:3, every call fans out within its chosen scheme."""
+    ports2 = [r.start_echo_server(0) for _ in range(2)]
+    hits3 = c.dynamic_partition(ports2, three_ports, 200)
+    assert hits3 >= 0, hits3
+    # expected 3/5 of 200 = 120; allow generous binomial slack
+    assert 80 <= hits3 <= 160, hits3
+
+
 @pytest.mark.parametrize("lb", ["rr", "random", "p2c", "la", "wrr", "c_hash"])
 def test_lb_spreads_load(three_ports, lb):
     n = c.lb_spread(lb, three_ports, 60)
