@@ -14,6 +14,10 @@ struct BenchResult {
   std::string first_error;
 };
 int start_echo_server(int port);
+int start_nshead_server();
+int protocol_call(const std::string& addr, const std::string& protocol,
+                  const std::string& method, const std::string& payload, int compress,
+                  std::string* response_out, std::string* err);
 int echo_once(const std::string& addr, const std::string& payload, int timeout_ms,
               std::string* response_out, int64_t* latency_us);
 int call_method_once(const std::string& addr, const std::string& method,
@@ -30,6 +34,22 @@ void bind_rpc(py::module_& m) {
   auto r = m.def_submodule("rpc");
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
         py::call_guard<py::gil_scoped_release>());
+  r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("protocol_call",
+        [](const std::string& addr, const std::string& protocol, const std::string& method,
+           const std::string& payload, int compress) {
+          std::string resp, err;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::protocol_call(addr, protocol, method, payload, compress, &resp,
+                                             &err);
+          }
+          return py::make_tuple(rc, py::bytes(resp), err);
+        },
+        py::arg("addr"), py::arg("protocol"), py::arg("method"), py::arg("payload"),
+        py::arg("compress") = 0);
   r.def("echo_once",
         [](const std::string& addr, py::bytes payload, int timeout_ms) {
           char* ptr;

@@ -25,6 +25,9 @@ int Channel::Init(EndPoint ep, const ChannelOptions* options) {
   if (options_.protocol == "redis") policy::RegisterRedisProtocol();
   if (options_.protocol == "thrift") policy::RegisterThriftProtocol();
   if (options_.protocol == "grpc" || options_.protocol == "h2") policy::RegisterGrpcClientProtocol();
+  if (options_.protocol == "hulu_pbrpc") policy::RegisterHuluProtocol();
+  if (options_.protocol == "sofa_pbrpc") policy::RegisterSofaProtocol();
+  if (options_.protocol == "nshead") policy::RegisterNsheadProtocol();
   protocol_index_ = FindProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
   return 0;
@@ -164,7 +167,8 @@ void IssueRPC(Controller* cntl) {
     }
   }
   if (!sock) {
-    if (GetClientSocket(ep, &sock, cntl->call.connection_shard, cntl->call.ssl) != 0) {
+    if (GetClientSocket(ep, &sock, cntl->call.connection_shard, cntl->call.ssl,
+                        cntl->call.protocol_index) != 0) {
       // Conduct the failure through the session so retry/ending logic runs.
       session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
       return;

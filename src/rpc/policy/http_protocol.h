@@ -32,6 +32,11 @@ void RegisterHttpProtocol();
 void RegisterH2Protocol();  // h2 + gRPC (policy/h2_protocol.cc)
 void RegisterThriftProtocol();  // framed TBinary (policy/thrift_protocol.cc)
 void RegisterGrpcClientProtocol();  // h2/gRPC client (policy/h2_client.cc)
+// Legacy Baidu protocols (policy/legacy_protocols.cc): hulu_pbrpc,
+// sofa_pbrpc, nshead (raw-body service, FIFO-correlated).
+void RegisterHuluProtocol();
+void RegisterSofaProtocol();
+void RegisterNsheadProtocol();
 
 // Implemented in rpc/builtin/builtin_services.cc; returns true if the path
 // matched a builtin page.

@@ -77,6 +77,47 @@ int start_echo_server(int port) {
   return server->listen_address().port;
 }
 
+// Starts an echo server that ALSO serves nshead raw-body echo (body is
+// echoed back with "N:" prefixed). Returns port.
+int start_nshead_server() {
+  Server* server = new Server;
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  opts.nshead_handler = [](const IOBuf& req, IOBuf* resp) {
+    resp->append("N:");
+    resp->append(req);
+  };
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
+// One sync call over an arbitrary wire protocol (std/hulu_pbrpc/sofa_pbrpc/
+// nshead/...). Returns 0 on success.
+int protocol_call(const std::string& addr, const std::string& protocol,
+                  const std::string& method, const std::string& payload, int compress,
+                  std::string* response_out, std::string* err) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.protocol = protocol;
+  opts.timeout_ms = 3000;
+  opts.max_retry = 0;
+  if (channel.Init(addr.c_str(), &opts) != 0) {
+    if (err != nullptr) *err = "Init failed";
+    return -1;
+  }
+  Controller cntl;
+  cntl.set_request_compress_type((CompressType)compress);
+  IOBuf request, response;
+  request.append(payload);
+  channel.CallMethod(method, &cntl, &request, &response, nullptr);
+  if (cntl.Failed()) {
+    if (err != nullptr) *err = cntl.ErrorText();
+    return cntl.ErrorCode();
+  }
+  if (response_out != nullptr) *response_out = response.to_string();
+  return 0;
+}
+
 // One sync echo; returns 0 on success and fills latency_us, else error code.
 int echo_once(const std::string& addr, const std::string& payload, int timeout_ms,
               std::string* response_out, int64_t* latency_us) {

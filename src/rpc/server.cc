@@ -98,6 +98,9 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   policy::RegisterStdProtocol();
   policy::RegisterH2Protocol();
   policy::RegisterThriftProtocol();
+  policy::RegisterHuluProtocol();
+  policy::RegisterSofaProtocol();
+  if (options_.nshead_handler) policy::RegisterNsheadProtocol();
   if (options_.redis_service != nullptr) policy::RegisterRedisProtocol();
   int listen_fd = tcp_listen(ep);
   if (listen_fd < 0) {

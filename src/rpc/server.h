@@ -62,6 +62,9 @@ struct ServerOptions {
   // set => every accepted connection speaks TLS (rpc/ssl_util.h).
   std::string ssl_cert;
   std::string ssl_key;
+  // nshead raw-body service (parity: reference ServerOptions::nshead_service):
+  // called with the request body; fills the response body.
+  std::function<void(const IOBuf& req_body, IOBuf* resp_body)> nshead_handler;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)

@@ -125,15 +125,21 @@ bool IsEndpointIsolated(const EndPoint& ep) {
   return h != nullptr && h->isolated.load(std::memory_order_acquire);
 }
 
-int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ssl) {
+int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ssl,
+                    int protocol_index) {
   if (IsEndpointIsolated(ep)) {
     errno = EHOSTDOWN;
     return -1;
   }
   ClientSocketMap& m = the_map();
-  // TLS connections never share plaintext sockets: fold the flag into the
-  // map key (shards are small, bit 30 is free).
-  const auto key = std::make_pair(ep, shard | (ssl ? (1 << 30) : 0));
+  // Key = (endpoint, shard, ssl, protocol) folded into one int (parity:
+  // the reference keys client sockets by ChannelSignature). Distinct wire
+  // protocols must not share a connection — a response smaller than
+  // another protocol's minimum header would starve behind the socket's
+  // preferred-protocol parse. Shards are small ints: bits 20-27 carry the
+  // protocol index, bit 30 carries TLS.
+  const int proto_bits = protocol_index >= 0 ? ((protocol_index + 1) & 0xff) << 20 : 0;
+  const auto key = std::make_pair(ep, shard | proto_bits | (ssl ? (1 << 30) : 0));
   {
     std::lock_guard<std::mutex> lk(m.mu);
     auto it = m.sockets.find(key);

@@ -1,0 +1,89 @@
+"""Legacy Baidu wire protocols (reference policy/hulu_pbrpc_protocol.cpp,
+sofa_pbrpc_protocol.cpp, nshead_protocol.cpp): hulu 12-byte LE header,
+sofa 24-byte LE header, nshead 36-byte struct with FIFO correlation.
+One server port speaks std + hulu + sofa (+ nshead where enabled)
+simultaneously — protocol sniffing by magic, like the reference."""
+import struct
+
+import brpc_amd as b
+import pytest
+
+r = b.core.rpc
+
+
+@pytest.fixture(scope="module")
+def port():
+    return r.start_nshead_server()  # echo service + nshead raw echo
+
+
+def test_hulu_echo(port):
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "hulu_pbrpc",
+                                    "EchoService.Echo", b"hulu-hi")
+    assert rc == 0, err
+    assert resp == b"hulu-hi"
+
+
+def test_hulu_unknown_method(port):
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "hulu_pbrpc",
+                                    "EchoService.Nope", b"x")
+    assert rc != 0
+    assert "unknown method" in err
+
+
+def test_hulu_snappy_compress(port):
+    payload = b"z" * 5000
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "hulu_pbrpc",
+                                    "EchoService.Echo", payload, compress=1)
+    assert rc == 0, err
+    assert resp == payload
+
+
+def test_sofa_echo(port):
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "sofa_pbrpc",
+                                    "EchoService.Echo", b"sofa-hi")
+    assert rc == 0, err
+    assert resp == b"sofa-hi"
+
+
+def test_sofa_error_propagates(port):
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "sofa_pbrpc",
+                                    "EchoService.Fail", b"x")
+    assert rc != 0
+    assert "asked for it" in err
+
+
+def test_nshead_raw_echo(port):
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "nshead", "ignored",
+                                    b"raw-body")
+    assert rc == 0, err
+    assert resp == b"N:raw-body"
+
+
+def test_nshead_wire_format_by_hand(port):
+    """Speak nshead with a raw socket: 36-byte LE header, magic 0xfb709394."""
+    import socket
+    body = b"manual"
+    head = struct.pack("<HHI16sIII", 7, 1, 42, b"pytest".ljust(16, b"\0"),
+                       0xFB709394, 0, len(body))
+    s = socket.create_connection(("127.0.0.1", port), timeout=5)
+    s.sendall(head + body)
+    resp_head = b""
+    while len(resp_head) < 36:
+        chunk = s.recv(36 - len(resp_head))
+        assert chunk
+        resp_head += chunk
+    rid, ver, log_id, prov, magic, _res, blen = struct.unpack("<HHI16sIII", resp_head)
+    assert magic == 0xFB709394
+    assert (rid, ver, log_id) == (7, 1, 42)  # echoed back
+    got = b""
+    while len(got) < blen:
+        got += s.recv(blen - len(got))
+    assert got == b"N:manual"
+    s.close()
+
+
+def test_std_still_works_on_same_port(port):
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "std",
+                                    "EchoService.Echo", b"std-hi")
+    assert rc == 0, err
+    assert resp == b"std-hi"
