@@ -25,10 +25,11 @@ int Channel::Init(EndPoint ep, const ChannelOptions* options) {
   if (options_.protocol == "redis") policy::RegisterRedisProtocol();
   if (options_.protocol == "thrift") policy::RegisterThriftProtocol();
   if (options_.protocol == "grpc" || options_.protocol == "h2") policy::RegisterGrpcClientProtocol();
+  if (options_.protocol == "http") policy::RegisterHttpProtocol();
   if (options_.protocol == "hulu_pbrpc") policy::RegisterHuluProtocol();
   if (options_.protocol == "sofa_pbrpc") policy::RegisterSofaProtocol();
   if (options_.protocol == "nshead") policy::RegisterNsheadProtocol();
-  protocol_index_ = FindProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
+  protocol_index_ = FindClientProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
   return 0;
 }
@@ -46,7 +47,7 @@ int Channel::Init(const char* naming_url, const char* lb_name, const ChannelOpti
   if (options != nullptr) options_ = *options;
   single_server_ = false;
   policy::RegisterStdProtocol();
-  protocol_index_ = FindProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
+  protocol_index_ = FindClientProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
   lb_ = LoadBalancerWithNaming::Create(naming_url, lb_name);
   if (lb_ == nullptr) return -1;

@@ -204,7 +204,7 @@ ssize_t c_body_read(nghttp2_session*, int32_t, uint8_t* buf, size_t length,
 
 // ---- issue + parse hooks ----
 
-int IssueGrpcRequest(Socket* sock, Controller* cntl, uint64_t cid) {
+int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
   if (!ngc().ok) return -1;
   H2ClientCtx* ctx = (H2ClientCtx*)sock->protocol_ctx;
   if (ctx == nullptr) {
@@ -221,34 +221,47 @@ int IssueGrpcRequest(Socket* sock, Controller* cntl, uint64_t cid) {
     sock->protocol_ctx = ctx;
     sock->protocol_ctx_deleter = [](void* p) { h2c_unref((H2ClientCtx*)p); };
   }
-  // gRPC frame the payload.
   std::string payload = cntl->call.request_buf.to_string();
   H2ClientBody* body = new H2ClientBody;
-  char frame[5];
-  frame[0] = 0;
-  frame[1] = (char)(payload.size() >> 24);
-  frame[2] = (char)(payload.size() >> 16);
-  frame[3] = (char)(payload.size() >> 8);
-  frame[4] = (char)payload.size();
-  body->data.assign(frame, 5);
-  body->data += payload;
+  if (grpc) {
+    // gRPC frame the payload (5-byte length prefix).
+    char frame[5];
+    frame[0] = 0;
+    frame[1] = (char)(payload.size() >> 24);
+    frame[2] = (char)(payload.size() >> 16);
+    frame[3] = (char)(payload.size() >> 8);
+    frame[4] = (char)payload.size();
+    body->data.assign(frame, 5);
+    body->data += payload;
+  } else {
+    body->data = payload;
+  }
 
-  std::string path = "/" + cntl->call.service_name + "/" + cntl->call.method_name;
+  const std::string& svc = cntl->call.service_name;
+  const std::string& mn = cntl->call.method_name;
+  std::string path = !mn.empty() && mn[0] == '/' ? mn
+                     : svc.empty() && !grpc      ? "/" + mn
+                                                 : "/" + svc + "/" + mn;
   std::string authority = endpoint2str(cntl->remote_side());
   // NOTE: every value must outlive submit_request — named locals, never
   // temporaries (nghttp2_nv holds raw pointers).
-  std::string v_method = "POST", v_scheme = "http", v_ct = "application/grpc",
-              v_te = "trailers", v_ua = "brpc-amd-grpc/1.0";
+  std::string v_method = grpc || !payload.empty() ? "POST" : "GET", v_scheme = "http",
+              v_ct = grpc ? "application/grpc" : "application/octet-stream",
+              v_te = "trailers", v_ua = grpc ? "brpc-amd-grpc/1.0" : "brpc-amd-h2/1.0";
   nghttp2_nv nvs[7] = {
       cnv(":method", v_method), cnv(":scheme", v_scheme), cnv(":path", path),
       cnv(":authority", authority), cnv("content-type", v_ct), cnv("te", v_te),
       cnv("user-agent", v_ua),
   };
+  // plain h2 omits "te: trailers" (gRPC-specific).
+  const size_t nnv = grpc ? 7 : 6;
+  if (!grpc) nvs[5] = nvs[6];  // drop te, keep user-agent
   nghttp2_data_provider prd;
   prd.source.ptr = body;
   prd.read_callback = c_body_read;
   std::lock_guard<std::mutex> lk(ctx->mu);
-  int32_t stream_id = ngc().submit_request(ctx->session, nullptr, nvs, 7, &prd, nullptr);
+  int32_t stream_id =
+      ngc().submit_request(ctx->session, nullptr, nvs, nnv, &prd, nullptr);
   if (stream_id < 0) {
     delete body;
     return -1;
@@ -290,11 +303,24 @@ void RegisterGrpcClientProtocol() {
     }
     Protocol p;
     p.parse = ParseGrpcClient;
-    p.issue_request = IssueGrpcRequest;
+    p.issue_request = [](Socket* s, Controller* c, uint64_t cid) {
+      return IssueH2Request(s, c, cid, /*grpc=*/true);
+    };
     p.support_server = false;
     p.support_client = true;
     p.name = "grpc";
     RegisterProtocol(p);
+    // Plain HTTP-semantics h2 client (parity: reference protocol "h2"):
+    // same nghttp2 session machinery, no gRPC framing or trailers.
+    Protocol h2;
+    h2.parse = ParseGrpcClient;
+    h2.issue_request = [](Socket* s, Controller* c, uint64_t cid) {
+      return IssueH2Request(s, c, cid, /*grpc=*/false);
+    };
+    h2.support_server = false;
+    h2.support_client = true;
+    h2.name = "h2";
+    RegisterProtocol(h2);
   });
 }
 

@@ -20,6 +20,9 @@
 #include "rpc/wire.h"
 
 namespace bam {
+
+void EndRPC(Controller* cntl, SessionId locked_id);  // channel.cc
+
 namespace policy {
 
 namespace {
@@ -84,7 +87,142 @@ bool parse_headers(const std::string& head, HttpRequest* out) {
   return true;
 }
 
-ParseResult ParseHttpMessage(IOBuf* source, Socket* /*sock*/, bool /*eof*/) {
+// ---- client side (HTTP/1.1 responses, FIFO-correlated) ----
+// Parity: reference http_rpc_protocol.cpp client half. Supports
+// Content-Length and chunked transfer-coding.
+
+struct HttpClientMessage : public InputMessageBase {
+  int status = 0;
+  std::string reason;
+  std::map<std::string, std::string> headers;
+  IOBuf body;
+};
+
+ParseResult ParseHttpResponse(IOBuf* source, Socket* /*sock*/, bool /*eof*/) {
+  char probe[8];
+  size_t n = std::min<size_t>(source->size(), 8);
+  if (n < 5) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  const char* p = (const char*)source->fetch(probe, n);
+  if (memcmp(p, "HTTP/", 5) != 0) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  std::string head;
+  size_t scan = std::min<size_t>(source->size(), 64 << 10);
+  source->copy_to(&head, scan, 0);
+  size_t hend = find_header_end(head);
+  if (hend == std::string::npos) {
+    if (scan >= (64 << 10)) return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+    return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  }
+  // status line: HTTP/1.1 200 OK
+  size_t sp1 = head.find(' ');
+  size_t eol = head.find("\r\n");
+  if (sp1 == std::string::npos || eol == std::string::npos || sp1 > eol)
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  int status = atoi(head.c_str() + sp1 + 1);
+  std::map<std::string, std::string> headers;
+  size_t line = eol + 2;
+  while (line + 2 <= hend) {
+    size_t le = head.find("\r\n", line);
+    if (le == std::string::npos || le >= hend) break;
+    std::string l = head.substr(line, le - line);
+    line = le + 2;
+    if (l.empty()) break;
+    size_t c = l.find(':');
+    if (c == std::string::npos) continue;
+    std::string k = l.substr(0, c);
+    for (char& ch : k) ch = (char)tolower((unsigned char)ch);
+    size_t v0 = l.find_first_not_of(' ', c + 1);
+    headers[k] = v0 == std::string::npos ? "" : l.substr(v0);
+  }
+  IOBuf body;
+  size_t consumed = hend;
+  auto te = headers.find("transfer-encoding");
+  if (te != headers.end() && te->second.find("chunked") != std::string::npos) {
+    // Decode chunked coding; needs the terminating 0-chunk in the buffer.
+    std::string all;
+    source->copy_to(&all, (size_t)-1, 0);
+    size_t pos = hend;
+    for (;;) {
+      size_t le = all.find("\r\n", pos);
+      if (le == std::string::npos) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+      size_t chunk_len = strtoul(all.c_str() + pos, nullptr, 16);
+      pos = le + 2;
+      if (chunk_len == 0) {
+        // trailer section ends with CRLF
+        size_t fin = all.find("\r\n", pos);
+        if (fin == std::string::npos) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+        consumed = fin + 2;
+        break;
+      }
+      if (all.size() < pos + chunk_len + 2)
+        return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+      body.append(all.data() + pos, chunk_len);
+      pos += chunk_len + 2;
+    }
+    source->pop_front(consumed);
+  } else {
+    size_t content_len = 0;
+    auto cl = headers.find("content-length");
+    if (cl != headers.end()) content_len = (size_t)atoll(cl->second.c_str());
+    if (source->size() < hend + content_len)
+      return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+    source->pop_front(hend);
+    source->cutn(&body, content_len);
+  }
+  HttpClientMessage* msg = new HttpClientMessage;
+  msg->status = status;
+  msg->headers.swap(headers);
+  msg->body.swap(body);
+  return ParseResult::make_ok(msg);
+}
+
+void ProcessHttpResponse(InputMessageBase* mb) {
+  HttpClientMessage* msg = (HttpClientMessage*)mb;
+  SocketUniquePtr sock;
+  SessionId cid = 0;
+  if (Socket::Address(msg->socket_id, &sock) == 0) cid = sock->pop_pipeline();
+  void* data = nullptr;
+  if (cid == 0 || session_lock(cid, &data) != 0) {
+    delete msg;
+    return;
+  }
+  Controller* cntl = (Controller*)data;
+  sock->remove_pending_session(cid);
+  if (msg->status != 200) {
+    cntl->SetFailed(EHTTP, "HTTP status " + std::to_string(msg->status) + ": " +
+                               msg->body.to_string().substr(0, 200));
+  } else if (cntl->call.response != nullptr) {
+    cntl->call.response->clear();
+    cntl->call.response->swap(msg->body);
+  }
+  delete msg;
+  EndRPC(cntl, cid);
+}
+
+void PackHttp1Request(IOBuf* out, Controller* cntl, uint64_t /*fifo-correlated*/) {
+  const std::string& svc = cntl->call.service_name;
+  const std::string& m = cntl->call.method_name;
+  std::string path = !m.empty() && m[0] == '/' ? m
+                     : svc.empty()             ? "/" + m
+                                               : "/" + svc + "/" + m;
+  const IOBuf& body = cntl->call.request_buf;
+  std::string head;
+  head.reserve(256);
+  head += (body.empty() ? "GET " : "POST ") + path + " HTTP/1.1\r\n";
+  head += "Host: " + endpoint2str(cntl->remote_side()) + "\r\n";
+  head += "User-Agent: brpc-amd/1.0\r\n";
+  head += "Accept: */*\r\n";
+  if (!body.empty()) {
+    head += "Content-Type: application/octet-stream\r\n";
+    head += "Content-Length: " + std::to_string(body.size()) + "\r\n";
+  }
+  head += "\r\n";
+  out->append(head);
+  out->append(body);
+}
+
+ParseResult ParseHttpMessage(IOBuf* source, Socket* sock, bool eof) {
+  // Client-side sockets (no owning server) carry responses.
+  if (sock->user() == nullptr) return ParseHttpResponse(source, sock, eof);
   // Cheap probe for an HTTP method prefix.
   char probe[8];
   size_t n = std::min<size_t>(source->size(), 8);
@@ -215,9 +353,11 @@ void RegisterHttpProtocol() {
     Protocol p;
     p.parse = ParseHttpMessage;
     p.process_request = ProcessHttpRequest;
-    p.process_response = nullptr;  // client-side HTTP: later phase
+    p.process_response = ProcessHttpResponse;
+    p.pack_request = PackHttp1Request;
+    p.client_pipelined = true;  // HTTP/1.1 responses match requests FIFO
     p.support_server = true;
-    p.support_client = false;
+    p.support_client = true;
     p.name = "http";
     RegisterProtocol(p);
   });

@@ -31,6 +31,21 @@ int FindProtocolIndex(const std::string& name) {
   return -1;
 }
 
+int FindClientProtocolIndex(const std::string& name) {
+  // A name can be registered twice (e.g. "h2": server parse entry AND the
+  // nghttp2 client entry). Channels must bind to one that can issue
+  // client requests.
+  int n = g_nprotocols.load(std::memory_order_acquire);
+  int any = -1;
+  for (int i = 0; i < n; ++i) {
+    if (name == g_protocols[i].name) {
+      if (g_protocols[i].support_client) return i;
+      if (any < 0) any = i;
+    }
+  }
+  return any;
+}
+
 int ProtocolCount() { return g_nprotocols.load(std::memory_order_acquire); }
 
 }  // namespace bam

@@ -68,6 +68,8 @@ static const int kMaxProtocols = 16;
 int RegisterProtocol(const Protocol& p);
 const Protocol* GetProtocol(int index);
 int FindProtocolIndex(const std::string& name);
+// Prefers an entry with support_client (names can be double-registered).
+int FindClientProtocolIndex(const std::string& name);
 int ProtocolCount();
 
 }  // namespace bam
