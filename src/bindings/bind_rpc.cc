@@ -217,6 +217,7 @@ namespace rpctest {
 int start_stream_echo_server();
 int stream_echo_test(int port, int nframes, int frame_size, std::string* err);
 double stream_throughput_test(int port, int nframes, int frame_size);
+double stream_throughput_hbm_test(int port, int nframes, int frame_size, bool sink_to_hbm);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -234,6 +235,8 @@ void bind_rpc_stream(py::module_& m) {
           }
           return py::make_tuple(rc, err);
         });
+  s.def("throughput_hbm", &bam::rpctest::stream_throughput_hbm_test,
+        py::call_guard<py::gil_scoped_release>());
   s.def("throughput", &bam::rpctest::stream_throughput_test,
         py::call_guard<py::gil_scoped_release>());
 }

@@ -75,3 +75,14 @@ def test_hbm_echo_bench_smoke():
 def test_gpu_snappy_cross_check(n, mode):
     """GPU-compressed streams decompress with the host codec and vice versa."""
     assert g.snappy_cross_check(n, mode, 0)
+
+
+def test_stream_hbm_throughput():
+    """BASELINE config 3 (single-GPU analogue): streaming RPC with 1 MB
+    frames resident in HBM; write path stages D2H via the direct-gather
+    kernel, OpenSinkHbm re-uploads every frame into HBM on the server."""
+    port = b.core.stream.start_server()
+    mbps = b.core.stream.throughput_hbm(port, 64, 1 << 20, False)
+    assert mbps > 100, mbps  # staging must not collapse below 0.1 GB/s
+    mbps2 = b.core.stream.throughput_hbm(port, 64, 1 << 20, True)
+    assert mbps2 > 50, mbps2
