@@ -15,6 +15,11 @@ struct BenchResult {
 };
 int start_echo_server(int port);
 int start_nshead_server();
+int start_shm_echo_server(const std::string& name);
+int shm_call(const std::string& name, const std::string& method, const std::string& payload,
+             std::string* response_out, std::string* err);
+int shm_echo_bench(const std::string& name, int payload, int concurrency, int calls,
+                   double* qps, int64_t* p99_us, int* errors_out);
 int protocol_call(const std::string& addr, const std::string& protocol,
                   const std::string& method, const std::string& payload, int compress,
                   std::string* response_out, std::string* err);
@@ -36,6 +41,36 @@ void bind_rpc(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
         py::call_guard<py::gil_scoped_release>());
+  r.def("start_shm_server", &bam::rpctest::start_shm_echo_server,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("shm_call",
+        [](const std::string& name, const std::string& method, const std::string& payload) {
+          std::string resp, err;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::shm_call(name, method, payload, &resp, &err);
+          }
+          return py::make_tuple(rc, py::bytes(resp), err);
+        });
+  r.def("shm_echo_bench",
+        [](const std::string& name, int payload, int concurrency, int calls) {
+          double qps = 0;
+          int64_t p99 = 0;
+          int errors = 0;
+          int rc;
+          {
+            py::gil_scoped_release rel;
+            rc = bam::rpctest::shm_echo_bench(name, payload, concurrency, calls, &qps, &p99,
+                                              &errors);
+          }
+          py::dict d;
+          d["rc"] = rc;
+          d["qps"] = qps;
+          d["p99_us"] = p99;
+          d["errors"] = errors;
+          return d;
+        });
   r.def("protocol_call",
         [](const std::string& addr, const std::string& protocol, const std::string& method,
            const std::string& payload, int compress) {

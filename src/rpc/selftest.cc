@@ -15,6 +15,7 @@
 #include "rpc/channel.h"
 #include "rpc/controller.h"
 #include "rpc/server.h"
+#include "rpc/shm_ring.h"
 
 namespace bam {
 namespace rpctest {
@@ -115,6 +116,77 @@ int protocol_call(const std::string& addr, const std::string& protocol,
     return cntl.ErrorCode();
   }
   if (response_out != nullptr) *response_out = response.to_string();
+  return 0;
+}
+
+// ---- shm ring transport (UBRing analogue, rpc/shm_ring.h) ----
+
+int start_shm_echo_server(const std::string& name) {
+  Server* server = new Server;  // never bound to TCP: shm-only dispatch
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  return shm::ServeShm(name, server);
+}
+
+int shm_call(const std::string& name, const std::string& method, const std::string& payload,
+             std::string* response_out, std::string* err) {
+  shm::ShmChannel ch;
+  if (ch.Init(name) != 0) {
+    *err = "shm connect failed";
+    return -1;
+  }
+  IOBuf req, resp;
+  req.append(payload);
+  int rc = ch.Call(method, req, &resp, 3000000, err);
+  if (rc == 0 && response_out != nullptr) *response_out = resp.to_string();
+  return rc;
+}
+
+// Throughput/latency: `concurrency` fibers each issue calls/concurrency
+// sync echoes over ONE shm connection. Returns {qps, p99_us, errors}.
+namespace {
+struct ShmBenchArg {
+  shm::ShmChannel* ch;
+  std::string payload;
+  int calls;
+  std::atomic<int>* errors;
+  std::vector<int64_t>* lat;  // per-fiber slice, preallocated
+  size_t lat_off;
+};
+void shm_bench_fiber(void* raw) {
+  ShmBenchArg* a = (ShmBenchArg*)raw;
+  IOBuf req;
+  req.append(a->payload);
+  for (int i = 0; i < a->calls; ++i) {
+    IOBuf resp;
+    int64_t t0 = monotonic_time_us();
+    int rc = a->ch->Call("EchoService.Echo", req, &resp, 5000000, nullptr);
+    (*a->lat)[a->lat_off + i] = monotonic_time_us() - t0;
+    if (rc != 0 || resp.size() != a->payload.size()) a->errors->fetch_add(1);
+  }
+}
+}  // namespace
+
+int shm_echo_bench(const std::string& name, int payload, int concurrency, int calls,
+                   double* qps, int64_t* p99_us, int* errors_out) {
+  shm::ShmChannel ch;
+  if (ch.Init(name) != 0) return -1;
+  std::string pay((size_t)payload, 'u');
+  int per = calls / concurrency;
+  std::vector<int64_t> lat((size_t)per * concurrency, 0);
+  std::atomic<int> errors{0};
+  std::vector<ShmBenchArg> args((size_t)concurrency);
+  std::vector<fiber_t> tids((size_t)concurrency);
+  int64_t t0 = monotonic_time_us();
+  for (int i = 0; i < concurrency; ++i) {
+    args[i] = ShmBenchArg{&ch, pay, per, &errors, &lat, (size_t)i * per};
+    fiber_start_background(&tids[i], shm_bench_fiber, &args[i]);
+  }
+  for (int i = 0; i < concurrency; ++i) fiber_join(tids[i]);
+  int64_t elapsed = monotonic_time_us() - t0;
+  std::sort(lat.begin(), lat.end());
+  *qps = (double)lat.size() * 1e6 / (double)elapsed;
+  *p99_us = lat[(size_t)((double)lat.size() * 0.99)];
+  *errors_out = errors.load();
   return 0;
 }
 

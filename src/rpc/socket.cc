@@ -323,7 +323,14 @@ ssize_t Socket::write_bytes(IOBuf* data, bool may_block) {
   char tmp[16384];
   size_t n = data->copy_to(tmp, sizeof(tmp), 0);
   if (n == 0) return 0;
-  ssize_t rc = ssl::Write(ssl_, tmp, n);
+  ssize_t rc;
+  {
+    // One SSL object is NOT safe under concurrent SSL_read/SSL_write
+    // (TLS1.3 session tickets arrive exactly when the first writes go
+    // out); ssl_hs_mu_ serializes every SSL operation on this socket.
+    std::lock_guard<std::mutex> lk(ssl_hs_mu_);
+    rc = ssl::Write(ssl_, tmp, n);
+  }
   if (rc > 0) data->pop_front((size_t)rc);
   return rc;
 }
@@ -332,7 +339,11 @@ ssize_t Socket::read_bytes(IOBuf* out, size_t max) {
   if (ssl_ == nullptr) return out->append_from_file_descriptor(fd(), max);
   char tmp[16384];
   size_t want = max < sizeof(tmp) ? max : sizeof(tmp);
-  ssize_t rc = ssl::Read(ssl_, tmp, want);
+  ssize_t rc;
+  {
+    std::lock_guard<std::mutex> lk(ssl_hs_mu_);  // see write_bytes
+    rc = ssl::Read(ssl_, tmp, want);
+  }
   if (rc > 0) out->append(tmp, (size_t)rc);
   return rc;
 }
