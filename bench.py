@@ -71,11 +71,14 @@ def main():
     pooled = not use_gpu
 
     def run_step():
-        res = b.core.rpc.echo_bench(addr, args.payload, args.concurrency,
-                                    args.calls_per_step, 30000, method, hbm_req, pooled)
-        if res["errors"]:
-            raise RuntimeError("bench errors: n=%s first=%s" % (res["errors"], res.get("first_error")))
-        return res
+        last = None
+        for attempt in range(2):  # one retry: a loaded box can time out a stray call
+            res = b.core.rpc.echo_bench(addr, args.payload, args.concurrency,
+                                        args.calls_per_step, 30000, method, hbm_req, pooled)
+            if not res["errors"]:
+                return res
+            last = res
+        raise RuntimeError("bench errors: n=%s first=%s" % (last["errors"], last.get("first_error")))
 
     for _ in range(args.warmup):
         run_step()

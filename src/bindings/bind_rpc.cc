@@ -15,6 +15,7 @@ struct BenchResult {
 };
 int start_echo_server(int port);
 int start_nshead_server();
+int start_mongo_echo_server();
 int start_shm_echo_server(const std::string& name);
 int shm_call(const std::string& name, const std::string& method, const std::string& payload,
              std::string* response_out, std::string* err);
@@ -40,6 +41,8 @@ void bind_rpc(py::module_& m) {
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("start_mongo_server", &bam::rpctest::start_mongo_echo_server,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_shm_server", &bam::rpctest::start_shm_echo_server,
         py::call_guard<py::gil_scoped_release>());

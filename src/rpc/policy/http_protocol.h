@@ -37,6 +37,8 @@ void RegisterGrpcClientProtocol();  // h2/gRPC client (policy/h2_client.cc)
 void RegisterHuluProtocol();
 void RegisterSofaProtocol();
 void RegisterNsheadProtocol();
+void RegisterMongoProtocol();  // server-side (policy/mongo_protocol.cc)
+void RegisterEspProtocol();    // client-side (policy/legacy_protocols.cc)
 
 // Implemented in rpc/builtin/builtin_services.cc; returns true if the path
 // matched a builtin page.

@@ -557,6 +557,71 @@ void PackNsheadRequest(IOBuf* out, Controller* cntl, uint64_t /*fifo-correlated*
   pack_nshead(out, 0, 0, (uint32_t)cntl->log_id(), nullptr, cntl->call.request_buf);
 }
 
+// ==================== esp (client) ====================
+// Parity: reference policy/esp_protocol.cpp + esp_head.h — 32-byte packed
+// little-endian head {from u64, to u64, msg u32, msg_id u64, body_len i32}
+// followed by a raw body. The head has NO magic, so the parse is gated to
+// sockets created for the esp protocol (client_protocol_hint). Responses
+// match requests FIFO per connection (the reference allows one pending
+// call per pooled connection; FIFO is the superset of that).
+
+#pragma pack(push, 1)
+struct EspHead {
+  uint64_t from = 0;
+  uint64_t to = 0;
+  uint32_t msg = 0;
+  uint64_t msg_id = 0;
+  int32_t body_len = 0;
+};
+#pragma pack(pop)
+static_assert(sizeof(EspHead) == 32, "esp head is 32 packed bytes");
+
+int g_esp_protocol_index = -1;
+
+struct EspMsg : public InputMessageBase {
+  EspHead head;
+  IOBuf body;
+};
+
+ParseResult ParseEsp(IOBuf* source, Socket* sock, bool) {
+  if (sock->client_protocol_hint != g_esp_protocol_index)
+    return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  char aux[sizeof(EspHead)];
+  if (source->size() < sizeof(EspHead))
+    return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  const char* h = (const char*)source->fetch(aux, sizeof(EspHead));
+  EspHead head;
+  memcpy(&head, h, sizeof(head));
+  if (head.body_len < 0 || head.body_len > (64 << 20))
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  if (source->size() < sizeof(EspHead) + (size_t)head.body_len)
+    return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+  EspMsg* msg = new EspMsg;
+  msg->head = head;
+  source->pop_front(sizeof(EspHead));
+  source->cutn(&msg->body, (size_t)head.body_len);
+  return ParseResult::make_ok(msg);
+}
+
+void ProcessEspResponse(InputMessageBase* mb) {
+  EspMsg* msg = (EspMsg*)mb;
+  SocketUniquePtr sock;
+  SessionId cid = 0;
+  if (Socket::Address(msg->socket_id, &sock) == 0) cid = sock->pop_pipeline();
+  if (cid != 0) finish_client_call(cid, msg->socket_id, 0, "", &msg->body, 0);
+  delete msg;
+}
+
+void PackEspRequest(IOBuf* out, Controller* cntl, uint64_t correlation_id) {
+  EspHead head;
+  // method name carries the numeric esp `msg` selector
+  head.msg = (uint32_t)strtoul(cntl->call.method_name.c_str(), nullptr, 10);
+  head.msg_id = correlation_id;
+  head.body_len = (int32_t)cntl->call.request_buf.size();
+  out->append(&head, sizeof(head));
+  out->append(cntl->call.request_buf);
+}
+
 }  // namespace
 
 void RegisterHuluProtocol() {
@@ -586,6 +651,21 @@ void RegisterSofaProtocol() {
     p.support_client = true;
     p.name = "sofa_pbrpc";
     RegisterProtocol(p);
+  });
+}
+
+void RegisterEspProtocol() {
+  static std::once_flag once;
+  std::call_once(once, [] {
+    Protocol p;
+    p.parse = ParseEsp;
+    p.process_response = ProcessEspResponse;
+    p.pack_request = PackEspRequest;
+    p.client_pipelined = true;
+    p.support_server = false;
+    p.support_client = true;
+    p.name = "esp";
+    g_esp_protocol_index = RegisterProtocol(p);
   });
 }
 

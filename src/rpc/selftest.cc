@@ -119,6 +119,42 @@ int protocol_call(const std::string& addr, const std::string& protocol,
   return 0;
 }
 
+// ---- mongo server (adaptor parity) ----
+// Replies {ok: 1.0} to OP_QUERY ("ismaster"-style) and OP_MSG; records
+// fire-and-forget inserts into a counter readable via OP_QUERY on
+// "insert_count" (session-free demo of the adaptor contract).
+int start_mongo_echo_server() {
+  Server* server = new Server;
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  static std::atomic<int> g_inserts{0};
+  opts.mongo_handler = [](const MongoHeader& head, const IOBuf& body, MongoReply* reply) {
+    if (head.op_code == 2002) {  // OP_INSERT: fire-and-forget
+      g_inserts.fetch_add(1, std::memory_order_relaxed);
+      return;
+    }
+    // minimal BSON {"ok": 1.0, "n": <inserts>} (double 0x01, int32 0x10)
+    std::string doc;
+    auto i32 = [&](int32_t v) { doc.append((const char*)&v, 4); };
+    double one = 1.0;
+    std::string elems;
+    elems.push_back('\x01');
+    elems.append("ok", 3);
+    elems.append((const char*)&one, 8);
+    elems.push_back('\x10');
+    elems.append("n", 2);
+    int32_t n = g_inserts.load(std::memory_order_relaxed);
+    elems.append((const char*)&n, 4);
+    i32((int32_t)(4 + elems.size() + 1));
+    doc += elems;
+    doc.push_back('\0');
+    if (head.op_code == 2013) reply->body.append("\0", 1);  // OP_MSG section kind 0
+    reply->body.append(doc);
+  };
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
 // ---- shm ring transport (UBRing analogue, rpc/shm_ring.h) ----
 
 int start_shm_echo_server(const std::string& name) {

@@ -101,6 +101,7 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   policy::RegisterHuluProtocol();
   policy::RegisterSofaProtocol();
   if (options_.nshead_handler) policy::RegisterNsheadProtocol();
+  if (options_.mongo_handler) policy::RegisterMongoProtocol();
   if (options_.redis_service != nullptr) policy::RegisterRedisProtocol();
   int listen_fd = tcp_listen(ep);
   if (listen_fd < 0) {

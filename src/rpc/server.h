@@ -44,6 +44,25 @@ enum ServiceOwnership { SERVER_OWNS_SERVICE, SERVER_DOESNT_OWN_SERVICE };
 
 class RedisService;
 
+// Mongo wire support (parity: reference mongo_service_adaptor.h — the
+// framework frames the 16-byte mongo header and hands the RAW body to the
+// user handler; BSON interpretation is the user's business).
+struct MongoHeader {
+  int32_t message_length = 0;
+  int32_t request_id = 0;
+  int32_t response_to = 0;
+  int32_t op_code = 0;
+};
+struct MongoReply {
+  IOBuf body;               // reply documents / OP_MSG sections (raw bytes)
+  int32_t response_flags = 0;
+  int64_t cursor_id = 0;
+  int32_t number_returned = 1;
+  bool always_reply = false;  // reply even with an empty body
+};
+typedef std::function<void(const MongoHeader& head, const IOBuf& body, MongoReply* reply)>
+    MongoHandlerFn;
+
 // Runs before every request handler (parity: reference brpc/interceptor.h).
 // Return false to reject; fill *error_code/*error_text.
 typedef std::function<bool(Controller* cntl, int* error_code, std::string* error_text)>
@@ -65,6 +84,8 @@ struct ServerOptions {
   // nshead raw-body service (parity: reference ServerOptions::nshead_service):
   // called with the request body; fills the response body.
   std::function<void(const IOBuf& req_body, IOBuf* resp_body)> nshead_handler;
+  // mongo service adaptor (see MongoHandlerFn above).
+  MongoHandlerFn mongo_handler;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)

@@ -84,6 +84,7 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   s->remote_side_ = options.remote_side;
   s->read_buf_.clear();
   s->preferred_protocol_index = -1;
+  s->client_protocol_hint = -1;
   s->protocol_ctx = nullptr;
   s->protocol_ctx_deleter = nullptr;
   s->input_events_.store(0, std::memory_order_relaxed);

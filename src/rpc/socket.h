@@ -102,6 +102,10 @@ class Socket {
   // Read buffer for InputMessenger.
   IOBuf& read_buf() { return read_buf_; }
   int preferred_protocol_index = -1;
+  // For client sockets: the protocol this connection was created for
+  // (socket map key). Lets magic-less client protocols (esp) gate their
+  // parse to their own connections.
+  int client_protocol_hint = -1;
   // Per-connection protocol state (e.g. the h2 session); freed by the
   // deleter when the socket is recycled.
   void* protocol_ctx = nullptr;

@@ -158,6 +158,7 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ss
   SocketId sid;
   if (Socket::Create(opts, &sid) != 0) return -1;
   if (Socket::Address(sid, out) != 0) return -1;
+  (*out)->client_protocol_hint = protocol_index;
   if (ssl) {
     static void* g_client_ctx = ssl::NewClientCtx();  // process-lifetime
     void* h = g_client_ctx != nullptr
