@@ -198,6 +198,7 @@ void bind_util(py::module_& m) {
 #include "base/codecs.h"
 #include "base/containers.h"
 #include "var/variable.h"
+#include "base/mcpack.h"
 
 void bind_codecs(py::module_& m) {
   auto c = m.def_submodule("codecs");
@@ -212,6 +213,82 @@ void bind_codecs(py::module_& m) {
     return py::bytes(out);
   });
   c.def("sha1_hex", [](py::bytes data) { return bam::SHA1HexDigest(data.cast<std::string>()); });
+
+  // mcpack v2 codec (base/mcpack.h): python structures <-> mcpack bytes.
+  struct McpackConv {
+    static bam::mcpack::Value from_py(py::handle h) {
+      namespace mc = bam::mcpack;
+      if (h.is_none()) return mc::Value();
+      if (py::isinstance<py::bool_>(h)) return mc::Value::Bool(h.cast<bool>());
+      if (py::isinstance<py::int_>(h)) {
+        long long v = h.cast<long long>();
+        return mc::Value::Int(v);
+      }
+      if (py::isinstance<py::float_>(h)) return mc::Value::Double(h.cast<double>());
+      if (py::isinstance<py::bytes>(h)) return mc::Value::Bin(h.cast<std::string>());
+      if (py::isinstance<py::str>(h)) return mc::Value::Str(h.cast<std::string>());
+      if (py::isinstance<py::dict>(h)) {
+        mc::Value v = mc::Value::Object();
+        for (auto kv : h.cast<py::dict>()) {
+          v.obj[kv.first.cast<std::string>()] = from_py(kv.second);
+        }
+        return v;
+      }
+      if (py::isinstance<py::list>(h) || py::isinstance<py::tuple>(h)) {
+        mc::Value v = mc::Value::Array();
+        for (auto it : h.cast<py::sequence>()) v.arr.push_back(from_py(it));
+        return v;
+      }
+      throw std::runtime_error("mcpack: unsupported python type");
+    }
+    static py::object to_py(const bam::mcpack::Value& v) {
+      namespace mc = bam::mcpack;
+      switch (v.type) {
+        case mc::Value::NIL: return py::none();
+        case mc::Value::BOOL: return py::bool_(v.b);
+        case mc::Value::INT: return py::int_(v.i);
+        case mc::Value::UINT: return py::int_(v.u);
+        case mc::Value::DOUBLE: return py::float_(v.d);
+        case mc::Value::STRING: return py::str(v.str);
+        case mc::Value::BINARY: return py::bytes(v.str);
+        case mc::Value::OBJECT: {
+          py::dict d;
+          for (const auto& kv : v.obj) d[py::str(kv.first)] = to_py(kv.second);
+          return d;
+        }
+        case mc::Value::ARRAY: {
+          py::list l;
+          for (const auto& it : v.arr) l.append(to_py(it));
+          return l;
+        }
+      }
+      return py::none();
+    }
+  };
+  c.def("mcpack_dumps", [](py::dict d) {
+    bam::mcpack::Value v = McpackConv::from_py(d);
+    std::string out;
+    if (!bam::mcpack::Serialize(v, &out)) throw std::runtime_error("mcpack serialize failed");
+    return py::bytes(out);
+  });
+  c.def("mcpack_loads", [](py::bytes data) {
+    std::string s = data.cast<std::string>();
+    bam::mcpack::Value v;
+    std::string err;
+    if (!bam::mcpack::Parse(s.data(), s.size(), &v, &err))
+      throw std::runtime_error("mcpack parse failed: " + err);
+    return McpackConv::to_py(v);
+  });
+  c.def("mcpack_to_json", [](py::bytes data) {
+    std::string s = data.cast<std::string>();
+    bam::mcpack::Value v;
+    std::string err;
+    if (!bam::mcpack::Parse(s.data(), s.size(), &v, &err))
+      throw std::runtime_error("mcpack parse failed: " + err);
+    std::string out;
+    bam::mcpack::ToJson(v, &out);
+    return out;
+  });
   c.def("murmur3_32", [](py::bytes data, uint32_t seed) {
     std::string s = data.cast<std::string>();
     return bam::MurmurHash3_32(s.data(), s.size(), seed);

@@ -1,14 +1,16 @@
 """Fuzz-harness parity (SURVEY §4 ≙ reference test/fuzzing/): each protocol
 parser has a libFuzzer target under tests/fuzz/ built by `make fuzz` with
 ASan; this test builds them and runs each for a few seconds. Regression
-memo: fuzz_snappy found the preamble malloc-bomb fixed in base/snappy.cc."""
+memo: fuzz_snappy found the preamble malloc-bomb fixed in base/snappy.cc;
+fuzz_mcpack found a parse_primitive stack overflow (type 0x3a)."""
 import os
 import subprocess
 
 import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-TARGETS = ["fuzz_rpc_meta", "fuzz_http", "fuzz_redis", "fuzz_json", "fuzz_snappy"]
+TARGETS = ["fuzz_rpc_meta", "fuzz_http", "fuzz_redis", "fuzz_json", "fuzz_snappy",
+           "fuzz_mcpack"]
 
 
 @pytest.mark.slow
