@@ -1,3 +1,4 @@
+#include "rpc/mysql_client.h"
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
@@ -16,6 +17,7 @@ struct BenchResult {
 int start_echo_server(int port);
 int start_nshead_server();
 int start_mongo_echo_server();
+// (mysql client bound directly below via rpc/mysql_client.h)
 int start_shm_echo_server(const std::string& name);
 int shm_call(const std::string& name, const std::string& method, const std::string& payload,
              std::string* response_out, std::string* err);
@@ -38,6 +40,32 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
 
 void bind_rpc(py::module_& m) {
   auto r = m.def_submodule("rpc");
+  py::class_<bam::MysqlResult>(r, "MysqlResult")
+      .def_readonly("ok", &bam::MysqlResult::ok)
+      .def_readonly("affected_rows", &bam::MysqlResult::affected_rows)
+      .def_readonly("last_insert_id", &bam::MysqlResult::last_insert_id)
+      .def_readonly("error_code", &bam::MysqlResult::error_code)
+      .def_readonly("error_message", &bam::MysqlResult::error_message)
+      .def_readonly("columns", &bam::MysqlResult::columns)
+      .def_readonly("rows", &bam::MysqlResult::rows);
+  py::class_<bam::MysqlClient>(r, "MysqlClient")
+      .def(py::init<>())
+      .def("connect", &bam::MysqlClient::Connect, py::arg("host"), py::arg("port"),
+           py::arg("user"), py::arg("password"), py::arg("db") = "",
+           py::arg("timeout_ms") = 3000, py::call_guard<py::gil_scoped_release>())
+      .def("query",
+           [](bam::MysqlClient& c, const std::string& sql) {
+             bam::MysqlResult res;
+             {
+               py::gil_scoped_release rel;
+               c.Query(sql, &res);
+             }
+             return res;
+           })
+      .def("ping", &bam::MysqlClient::Ping, py::call_guard<py::gil_scoped_release>())
+      .def("close", &bam::MysqlClient::Close)
+      .def("connected", &bam::MysqlClient::connected)
+      .def("server_version", &bam::MysqlClient::server_version);
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,

@@ -1,0 +1,341 @@
+// brpc_amd: MySQL client implementation (see mysql_client.h).
+#include "rpc/mysql_client.h"
+
+#include <errno.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
+#include <poll.h>
+#include <string.h>
+#include <sys/socket.h>
+#include <unistd.h>
+
+#include "base/codecs.h"
+#include "base/endpoint.h"
+
+namespace bam {
+
+namespace {
+
+// capability flags (protocol constants)
+constexpr uint32_t kClientLongPassword = 0x1;
+constexpr uint32_t kClientProtocol41 = 0x200;
+constexpr uint32_t kClientSecureConnection = 0x8000;
+constexpr uint32_t kClientPluginAuth = 0x80000;
+constexpr uint32_t kClientConnectWithDb = 0x8;
+
+constexpr uint8_t kComQuit = 0x01;
+constexpr uint8_t kComInitDb = 0x02;
+constexpr uint8_t kComQuery = 0x03;
+constexpr uint8_t kComPing = 0x0e;
+
+// mysql_native_password: SHA1(pwd) XOR SHA1(salt + SHA1(SHA1(pwd)))
+std::string native_scramble(const std::string& password, const std::string& salt) {
+  if (password.empty()) return "";
+  std::string h1 = SHA1Hash(password);
+  std::string h2 = SHA1Hash(h1);
+  std::string h3 = SHA1Hash(salt + h2);
+  std::string out(h1.size(), '\0');
+  for (size_t i = 0; i < h1.size(); ++i) out[i] = h1[i] ^ h3[i];
+  return out;
+}
+
+// length-encoded integer; returns bytes consumed, 0 on error, -1 for NULL.
+int lenc_int(const char* p, size_t n, uint64_t* v) {
+  if (n == 0) return 0;
+  uint8_t first = (uint8_t)p[0];
+  if (first < 0xfb) {
+    *v = first;
+    return 1;
+  }
+  if (first == 0xfb) return -1;  // NULL
+  if (first == 0xfc) {
+    if (n < 3) return 0;
+    *v = (uint8_t)p[1] | ((uint64_t)(uint8_t)p[2] << 8);
+    return 3;
+  }
+  if (first == 0xfd) {
+    if (n < 4) return 0;
+    *v = (uint8_t)p[1] | ((uint64_t)(uint8_t)p[2] << 8) | ((uint64_t)(uint8_t)p[3] << 16);
+    return 4;
+  }
+  if (n < 9) return 0;
+  uint64_t x = 0;
+  memcpy(&x, p + 1, 8);
+  *v = x;
+  return 9;
+}
+
+// length-encoded string; consumed bytes or 0; null=true for NULL value.
+int lenc_str(const char* p, size_t n, std::string* out, bool* null) {
+  uint64_t len;
+  int k = lenc_int(p, n, &len);
+  if (k == -1) {
+    *null = true;
+    return 1;
+  }
+  if (k == 0 || n < (size_t)k + len) return 0;
+  *null = false;
+  out->assign(p + k, len);
+  return k + (int)len;
+}
+
+}  // namespace
+
+MysqlClient::~MysqlClient() { Close(); }
+
+void MysqlClient::Close() {
+  if (fd_ >= 0) {
+    std::string quit(1, (char)kComQuit);
+    write_packet(quit, 0);
+    ::close(fd_);
+    fd_ = -1;
+  }
+}
+
+int MysqlClient::read_n(void* buf, size_t n) {
+  char* p = (char*)buf;
+  size_t got = 0;
+  while (got < n) {
+    struct pollfd pfd{fd_, POLLIN, 0};
+    int pr = ::poll(&pfd, 1, timeout_ms_);
+    if (pr <= 0) return -1;
+    ssize_t r = ::recv(fd_, p + got, n - got, 0);
+    if (r <= 0) return -1;
+    got += (size_t)r;
+  }
+  return 0;
+}
+
+int MysqlClient::read_packet(std::string* payload, uint8_t* seq) {
+  uint8_t head[4];
+  if (read_n(head, 4) != 0) return -1;
+  uint32_t len = head[0] | ((uint32_t)head[1] << 8) | ((uint32_t)head[2] << 16);
+  *seq = head[3];
+  payload->resize(len);
+  if (len != 0 && read_n(&(*payload)[0], len) != 0) return -1;
+  return 0;
+}
+
+int MysqlClient::write_packet(const std::string& payload, uint8_t seq) {
+  uint8_t head[4] = {(uint8_t)(payload.size() & 0xff), (uint8_t)((payload.size() >> 8) & 0xff),
+                     (uint8_t)((payload.size() >> 16) & 0xff), seq};
+  std::string buf((const char*)head, 4);
+  buf += payload;
+  size_t off = 0;
+  while (off < buf.size()) {
+    ssize_t w = ::send(fd_, buf.data() + off, buf.size() - off, MSG_NOSIGNAL);
+    if (w <= 0) {
+      if (errno == EAGAIN || errno == EINTR) continue;
+      return -1;
+    }
+    off += (size_t)w;
+  }
+  return 0;
+}
+
+int MysqlClient::Connect(const std::string& host, int port, const std::string& user,
+                         const std::string& password, const std::string& db,
+                         int timeout_ms) {
+  Close();
+  timeout_ms_ = timeout_ms;
+  EndPoint ep;
+  if (hostname2endpoint(host.c_str(), port, &ep) != 0) return -1;
+  fd_ = ::socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd_ < 0) return -1;
+  struct sockaddr_in sa;
+  memset(&sa, 0, sizeof(sa));
+  sa.sin_family = AF_INET;
+  sa.sin_addr = ep.ip;
+  sa.sin_port = htons((uint16_t)ep.port);
+  if (::connect(fd_, (struct sockaddr*)&sa, sizeof(sa)) != 0) {
+    Close();
+    return -1;
+  }
+  int one = 1;
+  setsockopt(fd_, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
+
+  // ---- HandshakeV10 ----
+  std::string hs;
+  uint8_t seq;
+  if (read_packet(&hs, &seq) != 0 || hs.size() < 33 || (uint8_t)hs[0] != 10) {
+    Close();
+    return -1;
+  }
+  size_t pos = 1;
+  size_t z = hs.find('\0', pos);
+  if (z == std::string::npos) {
+    Close();
+    return -1;
+  }
+  server_version_ = hs.substr(pos, z - pos);
+  pos = z + 1;
+  if (hs.size() < pos + 4 + 8 + 1 + 2) {
+    Close();
+    return -1;
+  }
+  pos += 4;  // thread id
+  std::string salt = hs.substr(pos, 8);
+  pos += 8 + 1;          // auth-plugin-data-part-1 + filler
+  pos += 2;              // capability_flags_1
+  if (hs.size() >= pos + 1 + 2 + 2 + 1 + 10) {
+    pos += 1 + 2 + 2;    // charset + status + capability_flags_2
+    uint8_t auth_len = (uint8_t)hs[pos];
+    pos += 1 + 10;       // auth data len + reserved
+    // auth-plugin-data-part-2: max(13, auth_len - 8), includes trailing '\0'
+    size_t part2 = auth_len > 8 ? (size_t)auth_len - 8 : 13;
+    if (part2 < 13) part2 = 13;
+    if (hs.size() >= pos + part2) {
+      salt += hs.substr(pos, part2 - 1);  // drop the trailing '\0'
+    }
+  }
+
+  // ---- HandshakeResponse41 ----
+  uint32_t caps = kClientLongPassword | kClientProtocol41 | kClientSecureConnection |
+                  kClientPluginAuth;
+  if (!db.empty()) caps |= kClientConnectWithDb;
+  std::string resp;
+  resp.append((const char*)&caps, 4);
+  uint32_t max_packet = 16 << 20;
+  resp.append((const char*)&max_packet, 4);
+  resp.push_back(33);  // utf8_general_ci
+  resp.append(23, '\0');
+  resp.append(user);
+  resp.push_back('\0');
+  std::string scramble = native_scramble(password, salt);
+  resp.push_back((char)scramble.size());
+  resp.append(scramble);
+  if (!db.empty()) {
+    resp.append(db);
+    resp.push_back('\0');
+  }
+  resp.append("mysql_native_password");
+  resp.push_back('\0');
+  if (write_packet(resp, (uint8_t)(seq + 1)) != 0) {
+    Close();
+    return -1;
+  }
+
+  std::string fin;
+  if (read_packet(&fin, &seq) != 0 || fin.empty()) {
+    Close();
+    return -1;
+  }
+  if ((uint8_t)fin[0] == 0xff) {  // ERR
+    int code = fin.size() >= 3 ? ((uint8_t)fin[1] | ((int)(uint8_t)fin[2] << 8)) : -1;
+    Close();
+    return code;
+  }
+  if ((uint8_t)fin[0] != 0x00) {  // auth-switch etc. unsupported
+    Close();
+    return -1;
+  }
+  return 0;
+}
+
+int MysqlClient::Ping() {
+  if (fd_ < 0) return -1;
+  if (write_packet(std::string(1, (char)kComPing), 0) != 0) return -1;
+  std::string p;
+  uint8_t seq;
+  if (read_packet(&p, &seq) != 0 || p.empty()) return -1;
+  return (uint8_t)p[0] == 0x00 ? 0 : -1;
+}
+
+int MysqlClient::Query(const std::string& sql, MysqlResult* out) {
+  *out = MysqlResult();
+  if (fd_ < 0) {
+    out->error_code = -1;
+    out->error_message = "not connected";
+    return -1;
+  }
+  std::string cmd(1, (char)kComQuery);
+  cmd += sql;
+  if (write_packet(cmd, 0) != 0) {
+    out->error_code = -1;
+    out->error_message = "write failed";
+    return -1;
+  }
+  std::string p;
+  uint8_t seq;
+  if (read_packet(&p, &seq) != 0 || p.empty()) {
+    out->error_code = -1;
+    out->error_message = "read failed";
+    return -1;
+  }
+  uint8_t first = (uint8_t)p[0];
+  if (first == 0xff) {  // ERR: code u16, sql-state marker'#'+5, message
+    out->error_code = p.size() >= 3 ? ((uint8_t)p[1] | ((int)(uint8_t)p[2] << 8)) : -1;
+    size_t mpos = 3;
+    if (p.size() > 3 && p[3] == '#') mpos = 9;
+    out->error_message = p.substr(mpos);
+    return out->error_code;
+  }
+  if (first == 0x00) {  // OK: affected_rows lenc, last_insert_id lenc
+    out->ok = true;
+    size_t pos = 1;
+    uint64_t v = 0;
+    int k = lenc_int(p.data() + pos, p.size() - pos, &v);
+    if (k > 0) {
+      out->affected_rows = v;
+      pos += k;
+      k = lenc_int(p.data() + pos, p.size() - pos, &v);
+      if (k > 0) out->last_insert_id = v;
+    }
+    return 0;
+  }
+  // resultset: first packet = column count (lenc)
+  uint64_t ncols = 0;
+  if (lenc_int(p.data(), p.size(), &ncols) <= 0 || ncols == 0 || ncols > 4096) {
+    out->error_code = -1;
+    out->error_message = "bad column count";
+    return -1;
+  }
+  // column definitions: catalog, schema, table, org_table, name, org_name...
+  for (uint64_t i = 0; i < ncols; ++i) {
+    if (read_packet(&p, &seq) != 0) return -1;
+    size_t pos = 0;
+    std::string field;
+    bool null;
+    for (int f = 0; f < 5; ++f) {  // 5th lenc-string = column name
+      int k = lenc_str(p.data() + pos, p.size() - pos, &field, &null);
+      if (k <= 0) {
+        field.clear();
+        break;
+      }
+      pos += (size_t)k;
+    }
+    out->columns.push_back(field);
+  }
+  // optional EOF (pre-DEPRECATE_EOF servers)
+  if (read_packet(&p, &seq) != 0) return -1;
+  bool was_eof = !p.empty() && (uint8_t)p[0] == 0xfe && p.size() < 9;
+  for (;;) {
+    if (!was_eof) {
+      // p already holds a row (or terminator)
+      was_eof = true;  // only skip the pre-read once
+    } else {
+      if (read_packet(&p, &seq) != 0) return -1;
+    }
+    if (!p.empty() && (uint8_t)p[0] == 0xfe && p.size() < 9) break;  // EOF
+    if (!p.empty() && (uint8_t)p[0] == 0xff) {                      // mid-stream ERR
+      out->error_code = p.size() >= 3 ? ((uint8_t)p[1] | ((int)(uint8_t)p[2] << 8)) : -1;
+      out->error_message = p.substr(p.size() > 9 && p[3] == '#' ? 9 : 3);
+      return out->error_code;
+    }
+    std::vector<std::string> row;
+    size_t pos = 0;
+    for (uint64_t i = 0; i < ncols; ++i) {
+      std::string v;
+      bool null = false;
+      int k = lenc_str(p.data() + pos, p.size() - pos, &v, &null);
+      if (k <= 0) break;
+      pos += (size_t)k;
+      row.push_back(null ? "" : v);
+    }
+    out->rows.push_back(std::move(row));
+  }
+  out->ok = true;
+  return 0;
+}
+
+}  // namespace bam
