@@ -1,4 +1,5 @@
 #include "rpc/mysql_client.h"
+#include "rpc/rtmp_client.h"
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
@@ -16,6 +17,7 @@ struct BenchResult {
 };
 int start_echo_server(int port);
 int start_nshead_server();
+int start_rtmp_server();
 int start_mongo_echo_server();
 // (mysql client bound directly below via rpc/mysql_client.h)
 int start_shm_echo_server(const std::string& name);
@@ -66,9 +68,37 @@ void bind_rpc(py::module_& m) {
       .def("close", &bam::MysqlClient::Close)
       .def("connected", &bam::MysqlClient::connected)
       .def("server_version", &bam::MysqlClient::server_version);
+  py::class_<bam::RtmpClient>(r, "RtmpClient")
+      .def(py::init<>())
+      .def("connect", &bam::RtmpClient::Connect, py::arg("host"), py::arg("port"),
+           py::arg("app"), py::arg("timeout_ms") = 3000,
+           py::call_guard<py::gil_scoped_release>())
+      .def("publish", &bam::RtmpClient::Publish, py::call_guard<py::gil_scoped_release>())
+      .def("play", &bam::RtmpClient::Play, py::call_guard<py::gil_scoped_release>())
+      .def("push_frame",
+           [](bam::RtmpClient& c, int type, uint32_t ts, py::bytes payload) {
+             std::string p = payload.cast<std::string>();
+             py::gil_scoped_release rel;
+             return c.PushFrame((uint8_t)type, ts, p);
+           })
+      .def("poll_frame",
+           [](bam::RtmpClient& c, int timeout_ms) -> py::object {
+             bam::rtmp::Message m;
+             int rc;
+             {
+               py::gil_scoped_release rel;
+               rc = c.PollFrame(&m, timeout_ms);
+             }
+             if (rc != 0) return py::none();
+             return py::make_tuple((int)m.type, m.timestamp, py::bytes(m.payload));
+           },
+           py::arg("timeout_ms") = 3000)
+      .def("close", &bam::RtmpClient::Close);
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("start_rtmp_server", &bam::rpctest::start_rtmp_server,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_mongo_server", &bam::rpctest::start_mongo_echo_server,
         py::call_guard<py::gil_scoped_release>());

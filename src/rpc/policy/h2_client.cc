@@ -204,8 +204,17 @@ ssize_t c_body_read(nghttp2_session*, int32_t, uint8_t* buf, size_t length,
 
 // ---- issue + parse hooks ----
 
+int g_grpc_protocol_index = -1;
+int g_h2c_protocol_index = -1;
+
+bool own_client_ctx(Socket* sock) {
+  return sock->protocol_ctx_owner == g_grpc_protocol_index ||
+         sock->protocol_ctx_owner == g_h2c_protocol_index;
+}
+
 int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
   if (!ngc().ok) return -1;
+  if (sock->protocol_ctx != nullptr && !own_client_ctx(sock)) return -1;
   H2ClientCtx* ctx = (H2ClientCtx*)sock->protocol_ctx;
   if (ctx == nullptr) {
     ctx = new H2ClientCtx;
@@ -220,6 +229,7 @@ int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
     ngc().submit_settings(ctx->session, NGHTTP2_FLAG_NONE, nullptr, 0);
     sock->protocol_ctx = ctx;
     sock->protocol_ctx_deleter = [](void* p) { h2c_unref((H2ClientCtx*)p); };
+    sock->protocol_ctx_owner = grpc ? g_grpc_protocol_index : g_h2c_protocol_index;
   }
   std::string payload = cntl->call.request_buf.to_string();
   H2ClientBody* body = new H2ClientBody;
@@ -275,6 +285,7 @@ struct H2ClientPump : public InputMessageBase {};
 
 ParseResult ParseGrpcClient(IOBuf* source, Socket* sock, bool /*eof*/) {
   if (sock->user() != nullptr) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  if (!own_client_ctx(sock)) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
   H2ClientCtx* ctx = (H2ClientCtx*)sock->protocol_ctx;
   if (ctx == nullptr) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
   std::string bytes = source->to_string();
@@ -309,7 +320,7 @@ void RegisterGrpcClientProtocol() {
     p.support_server = false;
     p.support_client = true;
     p.name = "grpc";
-    RegisterProtocol(p);
+    g_grpc_protocol_index = RegisterProtocol(p);
     // Plain HTTP-semantics h2 client (parity: reference protocol "h2"):
     // same nghttp2 session machinery, no gRPC framing or trailers.
     Protocol h2;
@@ -320,7 +331,7 @@ void RegisterGrpcClientProtocol() {
     h2.support_server = false;
     h2.support_client = true;
     h2.name = "h2";
-    RegisterProtocol(h2);
+    g_h2c_protocol_index = RegisterProtocol(h2);
   });
 }
 

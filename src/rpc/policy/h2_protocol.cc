@@ -332,8 +332,12 @@ struct H2PumpMessage : public InputMessageBase {
 
 const char kPreface[] = "PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n";  // 24 bytes
 
+int g_h2_protocol_index = -1;
+
 ParseResult ParseH2(IOBuf* source, Socket* sock, bool /*eof*/) {
   if (sock->user() == nullptr) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  if (sock->protocol_ctx != nullptr && sock->protocol_ctx_owner != g_h2_protocol_index)
+    return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);  // another protocol's connection
   H2Ctx* ctx = (H2Ctx*)sock->protocol_ctx;
   if (ctx == nullptr) {
     // new connection: require the client preface
@@ -355,6 +359,7 @@ ParseResult ParseH2(IOBuf* source, Socket* sock, bool /*eof*/) {
     ng().submit_settings(ctx->session, NGHTTP2_FLAG_NONE, nullptr, 0);
     sock->protocol_ctx = ctx;
     sock->protocol_ctx_deleter = [](void* p) { h2_ctx_unref((H2Ctx*)p); };
+    sock->protocol_ctx_owner = g_h2_protocol_index;
   }
   // Feed everything we have into the session.
   std::string bytes = source->to_string();
@@ -413,7 +418,7 @@ void RegisterH2Protocol() {
     p.support_server = true;
     p.support_client = false;
     p.name = "h2";
-    RegisterProtocol(p);
+    g_h2_protocol_index = RegisterProtocol(p);
   });
 }
 

@@ -119,6 +119,16 @@ int protocol_call(const std::string& addr, const std::string& protocol,
   return 0;
 }
 
+// ---- rtmp media server ----
+int start_rtmp_server() {
+  Server* server = new Server;
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  opts.enable_rtmp = true;
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
 // ---- mongo server (adaptor parity) ----
 // Replies {ok: 1.0} to OP_QUERY ("ismaster"-style) and OP_MSG; records
 // fire-and-forget inserts into a counter readable via OP_QUERY on

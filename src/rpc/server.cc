@@ -102,6 +102,7 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   policy::RegisterSofaProtocol();
   if (options_.nshead_handler) policy::RegisterNsheadProtocol();
   if (options_.mongo_handler) policy::RegisterMongoProtocol();
+  if (options_.enable_rtmp) policy::RegisterRtmpProtocol();
   if (options_.redis_service != nullptr) policy::RegisterRedisProtocol();
   int listen_fd = tcp_listen(ep);
   if (listen_fd < 0) {

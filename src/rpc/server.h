@@ -86,6 +86,10 @@ struct ServerOptions {
   std::function<void(const IOBuf& req_body, IOBuf* resp_body)> nshead_handler;
   // mongo service adaptor (see MongoHandlerFn above).
   MongoHandlerFn mongo_handler;
+  // RTMP media endpoint (policy/rtmp_protocol.cc): handshake + chunk
+  // streams + connect/createStream/publish/play and a built-in
+  // publish->play relay hub.
+  bool enable_rtmp = false;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)

@@ -87,6 +87,7 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   s->client_protocol_hint = -1;
   s->protocol_ctx = nullptr;
   s->protocol_ctx_deleter = nullptr;
+  s->protocol_ctx_owner = -1;
   s->input_events_.store(0, std::memory_order_relaxed);
   s->ssl_ = nullptr;
   s->ssl_state_.store(0, std::memory_order_relaxed);
@@ -183,6 +184,7 @@ void Socket::Recycle() {
   }
   protocol_ctx = nullptr;
   protocol_ctx_deleter = nullptr;
+  protocol_ctx_owner = -1;
   int fd = fd_.load(std::memory_order_acquire);
   if (fd >= 0) {
     ::close(fd);

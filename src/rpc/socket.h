@@ -107,9 +107,14 @@ class Socket {
   // parse to their own connections.
   int client_protocol_hint = -1;
   // Per-connection protocol state (e.g. the h2 session); freed by the
-  // deleter when the socket is recycled.
+  // deleter when the socket is recycled. protocol_ctx_owner is the
+  // registry index of the protocol that installed it: every protocol MUST
+  // check the owner before reinterpreting the pointer (an RTMP context
+  // reinterpreted as an h2 session locks garbage as a mutex — found by
+  // the ASan harness).
   void* protocol_ctx = nullptr;
   void (*protocol_ctx_deleter)(void*) = nullptr;
+  int protocol_ctx_owner = -1;
 
   // Server-side auth result, set on the first verified request of the
   // connection (rpc/authenticator.h); owned by the socket, freed at

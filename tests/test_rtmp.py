@@ -1,0 +1,73 @@
+"""RTMP media stack (reference brpc/rtmp.cpp + policy/rtmp_protocol.cpp,
+clean-room subset): C0C1C2 handshake, chunk streams, AMF0 commands,
+publish->play relay hub. One port serves RTMP alongside baidu_std."""
+import threading
+import time
+
+import brpc_amd as b
+import pytest
+
+r = b.core.rpc
+
+AUDIO, VIDEO, DATA = 8, 9, 18
+
+
+@pytest.fixture(scope="module")
+def port():
+    return r.start_rtmp_server()
+
+
+def test_connect_and_publish(port):
+    pub = r.RtmpClient()
+    assert pub.connect("127.0.0.1", port, "live") == 0
+    assert pub.publish("streamA") == 0
+    pub.close()
+
+
+def test_publish_play_relay(port):
+    pub = r.RtmpClient()
+    assert pub.connect("127.0.0.1", port, "live") == 0
+    assert pub.publish("cam1") == 0
+    ply = r.RtmpClient()
+    assert ply.connect("127.0.0.1", port, "live") == 0
+    assert ply.play("cam1") == 0
+    # media flows publisher -> server -> player
+    frames = [(VIDEO, 100, b"\x17" + b"v" * 500),
+              (AUDIO, 120, b"\xaf" + b"a" * 100),
+              (VIDEO, 140, b"\x27" + b"q" * 5000)]  # crosses chunk boundary
+    for t, ts, payload in frames:
+        assert pub.push_frame(t, ts, payload) == 0
+    for t, ts, payload in frames:
+        got = ply.poll_frame(5000)
+        assert got is not None, "frame lost"
+        assert got[0] == t
+        assert got[1] == ts
+        assert got[2] == payload
+    pub.close()
+    ply.close()
+
+
+def test_two_players_both_receive(port):
+    pub = r.RtmpClient()
+    assert pub.connect("127.0.0.1", port, "live") == 0
+    assert pub.publish("multi") == 0
+    players = []
+    for _ in range(2):
+        p = r.RtmpClient()
+        assert p.connect("127.0.0.1", port, "live") == 0
+        assert p.play("multi") == 0
+        players.append(p)
+    assert pub.push_frame(DATA, 1, b"meta" * 10) == 0
+    for p in players:
+        got = p.poll_frame(5000)
+        assert got is not None
+        assert got[2] == b"meta" * 10
+        p.close()
+    pub.close()
+
+
+def test_std_rpc_shares_port(port):
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "std",
+                                    "EchoService.Echo", b"rtmp-port")
+    assert rc == 0, err
+    assert resp == b"rtmp-port"
