@@ -57,8 +57,10 @@ class ObjectPool {
   };
 
   static Global& global() {
-    static Global g;
-    return g;
+    // Leaked on purpose: thread-exit Local destructors and daemon threads
+    // return objects during static destruction (see resource_pool.h).
+    static Global* g = new Global;
+    return *g;
   }
   static Local& local() {
     static thread_local Local lc;

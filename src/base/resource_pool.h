@@ -75,8 +75,12 @@ class ResourcePool {
     }
   };
   static Singleton& singleton() {
-    static Singleton s;
-    return s;
+    // Heap-allocated and deliberately leaked: daemon threads (timer
+    // thread, fiber workers) keep returning resources during static
+    // destruction at process exit — destroying this vector/mutex then is
+    // a use-after-free (caught by ASan in the test suite).
+    static Singleton* s = new Singleton;
+    return *s;
   }
 };
 
