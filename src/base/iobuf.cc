@@ -28,12 +28,14 @@ void set_byte_mover(ByteMoverFns fns) { g_mover = fns; }
 
 static UploadAsyncFn g_upload_async = nullptr;
 void set_upload_async(UploadAsyncFn fn) {
-  // Default OFF: same-box A/B showed per-response upload kernels serialize
-  // on the staging stream and the batch gather must drain them, losing to
-  // the runtime's internal SDMA parallelism (137k vs 94k QPS at 64B/c32).
-  // BAM_UPLOAD_ASYNC=1 re-enables for experiments.
+  // Default ON since the upload leg moved to its OWN stream (uploads
+  // pipeline with no host sync; the batch gather host-waits for the upload
+  // ticket, which is almost always already passed). Same-box A/B:
+  // 64B/c64 150.8k vs 131.2k QPS, 16KB/c32 118.0k vs 83.3k (+42%).
+  // The first single-stream attempt LOST (gathers drained the upload queue
+  // in stream order) — BAM_UPLOAD_ASYNC=0 restores sync hipMemcpy uploads.
   const char* e = getenv("BAM_UPLOAD_ASYNC");
-  if (e == nullptr || e[0] != '1') return;
+  if (e != nullptr && e[0] == '0') return;
   g_upload_async = fn;
 }
 
