@@ -320,6 +320,8 @@ void bind_memcache(py::module_& m) {
   py::class_<bam::MemcacheClient>(m, "MemcacheClient")
       .def(py::init<const std::string&, int>(), py::arg("addr"), py::arg("timeout_ms") = 1000)
       .def("ok", &bam::MemcacheClient::ok)
+      .def("sasl_auth_plain", &bam::MemcacheClient::SaslAuthPlain,
+           py::call_guard<py::gil_scoped_release>())
       .def("set", [](bam::MemcacheClient& c, const std::string& key, const std::string& value,
                      uint32_t flags, uint32_t exptime) {
         return c.Set(key, value, flags, exptime);

@@ -30,6 +30,7 @@ int Channel::Init(EndPoint ep, const ChannelOptions* options) {
   if (options_.protocol == "sofa_pbrpc") policy::RegisterSofaProtocol();
   if (options_.protocol == "nshead") policy::RegisterNsheadProtocol();
   if (options_.protocol == "esp") policy::RegisterEspProtocol();
+  if (options_.protocol == "nova_pbrpc") policy::RegisterNovaProtocol();
   protocol_index_ = FindClientProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
   return 0;

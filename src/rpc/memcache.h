@@ -36,6 +36,12 @@ class MemcacheClient {
   int RawCall(uint8_t opcode, const std::string& key, const std::string& value,
               const std::string& extras, MemcacheResponse* out);
 
+  // Couchbase parity (reference policy/couchbase_authenticator.cpp +
+  // couchbase_protocol.cpp): SASL PLAIN authentication over the memcache
+  // binary protocol (opcode 0x21, mechanism "PLAIN",
+  // value = "\0user\0password"). 0 on success.
+  int SaslAuthPlain(const std::string& user, const std::string& password);
+
  private:
   Channel channel_;
   bool init_ok_ = false;

@@ -39,6 +39,7 @@ void RegisterSofaProtocol();
 void RegisterNsheadProtocol();
 void RegisterMongoProtocol();  // server-side (policy/mongo_protocol.cc)
 void RegisterEspProtocol();    // client-side (policy/legacy_protocols.cc)
+void RegisterNovaProtocol();   // client-side nshead variant (method index in reserved)
 void RegisterRtmpProtocol();   // server-side (policy/rtmp_protocol.cc)
 
 // Implemented in rpc/builtin/builtin_services.cc; returns true if the path

@@ -557,6 +557,41 @@ void PackNsheadRequest(IOBuf* out, Controller* cntl, uint64_t /*fifo-correlated*
   pack_nshead(out, 0, 0, (uint32_t)cntl->log_id(), nullptr, cntl->call.request_buf);
 }
 
+// ==================== nova_pbrpc (client) ====================
+// Parity: reference policy/nova_pbrpc_protocol.cpp — nshead framing with
+// NO meta message: nshead.reserved carries the method index,
+// nshead.version==1 marks a snappy-compressed protobuf body; responses
+// come back nshead-framed, FIFO-correlated (nshead has no correlation
+// id). Server side parity is the NsheadService adaptor
+// (ServerOptions::nshead_handler), exactly like the reference's
+// NovaServiceAdaptor over NsheadService.
+
+void PackNovaRequest(IOBuf* out, Controller* cntl, uint64_t /*fifo*/) {
+  uint32_t method_index = (uint32_t)strtoul(cntl->call.method_name.c_str(), nullptr, 10);
+  IOBuf body = cntl->call.request_buf;
+  uint16_t version = 0;
+  if (cntl->request_compress_type() == COMPRESS_TYPE_SNAPPY) {
+    RegisterBuiltinCompressHandlers();
+    IOBuf compressed;
+    if (ApplyCompress(COMPRESS_TYPE_SNAPPY, body, &compressed)) {
+      body.swap(compressed);
+      version = 1;
+    }
+  }
+  char h[36];
+  memset(h, 0, sizeof(h));
+  memcpy(h + 2, &version, 2);
+  uint32_t log_id = (uint32_t)cntl->log_id();
+  memcpy(h + 4, &log_id, 4);
+  uint32_t magic = 0xfb709394u;
+  memcpy(h + 24, &magic, 4);
+  memcpy(h + 28, &method_index, 4);  // reserved = method index
+  uint32_t blen = (uint32_t)body.size();
+  memcpy(h + 32, &blen, 4);
+  out->append(h, sizeof(h));
+  out->append(std::move(body));
+}
+
 // ==================== esp (client) ====================
 // Parity: reference policy/esp_protocol.cpp + esp_head.h — 32-byte packed
 // little-endian head {from u64, to u64, msg u32, msg_id u64, body_len i32}
@@ -650,6 +685,21 @@ void RegisterSofaProtocol() {
     p.support_server = true;
     p.support_client = true;
     p.name = "sofa_pbrpc";
+    RegisterProtocol(p);
+  });
+}
+
+void RegisterNovaProtocol() {
+  static std::once_flag once;
+  std::call_once(once, [] {
+    Protocol p;
+    p.parse = ParseNshead;        // same frame; adaptor semantics differ
+    p.process_response = ProcessNsheadResponse;
+    p.pack_request = PackNovaRequest;
+    p.client_pipelined = true;
+    p.support_server = false;  // server side = nshead_handler adaptor
+    p.support_client = true;
+    p.name = "nova_pbrpc";
     RegisterProtocol(p);
   });
 }

@@ -162,6 +162,18 @@ MemcacheClient::MemcacheClient(const std::string& addr, int timeout_ms) {
   init_ok_ = channel_.Init(addr.c_str(), &opts) == 0;
 }
 
+int MemcacheClient::SaslAuthPlain(const std::string& user, const std::string& password) {
+  std::string token;
+  token.push_back('\0');
+  token += user;
+  token.push_back('\0');
+  token += password;
+  MemcacheResponse resp;
+  int rc = RawCall(0x21, "PLAIN", token, "", &resp);
+  if (rc != 0) return rc;
+  return resp.status == 0 ? 0 : 10000 + resp.status;
+}
+
 int MemcacheClient::RawCall(uint8_t opcode, const std::string& key, const std::string& value,
                             const std::string& extras, MemcacheResponse* out) {
   if (!init_ok_) return -1;

@@ -87,3 +87,48 @@ def test_std_still_works_on_same_port(port):
                                     "EchoService.Echo", b"std-hi")
     assert rc == 0, err
     assert resp == b"std-hi"
+
+
+def test_nova_pbrpc_client():
+    """nova_pbrpc (reference policy/nova_pbrpc_protocol.cpp): nshead frame,
+    method index in `reserved`, version==1 => snappy body, FIFO replies.
+    Scripted server checks the fields and echoes."""
+    import socket
+    import threading
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(2)
+    seen = {}
+
+    def run():
+        c, _ = srv.accept()
+        buf = b""
+        while True:
+            try:
+                chunk = c.recv(65536)
+            except OSError:
+                return
+            if not chunk:
+                return
+            buf += chunk
+            while len(buf) >= 36:
+                rid, ver, log_id, prov, magic, reserved, blen = struct.unpack(
+                    "<HHI16sIII", buf[:36])
+                if len(buf) < 36 + blen:
+                    break
+                body = buf[36:36 + blen]
+                buf = buf[36 + blen:]
+                seen["reserved"] = reserved
+                seen["version"] = ver
+                reply = b"nova:" + body
+                c.sendall(struct.pack("<HHI16sIII", 0, 0, log_id, b"\0" * 16,
+                                      0xFB709394, 0, len(reply)) + reply)
+
+    threading.Thread(target=run, daemon=True).start()
+    port2 = srv.getsockname()[1]
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port2, "nova_pbrpc", "7", b"pbbody")
+    assert rc == 0, err
+    assert resp == b"nova:pbbody"
+    assert seen["reserved"] == 7
+    assert seen["version"] == 0
+    srv.close()

@@ -52,6 +52,11 @@ def start_fake_memcached():
                         status = 1
                 elif op == 0x0B:  # version
                     rv = b"1.6.0-fake"
+                elif op == 0x21:  # SASL auth (couchbase-style PLAIN)
+                    if key == b"PLAIN" and value == b"\x00cb_user\x00cb_pass":
+                        rv = b"Authenticated"
+                    else:
+                        status = 0x20  # auth error
                 resp_body = rx + rv
                 hdr = struct.pack(">BBHBBHIIQ", 0x81, op, 0, len(rx), 0, status,
                                   len(resp_body), 0, 0)
@@ -96,3 +101,13 @@ def test_memcache_many_pipelined(mc_port):
         assert c.set(f"key{i}", b"v%d" % i) == 0
     for i in range(50):
         assert c.get(f"key{i}") == b"v%d" % i
+
+
+def test_couchbase_sasl_auth(mc_port):
+    """Couchbase parity (reference policy/couchbase_authenticator.cpp):
+    SASL PLAIN over the memcache binary protocol."""
+    c = b.MemcacheClient("127.0.0.1:%d" % mc_port)
+    assert c.ok()
+    assert c.sasl_auth_plain("cb_user", "cb_pass") == 0
+    rc = c.sasl_auth_plain("cb_user", "wrong")
+    assert rc == 10000 + 0x20
