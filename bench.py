@@ -48,7 +48,13 @@ def main():
         import torch.distributed as dist_mod
 
         dist = dist_mod
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # Every rank must pick the SAME backend: on a box with fewer GPUs
+        # than ranks, cuda.is_available() diverges across ranks (each rank
+        # is masked to its own device index). BAM_BENCH_BACKEND forces it
+        # (the 2-rank CPU test sets gloo); otherwise trust the full-node
+        # assumption (driver runs one rank per real GPU).
+        backend = os.environ.get("BAM_BENCH_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo")
         dist.init_process_group(backend=backend)
 
     ndev = b.core.gpu.initialize()

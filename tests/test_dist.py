@@ -11,6 +11,7 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 def test_bench_two_ranks_gloo():
     env = dict(os.environ)
+    env["BAM_BENCH_BACKEND"] = "gloo"  # ranks > GPUs on a 1-GPU box must agree
     env["MASTER_ADDR"] = "127.0.0.1"
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
@@ -58,7 +59,9 @@ def test_stream_p2p_two_ranks_gloo():
         "--master-addr", "127.0.0.1", "--master-port", str(port),
         "tools/stream_xgmi_bench.py", "--frames", "20", "--warmup", "2",
     ]
-    out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True, timeout=300)
+    env = dict(os.environ)
+    env["BAM_BENCH_BACKEND"] = "gloo"  # 2 ranks on a 1-GPU box must agree
+    out = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True, text=True, timeout=300)
     assert out.returncode == 0, out.stderr[-3000:]
     line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
     res = json.loads(line)

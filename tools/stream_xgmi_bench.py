@@ -36,7 +36,8 @@ def main():
     args = ap.parse_args()
 
     use_gpu = torch.cuda.is_available()
-    backend = "nccl" if use_gpu else "gloo"
+    backend = os.environ.get("BAM_BENCH_BACKEND") or ("nccl" if use_gpu else "gloo")
+    use_gpu = use_gpu and backend == "nccl"
     dist.init_process_group(backend=backend)
     rank = dist.get_rank()
     dev = torch.device("cuda:0") if use_gpu else torch.device("cpu")
