@@ -81,9 +81,10 @@ class PyServer {
 
   int start(int port, int max_concurrency, const std::string& auth_user,
             const std::string& auth_password, const std::string& ssl_cert,
-            const std::string& ssl_key) {
+            const std::string& ssl_key, const std::string& adaptive_max_concurrency) {
     ServerOptions opts;
     opts.max_concurrency = max_concurrency;
+    opts.adaptive_max_concurrency = adaptive_max_concurrency;
     if (!auth_user.empty()) {
       auth_.reset(new PasswordAuthenticator(auth_user, auth_password));
       opts.auth = auth_.get();
@@ -169,7 +170,8 @@ void bind_api(py::module_& m) {
            py::arg("handler"))
       .def("start", &PyServer::start, py::arg("port") = 0, py::arg("max_concurrency") = 0,
            py::arg("auth_user") = "", py::arg("auth_password") = "",
-           py::arg("ssl_cert") = "", py::arg("ssl_key") = "")
+           py::arg("ssl_cert") = "", py::arg("ssl_key") = "",
+           py::arg("adaptive_max_concurrency") = "")
       .def("stop", &PyServer::stop)
       .def("running", &PyServer::running)
       .def("port", &PyServer::port)

@@ -7,6 +7,7 @@
 #include "rpc/authenticator.h"
 #include "rpc/channel.h"
 #include "rpc/controller.h"
+#include "rpc/concurrency_limiter.h"
 #include "rpc/server.h"
 #include "rpc/compress.h"
 #include "rpc/policy/http_protocol.h"
@@ -239,6 +240,10 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
     cntl->server_->nprocessed.fetch_add(1, std::memory_order_relaxed);
   }
   if (status != nullptr) *status << (monotonic_time_us() - start_us);
+  if (cntl->server_ != nullptr && cntl->server_->limiter() != nullptr &&
+      cntl->concurrency_counted_) {
+    cntl->server_->limiter()->OnResponse(cntl->ErrorCode(), monotonic_time_us() - start_us);
+  }
   if (rpcz::enabled()) {
     rpcz::Span span;
     span.start_us = start_us;
@@ -339,6 +344,13 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   if (server->max_concurrency() > 0 &&
       server->concurrency.load(std::memory_order_relaxed) >= server->max_concurrency()) {
     cntl->SetFailed(ELIMIT, "reached server max_concurrency");
+    delete msg;
+    done->Run();
+    return;
+  }
+  if (server->limiter() != nullptr &&
+      !server->limiter()->OnRequest(server->concurrency.load(std::memory_order_relaxed) + 1)) {
+    cntl->SetFailed(ELIMIT, "rejected by adaptive concurrency limiter");
     delete msg;
     done->Run();
     return;

@@ -1,6 +1,7 @@
 #include "rpc/server.h"
 
 #include "rpc/ssl_util.h"
+#include "rpc/concurrency_limiter.h"
 
 #include <sys/socket.h>
 #include <unistd.h>
@@ -88,6 +89,9 @@ int Server::Start(int port, const ServerOptions* opt) {
 int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   if (IsRunning()) return -1;
   if (opt != nullptr) options_ = *opt;
+  if (limiter_ == nullptr) {
+    limiter_ = ConcurrencyLimiter::Create(options_.adaptive_max_concurrency);
+  }
   if (!options_.ssl_cert.empty() || !options_.ssl_key.empty()) {
     ssl_ctx_ = ssl::NewServerCtx(options_.ssl_cert, options_.ssl_key);
     if (ssl_ctx_ == nullptr) {

@@ -90,6 +90,9 @@ struct ServerOptions {
   // streams + connect/createStream/publish/play and a built-in
   // publish->play relay hub.
   bool enable_rtmp = false;
+  // Adaptive concurrency (rpc/concurrency_limiter.h): "" | "auto" |
+  // "timeout:<ms>" | "<n>". Applies on top of max_concurrency.
+  std::string adaptive_max_concurrency;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)
@@ -120,6 +123,7 @@ class Server {
   std::atomic<int32_t> concurrency{0};
   int max_concurrency() const { return options_.max_concurrency; }
   const ServerOptions& options() const { return options_; }
+  class ConcurrencyLimiter* limiter() const { return limiter_; }
   InputMessenger* messenger() { return &messenger_; }
   RedisService* redis_service() const { return options_.redis_service; }
 
@@ -138,6 +142,7 @@ class Server {
   std::atomic<bool> running_{false};
   InputMessenger messenger_;
   void* ssl_ctx_ = nullptr;  // SSL_CTX* when TLS enabled (never freed: sockets may outlive Stop)
+  class ConcurrencyLimiter* limiter_ = nullptr;  // never freed (sockets may outlive Stop)
 };
 
 }  // namespace bam
