@@ -23,6 +23,7 @@
 #include <string>
 
 #include "base/iobuf.h"
+#include "base/mcpack.h"
 #include "base/logging.h"
 #include "base/time.h"
 #include "rpc/compress.h"
@@ -481,6 +482,9 @@ void PackSofaRequest(IOBuf* out, Controller* cntl, uint64_t correlation_id) {
 constexpr uint32_t kNsheadMagic = 0xfb709394u;
 constexpr size_t kNsheadLen = 36;
 
+int g_nshead_protocol_index = -1;
+int g_nova_protocol_index = -1;
+
 struct NsheadMessage : public InputMessageBase {
   uint16_t id = 0;
   uint16_t version = 0;
@@ -489,7 +493,8 @@ struct NsheadMessage : public InputMessageBase {
   IOBuf body;
 };
 
-ParseResult ParseNshead(IOBuf* source, Socket*, bool) {
+// Frame-only parse shared by nshead / nova / ubrpc (no ownership gate).
+ParseResult parse_nshead_frame(IOBuf* source) {
   char aux[kNsheadLen];
   if (source->size() < kNsheadLen) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
   const char* h = (const char*)source->fetch(aux, kNsheadLen);
@@ -507,6 +512,16 @@ ParseResult ParseNshead(IOBuf* source, Socket*, bool) {
   source->pop_front(kNsheadLen);
   source->cutn(&msg->body, body_len);
   return ParseResult::make_ok(msg);
+}
+
+ParseResult ParseNshead(IOBuf* source, Socket* sock, bool) {
+  // Client side: ubrpc shares this frame with its OWN id-correlated parse —
+  // the FIFO path only claims connections created for nshead/nova.
+  if (sock->user() == nullptr && sock->client_protocol_hint != g_nshead_protocol_index &&
+      sock->client_protocol_hint != g_nova_protocol_index) {
+    return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  }
+  return parse_nshead_frame(source);
 }
 
 void pack_nshead(IOBuf* out, uint16_t id, uint16_t version, uint32_t log_id,
@@ -590,6 +605,94 @@ void PackNovaRequest(IOBuf* out, Controller* cntl, uint64_t /*fifo*/) {
   memcpy(h + 32, &blen, 4);
   out->append(h, sizeof(h));
   out->append(std::move(body));
+}
+
+// ==================== ubrpc (client) ====================
+// Parity: reference policy/ubrpc2pb_protocol.cpp — nshead (version 1000)
+// framing a compack/mcpack object:
+//   request:  {header:{...}, content:[{service_name, id, method, params{...}}]}
+//   response: {content:[{id, (code,message) | result/params...}]}
+// Correlation rides content[0].id (a REAL correlation id, unlike plain
+// nshead's FIFO). We are payload-centric: params = {"req": <binary>}; the
+// response payload handed back is the re-serialized content[0] object.
+
+int g_ubrpc_protocol_index = -1;
+
+struct UbrpcMessage : public InputMessageBase {
+  int64_t id = 0;
+  int code = 0;
+  std::string message;
+  IOBuf payload;  // mcpack of content[0]
+};
+
+ParseResult ParseUbrpc(IOBuf* source, Socket* sock, bool eof) {
+  // reuse the nshead frame; gate on our own hint so plain nshead keeps
+  // working independently
+  if (sock->client_protocol_hint != g_ubrpc_protocol_index)
+    return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  (void)eof;
+  ParseResult raw = parse_nshead_frame(source);
+  if (raw.error != PARSE_OK) return raw;
+  NsheadMessage* nm = (NsheadMessage*)raw.msg;
+  std::string body = nm->body.to_string();
+  delete nm;
+  mcpack::Value root;
+  if (!mcpack::Parse(body.data(), body.size(), &root) ||
+      root.obj.count("content") == 0 || root.obj["content"].arr.empty()) {
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  }
+  mcpack::Value& c0 = root.obj["content"].arr[0];
+  UbrpcMessage* msg = new UbrpcMessage;
+  auto it = c0.obj.find("id");
+  if (it != c0.obj.end()) msg->id = it->second.type == mcpack::Value::UINT
+                                        ? (int64_t)it->second.u
+                                        : it->second.i;
+  it = c0.obj.find("code");
+  if (it != c0.obj.end()) msg->code = (int)(it->second.type == mcpack::Value::UINT
+                                                ? (int64_t)it->second.u
+                                                : it->second.i);
+  it = c0.obj.find("message");
+  if (it != c0.obj.end()) msg->message = it->second.str;
+  std::string re;
+  mcpack::Serialize(c0.type == mcpack::Value::OBJECT ? c0 : mcpack::Value::Object(), &re);
+  msg->payload.append(re);
+  return ParseResult::make_ok(msg);
+}
+
+void ProcessUbrpcResponse(InputMessageBase* mb) {
+  UbrpcMessage* msg = (UbrpcMessage*)mb;
+  finish_client_call((SessionId)msg->id, msg->socket_id,
+                     msg->code != 0 ? msg->code : 0, msg->message, &msg->payload, 0);
+  delete msg;
+}
+
+void PackUbrpcRequest(IOBuf* out, Controller* cntl, uint64_t correlation_id) {
+  mcpack::Value root = mcpack::Value::Object();
+  mcpack::Value header = mcpack::Value::Object();
+  header.obj["connection"] = mcpack::Value::Bool(true);
+  root.obj["header"] = std::move(header);
+  mcpack::Value c0 = mcpack::Value::Object();
+  c0.obj["service_name"] = mcpack::Value::Str(cntl->call.service_name);
+  c0.obj["method"] = mcpack::Value::Str(cntl->call.method_name);
+  c0.obj["id"] = mcpack::Value::Int((int64_t)correlation_id);
+  mcpack::Value params = mcpack::Value::Object();
+  params.obj["req"] = mcpack::Value::Bin(cntl->call.request_buf.to_string());
+  c0.obj["params"] = std::move(params);
+  mcpack::Value content = mcpack::Value::Array();
+  content.arr.push_back(std::move(c0));
+  root.obj["content"] = std::move(content);
+  std::string body;
+  mcpack::Serialize(root, &body);
+  char h[36];
+  memset(h, 0, sizeof(h));
+  uint16_t version = 1000;  // UBRPC_NSHEAD_VERSION
+  memcpy(h + 2, &version, 2);
+  uint32_t magic = 0xfb709394u;
+  memcpy(h + 24, &magic, 4);
+  uint32_t blen = (uint32_t)body.size();
+  memcpy(h + 32, &blen, 4);
+  out->append(h, sizeof(h));
+  out->append(body);
 }
 
 // ==================== esp (client) ====================
@@ -689,6 +792,20 @@ void RegisterSofaProtocol() {
   });
 }
 
+void RegisterUbrpcProtocol() {
+  static std::once_flag once;
+  std::call_once(once, [] {
+    Protocol p;
+    p.parse = ParseUbrpc;
+    p.process_response = ProcessUbrpcResponse;
+    p.pack_request = PackUbrpcRequest;
+    p.support_server = false;
+    p.support_client = true;
+    p.name = "ubrpc";
+    g_ubrpc_protocol_index = RegisterProtocol(p);
+  });
+}
+
 void RegisterNovaProtocol() {
   static std::once_flag once;
   std::call_once(once, [] {
@@ -700,7 +817,7 @@ void RegisterNovaProtocol() {
     p.support_server = false;  // server side = nshead_handler adaptor
     p.support_client = true;
     p.name = "nova_pbrpc";
-    RegisterProtocol(p);
+    g_nova_protocol_index = RegisterProtocol(p);
   });
 }
 
@@ -731,7 +848,7 @@ void RegisterNsheadProtocol() {
     p.support_server = true;
     p.support_client = true;
     p.name = "nshead";
-    RegisterProtocol(p);
+    g_nshead_protocol_index = RegisterProtocol(p);
   });
 }
 

@@ -132,3 +132,82 @@ def test_nova_pbrpc_client():
     assert seen["reserved"] == 7
     assert seen["version"] == 0
     srv.close()
+
+
+def test_ubrpc_client():
+    """ubrpc (reference policy/ubrpc2pb_protocol.cpp): nshead(version=1000)
+    + mcpack {header, content:[{service_name, method, id, params}]};
+    response correlated by content[0].id. The scripted server round-trips
+    real mcpack via the codec bindings."""
+    import socket
+    import threading
+    c = b.core.codecs
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(2)
+
+    def run():
+        conn, _ = srv.accept()
+        buf = b""
+        while True:
+            try:
+                chunk = conn.recv(65536)
+            except OSError:
+                return
+            if not chunk:
+                return
+            buf += chunk
+            while len(buf) >= 36:
+                _, ver, _, _, magic, _, blen = struct.unpack("<HHI16sIII", buf[:36])
+                if len(buf) < 36 + blen:
+                    break
+                assert magic == 0xFB709394 and ver == 1000
+                body = buf[36:36 + blen]
+                buf = buf[36 + blen:]
+                req = c.mcpack_loads(body)
+                c0 = req["content"][0]
+                assert c0["service_name"] == "Calc" and c0["method"] == "Mul"
+                reply = c.mcpack_dumps({"content": [{
+                    "id": c0["id"], "code": 0,
+                    "result": c0["params"]["req"] + b"*2"}]})
+                conn.sendall(struct.pack("<HHI16sIII", 0, 1000, 0, b"\0" * 16,
+                                         0xFB709394, 0, len(reply)) + reply)
+
+    threading.Thread(target=run, daemon=True).start()
+    port = srv.getsockname()[1]
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "ubrpc", "Calc.Mul", b"seven")
+    assert rc == 0, err
+    got = b.core.codecs.mcpack_loads(resp)
+    assert got["result"] == b"seven*2"
+    srv.close()
+
+
+def test_ubrpc_error_propagates():
+    import socket
+    import threading
+    c = b.core.codecs
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(2)
+
+    def run():
+        conn, _ = srv.accept()
+        buf = b""
+        while len(buf) < 36:
+            buf += conn.recv(65536)
+        _, _, _, _, _, _, blen = struct.unpack("<HHI16sIII", buf[:36])
+        while len(buf) < 36 + blen:
+            buf += conn.recv(65536)
+        req = c.mcpack_loads(buf[36:36 + blen])
+        rid = req["content"][0]["id"]
+        reply = c.mcpack_dumps({"content": [{"id": rid, "code": 2001,
+                                             "message": "no such idl method"}]})
+        conn.sendall(struct.pack("<HHI16sIII", 0, 1000, 0, b"\0" * 16,
+                                 0xFB709394, 0, len(reply)) + reply)
+
+    threading.Thread(target=run, daemon=True).start()
+    port = srv.getsockname()[1]
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "ubrpc", "X.Y", b"q")
+    assert rc == 2001
+    assert "no such idl method" in err
+    srv.close()
