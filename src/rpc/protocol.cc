@@ -1,5 +1,7 @@
 #include "rpc/protocol.h"
 
+#include "base/logging.h"
+
 #include <atomic>
 #include <cstring>
 
@@ -12,6 +14,8 @@ static std::atomic<int> g_nprotocols{0};
 
 int RegisterProtocol(const Protocol& p) {
   int n = g_nprotocols.load(std::memory_order_acquire);
+  CHECK(n < kMaxProtocols) << "protocol registry full (" << n
+                           << "): raise kMaxProtocols";  // silent -1 once cost rtmp its slot
   if (n >= kMaxProtocols) return -1;
   g_protocols[n] = p;
   g_nprotocols.store(n + 1, std::memory_order_release);

@@ -62,7 +62,7 @@ struct Protocol {
   const char* name = "unknown";
 };
 
-static const int kMaxProtocols = 16;
+static const int kMaxProtocols = 32;
 
 // Registration order = parse attempt order. Returns index or -1.
 int RegisterProtocol(const Protocol& p);
