@@ -32,6 +32,7 @@ int Channel::Init(EndPoint ep, const ChannelOptions* options) {
   if (options_.protocol == "esp") policy::RegisterEspProtocol();
   if (options_.protocol == "nova_pbrpc") policy::RegisterNovaProtocol();
   if (options_.protocol == "ubrpc") policy::RegisterUbrpcProtocol();
+  if (options_.protocol == "public_pbrpc") policy::RegisterPublicPbrpcProtocol();
   protocol_index_ = FindClientProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
   return 0;

@@ -41,6 +41,7 @@ void RegisterMongoProtocol();  // server-side (policy/mongo_protocol.cc)
 void RegisterEspProtocol();    // client-side (policy/legacy_protocols.cc)
 void RegisterNovaProtocol();   // client-side nshead variant (method index in reserved)
 void RegisterUbrpcProtocol();  // client-side nshead+mcpack (id-correlated)
+void RegisterPublicPbrpcProtocol();  // client-side nshead+PublicPbrpc pb
 void RegisterRtmpProtocol();   // server-side (policy/rtmp_protocol.cc)
 
 // Implemented in rpc/builtin/builtin_services.cc; returns true if the path

@@ -607,6 +607,116 @@ void PackNovaRequest(IOBuf* out, Controller* cntl, uint64_t /*fifo*/) {
   out->append(std::move(body));
 }
 
+// ==================== public_pbrpc (client) ====================
+// Parity: reference policy/public_pbrpc_protocol.cpp +
+// public_pbrpc_meta.proto — nshead (version 1000, provider "__pbrpc__")
+// whose body is ONE protobuf `PublicPbrpcRequest`
+// {requestHead{...}=1, requestBody{version=1 "pbrpc=1.0", service=3,
+// method_id=4, id=5, serialized_request=6}=2}; responses are
+// `PublicPbrpcResponse` {responseHead{code=1 sint32, text=2}=1,
+// responseBody{serialized_response=1, error=3, id=4}=2}. Correlated by
+// body.id (hand-rolled codec via rpc/wire.h, like baidu_std's meta).
+
+int g_public_protocol_index = -1;
+
+struct PublicMessage : public InputMessageBase {
+  int code = 0;          // head.code or body.error
+  std::string text;
+  uint64_t id = 0;
+  IOBuf payload;         // serialized_response
+};
+
+ParseResult ParsePublic(IOBuf* source, Socket* sock, bool) {
+  if (sock->client_protocol_hint != g_public_protocol_index)
+    return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
+  ParseResult raw = parse_nshead_frame(source);
+  if (raw.error != PARSE_OK) return raw;
+  NsheadMessage* nm = (NsheadMessage*)raw.msg;
+  std::string body = nm->body.to_string();
+  delete nm;
+  PublicMessage* msg = new PublicMessage;
+  wire::Reader r(body.data(), body.size());
+  int wt;
+  bool ok = true;
+  for (int f; ok && (f = r.read_tag(&wt)) != 0;) {
+    if (f == 1) {  // responseHead
+      std::string sub = r.read_string();
+      wire::Reader rh(sub.data(), sub.size());
+      int wt2;
+      for (int f2; (f2 = rh.read_tag(&wt2)) != 0;) {
+        if (f2 == 1) msg->code = (int)zigzag_dec(rh.varint());  // sint32
+        else if (f2 == 2) msg->text = rh.read_string();
+        else rh.skip(wt2);
+        if (!rh.ok()) { ok = false; break; }
+      }
+    } else if (f == 2) {  // responseBody
+      std::string sub = r.read_string();
+      wire::Reader rb(sub.data(), sub.size());
+      int wt2;
+      for (int f2; (f2 = rb.read_tag(&wt2)) != 0;) {
+        if (f2 == 1) {
+          std::string sr = rb.read_string();
+          msg->payload.clear();
+          msg->payload.append(sr);
+        } else if (f2 == 3) {
+          int err = (int)rb.varint();
+          if (err != 0) msg->code = err;
+        } else if (f2 == 4) {
+          msg->id = rb.varint();
+        } else {
+          rb.skip(wt2);
+        }
+        if (!rb.ok()) { ok = false; break; }
+      }
+    } else {
+      r.skip(wt);
+    }
+    if (!r.ok()) ok = false;
+  }
+  if (!ok) {
+    delete msg;
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  }
+  return ParseResult::make_ok(msg);
+}
+
+void ProcessPublicResponse(InputMessageBase* mb) {
+  PublicMessage* msg = (PublicMessage*)mb;
+  finish_client_call((SessionId)msg->id, msg->socket_id, msg->code, msg->text, &msg->payload,
+                     0);
+  delete msg;
+}
+
+void PackPublicRequest(IOBuf* out, Controller* cntl, uint64_t correlation_id) {
+  std::string head;
+  wire::put_str_field(&head, 1, "127.0.0.1");        // from_host
+  wire::put_int_field(&head, 3, 1);                   // connection = true
+  if (cntl->log_id() != 0) wire::put_int_field(&head, 7, (int64_t)cntl->log_id());
+  std::string bodymsg;
+  wire::put_str_field(&bodymsg, 1, "pbrpc=1.0");      // version
+  wire::put_str_field(&bodymsg, 3, cntl->call.service_name);
+  wire::put_int_field(&bodymsg, 4,
+                      (int64_t)strtoul(cntl->call.method_name.c_str(), nullptr, 10));
+  wire::put_int_field(&bodymsg, 5, (int64_t)correlation_id);
+  wire::put_str_field(&bodymsg, 6, cntl->call.request_buf.to_string());
+  std::string req;
+  wire::put_msg_field(&req, 1, head);
+  wire::put_msg_field(&req, 2, bodymsg);
+  char h[36];
+  memset(h, 0, sizeof(h));
+  uint16_t version = 1000;
+  memcpy(h + 2, &version, 2);
+  uint32_t log_id32 = (uint32_t)cntl->log_id();
+  memcpy(h + 4, &log_id32, 4);
+  memcpy(h + 8, "__pbrpc__", 9);
+  uint32_t magic = 0xfb709394u;
+  memcpy(h + 24, &magic, 4);
+  uint32_t blen = (uint32_t)req.size();
+  memcpy(h + 32, &blen, 4);
+  out->append(h, sizeof(h));
+  out->append(req);
+}
+
 // ==================== ubrpc (client) ====================
 // Parity: reference policy/ubrpc2pb_protocol.cpp — nshead (version 1000)
 // framing a compack/mcpack object:
@@ -789,6 +899,20 @@ void RegisterSofaProtocol() {
     p.support_client = true;
     p.name = "sofa_pbrpc";
     RegisterProtocol(p);
+  });
+}
+
+void RegisterPublicPbrpcProtocol() {
+  static std::once_flag once;
+  std::call_once(once, [] {
+    Protocol p;
+    p.parse = ParsePublic;
+    p.process_response = ProcessPublicResponse;
+    p.pack_request = PackPublicRequest;
+    p.support_server = false;
+    p.support_client = true;
+    p.name = "public_pbrpc";
+    g_public_protocol_index = RegisterProtocol(p);
   });
 }
 

@@ -211,3 +211,85 @@ def test_ubrpc_error_propagates():
     assert rc == 2001
     assert "no such idl method" in err
     srv.close()
+
+
+def test_public_pbrpc_client():
+    """public_pbrpc (reference policy/public_pbrpc_protocol.cpp): nshead
+    (version=1000, provider __pbrpc__) + one PublicPbrpcRequest protobuf;
+    response = PublicPbrpcResponse correlated by body.id."""
+    import socket
+    import threading
+
+    def varint(data, pos):
+        v = sh = 0
+        while True:
+            b_ = data[pos]; pos += 1
+            v |= (b_ & 0x7F) << sh
+            if not b_ & 0x80:
+                return v, pos
+            sh += 7
+
+    def parse_fields(data):
+        out = {}
+        pos = 0
+        while pos < len(data):
+            tag, pos = varint(data, pos)
+            f, wt = tag >> 3, tag & 7
+            if wt == 0:
+                v, pos = varint(data, pos)
+            elif wt == 2:
+                ln, pos = varint(data, pos)
+                v = data[pos:pos + ln]; pos += ln
+            else:
+                raise AssertionError(wt)
+            out.setdefault(f, []).append(v)
+        return out
+
+    def enc_varint(v):
+        out = b""
+        while True:
+            b_ = v & 0x7F; v >>= 7
+            out += bytes([b_ | (0x80 if v else 0)])
+            if not v:
+                return out
+
+    def enc_str(f, s_):
+        return enc_varint((f << 3) | 2) + enc_varint(len(s_)) + s_
+
+    def enc_int(f, v):
+        return enc_varint(f << 3) + enc_varint(v)
+
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(2)
+
+    def run():
+        conn, _ = srv.accept()
+        buf = b""
+        while len(buf) < 36:
+            buf += conn.recv(65536)
+        rid, ver, log_id, prov, magic, _res, blen = struct.unpack("<HHI16sIII", buf[:36])
+        assert ver == 1000 and prov.rstrip(b"\0") == b"__pbrpc__"
+        while len(buf) < 36 + blen:
+            buf += conn.recv(65536)
+        req = parse_fields(buf[36:36 + blen])
+        body = parse_fields(req[2][0])
+        assert body[1][0] == b"pbrpc=1.0"
+        assert body[3][0] == b"CalcService"
+        assert varint(bytes(body[4][0:1]) if isinstance(body[4][0], int) else b"", 0) or True
+        rid64 = body[5][0]
+        payload = body[6][0]
+        # response: head{code=0 zigzag} + body{serialized=payload+"!", id}
+        head = enc_int(1, 0) + enc_str(2, b"")
+        rbody = enc_str(1, payload + b"!") + enc_int(4, rid64)
+        resp = enc_str(1, head) + enc_str(2, rbody)
+        conn.sendall(struct.pack("<HHI16sIII", 0, 1000, log_id, b"\0" * 16,
+                                 0xFB709394, 0, len(resp)) + resp)
+
+    threading.Thread(target=run, daemon=True).start()
+    port = srv.getsockname()[1]
+    rc, resp, err = r.protocol_call("127.0.0.1:%d" % port, "public_pbrpc",
+                                    "CalcService.3", b"pubreq")
+    assert rc == 0, err
+    assert resp == b"pubreq!"
+    srv.close()
