@@ -66,6 +66,18 @@ class PyServer {
  public:
   PyServer() : server_(new Server) {}
 
+  void set_method_max_concurrency(const std::string& full_method, int limit) {
+    server_->SetMethodMaxConcurrency(full_method, limit);
+  }
+
+  void add_restful_mapping(const std::string& service, const std::string& mappings) {
+    auto it = services_.find(service);
+    if (it == services_.end()) throw std::runtime_error("unknown service " + service);
+    // re-register the mappings on the existing service object
+    if (server_->AddServiceRestfulOnly(it->second, mappings) != 0)
+      throw std::runtime_error("bad restful mapping: " + mappings);
+  }
+
   void add_method(const std::string& service, const std::string& method, py::function fn) {
     Service* svc;
     auto it = services_.find(service);
@@ -168,6 +180,8 @@ void bind_api(py::module_& m) {
       .def(py::init<>())
       .def("add_method", &PyServer::add_method, py::arg("service"), py::arg("method"),
            py::arg("handler"))
+      .def("set_method_max_concurrency", &PyServer::set_method_max_concurrency)
+      .def("add_restful_mapping", &PyServer::add_restful_mapping)
       .def("start", &PyServer::start, py::arg("port") = 0, py::arg("max_concurrency") = 0,
            py::arg("auth_user") = "", py::arg("auth_password") = "",
            py::arg("ssl_cert") = "", py::arg("ssl_key") = "",

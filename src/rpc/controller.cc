@@ -23,6 +23,7 @@ void Controller::Reset() {
   response_stream_id_ = 0;
   auth_context_ = nullptr;
   concurrency_counted_ = false;
+  method_gate_entered_ = false;
   call = Call();
 }
 

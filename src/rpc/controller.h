@@ -100,6 +100,7 @@ class Controller {
   uint64_t response_stream_id_ = 0;  // stream accepted by this server (StreamAccept)
   int64_t server_cid_ = 0;  // correlation id to echo back
   bool concurrency_counted_ = false;  // incremented server->concurrency (must decrement)
+  bool method_gate_entered_ = false;  // passed BeginMethod (must EndMethod)
   // Connection's verified auth result (owned by the server Socket);
   // nullptr when the server has no authenticator. (rpc/authenticator.h)
   const class AuthContext* auth_context_ = nullptr;

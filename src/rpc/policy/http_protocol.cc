@@ -291,12 +291,16 @@ void ProcessHttpRequest(InputMessageBase* msg_base) {
 
   bool handled = DispatchBuiltinService(server, msg->req, &resp);
   if (!handled && server != nullptr) {
-    // RPC over HTTP: /Service/Method
-    std::string path = msg->req.path;
-    if (!path.empty() && path[0] == '/') path = path.substr(1);
-    size_t slash = path.find('/');
-    std::string svc = slash == std::string::npos ? "" : path.substr(0, slash);
-    std::string method = slash == std::string::npos ? path : path.substr(slash + 1);
+    // restful mappings first (AddService(..., "/v1/x => Method")), then
+    // the default /Service/Method form
+    std::string svc, method;
+    if (!server->MapRestfulPath(msg->req.path, &svc, &method)) {
+      std::string path = msg->req.path;
+      if (!path.empty() && path[0] == '/') path = path.substr(1);
+      size_t slash = path.find('/');
+      svc = slash == std::string::npos ? "" : path.substr(0, slash);
+      method = slash == std::string::npos ? path : path.substr(slash + 1);
+    }
     const MethodFn* fn = server->FindMethod(svc, method);
     if (fn != nullptr) {
       Controller cntl;

@@ -244,6 +244,9 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
       cntl->concurrency_counted_) {
     cntl->server_->limiter()->OnResponse(cntl->ErrorCode(), monotonic_time_us() - start_us);
   }
+  if (cntl->server_ != nullptr && cntl->method_gate_entered_) {
+    cntl->server_->EndMethod(cntl->call.service_name, cntl->call.method_name);
+  }
   if (rpcz::enabled()) {
     rpcz::Span span;
     span.start_us = start_us;
@@ -355,6 +358,14 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
     done->Run();
     return;
   }
+  if (!server->BeginMethod(msg->meta.service_name, msg->meta.method_name)) {
+    cntl->SetFailed(ELIMIT, "reached method_max_concurrency of " +
+                                msg->meta.service_name + "." + msg->meta.method_name);
+    delete msg;
+    done->Run();
+    return;
+  }
+  cntl->method_gate_entered_ = true;
   if (server->options().interceptor) {
     int ec = 0;
     std::string etext;

@@ -31,6 +31,94 @@ int Server::AddService(Service* service, ServiceOwnership ownership) {
   return 0;
 }
 
+int Server::AddService(Service* service, ServiceOwnership ownership,
+                       const std::string& restful_mappings) {
+  if (AddService(service, ownership) != 0) return -1;
+  return AddServiceRestfulOnly(service, restful_mappings);
+}
+
+int Server::AddServiceRestfulOnly(Service* service, const std::string& restful_mappings) {
+  // "PATH => Method [, PATH => Method]..."
+  size_t pos = 0;
+  while (pos < restful_mappings.size()) {
+    size_t comma = restful_mappings.find(',', pos);
+    std::string item = restful_mappings.substr(
+        pos, comma == std::string::npos ? std::string::npos : comma - pos);
+    pos = comma == std::string::npos ? restful_mappings.size() : comma + 1;
+    size_t arrow = item.find("=>");
+    if (arrow == std::string::npos) return -1;
+    auto trim = [](std::string x) {
+      size_t a = x.find_first_not_of(" \t");
+      size_t b = x.find_last_not_of(" \t");
+      return a == std::string::npos ? std::string() : x.substr(a, b - a + 1);
+    };
+    std::string path = trim(item.substr(0, arrow));
+    std::string method = trim(item.substr(arrow + 2));
+    if (path.empty() || method.empty() || service->FindMethod(method) == nullptr) return -1;
+    if (path.size() >= 2 && path.compare(path.size() - 2, 2, "/*") == 0) {
+      restful_prefix_.emplace_back(path.substr(0, path.size() - 1), service->name(), method);
+    } else {
+      restful_exact_[path] = {service->name(), method};
+    }
+  }
+  return 0;
+}
+
+bool Server::MapRestfulPath(const std::string& path, std::string* service,
+                            std::string* method) const {
+  auto it = restful_exact_.find(path);
+  if (it != restful_exact_.end()) {
+    *service = it->second.first;
+    *method = it->second.second;
+    return true;
+  }
+  for (const auto& t : restful_prefix_) {
+    const std::string& prefix = std::get<0>(t);
+    if (path.compare(0, prefix.size(), prefix) == 0) {
+      *service = std::get<1>(t);
+      *method = std::get<2>(t);
+      return true;
+    }
+  }
+  return false;
+}
+
+void Server::SetMethodMaxConcurrency(const std::string& full_method, int32_t limit) {
+  std::lock_guard<std::mutex> lk(gates_mu_);
+  MethodGate*& g = method_gates_[full_method];
+  if (g == nullptr) g = new MethodGate;
+  g->max = limit;
+}
+
+bool Server::BeginMethod(const std::string& service, const std::string& method) {
+  MethodGate* g;
+  {
+    std::lock_guard<std::mutex> lk(gates_mu_);
+    if (method_gates_.empty()) return true;
+    auto it = method_gates_.find(service + "." + method);
+    if (it == method_gates_.end()) return true;
+    g = it->second;
+  }
+  if (g->max <= 0) return true;
+  if (g->current.fetch_add(1, std::memory_order_relaxed) >= g->max) {
+    g->current.fetch_sub(1, std::memory_order_relaxed);
+    return false;
+  }
+  return true;
+}
+
+void Server::EndMethod(const std::string& service, const std::string& method) {
+  MethodGate* g;
+  {
+    std::lock_guard<std::mutex> lk(gates_mu_);
+    if (method_gates_.empty()) return;
+    auto it = method_gates_.find(service + "." + method);
+    if (it == method_gates_.end()) return;
+    g = it->second;
+  }
+  if (g->max > 0) g->current.fetch_sub(1, std::memory_order_relaxed);
+}
+
 const MethodFn* Server::FindMethod(const std::string& service, const std::string& method,
                                    Service** svc_out) const {
   auto it = services_.find(service);

@@ -9,6 +9,8 @@
 #include <map>
 #include <memory>
 #include <string>
+#include <tuple>
+#include <vector>
 
 #include "base/endpoint.h"
 #include "rpc/controller.h"
@@ -103,6 +105,23 @@ class Server {
   ~Server();
 
   int AddService(Service* service, ServiceOwnership ownership);
+  // Restful URL mapping (parity: reference AddService(..., restful_mappings)
+  // "/v1/echo => Echo, /store/* => Put"): maps exact paths or trailing-*
+  // prefixes onto this service's methods for the HTTP/h2 protocols.
+  int AddService(Service* service, ServiceOwnership ownership,
+                 const std::string& restful_mappings);
+  // Adds restful mappings for a service already registered via AddService.
+  int AddServiceRestfulOnly(Service* service, const std::string& restful_mappings);
+  // Resolves a restful path; false if unmapped.
+  bool MapRestfulPath(const std::string& path, std::string* service,
+                      std::string* method) const;
+  // Per-method concurrency cap (parity: reference method_max_concurrency /
+  // MethodStatus gate). full_method = "Service.Method".
+  void SetMethodMaxConcurrency(const std::string& full_method, int32_t limit);
+  // Admission for one method; returns false (ELIMIT) when capped. Paired
+  // with EndMethod from the response path.
+  bool BeginMethod(const std::string& service, const std::string& method);
+  void EndMethod(const std::string& service, const std::string& method);
   int Start(int port, const ServerOptions* opt);  // port 0 = pick free port
   int Start(const EndPoint& ep, const ServerOptions* opt);
   int Stop(int wait_ms = 0);
@@ -136,6 +155,14 @@ class Server {
   std::map<Service*, ServiceOwnership> ownership_;
   std::mutex status_mu_;
   std::map<std::string, var::LatencyRecorder*> method_status_;
+  std::map<std::string, std::pair<std::string, std::string>> restful_exact_;
+  std::vector<std::tuple<std::string, std::string, std::string>> restful_prefix_;
+  struct MethodGate {
+    std::atomic<int32_t> current{0};
+    int32_t max = 0;
+  };
+  mutable std::mutex gates_mu_;
+  std::map<std::string, MethodGate*> method_gates_;
   ServerOptions options_;
   EndPoint listen_ep_;
   SocketId listen_socket_ = 0;
