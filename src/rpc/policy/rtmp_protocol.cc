@@ -42,6 +42,7 @@ struct RtmpCtx {
   std::string app;
   std::string publishing;  // stream name if this connection publishes
   std::string playing;     // stream name if this connection plays
+  SocketId socket_id = 0;  // for hub cleanup at recycle
 };
 
 // ---- relay hub ----
@@ -112,9 +113,12 @@ ParseResult ParseRtmp(IOBuf* source, Socket* sock, bool) {
     source->copy_to(&c0, 1, 0);
     if (c0 != 0x03) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
     ctx = new RtmpCtx;
+    ctx->socket_id = sock->id();
     sock->protocol_ctx = ctx;
     sock->protocol_ctx_deleter = [](void* p) {
       RtmpCtx* c = (RtmpCtx*)p;
+      // a player that vanished without deleteStream must leave the hub
+      if (!c->playing.empty()) hub_remove_player(c->playing, c->socket_id);
       delete c;
     };
     sock->protocol_ctx_owner = g_rtmp_protocol_index;
