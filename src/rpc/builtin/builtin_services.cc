@@ -3,6 +3,8 @@
 // server.cpp:501-603). This file serves the core set: /index, /status,
 // /vars, /flags, /health, /version, /connections, /protobufs, /fibers
 // (≙ /bthreads), /memory, /brpc_metrics (Prometheus), /rpcz (spans).
+#include <string.h>
+
 #include <sstream>
 
 #include "base/flags.h"
@@ -18,6 +20,7 @@
 
 namespace bam {
 std::string dump_fiber_stacks(int max_fibers);  // fiber/tracer.cc
+std::string CpuProfile(int seconds, int hz);    // rpc/cpu_profiler.cc
 namespace policy {
 
 namespace {
@@ -29,8 +32,8 @@ void page_index(HttpResponse* resp) {
   std::ostringstream os;
   os << "<html><head><title>brpc_amd</title></head><body><h1>brpc_amd server</h1><ul>";
   const char* pages[] = {"status", "vars",   "flags",  "health",       "version",
-                         "connections", "protobufs", "fibers", "memory", "rpcz",
-                         "brpc_metrics"};
+                         "connections", "protobufs", "fibers", "memory", "threads",
+                         "hotspots/cpu", "rpcz", "brpc_metrics"};
   for (const char* p : pages) os << "<li><a href=\"/" << p << "\">/" << p << "</a></li>";
   os << "</ul></body></html>";
   resp->body.append(os.str());
@@ -140,6 +143,24 @@ void page_fibers(const HttpRequest& req, HttpResponse* resp) {
   resp->body.append(os.str());
 }
 
+// /threads: pthread inventory (parity: builtin/threads_service.cpp).
+void page_threads(HttpResponse* resp) {
+  std::ostringstream os;
+  FILE* f = fopen("/proc/self/status", "r");
+  if (f != nullptr) {
+    char line[256];
+    while (fgets(line, sizeof(line), f) != nullptr) {
+      if (strncmp(line, "Threads:", 8) == 0 || strncmp(line, "VmRSS:", 6) == 0 ||
+          strncmp(line, "VmSize:", 7) == 0) {
+        os << line;
+      }
+    }
+    fclose(f);
+  }
+  os << "fiber_workers: " << fiber_get_concurrency() << "\n";
+  resp->body.append(os.str());
+}
+
 void page_memory(HttpResponse* resp) {
   std::ostringstream os;
   os << "iobuf_block_count: " << IOBuf::block_count() << "\n";
@@ -195,6 +216,14 @@ bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse
     page_fibers(req, resp);
   } else if (p == "/memory") {
     page_memory(resp);
+  } else if (p == "/threads") {
+    page_threads(resp);
+  } else if (p == "/hotspots/cpu" || p == "/hotspots") {
+    // parity: builtin/hotspots_service.cpp, self-contained SIGPROF sampler
+    int seconds = 1;
+    auto it = req.query.find("seconds");
+    if (it != req.query.end()) seconds = atoi(it->second.c_str());
+    resp->body.append(CpuProfile(seconds, 200));
   } else if (p == "/brpc_metrics" || p == "/metrics") {
     page_metrics(resp);
   } else if (p == "/rpcz") {

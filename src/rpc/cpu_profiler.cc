@@ -1,0 +1,132 @@
+// brpc_amd: built-in CPU hotspot sampler.
+// Parity: reference builtin/hotspots_service.cpp (gperftools
+// ProfilerStart, weak-linked). Self-contained redesign — no gperftools in
+// the image: an ITIMER_PROF / SIGPROF handler captures backtraces into a
+// preallocated lock-free slot array for N seconds; the report aggregates
+// by symbolized frame (dladdr) with self/cumulative counts. Exposed at
+// /hotspots/cpu?seconds=N (builtin_services.cc).
+#include <dlfcn.h>
+#include <execinfo.h>
+#include <signal.h>
+#include <string.h>
+#include <sys/time.h>
+
+#include <algorithm>
+#include <atomic>
+#include <map>
+#include <mutex>
+#include <sstream>
+#include <string>
+#include <vector>
+
+#include "fiber/fiber.h"
+
+namespace bam {
+
+namespace {
+
+constexpr int kMaxDepth = 24;
+constexpr int kMaxSamples = 40000;
+
+struct Sample {
+  void* frames[kMaxDepth];
+  int depth;
+};
+
+Sample* g_samples = nullptr;          // preallocated outside the handler
+std::atomic<int> g_sample_idx{0};
+std::atomic<bool> g_active{false};
+
+void sigprof_handler(int, siginfo_t*, void*) {
+  if (!g_active.load(std::memory_order_relaxed)) return;
+  int i = g_sample_idx.fetch_add(1, std::memory_order_relaxed);
+  if (i >= kMaxSamples) return;
+  // backtrace() is not strictly async-signal-safe but is the standard
+  // practice for SIGPROF samplers (gperftools does the equivalent via
+  // libunwind); frames land in preallocated memory, no malloc here after
+  // the first call (primed in CpuProfile below).
+  g_samples[i].depth = backtrace(g_samples[i].frames, kMaxDepth);
+}
+
+std::string frame_name(void* ip) {
+  Dl_info info;
+  if (dladdr(ip, &info) != 0 && info.dli_sname != nullptr) return info.dli_sname;
+  char buf[32];
+  snprintf(buf, sizeof(buf), "%p", ip);
+  return buf;
+}
+
+}  // namespace
+
+// Samples the process for `seconds` (clamped to [1,30]) at `hz` and
+// returns a text report. Serializes concurrent profile requests.
+std::string CpuProfile(int seconds, int hz) {
+  static std::mutex mu;
+  std::lock_guard<std::mutex> lk(mu);
+  if (seconds < 1) seconds = 1;
+  if (seconds > 30) seconds = 30;
+  if (hz < 10) hz = 10;
+  if (hz > 1000) hz = 1000;
+  if (g_samples == nullptr) g_samples = new Sample[kMaxSamples];
+  {
+    // prime backtrace()'s lazy libgcc initialization outside the handler
+    void* prime[4];
+    backtrace(prime, 4);
+  }
+  g_sample_idx.store(0, std::memory_order_relaxed);
+
+  struct sigaction sa, old_sa;
+  memset(&sa, 0, sizeof(sa));
+  sa.sa_sigaction = sigprof_handler;
+  sa.sa_flags = SA_SIGINFO | SA_RESTART;
+  sigaction(SIGPROF, &sa, &old_sa);
+  struct itimerval it, old_it;
+  it.it_interval.tv_sec = 0;
+  it.it_interval.tv_usec = 1000000 / hz;
+  it.it_value = it.it_interval;
+  g_active.store(true, std::memory_order_release);
+  setitimer(ITIMER_PROF, &it, &old_it);
+
+  fiber_usleep((uint64_t)seconds * 1000000);
+
+  g_active.store(false, std::memory_order_release);
+  memset(&it, 0, sizeof(it));
+  setitimer(ITIMER_PROF, &it, nullptr);
+  sigaction(SIGPROF, &old_sa, nullptr);
+
+  const int n = std::min(g_sample_idx.load(std::memory_order_relaxed), kMaxSamples);
+  std::map<std::string, std::pair<int, int>> agg;  // name -> {self, cumulative}
+  for (int i = 0; i < n; ++i) {
+    const Sample& s = g_samples[i];
+    // frames[0..1] are the handler itself; self = first frame past them
+    bool first = true;
+    std::map<std::string, bool> seen;  // cumulative counts once per sample
+    for (int d = 2; d < s.depth; ++d) {
+      std::string name = frame_name(s.frames[d]);
+      if (first) {
+        ++agg[name].first;
+        first = false;
+      }
+      if (!seen[name]) {
+        ++agg[name].second;
+        seen[name] = true;
+      }
+    }
+  }
+  std::vector<std::pair<std::string, std::pair<int, int>>> rows(agg.begin(), agg.end());
+  std::sort(rows.begin(), rows.end(),
+            [](const auto& a, const auto& b) { return a.second.first > b.second.first; });
+  std::ostringstream os;
+  os << "cpu profile: " << n << " samples @ " << hz << " Hz over " << seconds << " s\n";
+  os << "  self   cum  symbol\n";
+  int emitted = 0;
+  for (const auto& r : rows) {
+    if (r.second.first == 0 && emitted > 60) break;
+    os << "  " << r.second.first << "  " << r.second.second << "  " << r.first << "\n";
+    if (++emitted >= 100) break;
+  }
+  if (n == 0) os << "(no samples: process mostly idle or blocked — ITIMER_PROF counts CPU time)\n";
+  return os.str();
+}
+
+}  // namespace bam

@@ -121,3 +121,34 @@ def test_fiber_stack_tracer(base):
     assert "fiber #" in body
     # the parked fiber's stack should include the butex/usleep path
     assert "butex" in body or "usleep" in body or "sched" in body, body[:2000]
+
+
+def test_threads_page(base):
+    code, body = get(base + "/threads")
+    assert code == 200
+    assert "Threads:" in body
+    assert "fiber_workers:" in body
+
+
+def test_hotspots_cpu_profile(base):
+    import threading
+    # burn CPU in the background so the PROF timer has something to sample
+    stop = [False]
+
+    def burn():
+        x = 0
+        while not stop[0]:
+            x = (x * 1103515245 + 12345) & 0xFFFFFFFF
+
+    ts = [threading.Thread(target=burn) for _ in range(2)]
+    for t in ts:
+        t.start()
+    try:
+        code, body = get(base + "/hotspots/cpu?seconds=1", timeout=15)
+        assert code == 200
+        assert "cpu profile:" in body
+        assert "samples @" in body
+    finally:
+        stop[0] = True
+        for t in ts:
+            t.join()
