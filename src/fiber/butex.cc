@@ -125,6 +125,15 @@ void butex_destroy(std::atomic<int>* v) {
   delete b;
 }
 
+// Contention accounting (parity: the reference's contention profiler —
+// instrumented bthread_mutex feeding /hotspots/contention). We keep
+// process-global park counts + parked time, exposed as vars
+// fiber_butex_{waits,wait_us} (registered in server.cc).
+static std::atomic<int64_t> g_butex_waits{0};
+static std::atomic<int64_t> g_butex_wait_us{0};
+int64_t butex_total_waits() { return g_butex_waits.load(std::memory_order_relaxed); }
+int64_t butex_total_wait_us() { return g_butex_wait_us.load(std::memory_order_relaxed); }
+
 int butex_wait(std::atomic<int>* v, int expected, const int64_t* abstime_us) {
   Butex* b = container_of_value(v);
   if (b->value.load(std::memory_order_acquire) != expected) {
@@ -147,8 +156,11 @@ int butex_wait(std::atomic<int>* v, int expected, const int64_t* abstime_us) {
       w.has_abstime = true;
       w.abstime_us = *abstime_us;
     }
+    g_butex_waits.fetch_add(1, std::memory_order_relaxed);
+    const int64_t park_t0 = monotonic_time_us();
     g->sched(remained_add_waiter, &w);
     // Resumed (possibly on another worker).
+    g_butex_wait_us.fetch_add(monotonic_time_us() - park_t0, std::memory_order_relaxed);
     if (w.timer_id != 0) timer_delete(w.timer_id);
     if (w.state.load(std::memory_order_acquire) == 2) {
       errno = ETIMEDOUT;
@@ -157,6 +169,8 @@ int butex_wait(std::atomic<int>* v, int expected, const int64_t* abstime_us) {
     return 0;
   }
   // Pthread path: futex on a private per-waiter word.
+  g_butex_waits.fetch_add(1, std::memory_order_relaxed);
+  const int64_t park_t0 = monotonic_time_us();
   ButexWaiter w;
   w.type = ButexWaiter::PTHREAD;
   w.expected = expected;
@@ -187,6 +201,7 @@ int butex_wait(std::atomic<int>* v, int expected, const int64_t* abstime_us) {
           }
         }
         if (removed) {
+          g_butex_wait_us.fetch_add(monotonic_time_us() - park_t0, std::memory_order_relaxed);
           errno = ETIMEDOUT;
           return -1;
         }
@@ -200,6 +215,7 @@ int butex_wait(std::atomic<int>* v, int expected, const int64_t* abstime_us) {
     }
     sys_futex(&w.sig, FUTEX_WAIT, 0, prel);
   }
+  g_butex_wait_us.fetch_add(monotonic_time_us() - park_t0, std::memory_order_relaxed);
   if (w.state.load(std::memory_order_acquire) == 2) {
     errno = ETIMEDOUT;
     return -1;

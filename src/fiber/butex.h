@@ -20,6 +20,10 @@ void butex_destroy(std::atomic<int>* b);
 // ETIMEDOUT if abstime (CLOCK_MONOTONIC us) passed.
 int butex_wait(std::atomic<int>* b, int expected, const int64_t* abstime_us = nullptr);
 
+// Contention totals (fiber parks + parked microseconds).
+int64_t butex_total_waits();
+int64_t butex_total_wait_us();
+
 int butex_wake(std::atomic<int>* b);      // wake one; returns #woken
 int butex_wake_all(std::atomic<int>* b);  // returns #woken
 

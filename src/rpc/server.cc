@@ -1,6 +1,7 @@
 #include "rpc/server.h"
 
 #include "rpc/ssl_util.h"
+#include "fiber/butex.h"
 #include "rpc/concurrency_limiter.h"
 
 #include <sys/socket.h>
@@ -177,6 +178,13 @@ int Server::Start(int port, const ServerOptions* opt) {
 int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   if (IsRunning()) return -1;
   if (opt != nullptr) options_ = *opt;
+  // process-wide contention vars (≙ contention profiler surface)
+  static var::PassiveStatus* g_waits = new var::PassiveStatus(
+      "fiber_butex_waits", [] { return std::to_string(butex_total_waits()); });
+  static var::PassiveStatus* g_wait_us = new var::PassiveStatus(
+      "fiber_butex_wait_us", [] { return std::to_string(butex_total_wait_us()); });
+  (void)g_waits;
+  (void)g_wait_us;
   if (limiter_ == nullptr) {
     limiter_ = ConcurrencyLimiter::Create(options_.adaptive_max_concurrency);
   }

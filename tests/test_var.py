@@ -23,3 +23,22 @@ def test_server_method_status_exposed():
     assert "_count" in dump
     count = v.describe(f"rpc_server_{port}_EchoService.Echo_count")
     assert count is not None and int(count) >= 10
+
+
+def test_contention_vars_exposed():
+    """Contention surface (≙ reference contention profiler feed): butex
+    park counts + parked time as process vars."""
+    import urllib.request
+    srv = b.Server()
+    srv.add_method("CV", "Echo", lambda req, att: (req, b""))
+    port = srv.start(0)
+    ch = b.Channel("127.0.0.1:%d" % port)
+    for i in range(30):
+        ch.call("CV.Echo", b"x")
+    body = urllib.request.urlopen(
+        "http://127.0.0.1:%d/vars/fiber_butex" % port, timeout=5).read().decode()
+    assert "fiber_butex_waits" in body
+    assert "fiber_butex_wait_us" in body
+    waits = int([l for l in body.splitlines() if l.startswith("fiber_butex_waits")][0].split(":")[1])
+    assert waits > 0
+    srv.stop()
