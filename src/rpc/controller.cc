@@ -1,5 +1,7 @@
 #include "rpc/controller.h"
 
+#include "rpc/server.h"
+
 namespace bam {
 
 void Controller::Reset() {
@@ -28,5 +30,27 @@ void Controller::Reset() {
 }
 
 Controller::~Controller() {}
+
+
+
+void* Controller::session_local_data() {
+  if (server_ == nullptr || !server_->options().session_local_data_factory) return nullptr;
+  SocketUniquePtr sock;
+  if (Socket::Address(server_socket_, &sock) != 0) return nullptr;
+  void* d = sock->session_local_data.load(std::memory_order_acquire);
+  if (d != nullptr) return d;
+  void* fresh = server_->options().session_local_data_factory();
+  if (fresh == nullptr) return nullptr;
+  void* expected = nullptr;
+  sock->session_local_deleter = server_->options().session_local_data_deleter;
+  if (sock->session_local_data.compare_exchange_strong(expected, fresh,
+                                                       std::memory_order_acq_rel)) {
+    return fresh;
+  }
+  // lost the install race
+  if (server_->options().session_local_data_deleter)
+    server_->options().session_local_data_deleter(fresh);
+  return expected;
+}
 
 }  // namespace bam

@@ -70,6 +70,11 @@ class Controller {
 
   bool is_server_side() const { return server_ != nullptr; }
 
+  // Per-connection user state (reference Controller::session_local_data):
+  // lazily created from ServerOptions::session_local_data_factory; same
+  // pointer for every request on one connection. nullptr without a factory.
+  void* session_local_data();
+
   // ---- internals (channel / protocol / server plumbing) ----
   struct Call {
     SessionId cid = 0;              // whole-call session

@@ -119,6 +119,27 @@ int protocol_call(const std::string& addr, const std::string& protocol,
   return 0;
 }
 
+// ---- session-local data (per-connection counter) ----
+int start_session_counter_server() {
+  Server* server = new Server;
+  Service* svc = new Service("Sess");
+  svc->AddMethod("Count", [](Controller* cntl, const IOBuf&, IOBuf* resp, Closure* done) {
+    int64_t* n = (int64_t*)cntl->session_local_data();
+    if (n == nullptr) {
+      cntl->SetFailed(EINTERNAL, "no session data");
+    } else {
+      resp->append(std::to_string(++*n));
+    }
+    done->Run();
+  });
+  server->AddService(svc, SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  opts.session_local_data_factory = [] { return (void*)new int64_t(0); };
+  opts.session_local_data_deleter = [](void* p) { delete (int64_t*)p; };
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
 // ---- rtmp media server ----
 int start_rtmp_server() {
   Server* server = new Server;

@@ -95,6 +95,12 @@ struct ServerOptions {
   // Adaptive concurrency (rpc/concurrency_limiter.h): "" | "auto" |
   // "timeout:<ms>" | "<n>". Applies on top of max_concurrency.
   std::string adaptive_max_concurrency;
+  // Per-connection user state (parity: reference
+  // ServerOptions::session_local_data_factory): created lazily on first
+  // Controller::session_local_data() of a connection, destroyed when the
+  // connection recycles.
+  std::function<void*()> session_local_data_factory;
+  std::function<void(void*)> session_local_data_deleter;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)

@@ -91,6 +91,8 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   s->input_events_.store(0, std::memory_order_relaxed);
   s->ssl_ = nullptr;
   s->ssl_state_.store(0, std::memory_order_relaxed);
+  s->session_local_data.store(nullptr, std::memory_order_relaxed);
+  s->session_local_deleter = nullptr;
   s->write_head_.store(nullptr, std::memory_order_relaxed);
   s->in_bytes = 0;
   s->out_bytes = 0;
@@ -192,6 +194,10 @@ void Socket::Recycle() {
   }
   read_buf_.clear();
   delete (AuthContext*)auth_ctx_.exchange(nullptr, std::memory_order_acq_rel);
+  if (void* sld = session_local_data.exchange(nullptr, std::memory_order_acq_rel)) {
+    if (session_local_deleter) session_local_deleter(sld);
+  }
+  session_local_deleter = nullptr;
   ssl::FreeSsl(ssl_);
   ssl_ = nullptr;
   ssl_state_.store(0, std::memory_order_relaxed);

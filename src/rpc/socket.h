@@ -205,6 +205,14 @@ class Socket {
   EndPoint local_side_;
   void* user_ = nullptr;
   std::atomic<void*> auth_ctx_{nullptr};
+
+ public:
+  // Per-connection user data (≙ reference session_local_data): installed
+  // by Controller::session_local_data(), freed at recycle via the deleter.
+  std::atomic<void*> session_local_data{nullptr};
+  std::function<void(void*)> session_local_deleter;
+
+ private:
   void* ssl_ = nullptr;                 // SSL* when TLS is enabled
   std::atomic<int> ssl_state_{0};       // 0 off, 1 handshaking, 2 ready
   std::mutex ssl_hs_mu_;                // serializes handshake stepping

@@ -97,3 +97,15 @@ def test_unknown_method_raises(py_server):
     with pytest.raises(b.RpcError) as ei:
         ch.call("PySvc.Nope", b"x")
     assert ei.value.args[0] == 1002  # ENOMETHOD
+
+
+def test_session_local_data_per_connection():
+    """≙ reference ServerOptions::session_local_data_factory +
+    Controller::session_local_data: one lazily-created object per
+    connection, destroyed at connection recycle."""
+    r = b.core.rpc
+    port = r.start_session_counter_server()
+    # same (cached single) connection: counter grows
+    vals = [r.protocol_call("127.0.0.1:%d" % port, "std", "Sess.Count", b"")[1]
+            for _ in range(3)]
+    assert vals == [b"1", b"2", b"3"], vals
