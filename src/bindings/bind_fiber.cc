@@ -12,6 +12,8 @@ bool butex_wake_test();
 bool butex_timeout_test();
 int64_t mutex_test(int nfibers, int iters);
 bool countdown_test(int n);
+bool semaphore_test();
+bool rwlock_test(int nreaders, int nwriters, int iters);
 bool timer_test();
 bool fiber_key_test();
 }  // namespace selftest
@@ -29,6 +31,10 @@ void bind_fiber(py::module_& m) {
   f.def("butex_timeout_test", &bam::selftest::butex_timeout_test,
         py::call_guard<py::gil_scoped_release>());
   f.def("mutex_test", &bam::selftest::mutex_test, py::call_guard<py::gil_scoped_release>());
+  f.def("semaphore_test", &bam::selftest::semaphore_test,
+        py::call_guard<py::gil_scoped_release>());
+  f.def("rwlock_test", &bam::selftest::rwlock_test,
+        py::call_guard<py::gil_scoped_release>());
   f.def("countdown_test", &bam::selftest::countdown_test,
         py::call_guard<py::gil_scoped_release>());
   f.def("timer_test", &bam::selftest::timer_test, py::call_guard<py::gil_scoped_release>());

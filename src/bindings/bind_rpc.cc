@@ -17,6 +17,7 @@ struct BenchResult {
 };
 int start_echo_server(int port);
 int start_nshead_server();
+int retry_policy_test(int max_retry);
 int start_session_counter_server();
 int start_rtmp_server();
 int start_mongo_echo_server();
@@ -98,6 +99,8 @@ void bind_rpc(py::module_& m) {
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("retry_policy_test", &bam::rpctest::retry_policy_test,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_session_counter_server", &bam::rpctest::start_session_counter_server,
         py::call_guard<py::gil_scoped_release>());

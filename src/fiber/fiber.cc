@@ -1,6 +1,7 @@
 #include "fiber/fiber.h"
 
 #include <errno.h>
+#include <poll.h>
 #include <unistd.h>
 
 #include "base/time.h"
@@ -123,6 +124,26 @@ int64_t fiber_count_created() {
 }
 int64_t fiber_count_active() {
   return TaskControl::singleton()->nfibers_active.load(std::memory_order_relaxed);
+}
+
+
+
+int fiber_fd_wait(int fd, short events, int timeout_ms) {
+  const int64_t deadline = monotonic_time_us() + (int64_t)timeout_ms * 1000;
+  struct pollfd pfd;
+  for (;;) {
+    pfd.fd = fd;
+    pfd.events = events;
+    pfd.revents = 0;
+    int rc = ::poll(&pfd, 1, 0);
+    if (rc > 0) return (pfd.revents & (POLLERR | POLLNVAL)) ? -1 : 0;
+    if (rc < 0 && errno != EINTR) return -1;
+    if (timeout_ms >= 0 && monotonic_time_us() >= deadline) {
+      errno = ETIMEDOUT;
+      return -1;
+    }
+    fiber_usleep(500);
+  }
 }
 
 }  // namespace bam

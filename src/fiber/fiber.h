@@ -32,6 +32,13 @@ bool fiber_exists(fiber_t tid);
 
 int fiber_yield();
 int fiber_usleep(uint64_t us);
+
+// Waits for readability(POLLIN)/writability(POLLOUT) of fd without
+// blocking the worker pthread (parity: reference bthread_fd_wait).
+// Polls non-blockingly and parks the fiber between checks (0.5 ms
+// granularity — the RPC runtime's own fds use edge-triggered epoll; this
+// API is for user fds). Returns 0 when ready, -1 on timeout/error.
+int fiber_fd_wait(int fd, short events, int timeout_ms);
 fiber_t fiber_self();
 bool is_running_on_fiber();
 

@@ -68,6 +68,80 @@ bool urgent_test() {
   return a.child_ran_first;
 }
 
+// ---- semaphore + rwlock ----
+namespace {
+struct SemArg {
+  FiberSemaphore* sem;
+  std::atomic<int>* acquired;
+};
+void sem_fn(void* raw) {
+  SemArg* a = (SemArg*)raw;
+  a->sem->acquire();
+  a->acquired->fetch_add(1);
+}
+struct RwArg {
+  FiberRWLock* rw;
+  std::atomic<int>* readers_in;
+  std::atomic<int>* max_readers;
+  std::atomic<long long>* counter;
+  int iters;
+  bool writer;
+};
+void rw_fn(void* raw) {
+  RwArg* a = (RwArg*)raw;
+  for (int i = 0; i < a->iters; ++i) {
+    if (a->writer) {
+      a->rw->wrlock();
+      long long v = a->counter->load(std::memory_order_relaxed);
+      a->counter->store(v + 1, std::memory_order_relaxed);
+      if (a->readers_in->load() != 0) a->max_readers->store(-1);  // reader inside write CS!
+      a->rw->unlock();
+    } else {
+      a->rw->rdlock();
+      int in = a->readers_in->fetch_add(1) + 1;
+      int m = a->max_readers->load();
+      while (m >= 0 && in > m && !a->max_readers->compare_exchange_weak(m, in)) {
+      }
+      a->readers_in->fetch_sub(1);
+      a->rw->unlock();
+    }
+  }
+}
+}  // namespace
+
+bool semaphore_test() {
+  FiberSemaphore sem(0);
+  std::atomic<int> acquired{0};
+  fiber_t t[4];
+  SemArg a{&sem, &acquired};
+  for (int i = 0; i < 4; ++i) fiber_start_background(&t[i], sem_fn, &a);
+  fiber_usleep(20000);
+  if (acquired.load() != 0) return false;  // nothing released yet
+  sem.release(2);
+  fiber_usleep(50000);
+  if (acquired.load() != 2) return false;
+  sem.release(2);
+  for (int i = 0; i < 4; ++i) fiber_join(t[i]);
+  return acquired.load() == 4 && !sem.try_acquire();
+}
+
+bool rwlock_test(int nreaders, int nwriters, int iters) {
+  FiberRWLock rw;
+  std::atomic<int> readers_in{0}, max_readers{0};
+  std::atomic<long long> counter{0};
+  std::vector<fiber_t> tids;
+  std::vector<RwArg> args(nreaders + nwriters);
+  for (int i = 0; i < nreaders + nwriters; ++i) {
+    args[i] = RwArg{&rw, &readers_in, &max_readers, &counter, iters, i < nwriters};
+    fiber_t t;
+    fiber_start_background(&t, rw_fn, &args[i]);
+    tids.push_back(t);
+  }
+  for (fiber_t t : tids) fiber_join(t);
+  // writers mutually exclusive => counter exact; readers overlapped at least once
+  return counter.load() == (long long)nwriters * iters && max_readers.load() > 1;
+}
+
 // ---- usleep accuracy ----
 namespace {
 struct SleepArg {

@@ -125,4 +125,87 @@ class CountdownEvent {
   std::atomic<int>* word_;
 };
 
+// Counting semaphore (parity: reference bthread/semaphore — butex-backed;
+// blocking acquire parks the fiber, never the worker pthread).
+class FiberSemaphore {
+ public:
+  explicit FiberSemaphore(int initial = 0) : count_(butex_create()) {
+    count_->store(initial, std::memory_order_relaxed);
+  }
+  ~FiberSemaphore() { butex_destroy(count_); }
+
+  void release(int n = 1) {
+    count_->fetch_add(n, std::memory_order_release);
+    if (n == 1) butex_wake(count_);
+    else butex_wake_all(count_);
+  }
+
+  void acquire() {
+    for (;;) {
+      int v = count_->load(std::memory_order_acquire);
+      if (v > 0 && count_->compare_exchange_weak(v, v - 1, std::memory_order_acq_rel)) return;
+      if (v <= 0) butex_wait(count_, v, nullptr);
+    }
+  }
+
+  bool try_acquire() {
+    int v = count_->load(std::memory_order_acquire);
+    return v > 0 && count_->compare_exchange_strong(v, v - 1, std::memory_order_acq_rel);
+  }
+
+ private:
+  std::atomic<int>* count_;
+};
+
+// Reader-writer lock (parity: reference bthread/rwlock). Writer-preferring:
+// state = -1 writer held, 0 free, >0 reader count; writers_waiting_ blocks
+// new readers so writers cannot starve.
+class FiberRWLock {
+ public:
+  FiberRWLock() : state_(butex_create()), writers_waiting_(0) {
+    state_->store(0, std::memory_order_relaxed);
+  }
+  ~FiberRWLock() { butex_destroy(state_); }
+
+  void rdlock() {
+    for (;;) {
+      int v = state_->load(std::memory_order_acquire);
+      if (v >= 0 && writers_waiting_.load(std::memory_order_acquire) == 0) {
+        if (state_->compare_exchange_weak(v, v + 1, std::memory_order_acq_rel)) return;
+        continue;
+      }
+      butex_wait(state_, v, nullptr);
+    }
+  }
+
+  void wrlock() {
+    writers_waiting_.fetch_add(1, std::memory_order_acq_rel);
+    for (;;) {
+      int v = state_->load(std::memory_order_acquire);
+      if (v == 0 && state_->compare_exchange_weak(v, -1, std::memory_order_acq_rel)) {
+        writers_waiting_.fetch_sub(1, std::memory_order_acq_rel);
+        return;
+      }
+      if (v != 0) butex_wait(state_, v, nullptr);
+    }
+  }
+
+  void unlock() {
+    int v = state_->load(std::memory_order_acquire);
+    if (v == -1) {
+      state_->store(0, std::memory_order_release);
+    } else {
+      if (state_->fetch_sub(1, std::memory_order_acq_rel) != 1) {
+        // readers remain; only the last reader wakes writers
+        return;
+      }
+    }
+    butex_wake_all(state_);
+  }
+
+ private:
+  std::atomic<int>* state_;
+  std::atomic<int> writers_waiting_;
+};
+
 }  // namespace bam

@@ -125,7 +125,10 @@ static int HandleSessionError(SessionId id, void* data, int error_code) {
     session_unlock(id);
     return 0;
   }
-  if (error_code != ERPCTIMEDOUT && cntl->retry_count_ < cntl->max_retry_) {
+  const bool policy_says_retry =
+      cntl->call.retry_policy ? cntl->call.retry_policy(error_code, cntl->retry_count_)
+                              : error_code != ERPCTIMEDOUT;
+  if (policy_says_retry && cntl->retry_count_ < cntl->max_retry_) {
     ++cntl->retry_count_;
     session_bump_slot(id);
     if (cntl->call.lb != nullptr && cntl->call.server_ep.port != 0) {
@@ -224,6 +227,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->call.protocol_index = protocol_index_;
   cntl->call.auth = options_.auth;
   cntl->call.ssl = options_.ssl;
+  cntl->call.retry_policy = options_.retry_policy;
   if (options_.connection_type == "pooled") {
     static std::atomic<uint32_t> rr{0};
     cntl->call.connection_shard =

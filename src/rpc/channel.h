@@ -31,6 +31,11 @@ struct ChannelOptions {
   // TLS to the server (certificate verification off, like the reference's
   // default ChannelSSLOptions).
   bool ssl = false;
+  // Custom retry decision (parity: reference brpc/retry_policy.h
+  // RetryPolicy::DoRetry): called with the failing error_code and the
+  // attempt index (0-based); return false to stop retrying. nullptr =
+  // default policy (retry everything but deadline errors up to max_retry).
+  std::function<bool(int error_code, int attempt)> retry_policy;
 };
 
 class ChannelBase {

@@ -249,7 +249,15 @@ void h2_dispatch_fiber(void* raw) {
   std::string content_type = st.is_grpc ? "application/grpc" : "application/octet-stream";
 
   if (fn == nullptr) {
-    if (st.is_grpc) {
+    if (st.is_grpc && st.path == "/grpc.health.v1.Health/Check") {
+      // builtin gRPC health service (parity: reference grpc_health_check):
+      // HealthCheckResponse{status: SERVING} = field 1 varint 1
+      std::string payload("\x08\x01", 2);
+      char frame[5] = {0, 0, 0, 0, (char)payload.size()};
+      body->data.assign(frame, 5);
+      body->data += payload;
+      body->grpc_status = 0;
+    } else if (st.is_grpc) {
       body->grpc_status = 12;  // UNIMPLEMENTED
       body->grpc_message = "unknown method " + st.path;
     } else {

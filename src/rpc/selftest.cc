@@ -119,6 +119,29 @@ int protocol_call(const std::string& addr, const std::string& protocol,
   return 0;
 }
 
+// ---- retry policy hook ----
+// Calls a dead port with a policy that refuses retries; returns the
+// observed attempt count (policy calls) — must be exactly 1.
+int retry_policy_test(int max_retry) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = 2000;
+  opts.max_retry = max_retry;
+  static std::atomic<int> calls{0};
+  calls = 0;
+  opts.retry_policy = [](int /*error_code*/, int /*attempt*/) {
+    calls.fetch_add(1);
+    return false;  // never retry
+  };
+  if (channel.Init("127.0.0.1:1", &opts) != 0) return -1;
+  Controller cntl;
+  IOBuf req, resp;
+  req.append("x");
+  channel.CallMethod("EchoService.Echo", &cntl, &req, &resp, nullptr);
+  if (!cntl.Failed()) return -2;
+  return calls.load();
+}
+
 // ---- session-local data (per-connection counter) ----
 int start_session_counter_server() {
   Server* server = new Server;

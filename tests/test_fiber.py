@@ -65,3 +65,15 @@ def test_counters_move():
 def test_fiber_local_storage():
     # fiber_key (≙ bthread_key): per-fiber slots + exit-time destructors
     assert f.key_test()
+
+
+def test_semaphore():
+    """FiberSemaphore (≙ reference bthread/semaphore): blocked acquirers
+    park as fibers; release(n) wakes exactly that much capacity."""
+    assert f.semaphore_test()
+
+
+def test_rwlock():
+    """FiberRWLock (≙ reference bthread/rwlock): writers exclusive (exact
+    counter under racy read-modify-write), readers overlap."""
+    assert f.rwlock_test(8, 2, 300)

@@ -98,3 +98,12 @@ def test_our_grpc_client_unknown_method(addr):
     rc, resp, err = b.core.combo.grpc_call(addr, "EchoService/Missing", b"x")
     assert rc != 0
     assert "12" in err or "grpc-status" in err
+
+
+def test_builtin_grpc_health_check(addr):
+    """Builtin grpc.health.v1.Health/Check (≙ reference grpc_health_check):
+    responds SERVING without a user-registered service."""
+    with grpc.insecure_channel(addr) as ch:
+        call = _stub(ch, "/grpc.health.v1.Health/Check")
+        resp = call(b"", timeout=5)
+        assert resp == b"\x08\x01"  # status: SERVING

@@ -109,3 +109,10 @@ def test_session_local_data_per_connection():
     vals = [r.protocol_call("127.0.0.1:%d" % port, "std", "Sess.Count", b"")[1]
             for _ in range(3)]
     assert vals == [b"1", b"2", b"3"], vals
+
+
+def test_retry_policy_hook():
+    """ChannelOptions::retry_policy (≙ reference brpc/retry_policy.h):
+    a policy returning false stops retries after the first failure."""
+    n = b.core.rpc.retry_policy_test(3)
+    assert n == 1, n
