@@ -249,9 +249,12 @@ TaskControl::TaskControl() {
   }
   if (n <= 0) n = (int)std::thread::hardware_concurrency();
   if (n <= 0) n = 4;
-  // More workers than ~16 hurts: the steal scan is O(workers) and the
-  // spin-before-park burn multiplies (reference default is 8+1 workers).
-  if (n > 16) n = 16;
+  // Fewer workers win on big machines: the steal scan is O(workers) and
+  // the spin-before-park burn multiplies. Same-box measurements (MI355X
+  // host, echo bench): cap 8 vs 16 -> HBM p99 ~380 vs ~700-1850 us at
+  // equal QPS, host path +12% QPS. The reference's default is 8+1
+  // workers too. BAM_FIBER_WORKERS overrides.
+  if (n > 8) n = 8;
   start_workers(n);
 }
 
