@@ -1,3 +1,4 @@
+#include "rpc/load_balancer.h"
 #include "rpc/mysql_client.h"
 #include "rpc/rtmp_client.h"
 #include <pybind11/pybind11.h>
@@ -18,6 +19,7 @@ struct BenchResult {
 int start_echo_server(int port);
 int start_nshead_server();
 int retry_policy_test(int max_retry);
+// naming resolution (rpc/load_balancer.h)
 int start_session_counter_server();
 int start_rtmp_server();
 int start_mongo_echo_server();
@@ -100,6 +102,16 @@ void bind_rpc(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
         py::call_guard<py::gil_scoped_release>());
+  r.def("resolve_naming", [](const std::string& url) {
+    std::vector<std::string> out;
+    std::vector<bam::EndPoint> eps;
+    {
+      py::gil_scoped_release rel;
+      if (bam::ResolveNamingUrl(url, &eps) != 0) return out;
+    }
+    for (const auto& ep : eps) out.push_back(bam::endpoint2str(ep));
+    return out;
+  });
   r.def("retry_policy_test", &bam::rpctest::retry_policy_test,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_session_counter_server", &bam::rpctest::start_session_counter_server,

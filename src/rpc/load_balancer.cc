@@ -1,5 +1,10 @@
 #include "rpc/load_balancer.h"
 
+#include "rpc/channel.h"
+#include "rpc/controller.h"
+
+#include <netdb.h>
+#include <netinet/in.h>
 #include <stdio.h>
 #include <string.h>
 
@@ -8,6 +13,7 @@
 #include <mutex>
 
 #include "base/fast_rand.h"
+#include "base/json.h"
 #include "base/logging.h"
 
 namespace bam {
@@ -103,6 +109,65 @@ int ResolveNamingUrl(const std::string& url, std::vector<EndPoint>* out) {
     while ((n = fread(buf, 1, sizeof(buf), f)) > 0) content.append(buf, n);
     fclose(f);
     parse_csv(content);
+    return 0;
+  }
+  if (url.rfind("dns://", 0) == 0) {
+    // dns://host:port — ALL A records of host, each at :port (parity:
+    // reference policy/domain_naming_service.cpp). Re-resolved by the
+    // refresher fiber like every naming scheme here.
+    std::string rest = url.substr(6);
+    size_t colon = rest.find(':');
+    if (colon == std::string::npos) return -1;
+    std::string host = rest.substr(0, colon);
+    int port = atoi(rest.c_str() + colon + 1);
+    struct addrinfo hints, *res = nullptr;
+    memset(&hints, 0, sizeof(hints));
+    hints.ai_family = AF_INET;
+    hints.ai_socktype = SOCK_STREAM;
+    if (getaddrinfo(host.c_str(), nullptr, &hints, &res) != 0) return -1;
+    for (struct addrinfo* p = res; p != nullptr; p = p->ai_next) {
+      EndPoint ep;
+      ep.ip = ((struct sockaddr_in*)p->ai_addr)->sin_addr;
+      ep.port = port;
+      out->push_back(ep);
+    }
+    freeaddrinfo(res);
+    return out->empty() ? -1 : 0;
+  }
+  if (url.rfind("consul://", 0) == 0) {
+    // consul://host:port/service — GET /v1/health/service/<svc>?passing=1
+    // over our own HTTP client; entries come from Service.Address/Port
+    // (parity: reference policy/consul_naming_service.cpp).
+    std::string rest = url.substr(9);
+    size_t slash = rest.find('/');
+    if (slash == std::string::npos) return -1;
+    std::string agent = rest.substr(0, slash);
+    std::string service = rest.substr(slash + 1);
+    Channel ch;
+    ChannelOptions copt;
+    copt.protocol = "http";
+    copt.timeout_ms = 3000;
+    copt.max_retry = 0;
+    if (ch.Init(agent.c_str(), &copt) != 0) return -1;
+    Controller cntl;
+    IOBuf req, resp;
+    ch.CallMethod("/v1/health/service/" + service + "?passing=1", &cntl, &req, &resp,
+                  nullptr);
+    if (cntl.Failed()) return -1;
+    json::Value root;
+    if (!json::Parse(resp.to_string(), &root) || root.type != json::Value::ARRAY) return -1;
+    for (const json::Value& entry : *root.arr) {
+      if (entry.type != json::Value::OBJECT) continue;
+      auto sit = entry.obj->find("Service");
+      if (sit == entry.obj->end() || sit->second.type != json::Value::OBJECT) continue;
+      const json::Object& svc = *sit->second.obj;
+      auto ait = svc.find("Address");
+      auto pit = svc.find("Port");
+      if (ait == svc.end() || pit == svc.end()) continue;
+      EndPoint ep;
+      std::string hp = ait->second.str + ":" + std::to_string((int)pit->second.num);
+      if (str2endpoint(hp.c_str(), &ep) == 0) out->push_back(ep);
+    }
     return 0;
   }
   // bare "host:port" treated as a single-entry list

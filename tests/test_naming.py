@@ -1,0 +1,43 @@
+"""Naming services beyond list/file (reference policy/domain_naming_service
++ consul_naming_service): dns:// resolves every A record; consul:// pulls
+Service.Address/Port from the health API — served here by a scripted
+consul agent running on our own HTTP server."""
+import json
+
+import brpc_amd as b
+import pytest
+
+r = b.core.rpc
+
+
+def test_dns_naming_resolves_localhost():
+    port = 7777
+    eps = r.resolve_naming("dns://localhost:%d" % port)
+    assert eps and all(e.endswith(":%d" % port) for e in eps), eps
+
+
+def test_consul_naming():
+    # scripted consul agent: our own Server answering the health API
+    echo_port = r.start_echo_server(0)
+    agent = b.Server()
+
+    def health(req, att):
+        body = json.dumps([
+            {"Service": {"Address": "127.0.0.1", "Port": echo_port}},
+            {"Service": {"Address": "127.0.0.1", "Port": echo_port}},
+        ]).encode()
+        return body, b""
+
+    agent.add_method("v1", "health/service/echo", health)  # path form below
+    agent_port = agent.start(0)
+    # resolution goes through /v1/health/service/echo?passing=1 — register
+    # a restful mapping so the GET lands on the handler
+    agent.add_restful_mapping("v1", "/v1/health/service/echo => health/service/echo")
+    eps = b.core.rpc.resolve_naming("consul://127.0.0.1:%d/echo" % agent_port)
+    assert eps == ["127.0.0.1:%d" % echo_port] * 2, eps
+    agent.stop()
+
+
+def test_dns_resolver_direct():
+    eps = b.core.rpc.resolve_naming("dns://localhost:8123")
+    assert "127.0.0.1:8123" in eps
