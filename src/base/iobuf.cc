@@ -603,6 +603,39 @@ struct TlsBounce {
 thread_local TlsBounce tls_bounce;
 }  // namespace
 
+int IOBuf::cut_until(IOBuf* out, const std::string& delim) {
+  if (delim.empty()) return -1;
+  // KMP across block boundaries (self-overlapping delimiters like
+  // "\r\n\r\n" need the real failure function).
+  std::vector<size_t> lps(delim.size(), 0);
+  for (size_t i = 1, len = 0; i < delim.size();) {
+    if (delim[i] == delim[len]) {
+      lps[i++] = ++len;
+    } else if (len != 0) {
+      len = lps[len - 1];
+    } else {
+      lps[i++] = 0;
+    }
+  }
+  size_t match = 0;
+  size_t scanned = 0;
+  for (uint32_t i = 0; i < count_; ++i) {
+    const BlockRef& r = ref_at(i);
+    if (r.block->res != RES_HOST) return -1;  // device bytes are not scannable
+    const char* p = r.block->data + r.offset;
+    for (uint32_t k = 0; k < r.length; ++k) {
+      ++scanned;
+      while (match > 0 && p[k] != delim[match]) match = lps[match - 1];
+      if (p[k] == delim[match] && ++match == delim.size()) {
+        cutn(out, scanned - delim.size());
+        pop_front(delim.size());
+        return 0;
+      }
+    }
+  }
+  return -1;
+}
+
 ssize_t IOBuf::cut_into_file_descriptor(int fd, size_t size_hint) {
   if (empty()) return 0;
   struct iovec iov[kMaxIov];

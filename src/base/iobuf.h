@@ -106,6 +106,10 @@ class IOBuf {
   size_t cutn(void* out, size_t n);        // copies to host memory
   size_t cutn(std::string* out, size_t n);
   int cut1(char* c);
+  // Cuts everything up to AND INCLUDING the first occurrence of `delim`
+  // into *out (delim itself is consumed but not copied). Returns 0, or -1
+  // if the delimiter is absent (parity: reference IOBuf::cut_until).
+  int cut_until(IOBuf* out, const std::string& delim);
   size_t pop_front(size_t n);
   size_t pop_back(size_t n);
 

@@ -48,6 +48,11 @@ void bind_base(py::module_& m) {
         return py::bytes(s);
       })
       .def("pop_front", &IOBuf::pop_front)
+      .def("cut_until", [](IOBuf& b, py::bytes delim) -> py::object {
+        IOBuf out;
+        if (b.cut_until(&out, delim.cast<std::string>()) != 0) return py::none();
+        return py::bytes(out.to_string());
+      })
       .def("pop_back", &IOBuf::pop_back)
       .def("copy_to",
            [](const IOBuf& b, size_t n, size_t pos) {

@@ -126,3 +126,26 @@ class TestEndPoint:
 def test_fast_rand():
     vals = {b.fast_rand() for _ in range(100)}
     assert len(vals) == 100
+
+
+def test_iobuf_cut_until():
+    """IOBuf::cut_until (≙ reference iobuf cut_until): delimiter consumed,
+    body returned; KMP handles self-overlapping delimiters across blocks."""
+    buf = b.IOBuf()
+    buf.append(b"GET / HTTP/1.1\r\nHost: x\r\n\r\nBODY")
+    line = buf.cut_until(b"\r\n")
+    assert line == b"GET / HTTP/1.1"
+    rest = buf.cut_until(b"\r\n\r\n")
+    assert rest == b"Host: x"
+    assert buf.cutn(100) == b"BODY"
+    # self-overlap: "aab" inside "aaab"
+    buf2 = b.IOBuf()
+    buf2.append(b"aa")
+    buf2.append(b"ab!")  # crosses block boundary
+    assert buf2.cut_until(b"aab") == b"a"
+    assert buf2.cutn(10) == b"!"
+    # absent delimiter
+    buf3 = b.IOBuf()
+    buf3.append(b"xyz")
+    assert buf3.cut_until(b"\r\n") is None
+    assert buf3.cutn(10) == b"xyz"
