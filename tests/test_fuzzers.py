@@ -10,7 +10,7 @@ import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 TARGETS = ["fuzz_rpc_meta", "fuzz_http", "fuzz_redis", "fuzz_json", "fuzz_snappy",
-           "fuzz_mcpack"]
+           "fuzz_mcpack", "fuzz_cut_until"]
 
 
 @pytest.mark.slow
