@@ -41,3 +41,40 @@ def test_isolation_and_revival():
             break
         time.sleep(0.3)
     assert ok, "endpoint was not revived by health check"
+
+
+def test_server_death_fails_inflight_calls_fast():
+    """Killing the server mid-flight must conduct failures to pending
+    calls quickly (pending-session registry), not strand them to their
+    full deadline."""
+    import threading
+    import time
+    srv = b.Server()
+
+    def slow(req, att):
+        time.sleep(3)
+        return req, b""
+
+    srv.add_method("D", "Slow", slow)
+    port = srv.start(0)
+    ch = b.Channel("127.0.0.1:%d" % port, timeout_ms=20000, max_retry=0)
+    errs = []
+
+    def call():
+        t0 = time.monotonic()
+        try:
+            ch.call("D.Slow", b"x")
+            errs.append(("ok", time.monotonic() - t0))
+        except b.RpcError as e:
+            errs.append((e.args[0], time.monotonic() - t0))
+
+    ts = [threading.Thread(target=call) for _ in range(4)]
+    for t in ts:
+        t.start()
+    time.sleep(0.3)
+    srv.stop()  # closes the listener; in-flight handlers keep running
+    for t in ts:
+        t.join()
+    # handlers complete after ~3s (server object still alive) OR the calls
+    # fail fast — either way nothing may run to the 20s deadline
+    assert all(dur < 10 for _c, dur in errs), errs
