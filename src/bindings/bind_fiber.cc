@@ -13,7 +13,7 @@ bool butex_timeout_test();
 int64_t mutex_test(int nfibers, int iters);
 bool countdown_test(int n);
 bool semaphore_test();
-bool rwlock_test(int nreaders, int nwriters, int iters);
+int rwlock_test(int nreaders, int nwriters, int iters);
 bool timer_test();
 bool fiber_key_test();
 }  // namespace selftest
