@@ -129,12 +129,17 @@ bool poll_wait(HasWorkFn has_work, StopFn stop) {
     __builtin_ia32_pause();
 #endif
   }
-  struct timespec nap {0, 20000};  // 20 µs
-  while (!has_work()) {
+  // 20 µs naps while traffic is plausible, backing off to 1 ms after
+  // ~50 ms idle so an idle (or leaked) connection's poller costs ~nothing.
+  long nap_ns = 20000;
+  int idle = 0;
+  for (;;) {
+    if (has_work()) return true;
     if (stop()) return false;
+    struct timespec nap {0, nap_ns};
     nanosleep(&nap, nullptr);
+    if (++idle > 2500 && nap_ns < 1000000) nap_ns = 1000000;
   }
-  return true;
 }
 
 struct Segment {
