@@ -55,6 +55,8 @@ def main():
         # assumption (driver runs one rank per real GPU).
         backend = os.environ.get("BAM_BENCH_BACKEND") or (
             "nccl" if torch.cuda.is_available() else "gloo")
+        if backend == "nccl":
+            torch.cuda.set_device(0)  # each rank is masked to its own GPU
         dist.init_process_group(backend=backend)
 
     ndev = b.core.gpu.initialize()
@@ -107,12 +109,15 @@ def main():
     barrier_sync()
     elapsed = time.monotonic() - t0
 
-    # MAX elapsed over ranks; SUM of calls over ranks.
+    # MAX elapsed over ranks; SUM of calls over ranks. RCCL (backend
+    # "nccl") reduces DEVICE tensors only — CPU tensors here would abort
+    # the driver's 8-GPU scaling run.
     if dist is not None:
         import torch
 
-        te = torch.tensor([elapsed], dtype=torch.float64)
-        tc = torch.tensor([float(total_calls)], dtype=torch.float64)
+        dev = torch.device("cuda", 0) if dist.get_backend() == "nccl" else torch.device("cpu")
+        te = torch.tensor([elapsed], dtype=torch.float64, device=dev)
+        tc = torch.tensor([float(total_calls)], dtype=torch.float64, device=dev)
         dist.all_reduce(te, op=dist.ReduceOp.MAX)
         dist.all_reduce(tc, op=dist.ReduceOp.SUM)
         elapsed = te.item()
