@@ -38,6 +38,8 @@ def main():
     use_gpu = torch.cuda.is_available()
     backend = os.environ.get("BAM_BENCH_BACKEND") or ("nccl" if use_gpu else "gloo")
     use_gpu = use_gpu and backend == "nccl"
+    if use_gpu:
+        torch.cuda.set_device(0)  # each rank masked to its own GPU
     dist.init_process_group(backend=backend)
     rank = dist.get_rank()
     dev = torch.device("cuda:0") if use_gpu else torch.device("cpu")
