@@ -69,6 +69,27 @@ void bind_rpc(py::module_& m) {
              return res;
            })
       .def("ping", &bam::MysqlClient::Ping, py::call_guard<py::gil_scoped_release>())
+      .def("prepare",
+           [](bam::MysqlClient& c, const std::string& sql) {
+             int nparams = 0;
+             int64_t sid;
+             {
+               py::gil_scoped_release rel;
+               sid = c.Prepare(sql, &nparams);
+             }
+             return py::make_tuple(sid, nparams);
+           })
+      .def("execute_prepared",
+           [](bam::MysqlClient& c, int64_t sid, const std::vector<std::string>& params) {
+             bam::MysqlResult res;
+             {
+               py::gil_scoped_release rel;
+               c.ExecutePrepared(sid, params, &res);
+             }
+             return res;
+           })
+      .def("close_statement", &bam::MysqlClient::CloseStatement,
+           py::call_guard<py::gil_scoped_release>())
       .def("close", &bam::MysqlClient::Close)
       .def("connected", &bam::MysqlClient::connected)
       .def("server_version", &bam::MysqlClient::server_version);

@@ -12,6 +12,8 @@
 #include "base/codecs.h"
 #include "base/endpoint.h"
 
+#include <vector>
+
 namespace bam {
 
 namespace {
@@ -239,6 +241,179 @@ int MysqlClient::Ping() {
   uint8_t seq;
   if (read_packet(&p, &seq) != 0 || p.empty()) return -1;
   return (uint8_t)p[0] == 0x00 ? 0 : -1;
+}
+
+// ---- prepared statements (binary protocol) ----
+
+int64_t MysqlClient::Prepare(const std::string& sql, int* param_count) {
+  if (fd_ < 0) return -1;
+  std::string cmd(1, (char)0x16);  // COM_STMT_PREPARE
+  cmd += sql;
+  if (write_packet(cmd, 0) != 0) return -1;
+  std::string p;
+  uint8_t seq;
+  if (read_packet(&p, &seq) != 0 || p.empty()) return -1;
+  if ((uint8_t)p[0] == 0xff)
+    return p.size() >= 3 ? ((uint8_t)p[1] | ((int64_t)(uint8_t)p[2] << 8)) : -1;
+  if ((uint8_t)p[0] != 0x00 || p.size() < 12) return -1;
+  // PREPARE_OK: statement_id u32, num_columns u16, num_params u16
+  uint32_t stmt_id;
+  memcpy(&stmt_id, p.data() + 1, 4);
+  uint16_t ncols, nparams;
+  memcpy(&ncols, p.data() + 5, 2);
+  memcpy(&nparams, p.data() + 7, 2);
+  if (param_count != nullptr) *param_count = nparams;
+  // drain param + column definition packets (+ possible EOFs)
+  for (int i = 0; i < nparams; ++i)
+    if (read_packet(&p, &seq) != 0) return -1;
+  if (nparams > 0 && (read_packet(&p, &seq) != 0)) return -1;  // EOF
+  for (int i = 0; i < ncols; ++i)
+    if (read_packet(&p, &seq) != 0) return -1;
+  if (ncols > 0 && (read_packet(&p, &seq) != 0)) return -1;  // EOF
+  return (int64_t)stmt_id;
+}
+
+int MysqlClient::ExecutePrepared(int64_t stmt_id, const std::vector<std::string>& params,
+                                 MysqlResult* out) {
+  *out = MysqlResult();
+  if (fd_ < 0) return -1;
+  std::string cmd(1, (char)0x17);  // COM_STMT_EXECUTE
+  uint32_t sid = (uint32_t)stmt_id;
+  cmd.append((const char*)&sid, 4);
+  cmd.push_back(0);                         // CURSOR_TYPE_NO_CURSOR
+  uint32_t iter = 1;
+  cmd.append((const char*)&iter, 4);
+  if (!params.empty()) {
+    std::string null_bitmap((params.size() + 7) / 8, '\0');
+    cmd += null_bitmap;
+    cmd.push_back(1);  // new_params_bound
+    for (size_t i = 0; i < params.size(); ++i) {
+      cmd.push_back((char)0xfe);  // MYSQL_TYPE_STRING
+      cmd.push_back(0);
+    }
+    for (const std::string& v : params) {
+      // length-encoded string
+      if (v.size() < 0xfb) {
+        cmd.push_back((char)v.size());
+      } else {
+        cmd.push_back((char)0xfc);
+        uint16_t l = (uint16_t)v.size();
+        cmd.append((const char*)&l, 2);
+      }
+      cmd += v;
+    }
+  }
+  if (write_packet(cmd, 0) != 0) return -1;
+  std::string p;
+  uint8_t seq;
+  if (read_packet(&p, &seq) != 0 || p.empty()) return -1;
+  uint8_t first = (uint8_t)p[0];
+  if (first == 0xff) {
+    out->error_code = p.size() >= 3 ? ((uint8_t)p[1] | ((int)(uint8_t)p[2] << 8)) : -1;
+    out->error_message = p.substr(p.size() > 9 && p[3] == '#' ? 9 : 3);
+    return out->error_code;
+  }
+  if (first == 0x00) {
+    out->ok = true;
+    size_t pos = 1;
+    uint64_t v = 0;
+    int k = lenc_int(p.data() + pos, p.size() - pos, &v);
+    if (k > 0) {
+      out->affected_rows = v;
+      pos += (size_t)k;
+      if (lenc_int(p.data() + pos, p.size() - pos, &v) > 0) out->last_insert_id = v;
+    }
+    return 0;
+  }
+  // binary resultset: column count, defs, EOF, rows (0x00-headed), EOF
+  uint64_t ncols = 0;
+  if (lenc_int(p.data(), p.size(), &ncols) <= 0 || ncols == 0 || ncols > 4096) return -1;
+  std::vector<uint8_t> col_types;
+  for (uint64_t i = 0; i < ncols; ++i) {
+    if (read_packet(&p, &seq) != 0) return -1;
+    // column def: catalog..org_name lenc strings, then 0x0c fixed block
+    size_t pos = 0;
+    std::string field;
+    bool nul;
+    for (int f = 0; f < 6; ++f) {
+      int k = lenc_str(p.data() + pos, p.size() - pos, &field, &nul);
+      if (k <= 0) break;
+      if (f == 4) out->columns.push_back(field);
+      pos += (size_t)k;
+    }
+    // fixed block: filler(1) charset(2) length(4) type(1) ...
+    col_types.push_back(pos + 8 < p.size() ? (uint8_t)p[pos + 7] : 0xfe);
+  }
+  if (read_packet(&p, &seq) != 0) return -1;  // EOF after defs
+  for (;;) {
+    if (read_packet(&p, &seq) != 0) return -1;
+    if (!p.empty() && (uint8_t)p[0] == 0xfe && p.size() < 9) break;  // EOF
+    if (p.empty() || (uint8_t)p[0] != 0x00) return -1;
+    // binary row: 0x00 header + null bitmap (offset 2) + values
+    size_t nb = (ncols + 9) / 8;
+    if (p.size() < 1 + nb) return -1;
+    const uint8_t* bitmap = (const uint8_t*)p.data() + 1;
+    size_t pos = 1 + nb;
+    std::vector<std::string> row;
+    for (uint64_t c = 0; c < ncols; ++c) {
+      if (bitmap[(c + 2) / 8] & (1 << ((c + 2) % 8))) {
+        row.push_back("");
+        continue;
+      }
+      uint8_t t = col_types[c];
+      char buf[32];
+      if (t == 0x01) {  // TINY
+        row.push_back(std::to_string((int)(int8_t)p[pos]));
+        pos += 1;
+      } else if (t == 0x02) {  // SHORT
+        int16_t v16;
+        memcpy(&v16, p.data() + pos, 2);
+        row.push_back(std::to_string(v16));
+        pos += 2;
+      } else if (t == 0x03) {  // LONG
+        int32_t v32;
+        memcpy(&v32, p.data() + pos, 4);
+        row.push_back(std::to_string(v32));
+        pos += 4;
+      } else if (t == 0x08) {  // LONGLONG
+        int64_t v64;
+        memcpy(&v64, p.data() + pos, 8);
+        row.push_back(std::to_string((long long)v64));
+        pos += 8;
+      } else if (t == 0x04 || t == 0x05) {  // FLOAT/DOUBLE
+        double d = 0;
+        if (t == 0x04) {
+          float f;
+          memcpy(&f, p.data() + pos, 4);
+          d = f;
+          pos += 4;
+        } else {
+          memcpy(&d, p.data() + pos, 8);
+          pos += 8;
+        }
+        snprintf(buf, sizeof(buf), "%g", d);
+        row.push_back(buf);
+      } else {  // string-ish: length-encoded bytes
+        std::string v;
+        bool nul2;
+        int k = lenc_str(p.data() + pos, p.size() - pos, &v, &nul2);
+        if (k <= 0) return -1;
+        row.push_back(v);
+        pos += (size_t)k;
+      }
+    }
+    out->rows.push_back(std::move(row));
+  }
+  out->ok = true;
+  return 0;
+}
+
+void MysqlClient::CloseStatement(int64_t stmt_id) {
+  if (fd_ < 0) return;
+  std::string cmd(1, (char)0x19);  // COM_STMT_CLOSE (no response)
+  uint32_t sid = (uint32_t)stmt_id;
+  cmd.append((const char*)&sid, 4);
+  write_packet(cmd, 0);
 }
 
 int MysqlClient::Query(const std::string& sql, MysqlResult* out) {

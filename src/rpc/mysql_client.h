@@ -43,6 +43,17 @@ class MysqlClient {
   int Query(const std::string& sql, MysqlResult* out);
 
   int Ping();
+
+  // Prepared statements (COM_STMT_PREPARE / COM_STMT_EXECUTE, binary
+  // protocol). Params are sent as MYSQL_TYPE_STRING; result rows come
+  // back through the binary row format and are surfaced as text in
+  // MysqlResult (ints/strings; other column types are hex-dumped).
+  // Prepare returns a statement id (>0) or -1 / server errno.
+  int64_t Prepare(const std::string& sql, int* param_count = nullptr);
+  int ExecutePrepared(int64_t stmt_id, const std::vector<std::string>& params,
+                      MysqlResult* out);
+  void CloseStatement(int64_t stmt_id);
+
   void Close();
   bool connected() const { return fd_ >= 0; }
   const std::string& server_version() const { return server_version_; }

@@ -93,6 +93,42 @@ def _fake_mysql_server(results):
             if cmd == 0x0e:  # PING
                 c.sendall(_packet(b"\x00\x00\x00\x02\x00\x00\x00", 1))
                 continue
+            if cmd == 0x16:  # STMT_PREPARE
+                sql = body[1:].decode()
+                nparams = sql.count("?")
+                # PREPARE_OK: 00 stmt_id nr_cols nr_params 00 warnings
+                ok = struct.pack("<BIHHBH", 0, 77, 1, nparams, 0, 0)
+                c.sendall(_packet(ok, 1))
+                seqn = 2
+                for _ in range(nparams):
+                    c.sendall(_packet(_coldef(b"?"), seqn)); seqn += 1
+                if nparams:
+                    c.sendall(_packet(b"\xfe\x00\x00\x02\x00", seqn)); seqn += 1
+                c.sendall(_packet(_coldef(b"res"), seqn)); seqn += 1
+                c.sendall(_packet(b"\xfe\x00\x00\x02\x00", seqn))
+                continue
+            if cmd == 0x17:  # STMT_EXECUTE: echo params joined as one row
+                # parse: stmt_id u32, flags u8, iter u32, then bitmap+types+values
+                nparams_guess = 2
+                pos = 1 + 4 + 1 + 4
+                nb = (nparams_guess + 7) // 8
+                pos += nb + 1 + 2 * nparams_guess
+                vals = []
+                for _ in range(nparams_guess):
+                    ln = body[pos]; pos += 1
+                    vals.append(body[pos:pos + ln]); pos += ln
+                joined = b"|".join(vals)
+                seqn = 1
+                c.sendall(_packet(_lenc(1), seqn)); seqn += 1
+                c.sendall(_packet(_coldef(b"res"), seqn)); seqn += 1
+                c.sendall(_packet(b"\xfe\x00\x00\x02\x00", seqn)); seqn += 1
+                # binary row: 00 header, null bitmap (1 col -> 1 byte), lenc value
+                row = b"\x00" + b"\x00" + _lenc_str(joined)
+                c.sendall(_packet(row, seqn)); seqn += 1
+                c.sendall(_packet(b"\xfe\x00\x00\x02\x00", seqn))
+                continue
+            if cmd == 0x19:  # STMT_CLOSE: no response
+                continue
             if cmd == 0x03:  # QUERY
                 sql = body[1:].decode()
                 if sql.startswith("SELECT"):
@@ -160,4 +196,19 @@ def test_mysql_bad_password_rejected():
     rc = c.connect("127.0.0.1", port, USER, "wrong")
     assert rc == 1045
     assert not c.connected()
+    srv.close()
+
+
+def test_mysql_prepared_statements():
+    srv, port = _fake_mysql_server(([b"x"], []))
+    c = b.core.rpc.MysqlClient()
+    assert c.connect("127.0.0.1", port, USER, PASSWORD) == 0
+    sid, nparams = c.prepare("SELECT concat(?, ?)")
+    assert sid == 77
+    assert nparams == 2
+    res = c.execute_prepared(sid, ["abc", "xyz"])
+    assert res.ok, (res.error_code, res.error_message)
+    assert res.columns == ["res"]
+    assert res.rows == [["abc|xyz"]]
+    c.close_statement(sid)
     srv.close()
