@@ -9,6 +9,8 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <openssl/sha.h>
+
 #include "base/codecs.h"
 #include "base/endpoint.h"
 
@@ -39,6 +41,27 @@ std::string native_scramble(const std::string& password, const std::string& salt
   std::string out(h1.size(), '\0');
   for (size_t i = 0; i < h1.size(); ++i) out[i] = h1[i] ^ h3[i];
   return out;
+}
+
+// caching_sha2_password fast path:
+// XOR(SHA256(pwd), SHA256(SHA256(SHA256(pwd)) || nonce))
+std::string sha256s(const std::string& in) {
+  unsigned char d[SHA256_DIGEST_LENGTH];
+  SHA256((const unsigned char*)in.data(), in.size(), d);
+  return std::string((const char*)d, sizeof(d));
+}
+std::string caching_sha2_scramble(const std::string& password, const std::string& nonce) {
+  if (password.empty()) return "";
+  std::string h1 = sha256s(password);
+  std::string h2 = sha256s(sha256s(h1) + nonce);
+  std::string out(h1.size(), '\0');
+  for (size_t i = 0; i < h1.size(); ++i) out[i] = h1[i] ^ h2[i];
+  return out;
+}
+std::string scramble_for(const std::string& plugin, const std::string& password,
+                         const std::string& nonce) {
+  if (plugin == "caching_sha2_password") return caching_sha2_scramble(password, nonce);
+  return native_scramble(password, nonce);
 }
 
 // length-encoded integer; returns bytes consumed, 0 on error, -1 for NULL.
@@ -191,6 +214,21 @@ int MysqlClient::Connect(const std::string& host, int port, const std::string& u
     }
   }
 
+  // server's auth plugin name trails the salt (if CLIENT_PLUGIN_AUTH)
+  std::string server_plugin = "mysql_native_password";
+  {
+    size_t z2 = hs.find('\0', hs.size() > 64 ? hs.size() - 64 : 0);
+    // simplest robust read: last NUL-terminated token of the packet
+    size_t last_nul = hs.find_last_of('\0');
+    if (last_nul != std::string::npos && last_nul + 1 <= hs.size()) {
+      size_t prev = hs.find_last_of('\0', last_nul - 1);
+      std::string tail = hs.substr(prev + 1, last_nul - prev - 1);
+      if (tail == "caching_sha2_password" || tail == "mysql_native_password")
+        server_plugin = tail;
+    }
+    (void)z2;
+  }
+
   // ---- HandshakeResponse41 ----
   uint32_t caps = kClientLongPassword | kClientProtocol41 | kClientSecureConnection |
                   kClientPluginAuth;
@@ -203,35 +241,61 @@ int MysqlClient::Connect(const std::string& host, int port, const std::string& u
   resp.append(23, '\0');
   resp.append(user);
   resp.push_back('\0');
-  std::string scramble = native_scramble(password, salt);
+  std::string scramble = scramble_for(server_plugin, password, salt);
   resp.push_back((char)scramble.size());
   resp.append(scramble);
   if (!db.empty()) {
     resp.append(db);
     resp.push_back('\0');
   }
-  resp.append("mysql_native_password");
+  resp.append(server_plugin);
   resp.push_back('\0');
   if (write_packet(resp, (uint8_t)(seq + 1)) != 0) {
     Close();
     return -1;
   }
 
-  std::string fin;
-  if (read_packet(&fin, &seq) != 0 || fin.empty()) {
+  // ---- auth continuation: OK / ERR / AuthSwitch (0xFE) / AuthMoreData (0x01)
+  for (int hop = 0; hop < 4; ++hop) {
+    std::string fin;
+    if (read_packet(&fin, &seq) != 0 || fin.empty()) {
+      Close();
+      return -1;
+    }
+    uint8_t tag = (uint8_t)fin[0];
+    if (tag == 0x00) return 0;  // OK
+    if (tag == 0xff) {          // ERR
+      int code = fin.size() >= 3 ? ((uint8_t)fin[1] | ((int)(uint8_t)fin[2] << 8)) : -1;
+      Close();
+      return code;
+    }
+    if (tag == 0xfe) {  // AuthSwitchRequest: plugin\0 nonce
+      size_t z3 = fin.find('\0', 1);
+      if (z3 == std::string::npos) {
+        Close();
+        return -1;
+      }
+      std::string plugin = fin.substr(1, z3 - 1);
+      std::string nonce = fin.substr(z3 + 1);
+      while (!nonce.empty() && nonce.back() == '\0') nonce.pop_back();
+      std::string sc = scramble_for(plugin, password, nonce);
+      if (write_packet(sc, (uint8_t)(seq + 1)) != 0) {
+        Close();
+        return -1;
+      }
+      continue;
+    }
+    if (tag == 0x01) {  // AuthMoreData (caching_sha2)
+      if (fin.size() >= 2 && (uint8_t)fin[1] == 0x03) continue;  // fast auth ok -> OK next
+      // 0x04 = full auth needed (requires TLS or RSA key exchange)
+      Close();
+      return -1;
+    }
     Close();
     return -1;
   }
-  if ((uint8_t)fin[0] == 0xff) {  // ERR
-    int code = fin.size() >= 3 ? ((uint8_t)fin[1] | ((int)(uint8_t)fin[2] << 8)) : -1;
-    Close();
-    return code;
-  }
-  if ((uint8_t)fin[0] != 0x00) {  // auth-switch etc. unsupported
-    Close();
-    return -1;
-  }
-  return 0;
+  Close();
+  return -1;
 }
 
 int MysqlClient::Ping() {

@@ -212,3 +212,57 @@ def test_mysql_prepared_statements():
     assert res.rows == [["abc|xyz"]]
     c.close_statement(sid)
     srv.close()
+
+
+def test_mysql_caching_sha2_fast_auth():
+    """caching_sha2_password fast path: server advertises the plugin in
+    HandshakeV10; client scramble = XOR(SHA256(pwd),
+    SHA256(SHA256(SHA256(pwd)) + nonce)); server replies AuthMoreData
+    0x03 (fast auth success) then OK."""
+    import socket
+    import threading
+
+    def sha2_scramble(password, salt):
+        h1 = hashlib.sha256(password.encode()).digest()
+        h2 = hashlib.sha256(hashlib.sha256(h1).digest() + salt).digest()
+        return bytes(a ^ b_ for a, b_ in zip(h1, h2))
+
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+    result = {}
+
+    def run():
+        c, _ = srv.accept()
+        hs = bytes([10]) + b"8.4.0-fake\0" + struct.pack("<I", 5)
+        hs += SALT[:8] + b"\x00"
+        hs += struct.pack("<H", 0xFFFF)
+        hs += bytes([33]) + struct.pack("<H", 2) + struct.pack("<H", 0xFFFF >> 16)
+        hs += bytes([21]) + b"\x00" * 10
+        hs += SALT[8:20] + b"\x00"
+        hs += b"caching_sha2_password\x00"
+        c.sendall(_packet(hs, 0))
+        head = c.recv(4)
+        ln = struct.unpack("<I", head[:3] + b"\x00")[0]
+        resp = b""
+        while len(resp) < ln:
+            resp += c.recv(ln - len(resp))
+        p = 4 + 4 + 1 + 23
+        z = resp.index(b"\x00", p)
+        alen = resp[z + 1]
+        auth = resp[z + 2:z + 2 + alen]
+        result["auth_ok"] = auth == sha2_scramble(PASSWORD, SALT)
+        c.sendall(_packet(b"\x01\x03", 2))                      # fast auth success
+        c.sendall(_packet(b"\x00\x00\x00\x02\x00\x00\x00", 3))  # OK
+        # stay open for COM_QUIT
+        c.recv(64)
+        c.close()
+
+    threading.Thread(target=run, daemon=True).start()
+    port = srv.getsockname()[1]
+    c = b.core.rpc.MysqlClient()
+    rc = c.connect("127.0.0.1", port, USER, PASSWORD)
+    assert rc == 0
+    assert result["auth_ok"], "caching_sha2 scramble mismatch"
+    c.close()
+    srv.close()
