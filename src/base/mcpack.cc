@@ -143,6 +143,26 @@ bool serialize_field(const Value& v, const std::string& name, std::string* out, 
     }
     case Value::OBJECT:
     case Value::ARRAY: {
+      // Uniform INT32-able arrays emit the compact ISOARRAY form
+      // (IsoItemsHead{type} + packed values) like the reference writer.
+      if (v.type == Value::ARRAY && !v.arr.empty()) {
+        bool all_i32 = true;
+        for (const Value& it : v.arr) {
+          if (it.type != Value::INT || it.i < INT32_MIN || it.i > INT32_MAX) {
+            all_i32 = false;
+            break;
+          }
+        }
+        if (all_i32) {
+          out->push_back((char)kIsoArray);
+          out->push_back((char)name_size(name));
+          put_pod<uint32_t>(out, (uint32_t)(1 + 4 * v.arr.size()));
+          put_name(out, name);
+          out->push_back((char)kInt32);
+          for (const Value& it : v.arr) put_pod<int32_t>(out, (int32_t)it.i);
+          return true;
+        }
+      }
       out->push_back((char)(v.type == Value::OBJECT ? kObject : kArray));
       out->push_back((char)name_size(name));
       size_t size_pos = out->size();

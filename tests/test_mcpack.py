@@ -73,3 +73,15 @@ def test_to_json():
 def test_corrupt_rejected():
     with pytest.raises(RuntimeError):
         c.mcpack_loads(b"\x10\x00\xff\xff\xff\xff")
+
+
+def test_isoarray_roundtrip_and_layout():
+    """Uniform int arrays serialize as ISOARRAY (0x30 + IsoItemsHead) and
+    parse back identically."""
+    blob = c.mcpack_dumps({"v": [7, -9, 100000]})
+    assert c.mcpack_loads(blob) == {"v": [7, -9, 100000]}
+    # field head at offset 10: ISOARRAY, name "v\0", vsize = 1 + 3*4
+    assert blob[10] == 0x30
+    (vsize,) = struct.unpack_from("<I", blob, 12)
+    assert vsize == 13
+    assert blob[16 + 2] == 0x14  # item type INT32 after name
