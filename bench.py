@@ -22,7 +22,15 @@ import time
 # Pin this process to its GPU BEFORE importing torch / brpc_amd.
 _local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 if "WORLD_SIZE" in os.environ and int(os.environ.get("WORLD_SIZE", "1")) > 1:
-    os.environ.setdefault("HIP_VISIBLE_DEVICES", str(_local_rank))
+    _hvd = os.environ.get("HIP_VISIBLE_DEVICES")
+    if _hvd is None:
+        os.environ["HIP_VISIBLE_DEVICES"] = str(_local_rank)
+    elif "," in _hvd:
+        # launcher exported the full device list to every rank: re-mask so
+        # rank i owns the i-th entry (otherwise all ranks pile onto GPU 0)
+        devs = [d.strip() for d in _hvd.split(",") if d.strip()]
+        os.environ["HIP_VISIBLE_DEVICES"] = devs[_local_rank % len(devs)]
+    # single pre-set value: trust the launcher's per-rank masking
 
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import brpc_amd as b  # noqa: E402
