@@ -18,6 +18,7 @@ struct BenchResult {
 };
 int start_echo_server(int port);
 int start_nshead_server();
+int64_t cancel_test(int port);
 int retry_policy_test(int max_retry);
 // naming resolution (rpc/load_balancer.h)
 int start_session_counter_server();
@@ -122,6 +123,8 @@ void bind_rpc(py::module_& m) {
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("cancel_test", &bam::rpctest::cancel_test,
         py::call_guard<py::gil_scoped_release>());
   r.def("resolve_naming", [](const std::string& url) {
     std::vector<std::string> out;

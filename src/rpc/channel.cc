@@ -126,8 +126,9 @@ static int HandleSessionError(SessionId id, void* data, int error_code) {
     return 0;
   }
   const bool policy_says_retry =
-      cntl->call.retry_policy ? cntl->call.retry_policy(error_code, cntl->retry_count_)
-                              : error_code != ERPCTIMEDOUT;
+      cntl->call.retry_policy
+          ? cntl->call.retry_policy(error_code, cntl->retry_count_)
+          : (error_code != ERPCTIMEDOUT && error_code != ECANCELED_RPC);
   if (policy_says_retry && cntl->retry_count_ < cntl->max_retry_) {
     ++cntl->retry_count_;
     session_bump_slot(id);

@@ -53,4 +53,8 @@ void* Controller::session_local_data() {
   return expected;
 }
 
+void Controller::StartCancel() {
+  if (cid_ != 0) session_error(cid_, ECANCELED_RPC);
+}
+
 }  // namespace bam

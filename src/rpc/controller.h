@@ -62,6 +62,12 @@ class Controller {
   void SetFailed(const std::string& reason) { SetFailed(EINTERNAL, reason); }
 
   SessionId call_id() const { return cid_; }
+  // Client-side cancellation (parity: reference StartCancel/IsCanceled):
+  // conducts ECANCELED through the call's session — done runs (or the
+  // synchronous CallMethod returns) with ErrorCode()==ECANCELED unless a
+  // response won the race.
+  void StartCancel();
+  bool IsCanceled() const { return error_code_ == ECANCELED_RPC; }
   int64_t latency_us() const { return end_us_ - start_us_; }
   int retried_count() const { return retry_count_; }
 

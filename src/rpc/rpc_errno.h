@@ -21,6 +21,7 @@ enum RpcError {
   EEOF = 1014,
   EUNUSED = 1015,
   ESSL_ERR = 1016,
+  ECANCELED_RPC = 1017,  // StartCancel (client-side cancellation)
   EINTERNAL = 2001,     // server internal error
   ERESPONSE = 2002,     // bad response
   ELOGOFF = 2003,       // server stopping

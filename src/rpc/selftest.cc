@@ -119,6 +119,30 @@ int protocol_call(const std::string& addr, const std::string& protocol,
   return 0;
 }
 
+// ---- StartCancel ----
+// Sleeps 800 ms server-side; cancels after ~50 ms. Returns the observed
+// client-side latency in us when ErrorCode()==ECANCELED_RPC, else -code.
+int64_t cancel_test(int port) {
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = 5000;
+  opts.max_retry = 2;
+  if (channel.Init(("127.0.0.1:" + std::to_string(port)).c_str(), &opts) != 0) return -1;
+  Controller cntl;
+  IOBuf req, resp;
+  req.append("800");  // Sleep ms
+  CountdownEvent done_ev(1);
+  Closure* done = NewCallback([&done_ev] { done_ev.signal(); });
+  int64_t t0 = monotonic_time_us();
+  channel.CallMethod("EchoService.Sleep", &cntl, &req, &resp, done);
+  fiber_usleep(50000);
+  cntl.StartCancel();
+  done_ev.wait();
+  int64_t lat = monotonic_time_us() - t0;
+  if (cntl.ErrorCode() != ECANCELED_RPC) return -cntl.ErrorCode();
+  return lat;
+}
+
 // ---- retry policy hook ----
 // Calls a dead port with a policy that refuses retries; returns the
 // observed attempt count (policy calls) — must be exactly 1.

@@ -116,3 +116,13 @@ def test_retry_policy_hook():
     a policy returning false stops retries after the first failure."""
     n = b.core.rpc.retry_policy_test(3)
     assert n == 1, n
+
+
+def test_start_cancel():
+    """Controller::StartCancel (≙ reference StartCancel/IsCanceled):
+    cancels an in-flight call with ECANCELED well before the server's
+    800ms handler or the 5s deadline."""
+    port = b.core.rpc.start_echo_server(0)
+    lat_us = b.core.rpc.cancel_test(port)
+    assert lat_us > 0, "expected ECANCELED, got errno %d" % -lat_us
+    assert lat_us < 700000, lat_us  # canceled long before the sleep finished
