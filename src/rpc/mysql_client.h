@@ -6,9 +6,13 @@
 //    scramble: SHA1(pwd) XOR SHA1(salt + SHA1(SHA1(pwd)))
 //  * COM_QUERY with OK / ERR / resultset (column defs + text rows with
 //    length-encoded values, EOF or OK-terminated), COM_PING, COM_INIT_DB
-// Deltas vs reference: prepared statements (COM_STMT_*) and caching_sha2
-// auth are not implemented; transactions work as plain queries
-// (BEGIN/COMMIT/ROLLBACK).
+//  * caching_sha2_password fast path (XOR(SHA256(pwd),
+//    SHA256(SHA256(SHA256(pwd)) + nonce))) and AuthSwitchRequest loop
+//  * prepared statements: COM_STMT_PREPARE / EXECUTE / CLOSE with binary
+//    result rows (common column types decoded, others hex-dumped)
+// Deltas vs reference: caching_sha2 full-auth (RSA password exchange on
+// cache miss over plaintext) is not implemented — use the fast path or
+// TLS; transactions work as plain queries (BEGIN/COMMIT/ROLLBACK).
 #pragma once
 
 #include <stdint.h>
