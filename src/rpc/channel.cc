@@ -125,10 +125,14 @@ static int HandleSessionError(SessionId id, void* data, int error_code) {
     session_unlock(id);
     return 0;
   }
+  // Cancellation is unconditionally terminal (parity: reference StartCancel
+  // semantics) — a user retry_policy must not resurrect a call the caller
+  // just canceled, so it is short-circuited before the policy hook.
   const bool policy_says_retry =
-      cntl->call.retry_policy
-          ? cntl->call.retry_policy(error_code, cntl->retry_count_)
-          : (error_code != ERPCTIMEDOUT && error_code != ECANCELED_RPC);
+      error_code != ECANCELED_RPC &&
+      (cntl->call.retry_policy
+           ? cntl->call.retry_policy(error_code, cntl->retry_count_)
+           : error_code != ERPCTIMEDOUT);
   if (policy_says_retry && cntl->retry_count_ < cntl->max_retry_) {
     ++cntl->retry_count_;
     session_bump_slot(id);

@@ -12,6 +12,7 @@
 #include <algorithm>
 #include <mutex>
 
+#include "base/flags.h"
 #include "base/flat_map.h"
 #include "base/logging.h"
 #include "base/time.h"
@@ -25,7 +26,30 @@ void EndRPC(Controller* cntl, SessionId locked_id);  // channel.cc
 
 namespace policy {
 
+// Parity: reference FLAGS_max_body_size (http_message.cpp) — a declared
+// Content-Length (or chunked total) beyond this is rejected outright so a
+// hostile peer cannot grow read_buf_ without bound.
+BAM_DEFINE_int64(http_max_body_size, 64 << 20,
+                 "Reject HTTP messages whose body exceeds this many bytes");
+
 namespace {
+
+// Strict non-negative decimal parse for Content-Length / chunk totals.
+// Returns false on empty/garbage/negative/overflow/over-limit values.
+bool parse_body_size(const char* s, size_t* out) {
+  if (s == nullptr || *s == '\0') return false;
+  uint64_t v = 0;
+  const char* p = s;
+  for (; *p != '\0' && *p != '\r' && *p != ' '; ++p) {
+    if (*p < '0' || *p > '9') return false;
+    if (v > (UINT64_MAX - 9) / 10) return false;
+    v = v * 10 + (uint64_t)(*p - '0');
+  }
+  if (p == s) return false;
+  if ((int64_t)v > FLAG_http_max_body_size) return false;
+  *out = (size_t)v;
+  return true;
+}
 
 struct HttpMessage : public InputMessageBase {
   HttpRequest req;
@@ -144,7 +168,14 @@ ParseResult ParseHttpResponse(IOBuf* source, Socket* /*sock*/, bool /*eof*/) {
     for (;;) {
       size_t le = all.find("\r\n", pos);
       if (le == std::string::npos) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
-      size_t chunk_len = strtoul(all.c_str() + pos, nullptr, 16);
+      char* cend = nullptr;
+      unsigned long long chunk_ull = strtoull(all.c_str() + pos, &cend, 16);
+      if (cend == all.c_str() + pos || all[pos] == '-' ||
+          chunk_ull > (unsigned long long)FLAG_http_max_body_size ||
+          body.size() + chunk_ull > (size_t)FLAG_http_max_body_size) {
+        return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+      }
+      size_t chunk_len = (size_t)chunk_ull;
       pos = le + 2;
       if (chunk_len == 0) {
         // trailer section ends with CRLF
@@ -162,7 +193,8 @@ ParseResult ParseHttpResponse(IOBuf* source, Socket* /*sock*/, bool /*eof*/) {
   } else {
     size_t content_len = 0;
     auto cl = headers.find("content-length");
-    if (cl != headers.end()) content_len = (size_t)atoll(cl->second.c_str());
+    if (cl != headers.end() && !parse_body_size(cl->second.c_str(), &content_len))
+      return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
     if (source->size() < hend + content_len)
       return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
     source->pop_front(hend);
@@ -256,7 +288,10 @@ ParseResult ParseHttpMessage(IOBuf* source, Socket* sock, bool eof) {
   }
   size_t content_len = 0;
   auto it = msg->req.headers.find("content-length");
-  if (it != msg->req.headers.end()) content_len = (size_t)atoll(it->second.c_str());
+  if (it != msg->req.headers.end() && !parse_body_size(it->second.c_str(), &content_len)) {
+    delete msg;
+    return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+  }
   if (source->size() < hend + content_len) {
     delete msg;
     return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
@@ -278,6 +313,28 @@ const char* status_reason(int code) {
     case 503: return "Service Unavailable";
     default: return "Unknown";
   }
+}
+
+// Serializes and writes one HTTP/1.1 response. Addressed by SocketId so it
+// is safe to run from ANY thread at ANY time (async handlers complete after
+// ProcessHttpRequest returned; the socket may have died meanwhile).
+void SendHttpResponseToSocket(SocketId sid, HttpResponse* resp, bool keep_alive) {
+  SocketUniquePtr sock;
+  if (Socket::Address(sid, &sock) != 0) return;
+  std::string head;
+  head.reserve(256);
+  head += "HTTP/1.1 " + std::to_string(resp->status) + " " + status_reason(resp->status) +
+          "\r\n";
+  head += "Content-Type: " + resp->content_type + "\r\n";
+  head += "Content-Length: " + std::to_string(resp->body.size()) + "\r\n";
+  for (const auto& kv : resp->headers) head += kv.first + ": " + kv.second + "\r\n";
+  if (!keep_alive) head += "Connection: close\r\n";
+  head += "\r\n";
+  IOBuf out;
+  out.append(head);
+  out.append(std::move(resp->body));
+  sock->Write(&out);
+  if (!keep_alive) sock->SetFailed(0, "connection: close");
 }
 
 void ProcessHttpRequest(InputMessageBase* msg_base) {
@@ -306,49 +363,46 @@ void ProcessHttpRequest(InputMessageBase* msg_base) {
     }
     const MethodFn* fn = server->FindMethod(svc, method);
     if (fn != nullptr) {
-      Controller cntl;
-      cntl.server_ = server;
-      cntl.server_socket_ = sock->id();
-      cntl.remote_side_ = sock->remote_side();
-      IOBuf http_resp_body;
-      // synchronous closure: handlers that run inline complete before return
-      std::atomic<bool> done_flag{false};
-      Closure* done = NewCallback([&done_flag] { done_flag.store(true); });
-      (*fn)(&cntl, msg->req.body, &http_resp_body, done);
-      // NOTE: async handlers over HTTP are not yet supported; spin briefly.
-      for (int i = 0; i < 100000 && !done_flag.load(std::memory_order_acquire); ++i) {
-        usleep(100);
-      }
-      if (cntl.Failed()) {
-        resp.status = 500;
-        resp.body.append(cntl.ErrorText());
-      } else {
-        resp.body.append(std::move(http_resp_body));
-        resp.content_type = "application/octet-stream";
-      }
-      handled = true;
+      // Heap-allocated call state: async handlers may run `done` from
+      // another fiber long after this function returned (the reference
+      // sends the response from the done closure too; a stack capture
+      // here was a use-after-free).
+      struct HttpCallCtx {
+        Controller cntl;
+        IOBuf resp_body;
+        SocketId sid;
+        bool keep_alive;
+      };
+      HttpCallCtx* ctx = new HttpCallCtx;
+      ctx->cntl.server_ = server;
+      ctx->cntl.server_socket_ = sock->id();
+      ctx->cntl.remote_side_ = sock->remote_side();
+      ctx->sid = sock->id();
+      ctx->keep_alive = msg->req.keep_alive;
+      Closure* done = NewCallback([ctx] {
+        HttpResponse r;
+        r.status = 200;
+        if (ctx->cntl.Failed()) {
+          r.status = 500;
+          r.content_type = "text/plain";
+          r.body.append(ctx->cntl.ErrorText());
+        } else {
+          r.content_type = "application/octet-stream";
+          r.body.append(std::move(ctx->resp_body));
+        }
+        SendHttpResponseToSocket(ctx->sid, &r, ctx->keep_alive);
+        delete ctx;
+      });
+      (*fn)(&ctx->cntl, msg->req.body, &ctx->resp_body, done);
+      delete msg;
+      return;
     }
   }
   if (!handled) {
     resp.status = 404;
     resp.body.append("no such page/method: " + msg->req.path + "\n");
   }
-
-  // serialize response
-  std::string head;
-  head.reserve(256);
-  head += "HTTP/1.1 " + std::to_string(resp.status) + " " + status_reason(resp.status) +
-          "\r\n";
-  head += "Content-Type: " + resp.content_type + "\r\n";
-  head += "Content-Length: " + std::to_string(resp.body.size()) + "\r\n";
-  for (const auto& kv : resp.headers) head += kv.first + ": " + kv.second + "\r\n";
-  if (!msg->req.keep_alive) head += "Connection: close\r\n";
-  head += "\r\n";
-  IOBuf out;
-  out.append(head);
-  out.append(std::move(resp.body));
-  sock->Write(&out);
-  if (!msg->req.keep_alive) sock->SetFailed(0, "connection: close");
+  SendHttpResponseToSocket(msg->socket_id, &resp, msg->req.keep_alive);
   delete msg;
 }
 

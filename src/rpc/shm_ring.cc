@@ -100,16 +100,22 @@ bool ring_push(Ring& r, uint64_t cid, uint32_t aux, const void* a, size_t alen,
   return true;
 }
 
-// Consumer: pops one record; false if empty.
-bool ring_pop(Ring& r, RecHdr* h, std::string* body) {
+// Consumer: pops one record. Returns 1 = got one, 0 = empty, -1 = the
+// segment is corrupt (body_len from shared memory fails validation — any
+// same-user process can scribble on the mapping; never trust it).
+int ring_pop(Ring& r, RecHdr* h, std::string* body) {
   uint64_t tail = r.tail->v.load(std::memory_order_relaxed);
   uint64_t head = r.head->v.load(std::memory_order_acquire);
-  if (head == tail) return false;
+  if (head == tail) return 0;
+  if (head - tail < sizeof(*h)) return -1;
   ring_copy_out(r, tail, h, sizeof(*h));
+  // A valid producer never writes a record larger than half the ring
+  // (ring_push refuses), and the record must fit in the published span.
+  if (h->body_len > r.bytes / 2 || rec_size(h->body_len) > head - tail) return -1;
   body->resize(h->body_len);
   if (h->body_len != 0) ring_copy_out(r, tail + sizeof(*h), &(*body)[0], h->body_len);
   r.tail->v.store(tail + rec_size(h->body_len), std::memory_order_release);
-  return true;
+  return 1;
 }
 
 // Adaptive wait: spin with pause, then nap. Returns false on stop().
@@ -221,7 +227,14 @@ void shm_server_poll(void* raw) {
             [&] { return srv->stopping.load(std::memory_order_acquire); })) {
       break;
     }
-    while (ring_pop(seg.req, &h, &body)) {
+    int got;
+    while ((got = ring_pop(seg.req, &h, &body)) != 0) {
+      if (got < 0) {
+        LOG(ERROR) << "shm segment corrupt (bad record header); closing server ring";
+        seg.hdr->closed.store(1, std::memory_order_release);
+        srv->stopping.store(true, std::memory_order_release);
+        break;
+      }
       // body = method + payload
       size_t mlen = h.aux <= body.size() ? h.aux : body.size();
       std::string full_method = body.substr(0, mlen);
@@ -318,6 +331,7 @@ struct ShmChannel::Impl {
   PendShard& shard_of(uint64_t cid) { return pend[cid % kPendShards]; }
   std::atomic<uint64_t> next_cid{1};
   std::atomic<bool> stopping{false};
+  std::atomic<bool> corrupt{false};  // poller saw an invalid record header
   CountdownEvent poller_exited{1};
 };
 
@@ -338,7 +352,14 @@ void shm_client_poll(void* raw) {
             })) {
       break;
     }
-    while (ring_pop(impl->seg.resp, &h, &body)) {
+    int got;
+    while ((got = ring_pop(impl->seg.resp, &h, &body)) != 0) {
+      if (got < 0) {
+        LOG(ERROR) << "shm segment corrupt (bad record header); abandoning client ring";
+        impl->corrupt.store(true, std::memory_order_release);
+        impl->stopping.store(true, std::memory_order_release);
+        break;
+      }
       std::shared_ptr<PendingCall> pc;
       {
         auto& sh = impl->shard_of(h.cid);
@@ -405,6 +426,7 @@ int ShmChannel::Call(const std::string& full_method, const IOBuf& request, IOBuf
     }
     if (pushed) break;
     if (monotonic_time_us() > deadline ||
+        impl_->corrupt.load(std::memory_order_acquire) ||
         impl_->seg.hdr->closed.load(std::memory_order_acquire) != 0) {
       auto& sh = impl_->shard_of(cid);
       std::lock_guard<std::mutex> lk(sh.mu);

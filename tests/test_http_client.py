@@ -57,3 +57,36 @@ def test_server_error_maps_to_500(port):
                                     "EchoService.Fail", b"x")
     assert rc == EHTTP
     assert "500" in err
+
+
+def test_hostile_content_length_rejected(port):
+    """a negative or absurd Content-Length must close the
+    connection instead of growing read_buf_ forever (reference enforces
+    FLAGS_max_body_size, http_message.cpp)."""
+    import socket as pysock
+    for cl in ("-1", "1152921504606846976", "99999999999999999999", "12x"):
+        s = pysock.create_connection(("127.0.0.1", port), timeout=5)
+        req = ("POST /EchoService/Echo HTTP/1.1\r\nHost: x\r\n"
+               "Content-Length: %s\r\n\r\n" % cl).encode()
+        s.sendall(req)
+        s.settimeout(5)
+        # Server must drop the connection (parse error), not wait for a body.
+        try:
+            data = s.recv(4096)
+        except pysock.timeout:
+            raise AssertionError("server kept connection open for Content-Length=%s" % cl)
+        assert data == b"", (cl, data)
+        s.close()
+
+
+def test_normal_content_length_still_works(port):
+    import socket as pysock
+    s = pysock.create_connection(("127.0.0.1", port), timeout=5)
+    body = b"hello"
+    s.sendall(b"POST /EchoService/Echo HTTP/1.1\r\nHost: x\r\n"
+              b"Content-Length: %d\r\n\r\n%s" % (len(body), body))
+    s.settimeout(5)
+    data = s.recv(65536)
+    assert b"200" in data.split(b"\r\n", 1)[0]
+    assert data.endswith(body)
+    s.close()

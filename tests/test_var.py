@@ -21,7 +21,15 @@ def test_server_method_status_exposed():
         assert rc == 0
     dump = v.dump_exposed(f"rpc_server_{port}_EchoService.Echo")
     assert "_count" in dump
-    count = v.describe(f"rpc_server_{port}_EchoService.Echo_count")
+    # The method counter is bumped when the server finishes the call; the
+    # client can observe the response a hair earlier — poll briefly.
+    import time
+    count = None
+    for _ in range(100):
+        count = v.describe(f"rpc_server_{port}_EchoService.Echo_count")
+        if count is not None and int(count) >= 10:
+            break
+        time.sleep(0.02)
     assert count is not None and int(count) >= 10
 
 
