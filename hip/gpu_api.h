@@ -52,6 +52,20 @@ int bam_gpu_snappy_decompress(const void* src_dev, size_t n, void* dst_dev, size
 // payload generation without H2D traffic).
 int bam_gpu_fill(void* dst_dev, size_t n, uint64_t pattern, int dev);
 
+// ---- fiber-wait integration (see src/fiber/gpu_wait.h) ----
+// wait(flag, want, dev, kind): park the calling fiber until *flag >= want.
+// Returns 0 once satisfied; nonzero = could not park (caller falls back to
+// spinning). wake(dev, kind) is invoked from a HIP host-callback thread
+// when the stream carrying (dev, kind) reaches a requested wake marker.
+typedef int (*bam_fiber_wait_fn)(const volatile unsigned long long* flag,
+                                 unsigned long long want, int dev, int kind);
+typedef void (*bam_fiber_wake_fn)(int dev, int kind);
+void bam_gpu_set_fiber_wait(bam_fiber_wait_fn wait, bam_fiber_wake_fn wake);
+// Enqueues a wake marker on (dev, kind)'s stream; 0 on success. Stream
+// order guarantees the marker fires after every op enqueued before the
+// call — in particular after the kernel that publishes the waited ticket.
+int bam_gpu_request_wake(int dev, int kind);
+
 // Last error string (static buffer).
 const char* bam_gpu_last_error(void);
 

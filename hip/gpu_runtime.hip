@@ -14,6 +14,7 @@
 #include <vector>
 
 #include "gpu_api.h"
+#include "internal.h"
 
 static char g_err[256] = {0};
 
@@ -69,6 +70,83 @@ struct ScopedDevice {
 };
 
 }  // namespace
+
+// ---------------- fiber-wait / wake infrastructure ----------------
+// See internal.h. The core runtime (src/fiber/gpu_wait.cc) registers a
+// park/wake pair; until it does, waits fall back to bounded spinning.
+
+namespace {
+
+bam_fiber_wait_fn g_fiber_wait = nullptr;
+bam_fiber_wake_fn g_fiber_wake = nullptr;
+
+struct WakeSlot {
+  hipStream_t stream = nullptr;  // set once at stream creation
+  int dev = 0;
+  int kind = 0;
+};
+WakeSlot g_wake[kMaxDev][bamhip::kWakeKinds];
+std::mutex g_wake_mu;
+
+void wake_trampoline(void* p) {
+  WakeSlot* s = (WakeSlot*)p;
+  bam_fiber_wake_fn wk = __atomic_load_n(&g_fiber_wake, __ATOMIC_ACQUIRE);
+  if (wk != nullptr) wk(s->dev, s->kind);
+}
+
+}  // namespace
+
+namespace bamhip {
+
+void register_wake_stream(int dev, int kind, hipStream_t stream) {
+  if (dev < 0 || dev >= kMaxDev || kind < 0 || kind >= kWakeKinds) return;
+  std::lock_guard<std::mutex> lk(g_wake_mu);
+  g_wake[dev][kind].dev = dev;
+  g_wake[dev][kind].kind = kind;
+  __atomic_store_n(&g_wake[dev][kind].stream, stream, __ATOMIC_RELEASE);
+}
+
+bool wait_ticket(const volatile unsigned long long* flag, unsigned long long want,
+                 int dev, int kind, hipStream_t sync_stream) {
+  // Short spin first: small staging batches complete in single-digit µs,
+  // far below a park/wake round trip.
+  for (int spin = 0; spin < 4000; ++spin) {
+    if (*flag >= want) return true;
+#if defined(__x86_64__)
+    __builtin_ia32_pause();
+#endif
+  }
+  bam_fiber_wait_fn fw = __atomic_load_n(&g_fiber_wait, __ATOMIC_ACQUIRE);
+  if (fw != nullptr && fw(flag, want, dev, kind) == 0 && *flag >= want) return true;
+  // No fiber runtime (plain pthread) or park unavailable: bounded spin with
+  // a hard stream sync as last resort.
+  for (uint64_t spin = 0; *flag < want; ++spin) {
+    if (spin > 4000000) {
+      if (sync_stream == nullptr) return false;
+      if (hipStreamSynchronize(sync_stream) != hipSuccess || *flag < want) return false;
+      break;
+    }
+#if defined(__x86_64__)
+    __builtin_ia32_pause();
+#endif
+  }
+  return true;
+}
+
+}  // namespace bamhip
+
+extern "C" void bam_gpu_set_fiber_wait(bam_fiber_wait_fn wait, bam_fiber_wake_fn wake) {
+  __atomic_store_n(&g_fiber_wake, wake, __ATOMIC_RELEASE);
+  __atomic_store_n(&g_fiber_wait, wait, __ATOMIC_RELEASE);
+}
+
+extern "C" int bam_gpu_request_wake(int dev, int kind) {
+  if (dev < 0 || dev >= kMaxDev || kind < 0 || kind >= bamhip::kWakeKinds) return -1;
+  WakeSlot* s = &g_wake[dev][kind];
+  hipStream_t stream = __atomic_load_n(&s->stream, __ATOMIC_ACQUIRE);
+  if (stream == nullptr) return -1;
+  return hipLaunchHostFunc(stream, wake_trampoline, s) == hipSuccess ? 0 : -1;
+}
 
 extern "C" void* bam_gpu_alloc_hbm(uint32_t cap, int dev) {
   if (dev < 0 || dev >= kMaxDev) return nullptr;
@@ -361,33 +439,15 @@ struct DirectState {
 };
 DirectState g_direct[kMaxDev];
 
-// Spin until the upload stream drained up to `want` (up_flag is pinned).
-bool upload_wait(DirectState& st, unsigned long long want) {
-  for (uint64_t spin = 0; *st.up_flag < want; ++spin) {
-    if (spin > 4000000) {
-      if (hipStreamSynchronize(st.up_stream) != hipSuccess || *st.up_flag < want) return false;
-      break;
-    }
-#if defined(__x86_64__)
-    __builtin_ia32_pause();
-#endif
-  }
-  return true;
+// Waits until the upload stream drained up to `want` (up_flag is pinned):
+// short spin, then park the fiber (bamhip::wait_ticket).
+bool upload_wait(DirectState& st, unsigned long long want, int dev) {
+  return bamhip::wait_ticket(st.up_flag, want, dev, bamhip::kWakeUpload, st.up_stream);
 }
 
-// Spin until the device has drained every launch up to `want`; bounded,
-// with a hard hipStreamSynchronize fallback. Call with st.mu held.
-bool direct_wait(DirectState& st, unsigned long long want) {
-  for (uint64_t spin = 0; *st.flag < want; ++spin) {
-    if (spin > 4000000) {
-      if (hipStreamSynchronize(st.stream) != hipSuccess || *st.flag < want) return false;
-      break;
-    }
-#if defined(__x86_64__)
-    __builtin_ia32_pause();
-#endif
-  }
-  return true;
+// Waits until the device drained every gather launch up to `want`.
+bool direct_wait(DirectState& st, unsigned long long want, int dev) {
+  return bamhip::wait_ticket(st.flag, want, dev, bamhip::kWakeGather, st.stream);
 }
 
 size_t direct_max_bytes() {
@@ -427,6 +487,7 @@ int split_spans(const void* const* srcs, const size_t* lens, int nspans, DirectA
 // host VAs on this platform/config, disable the path for good.
 bool direct_init(DirectState& st, int dev) {
   if (hipStreamCreateWithFlags(&st.stream, hipStreamNonBlocking) != hipSuccess) return false;
+  bamhip::register_wake_stream(dev, bamhip::kWakeGather, st.stream);
   if (hipMalloc(&st.counter_dev, 8) != hipSuccess) return false;
   if (hipMemset(st.counter_dev, 0, 8) != hipSuccess) return false;
   void* f = nullptr;
@@ -438,6 +499,7 @@ bool direct_init(DirectState& st, int dev) {
     return false;
   st.ring = (char*)ring;
   if (hipStreamCreateWithFlags(&st.up_stream, hipStreamNonBlocking) != hipSuccess) return false;
+  bamhip::register_wake_stream(dev, bamhip::kWakeUpload, st.up_stream);
   if (hipMalloc(&st.up_counter_dev, 8) != hipSuccess) return false;
   if (hipMemset(st.up_counter_dev, 0, 8) != hipSuccess) return false;
   void* uf = nullptr;
@@ -500,7 +562,7 @@ int gather_direct(void* host_dst, const void* const* srcs, const size_t* lens,
   // this wait is almost always already satisfied.
   {
     unsigned long long want = st.up_ticket.load(std::memory_order_acquire);
-    if (want != 0 && *st.up_flag < want && !upload_wait(st, want)) return 1;
+    if (want != 0 && *st.up_flag < want && !upload_wait(st, want, dev)) return 1;
   }
   unsigned long long ticket;
   {
@@ -536,7 +598,7 @@ int gather_direct(void* host_dst, const void* const* srcs, const size_t* lens,
       return 1;
     }
   }
-  if (!direct_wait(st, ticket)) {
+  if (!direct_wait(st, ticket, dev)) {
     std::lock_guard<std::mutex> lk(st.mu);
     st.status = -1;
     return 1;
@@ -561,7 +623,7 @@ int upload_direct(void* dst_dev, const void* src, size_t n, int dev) {
   std::lock_guard<std::mutex> lk(st.up_mu);
   const int slot = (int)(st.nslots_used++ % kUploadSlots);
   // ring full wrap: the previous occupant's kernel must have drained
-  if (!upload_wait(st, st.slot_ticket[slot])) return 1;
+  if (!upload_wait(st, st.slot_ticket[slot], dev)) return 1;
   char* sp = st.ring + (size_t)slot * kUploadSlotBytes;
   ::memcpy(sp, src, n);
   st.up_launched += 1;
@@ -587,13 +649,13 @@ extern "C" void bam_gpu_quiesce(int dev) {
     want = st.ticket;
   }
   unsigned long long up_want = st.up_ticket.load(std::memory_order_acquire);
-  if (up_want != 0 && *st.up_flag < up_want && !upload_wait(st, up_want)) {
+  if (up_want != 0 && *st.up_flag < up_want && !upload_wait(st, up_want, dev)) {
     std::lock_guard<std::mutex> lk(st.mu);
     st.status = -1;
     return;
   }
   if (*st.flag >= want) return;
-  if (!direct_wait(st, want)) {
+  if (!direct_wait(st, want, dev)) {
     std::lock_guard<std::mutex> lk(st.mu);
     st.status = -1;
   }

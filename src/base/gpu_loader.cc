@@ -8,6 +8,7 @@
 
 #include "base/iobuf.h"
 #include "base/logging.h"
+#include "fiber/gpu_wait.h"
 
 namespace bam {
 namespace gpu {
@@ -70,6 +71,21 @@ void do_load() {
                                      int))must_sym(h, "bam_gpu_snappy_decompress");
   g_api.last_error = (const char* (*)(void))must_sym(h, "bam_gpu_last_error");
   if (!g_error.empty()) return;
+
+  // Fiber↔stream integration (fiber/gpu_wait.h): give the HIP lib a
+  // park/wake pair so its ticket waits yield the worker instead of
+  // spinning; give the core the wake-marker enqueue hook.
+  auto set_fiber_wait = (void (*)(int (*)(const volatile unsigned long long*,
+                                          unsigned long long, int, int),
+                                  void (*)(int, int)))dlsym(h, "bam_gpu_set_fiber_wait");
+  auto request_wake = (int (*)(int, int))dlsym(h, "bam_gpu_request_wake");
+  if (set_fiber_wait != nullptr && request_wake != nullptr) {
+    gpu_wait_set_request_fn(request_wake);
+    set_fiber_wait(
+        [](const volatile unsigned long long* flag, unsigned long long want, int dev,
+           int kind) { return gpu_fiber_wait_u64(flag, (uint64_t)want, dev, kind); },
+        gpu_fiber_wake);
+  }
 
   g_ndev = g_api.device_count();
   g_loaded = true;

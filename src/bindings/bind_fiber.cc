@@ -16,6 +16,7 @@ bool semaphore_test();
 int rwlock_test(int nreaders, int nwriters, int iters);
 bool timer_test();
 bool fiber_key_test();
+bool gpu_wait_selftest();
 }  // namespace selftest
 }  // namespace bam
 
@@ -39,6 +40,8 @@ void bind_fiber(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   f.def("timer_test", &bam::selftest::timer_test, py::call_guard<py::gil_scoped_release>());
   f.def("key_test", &bam::selftest::fiber_key_test, py::call_guard<py::gil_scoped_release>());
+  f.def("gpu_wait_test", &bam::selftest::gpu_wait_selftest,
+        py::call_guard<py::gil_scoped_release>());
   f.def("concurrency", &bam::fiber_get_concurrency);
   f.def("set_concurrency", &bam::fiber_set_concurrency);
   f.def("count_created", &bam::fiber_count_created);
