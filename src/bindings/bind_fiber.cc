@@ -1,6 +1,7 @@
 #include <pybind11/pybind11.h>
 
 #include "fiber/fiber.h"
+#include "fiber/gpu_wait.h"
 #include "bindings/bind.h"
 
 namespace bam {
@@ -42,6 +43,8 @@ void bind_fiber(py::module_& m) {
   f.def("key_test", &bam::selftest::fiber_key_test, py::call_guard<py::gil_scoped_release>());
   f.def("gpu_wait_test", &bam::selftest::gpu_wait_selftest,
         py::call_guard<py::gil_scoped_release>());
+  f.def("gpu_wait_parks", &bam::gpu_wait_parks);
+  f.def("gpu_wait_wake_requests", &bam::gpu_wait_wake_requests);
   f.def("concurrency", &bam::fiber_get_concurrency);
   f.def("set_concurrency", &bam::fiber_set_concurrency);
   f.def("count_created", &bam::fiber_count_created);

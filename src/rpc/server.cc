@@ -2,6 +2,7 @@
 
 #include "rpc/ssl_util.h"
 #include "fiber/butex.h"
+#include "fiber/gpu_wait.h"
 #include "rpc/concurrency_limiter.h"
 
 #include <sys/socket.h>
@@ -183,8 +184,14 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
       "fiber_butex_waits", [] { return std::to_string(butex_total_waits()); });
   static var::PassiveStatus* g_wait_us = new var::PassiveStatus(
       "fiber_butex_wait_us", [] { return std::to_string(butex_total_wait_us()); });
+  static var::PassiveStatus* g_gpu_parks = new var::PassiveStatus(
+      "gpu_wait_parks", [] { return std::to_string(gpu_wait_parks()); });
+  static var::PassiveStatus* g_gpu_wakes = new var::PassiveStatus(
+      "gpu_wait_wake_requests", [] { return std::to_string(gpu_wait_wake_requests()); });
   (void)g_waits;
   (void)g_wait_us;
+  (void)g_gpu_parks;
+  (void)g_gpu_wakes;
   if (limiter_ == nullptr) {
     limiter_ = ConcurrencyLimiter::Create(options_.adaptive_max_concurrency);
   }

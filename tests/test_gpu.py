@@ -86,3 +86,16 @@ def test_stream_hbm_throughput():
     assert mbps > 100, mbps  # staging must not collapse below 0.1 GB/s
     mbps2 = b.core.stream.throughput_hbm(port, 64, 1 << 20, True)
     assert mbps2 > 50, mbps2
+
+
+def test_gpu_wait_parks_on_real_hardware():
+    """fiber/gpu_wait: long device-side work must PARK the waiter (butex via
+    hipLaunchHostFunc wake), not spin. A ~512 MB span-copy gather far
+    exceeds the short-spin budget, so parks/wake-requests must advance."""
+    f = b.core.fiber
+    parks0 = f.gpu_wait_parks()
+    wakes0 = f.gpu_wait_wake_requests()
+    assert g.gather_matches(512 << 20, 2 << 20, 0)
+    assert f.gpu_wait_parks() > parks0 or f.gpu_wait_wake_requests() > wakes0, (
+        "big gather completed without ever parking — wait_ticket is not "
+        "routing through fiber/gpu_wait")
