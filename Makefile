@@ -52,7 +52,8 @@ build/hip/%.o: hip/%.hip
 
 $(HIP_SO): $(HIP_OBJS)
 	@if [ -n "$(HIP_OBJS)" ]; then \
-	  $(HIPCC) --offload-arch=$(GPU_ARCH) -shared -fPIC $(HIP_OBJS) -o $@ ; \
+	  $(HIPCC) --offload-arch=$(GPU_ARCH) -shared -fPIC $(HIP_OBJS) \
+	    -L/opt/rocm/lib -lrccl -Wl,-rpath,/opt/rocm/lib -o $@ ; \
 	else \
 	  echo "no hip sources yet"; \
 	fi

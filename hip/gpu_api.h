@@ -66,6 +66,23 @@ void bam_gpu_set_fiber_wait(bam_fiber_wait_fn wait, bam_fiber_wake_fn wake);
 // call — in particular after the kernel that publishes the waited ticket.
 int bam_gpu_request_wake(int dev, int kind);
 
+// ---- RCCL collectives over xGMI (hip/comm.hip) ----
+// One communicator per (process, GPU); ops enqueue on the communicator's
+// dedicated stream and PARK the calling fiber on the completion ticket.
+// Buffers are device (HBM) pointers.
+int bam_comm_uid(char out[128]);
+void* bam_comm_create(int nranks, int rank, const char uid[128], int dev);
+void bam_comm_destroy(void* comm);
+int bam_comm_rank(void* comm);
+int bam_comm_nranks(void* comm);
+int bam_comm_broadcast(void* comm, void* buf_dev, size_t n, int root);
+int bam_comm_allgather(void* comm, const void* send_dev, void* recv_dev, size_t per_rank);
+int bam_comm_send(void* comm, const void* buf_dev, size_t n, int peer);
+int bam_comm_recv(void* comm, void* buf_dev, size_t n, int peer);
+int bam_comm_sendrecv(void* comm, const void* sbuf, size_t sn, int speer, void* rbuf,
+                      size_t rn, int rpeer);
+const char* bam_comm_last_error(void);
+
 // Last error string (static buffer).
 const char* bam_gpu_last_error(void);
 

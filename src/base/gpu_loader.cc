@@ -72,6 +72,21 @@ void do_load() {
   g_api.last_error = (const char* (*)(void))must_sym(h, "bam_gpu_last_error");
   if (!g_error.empty()) return;
 
+  // RCCL comm surface — optional (older lib builds lack it).
+  g_api.comm_uid = (int (*)(char*))dlsym(h, "bam_comm_uid");
+  g_api.comm_create = (void* (*)(int, int, const char*, int))dlsym(h, "bam_comm_create");
+  g_api.comm_destroy = (void (*)(void*))dlsym(h, "bam_comm_destroy");
+  g_api.comm_rank = (int (*)(void*))dlsym(h, "bam_comm_rank");
+  g_api.comm_nranks = (int (*)(void*))dlsym(h, "bam_comm_nranks");
+  g_api.comm_broadcast = (int (*)(void*, void*, size_t, int))dlsym(h, "bam_comm_broadcast");
+  g_api.comm_allgather =
+      (int (*)(void*, const void*, void*, size_t))dlsym(h, "bam_comm_allgather");
+  g_api.comm_send = (int (*)(void*, const void*, size_t, int))dlsym(h, "bam_comm_send");
+  g_api.comm_recv = (int (*)(void*, void*, size_t, int))dlsym(h, "bam_comm_recv");
+  g_api.comm_sendrecv = (int (*)(void*, const void*, size_t, int, void*, size_t,
+                                 int))dlsym(h, "bam_comm_sendrecv");
+  g_api.comm_last_error = (const char* (*)(void))dlsym(h, "bam_comm_last_error");
+
   // Fiber↔stream integration (fiber/gpu_wait.h): give the HIP lib a
   // park/wake pair so its ticket waits yield the worker instead of
   // spinning; give the core the wake-marker enqueue hook.

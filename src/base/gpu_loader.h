@@ -28,6 +28,18 @@ struct GpuApi {
   int (*snappy_compress)(const void*, size_t, void*, size_t, size_t*, int);
   int (*snappy_decompress)(const void*, size_t, void*, size_t, size_t*, int);
   const char* (*last_error)(void);
+  // RCCL collectives (hip/comm.hip); null on builds without librccl.
+  int (*comm_uid)(char*);
+  void* (*comm_create)(int, int, const char*, int);
+  void (*comm_destroy)(void*);
+  int (*comm_rank)(void*);
+  int (*comm_nranks)(void*);
+  int (*comm_broadcast)(void*, void*, size_t, int);
+  int (*comm_allgather)(void*, const void*, void*, size_t);
+  int (*comm_send)(void*, const void*, size_t, int);
+  int (*comm_recv)(void*, void*, size_t, int);
+  int (*comm_sendrecv)(void*, const void*, size_t, int, void*, size_t, int);
+  const char* (*comm_last_error)(void);
 };
 
 // Loads the library (idempotent). Returns device count (0 = no GPU or no

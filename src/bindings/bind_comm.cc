@@ -1,0 +1,200 @@
+// Python surface for CommGroup (in-framework RCCL/xGMI collectives +
+// TCP-mesh control plane). Handles are process-local ints.
+#include <pybind11/pybind11.h>
+
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <string>
+
+#include "base/gpu_loader.h"
+#include "base/time.h"
+#include "bindings/bind.h"
+#include "rpc/comm_group.h"
+
+namespace {
+
+std::mutex g_mu;
+std::map<int, bam::CommGroup*> g_groups;
+int g_next = 1;
+
+bam::CommGroup* get(int h) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  auto it = g_groups.find(h);
+  return it == g_groups.end() ? nullptr : it->second;
+}
+
+// ---- GPU-side selftests / micro-benches (no torch in the data path) ----
+
+// Broadcast + AllGather on HBM buffers, verified against host patterns.
+bool gpu_collective_roundtrip(int h, size_t n) {
+  bam::CommGroup* g = get(h);
+  const bam::gpu::GpuApi* api = bam::gpu::api();
+  if (g == nullptr || api == nullptr) return false;
+  const int rank = g->rank(), nranks = g->nranks(), dev = 0;
+  char* buf = (char*)api->alloc_hbm((uint32_t)n, dev);
+  char* gathered = (char*)api->alloc_hbm((uint32_t)(n * nranks), dev);
+  if (buf == nullptr || gathered == nullptr) return false;
+  std::string host(n, 0);
+  bool ok = true;
+  // Broadcast: root fills 0xB7-pattern, everyone must read it back.
+  if (rank == 0) {
+    for (size_t i = 0; i < n; ++i) host[i] = (char)(0xB7 ^ (i & 0xff));
+    api->memcpy_res(buf, 2, dev, host.data(), 0, 0, n);
+  }
+  ok = ok && g->Broadcast(buf, n, 0) == 0;
+  std::string back(n, 1);
+  api->memcpy_res(&back[0], 0, 0, buf, 2, dev, n);
+  for (size_t i = 0; ok && i < n; ++i)
+    ok = back[i] == (char)(0xB7 ^ (i & 0xff));
+  // AllGather: every rank contributes its rank byte.
+  std::string mine(n, (char)(0x40 + rank));
+  api->memcpy_res(buf, 2, dev, mine.data(), 0, 0, n);
+  ok = ok && g->AllGather(buf, gathered, n) == 0;
+  std::string all(n * nranks, 0);
+  api->memcpy_res(&all[0], 0, 0, gathered, 2, dev, n * nranks);
+  for (int r = 0; ok && r < nranks; ++r)
+    for (size_t i = 0; ok && i < n; ++i)
+      ok = all[(size_t)r * n + i] == (char)(0x40 + r);
+  api->free_hbm(buf, (uint32_t)n, dev);
+  api->free_hbm(gathered, (uint32_t)(n * nranks), dev);
+  return ok;
+}
+
+// p2p frame exchange throughput (config-3 shape): each iteration moves one
+// `frame` of HBM bytes to `peer` and receives one back (full duplex over
+// xGMI). Returns GB/s of payload moved out of this rank.
+double gpu_p2p_gbps(int h, int peer, size_t frame, int iters) {
+  bam::CommGroup* g = get(h);
+  const bam::gpu::GpuApi* api = bam::gpu::api();
+  if (g == nullptr || api == nullptr || iters <= 0) return -1;
+  const int dev = 0;
+  char* sbuf = (char*)api->alloc_hbm((uint32_t)frame, dev);
+  char* rbuf = (char*)api->alloc_hbm((uint32_t)frame, dev);
+  if (sbuf == nullptr || rbuf == nullptr) return -1;
+  api->fill(sbuf, frame, 0x5a5a5a5a5a5a5a5aULL, dev);
+  // warm
+  if (g->SendRecv(sbuf, frame, peer, rbuf, frame, peer) != 0) return -1;
+  int64_t t0 = bam::monotonic_time_us();
+  for (int i = 0; i < iters; ++i) {
+    if (g->SendRecv(sbuf, frame, peer, rbuf, frame, peer) != 0) return -1;
+  }
+  int64_t us = bam::monotonic_time_us() - t0;
+  api->free_hbm(sbuf, (uint32_t)frame, dev);
+  api->free_hbm(rbuf, (uint32_t)frame, dev);
+  return (double)frame * iters / (us * 1e-6) / 1e9;
+}
+
+}  // namespace
+
+void bind_comm(py::module_& m) {
+  auto c = m.def_submodule("comm");
+  c.def(
+      "create",
+      [](int nranks, int rank, const std::string& backend, const std::string& host,
+         int base_port, int dev, int timeout_ms) {
+        bam::CommGroup::Options o;
+        o.nranks = nranks;
+        o.rank = rank;
+        o.backend = backend;
+        o.host = host;
+        o.base_port = base_port;
+        o.dev = dev;
+        o.connect_timeout_ms = timeout_ms;
+        std::string err;
+        bam::CommGroup* g = bam::CommGroup::Create(o, &err);
+        if (g == nullptr) throw std::runtime_error("CommGroup: " + err);
+        std::lock_guard<std::mutex> lk(g_mu);
+        int h = g_next++;
+        g_groups[h] = g;
+        return h;
+      },
+      py::arg("nranks"), py::arg("rank"), py::arg("backend") = "tcp",
+      py::arg("host") = "127.0.0.1", py::arg("base_port") = 0, py::arg("dev") = 0,
+      py::arg("timeout_ms") = 30000, py::call_guard<py::gil_scoped_release>());
+  c.def("destroy", [](int h) {
+    bam::CommGroup* g = nullptr;
+    {
+      std::lock_guard<std::mutex> lk(g_mu);
+      auto it = g_groups.find(h);
+      if (it != g_groups.end()) {
+        g = it->second;
+        g_groups.erase(it);
+      }
+    }
+    delete g;
+  });
+  c.def("rank", [](int h) { return get(h)->rank(); });
+  c.def("nranks", [](int h) { return get(h)->nranks(); });
+  c.def(
+      "barrier", [](int h) { return get(h)->Barrier(); },
+      py::call_guard<py::gil_scoped_release>());
+  // Host-buffer collectives (backend "tcp"; used by the CPU tests).
+  c.def(
+      "broadcast",
+      [](int h, py::bytes data, size_t n, int root) {
+        bam::CommGroup* g = get(h);
+        std::string buf(data);
+        buf.resize(n);
+        int rc;
+        {
+          py::gil_scoped_release rel;
+          rc = g->Broadcast(&buf[0], n, root);
+        }
+        if (rc != 0) throw std::runtime_error("broadcast failed");
+        return py::bytes(buf);
+      },
+      py::arg("h"), py::arg("data"), py::arg("n"), py::arg("root"));
+  c.def(
+      "allgather",
+      [](int h, py::bytes data) {
+        bam::CommGroup* g = get(h);
+        std::string mine(data);
+        std::string all(mine.size() * g->nranks(), 0);
+        int rc;
+        {
+          py::gil_scoped_release rel;
+          rc = g->AllGather(mine.data(), &all[0], mine.size());
+        }
+        if (rc != 0) throw std::runtime_error("allgather failed");
+        return py::bytes(all);
+      },
+      py::arg("h"), py::arg("data"));
+  c.def(
+      "send",
+      [](int h, int peer, py::bytes data) {
+        std::string buf(data);
+        py::gil_scoped_release rel;
+        if (get(h)->Send(buf.data(), buf.size(), peer) != 0)
+          throw std::runtime_error("send failed");
+      },
+      py::arg("h"), py::arg("peer"), py::arg("data"));
+  c.def(
+      "recv",
+      [](int h, int peer, size_t n) {
+        std::string buf(n, 0);
+        {
+          py::gil_scoped_release rel;
+          if (get(h)->Recv(&buf[0], n, peer) != 0) throw std::runtime_error("recv failed");
+        }
+        return py::bytes(buf);
+      },
+      py::arg("h"), py::arg("peer"), py::arg("n"));
+  c.def(
+      "host_broadcast",
+      [](int h, py::bytes data, int root) {
+        std::string blob(data);
+        {
+          py::gil_scoped_release rel;
+          if (get(h)->HostBroadcast(&blob, root) != 0)
+            throw std::runtime_error("host_broadcast failed");
+        }
+        return py::bytes(blob);
+      },
+      py::arg("h"), py::arg("data"), py::arg("root"));
+  // GPU data-plane checks (backend "rccl").
+  c.def("gpu_collective_roundtrip", &gpu_collective_roundtrip, py::arg("h"), py::arg("n"),
+        py::call_guard<py::gil_scoped_release>());
+  c.def("gpu_p2p_gbps", &gpu_p2p_gbps, py::arg("h"), py::arg("peer"), py::arg("frame"),
+        py::arg("iters") = 20, py::call_guard<py::gil_scoped_release>());
+}

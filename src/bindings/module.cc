@@ -17,4 +17,5 @@ PYBIND11_MODULE(_core, m) {
   bind_json2pb(m);
   bind_thrift(m);
   bind_codecs(m);
+  bind_comm(m);
 }

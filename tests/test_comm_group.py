@@ -1,0 +1,91 @@
+"""CommGroup (src/rpc/comm_group.*): the in-framework multi-GPU group.
+Data plane = RCCL over xGMI on GPUs (tests/test_gpu.py); here the SAME
+collective API runs on the TCP-mesh backend across real processes, which
+is exactly the control plane the RCCL path bootstraps through."""
+import os
+import socket
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def free_port_block(n=4):
+    socks, ports = [], []
+    for _ in range(n):
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        socks.append(s)
+        ports.append(s.getsockname()[1])
+    for s in socks:
+        s.close()
+    return ports[0]
+
+
+WORKER = r"""
+import sys
+sys.path.insert(0, %r)
+import brpc_amd as b
+c = b.core.comm
+rank, nranks, port = int(sys.argv[1]), int(sys.argv[2]), int(sys.argv[3])
+h = c.create(nranks, rank, "tcp", "127.0.0.1", port)
+assert c.rank(h) == rank and c.nranks(h) == nranks
+
+# broadcast: root 0 pushes a blob
+blob = b"bcast-payload" * 100 if rank == 0 else b""
+out = c.broadcast(h, blob, 1300, 0)
+assert out == b"bcast-payload" * 100, (rank, len(out))
+
+# allgather: every rank contributes rank-stamped bytes
+mine = bytes([0x40 + rank]) * 512
+allb = c.allgather(h, mine)
+assert len(allb) == 512 * nranks
+for r in range(nranks):
+    assert allb[r * 512:(r + 1) * 512] == bytes([0x40 + r]) * 512, r
+
+# p2p ring: send to (rank+1) %% n, recv from (rank-1) %% n
+import threading
+nxt, prv = (rank + 1) %% nranks, (rank - 1) %% nranks
+msg = b"ring-%%d" %% rank
+got = [None]
+t = threading.Thread(target=lambda: c.send(h, nxt, msg))
+t.start()
+got[0] = c.recv(h, prv, len(b"ring-%%d" %% prv))
+t.join()
+assert got[0] == b"ring-%%d" %% prv, (rank, got[0])
+
+assert c.barrier(h) == 0
+
+# host control plane
+hb = c.host_broadcast(h, b"uid-bytes" if rank == 0 else b"", 0)
+assert hb == b"uid-bytes"
+c.destroy(h)
+print("RANK_OK", rank)
+"""
+
+
+@pytest.mark.parametrize("nranks", [2, 3])
+def test_comm_group_multiprocess(nranks):
+    port = free_port_block(8) + 100  # clear of the probe sockets
+    procs = []
+    for r in range(nranks):
+        procs.append(subprocess.Popen(
+            [sys.executable, "-c", WORKER % REPO, str(r), str(nranks), str(port)],
+            cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True))
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=120)
+        assert p.returncode == 0, (r, err[-2000:], out[-500:])
+        assert "RANK_OK %d" % r in out
+
+
+def test_comm_group_single_rank():
+    import brpc_amd as b
+    c = b.core.comm
+    h = c.create(1, 0, "tcp", "127.0.0.1", free_port_block() + 100)
+    assert c.allgather(h, b"solo") == b"solo"
+    out = c.broadcast(h, b"x", 1, 0)
+    assert out == b"x"
+    assert c.barrier(h) == 0
+    c.destroy(h)

@@ -99,3 +99,20 @@ def test_gpu_wait_parks_on_real_hardware():
     assert f.gpu_wait_parks() > parks0 or f.gpu_wait_wake_requests() > wakes0, (
         "big gather completed without ever parking — wait_ticket is not "
         "routing through fiber/gpu_wait")
+
+
+def test_rccl_single_rank_collectives():
+    """In-framework RCCL (hip/comm.hip): communicator init + broadcast +
+    allgather + self-sendrecv on HBM buffers, world size 1 (the 1-GPU box;
+    the driver's 8-GPU scale run exercises nranks=8 through bench.py)."""
+    import socket
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    c = b.core.comm
+    h = c.create(1, 0, "rccl", "127.0.0.1", port + 1 if port < 65000 else 30000)
+    try:
+        assert c.gpu_collective_roundtrip(h, 1 << 20)
+        gbps = c.gpu_p2p_gbps(h, 0, 1 << 20, 10)
+        assert gbps > 0.5, gbps  # self-exchange is a device copy
+    finally:
+        c.destroy(h)
