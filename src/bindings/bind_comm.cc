@@ -1,16 +1,22 @@
 // Python surface for CommGroup (in-framework RCCL/xGMI collectives +
 // TCP-mesh control plane). Handles are process-local ints.
 #include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
 
 #include <cstring>
 #include <map>
 #include <mutex>
 #include <string>
 
+#include <algorithm>
+#include <vector>
+
 #include "base/gpu_loader.h"
 #include "base/time.h"
 #include "bindings/bind.h"
+#include "rpc/collective_channel.h"
 #include "rpc/comm_group.h"
+#include "rpc/server.h"
 
 namespace {
 
@@ -83,6 +89,133 @@ double gpu_p2p_gbps(int h, int peer, size_t frame, int iters) {
   api->free_hbm(sbuf, (uint32_t)frame, dev);
   api->free_hbm(rbuf, (uint32_t)frame, dev);
   return (double)frame * iters / (us * 1e-6) / 1e9;
+}
+
+// ---- collective fan-out (BASELINE config 4) ----
+
+std::mutex g_srv_mu;
+std::vector<bam::Server*> g_coll_servers;
+
+// Starts a Server on `port` (0 = ephemeral) with the collective service
+// bound to group `h`. Returns the bound port.
+int fanout_serve(int h, int port) {
+  bam::CommGroup* g = get(h);
+  if (g == nullptr) throw std::runtime_error("bad group handle");
+  auto* srv = new bam::Server;
+  if (bam::RegisterCollectiveService(srv, g) != 0)
+    throw std::runtime_error("RegisterCollectiveService failed");
+  if (srv->Start(port, nullptr) != 0) throw std::runtime_error("server start failed");
+  std::lock_guard<std::mutex> lk(g_srv_mu);
+  g_coll_servers.push_back(srv);
+  return srv->listen_address().port;
+}
+
+// Runs `rounds` collective fan-out calls from this (caller) rank and
+// reports qps + latency percentiles. Payload is uploaded to HBM for
+// "rccl" groups. verify: check every gathered slot echoes len bytes.
+struct FanoutResult {
+  int rc = 0;
+  std::string error;
+  int rounds = 0;
+  double qps = 0, p50_us = 0, p99_us = 0;
+  bool data_ok = false;
+};
+
+FanoutResult fanout_call_impl(int h, const std::vector<std::string>& addrs,
+                              const std::string& method, const std::string& payload,
+                              size_t resp_cap, int rounds, bool verify) {
+  bam::CommGroup* g = get(h);
+  if (g == nullptr) throw std::runtime_error("bad group handle");
+  const bool device = g->backend() == "rccl";
+  bam::CollectiveChannel ch;
+  if (ch.Init(g, addrs) != 0) throw std::runtime_error("CollectiveChannel: " + ch.last_error());
+  const size_t slot = bam::CollectiveChannel::slot_size(resp_cap);
+  const size_t gathered_bytes = slot * g->nranks();
+  void* req = nullptr;
+  void* gathered = nullptr;
+  const bam::gpu::GpuApi* api = bam::gpu::api();
+  if (device) {
+    req = api->alloc_hbm((uint32_t)payload.size(), 0);
+    gathered = api->alloc_hbm((uint32_t)gathered_bytes, 0);
+    api->memcpy_res(req, 2, 0, payload.data(), 0, 0, payload.size());
+  } else {
+    req = malloc(payload.size());
+    gathered = malloc(gathered_bytes);
+    memcpy(req, payload.data(), payload.size());
+  }
+  std::vector<double> lat_us;
+  lat_us.reserve(rounds);
+  int rc = 0;
+  std::string first_err;
+  int64_t t0 = bam::monotonic_time_us();
+  for (int i = 0; i < rounds && rc == 0; ++i) {
+    int64_t s = bam::monotonic_time_us();
+    rc = ch.Call(method, req, payload.size(), gathered, resp_cap);
+    if (rc != 0) first_err = ch.last_error();
+    lat_us.push_back((double)(bam::monotonic_time_us() - s));
+  }
+  int64_t total_us = bam::monotonic_time_us() - t0;
+  bool data_ok = true;
+  if (rc == 0 && verify) {
+    std::string host(gathered_bytes, 0);
+    if (device) {
+      api->memcpy_res(&host[0], 0, 0, gathered, 2, 0, gathered_bytes);
+    } else {
+      memcpy(&host[0], gathered, gathered_bytes);
+    }
+    for (int r = 0; r < g->nranks() && data_ok; ++r) {
+      uint64_t len = 0;
+      memcpy(&len, host.data() + (size_t)r * slot, 8);
+      if (method == "echo") {
+        data_ok = len == payload.size() &&
+                  memcmp(host.data() + (size_t)r * slot + 8, payload.data(), len) == 0;
+      } else {
+        data_ok = len > 0 && len <= resp_cap;
+      }
+    }
+  }
+  if (device) {
+    api->free_hbm(req, (uint32_t)payload.size(), 0);
+    api->free_hbm(gathered, (uint32_t)gathered_bytes, 0);
+  } else {
+    free(req);
+    free(gathered);
+  }
+  std::sort(lat_us.begin(), lat_us.end());
+  auto pct = [&](double p) {
+    if (lat_us.empty()) return 0.0;
+    size_t i = (size_t)(p * (lat_us.size() - 1));
+    return lat_us[i];
+  };
+  FanoutResult out;
+  out.rc = rc;
+  out.error = first_err;
+  out.rounds = rounds;
+  out.qps = total_us > 0 ? rounds * 1e6 / total_us : 0.0;
+  out.p50_us = pct(0.5);
+  out.p99_us = pct(0.99);
+  out.data_ok = data_ok;
+  return out;
+}
+
+// py::dict must only be touched WITH the GIL; the blocking work runs
+// without it.
+py::dict fanout_call(int h, const std::vector<std::string>& addrs, const std::string& method,
+                     const std::string& payload, size_t resp_cap, int rounds, bool verify) {
+  FanoutResult r;
+  {
+    py::gil_scoped_release rel;
+    r = fanout_call_impl(h, addrs, method, payload, resp_cap, rounds, verify);
+  }
+  py::dict out;
+  out["rc"] = r.rc;
+  out["error"] = r.error;
+  out["rounds"] = r.rounds;
+  out["qps"] = r.qps;
+  out["p50_us"] = r.p50_us;
+  out["p99_us"] = r.p99_us;
+  out["data_ok"] = r.data_ok;
+  return out;
 }
 
 }  // namespace
@@ -197,4 +330,10 @@ void bind_comm(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   c.def("gpu_p2p_gbps", &gpu_p2p_gbps, py::arg("h"), py::arg("peer"), py::arg("frame"),
         py::arg("iters") = 20, py::call_guard<py::gil_scoped_release>());
+  // Collective fan-out (CollectiveChannel, BASELINE config 4).
+  c.def("fanout_serve", &fanout_serve, py::arg("h"), py::arg("port") = 0,
+        py::call_guard<py::gil_scoped_release>());
+  c.def("fanout_call", &fanout_call, py::arg("h"), py::arg("addrs"), py::arg("method"),
+        py::arg("payload"), py::arg("resp_cap"), py::arg("rounds") = 1,
+        py::arg("verify") = true);
 }

@@ -1,0 +1,88 @@
+// brpc_amd: CollectiveChannel — ParallelChannel fan-out riding RCCL over
+// xGMI (BASELINE config 4; VERDICT round-1 top item).
+//
+// Shape: N ranks, one process per GPU, joined in a CommGroup. Every rank
+// except the caller runs a Server with the collective service registered.
+// One logical call =
+//   1. caller (group rank 0) fans a tiny CONTROL RPC out to every other
+//      rank over TCP (the reference ParallelChannel fan-out,
+//      parallel_channel.cpp:90-186 — here it only carries metadata),
+//   2. the 16 KB-class PAYLOAD moves as ONE RCCL broadcast of the caller's
+//      HBM buffer over xGMI (not N unicasts),
+//   3. every rank runs the named device op on its GPU-resident copy
+//      (echo, snappy round-trip, ... — an extensible registry),
+//   4. responses return as ONE RCCL all-gather into the caller's HBM,
+//   5. control RPCs ack; the caller merges the gathered slots.
+// Fibers park during the collectives (fiber/gpu_wait); no worker blocks.
+// With a "tcp" CommGroup the same machinery runs on host buffers — that is
+// the CPU test path and the no-GPU fallback.
+//
+// Failure model: like any collective runtime, a dead rank fails the GROUP
+// (the control RPC's timeout detects it); per-call fail_limit semantics
+// live in the TCP ParallelChannel, which remains available.
+#pragma once
+
+#include <functional>
+#include <memory>
+#include <string>
+#include <vector>
+
+#include "rpc/channel.h"
+#include "rpc/comm_group.h"
+
+namespace bam {
+
+class Server;
+
+// A device-side method: transform req (len bytes, device-resident for
+// "rccl" groups, host for "tcp") into resp (≤ resp_cap bytes, same
+// residency). Returns 0 and *resp_len on success. `dev` is the local GPU.
+typedef std::function<int(const void* req, size_t len, void* resp, size_t resp_cap,
+                          size_t* resp_len, int dev)>
+    CollectiveMethodFn;
+
+// Registers a named method in the process-wide collective registry.
+// Built-ins (registered on first use): "echo", "snappy_echo"
+// (decompress + recompress — the 16KB+snappy config's server work).
+void RegisterCollectiveMethod(const std::string& name, CollectiveMethodFn fn);
+
+// Server side: adds the "Collective.Round" control method to `server` and
+// binds it to `group`. The server participates in one collective round per
+// control RPC. Not owned; `group` must outlive the server.
+int RegisterCollectiveService(Server* server, CommGroup* group);
+
+class CollectiveChannel {
+ public:
+  struct Options {
+    int32_t timeout_ms = 5000;  // control-RPC timeout (detects dead ranks)
+  };
+
+  // `group`: this process must be a member (any rank may call, but all
+  // callers must be the same rank — one caller per group).
+  // `server_addrs[i]`: TCP address of group-rank i's server ("" for the
+  // caller's own rank).
+  int Init(CommGroup* group, const std::vector<std::string>& server_addrs,
+           const Options* opt = nullptr);
+
+  // One fan-out round. req: caller's payload (device ptr for "rccl"
+  // groups). Each rank's response lands in gathered[rank] as
+  // <u64 actual_len><payload...> within a resp_cap-sized slot; `gathered`
+  // must hold nranks * slot_size(resp_cap) bytes (same residency as req).
+  // Returns 0, or a nonzero errno-style code.
+  int Call(const std::string& method, const void* req, size_t req_len, void* gathered,
+           size_t resp_cap);
+
+  static size_t slot_size(size_t resp_cap) { return 8 + ((resp_cap + 7) & ~(size_t)7); }
+
+  int nranks() const { return group_ != nullptr ? group_->nranks() : 0; }
+  const std::string& last_error() const { return err_; }
+
+ private:
+  CommGroup* group_ = nullptr;
+  Options opt_;
+  std::vector<std::unique_ptr<ChannelBase>> controls_;  // index = group rank
+  std::string err_;
+  uint32_t round_ = 0;
+};
+
+}  // namespace bam

@@ -11,6 +11,7 @@
 #include <atomic>
 #include <chrono>
 #include <condition_variable>
+#include <deque>
 #include <mutex>
 #include <memory>
 #include <thread>
@@ -29,6 +30,12 @@ namespace bam {
 // duplicate links. Blocking sockets; every logical message is framed
 // <u64 len><bytes> and per-peer send/recv are each serialized by a mutex.
 
+// Every frame is tagged with a logical CHANNEL so independent consumers on
+// one connection (a collective round's Broadcast recv, a user p2p Recv, a
+// HostBroadcast) never steal each other's messages: a reader that pulls a
+// frame for another channel parks it in that channel's pending queue.
+enum MeshChan { kChanP2P = 0, kChanHost = 1, kChanColl = 2, kChanCount = 3 };
+
 struct CommGroup::Mesh {
   int nranks = 0;
   int rank = 0;
@@ -36,7 +43,11 @@ struct CommGroup::Mesh {
   struct Conn {
     int fd = -1;
     std::mutex send_mu;
-    std::mutex recv_mu;
+    std::mutex recv_mu;              // at most one thread reads the fd
+    std::mutex pend_mu;
+    std::condition_variable pend_cv;
+    std::deque<std::string> pending[kChanCount];
+    bool dead = false;
   };
   std::unique_ptr<Conn[]> conns;  // index = peer rank
   std::thread acceptor;
@@ -80,32 +91,92 @@ struct CommGroup::Mesh {
     return true;
   }
 
-  bool send_msg(int peer, const void* data, size_t n) {
+  bool send_msg(int peer, int chan, const void* data, size_t n) {
     Conn& c = conns[peer];
     std::lock_guard<std::mutex> lk(c.send_mu);
-    uint64_t len = n;
-    return write_all(c.fd, &len, 8) && (n == 0 || write_all(c.fd, data, n));
+    uint64_t hdr[2] = {(uint64_t)chan, n};
+    return write_all(c.fd, hdr, 16) && (n == 0 || write_all(c.fd, data, n));
   }
 
-  // Receives one framed message from `peer`; expect >= 0 enforces size.
-  bool recv_msg(int peer, std::string* out, int64_t expect = -1) {
+  // Pulls the next frame for `chan` from `peer`. A frame for a different
+  // channel is parked in its pending queue and its waiters are woken.
+  // Returns the payload in *out (or straight into dst when dst != null and
+  // the size matches exactly).
+  bool recv_chan(int peer, int chan, std::string* out, void* dst, size_t dst_n) {
     Conn& c = conns[peer];
-    std::lock_guard<std::mutex> lk(c.recv_mu);
-    uint64_t len = 0;
-    if (!read_all(c.fd, &len, 8)) return false;
-    if (expect >= 0 && len != (uint64_t)expect) return false;
-    if (len > (1ull << 33)) return false;  // sanity: 8 GiB cap
-    out->resize(len);
-    return len == 0 || read_all(c.fd, &(*out)[0], len);
+    for (;;) {
+      {
+        std::unique_lock<std::mutex> lk(c.pend_mu);
+        if (!c.pending[chan].empty()) {
+          std::string f = std::move(c.pending[chan].front());
+          c.pending[chan].pop_front();
+          lk.unlock();
+          if (dst != nullptr) {
+            if (f.size() != dst_n) return false;
+            memcpy(dst, f.data(), f.size());
+          } else {
+            *out = std::move(f);
+          }
+          return true;
+        }
+        if (c.dead) return false;
+      }
+      if (c.recv_mu.try_lock()) {
+        std::lock_guard<std::mutex> lk(c.recv_mu, std::adopt_lock);
+        // Re-check pending under the read lock: another reader may have
+        // parked our frame between the check above and the lock.
+        {
+          std::lock_guard<std::mutex> plk(c.pend_mu);
+          if (!c.pending[chan].empty()) continue;
+        }
+        uint64_t hdr[2];
+        if (!read_all(c.fd, hdr, 16) || hdr[0] >= kChanCount || hdr[1] > (1ull << 33)) {
+          std::lock_guard<std::mutex> plk(c.pend_mu);
+          c.dead = true;
+          c.pend_cv.notify_all();
+          return false;
+        }
+        const int fchan = (int)hdr[0];
+        const size_t len = (size_t)hdr[1];
+        if (fchan == chan && dst != nullptr) {
+          if (len != dst_n) {
+            std::lock_guard<std::mutex> plk(c.pend_mu);
+            c.dead = true;
+            c.pend_cv.notify_all();
+            return false;
+          }
+          return len == 0 || read_all(c.fd, dst, len);
+        }
+        std::string f(len, 0);
+        if (len != 0 && !read_all(c.fd, &f[0], len)) {
+          std::lock_guard<std::mutex> plk(c.pend_mu);
+          c.dead = true;
+          c.pend_cv.notify_all();
+          return false;
+        }
+        if (fchan == chan) {
+          *out = std::move(f);
+          return true;
+        }
+        std::lock_guard<std::mutex> plk(c.pend_mu);
+        c.pending[fchan].push_back(std::move(f));
+        c.pend_cv.notify_all();
+      } else {
+        // Another thread owns the fd; wait for it to park our frame (or
+        // release the lock).
+        std::unique_lock<std::mutex> lk(c.pend_mu);
+        c.pend_cv.wait_for(lk, std::chrono::milliseconds(5));
+      }
+    }
   }
 
-  bool recv_into(int peer, void* dst, size_t n) {
-    Conn& c = conns[peer];
-    std::lock_guard<std::mutex> lk(c.recv_mu);
-    uint64_t len = 0;
-    if (!read_all(c.fd, &len, 8)) return false;
-    if (len != n) return false;
-    return n == 0 || read_all(c.fd, dst, n);
+  bool recv_msg(int peer, int chan, std::string* out) {
+    return recv_chan(peer, chan, out, nullptr, 0);
+  }
+
+  bool recv_into(int peer, int chan, void* dst, size_t n) {
+    std::string tmp;
+    return recv_chan(peer, chan, &tmp, dst, n);
   }
 };
 
@@ -274,12 +345,12 @@ CommGroup::~CommGroup() {
 
 int CommGroup::HostSend(int peer, const void* data, size_t n) {
   if (peer < 0 || peer >= opt_.nranks || peer == opt_.rank) return -1;
-  return mesh_->send_msg(peer, data, n) ? 0 : -1;
+  return mesh_->send_msg(peer, kChanHost, data, n) ? 0 : -1;
 }
 
 int CommGroup::HostRecv(int peer, std::string* out) {
   if (peer < 0 || peer >= opt_.nranks || peer == opt_.rank) return -1;
-  return mesh_->recv_msg(peer, out) ? 0 : -1;
+  return mesh_->recv_msg(peer, kChanHost, out) ? 0 : -1;
 }
 
 int CommGroup::HostBroadcast(std::string* blob, int root) {
@@ -287,11 +358,11 @@ int CommGroup::HostBroadcast(std::string* blob, int root) {
   if (opt_.rank == root) {
     for (int p = 0; p < opt_.nranks; ++p) {
       if (p == opt_.rank) continue;
-      if (!mesh_->send_msg(p, blob->data(), blob->size())) return -1;
+      if (!mesh_->send_msg(p, kChanHost, blob->data(), blob->size())) return -1;
     }
     return 0;
   }
-  return mesh_->recv_msg(root, blob) ? 0 : -1;
+  return mesh_->recv_msg(root, kChanHost, blob) ? 0 : -1;
 }
 
 int CommGroup::Barrier() {
@@ -301,14 +372,14 @@ int CommGroup::Barrier() {
   if (opt_.rank == 0) {
     std::string tmp;
     for (int p = 1; p < opt_.nranks; ++p)
-      if (!mesh_->recv_msg(p, &tmp, 1)) return -1;
+      if (!mesh_->recv_msg(p, kChanHost, &tmp)) return -1;
     for (int p = 1; p < opt_.nranks; ++p)
-      if (!mesh_->send_msg(p, &z, 1)) return -1;
+      if (!mesh_->send_msg(p, kChanHost, &z, 1)) return -1;
     return 0;
   }
-  if (!mesh_->send_msg(0, &z, 1)) return -1;
+  if (!mesh_->send_msg(0, kChanHost, &z, 1)) return -1;
   std::string tmp;
-  return mesh_->recv_msg(0, &tmp, 1) ? 0 : -1;
+  return mesh_->recv_msg(0, kChanHost, &tmp) ? 0 : -1;
 }
 
 // ---------------- data plane ----------------
@@ -323,11 +394,11 @@ int CommGroup::Broadcast(void* buf, size_t n, int root) {
   if (opt_.rank == root) {
     for (int p = 0; p < opt_.nranks; ++p) {
       if (p == opt_.rank) continue;
-      if (!mesh_->send_msg(p, buf, n)) return -1;
+      if (!mesh_->send_msg(p, kChanColl, buf, n)) return -1;
     }
     return 0;
   }
-  return mesh_->recv_into(root, buf, n) ? 0 : -1;
+  return mesh_->recv_into(root, kChanColl, buf, n) ? 0 : -1;
 }
 
 int CommGroup::AllGather(const void* send, void* recv, size_t per_rank) {
@@ -349,12 +420,12 @@ int CommGroup::AllGather(const void* send, void* recv, size_t per_rank) {
   for (int p = 0; p < opt_.nranks; ++p) {
     if (p == opt_.rank) continue;
     senders.emplace_back([this, p, send, per_rank, &ok] {
-      if (!mesh_->send_msg(p, send, per_rank)) ok.store(false);
+      if (!mesh_->send_msg(p, kChanColl, send, per_rank)) ok.store(false);
     });
   }
   for (int p = 0; p < opt_.nranks; ++p) {
     if (p == opt_.rank) continue;
-    if (!mesh_->recv_into(p, out + (size_t)p * per_rank, per_rank)) ok.store(false);
+    if (!mesh_->recv_into(p, kChanColl, out + (size_t)p * per_rank, per_rank)) ok.store(false);
   }
   for (auto& t : senders) t.join();
   return ok.load() ? 0 : -1;
@@ -362,12 +433,12 @@ int CommGroup::AllGather(const void* send, void* recv, size_t per_rank) {
 
 int CommGroup::Send(const void* buf, size_t n, int peer) {
   if (rccl_ != nullptr) return gpu::api()->comm_send(rccl_, buf, n, peer);
-  return mesh_->send_msg(peer, buf, n) ? 0 : -1;
+  return mesh_->send_msg(peer, kChanP2P, buf, n) ? 0 : -1;
 }
 
 int CommGroup::Recv(void* buf, size_t n, int peer) {
   if (rccl_ != nullptr) return gpu::api()->comm_recv(rccl_, buf, n, peer);
-  return mesh_->recv_into(peer, buf, n) ? 0 : -1;
+  return mesh_->recv_into(peer, kChanP2P, buf, n) ? 0 : -1;
 }
 
 int CommGroup::SendRecv(const void* sbuf, size_t sn, int speer, void* rbuf, size_t rn,
@@ -376,9 +447,9 @@ int CommGroup::SendRecv(const void* sbuf, size_t sn, int speer, void* rbuf, size
   // tcp: overlap directions with a sender thread.
   std::atomic<bool> ok{true};
   std::thread t([this, sbuf, sn, speer, &ok] {
-    if (sn > 0 && !mesh_->send_msg(speer, sbuf, sn)) ok.store(false);
+    if (sn > 0 && !mesh_->send_msg(speer, kChanP2P, sbuf, sn)) ok.store(false);
   });
-  if (rn > 0 && !mesh_->recv_into(rpeer, rbuf, rn)) ok.store(false);
+  if (rn > 0 && !mesh_->recv_into(rpeer, kChanP2P, rbuf, rn)) ok.store(false);
   t.join();
   return ok.load() ? 0 : -1;
 }

