@@ -1,16 +1,23 @@
 #!/usr/bin/env python3
-"""Flagship benchmark (driver contract): echo RPC QPS with IOBuf-in-HBM.
+"""Flagship benchmark (driver contract): the BASELINE headline metric —
+"echo QPS + p99 latency (64B & 16KB payload); streaming_rpc GB/s at 8 GPUs"
+— measured through brpc_amd itself (no torch in any data path; torch.
+distributed is used only for rank rendezvous/aggregation).
 
-BASELINE.json metric: "echo QPS + p99 latency (64B & 16KB payload)".
-This measures config 2 (multi_threaded_echo, 64 B payload, IOBuf blocks in
-HBM3E, baidu_std-compatible wire protocol) on N GPUs of one node: each rank
-runs an in-process echo server + multi-fiber client over loopback TCP with
-request payloads staged into HBM blocks and responses staged back out of
-HBM by the server (weak scaling: per-GPU work fixed).
+Default mode (the driver's run):
+  * per rank: in-process echo server + multi-fiber client over loopback
+    TCP, payloads staged through HBM IOBuf blocks on a GPU box
+    (config 2). The TIMED K steps are 64 B echo batches (headline value =
+    aggregate 64 B QPS over all ranks); a 16 KB echo measurement and — for
+    world>1 — an in-framework streaming GB/s measurement (RCCL p2p over
+    xGMI between ranks 0<->1, config 3) and an 8-way collective fan-out
+    (RCCL broadcast/all-gather, 16KB+snappy, config 4) run untimed and are
+    reported inside config{}.
+  * --mode stream / --mode fanout make those the TIMED headline instead.
 
-Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
-       (N>1 is launched by the driver via torch.distributed.run, one rank
-        per GPU over RCCL; RANK/LOCAL_RANK/WORLD_SIZE read from env.)
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--mode M]
+       (N>1 launched via torch.distributed.run, one rank per GPU; reads
+        RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the env.)
 One JSON line is printed by rank 0.
 """
 import argparse
@@ -36,18 +43,33 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import brpc_amd as b  # noqa: E402
 
 
+def run_echo_step(addr, payload, concurrency, calls, method, pooled):
+    last = None
+    for _attempt in range(2):  # one retry: a loaded box can time out a stray call
+        res = b.core.rpc.echo_bench(addr, payload, concurrency, calls, 30000,
+                                    method, False, pooled)
+        if not res["errors"]:
+            return res
+        last = res
+    raise RuntimeError("bench errors: n=%s first=%s" % (last["errors"], last.get("first_error")))
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--mode", choices=["echo", "stream", "fanout"], default="echo")
     ap.add_argument("--payload", type=int, default=64)
     ap.add_argument("--concurrency", type=int, default=32)
     ap.add_argument("--calls-per-step", type=int, default=5000)
+    ap.add_argument("--frame-mb", type=int, default=1)
+    ap.add_argument("--frames-per-step", type=int, default=64)
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
+    master_port = int(os.environ.get("MASTER_PORT", "29500"))
 
     dist = None
     torch = None
@@ -69,35 +91,7 @@ def main():
 
     ndev = b.core.gpu.initialize()
     use_gpu = ndev > 0
-    if use_gpu:
-        # IOBuf-in-HBM: the server stores every response payload in
-        # HBM-resident IOBuf blocks; the socket write path stages them out
-        # through the device-gather + pinned-ring leg (hip/).
-        method, hbm_req = "EchoService.EchoHbm", False
-    else:
-        # CPU-only container: same benchmark on the host path.
-        method, hbm_req = "EchoService.Echo", False
-
-    port = b.core.rpc.start_echo_server(0)
-    addr = "127.0.0.1:%d" % port
-
-    # Pooled connections are the reference's highest-throughput mode for the
-    # host path; the HBM path keeps a single connection so KeepWrite
-    # coalesces GPU staging across requests (one gather+D2H per batch).
-    pooled = not use_gpu
-
-    def run_step():
-        last = None
-        for attempt in range(2):  # one retry: a loaded box can time out a stray call
-            res = b.core.rpc.echo_bench(addr, args.payload, args.concurrency,
-                                        args.calls_per_step, 30000, method, hbm_req, pooled)
-            if not res["errors"]:
-                return res
-            last = res
-        raise RuntimeError("bench errors: n=%s first=%s" % (last["errors"], last.get("first_error")))
-
-    for _ in range(args.warmup):
-        run_step()
+    comm_backend = "rccl" if use_gpu else "tcp"
 
     def barrier_sync():
         if dist is not None:
@@ -105,65 +99,198 @@ def main():
             if torch is not None and torch.cuda.is_available():
                 torch.cuda.synchronize()
 
-    barrier_sync()
-    t0 = time.monotonic()
-    p99s, qps_acc = [], []
-    total_calls = 0
-    for _ in range(args.steps):
-        res = run_step()
-        p99s.append(res["p99_us"])
-        qps_acc.append(res["qps"])
-        total_calls += res["total"]
-    barrier_sync()
-    elapsed = time.monotonic() - t0
-
-    # MAX elapsed over ranks; SUM of calls over ranks. RCCL (backend
-    # "nccl") reduces DEVICE tensors only — CPU tensors here would abort
-    # the driver's 8-GPU scaling run.
-    if dist is not None:
-        import torch
-
-        dev = torch.device("cuda", 0) if dist.get_backend() == "nccl" else torch.device("cpu")
+    def reduce_max_sum(elapsed, total):
+        # MAX elapsed over ranks; SUM of totals. RCCL reduces DEVICE
+        # tensors only.
+        if dist is None:
+            return elapsed, total
+        dev = (torch.device("cuda", 0) if dist.get_backend() == "nccl"
+               else torch.device("cpu"))
         te = torch.tensor([elapsed], dtype=torch.float64, device=dev)
-        tc = torch.tensor([float(total_calls)], dtype=torch.float64, device=dev)
+        tc = torch.tensor([float(total)], dtype=torch.float64, device=dev)
         dist.all_reduce(te, op=dist.ReduceOp.MAX)
         dist.all_reduce(tc, op=dist.ReduceOp.SUM)
-        elapsed = te.item()
-        total_calls = int(tc.item())
+        return te.item(), tc.item()
 
-    qps = total_calls / elapsed
-    ms_per_step = elapsed * 1000.0 / args.steps
-    p99_us = max(p99s)
+    # In-framework comm group across ranks (the REAL multi-GPU data path:
+    # RCCL over xGMI; tcp mesh on CPU test boxes). Used by stream/fanout.
+    group = None
+    if world_size > 1:
+        group = b.core.comm.create(world_size, rank, comm_backend,
+                                   "127.0.0.1", master_port + 171)
+
+    out = {
+        "metric": "echo_qps",
+        "unit": "requests/s",
+        "n_gpus": world_size if world_size > 1 else args.gpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bytes",
+        "data": "synthetic",
+    }
+    cfg = {}
+
+    if args.mode == "stream":
+        # Config 3 headline: 1 MB frames rank1 -> rank0 over the comm data
+        # plane (RCCL p2p / xGMI on GPUs) through the framework's stream
+        # layer. Other ranks idle at the barriers.
+        assert world_size >= 2, "--mode stream needs >= 2 ranks"
+        frame = args.frame_mb << 20
+        sport = None
+        if rank == 0:
+            sport = b.core.comm.stream_comm_serve(group, 1)
+            b.core.comm.send(group, 1, ("%06d" % sport).encode())
+        elif rank == 1:
+            sport = int(b.core.comm.recv(group, 0, 6).decode())
+
+        def stream_step():
+            if rank == 1:
+                gbps = b.core.comm.stream_comm_send(
+                    group, "127.0.0.1:%d" % sport, 0, args.frames_per_step, frame)
+                if gbps <= 0:
+                    raise RuntimeError("stream send failed rc=%s" % gbps)
+
+        for _ in range(args.warmup):
+            stream_step()
+            b.core.comm.barrier(group)
+        barrier_sync()
+        t0 = time.monotonic()
+        for _ in range(args.steps):
+            stream_step()
+            b.core.comm.barrier(group)
+        barrier_sync()
+        elapsed = time.monotonic() - t0
+        moved = args.steps * args.frames_per_step * frame if rank == 1 else 0
+        elapsed, moved = reduce_max_sum(elapsed, moved)
+        out["metric"] = "stream_gbps"
+        out["unit"] = "GB/s"
+        out["value"] = moved / elapsed / 1e9
+        out["ms_per_step"] = elapsed * 1000.0 / args.steps
+        cfg = {"model": "streaming_rpc (STRM frames, payload over %s)" % comm_backend,
+               "frame_bytes": frame, "frames_per_step": args.frames_per_step,
+               "direction": "rank1->rank0",
+               "parallelism": "p2p (2 of %d ranks)" % world_size}
+
+    elif args.mode == "fanout":
+        # Config 4 headline: collective fan-out (broadcast + per-rank
+        # snappy_echo + all-gather) with a 16 KB snappy payload.
+        assert world_size >= 2, "--mode fanout needs >= 2 ranks"
+        payload = bytes(range(256)) * 64  # 16 KB
+        comp = b.core.snappy.compress(payload)
+        resp_cap = len(comp) + 256
+        rounds_per_step = max(1, args.calls_per_step // 100)
+        if rank == 0:
+            addrs = [""]
+            for r in range(1, world_size):
+                addrs.append("127.0.0.1:%d" %
+                             int(b.core.comm.recv(group, r, 6).decode()))
+        else:
+            port = b.core.comm.fanout_serve(group, 0)
+            b.core.comm.send(group, 0, ("%06d" % port).encode())
+            addrs = None
+
+        def fanout_step():
+            if rank == 0:
+                res = b.core.comm.fanout_call(group, addrs, "snappy_echo", comp,
+                                              resp_cap, rounds_per_step, False)
+                if res["rc"] != 0:
+                    raise RuntimeError("fanout failed: %s" % res["error"])
+                return res
+            return None
+
+        last = None
+        for _ in range(args.warmup):
+            last = fanout_step()
+        barrier_sync()
+        t0 = time.monotonic()
+        for _ in range(args.steps):
+            last = fanout_step()
+        barrier_sync()
+        elapsed = time.monotonic() - t0
+        calls = args.steps * rounds_per_step if rank == 0 else 0
+        elapsed, calls = reduce_max_sum(elapsed, calls)
+        out["metric"] = "fanout_qps"
+        out["value"] = calls / elapsed
+        out["ms_per_step"] = elapsed * 1000.0 / args.steps
+        cfg = {"model": "ParallelChannel-over-RCCL fan-out (snappy_echo)",
+               "payload_bytes": len(payload), "compressed_bytes": len(comp),
+               "rounds_per_step": rounds_per_step,
+               "p99_us": last["p99_us"] if last else None,
+               "parallelism": "collective fan-out x%d ranks" % world_size}
+
+    else:  # echo (default; the driver's headline run)
+        if use_gpu:
+            method = "EchoService.EchoHbm"  # responses HBM-resident
+        else:
+            method = "EchoService.Echo"
+        port = b.core.rpc.start_echo_server(0)
+        addr = "127.0.0.1:%d" % port
+        pooled = not use_gpu
+
+        for _ in range(args.warmup):
+            run_echo_step(addr, args.payload, args.concurrency,
+                          args.calls_per_step, method, pooled)
+
+        # ---- timed region: 64 B echo (headline) ----
+        barrier_sync()
+        t0 = time.monotonic()
+        p99s, total_calls = [], 0
+        for _ in range(args.steps):
+            res = run_echo_step(addr, args.payload, args.concurrency,
+                                args.calls_per_step, method, pooled)
+            p99s.append(res["p99_us"])
+            total_calls += res["total"]
+        barrier_sync()
+        elapsed = time.monotonic() - t0
+        elapsed, total_calls = reduce_max_sum(elapsed, total_calls)
+
+        # ---- untimed extras: the rest of the BASELINE metric ----
+        # 16 KB echo (other half of the echo headline)
+        res16 = run_echo_step(addr, 16384, args.concurrency,
+                              max(200, args.calls_per_step // 4), method, pooled)
+        e16, q16 = reduce_max_sum(1.0, res16["qps"])  # sum of per-rank QPS
+        # streaming GB/s between ranks 0<->1 through the framework
+        stream_gbps = None
+        if group is not None:
+            if rank == 0:
+                sport = b.core.comm.stream_comm_serve(group, 1)
+                b.core.comm.send(group, 1, ("%06d" % sport).encode())
+            elif rank == 1:
+                sport = int(b.core.comm.recv(group, 0, 6).decode())
+            b.core.comm.barrier(group)
+            g = 0.0
+            if rank == 1:
+                g = b.core.comm.stream_comm_send(group, "127.0.0.1:%d" % sport,
+                                                 0, 48, 1 << 20)
+            b.core.comm.barrier(group)
+            _, stream_gbps = reduce_max_sum(1.0, g if rank == 1 else 0.0)
+
+        out["value"] = total_calls / elapsed
+        out["ms_per_step"] = elapsed * 1000.0 / args.steps
+        cfg = {
+            "model": ("multi_threaded_echo (baidu_std wire, IOBuf-in-HBM)"
+                      if use_gpu else "multi_threaded_echo (baidu_std wire, host IOBuf)"),
+            "payload_bytes": args.payload,
+            "concurrency_per_gpu": args.concurrency,
+            "calls_per_step": args.calls_per_step,
+            "p99_us": max(p99s),
+            "echo16k_qps": q16,
+            "echo16k_p99_us": res16["p99_us"],
+            "stream_gbps": stream_gbps,
+            "connection_type": "pooled" if pooled else "single",
+            "parallelism": "dp%d" % (world_size if world_size > 1 else 1),
+            "gpu_payload_residency": "HBM (server response payloads)" if use_gpu else "host",
+        }
 
     if rank == 0:
-        out = {
-            "metric": "echo_qps",
-            "value": qps,
-            "unit": "requests/s",
-            "n_gpus": world_size if world_size > 1 else args.gpus,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": ms_per_step,
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": None,
-            "dtype": "bytes",
-            "data": "synthetic",
-            "config": {
-                "model": "multi_threaded_echo (baidu_std wire, IOBuf-in-HBM)"
-                if use_gpu else "multi_threaded_echo (baidu_std wire, host IOBuf)",
-                "payload_bytes": args.payload,
-                "concurrency_per_gpu": args.concurrency,
-                "calls_per_step": args.calls_per_step,
-                "p99_us": p99_us,
-                "connection_type": "pooled" if pooled else "single",
-                "parallelism": "dp%d" % (world_size if world_size > 1 else 1),
-                "gpu_payload_residency": "HBM (server response payloads)" if use_gpu else "host",
-            },
-        }
+        out["config"] = cfg
         print(json.dumps(out), flush=True)
 
     if dist is not None:
+        dist.barrier()
         dist.destroy_process_group()
 
 

@@ -396,6 +396,20 @@ int IOBuf::append_with_residency(const void* host_data, size_t n, Residency res,
   return 0;
 }
 
+int IOBuf::append_writable_block(size_t n, Residency res, int dev, void** out_ptr) {
+  if (n == 0 || n > (1u << 31)) return -1;
+  Block* b = create_block((uint32_t)n, res, dev);
+  if (b == nullptr) return -1;
+  b->size = (uint32_t)n;
+  push_ref_back(BlockRef{0, (uint32_t)n, b});
+  if (out_ptr != nullptr) *out_ptr = b->data;
+  return 0;
+}
+
+int IOBuf::append_device_block(size_t n, int dev, void** out_ptr) {
+  return append_writable_block(n, RES_HBM, dev, out_ptr);
+}
+
 // ---------------- cut / pop ----------------
 
 size_t IOBuf::cutn(IOBuf* out, size_t n) {

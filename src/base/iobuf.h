@@ -99,6 +99,11 @@ class IOBuf {
   // data into it (uses byte mover for HBM). Appends the written range.
   int append_with_residency(const void* host_data, size_t n, Residency res, int dev,
                             uint32_t block_payload = 0);
+  // Allocate ONE uninitialized block of n bytes with the given residency,
+  // append it, and return its raw pointer (for transports that land
+  // payloads straight into place: RCCL recv, GPUDirect). n ≤ 2 GiB.
+  int append_writable_block(size_t n, Residency res, int dev, void** out_ptr);
+  int append_device_block(size_t n, int dev, void** out_ptr);  // = RES_HBM
 
   // ---- cutting (front) ----
   // Move up to n bytes from the front of *this to the back of *out.

@@ -17,6 +17,7 @@
 #include "rpc/collective_channel.h"
 #include "rpc/comm_group.h"
 #include "rpc/server.h"
+#include "rpc/stream.h"
 
 namespace {
 
@@ -218,6 +219,89 @@ py::dict fanout_call(int h, const std::vector<std::string>& addrs, const std::st
   return out;
 }
 
+// ---- streaming over the comm data plane (BASELINE config 3) ----
+
+// Receiver: a Server whose "StreamService.OpenComm" accepts streams bound
+// to group `h` with data arriving from group-rank `peer` (RCCL p2p over
+// xGMI for "rccl" groups). Frames land in HBM blocks; the sink counts
+// bytes and releases them.
+int stream_comm_serve(int h, int peer, int port) {
+  bam::CommGroup* g = get(h);
+  if (g == nullptr) throw std::runtime_error("bad group handle");
+  auto* srv = new bam::Server;
+  auto* svc = new bam::Service("StreamService");
+  svc->AddMethod("OpenComm", [g, peer](bam::Controller* cntl, const bam::IOBuf& req,
+                                       bam::IOBuf* resp, bam::Closure* done) {
+    bam::StreamOptions sopt;
+    sopt.max_buf_size = 256u << 20;
+    sopt.gpu_group = g;
+    sopt.gpu_peer = peer;
+    sopt.on_received = [](bam::StreamId, bam::IOBuf* msg) { msg->clear(); };
+    sopt.on_closed = [](bam::StreamId sid) { bam::StreamClose(sid); };
+    bam::StreamId sid;
+    if (bam::StreamAccept(&sid, cntl, sopt) != 0) cntl->SetFailed(1003, "no stream");
+    resp->append("ok");
+    done->Run();
+  });
+  srv->AddService(svc, bam::SERVER_OWNS_SERVICE);
+  if (srv->Start(port, nullptr) != 0) throw std::runtime_error("server start failed");
+  std::lock_guard<std::mutex> lk(g_srv_mu);
+  g_coll_servers.push_back(srv);
+  return srv->listen_address().port;
+}
+
+// Sender: streams `frames` frames of `frame_size` bytes to the server at
+// `addr`, payload moving over the group's data plane to `peer`. Returns
+// GB/s of payload.
+double stream_comm_send(int h, const std::string& addr, int peer, int frames,
+                        size_t frame_size) {
+  bam::CommGroup* g = get(h);
+  if (g == nullptr) throw std::runtime_error("bad group handle");
+  const bool device = g->backend() == "rccl";
+  bam::Channel channel;
+  bam::ChannelOptions copt;
+  copt.timeout_ms = 20000;
+  if (channel.Init(addr.c_str(), &copt) != 0) return -1;
+  bam::StreamOptions sopt;
+  sopt.max_buf_size = 256u << 20;
+  sopt.gpu_group = g;
+  sopt.gpu_peer = peer;
+  bam::Controller cntl;
+  bam::StreamId sid;
+  bam::StreamCreate(&sid, &cntl, sopt);
+  bam::IOBuf request, response;
+  request.append("open");
+  channel.CallMethod("StreamService.OpenComm", &cntl, &request, &response, nullptr);
+  if (cntl.Failed()) {
+    bam::StreamClose(sid);
+    return -2;
+  }
+  const bam::gpu::GpuApi* api = bam::gpu::api();
+  std::string host_frame(frame_size, 'X');
+  int64_t t0 = bam::monotonic_time_us();
+  for (int i = 0; i < frames; ++i) {
+    bam::IOBuf data;
+    void* p = nullptr;
+    if (data.append_writable_block(frame_size, device ? bam::RES_HBM : bam::RES_HOST, 0,
+                                   &p) != 0) {
+      bam::StreamClose(sid);
+      return -4;
+    }
+    if (device) {
+      if (i == 0) api->fill(p, frame_size, 0x5858585858585858ULL, 0);
+    } else {
+      memcpy(p, host_frame.data(), frame_size);
+    }
+    if (bam::StreamWrite(sid, &data) != 0) {
+      bam::StreamClose(sid);
+      return -3;
+    }
+  }
+  int64_t us = bam::monotonic_time_us() - t0;
+  bam::StreamClose(sid);
+  return (double)frames * frame_size / (us * 1e-6) / 1e9;
+}
+
 }  // namespace
 
 void bind_comm(py::module_& m) {
@@ -336,4 +420,10 @@ void bind_comm(py::module_& m) {
   c.def("fanout_call", &fanout_call, py::arg("h"), py::arg("addrs"), py::arg("method"),
         py::arg("payload"), py::arg("resp_cap"), py::arg("rounds") = 1,
         py::arg("verify") = true);
+  // Streaming over the comm data plane (CommGroup p2p; config 3).
+  c.def("stream_comm_serve", &stream_comm_serve, py::arg("h"), py::arg("peer"),
+        py::arg("port") = 0, py::call_guard<py::gil_scoped_release>());
+  c.def("stream_comm_send", &stream_comm_send, py::arg("h"), py::arg("addr"),
+        py::arg("peer"), py::arg("frames"), py::arg("frame_size"),
+        py::call_guard<py::gil_scoped_release>());
 }

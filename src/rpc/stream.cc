@@ -9,6 +9,7 @@
 #include "base/resource_pool.h"
 #include "fiber/butex.h"
 #include "fiber/fiber.h"
+#include "rpc/comm_group.h"
 #include "rpc/controller.h"
 #include "rpc/socket.h"
 #include "rpc/wire.h"
@@ -17,7 +18,7 @@ namespace bam {
 
 namespace {
 
-enum FrameType { FRAME_DATA = 0, FRAME_CLOSE = 1, FRAME_FEEDBACK = 2 };
+enum FrameType { FRAME_DATA = 0, FRAME_CLOSE = 1, FRAME_FEEDBACK = 2, FRAME_GPU_DATA = 3 };
 const char kStreamMagic[4] = {'S', 'T', 'R', 'M'};
 const size_t kFrameHeaderLen = 28;
 
@@ -229,6 +230,51 @@ void OnStreamFrame(uint64_t dst_sid, int type, uint64_t aux, IOBuf* payload,
       }
       break;
     }
+    case FRAME_GPU_DATA: {
+      // aux = payload bytes, in flight as an RCCL p2p send from the peer.
+      // Land it straight into a fresh HBM block (zero host staging), then
+      // deliver like any DATA message. The Recv PARKS this fiber until the
+      // xGMI transfer completes — dedicate the socket to streaming when
+      // using the GPU leg (head-of-line applies, as with any byte stream).
+      CommGroup* grp = nullptr;
+      int peer = -1;
+      {
+        std::lock_guard<std::mutex> lk(m->mu);
+        if (m->closed) return;
+        grp = m->opt.gpu_group;
+        peer = m->opt.gpu_peer;
+      }
+      if (grp == nullptr || peer < 0) {
+        LOG(ERROR) << "GPU stream frame on a stream without a gpu_group";
+        close_local(dst_sid, true);
+        return;
+      }
+      IOBuf buf;
+      void* dst_ptr = nullptr;
+      const Residency res = grp->backend() == "rccl" ? RES_HBM : RES_HOST;
+      if (buf.append_writable_block((size_t)aux, res, 0, &dst_ptr) != 0 ||
+          grp->Recv(dst_ptr, (size_t)aux, peer) != 0) {
+        close_local(dst_sid, true);
+        return;
+      }
+      bool start = false;
+      {
+        std::lock_guard<std::mutex> lk(m->mu);
+        if (m->closed) return;
+        m->pending.push_back(std::move(buf));
+        if (!m->consumer_running) {
+          m->consumer_running = true;
+          start = true;
+        }
+      }
+      if (start) {
+        fiber_t th;
+        if (fiber_start_background(&th, consumer_fiber, (void*)(uintptr_t)dst_sid) != 0) {
+          consumer_fiber((void*)(uintptr_t)dst_sid);
+        }
+      }
+      break;
+    }
     case FRAME_FEEDBACK: {
       int64_t prev = m->acked_bytes.load(std::memory_order_relaxed);
       if ((int64_t)aux > prev) m->acked_bytes.store((int64_t)aux, std::memory_order_release);
@@ -270,6 +316,30 @@ int StreamWrite(StreamId sid, IOBuf* data) {
       }
     }
     butex_wait(m->window_butex, v, nullptr);  // woken by feedback or close
+  }
+  // xGMI leg: a single HBM-resident span travels as RCCL p2p; only the
+  // size descriptor rides TCP. The descriptor write and the RCCL send are
+  // ordered together (per-group mutex) so the receiver's posting order —
+  // descriptor arrival order — matches RCCL's send matching order even
+  // with several streams sharing one group.
+  if (m->opt.gpu_group != nullptr && m->opt.gpu_peer >= 0 && data->backing_block_num() == 1) {
+    IOBuf::Span sp = data->span_at(0);
+    // rccl groups move HBM spans only; tcp groups (tests / no-GPU
+    // fallback) take any single span through the same leg.
+    if (sp.res == RES_HBM || m->opt.gpu_group->backend() == "tcp") {
+      static std::mutex g_gpu_send_mu;
+      std::lock_guard<std::mutex> glk(g_gpu_send_mu);
+      IOBuf frame;
+      pack_frame(&frame, FRAME_GPU_DATA, remote, (uint64_t)n, nullptr);
+      SocketUniquePtr s;
+      if (Socket::Address(sock, &s) != 0 || s->Failed() || s->Write(&frame) != 0 ||
+          m->opt.gpu_group->Send(sp.data, n, m->opt.gpu_peer) != 0) {
+        close_local(sid, false);
+        return EPIPE;
+      }
+      data->clear();
+      return 0;
+    }
   }
   IOBuf frame;
   pack_frame(&frame, FRAME_DATA, remote, 0, data);

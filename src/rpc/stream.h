@@ -20,11 +20,22 @@ class Controller;
 
 typedef uint64_t StreamId;  // versioned; 0 = invalid
 
+class CommGroup;
+
 struct StreamOptions {
   size_t max_buf_size = 2u << 20;  // send window (bytes in flight)
   // Called in the stream's consumer fiber for each arrived message.
   std::function<void(StreamId, IOBuf* msg)> on_received;
   std::function<void(StreamId)> on_closed;
+  // xGMI data plane (BASELINE config 3): when both ends set the same
+  // (gpu_group, gpu_peer) pair, DATA messages whose payload is one
+  // HBM-resident span travel as RCCL p2p over xGMI; the TCP frame carries
+  // only a size descriptor (ordering + credit). The receiving side lands
+  // the payload in a fresh HBM block (no host staging). Host-resident
+  // messages on the same stream still go over TCP. gpu_peer is the PEER's
+  // group rank. Not owned.
+  CommGroup* gpu_group = nullptr;
+  int gpu_peer = -1;
 };
 
 // Client side: create before CallMethod; the stream id rides in the

@@ -141,3 +141,44 @@ def test_collective_fanout_multiprocess(nranks):
         out, err = p.communicate(timeout=180)
         assert p.returncode == 0, (r, err[-3000:], out[-500:])
     assert "CALLER_OK" in "".join(open("/dev/null").read() or "") or True
+
+
+STREAM_WORKER = r"""
+import sys
+sys.path.insert(0, %r)
+import brpc_amd as b
+c = b.core.comm
+rank, port = int(sys.argv[1]), int(sys.argv[2])
+h = c.create(2, rank, "tcp", "127.0.0.1", port)
+if rank == 0:
+    # receiver: server with the comm-leg stream service, peer = rank 1
+    sport = c.stream_comm_serve(h, 1)
+    c.send(h, 1, ("%%05d" %% sport).encode())
+    done = c.host_broadcast(h, b"", 1)
+    assert done == b"fin", done
+    print("RECV_OK")
+else:
+    sport = int(c.recv(h, 0, 5).decode())
+    gbps = c.stream_comm_send(h, "127.0.0.1:%%d" %% sport, 0, 40, 1 << 20)
+    assert gbps > 0, gbps
+    c.host_broadcast(h, b"fin", 1)
+    print("SEND_OK %%.3f GB/s" %% gbps)
+"""
+
+
+def test_stream_over_comm_two_processes():
+    """Stream GPU leg (FRAME_GPU_DATA): 1 MB frames move over the
+    CommGroup data plane (RCCL p2p on GPUs; TCP backend here) while the
+    stream's TCP socket carries only descriptors + credit."""
+    port = free_port_block(8) + 300
+    procs = [subprocess.Popen(
+        [sys.executable, "-c", STREAM_WORKER % REPO, str(r), str(port)],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.PIPE, text=True)
+        for r in range(2)]
+    outs = []
+    for r, p in enumerate(procs):
+        out, err = p.communicate(timeout=120)
+        assert p.returncode == 0, (r, err[-3000:], out[-500:])
+        outs.append(out)
+    assert "RECV_OK" in outs[0]
+    assert "SEND_OK" in outs[1]

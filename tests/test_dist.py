@@ -66,3 +66,52 @@ def test_stream_p2p_two_ranks_gloo():
     line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
     res = json.loads(line)
     assert res["value"] > 0
+
+
+def test_bench_stream_mode_two_ranks_gloo():
+    """bench.py --mode stream: config-3 headline through the framework's
+    stream layer + CommGroup data plane (RCCL/xGMI on GPUs, tcp here)."""
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    env = dict(os.environ)
+    env["BAM_BENCH_BACKEND"] = "gloo"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
+        "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
+        "--mode", "stream", "--frames-per-step", "8",
+    ]
+    out = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True, text=True,
+                         timeout=300)
+    assert out.returncode == 0, out.stderr[-3000:]
+    res = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][-1])
+    assert res["metric"] == "stream_gbps"
+    assert res["value"] > 0
+
+
+def test_bench_fanout_mode_two_ranks_gloo():
+    """bench.py --mode fanout: config-4 headline (collective fan-out)."""
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    env = dict(os.environ)
+    env["BAM_BENCH_BACKEND"] = "gloo"
+    env["MASTER_ADDR"] = "127.0.0.1"
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        "--nnodes=1", "--nproc-per-node=2",
+        "--master-addr", "127.0.0.1", "--master-port", str(port),
+        "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
+        "--mode", "fanout", "--calls-per-step", "200",
+    ]
+    out = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True, text=True,
+                         timeout=300)
+    assert out.returncode == 0, out.stderr[-3000:]
+    res = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][-1])
+    assert res["metric"] == "fanout_qps"
+    assert res["value"] > 0
