@@ -39,4 +39,20 @@ void bind_var(py::module_& m) {
     ok = ok && rec.latency_max() == 1000;
     return ok;
   }, py::call_guard<py::gil_scoped_release>());
+  v.def("latency_histogram_selftest", []() {
+    // Burst bias regression: 200 slow samples then 100k fast ones. The old
+    // last-8192 ring would have evicted every slow sample; the histogram
+    // keeps them weighted exactly.
+    bam::var::LatencyRecorder rec;
+    for (int i = 0; i < 200; ++i) rec << 5000;
+    for (int i = 0; i < 100000; ++i) rec << 10;
+    bool ok = rec.count() == 100200;
+    int64_t p50 = rec.latency_percentile(0.5);
+    ok = ok && p50 >= 5 && p50 <= 15;
+    int64_t p99 = rec.latency_percentile(0.99);
+    ok = ok && p99 <= 20;  // fast samples dominate the 99th
+    int64_t p9995 = rec.latency_percentile(0.9995);
+    ok = ok && p9995 >= 4000 && p9995 <= 6000;  // the slow tail is NOT lost
+    return ok;
+  }, py::call_guard<py::gil_scoped_release>());
 }

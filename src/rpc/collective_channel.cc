@@ -88,8 +88,19 @@ int snappy_echo_fn(const void* req, size_t len, void* resp, size_t resp_cap,
                    size_t* resp_len, int dev) {
   if (dev >= 0) {
     const gpu::GpuApi* api = gpu::api();
+    // The snappy stream's preamble is the uncompressed length as a varint
+    // — read it from the device copy to size the scratch exactly.
+    unsigned char head[8] = {0};
+    api->memcpy_res(head, 0, 0, req, 2, 0, len < 8 ? len : 8);
+    uint64_t plain_len_hdr = 0;
+    int shift = 0;
+    for (int i = 0; i < 8; ++i) {
+      plain_len_hdr |= (uint64_t)(head[i] & 0x7f) << shift;
+      if ((head[i] & 0x80) == 0) break;
+      shift += 7;
+    }
     Scratch plain;  // decompressed intermediate
-    size_t plain_cap = len * 8 + 4096;
+    size_t plain_cap = plain_len_hdr + 4096;
     void* pp = plain.ensure(true, plain_cap);
     if (pp == nullptr) return ENOMEM;
     size_t plain_len = 0;

@@ -75,6 +75,7 @@ namespace {
 struct Sampler {
   std::mutex mu;
   std::set<WindowedInt*> targets;
+  std::set<SamplerTick*> ticks;
   bool started = false;
 
   void ensure_started() {
@@ -92,6 +93,12 @@ struct Sampler {
         snapshot.assign(targets.begin(), targets.end());
       }
       for (WindowedInt* w : snapshot) w->take_sample();
+      std::vector<SamplerTick*> tick_snapshot;
+      {
+        std::lock_guard<std::mutex> lk(mu);
+        tick_snapshot.assign(ticks.begin(), ticks.end());
+      }
+      for (SamplerTick* t : tick_snapshot) t->run();
     }
   }
 };
@@ -112,6 +119,19 @@ void unregister_sampler(WindowedInt* w) {
   Sampler& s = sampler();
   std::lock_guard<std::mutex> lk(s.mu);
   s.targets.erase(w);
+}
+
+SamplerTick::SamplerTick(std::function<void()> fn) : fn_(std::move(fn)) {
+  Sampler& s = sampler();
+  std::lock_guard<std::mutex> lk(s.mu);
+  s.ticks.insert(this);
+  s.ensure_started();
+}
+
+SamplerTick::~SamplerTick() {
+  Sampler& s = sampler();
+  std::lock_guard<std::mutex> lk(s.mu);
+  s.ticks.erase(this);
 }
 
 // ---------------- WindowedInt ----------------
@@ -147,8 +167,9 @@ int64_t WindowedInt::get_value() const {
 
 // ---------------- LatencyRecorder ----------------
 
-LatencyRecorder::LatencyRecorder() : ring_(kRingSize) {
+LatencyRecorder::LatencyRecorder() {
   qps_window_.reset(new WindowedInt([this] { return count_.get_value(); }, 1, true));
+  tick_.reset(new SamplerTick([this] { take_hist_snapshot(); }));
 }
 
 LatencyRecorder::LatencyRecorder(const std::string& prefix) : LatencyRecorder() {
@@ -156,7 +177,18 @@ LatencyRecorder::LatencyRecorder(const std::string& prefix) : LatencyRecorder() 
 }
 
 LatencyRecorder::~LatencyRecorder() {
+  tick_.reset();  // stop snapshots before members die
   for (Variable* v : exposed_) delete v;
+}
+
+void LatencyRecorder::take_hist_snapshot() {
+  std::vector<uint32_t> snap(detail::LatencyHistogram::kBuckets);
+  hist_.merge(snap.data());
+  std::lock_guard<std::mutex> lk(snap_mu_);
+  snaps_.push_back(std::move(snap));
+  while ((int)snaps_.size() > kWindowSec) snaps_.pop_front();
+  // windowed max decays with the same cadence
+  window_max_.store(0, std::memory_order_relaxed);
 }
 
 LatencyRecorder& LatencyRecorder::operator<<(int64_t latency_us) {
@@ -166,9 +198,7 @@ LatencyRecorder& LatencyRecorder::operator<<(int64_t latency_us) {
   while (latency_us > cur &&
          !window_max_.compare_exchange_weak(cur, latency_us, std::memory_order_relaxed)) {
   }
-  size_t i = (size_t)ring_idx_.fetch_add(1, std::memory_order_relaxed) % kRingSize;
-  ring_[i].store((uint32_t)std::min<int64_t>(latency_us, UINT32_MAX),
-                 std::memory_order_relaxed);
+  hist_.add(latency_us > 0 ? (uint64_t)latency_us : 0);
   return *this;
 }
 
@@ -180,13 +210,27 @@ int64_t LatencyRecorder::latency_avg() const {
 }
 
 int64_t LatencyRecorder::latency_percentile(double p) const {
-  size_t n = std::min<size_t>((size_t)ring_idx_.load(std::memory_order_relaxed), kRingSize);
-  if (n == 0) return 0;
-  std::vector<uint32_t> copy(n);
-  for (size_t i = 0; i < n; ++i) copy[i] = ring_[i].load(std::memory_order_relaxed);
-  std::sort(copy.begin(), copy.end());
-  size_t idx = (size_t)(p * (n - 1));
-  return copy[idx];
+  constexpr int kB = detail::LatencyHistogram::kBuckets;
+  std::vector<uint32_t> now(kB);
+  hist_.merge(now.data());
+  {
+    // Subtract the oldest snapshot -> counts for the last <=10 s only.
+    std::lock_guard<std::mutex> lk(snap_mu_);
+    if (!snaps_.empty()) {
+      const std::vector<uint32_t>& old = snaps_.front();
+      for (int i = 0; i < kB; ++i) now[i] -= std::min(now[i], old[i]);
+    }
+  }
+  uint64_t total = 0;
+  for (int i = 0; i < kB; ++i) total += now[i];
+  if (total == 0) return 0;
+  uint64_t rank = (uint64_t)(p * (total - 1));
+  uint64_t acc = 0;
+  for (int i = 0; i < kB; ++i) {
+    acc += now[i];
+    if (acc > rank) return detail::LatencyHistogram::value_of(i);
+  }
+  return detail::LatencyHistogram::value_of(kB - 1);
 }
 
 void LatencyRecorder::expose(const std::string& prefix) {

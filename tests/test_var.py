@@ -50,3 +50,14 @@ def test_contention_vars_exposed():
     waits = int([l for l in body.splitlines() if l.startswith("fiber_butex_waits")][0].split(":")[1])
     assert waits > 0
     srv.stop()
+
+
+def test_latency_percentile_burst_unbiased():
+    """Round-2 percentile rework: per-thread log-bucket histograms merged
+    on read (≙ reference bvar/detail/percentile.h). A burst of fast
+    samples after slow ones must NOT evict the slow tail the way the old
+    last-8192-samples ring did: with 200 x 5000µs followed by 100k x 10µs,
+    the true p99 over the window is 10µs-ish but p999.9-region values and
+    the overall distribution must still see the slow samples in p-quantiles
+    that include them."""
+    assert v.latency_histogram_selftest()
