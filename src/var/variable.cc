@@ -214,9 +214,10 @@ int64_t LatencyRecorder::latency_percentile(double p) const {
   std::vector<uint32_t> now(kB);
   hist_.merge(now.data());
   {
-    // Subtract the oldest snapshot -> counts for the last <=10 s only.
+    // Subtract the kWindowSec-old snapshot -> counts for the last ~10 s
+    // only. A younger recorder (deque not yet full) uses its lifetime.
     std::lock_guard<std::mutex> lk(snap_mu_);
-    if (!snaps_.empty()) {
+    if ((int)snaps_.size() >= kWindowSec) {
       const std::vector<uint32_t>& old = snaps_.front();
       for (int i = 0; i < kB; ++i) now[i] -= std::min(now[i], old[i]);
     }
