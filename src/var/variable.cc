@@ -12,6 +12,13 @@
 namespace bam {
 namespace var {
 
+namespace detail {
+uint64_t next_combiner_id() {
+  static std::atomic<uint64_t> id{1};
+  return id.fetch_add(1, std::memory_order_relaxed);
+}
+}  // namespace detail
+
 // ---------------- registry ----------------
 
 namespace {

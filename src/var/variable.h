@@ -49,6 +49,11 @@ class Variable {
 
 namespace detail {
 
+// Unique id per combiner instance: thread-local cell caches are keyed by
+// (address, id) so a new combiner reusing a destroyed one's address never
+// sees a stale Cell* (ABA across recorder lifetimes).
+uint64_t next_combiner_id();
+
 // Per-thread cell list; read = combine over all cells (+ sum of cells from
 // dead threads). Writes touch only the caller's cell.
 template <typename T, typename Op>
@@ -73,16 +78,16 @@ class AgentCombiner {
   }
 
   Cell* local_cell() {
-    // tls map: combiner -> cell
-    static thread_local std::map<const void*, Cell*> tls;
+    // tls map: combiner address -> (instance id, cell)
+    static thread_local std::map<const void*, std::pair<uint64_t, Cell*>> tls;
     auto it = tls.find(this);
-    if (it != tls.end()) return it->second;
+    if (it != tls.end() && it->second.first == id_) return it->second.second;
     auto cell = std::make_shared<Cell>();
     {
       std::lock_guard<std::mutex> lk(mu_);
       cells_.push_back(cell);
     }
-    tls[this] = cell.get();
+    tls[this] = {id_, cell.get()};
     return cell.get();
   }
 
@@ -90,6 +95,7 @@ class AgentCombiner {
   mutable std::mutex mu_;
   std::vector<std::shared_ptr<Cell>> cells_;
   T terminated_ = T();
+  const uint64_t id_ = next_combiner_id();
 };
 
 // Per-thread log-bucketed latency histogram, merged on read. Clean-room
@@ -147,20 +153,21 @@ class LatencyHistogram {
 
  private:
   Cell* local_cell() {
-    static thread_local std::map<const void*, Cell*> tls;
+    static thread_local std::map<const void*, std::pair<uint64_t, Cell*>> tls;
     auto it = tls.find(this);
-    if (it != tls.end()) return it->second;
+    if (it != tls.end() && it->second.first == id_) return it->second.second;
     auto cell = std::make_shared<Cell>();
     {
       std::lock_guard<std::mutex> lk(mu_);
       cells_.push_back(cell);
     }
-    tls[this] = cell.get();
+    tls[this] = {id_, cell.get()};
     return cell.get();
   }
 
   mutable std::mutex mu_;
   std::vector<std::shared_ptr<Cell>> cells_;
+  const uint64_t id_ = next_combiner_id();
 };
 
 struct AddOp {
