@@ -11,6 +11,9 @@
 namespace bam {
 
 BAM_DEFINE_bool(enable_rpcz, true, "record per-RPC spans for /rpcz");
+BAM_DEFINE_int64(rpcz_sample_mod, 16,
+                 "record 1 of every N spans per thread (1 = all; parity: the "
+                 "reference samples spans through a budgeted bvar collector)");
 BAM_DEFINE_int64(rpcz_max_spans, 2048, "max spans kept in the rpcz ring");
 
 namespace rpcz {
@@ -33,6 +36,11 @@ void set_enabled(bool on) { FLAG_enable_rpcz = on; }
 
 void RecordSpan(const Span& span) {
   if (!FLAG_enable_rpcz) return;
+  // Per-thread sampling keeps the global ring mutex off the hot path
+  // (2 spans/call x >100k QPS would serialize on it).
+  static thread_local uint64_t tl_counter = 0;
+  int64_t mod = FLAG_rpcz_sample_mod;
+  if (mod > 1 && (tl_counter++ % (uint64_t)mod) != 0) return;
   Ring& r = ring();
   std::lock_guard<std::mutex> lk(r.mu);
   size_t cap = (size_t)FLAG_rpcz_max_spans;

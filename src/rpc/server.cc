@@ -90,9 +90,11 @@ void Server::SetMethodMaxConcurrency(const std::string& full_method, int32_t lim
   MethodGate*& g = method_gates_[full_method];
   if (g == nullptr) g = new MethodGate;
   g->max = limit;
+  gate_count_.store((int)method_gates_.size(), std::memory_order_release);
 }
 
 bool Server::BeginMethod(const std::string& service, const std::string& method) {
+  if (!has_method_gates()) return true;  // common case: no per-method caps
   MethodGate* g;
   {
     std::lock_guard<std::mutex> lk(gates_mu_);
@@ -110,6 +112,7 @@ bool Server::BeginMethod(const std::string& service, const std::string& method) 
 }
 
 void Server::EndMethod(const std::string& service, const std::string& method) {
+  if (!has_method_gates()) return;
   MethodGate* g;
   {
     std::lock_guard<std::mutex> lk(gates_mu_);

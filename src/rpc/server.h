@@ -128,6 +128,7 @@ class Server {
   // with EndMethod from the response path.
   bool BeginMethod(const std::string& service, const std::string& method);
   void EndMethod(const std::string& service, const std::string& method);
+  bool has_method_gates() const { return gate_count_.load(std::memory_order_acquire) > 0; }
   int Start(int port, const ServerOptions* opt);  // port 0 = pick free port
   int Start(const EndPoint& ep, const ServerOptions* opt);
   int Stop(int wait_ms = 0);
@@ -169,6 +170,7 @@ class Server {
   };
   mutable std::mutex gates_mu_;
   std::map<std::string, MethodGate*> method_gates_;
+  std::atomic<int> gate_count_{0};
   ServerOptions options_;
   EndPoint listen_ep_;
   SocketId listen_socket_ = 0;

@@ -53,6 +53,8 @@ HealthRegistry& health_registry() {
   return *r;
 }
 
+std::atomic<bool> g_any_errors{false};  // no lookup on success until an error exists
+
 EndpointHealth* health_of(const EndPoint& ep, bool create) {
   HealthRegistry& r = health_registry();
   std::lock_guard<std::mutex> lk(r.mu);
@@ -103,10 +105,14 @@ void health_check_fiber(void* raw) {
 void ReportClientCallResult(const EndPoint& ep, bool network_error) {
   if (!FLAG_enable_circuit_breaker) return;
   if (!network_error) {
+    // Success: skip the registry mutex entirely unless some endpoint has
+    // ever errored (the overwhelmingly common case at high QPS).
+    if (!g_any_errors.load(std::memory_order_acquire)) return;
     EndpointHealth* h = health_of(ep, false);
     if (h != nullptr) h->consecutive_errors.store(0, std::memory_order_relaxed);
     return;
   }
+  g_any_errors.store(true, std::memory_order_release);
   EndpointHealth* h = health_of(ep, true);
   int n = h->consecutive_errors.fetch_add(1, std::memory_order_relaxed) + 1;
   if (n >= FLAG_circuit_breaker_max_failures &&
