@@ -18,6 +18,7 @@ int rwlock_test(int nreaders, int nwriters, int iters);
 bool timer_test();
 bool fiber_key_test();
 bool gpu_wait_selftest();
+int64_t fd_wait_selftest();
 }  // namespace selftest
 }  // namespace bam
 
@@ -42,6 +43,8 @@ void bind_fiber(py::module_& m) {
   f.def("timer_test", &bam::selftest::timer_test, py::call_guard<py::gil_scoped_release>());
   f.def("key_test", &bam::selftest::fiber_key_test, py::call_guard<py::gil_scoped_release>());
   f.def("gpu_wait_test", &bam::selftest::gpu_wait_selftest,
+        py::call_guard<py::gil_scoped_release>());
+  f.def("fd_wait_test", &bam::selftest::fd_wait_selftest,
         py::call_guard<py::gil_scoped_release>());
   f.def("gpu_wait_parks", &bam::gpu_wait_parks);
   f.def("gpu_wait_wake_requests", &bam::gpu_wait_wake_requests);

@@ -2,6 +2,10 @@
 
 #include <errno.h>
 #include <poll.h>
+#include <sys/epoll.h>
+#include <vector>
+#include <thread>
+#include <mutex>
 #include <unistd.h>
 
 #include "base/time.h"
@@ -128,22 +132,140 @@ int64_t fiber_count_active() {
 
 
 
-int fiber_fd_wait(int fd, short events, int timeout_ms) {
-  const int64_t deadline = monotonic_time_us() + (int64_t)timeout_ms * 1000;
-  struct pollfd pfd;
-  for (;;) {
-    pfd.fd = fd;
-    pfd.events = events;
-    pfd.revents = 0;
-    int rc = ::poll(&pfd, 1, 0);
-    if (rc > 0) return (pfd.revents & (POLLERR | POLLNVAL)) ? -1 : 0;
-    if (rc < 0 && errno != EINTR) return -1;
-    if (timeout_ms >= 0 && monotonic_time_us() >= deadline) {
-      errno = ETIMEDOUT;
-      return -1;
+// Epoll-integrated user-fd wait (parity: reference bthread_fd_wait,
+// bthread/fd.cpp — bthread owns its own epoll, independent of the RPC
+// dispatcher). A waiter parks on a pooled butex; the fd-wait epoll thread
+// wakes it when the fd turns ready. Versioned slots make a late epoll
+// event harmless after the waiter left.
+namespace {
+
+struct FdWaitSlot {
+  std::atomic<int>* butex = nullptr;
+  std::atomic<uint32_t> ver{0};
+};
+
+constexpr int kFdWaitSlots = 1024;
+
+struct FdWaitLoop {
+  int epfd = -1;
+  FdWaitSlot slots[kFdWaitSlots];
+  std::mutex free_mu;
+  std::vector<int> freelist;
+
+  FdWaitLoop() {
+    epfd = epoll_create1(EPOLL_CLOEXEC);
+    freelist.reserve(kFdWaitSlots);
+    for (int i = kFdWaitSlots - 1; i >= 0; --i) {
+      slots[i].butex = butex_create();
+      freelist.push_back(i);
     }
-    fiber_usleep(500);
+    std::thread([this] { run(); }).detach();
   }
+
+  void run() {
+    struct epoll_event evs[64];
+    for (;;) {
+      int n = epoll_wait(epfd, evs, 64, -1);
+      if (n < 0) {
+        if (errno == EINTR) continue;
+        return;
+      }
+      for (int i = 0; i < n; ++i) {
+        const uint64_t tag = evs[i].data.u64;
+        const int idx = (int)(tag & 0xffffffffu);
+        const uint32_t ver = (uint32_t)(tag >> 32);
+        if (idx < 0 || idx >= kFdWaitSlots) continue;
+        FdWaitSlot& s = slots[idx];
+        if (s.ver.load(std::memory_order_acquire) != ver) continue;  // stale
+        s.butex->fetch_add(1, std::memory_order_release);
+        butex_wake_all(s.butex);
+      }
+    }
+  }
+
+  int acquire() {
+    std::lock_guard<std::mutex> lk(free_mu);
+    if (freelist.empty()) return -1;
+    int idx = freelist.back();
+    freelist.pop_back();
+    return idx;
+  }
+
+  void release(int idx) {
+    std::lock_guard<std::mutex> lk(free_mu);
+    freelist.push_back(idx);
+  }
+};
+
+FdWaitLoop& fd_wait_loop() {
+  static FdWaitLoop* l = new FdWaitLoop;
+  return *l;
+}
+
+}  // namespace
+
+int fiber_fd_wait(int fd, short events, int timeout_ms) {
+  const int64_t deadline =
+      timeout_ms >= 0 ? monotonic_time_us() + (int64_t)timeout_ms * 1000 : 0;
+  // Fast path: already ready.
+  struct pollfd pfd;
+  pfd.fd = fd;
+  pfd.events = events;
+  pfd.revents = 0;
+  int rc = ::poll(&pfd, 1, 0);
+  if (rc > 0) return (pfd.revents & POLLNVAL) ? -1 : 0;
+  if (rc < 0 && errno != EINTR) return -1;
+
+  FdWaitLoop& loop = fd_wait_loop();
+  const int idx = loop.acquire();
+  if (idx < 0) {
+    // All slots busy (pathological): degrade to coarse polling.
+    for (;;) {
+      pfd.revents = 0;
+      rc = ::poll(&pfd, 1, 20);
+      if (rc > 0) return (pfd.revents & POLLNVAL) ? -1 : 0;
+      if (rc < 0 && errno != EINTR) return -1;
+      if (timeout_ms >= 0 && monotonic_time_us() >= deadline) {
+        errno = ETIMEDOUT;
+        return -1;
+      }
+    }
+  }
+  FdWaitSlot& slot = loop.slots[idx];
+  const uint32_t ver = slot.ver.fetch_add(1, std::memory_order_acq_rel) + 1;
+  struct epoll_event ev;
+  ev.events = EPOLLONESHOT | (events & POLLIN ? EPOLLIN : 0u) |
+              (events & POLLOUT ? EPOLLOUT : 0u) | EPOLLRDHUP;
+  ev.data.u64 = ((uint64_t)ver << 32) | (uint32_t)idx;
+  bool armed = epoll_ctl(loop.epfd, EPOLL_CTL_ADD, fd, &ev) == 0;
+  if (!armed && errno == EEXIST) armed = epoll_ctl(loop.epfd, EPOLL_CTL_MOD, fd, &ev) == 0;
+  int result = -1;
+  if (!armed) {
+    errno = EINVAL;
+  } else {
+    for (;;) {
+      const int v = slot.butex->load(std::memory_order_acquire);
+      pfd.revents = 0;
+      rc = ::poll(&pfd, 1, 0);
+      if (rc > 0) {
+        result = (pfd.revents & POLLNVAL) ? -1 : 0;
+        break;
+      }
+      if (rc < 0 && errno != EINTR) break;
+      if (timeout_ms >= 0 && monotonic_time_us() >= deadline) {
+        errno = ETIMEDOUT;
+        break;
+      }
+      int64_t abst = timeout_ms >= 0 ? deadline : monotonic_time_us() + 2000000;
+      butex_wait(slot.butex, v, &abst);
+      // Spurious wake / re-arm after ONESHOT delivery for the next loop.
+      epoll_ctl(loop.epfd, EPOLL_CTL_MOD, fd, &ev);
+    }
+    epoll_ctl(loop.epfd, EPOLL_CTL_DEL, fd, nullptr);
+  }
+  slot.ver.fetch_add(1, std::memory_order_release);  // invalidate late events
+  loop.release(idx);
+  return result;
 }
 
 }  // namespace bam

@@ -98,3 +98,12 @@ def test_gpu_wait_parks_and_wakes():
     here with a fake wake hook; the HIP-side integration is covered by
     tests/test_gpu.py on a real MI355X."""
     assert f.gpu_wait_test()
+
+
+def test_fd_wait_epoll_integrated():
+    """fiber_fd_wait is epoll-backed (≙ reference bthread_fd_wait): a
+    blocked waiter parks on a butex and wakes when the fd turns readable —
+    well under the old 500µs poll granularity — and timeouts fire."""
+    lat_us = f.fd_wait_test()
+    assert lat_us >= 0, lat_us
+    assert lat_us < 100_000, lat_us  # generous bound for a loaded CI box
