@@ -93,8 +93,10 @@ class PyServer {
 
   int start(int port, int max_concurrency, const std::string& auth_user,
             const std::string& auth_password, const std::string& ssl_cert,
-            const std::string& ssl_key, const std::string& adaptive_max_concurrency) {
+            const std::string& ssl_key, const std::string& adaptive_max_concurrency,
+            const std::string& socket_mode) {
     ServerOptions opts;
+    opts.socket_mode = socket_mode;
     opts.max_concurrency = max_concurrency;
     opts.adaptive_max_concurrency = adaptive_max_concurrency;
     if (!auth_user.empty()) {
@@ -122,9 +124,10 @@ class PyChannel {
  public:
   PyChannel(const std::string& addr, const std::string& lb, int timeout_ms, int max_retry,
             int backup_request_ms, int compress, const std::string& auth_user,
-            const std::string& auth_password, bool ssl) {
+            const std::string& auth_password, bool ssl, const std::string& socket_mode) {
     ChannelOptions opts;
     opts.ssl = ssl;
+    opts.socket_mode = socket_mode;
     opts.timeout_ms = timeout_ms;
     opts.max_retry = max_retry;
     opts.backup_request_ms = backup_request_ms;
@@ -185,7 +188,7 @@ void bind_api(py::module_& m) {
       .def("start", &PyServer::start, py::arg("port") = 0, py::arg("max_concurrency") = 0,
            py::arg("auth_user") = "", py::arg("auth_password") = "",
            py::arg("ssl_cert") = "", py::arg("ssl_key") = "",
-           py::arg("adaptive_max_concurrency") = "")
+           py::arg("adaptive_max_concurrency") = "", py::arg("socket_mode") = "")
       .def("stop", &PyServer::stop)
       .def("running", &PyServer::running)
       .def("port", &PyServer::port)
@@ -193,11 +196,11 @@ void bind_api(py::module_& m) {
 
   py::class_<PyChannel>(m, "Channel")
       .def(py::init<const std::string&, const std::string&, int, int, int, int,
-                    const std::string&, const std::string&, bool>(),
+                    const std::string&, const std::string&, bool, const std::string&>(),
            py::arg("addr"), py::arg("lb") = "", py::arg("timeout_ms") = 500,
            py::arg("max_retry") = 3, py::arg("backup_request_ms") = -1,
            py::arg("compress") = 0, py::arg("auth_user") = "", py::arg("auth_password") = "",
-           py::arg("ssl") = false)
+           py::arg("ssl") = false, py::arg("socket_mode") = "")
       .def("call", &PyChannel::call, py::arg("method"), py::arg("request"),
            py::arg("attachment") = std::string(), py::arg("timeout_ms") = 0,
            py::arg("log_id") = 0);

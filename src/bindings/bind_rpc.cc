@@ -5,6 +5,7 @@
 #include <pybind11/stl.h>
 
 #include "bindings/bind.h"
+#include "rpc/rdma_transport.h"
 
 namespace bam {
 namespace rpctest {
@@ -47,6 +48,7 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
 
 void bind_rpc(py::module_& m) {
   auto r = m.def_submodule("rpc");
+  r.def("rdma_live_recv_blocks", &bam::rdma::live_recv_blocks);
   py::class_<bam::MysqlResult>(r, "MysqlResult")
       .def_readonly("ok", &bam::MysqlResult::ok)
       .def_readonly("affected_rows", &bam::MysqlResult::affected_rows)

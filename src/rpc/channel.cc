@@ -181,7 +181,7 @@ void IssueRPC(Controller* cntl) {
   }
   if (!sock) {
     if (GetClientSocket(ep, &sock, cntl->call.connection_shard, cntl->call.ssl,
-                        cntl->call.protocol_index) != 0) {
+                        cntl->call.protocol_index, cntl->call.socket_mode) != 0) {
       // Conduct the failure through the session so retry/ending logic runs.
       session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
       return;
@@ -232,6 +232,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->call.protocol_index = protocol_index_;
   cntl->call.auth = options_.auth;
   cntl->call.ssl = options_.ssl;
+  cntl->call.socket_mode = options_.socket_mode.empty() ? nullptr : options_.socket_mode.c_str();
   cntl->call.retry_policy = options_.retry_policy;
   if (options_.connection_type == "pooled") {
     static std::atomic<uint32_t> rr{0};

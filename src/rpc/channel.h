@@ -25,6 +25,10 @@ struct ChannelOptions {
   std::string connection_type = "single";
   int connection_pool_size = 8;
   std::string connection_group;
+  // Byte transport: "" = TCP (default), "rdma_mock" = the RDMA endpoint
+  // machinery over the in-process mock provider (rpc/rdma_transport.h);
+  // a verbs provider slots in the same way on RDMA-capable hosts.
+  std::string socket_mode;
   // Client credential source (rpc/authenticator.h); not owned. When set,
   // every request carries RpcMeta.authentication_data.
   const class Authenticator* auth = nullptr;

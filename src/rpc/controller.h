@@ -101,6 +101,7 @@ class Controller {
     const class Authenticator* auth = nullptr;       // from ChannelOptions
     std::string auth_data;                           // credential for this attempt
     bool ssl = false;                                // TLS client connection
+    const char* socket_mode = nullptr;               // "" / "rdma_mock"
     std::function<bool(int, int)> retry_policy;      // from ChannelOptions
   };
   Call call;

@@ -10,6 +10,7 @@
 
 #include "base/logging.h"
 #include "rpc/event_dispatcher.h"
+#include "rpc/rdma_transport.h"
 #include "rpc/policy/std_protocol.h"
 #include "rpc/policy/http_protocol.h"
 #include "rpc/redis.h"
@@ -158,6 +159,23 @@ void Server::OnNewConnections(Socket* listen_socket) {
     if (Socket::Create(opts, &sid) != 0) {
       ::close(fd);
       continue;
+    }
+    if (!server->options_.socket_mode.empty()) {
+      SocketUniquePtr sp;
+      if (Socket::Address(sid, &sp) == 0 &&
+          server->options_.socket_mode == "rdma_mock") {
+        std::string terr;
+        Transport* t = rdma::CreateRdmaTransport(sp.get(), rdma::mock_provider(), 16,
+                                                 64 << 10, &terr);
+        if (t != nullptr) {
+          sp->set_transport(t);
+          // The peer may have paired and delivered before the transport
+          // was installed — kick the input path once to drain any early data.
+          sp->on_input_event();
+        } else {
+          sp->SetFailed(ECONNRESET, ("rdma transport: " + terr).c_str());
+        }
+      }
     }
     if (server->ssl_ctx_ != nullptr) {
       SocketUniquePtr sp;

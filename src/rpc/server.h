@@ -72,6 +72,9 @@ typedef std::function<bool(Controller* cntl, int* error_code, std::string* error
 
 struct ServerOptions {
   int idle_timeout_sec = -1;
+  // "" = TCP; "rdma_mock" upgrades every accepted connection to the RDMA
+  // endpoint (mock provider; parity: reference ServerOptions.socket_mode).
+  std::string socket_mode;
   int max_concurrency = 0;          // 0 = unlimited
   bool has_builtin_services = true;
   RedisService* redis_service = nullptr;  // serve RESP on the same port

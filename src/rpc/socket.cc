@@ -18,6 +18,7 @@
 #include "fiber/fiber.h"
 #include "fiber/session.h"
 #include "rpc/event_dispatcher.h"
+#include "rpc/transport.h"
 #include "rpc/rpc_errno.h"
 
 namespace bam {
@@ -93,6 +94,7 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   s->ssl_state_.store(0, std::memory_order_relaxed);
   s->session_local_data.store(nullptr, std::memory_order_relaxed);
   s->session_local_deleter = nullptr;
+  s->transport_ = options.transport;
   s->write_head_.store(nullptr, std::memory_order_relaxed);
   s->in_bytes = 0;
   s->out_bytes = 0;
@@ -181,6 +183,8 @@ int Socket::SetFailed(int error_code, const char* error_text) {
 }
 
 void Socket::Recycle() {
+  delete transport_;
+  transport_ = nullptr;
   if (protocol_ctx != nullptr && protocol_ctx_deleter != nullptr) {
     protocol_ctx_deleter(protocol_ctx);
   }
@@ -318,6 +322,7 @@ int Socket::ssl_handshake_wait() {
 }
 
 ssize_t Socket::write_bytes(IOBuf* data, bool may_block) {
+  if (transport_ != nullptr) return transport_->CutFromIOBuf(this, data);
   if (ssl_ == nullptr) return data->cut_into_file_descriptor(fd());
   if (ssl_state_.load(std::memory_order_acquire) != 2) {
     if (!may_block) {
@@ -345,6 +350,7 @@ ssize_t Socket::write_bytes(IOBuf* data, bool may_block) {
 }
 
 ssize_t Socket::read_bytes(IOBuf* out, size_t max) {
+  if (transport_ != nullptr) return transport_->AppendToIOBuf(this, out, max);
   if (ssl_ == nullptr) return out->append_from_file_descriptor(fd(), max);
   char tmp[16384];
   size_t want = max < sizeof(tmp) ? max : sizeof(tmp);
@@ -358,6 +364,7 @@ ssize_t Socket::read_bytes(IOBuf* out, size_t max) {
 }
 
 int Socket::wait_epoll_out(int64_t abstime_us) {
+  if (transport_ != nullptr) return transport_->WaitWritable(this, abstime_us);
   int v = epollout_butex_->load(std::memory_order_acquire);
   if (Failed()) return -1;
   // Edge-triggered EPOLLOUT may have fired before we captured v (e.g. the

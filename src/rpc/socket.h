@@ -60,6 +60,11 @@ struct SocketOptions {
   void* user = nullptr;           // owner cookie (InputMessenger, Acceptor)
   std::function<void(Socket*)> on_edge_triggered_events;  // readable callback
   std::function<void(SocketId)> on_failed;                // teardown hook
+  // Byte transport under this socket (rpc/transport.h). nullptr = the
+  // inlined TCP path (parity: reference socket.cpp:1752 `_conn ?
+  // CutMessageIntoFileDescriptor : _transport->CutFromIOBuf`). Owned by
+  // the socket; deleted at recycle.
+  class Transport* transport = nullptr;
 };
 
 class Socket {
@@ -160,6 +165,8 @@ class Socket {
   // (returns EAGAIN so the caller hands off to KeepWrite).
   ssize_t write_bytes(IOBuf* data, bool may_block);
   ssize_t read_bytes(IOBuf* out, size_t max);
+  class Transport* transport() const { return transport_; }
+  void set_transport(class Transport* t) { transport_ = t; }  // takes ownership
 
   // Unwritten bytes queued on this socket; Write fails with EOVERCROWDED
   // above -socket_max_unwritten_bytes (parity: reference socket.cpp:1640).
@@ -213,6 +220,7 @@ class Socket {
   std::function<void(void*)> session_local_deleter;
 
  private:
+  class Transport* transport_ = nullptr;  // owned; null = inline TCP
   void* ssl_ = nullptr;                 // SSL* when TLS is enabled
   std::atomic<int> ssl_state_{0};       // 0 off, 1 handshaking, 2 ready
   std::mutex ssl_hs_mu_;                // serializes handshake stepping

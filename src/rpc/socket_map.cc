@@ -1,5 +1,9 @@
 #include "rpc/socket_map.h"
 
+#include <string.h>
+
+#include "rpc/rdma_transport.h"
+
 #include <errno.h>
 #include <poll.h>
 #include <unistd.h>
@@ -132,7 +136,7 @@ bool IsEndpointIsolated(const EndPoint& ep) {
 }
 
 int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ssl,
-                    int protocol_index) {
+                    int protocol_index, const char* socket_mode) {
   if (IsEndpointIsolated(ep)) {
     errno = EHOSTDOWN;
     return -1;
@@ -145,7 +149,9 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ss
   // preferred-protocol parse. Shards are small ints: bits 20-27 carry the
   // protocol index, bit 30 carries TLS.
   const int proto_bits = protocol_index >= 0 ? ((protocol_index + 1) & 0xff) << 20 : 0;
-  const auto key = std::make_pair(ep, shard | proto_bits | (ssl ? (1 << 30) : 0));
+  const bool rdma = socket_mode != nullptr && strcmp(socket_mode, "rdma_mock") == 0;
+  const auto key = std::make_pair(
+      ep, shard | proto_bits | (ssl ? (1 << 30) : 0) | (rdma ? (1 << 29) : 0));
   {
     std::lock_guard<std::mutex> lk(m.mu);
     auto it = m.sockets.find(key);
@@ -165,6 +171,17 @@ int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ss
   if (Socket::Create(opts, &sid) != 0) return -1;
   if (Socket::Address(sid, out) != 0) return -1;
   (*out)->client_protocol_hint = protocol_index;
+  if (socket_mode != nullptr && strcmp(socket_mode, "rdma_mock") == 0) {
+    std::string terr;
+    Transport* t = rdma::CreateRdmaTransport((*out).get(), rdma::mock_provider(), 16,
+                                             64 << 10, &terr);
+    if (t == nullptr) {
+      (*out)->SetFailed(ECONNRESET, ("rdma transport: " + terr).c_str());
+      out->reset(nullptr);
+      return -1;
+    }
+    (*out)->set_transport(t);
+  }
   if (ssl) {
     static void* g_client_ctx = ssl::NewClientCtx();  // process-lifetime
     void* h = g_client_ctx != nullptr

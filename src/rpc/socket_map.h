@@ -12,7 +12,8 @@ namespace bam {
 // connection (parity: reference pooled connection_type — multiple
 // connections to one server spread parse/write parallelism).
 int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard = 0,
-                    bool ssl = false, int protocol_index = -1);
+                    bool ssl = false, int protocol_index = -1,
+                    const char* socket_mode = nullptr /* "rdma_mock" */);
 
 // Drops the cached socket for ep (e.g. after failure).
 void RemoveClientSocket(const EndPoint& ep, SocketId expected);
