@@ -46,6 +46,41 @@ struct SpanBatch {
   unsigned long long ticket;
 };
 
+__device__ __forceinline__ void copy_one_span(const SpanRec& s) {
+  const unsigned int tid = threadIdx.x;
+  const unsigned int nt = blockDim.x;
+  if ((((uintptr_t)s.src ^ (uintptr_t)s.dst) & 15) == 0) {
+    unsigned int head = (16 - ((uintptr_t)s.src & 15)) & 15;
+    if (head > s.len) head = s.len;
+    for (unsigned int i = tid; i < head; i += nt) s.dst[i] = s.src[i];
+    const unsigned int nvec = (s.len - head) / 16;
+    const uint4* vsrc = (const uint4*)(s.src + head);
+    uint4* vdst = (uint4*)(s.dst + head);
+    for (unsigned int i = tid; i < nvec; i += nt) vdst[i] = vsrc[i];
+    for (unsigned int i = head + nvec * 16 + tid; i < s.len; i += nt) s.dst[i] = s.src[i];
+  } else {
+    for (unsigned int i = tid; i < s.len; i += nt) s.dst[i] = s.src[i];
+  }
+}
+
+__global__ void copy_spans_table_kernel(const SpanRec* table, int n,
+                                        unsigned long long* counter,
+                                        volatile unsigned long long* flag,
+                                        unsigned long long expect,
+                                        unsigned long long ticket) {
+  copy_one_span(table[blockIdx.x]);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    __threadfence_system();
+    unsigned long long done = atomicAdd(counter, 1ull) + 1;
+    if (done == expect) {
+      *flag = ticket;
+      __threadfence_system();
+    }
+  }
+  (void)n;
+}
+
 __global__ void copy_spans_kernel(SpanBatch a) {
   const SpanRec s = a.spans[blockIdx.x];
   const unsigned int tid = threadIdx.x;
@@ -82,6 +117,12 @@ struct SpanCtx {
   unsigned long long launched = 0;
   unsigned long long ticket = 0;
   int status = 0;  // 0 untried, 1 ok, -1 unavailable
+  // Large batches: span table staged through pinned memory into a device
+  // table, ONE launch regardless of span count (a 64 MB/8 KB gather is
+  // 8192 spans — kernarg batches of 48 would cost 171 launches).
+  SpanRec* table_dev = nullptr;
+  SpanRec* table_pinned = nullptr;
+  size_t table_cap = 0;  // entries
 };
 SpanCtx g_ctx[kMaxDev][kCtxPerDev];
 std::atomic<unsigned int> g_rr[kMaxDev];
@@ -120,10 +161,32 @@ int run_spans(int nspans, NextFn next, int dev) {
       if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
       return -1;
     }
-    SpanBatch a;
-    int batch = 0;
-    auto flush = [&]() {
-      if (batch == 0) return;
+    // Count ≤kSubSpan pieces first to pick the path.
+    size_t npieces = 0;
+    for (int i = 0; i < nspans; ++i) {
+      SpanIn s = next(i);
+      npieces += (s.len + kSubSpan - 1) / kSubSpan;
+    }
+    if (npieces == 0) {
+      if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
+      return 0;
+    }
+    if (npieces <= (size_t)kSpanArgMax) {
+      // Latency path: spans by kernarg, one launch.
+      SpanBatch a;
+      int batch = 0;
+      for (int i = 0; i < nspans; ++i) {
+        SpanIn s = next(i);
+        size_t off = 0;
+        while (off < s.len) {
+          size_t piece = s.len - off < kSubSpan ? s.len - off : kSubSpan;
+          a.spans[batch].src = s.src + off;
+          a.spans[batch].dst = s.dst + off;
+          a.spans[batch].len = (unsigned int)piece;
+          off += piece;
+          ++batch;
+        }
+      }
       c.launched += (unsigned long long)batch;
       c.ticket += 1;
       a.counter = c.counter_dev;
@@ -132,24 +195,59 @@ int run_spans(int nspans, NextFn next, int dev) {
       a.ticket = c.ticket;
       hipLaunchKernelGGL(copy_spans_kernel, dim3(batch), dim3(256), 0, c.stream, a);
       if (hipGetLastError() != hipSuccess) {
-        c.status = -1;  // ticket bookkeeping is now unusable; disable ctx
+        c.status = -1;
         rc = -1;
       }
-      batch = 0;
-    };
-    for (int i = 0; i < nspans && rc == 0; ++i) {
-      SpanIn s = next(i);
-      size_t off = 0;
-      while (off < s.len && rc == 0) {
-        size_t piece = s.len - off < kSubSpan ? s.len - off : kSubSpan;
-        a.spans[batch].src = s.src + off;
-        a.spans[batch].dst = s.dst + off;
-        a.spans[batch].len = (unsigned int)piece;
-        off += piece;
-        if (++batch == kSpanArgMax) flush();
+    } else {
+      // Throughput path: stage the span table through pinned memory into
+      // the device table, ONE launch for every piece.
+      if (c.table_cap < npieces) {
+        size_t want = npieces * 2;
+        if (c.table_dev != nullptr) hipFree(c.table_dev);
+        if (c.table_pinned != nullptr) hipHostFree(c.table_pinned);
+        c.table_dev = nullptr;
+        c.table_pinned = nullptr;
+        c.table_cap = 0;
+        if (hipMalloc(&c.table_dev, want * sizeof(SpanRec)) != hipSuccess ||
+            hipHostMalloc((void**)&c.table_pinned, want * sizeof(SpanRec),
+                          hipHostMallocDefault) != hipSuccess) {
+          c.status = -1;
+          rc = -1;
+        } else {
+          c.table_cap = want;
+        }
+      }
+      if (rc == 0) {
+        size_t w = 0;
+        for (int i = 0; i < nspans; ++i) {
+          SpanIn s = next(i);
+          size_t off = 0;
+          while (off < s.len) {
+            size_t piece = s.len - off < kSubSpan ? s.len - off : kSubSpan;
+            c.table_pinned[w].src = s.src + off;
+            c.table_pinned[w].dst = s.dst + off;
+            c.table_pinned[w].len = (unsigned int)piece;
+            ++w;
+            off += piece;
+          }
+        }
+        c.launched += (unsigned long long)npieces;
+        c.ticket += 1;
+        if (hipMemcpyAsync(c.table_dev, c.table_pinned, npieces * sizeof(SpanRec),
+                           hipMemcpyHostToDevice, c.stream) != hipSuccess) {
+          c.status = -1;
+          rc = -1;
+        } else {
+          hipLaunchKernelGGL(copy_spans_table_kernel, dim3((uint32_t)npieces), dim3(256), 0,
+                             c.stream, c.table_dev, (int)npieces, c.counter_dev, c.flag,
+                             c.launched, c.ticket);
+          if (hipGetLastError() != hipSuccess) {
+            c.status = -1;
+            rc = -1;
+          }
+        }
       }
     }
-    if (rc == 0) flush();
     wait_ticket_val = c.ticket;
   }
   if (rc == 0 && wait_ticket_val != 0 &&

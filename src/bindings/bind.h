@@ -18,3 +18,4 @@ void bind_json2pb(py::module_& m);
 void bind_thrift(py::module_& m);
 void bind_codecs(py::module_& m);
 void bind_comm(py::module_& m);
+void bind_proto(py::module_& m);

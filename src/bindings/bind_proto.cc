@@ -1,0 +1,108 @@
+// Python surface for the self-contained protobuf runtime (base/proto.h).
+// Tests use the installed python google.protobuf as a wire/JSON oracle.
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <memory>
+
+#include "base/proto.h"
+#include "bindings/bind.h"
+
+namespace {
+
+struct PyPool {
+  std::shared_ptr<bam::proto::DescriptorPool> pool =
+      std::make_shared<bam::proto::DescriptorPool>();
+};
+
+struct PyMsg {
+  std::shared_ptr<bam::proto::DescriptorPool> pool;  // keep alive
+  std::unique_ptr<bam::proto::DynMessage> msg;
+};
+
+}  // namespace
+
+void bind_proto(py::module_& m) {
+  auto p = m.def_submodule("proto");
+
+  py::class_<PyPool>(p, "Pool")
+      .def(py::init<>())
+      .def("parse",
+           [](PyPool& self, const std::string& text) {
+             std::string err;
+             if (self.pool->ParseProtoText(text, &err) != 0)
+               throw std::runtime_error("proto parse: " + err);
+           })
+      .def("messages", [](PyPool& self) { return self.pool->message_names(); })
+      .def("services", [](PyPool& self) { return self.pool->service_names(); })
+      .def("service_methods",
+           [](PyPool& self, const std::string& name) {
+             const bam::proto::ServiceDef* s = self.pool->FindService(name);
+             if (s == nullptr) throw std::runtime_error("no service " + name);
+             std::vector<std::vector<std::string>> out;
+             for (const auto& mth : s->methods)
+               out.push_back({mth.name, mth.input_type, mth.output_type});
+             return out;
+           })
+      .def("new_message", [](PyPool& self, const std::string& full_name) {
+        const bam::proto::MessageDef* d = self.pool->FindMessage(full_name);
+        if (d == nullptr) throw std::runtime_error("no message " + full_name);
+        PyMsg out;
+        out.pool = self.pool;
+        out.msg.reset(new bam::proto::DynMessage(self.pool.get(), d));
+        return out;
+      });
+
+  py::class_<PyMsg>(p, "Message")
+      .def("parse_wire",
+           [](PyMsg& self, py::bytes data) {
+             std::string s(data);
+             self.msg->clear();
+             if (!self.msg->ParseWire(s.data(), s.size()))
+               throw std::runtime_error("wire parse failed");
+           })
+      .def("serialize_wire",
+           [](PyMsg& self) {
+             std::string out;
+             self.msg->SerializeWire(&out);
+             return py::bytes(out);
+           })
+      .def("from_json",
+           [](PyMsg& self, const std::string& text) {
+             std::string err;
+             if (!self.msg->FromJson(text, &err))
+               throw std::runtime_error("from_json: " + err);
+           })
+      .def("to_json",
+           [](PyMsg& self, bool original_names) {
+             std::string out;
+             self.msg->ToJson(&out, original_names);
+             return out;
+           },
+           py::arg("original_names") = false)
+      .def("get_int", [](PyMsg& self, const std::string& n, size_t i) {
+        return self.msg->get_int(n, i);
+      }, py::arg("name"), py::arg("idx") = 0)
+      .def("get_str", [](PyMsg& self, const std::string& n, size_t i) {
+        return py::bytes(self.msg->get_str(n, i));
+      }, py::arg("name"), py::arg("idx") = 0)
+      .def("get_double", [](PyMsg& self, const std::string& n, size_t i) {
+        return self.msg->get_double(n, i);
+      }, py::arg("name"), py::arg("idx") = 0)
+      .def("count", [](PyMsg& self, const std::string& n) { return self.msg->count(n); })
+      .def("set_int", [](PyMsg& self, const std::string& n, int64_t v) {
+        self.msg->set_int(n, v);
+      })
+      .def("set_str", [](PyMsg& self, const std::string& n, py::bytes v) {
+        self.msg->set_str(n, std::string(v));
+      })
+      .def("set_double", [](PyMsg& self, const std::string& n, double v) {
+        self.msg->set_double(n, v);
+      })
+      .def("add_int", [](PyMsg& self, const std::string& n, int64_t v) {
+        self.msg->add_int(n, v);
+      })
+      .def("add_str", [](PyMsg& self, const std::string& n, py::bytes v) {
+        self.msg->add_str(n, std::string(v));
+      });
+}

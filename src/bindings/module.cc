@@ -18,4 +18,5 @@ PYBIND11_MODULE(_core, m) {
   bind_thrift(m);
   bind_codecs(m);
   bind_comm(m);
+  bind_proto(m);
 }

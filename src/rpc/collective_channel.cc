@@ -106,8 +106,19 @@ int snappy_echo_fn(const void* req, size_t len, void* resp, size_t resp_cap,
     size_t plain_len = 0;
     if (api->snappy_decompress(req, len, pp, plain_cap, &plain_len, dev) != 0)
       return EINVAL;
-    if (api->snappy_compress(pp, plain_len, resp, resp_cap, resp_len, dev) != 0)
+    // The device compressor needs worst-case output room (like host
+    // MaxCompressedLength); compress into scratch, then move the actual
+    // bytes into the (tightly sized) response slot.
+    Scratch comp;
+    size_t comp_cap = plain_len + plain_len / 3 + 128;
+    void* cp = comp.ensure(true, comp_cap);
+    if (cp == nullptr) return ENOMEM;
+    size_t comp_len = 0;
+    if (api->snappy_compress(pp, plain_len, cp, comp_cap, &comp_len, dev) != 0)
       return EINVAL;
+    if (comp_len > resp_cap) return EINVAL;
+    api->memcpy_res(resp, 2, 0, cp, 2, 0, comp_len);
+    *resp_len = comp_len;
     return 0;
   }
   std::string in((const char*)req, len), plain, out;
