@@ -10,7 +10,7 @@ HIPCC     := /opt/rocm/bin/hipcc
 GPU_ARCH  := gfx950
 
 CXXFLAGS  := -O2 -g -std=c++17 -fPIC -pthread -Wall -Wno-unused-function \
-             -Isrc -I/usr/include -I/opt/conda/include -msse4.2 -fno-omit-frame-pointer -MMD -MP
+             -Isrc -I. -I/usr/include -I/opt/conda/include -msse4.2 -fno-omit-frame-pointer -MMD -MP
 LDFLAGS   := -shared -pthread -ldl -lz -lssl -lcrypto
 
 CORE_SRCS := $(wildcard src/base/*.cc) $(wildcard src/fiber/*.cc) $(wildcard src/rpc/*.cc) \
@@ -71,7 +71,7 @@ clean:
 # into build/fuzz/bin/.
 FUZZ_CLANG   := /opt/rocm/lib/llvm/bin/clang++
 FUZZ_FLAGS   := -O1 -g -std=c++17 -fPIC -pthread -fsanitize=address,fuzzer-no-link \
-                -fno-omit-frame-pointer -Isrc -I/usr/include -I/opt/conda/include -MMD -MP
+                -fno-omit-frame-pointer -Isrc -I. -I/usr/include -I/opt/conda/include -MMD -MP
 FUZZ_OBJDIR  := build/fuzz
 FUZZ_OBJS    := $(patsubst src/%.cc,$(FUZZ_OBJDIR)/%.o,$(CORE_SRCS)) $(FUZZ_OBJDIR)/fiber/context.o
 FUZZ_BINS    := $(patsubst tests/fuzz/%.cc,$(FUZZ_OBJDIR)/bin/%,$(wildcard tests/fuzz/*.cc))

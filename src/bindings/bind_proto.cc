@@ -44,6 +44,38 @@ void bind_proto(py::module_& m) {
                out.push_back({mth.name, mth.input_type, mth.output_type});
              return out;
            })
+      .def("describe_message",
+           [](PyPool& self, const std::string& name) {
+             const bam::proto::MessageDef* d = self.pool->FindMessage(name);
+             if (d == nullptr) throw std::runtime_error("no message " + name);
+             py::list out;
+             for (const auto& f : d->fields) {
+               py::dict fd;
+               fd["name"] = f.name;
+               fd["json_name"] = f.json_name;
+               fd["number"] = f.number;
+               fd["type"] = (int)f.type;
+               fd["repeated"] = f.repeated;
+               fd["packed"] = f.packed;
+               fd["is_map"] = f.is_map;
+               fd["oneof_index"] = f.oneof_index;
+               fd["type_name"] = f.type_name;
+               out.append(fd);
+             }
+             return out;
+           })
+      .def("enum_values",
+           [](PyPool& self, const std::string& name) {
+             const bam::proto::EnumDef* e = self.pool->FindEnum(name);
+             if (e == nullptr) throw std::runtime_error("no enum " + name);
+             return e->values;
+           })
+      .def("enums",
+           [](PyPool& self) {
+             std::vector<std::string> out;
+             for (const auto& kv : self.pool->enums_) out.push_back(kv.first);
+             return out;
+           })
       .def("new_message", [](PyPool& self, const std::string& full_name) {
         const bam::proto::MessageDef* d = self.pool->FindMessage(full_name);
         if (d == nullptr) throw std::runtime_error("no message " + full_name);

@@ -40,6 +40,7 @@ int call_method_once(const std::string& addr, const std::string& method,
                      const std::string& payload, int timeout_ms, int max_retry,
                      std::string* response_out, std::string* error_text);
 bool attachment_test(const std::string& addr);
+bool pb_stub_test(std::string* err);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
                        bool hbm_request, bool pooled);
@@ -49,6 +50,15 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
 void bind_rpc(py::module_& m) {
   auto r = m.def_submodule("rpc");
   r.def("rdma_live_recv_blocks", &bam::rdma::live_recv_blocks);
+  r.def("pb_stub_test", []() {
+    std::string err;
+    bool ok;
+    {
+      py::gil_scoped_release rel;
+      ok = bam::rpctest::pb_stub_test(&err);
+    }
+    return py::make_tuple(ok, err);
+  });
   py::class_<bam::MysqlResult>(r, "MysqlResult")
       .def_readonly("ok", &bam::MysqlResult::ok)
       .def_readonly("affected_rows", &bam::MysqlResult::affected_rows)
