@@ -431,6 +431,30 @@ void bind_json2pb(py::module_& m) {
       throw std::runtime_error("pb_to_json: " + err);
     return text;
   });
+  // Descriptor-driven (any runtime-parsed .proto; no hand schema).
+  j.def("json_to_pb_proto", [](const std::string& proto_src, const std::string& message,
+                               const std::string& json_text) {
+    bam::proto::DescriptorPool pool;
+    std::string err;
+    if (pool.ParseProtoText(proto_src, &err) != 0)
+      throw std::runtime_error("proto parse: " + err);
+    std::string wire;
+    if (!bam::json2pb::JsonToPbByDescriptor(pool, message, json_text, &wire, &err))
+      throw std::runtime_error("json_to_pb: " + err);
+    return py::bytes(wire);
+  });
+  j.def("pb_to_json_proto", [](const std::string& proto_src, const std::string& message,
+                               py::bytes wire) {
+    bam::proto::DescriptorPool pool;
+    std::string err;
+    if (pool.ParseProtoText(proto_src, &err) != 0)
+      throw std::runtime_error("proto parse: " + err);
+    std::string text;
+    if (!bam::json2pb::PbToJsonByDescriptor(pool, message, wire.cast<std::string>(), &text,
+                                            &err))
+      throw std::runtime_error("pb_to_json: " + err);
+    return text;
+  });
 }
 
 // ---- thrift passthrough client ----

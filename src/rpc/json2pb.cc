@@ -187,3 +187,46 @@ bool decode_object(const Schema& schema, const char* data, size_t n, json::Value
 
 }  // namespace json2pb
 }  // namespace bam
+
+namespace bam {
+namespace json2pb {
+
+bool JsonToPbByDescriptor(const proto::DescriptorPool& pool,
+                          const std::string& message_full_name, const std::string& json_text,
+                          std::string* wire, std::string* error) {
+  const proto::MessageDef* def = pool.FindMessage(message_full_name);
+  if (def == nullptr) {
+    if (error != nullptr) *error = "unknown message " + message_full_name;
+    return false;
+  }
+  proto::DynMessage msg(&pool, def);
+  std::string err;
+  if (!msg.FromJson(json_text, &err)) {
+    if (error != nullptr) *error = err;
+    return false;
+  }
+  wire->clear();
+  msg.SerializeWire(wire);
+  return true;
+}
+
+bool PbToJsonByDescriptor(const proto::DescriptorPool& pool,
+                          const std::string& message_full_name, const std::string& wire,
+                          std::string* json_text, std::string* error) {
+  const proto::MessageDef* def = pool.FindMessage(message_full_name);
+  if (def == nullptr) {
+    if (error != nullptr) *error = "unknown message " + message_full_name;
+    return false;
+  }
+  proto::DynMessage msg(&pool, def);
+  if (!msg.ParseWire(wire.data(), wire.size())) {
+    if (error != nullptr) *error = "malformed wire bytes";
+    return false;
+  }
+  json_text->clear();
+  msg.ToJson(json_text);
+  return true;
+}
+
+}  // namespace json2pb
+}  // namespace bam

@@ -11,6 +11,8 @@
 #include <string>
 #include <vector>
 
+#include "base/proto.h"
+
 namespace bam {
 namespace json2pb {
 
@@ -30,6 +32,16 @@ bool JsonToPb(const Schema& schema, const std::string& json_text, std::string* w
 // protobuf wire bytes -> JSON text.
 bool PbToJson(const Schema& schema, const std::string& wire, std::string* json_text,
               std::string* error = nullptr);
+
+// ---- descriptor-driven (round 2; parity: reference json2pb works on any
+// pb Message via descriptors — here via base/proto.h's DescriptorPool,
+// so ANY .proto parses at runtime, no hand-written schema needed). ----
+bool JsonToPbByDescriptor(const ::bam::proto::DescriptorPool& pool,
+                          const std::string& message_full_name, const std::string& json_text,
+                          std::string* wire, std::string* error = nullptr);
+bool PbToJsonByDescriptor(const ::bam::proto::DescriptorPool& pool,
+                          const std::string& message_full_name, const std::string& wire,
+                          std::string* json_text, std::string* error = nullptr);
 
 }  // namespace json2pb
 }  // namespace bam
