@@ -19,3 +19,4 @@ void bind_thrift(py::module_& m);
 void bind_codecs(py::module_& m);
 void bind_comm(py::module_& m);
 void bind_proto(py::module_& m);
+void bind_hpack(py::module_& m);

@@ -6,6 +6,7 @@
 #include <memory>
 
 #include "base/proto.h"
+#include "rpc/policy/hpack.h"
 #include "bindings/bind.h"
 
 namespace {
@@ -21,6 +22,50 @@ struct PyMsg {
 };
 
 }  // namespace
+
+void bind_hpack(py::module_& m) {
+  auto h = m.def_submodule("hpack");
+  h.def("huffman_encode", [](py::bytes data) {
+    std::string out;
+    bam::hpack::HuffmanEncode(std::string(data), &out);
+    return py::bytes(out);
+  });
+  h.def("huffman_decode", [](py::bytes data) {
+    std::string in(data), out;
+    if (!bam::hpack::HuffmanDecode(in.data(), in.size(), &out))
+      throw std::runtime_error("huffman decode failed");
+    return py::bytes(out);
+  });
+  h.def("huffman_table", []() {
+    std::vector<std::pair<uint32_t, int>> t;
+    bam::hpack::HuffmanTable(&t);
+    return t;
+  });
+  h.def("encode_int", [](uint64_t v, int prefix, int flags) {
+    std::string out;
+    bam::hpack::EncodeInt(&out, v, prefix, (uint8_t)flags);
+    return py::bytes(out);
+  });
+  py::class_<bam::hpack::Encoder>(h, "Encoder")
+      .def(py::init<size_t>(), py::arg("max_table_size") = 4096)
+      .def("encode", [](bam::hpack::Encoder& e,
+                        const std::vector<std::pair<std::string, std::string>>& headers) {
+        std::string out;
+        e.Encode(headers, &out);
+        return py::bytes(out);
+      });
+  py::class_<bam::hpack::Decoder>(h, "Decoder")
+      .def(py::init<size_t>(), py::arg("max_table_size") = 4096)
+      .def("decode", [](bam::hpack::Decoder& d, py::bytes block) {
+        std::string in(block);
+        std::vector<bam::hpack::Header> out;
+        if (!d.Decode(in.data(), in.size(), &out))
+          throw std::runtime_error("hpack decode failed");
+        py::list res;
+        for (auto& kv : out) res.append(py::make_tuple(py::bytes(kv.first), py::bytes(kv.second)));
+        return res;
+      });
+}
 
 void bind_proto(py::module_& m) {
   auto p = m.def_submodule("proto");

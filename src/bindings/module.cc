@@ -19,4 +19,5 @@ PYBIND11_MODULE(_core, m) {
   bind_codecs(m);
   bind_comm(m);
   bind_proto(m);
+  bind_hpack(m);
 }
