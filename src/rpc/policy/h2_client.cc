@@ -1,19 +1,20 @@
-// brpc_amd: HTTP/2 / gRPC CLIENT protocol (channel protocol "grpc").
+// brpc_amd: HTTP/2 / gRPC CLIENT protocol (channel protocols "grpc"/"h2"),
+// fully in-tree on rpc/policy/h2_session.* + hpack.* (round 1 dlopened
+// libnghttp2; that dependency is gone).
 // Parity: reference h2/gRPC client paths in policy/http2_rpc_protocol.cpp.
-// One nghttp2 client session per connection (socket protocol_ctx);
-// requests map stream_id -> correlation session; responses complete on
-// stream close with grpc-status trailer handling. Interop-tested against
-// the official grpc python SERVER (tests/test_grpc.py).
-#include <dlfcn.h>
-#include <nghttp2/nghttp2.h>
-
+// One session per connection (socket protocol_ctx); requests map
+// stream_id -> correlation session; responses complete on END_STREAM with
+// grpc-status trailer handling. Interop-tested against the official grpc
+// python SERVER (tests/test_grpc.py).
 #include <map>
 #include <mutex>
+#include <vector>
 
 #include "base/logging.h"
 #include "fiber/session.h"
 #include "rpc/channel.h"
 #include "rpc/controller.h"
+#include "rpc/policy/h2_session.h"
 #include "rpc/server.h"
 
 namespace bam {
@@ -23,51 +24,6 @@ void EndRPC(Controller* cntl, SessionId locked_id);  // channel.cc
 namespace policy {
 
 namespace {
-
-struct NgcApi {
-  int (*callbacks_new)(nghttp2_session_callbacks**);
-  void (*callbacks_del)(nghttp2_session_callbacks*);
-  void (*set_on_frame_recv)(nghttp2_session_callbacks*, nghttp2_on_frame_recv_callback);
-  void (*set_on_header)(nghttp2_session_callbacks*, nghttp2_on_header_callback);
-  void (*set_on_data_chunk)(nghttp2_session_callbacks*, nghttp2_on_data_chunk_recv_callback);
-  void (*set_on_stream_close)(nghttp2_session_callbacks*, nghttp2_on_stream_close_callback);
-  int (*client_new)(nghttp2_session**, const nghttp2_session_callbacks*, void*);
-  void (*session_del)(nghttp2_session*);
-  ssize_t (*mem_recv)(nghttp2_session*, const uint8_t*, size_t);
-  ssize_t (*mem_send)(nghttp2_session*, const uint8_t**);
-  int (*submit_settings)(nghttp2_session*, uint8_t, const nghttp2_settings_entry*, size_t);
-  int32_t (*submit_request)(nghttp2_session*, const nghttp2_priority_spec*,
-                            const nghttp2_nv*, size_t, const nghttp2_data_provider*, void*);
-  bool ok = false;
-};
-
-NgcApi& ngc() {
-  static NgcApi api = [] {
-    NgcApi a;
-    void* h = dlopen("libnghttp2.so.14", RTLD_NOW | RTLD_GLOBAL);
-    if (h == nullptr) h = dlopen("libnghttp2.so", RTLD_NOW | RTLD_GLOBAL);
-    if (h == nullptr) return a;
-#define NGC_SYM(field, name)            \
-  *(void**)(&a.field) = dlsym(h, name); \
-  if (a.field == nullptr) return a;
-    NGC_SYM(callbacks_new, "nghttp2_session_callbacks_new")
-    NGC_SYM(callbacks_del, "nghttp2_session_callbacks_del")
-    NGC_SYM(set_on_frame_recv, "nghttp2_session_callbacks_set_on_frame_recv_callback")
-    NGC_SYM(set_on_header, "nghttp2_session_callbacks_set_on_header_callback")
-    NGC_SYM(set_on_data_chunk, "nghttp2_session_callbacks_set_on_data_chunk_recv_callback")
-    NGC_SYM(set_on_stream_close, "nghttp2_session_callbacks_set_on_stream_close_callback")
-    NGC_SYM(client_new, "nghttp2_session_client_new")
-    NGC_SYM(session_del, "nghttp2_session_del")
-    NGC_SYM(mem_recv, "nghttp2_session_mem_recv")
-    NGC_SYM(mem_send, "nghttp2_session_mem_send")
-    NGC_SYM(submit_settings, "nghttp2_submit_settings")
-    NGC_SYM(submit_request, "nghttp2_submit_request")
-#undef NGC_SYM
-    a.ok = true;
-    return a;
-  }();
-  return api;
-}
 
 struct H2ClientStream {
   uint64_t cid = 0;
@@ -80,85 +36,32 @@ struct H2ClientStream {
 
 struct H2ClientCtx {
   std::mutex mu;
-  nghttp2_session* session = nullptr;
+  H2Session* session = nullptr;
   SocketId socket_id = 0;
   std::map<int32_t, H2ClientStream> streams;
   // closed streams are completed OUTSIDE mu (a completion can trigger a
-  // retry that re-enters IssueGrpcRequest -> mu: lock-order safety).
+  // retry that re-enters IssueH2Request -> mu: lock-order safety).
   std::vector<std::pair<H2ClientStream, uint32_t>> done_list;
   std::atomic<int> refs{1};
 
-  ~H2ClientCtx() {
-    if (session != nullptr) ngc().session_del(session);
-  }
+  ~H2ClientCtx() { delete session; }
 };
 
 void h2c_unref(H2ClientCtx* ctx) {
   if (ctx->refs.fetch_sub(1, std::memory_order_acq_rel) == 1) delete ctx;
 }
 
-struct H2ClientBody {
-  std::string data;
-  size_t offset = 0;
-};
-
-nghttp2_nv cnv(const char* name, const std::string& value) {
-  nghttp2_nv nv;
-  nv.name = (uint8_t*)name;
-  nv.namelen = strlen(name);
-  nv.value = (uint8_t*)value.data();
-  nv.valuelen = value.size();
-  nv.flags = NGHTTP2_NV_FLAG_NONE;
-  return nv;
-}
-
 void flush_client_locked(H2ClientCtx* ctx) {
+  if (!ctx->session->has_output()) return;
   SocketUniquePtr sock;
   if (Socket::Address(ctx->socket_id, &sock) != 0) return;
-  IOBuf out;
-  for (;;) {
-    const uint8_t* data = nullptr;
-    ssize_t n = ngc().mem_send(ctx->session, &data);
-    if (n <= 0) break;
-    out.append(data, (size_t)n);
+  std::string bytes;
+  ctx->session->TakeOutput(&bytes);
+  if (!bytes.empty()) {
+    IOBuf out;
+    out.append(bytes);
+    sock->Write(&out);
   }
-  if (!out.empty()) sock->Write(&out);
-}
-
-// ---- callbacks (ctx->mu held by the pump) ----
-
-int c_on_header(nghttp2_session*, const nghttp2_frame* frame, const uint8_t* name,
-                size_t namelen, const uint8_t* value, size_t valuelen, uint8_t,
-                void* user_data) {
-  H2ClientCtx* ctx = (H2ClientCtx*)user_data;
-  auto it = ctx->streams.find(frame->hd.stream_id);
-  if (it == ctx->streams.end()) return 0;
-  std::string key((const char*)name, namelen);
-  std::string val((const char*)value, valuelen);
-  if (key == "grpc-status") it->second.grpc_status = atoi(val.c_str());
-  else if (key == "grpc-message") it->second.grpc_message = val;
-  else if (key == ":status") it->second.http_status = atoi(val.c_str());
-  else if (key == "content-type" && val.rfind("application/grpc", 0) == 0)
-    it->second.is_grpc_response = true;
-  return 0;
-}
-
-int c_on_data(nghttp2_session*, uint8_t, int32_t stream_id, const uint8_t* data, size_t len,
-              void* user_data) {
-  H2ClientCtx* ctx = (H2ClientCtx*)user_data;
-  auto it = ctx->streams.find(stream_id);
-  if (it != ctx->streams.end()) it->second.body.append(data, len);
-  return 0;
-}
-
-int c_on_stream_close(nghttp2_session*, int32_t stream_id, uint32_t error_code,
-                      void* user_data) {
-  H2ClientCtx* ctx = (H2ClientCtx*)user_data;
-  auto it = ctx->streams.find(stream_id);
-  if (it == ctx->streams.end()) return 0;
-  ctx->done_list.emplace_back(std::move(it->second), error_code);
-  ctx->streams.erase(it);
-  return 0;
 }
 
 // Runs with ctx->mu RELEASED (see struct comment).
@@ -187,21 +90,6 @@ void complete_stream(H2ClientCtx* ctx, H2ClientStream&& st, uint32_t error_code)
   EndRPC(cntl, st.cid);
 }
 
-ssize_t c_body_read(nghttp2_session*, int32_t, uint8_t* buf, size_t length,
-                    uint32_t* data_flags, nghttp2_data_source* source, void*) {
-  H2ClientBody* body = (H2ClientBody*)source->ptr;
-  size_t left = body->data.size() - body->offset;
-  size_t n = left < length ? left : length;
-  memcpy(buf, body->data.data() + body->offset, n);
-  body->offset += n;
-  if (body->offset >= body->data.size()) {
-    *data_flags |= NGHTTP2_DATA_FLAG_EOF;
-    delete body;
-    source->ptr = nullptr;
-  }
-  return (ssize_t)n;
-}
-
 // ---- issue + parse hooks ----
 
 int g_grpc_protocol_index = -1;
@@ -213,26 +101,44 @@ bool own_client_ctx(Socket* sock) {
 }
 
 int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
-  if (!ngc().ok) return -1;
   if (sock->protocol_ctx != nullptr && !own_client_ctx(sock)) return -1;
   H2ClientCtx* ctx = (H2ClientCtx*)sock->protocol_ctx;
   if (ctx == nullptr) {
     ctx = new H2ClientCtx;
     ctx->socket_id = sock->id();
-    nghttp2_session_callbacks* cbs = nullptr;
-    ngc().callbacks_new(&cbs);
-    ngc().set_on_header(cbs, c_on_header);
-    ngc().set_on_data_chunk(cbs, c_on_data);
-    ngc().set_on_stream_close(cbs, c_on_stream_close);
-    ngc().client_new(&ctx->session, cbs, ctx);
-    ngc().callbacks_del(cbs);
-    ngc().submit_settings(ctx->session, NGHTTP2_FLAG_NONE, nullptr, 0);
+    H2Session::Callbacks cbs;
+    cbs.on_header = [ctx](int32_t sid, const std::string& key, const std::string& val) {
+      auto it = ctx->streams.find(sid);
+      if (it == ctx->streams.end()) return;
+      if (key == "grpc-status") it->second.grpc_status = atoi(val.c_str());
+      else if (key == "grpc-message") it->second.grpc_message = val;
+      else if (key == ":status") it->second.http_status = atoi(val.c_str());
+      else if (key == "content-type" && val.rfind("application/grpc", 0) == 0)
+        it->second.is_grpc_response = true;
+    };
+    cbs.on_data = [ctx](int32_t sid, const char* data, size_t n) {
+      auto it = ctx->streams.find(sid);
+      if (it != ctx->streams.end()) it->second.body.append(data, n);
+    };
+    cbs.on_end_stream = [ctx](int32_t sid) {
+      auto it = ctx->streams.find(sid);
+      if (it == ctx->streams.end()) return;
+      ctx->done_list.emplace_back(std::move(it->second), 0);
+      ctx->streams.erase(it);
+    };
+    cbs.on_rst = [ctx](int32_t sid, uint32_t error) {
+      auto it = ctx->streams.find(sid);
+      if (it == ctx->streams.end()) return;
+      ctx->done_list.emplace_back(std::move(it->second), error != 0 ? error : 1);
+      ctx->streams.erase(it);
+    };
+    ctx->session = new H2Session(/*server=*/false, cbs);
     sock->protocol_ctx = ctx;
     sock->protocol_ctx_deleter = [](void* p) { h2c_unref((H2ClientCtx*)p); };
     sock->protocol_ctx_owner = grpc ? g_grpc_protocol_index : g_h2c_protocol_index;
   }
   std::string payload = cntl->call.request_buf.to_string();
-  H2ClientBody* body = new H2ClientBody;
+  std::string body;
   if (grpc) {
     // gRPC frame the payload (5-byte length prefix).
     char frame[5];
@@ -241,10 +147,10 @@ int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
     frame[2] = (char)(payload.size() >> 16);
     frame[3] = (char)(payload.size() >> 8);
     frame[4] = (char)payload.size();
-    body->data.assign(frame, 5);
-    body->data += payload;
+    body.assign(frame, 5);
+    body += payload;
   } else {
-    body->data = payload;
+    body = payload;
   }
 
   const std::string& svc = cntl->call.service_name;
@@ -254,35 +160,22 @@ int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
                      : svc.empty() && !grpc          ? "/" + mn
                                                      : "/" + svc + "/" + mn;
   std::string authority = endpoint2str(cntl->remote_side());
-  // NOTE: every value must outlive submit_request — named locals, never
-  // temporaries (nghttp2_nv holds raw pointers).
-  std::string v_method = grpc || !payload.empty() ? "POST" : "GET", v_scheme = "http",
-              v_ct = grpc ? "application/grpc" : "application/octet-stream",
-              v_te = "trailers", v_ua = grpc ? "brpc-amd-grpc/1.0" : "brpc-amd-h2/1.0";
-  nghttp2_nv nvs[7] = {
-      cnv(":method", v_method), cnv(":scheme", v_scheme), cnv(":path", path),
-      cnv(":authority", authority), cnv("content-type", v_ct), cnv("te", v_te),
-      cnv("user-agent", v_ua),
+  std::vector<hpack::Header> headers = {
+      {":method", grpc || !payload.empty() ? "POST" : "GET"},
+      {":scheme", "http"},
+      {":path", path},
+      {":authority", authority},
+      {"content-type", grpc ? "application/grpc" : "application/octet-stream"},
   };
-  // plain h2 omits "te: trailers" (gRPC-specific).
-  const size_t nnv = grpc ? 7 : 6;
-  if (!grpc) nvs[5] = nvs[6];  // drop te, keep user-agent
-  nghttp2_data_provider prd;
-  prd.source.ptr = body;
-  prd.read_callback = c_body_read;
+  if (grpc) headers.push_back({"te", "trailers"});
+  headers.push_back({"user-agent", grpc ? "brpc-amd-grpc/1.0" : "brpc-amd-h2/1.0"});
   std::lock_guard<std::mutex> lk(ctx->mu);
-  int32_t stream_id =
-      ngc().submit_request(ctx->session, nullptr, nvs, nnv, &prd, nullptr);
-  if (stream_id < 0) {
-    delete body;
-    return -1;
-  }
+  int32_t stream_id = ctx->session->SubmitRequest(headers, body, /*end_stream=*/true);
+  if (stream_id < 0) return -1;
   ctx->streams[stream_id].cid = cid;
   flush_client_locked(ctx);
   return 0;
 }
-
-struct H2ClientPump : public InputMessageBase {};
 
 ParseResult ParseGrpcClient(IOBuf* source, Socket* sock, bool /*eof*/) {
   if (sock->user() != nullptr) return ParseResult::make_error(PARSE_ERROR_TRY_OTHERS);
@@ -293,7 +186,7 @@ ParseResult ParseGrpcClient(IOBuf* source, Socket* sock, bool /*eof*/) {
   std::vector<std::pair<H2ClientStream, uint32_t>> done;
   {
     std::lock_guard<std::mutex> lk(ctx->mu);
-    ssize_t consumed = ngc().mem_recv(ctx->session, (const uint8_t*)bytes.data(), bytes.size());
+    ssize_t consumed = ctx->session->Consume(bytes.data(), bytes.size());
     if (consumed < 0) return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
     source->pop_front((size_t)consumed);
     flush_client_locked(ctx);  // acks, window updates
@@ -309,10 +202,6 @@ ParseResult ParseGrpcClient(IOBuf* source, Socket* sock, bool /*eof*/) {
 void RegisterGrpcClientProtocol() {
   static std::once_flag flag;
   std::call_once(flag, [] {
-    if (!ngc().ok) {
-      LOG(WARNING) << "libnghttp2 unavailable; grpc client protocol disabled";
-      return;
-    }
     Protocol p;
     p.parse = ParseGrpcClient;
     p.issue_request = [](Socket* s, Controller* c, uint64_t cid) {
@@ -323,7 +212,7 @@ void RegisterGrpcClientProtocol() {
     p.name = "grpc";
     g_grpc_protocol_index = RegisterProtocol(p);
     // Plain HTTP-semantics h2 client (parity: reference protocol "h2"):
-    // same nghttp2 session machinery, no gRPC framing or trailers.
+    // same session machinery, no gRPC framing or trailers.
     Protocol h2;
     h2.parse = ParseGrpcClient;
     h2.issue_request = [](Socket* s, Controller* c, uint64_t cid) {

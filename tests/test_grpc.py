@@ -107,3 +107,17 @@ def test_builtin_grpc_health_check(addr):
         call = _stub(ch, "/grpc.health.v1.Health/Check")
         resp = call(b"", timeout=5)
         assert resp == b"\x08\x01"  # status: SERVING
+
+
+def test_grpc_flow_control_large_response(addr):
+    """>1 MB response exceeds the client's default 64 KB h2 windows: the
+    in-tree session must queue DATA and resume on WINDOW_UPDATE."""
+    import grpc
+    channel = grpc.insecure_channel(addr)
+    big = bytes(range(256)) * 5000  # 1.25 MB
+    fut = channel.unary_unary("/EchoService/Echo",
+                              request_serializer=lambda x: x,
+                              response_deserializer=lambda x: x)
+    resp = fut(big, timeout=20)
+    assert resp == big
+    channel.close()
