@@ -24,6 +24,33 @@ static void set_err(const char* what, hipError_t e) {
 
 extern "C" const char* bam_gpu_last_error(void) { return g_err; }
 
+// ---- telemetry (rendered by the /hotspots/gpu builtin page) ----
+namespace {
+std::atomic<uint64_t> g_stat_gathers{0};
+std::atomic<uint64_t> g_stat_gather_bytes{0};
+std::atomic<uint64_t> g_stat_gather_fallbacks{0};
+std::atomic<uint64_t> g_stat_uploads{0};
+std::atomic<uint64_t> g_stat_upload_bytes{0};
+std::atomic<uint64_t> g_stat_hbm_blocks{0};
+std::atomic<uint64_t> g_stat_hbm_bytes{0};
+}  // namespace
+
+extern "C" const char* bam_gpu_stats_text(void) {
+  static char buf[512];
+  snprintf(buf, sizeof(buf),
+           "gather_batches: %llu\ngather_bytes: %llu\ngather_fallbacks: %llu\n"
+           "uploads_async: %llu\nupload_bytes: %llu\n"
+           "hbm_blocks_live: %llu\nhbm_bytes_live: %llu\n",
+           (unsigned long long)g_stat_gathers.load(),
+           (unsigned long long)g_stat_gather_bytes.load(),
+           (unsigned long long)g_stat_gather_fallbacks.load(),
+           (unsigned long long)g_stat_uploads.load(),
+           (unsigned long long)g_stat_upload_bytes.load(),
+           (unsigned long long)g_stat_hbm_blocks.load(),
+           (unsigned long long)g_stat_hbm_bytes.load());
+  return buf;
+}
+
 extern "C" int bam_gpu_device_count(void) {
   static int count = [] {
     int n = 0;
@@ -164,6 +191,8 @@ extern "C" void* bam_gpu_alloc_hbm(uint32_t cap, int dev) {
   }
   DevPool& pool = g_pools[dev];
   std::lock_guard<std::mutex> lk(pool.mu);
+  g_stat_hbm_blocks.fetch_add(1, std::memory_order_relaxed);
+  g_stat_hbm_bytes.fetch_add(kClasses[cls], std::memory_order_relaxed);
   if (!pool.freelists[cls].empty()) {
     void* p = pool.freelists[cls].back();
     pool.freelists[cls].pop_back();
@@ -195,6 +224,8 @@ extern "C" void bam_gpu_free_hbm(void* p, uint32_t cap, int dev) {
   }
   DevPool& pool = g_pools[dev];
   std::lock_guard<std::mutex> lk(pool.mu);
+  g_stat_hbm_blocks.fetch_sub(1, std::memory_order_relaxed);
+  g_stat_hbm_bytes.fetch_sub(kClasses[cls], std::memory_order_relaxed);
   pool.freelists[cls].push_back(p);
 }
 
@@ -663,7 +694,12 @@ extern "C" void bam_gpu_quiesce(int dev) {
 
 extern "C" int bam_gpu_upload_async(void* dst_dev, const void* src, size_t n, int dev) {
   ScopedDevice sd(dev);
-  return upload_direct(dst_dev, src, n, dev);
+  int rc = upload_direct(dst_dev, src, n, dev);
+  if (rc == 0) {
+    g_stat_uploads.fetch_add(1, std::memory_order_relaxed);
+    g_stat_upload_bytes.fetch_add(n, std::memory_order_relaxed);
+  }
+  return rc;
 }
 
 extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
@@ -671,11 +707,14 @@ extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
   size_t total = 0;
   for (int i = 0; i < nspans; ++i) total += lens[i];
   if (total == 0) return 0;
+  g_stat_gathers.fetch_add(1, std::memory_order_relaxed);
+  g_stat_gather_bytes.fetch_add(total, std::memory_order_relaxed);
   ScopedDevice sd(dev);
   if (total <= direct_max_bytes() && nspans <= kDirectMaxSpans &&
       gather_direct(host_dst, srcs, lens, nspans, dev) == 0) {
     return 0;
   }
+  g_stat_gather_fallbacks.fetch_add(1, std::memory_order_relaxed);
   // Fallback paths read HBM outside the staging stream: drain uploads first.
   bam_gpu_quiesce(dev);
   if (nspans == 1) {

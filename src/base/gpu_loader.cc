@@ -86,6 +86,7 @@ void do_load() {
   g_api.comm_sendrecv = (int (*)(void*, const void*, size_t, int, void*, size_t,
                                  int))dlsym(h, "bam_comm_sendrecv");
   g_api.comm_last_error = (const char* (*)(void))dlsym(h, "bam_comm_last_error");
+  g_api.stats_text = (const char* (*)(void))dlsym(h, "bam_gpu_stats_text");
 
   // Fiber↔stream integration (fiber/gpu_wait.h): give the HIP lib a
   // park/wake pair so its ticket waits yield the worker instead of

@@ -40,6 +40,7 @@ struct GpuApi {
   int (*comm_recv)(void*, void*, size_t, int);
   int (*comm_sendrecv)(void*, const void*, size_t, int, void*, size_t, int);
   const char* (*comm_last_error)(void);
+  const char* (*stats_text)(void);  // telemetry for /hotspots/gpu (optional)
 };
 
 // Loads the library (idempotent). Returns device count (0 = no GPU or no

@@ -1,6 +1,7 @@
 #include "fiber/butex.h"
 
 #include <errno.h>
+#include <execinfo.h>
 
 #include <mutex>
 
@@ -134,6 +135,34 @@ static std::atomic<int64_t> g_butex_wait_us{0};
 int64_t butex_total_waits() { return g_butex_waits.load(std::memory_order_relaxed); }
 int64_t butex_total_wait_us() { return g_butex_wait_us.load(std::memory_order_relaxed); }
 
+// Sampled contention sites: cheap ring, 1/64 parks pay a backtrace().
+namespace {
+constexpr int kContentionRing = 4096;
+std::mutex g_cont_mu;
+ContentionSample g_cont_ring[kContentionRing];
+std::atomic<uint64_t> g_cont_n{0};
+
+void record_contention(int64_t wait_us) {
+  static thread_local uint32_t tl_counter = 0;
+  if ((tl_counter++ & 63) != 0) return;
+  ContentionSample s;
+  s.nframes = backtrace(s.frames, 4);
+  s.wait_us = wait_us;
+  uint64_t idx = g_cont_n.fetch_add(1, std::memory_order_relaxed) % kContentionRing;
+  std::lock_guard<std::mutex> lk(g_cont_mu);
+  g_cont_ring[idx] = s;
+}
+}  // namespace
+
+size_t butex_contention_samples(ContentionSample* out, size_t max) {
+  uint64_t n = g_cont_n.load(std::memory_order_relaxed);
+  size_t have = n < (uint64_t)kContentionRing ? (size_t)n : (size_t)kContentionRing;
+  if (have > max) have = max;
+  std::lock_guard<std::mutex> lk(g_cont_mu);
+  for (size_t i = 0; i < have; ++i) out[i] = g_cont_ring[i];
+  return have;
+}
+
 int butex_wait(std::atomic<int>* v, int expected, const int64_t* abstime_us) {
   Butex* b = container_of_value(v);
   if (b->value.load(std::memory_order_acquire) != expected) {
@@ -160,7 +189,9 @@ int butex_wait(std::atomic<int>* v, int expected, const int64_t* abstime_us) {
     const int64_t park_t0 = monotonic_time_us();
     g->sched(remained_add_waiter, &w);
     // Resumed (possibly on another worker).
-    g_butex_wait_us.fetch_add(monotonic_time_us() - park_t0, std::memory_order_relaxed);
+    const int64_t parked = monotonic_time_us() - park_t0;
+    g_butex_wait_us.fetch_add(parked, std::memory_order_relaxed);
+    record_contention(parked);
     if (w.timer_id != 0) timer_delete(w.timer_id);
     if (w.state.load(std::memory_order_acquire) == 2) {
       errno = ETIMEDOUT;

@@ -24,6 +24,17 @@ int butex_wait(std::atomic<int>* b, int expected, const int64_t* abstime_us = nu
 int64_t butex_total_waits();
 int64_t butex_total_wait_us();
 
+// Sampled contention SITES (parity: the reference contention profiler fed
+// by instrumented bthread_mutex): 1 of every 64 parks records the caller
+// backtrace + parked time. Renders at /contention.
+struct ContentionSample {
+  void* frames[4];
+  int nframes;
+  int64_t wait_us;
+};
+// Copies up to `max` recent samples into out; returns the count copied.
+size_t butex_contention_samples(ContentionSample* out, size_t max);
+
 int butex_wake(std::atomic<int>* b);      // wake one; returns #woken
 int butex_wake_all(std::atomic<int>* b);  // returns #woken
 

@@ -12,7 +12,10 @@
 #include "base/iobuf.h"
 #include "base/time.h"
 #include "fiber/fiber.h"
+#include <malloc.h>
+
 #include "rpc/policy/http_protocol.h"
+#include "fiber/gpu_wait.h"
 #include "rpc/rpcz.h"
 #include "rpc/server.h"
 #include "rpc/socket.h"
@@ -21,6 +24,7 @@
 namespace bam {
 std::string dump_fiber_stacks(int max_fibers);  // fiber/tracer.cc
 std::string CpuProfile(int seconds, int hz);    // rpc/cpu_profiler.cc
+std::string ContentionProfile();                // rpc/cpu_profiler.cc
 namespace policy {
 
 namespace {
@@ -33,7 +37,8 @@ void page_index(HttpResponse* resp) {
   os << "<html><head><title>brpc_amd</title></head><body><h1>brpc_amd server</h1><ul>";
   const char* pages[] = {"status", "vars",   "flags",  "health",       "version",
                          "connections", "protobufs", "fibers", "memory", "threads",
-                         "hotspots/cpu", "rpcz", "brpc_metrics"};
+                         "hotspots/cpu", "hotspots/contention", "hotspots/gpu",
+                         "rpcz", "brpc_metrics"};
   for (const char* p : pages) os << "<li><a href=\"/" << p << "\">/" << p << "</a></li>";
   os << "</ul></body></html>";
   resp->body.append(os.str());
@@ -162,9 +167,19 @@ void page_threads(HttpResponse* resp) {
 }
 
 void page_memory(HttpResponse* resp) {
+  // parity: reference /memory (tcmalloc extension dump) — here glibc
+  // mallinfo2 + the framework's own pools.
   std::ostringstream os;
+  struct mallinfo2 mi = mallinfo2();
+  os << "malloc_arena_bytes: " << mi.arena << "\n";
+  os << "malloc_in_use_bytes: " << mi.uordblks << "\n";
+  os << "malloc_free_bytes: " << mi.fordblks << "\n";
+  os << "malloc_mmap_bytes: " << mi.hblkhd << "\n";
   os << "iobuf_block_count: " << IOBuf::block_count() << "\n";
   os << "iobuf_block_memory: " << IOBuf::block_memory() << "\n";
+  if (gpu::api() != nullptr && gpu::api()->stats_text != nullptr) {
+    os << "---- gpu ----\n" << gpu::api()->stats_text();
+  }
   resp->body.append(os.str());
 }
 
@@ -223,7 +238,27 @@ bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse
     int seconds = 1;
     auto it = req.query.find("seconds");
     if (it != req.query.end()) seconds = atoi(it->second.c_str());
-    resp->body.append(CpuProfile(seconds, 200));
+    int hz = 200;
+    auto hzit = req.query.find("hz");
+    if (hzit != req.query.end()) hz = atoi(hzit->second.c_str());
+    resp->body.append(CpuProfile(seconds, hz));
+  } else if (p == "/hotspots/contention" || p == "/contention") {
+    // parity: contention profiler (instrumented sync primitives)
+    resp->body.append(ContentionProfile());
+  } else if (p == "/hotspots/gpu" || p == "/gpustats") {
+    // MI355X-native analogue of a GPU hotspots page: the HIP runtime's
+    // own telemetry (staging gathers, async uploads, HBM pool occupancy)
+    // + scheduler GPU-wait integration counters. Kernel-level timing
+    // comes from rocprofv3 offline (profiles/).
+    std::ostringstream os;
+    if (gpu::api() != nullptr && gpu::api()->stats_text != nullptr) {
+      os << gpu::api()->stats_text();
+    } else {
+      os << "no GPU runtime loaded\n";
+    }
+    os << "gpu_wait_parks: " << gpu_wait_parks() << "\n";
+    os << "gpu_wait_wake_requests: " << gpu_wait_wake_requests() << "\n";
+    resp->body.append(os.str());
   } else if (p == "/brpc_metrics" || p == "/metrics") {
     page_metrics(resp);
   } else if (p == "/rpcz") {

@@ -5,6 +5,7 @@
 // preallocated lock-free slot array for N seconds; the report aggregates
 // by symbolized frame (dladdr) with self/cumulative counts. Exposed at
 // /hotspots/cpu?seconds=N (builtin_services.cc).
+#include <cxxabi.h>
 #include <dlfcn.h>
 #include <execinfo.h>
 #include <signal.h>
@@ -19,6 +20,7 @@
 #include <string>
 #include <vector>
 
+#include "fiber/butex.h"
 #include "fiber/fiber.h"
 
 namespace bam {
@@ -50,13 +52,70 @@ void sigprof_handler(int, siginfo_t*, void*) {
 
 std::string frame_name(void* ip) {
   Dl_info info;
-  if (dladdr(ip, &info) != 0 && info.dli_sname != nullptr) return info.dli_sname;
+  if (dladdr(ip, &info) != 0 && info.dli_sname != nullptr) {
+    int status = 0;
+    char* dem = abi::__cxa_demangle(info.dli_sname, nullptr, nullptr, &status);
+    if (status == 0 && dem != nullptr) {
+      std::string out(dem);
+      free(dem);
+      return out;
+    }
+    return info.dli_sname;
+  }
   char buf[32];
   snprintf(buf, sizeof(buf), "%p", ip);
   return buf;
 }
 
 }  // namespace
+
+// Renders the sampled butex contention sites (parity: reference
+// /hotspots/contention fed by instrumented bthread_mutex): top callsites
+// by total parked time among the last 4096 sampled parks (1/64 sampling).
+std::string ContentionProfile() {
+  std::vector<ContentionSample> samples(4096);
+  size_t n = butex_contention_samples(samples.data(), samples.size());
+  struct Site {
+    int64_t total_us = 0;
+    int64_t count = 0;
+    void* frames[4];
+    int nframes = 0;
+  };
+  std::map<std::string, Site> sites;  // keyed by caller frame chain addrs
+  for (size_t i = 0; i < n; ++i) {
+    const ContentionSample& s = samples[i];
+    char key[80];
+    // frames[0] is butex internals; aggregate by the 3 caller frames.
+    snprintf(key, sizeof(key), "%p|%p|%p", s.nframes > 1 ? s.frames[1] : nullptr,
+             s.nframes > 2 ? s.frames[2] : nullptr, s.nframes > 3 ? s.frames[3] : nullptr);
+    Site& site = sites[key];
+    site.total_us += s.wait_us;
+    site.count += 1;
+    if (site.nframes == 0) {
+      site.nframes = s.nframes;
+      memcpy(site.frames, s.frames, sizeof(s.frames));
+    }
+  }
+  std::vector<const Site*> order;
+  for (const auto& kv : sites) order.push_back(&kv.second);
+  std::sort(order.begin(), order.end(),
+            [](const Site* a, const Site* b) { return a->total_us > b->total_us; });
+  std::ostringstream os;
+  os << "contention profile: " << n << " sampled parks (1/64 sampling), "
+     << sites.size() << " sites; totals are sampled-us (x64 ~= real)\n";
+  os << "  total_us   count  callsite\n";
+  int shown = 0;
+  for (const Site* site : order) {
+    if (++shown > 30) break;
+    os << "  " << site->total_us << "  " << site->count << "  ";
+    for (int d = 1; d < site->nframes && d < 4; ++d) {
+      if (d > 1) os << " <- ";
+      os << frame_name(site->frames[d]);
+    }
+    os << "\n";
+  }
+  return os.str();
+}
 
 // Samples the process for `seconds` (clamped to [1,30]) at `hz` and
 // returns a text report. Serializes concurrent profile requests.

@@ -152,3 +152,36 @@ def test_hotspots_cpu_profile(base):
         stop[0] = True
         for t in ts:
             t.join()
+
+
+def test_contention_page():
+    """/hotspots/contention: sampled butex park sites with symbolized
+    callsites (≙ reference contention profiler)."""
+    import urllib.request
+    port = r.start_echo_server(0)
+    addr = "127.0.0.1:%d" % port
+    for _ in range(300):
+        r.echo_once(addr, b"x" * 64, 2000)
+    body = urllib.request.urlopen(
+        "http://127.0.0.1:%d/hotspots/contention" % port, timeout=10).read().decode()
+    assert "contention profile" in body
+    assert "sampled parks" in body
+
+
+def test_gpu_stats_page():
+    """/hotspots/gpu: HIP runtime telemetry + gpu_wait counters (shows
+    'no GPU runtime' on CPU boxes, real counters on the MI355X)."""
+    import urllib.request
+    port = r.start_echo_server(0)
+    body = urllib.request.urlopen(
+        "http://127.0.0.1:%d/hotspots/gpu" % port, timeout=5).read().decode()
+    assert "gpu_wait_parks" in body
+
+
+def test_memory_page_has_malloc_stats():
+    import urllib.request
+    port = r.start_echo_server(0)
+    body = urllib.request.urlopen(
+        "http://127.0.0.1:%d/memory" % port, timeout=5).read().decode()
+    assert "malloc_in_use_bytes" in body
+    assert "iobuf_block_count" in body
