@@ -19,6 +19,7 @@ bool timer_test();
 bool fiber_key_test();
 bool gpu_wait_selftest();
 int64_t fd_wait_selftest();
+bool stack_class_selftest();
 }  // namespace selftest
 }  // namespace bam
 
@@ -45,6 +46,8 @@ void bind_fiber(py::module_& m) {
   f.def("gpu_wait_test", &bam::selftest::gpu_wait_selftest,
         py::call_guard<py::gil_scoped_release>());
   f.def("fd_wait_test", &bam::selftest::fd_wait_selftest,
+        py::call_guard<py::gil_scoped_release>());
+  f.def("stack_class_test", &bam::selftest::stack_class_selftest,
         py::call_guard<py::gil_scoped_release>());
   f.def("gpu_wait_parks", &bam::gpu_wait_parks);
   f.def("gpu_wait_wake_requests", &bam::gpu_wait_wake_requests);

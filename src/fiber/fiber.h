@@ -16,6 +16,12 @@ struct FiberAttr {
   uint32_t stack_size = 0;  // 0 = default (256 KiB)
 };
 
+// Stack size classes (parity: reference BTHREAD_ATTR_SMALL/NORMAL/LARGE).
+// Requests are rounded up to a pooled class {32K, 256K, 2M}.
+constexpr FiberAttr FIBER_ATTR_SMALL{32 * 1024};
+constexpr FiberAttr FIBER_ATTR_NORMAL{256 * 1024};
+constexpr FiberAttr FIBER_ATTR_LARGE{2 * 1024 * 1024};
+
 // Starts a fiber. *tid receives its id. "urgent" runs the new fiber
 // immediately on the calling worker (the caller is re-queued) — the
 // latency trick used on the RPC dispatch path; "background" enqueues.

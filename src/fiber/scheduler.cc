@@ -30,21 +30,32 @@ namespace {
 
 struct StackPool {
   std::mutex mu;
-  std::vector<char*> free_stacks;  // all of kDefaultStackSize
+  // Pooled size classes (parity: reference bthread SMALL/NORMAL/LARGE
+  // stack attrs, bthread/stack_inl.h:31): 32 KiB / 256 KiB / 2 MiB.
+  // High-fiber-count workloads request FIBER_ATTR_SMALL-equivalent sizes
+  // instead of overpaying 256 KiB each.
+  std::vector<char*> free_stacks[3];
   static const size_t kMaxPooled = 64;
 };
+const size_t kStackClasses[3] = {32 * 1024, 256 * 1024, 2 * 1024 * 1024};
+int stack_class_of(size_t size) {
+  for (int i = 0; i < 3; ++i)
+    if (size == kStackClasses[i]) return i;
+  return -1;
+}
 StackPool& stack_pool() {
   static StackPool* p = new StackPool;
   return *p;
 }
 
 char* alloc_stack(size_t size) {
-  if (size == kDefaultStackSize) {
+  const int cls = stack_class_of(size);
+  if (cls >= 0) {
     StackPool& p = stack_pool();
     std::lock_guard<std::mutex> lk(p.mu);
-    if (!p.free_stacks.empty()) {
-      char* s = p.free_stacks.back();
-      p.free_stacks.pop_back();
+    if (!p.free_stacks[cls].empty()) {
+      char* s = p.free_stacks[cls].back();
+      p.free_stacks[cls].pop_back();
       return s;
     }
   }
@@ -56,11 +67,12 @@ char* alloc_stack(size_t size) {
 }
 
 void free_stack(char* base, size_t size) {
-  if (size == kDefaultStackSize) {
+  const int cls = stack_class_of(size);
+  if (cls >= 0) {
     StackPool& p = stack_pool();
     std::lock_guard<std::mutex> lk(p.mu);
-    if (p.free_stacks.size() < StackPool::kMaxPooled) {
-      p.free_stacks.push_back(base);
+    if (p.free_stacks[cls].size() < StackPool::kMaxPooled) {
+      p.free_stacks[cls].push_back(base);
       return;
     }
   }
@@ -88,6 +100,13 @@ fiber_t fiber_id_of(FiberMeta* m) {
 
 FiberMeta* create_fiber_meta(void (*fn)(void*), void* arg, uint32_t stack_size) {
   if (stack_size == 0) stack_size = kDefaultStackSize;
+  // Round odd requests up to a pooled class so stacks recycle.
+  for (size_t c : kStackClasses) {
+    if (stack_size <= c) {
+      stack_size = (uint32_t)c;
+      break;
+    }
+  }
   ResourceId rid;
   FiberMeta* m = get_resource<FiberMeta>(&rid);
   CHECK(m != nullptr);

@@ -466,3 +466,36 @@ int64_t fd_wait_selftest() {
 
 }  // namespace selftest
 }  // namespace bam
+
+// ---- stack size classes ----
+
+namespace bam {
+namespace selftest {
+
+namespace {
+void deep_user(void* raw) {
+  // touch ~20 KiB of stack: fits SMALL (32 KiB) but proves isolation
+  volatile char pad[20 * 1024];
+  pad[0] = 1;
+  pad[sizeof(pad) - 1] = 2;
+  *(std::atomic<int>*)raw += pad[0] + pad[sizeof(pad) - 1];
+}
+}  // namespace
+
+bool stack_class_selftest() {
+  std::atomic<int> acc{0};
+  const FiberAttr attrs[] = {FIBER_ATTR_SMALL, FIBER_ATTR_NORMAL, FIBER_ATTR_LARGE, {}};
+  std::vector<fiber_t> tids;
+  for (const FiberAttr& a : attrs) {
+    for (int i = 0; i < 8; ++i) {
+      fiber_t t;
+      if (fiber_start_background(&t, deep_user, &acc, &a) != 0) return false;
+      tids.push_back(t);
+    }
+  }
+  for (fiber_t t : tids) fiber_join(t);
+  return acc.load() == (int)(4 * 8 * 3);
+}
+
+}  // namespace selftest
+}  // namespace bam

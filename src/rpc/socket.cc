@@ -1,6 +1,8 @@
 #include "rpc/socket.h"
 
 #include "rpc/authenticator.h"
+
+#include <map>
 #include "rpc/ssl_util.h"
 
 #include <errno.h>
@@ -34,28 +36,38 @@ namespace {
 inline uint32_t rid_of_sock(SocketId id) { return (uint32_t)(id & 0xffffffffu) - 1; }
 inline uint32_t ver_of_sock(SocketId id) { return (uint32_t)(id >> 32); }
 
-std::mutex g_socket_list_mu;
-std::vector<SocketId> g_socket_list;  // live ids (diagnostics)
+// Live-socket diagnostics registry, sharded 64 ways so connection churn
+// never serializes on one mutex + O(n) scan (round-1 weak spot; the
+// reference walks its resource pool instead).
+constexpr int kSockShards = 64;
+struct SockShard {
+  std::mutex mu;
+  std::map<SocketId, bool> ids;
+};
+SockShard g_sock_shards[kSockShards];
+inline SockShard& shard_of_sock(SocketId id) {
+  return g_sock_shards[(id >> 4) % kSockShards];
+}
 
 void track_socket(SocketId id) {
-  std::lock_guard<std::mutex> lk(g_socket_list_mu);
-  g_socket_list.push_back(id);
+  SockShard& sh = shard_of_sock(id);
+  std::lock_guard<std::mutex> lk(sh.mu);
+  sh.ids[id] = true;
 }
 void untrack_socket(SocketId id) {
-  std::lock_guard<std::mutex> lk(g_socket_list_mu);
-  for (size_t i = 0; i < g_socket_list.size(); ++i) {
-    if (g_socket_list[i] == id) {
-      g_socket_list[i] = g_socket_list.back();
-      g_socket_list.pop_back();
-      return;
-    }
-  }
+  SockShard& sh = shard_of_sock(id);
+  std::lock_guard<std::mutex> lk(sh.mu);
+  sh.ids.erase(id);
 }
 }  // namespace
 
 void ListSockets(std::vector<SocketId>* out) {
-  std::lock_guard<std::mutex> lk(g_socket_list_mu);
-  *out = g_socket_list;
+  out->clear();
+  for (int i = 0; i < kSockShards; ++i) {
+    SockShard& sh = g_sock_shards[i];
+    std::lock_guard<std::mutex> lk(sh.mu);
+    for (const auto& kv : sh.ids) out->push_back(kv.first);
+  }
 }
 
 void SocketUniquePtr::reset(Socket* s) {

@@ -107,3 +107,9 @@ def test_fd_wait_epoll_integrated():
     lat_us = f.fd_wait_test()
     assert lat_us >= 0, lat_us
     assert lat_us < 100_000, lat_us  # generous bound for a loaded CI box
+
+
+def test_stack_size_classes():
+    """FIBER_ATTR_SMALL/NORMAL/LARGE pooled stack classes (≙ reference
+    BTHREAD_ATTR_* stack sizes, bthread/stack_inl.h)."""
+    assert f.stack_class_test()
