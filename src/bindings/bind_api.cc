@@ -372,6 +372,7 @@ void bind_memcache(py::module_& m) {
 
 // ---- json2pb ----
 #include "rpc/json2pb.h"
+#include "rpc/thrift_codec.h"
 
 namespace {
 
@@ -457,8 +458,141 @@ void bind_json2pb(py::module_& m) {
   });
 }
 
+// ---- thrift struct codec (rpc/thrift_codec.h) ----
+// JSON-described structs: keys are "id:type" with type in {bool, byte,
+// i16, i32, i64, double, str, struct, list:<t>, set:<t>, map:<kt>:<vt>}.
+namespace {
+
+bam::thrift::TType ttype_of(const std::string& t) {
+  using namespace bam::thrift;
+  if (t == "bool") return T_BOOL;
+  if (t == "byte") return T_BYTE;
+  if (t == "i16") return T_I16;
+  if (t == "i32") return T_I32;
+  if (t == "i64") return T_I64;
+  if (t == "double") return T_DOUBLE;
+  if (t == "str") return T_STRING;
+  if (t == "struct") return T_STRUCT;
+  throw std::runtime_error("bad thrift type " + t);
+}
+
+bam::thrift::TValue tvalue_from_py(const std::string& type_desc, py::handle v);
+
+void tstruct_from_py(py::dict d, bam::thrift::TStruct* out) {
+  for (auto item : d) {
+    std::string key = item.first.cast<std::string>();
+    size_t colon = key.find(':');
+    if (colon == std::string::npos) throw std::runtime_error("key needs id:type");
+    int16_t id = (int16_t)atoi(key.substr(0, colon).c_str());
+    out->emplace_back(id, tvalue_from_py(key.substr(colon + 1), item.second));
+  }
+}
+
+bam::thrift::TValue tvalue_from_py(const std::string& td, py::handle v) {
+  using namespace bam::thrift;
+  if (td == "bool") return TValue::Bool(v.cast<bool>());
+  if (td == "byte") return TValue::Byte((int8_t)v.cast<int64_t>());
+  if (td == "i16") return TValue::I16((int16_t)v.cast<int64_t>());
+  if (td == "i32") return TValue::I32((int32_t)v.cast<int64_t>());
+  if (td == "i64") return TValue::I64(v.cast<int64_t>());
+  if (td == "double") return TValue::Double(v.cast<double>());
+  if (td == "str") return TValue::Str(v.cast<std::string>());
+  if (td == "struct") {
+    TValue x = TValue::Struct();
+    tstruct_from_py(v.cast<py::dict>(), x.st.get());
+    return x;
+  }
+  if (td.rfind("list:", 0) == 0 || td.rfind("set:", 0) == 0) {
+    const bool is_set = td[0] == 's';
+    std::string et = td.substr(td.find(':') + 1);
+    TValue x = is_set ? TValue::Set(ttype_of(et)) : TValue::List(ttype_of(et));
+    for (auto e : v.cast<py::list>()) x.list->push_back(tvalue_from_py(et, e));
+    return x;
+  }
+  if (td.rfind("map:", 0) == 0) {
+    size_t c1 = td.find(':'), c2 = td.find(':', c1 + 1);
+    std::string kt = td.substr(c1 + 1, c2 - c1 - 1), vt = td.substr(c2 + 1);
+    TValue x = TValue::Map(ttype_of(kt), ttype_of(vt));
+    for (auto e : v.cast<py::dict>())
+      x.map->emplace_back(tvalue_from_py(kt, e.first), tvalue_from_py(vt, e.second));
+    return x;
+  }
+  throw std::runtime_error("bad thrift type " + td);
+}
+
+py::object tvalue_to_py(const bam::thrift::TValue& v);
+
+py::dict tstruct_to_py(const bam::thrift::TStruct& st) {
+  using namespace bam::thrift;
+  py::dict d;
+  for (const auto& kv : st) {
+    const TValue& v = kv.second;
+    std::string t;
+    switch (v.type) {
+      case T_BOOL: t = "bool"; break;
+      case T_BYTE: t = "byte"; break;
+      case T_I16: t = "i16"; break;
+      case T_I32: t = "i32"; break;
+      case T_I64: t = "i64"; break;
+      case T_DOUBLE: t = "double"; break;
+      case T_STRING: t = "str"; break;
+      case T_STRUCT: t = "struct"; break;
+      case T_LIST: t = "list:?"; break;
+      case T_SET: t = "set:?"; break;
+      case T_MAP: t = "map:?:?"; break;
+      default: t = "?";
+    }
+    d[py::str(std::to_string(kv.first) + ":" + t)] = tvalue_to_py(v);
+  }
+  return d;
+}
+
+py::object tvalue_to_py(const bam::thrift::TValue& v) {
+  using namespace bam::thrift;
+  switch (v.type) {
+    case T_BOOL: return py::bool_(v.i != 0);
+    case T_BYTE:
+    case T_I16:
+    case T_I32:
+    case T_I64: return py::int_(v.i);
+    case T_DOUBLE: return py::float_(v.d);
+    case T_STRING: return py::bytes(v.s);
+    case T_STRUCT: return tstruct_to_py(v.st != nullptr ? *v.st : TStruct());
+    case T_LIST:
+    case T_SET: {
+      py::list l;
+      if (v.list != nullptr)
+        for (const auto& e : *v.list) l.append(tvalue_to_py(e));
+      return std::move(l);
+    }
+    case T_MAP: {
+      py::dict d;
+      if (v.map != nullptr)
+        for (const auto& e : *v.map) d[tvalue_to_py(e.first)] = tvalue_to_py(e.second);
+      return std::move(d);
+    }
+    default: return py::none();
+  }
+}
+
+}  // namespace
+
 // ---- thrift passthrough client ----
 void bind_thrift(py::module_& m) {
+  m.def("thrift_struct_encode", [](py::dict d) {
+    bam::thrift::TStruct st;
+    tstruct_from_py(d, &st);
+    std::string out;
+    bam::thrift::WriteStruct(st, &out);
+    return py::bytes(out);
+  });
+  m.def("thrift_struct_decode", [](py::bytes data) {
+    std::string in(data);
+    bam::thrift::TStruct st;
+    if (!bam::thrift::ReadStruct(in.data(), in.size(), &st))
+      throw std::runtime_error("malformed thrift struct");
+    return tstruct_to_py(st);
+  });
   m.def("thrift_call", [](const std::string& addr, const std::string& method,
                           const std::string& payload, int timeout_ms) {
     bam::ChannelOptions opts;

@@ -66,3 +66,79 @@ def test_thrift_raw_socket_client(thrift_port):
     seqid, = struct.unpack(">I", data[12 + nlen:16 + nlen])
     assert seqid == 7
     assert data[16 + nlen:] == tbinary_string_struct(0, b"RAW")
+
+
+def test_thrift_struct_codec_roundtrip():
+    """Round-2 thrift struct codec (rpc/thrift_codec.*): real TBinary
+    struct build/inspect instead of opaque passthrough (round-1 gap)."""
+    t = b.core
+    src = {
+        "1:i32": -42,
+        "2:str": b"hello thrift",
+        "3:bool": True,
+        "4:double": 2.5,
+        "5:i64": -(1 << 45),
+        "6:list:i32": [1, 2, 300000],
+        "7:map:str:i64": {b"a": 1, b"bb": -2},
+        "8:struct": {"1:str": b"nested", "2:i16": 7},
+        "9:set:str": [b"x", b"y"],
+    }
+    wire = t.thrift_struct_encode(src)
+    back = t.thrift_struct_decode(wire)
+    assert back["1:i32"] == -42
+    assert back["2:str"] == b"hello thrift"
+    assert back["3:bool"] is True
+    assert back["4:double"] == 2.5
+    assert back["5:i64"] == -(1 << 45)
+    assert back["6:list:?"] == [1, 2, 300000]
+    assert back["7:map:?:?"] == {b"a": 1, b"bb": -2}
+    assert back["8:struct"]["1:str"] == b"nested"
+    assert back["8:struct"]["2:i16"] == 7
+    assert sorted(back["9:set:?"]) == [b"x", b"y"]
+
+
+def test_thrift_struct_canonical_bytes():
+    """Wire bytes match the canonical TBinaryProtocol layout (so real
+    thrift runtimes parse them): field hdr = <type u8><id i16be>."""
+    t = b.core
+    wire = t.thrift_struct_encode({"1:i32": 5})
+    assert wire == bytes([8, 0, 1, 0, 0, 0, 5, 0])  # T_I32, id 1, 5, T_STOP
+    wire2 = t.thrift_struct_encode({"3:str": b"ab"})
+    assert wire2 == bytes([11, 0, 3, 0, 0, 0, 2]) + b"ab" + bytes([0])
+    wire3 = t.thrift_struct_encode({"2:list:i16": [1, 2]})
+    assert wire3 == bytes([15, 0, 2, 6, 0, 0, 0, 2, 0, 1, 0, 2, 0])
+
+
+def test_thrift_struct_malformed_rejected():
+    import pytest as _pytest
+    t = b.core
+    for bad in [bytes([8, 0, 1, 0, 0]),        # truncated i32
+                bytes([11, 0, 1, 0xFF, 0xFF, 0xFF, 0xFF, 0]),  # negative strlen
+                bytes([99, 0, 1, 0])]:         # unknown type
+        with _pytest.raises(RuntimeError):
+            t.thrift_struct_decode(bad)
+
+
+def test_thrift_rpc_with_struct_payload():
+    """End-to-end: a thrift call whose args/result are REAL structs built
+    with the codec, over the framed TBinary protocol."""
+    t = b.core
+    srv = b.Server()
+
+    def echo(req, att):
+        # the handler INSPECTS the request with the codec and builds a
+        # typed result struct (not opaque passthrough)
+        d = t.thrift_struct_decode(req)
+        return t.thrift_struct_encode({
+            "0:str": d["1:str"].upper(),
+            "1:i32": d["2:i32"] + 1,
+        })
+
+    srv.add_method("thrift", "Echo", echo)
+    port = srv.start(0)
+    args = t.thrift_struct_encode({"1:str": b"codec-payload", "2:i32": 99})
+    resp = b.thrift_call("127.0.0.1:%d" % port, "Echo", args, 3000)
+    back = t.thrift_struct_decode(resp)
+    assert back["0:str"] == b"CODEC-PAYLOAD"
+    assert back["1:i32"] == 100
+    srv.stop()
