@@ -62,6 +62,13 @@ int Channel::Init(const char* naming_url, const char* lb_name, const ChannelOpti
 
 static void split_full_method(const std::string& full, std::string* service,
                               std::string* method) {
+  // Absolute paths ("/v1/x/y.txt", http-ish protocols) stay whole —
+  // splitting on the last '.'/'/' would mangle dotted file components.
+  if (!full.empty() && full[0] == '/') {
+    *service = full;
+    method->clear();
+    return;
+  }
   size_t pos = full.find_last_of("./");
   if (pos == std::string::npos) {
     *service = "";

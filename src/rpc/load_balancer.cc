@@ -170,6 +170,76 @@ int ResolveNamingUrl(const std::string& url, std::vector<EndPoint>* out) {
     }
     return 0;
   }
+  if (url.rfind("nacos://", 0) == 0) {
+    // nacos://host:port/serviceName[?namespaceId=..&groupName=..] —
+    // GET /nacos/v1/ns/instance/list?serviceName=<svc>&healthyOnly=true
+    // (parity: reference policy/nacos_naming_service.cpp); entries come
+    // from hosts[].{ip,port}, healthy/enabled only.
+    std::string rest = url.substr(8);
+    size_t slash = rest.find('/');
+    if (slash == std::string::npos) return -1;
+    std::string agent = rest.substr(0, slash);
+    std::string svc_and_params = rest.substr(slash + 1);
+    std::string extra;
+    size_t q = svc_and_params.find('?');
+    if (q != std::string::npos) {
+      extra = "&" + svc_and_params.substr(q + 1);
+      svc_and_params = svc_and_params.substr(0, q);
+    }
+    Channel ch;
+    ChannelOptions copt;
+    copt.protocol = "http";
+    copt.timeout_ms = 3000;
+    copt.max_retry = 0;
+    if (ch.Init(agent.c_str(), &copt) != 0) return -1;
+    Controller cntl;
+    IOBuf req, resp;
+    ch.CallMethod("/nacos/v1/ns/instance/list?serviceName=" + svc_and_params +
+                      "&healthyOnly=true" + extra,
+                  &cntl, &req, &resp, nullptr);
+    if (cntl.Failed()) return -1;
+    json::Value root;
+    if (!json::Parse(resp.to_string(), &root) || root.type != json::Value::OBJECT) return -1;
+    auto hit = root.obj->find("hosts");
+    if (hit == root.obj->end() || hit->second.type != json::Value::ARRAY) return -1;
+    for (const json::Value& host : *hit->second.arr) {
+      if (host.type != json::Value::OBJECT) continue;
+      const json::Object& ho = *host.obj;
+      auto en = ho.find("enabled");
+      if (en != ho.end() && en->second.type == json::Value::BOOL && !en->second.b) continue;
+      auto he = ho.find("healthy");
+      if (he != ho.end() && he->second.type == json::Value::BOOL && !he->second.b) continue;
+      auto iit = ho.find("ip");
+      auto pit = ho.find("port");
+      if (iit == ho.end() || pit == ho.end()) continue;
+      EndPoint ep;
+      std::string hp = iit->second.str + ":" + std::to_string((int)pit->second.num);
+      if (str2endpoint(hp.c_str(), &ep) == 0) out->push_back(ep);
+    }
+    return 0;
+  }
+  if (url.rfind("remotefile://", 0) == 0) {
+    // remotefile://host:port/path — fetch a server-list file over HTTP
+    // (parity: reference policy/remote_file_naming_service.cpp); same
+    // line format as file://.
+    std::string rest = url.substr(13);
+    size_t slash = rest.find('/');
+    if (slash == std::string::npos) return -1;
+    std::string agent = rest.substr(0, slash);
+    std::string path = rest.substr(slash);
+    Channel ch;
+    ChannelOptions copt;
+    copt.protocol = "http";
+    copt.timeout_ms = 3000;
+    copt.max_retry = 0;
+    if (ch.Init(agent.c_str(), &copt) != 0) return -1;
+    Controller cntl;
+    IOBuf req, resp;
+    ch.CallMethod(path, &cntl, &req, &resp, nullptr);
+    if (cntl.Failed()) return -1;
+    parse_csv(resp.to_string());
+    return 0;
+  }
   // bare "host:port" treated as a single-entry list
   EndPoint ep;
   if (str2endpoint(url.c_str(), &ep) == 0) {

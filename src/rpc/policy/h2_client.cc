@@ -156,7 +156,7 @@ int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
   const std::string& svc = cntl->call.service_name;
   const std::string& mn = cntl->call.method_name;
   std::string path = !mn.empty() && mn[0] == '/' ? mn
-                     : !svc.empty() && svc[0] == '/' ? svc + "/" + mn
+                     : !svc.empty() && svc[0] == '/' ? (mn.empty() ? svc : svc + "/" + mn)
                      : svc.empty() && !grpc          ? "/" + mn
                                                      : "/" + svc + "/" + mn;
   std::string authority = endpoint2str(cntl->remote_side());

@@ -233,10 +233,9 @@ void ProcessHttpResponse(InputMessageBase* mb) {
 void PackHttp1Request(IOBuf* out, Controller* cntl, uint64_t /*fifo-correlated*/) {
   const std::string& svc = cntl->call.service_name;
   const std::string& m = cntl->call.method_name;
-  // split_full_method cuts at the LAST '/', so an absolute-path method
-  // ("/v1/x/y") arrives as svc="/v1/x", m="y" — rejoin without doubling.
+  // Absolute-path methods arrive whole in svc (split_full_method).
   std::string path = !m.empty() && m[0] == '/' ? m
-                     : !svc.empty() && svc[0] == '/' ? svc + "/" + m
+                     : !svc.empty() && svc[0] == '/' ? (m.empty() ? svc : svc + "/" + m)
                      : svc.empty()                   ? "/" + m
                                                      : "/" + svc + "/" + m;
   const IOBuf& body = cntl->call.request_buf;

@@ -41,3 +41,46 @@ def test_consul_naming():
 def test_dns_resolver_direct():
     eps = b.core.rpc.resolve_naming("dns://localhost:8123")
     assert "127.0.0.1:8123" in eps
+
+
+def test_nacos_naming():
+    """nacos:// (parity: reference policy/nacos_naming_service.cpp):
+    instance list API against a scripted nacos on our own HTTP server;
+    unhealthy/disabled instances filtered."""
+    echo_port = r.start_echo_server(0)
+    agent = b.Server()
+
+    def instances(req, att):
+        body = json.dumps({"hosts": [
+            {"ip": "127.0.0.1", "port": echo_port, "healthy": True, "enabled": True},
+            {"ip": "127.0.0.1", "port": 1, "healthy": False, "enabled": True},
+            {"ip": "127.0.0.1", "port": 2, "healthy": True, "enabled": False},
+            {"ip": "127.0.0.1", "port": echo_port + 0, "healthy": True, "enabled": True},
+        ]}).encode()
+        return body, b""
+
+    agent.add_method("nacos", "instances", instances)
+    agent.add_restful_mapping("nacos", "/nacos/v1/ns/instance/list => instances")
+    agent_port = agent.start(0)
+    eps = b.core.rpc.resolve_naming("nacos://127.0.0.1:%d/echo-svc" % agent_port)
+    assert eps == ["127.0.0.1:%d" % echo_port] * 2, eps
+    agent.stop()
+
+
+def test_remotefile_naming():
+    """remotefile:// (parity: reference remote_file_naming_service): a
+    server-list file fetched over HTTP, same format as file://."""
+    echo_port = r.start_echo_server(0)
+    agent = b.Server()
+
+    def listing(req, att):
+        return ("127.0.0.1:%d\n# comment\n127.0.0.1:%d\n" %
+                (echo_port, echo_port)).encode(), b""
+
+    agent.add_method("files", "servers", listing)
+    agent.add_restful_mapping("files", "/lists/servers.txt => servers")
+    agent_port = agent.start(0)
+    eps = b.core.rpc.resolve_naming(
+        "remotefile://127.0.0.1:%d/lists/servers.txt" % agent_port)
+    assert eps == ["127.0.0.1:%d" % echo_port] * 2, eps
+    agent.stop()
