@@ -9,6 +9,10 @@
 #include <sys/socket.h>
 #include <unistd.h>
 
+#include <openssl/bio.h>
+#include <openssl/evp.h>
+#include <openssl/pem.h>
+#include <openssl/rsa.h>
 #include <openssl/sha.h>
 
 #include "base/codecs.h"
@@ -256,7 +260,9 @@ int MysqlClient::Connect(const std::string& host, int port, const std::string& u
   }
 
   // ---- auth continuation: OK / ERR / AuthSwitch (0xFE) / AuthMoreData (0x01)
-  for (int hop = 0; hop < 4; ++hop) {
+  std::string current_nonce = salt;
+  bool sent_pubkey_request = false;
+  for (int hop = 0; hop < 6; ++hop) {
     std::string fin;
     if (read_packet(&fin, &seq) != 0 || fin.empty()) {
       Close();
@@ -278,6 +284,7 @@ int MysqlClient::Connect(const std::string& host, int port, const std::string& u
       std::string plugin = fin.substr(1, z3 - 1);
       std::string nonce = fin.substr(z3 + 1);
       while (!nonce.empty() && nonce.back() == '\0') nonce.pop_back();
+      current_nonce = nonce;
       std::string sc = scramble_for(plugin, password, nonce);
       if (write_packet(sc, (uint8_t)(seq + 1)) != 0) {
         Close();
@@ -287,7 +294,56 @@ int MysqlClient::Connect(const std::string& host, int port, const std::string& u
     }
     if (tag == 0x01) {  // AuthMoreData (caching_sha2)
       if (fin.size() >= 2 && (uint8_t)fin[1] == 0x03) continue;  // fast auth ok -> OK next
-      // 0x04 = full auth needed (requires TLS or RSA key exchange)
+      if (fin.size() >= 2 && (uint8_t)fin[1] == 0x04 && !sent_pubkey_request) {
+        // FULL auth over plain TCP (parity: reference policy/mysql full
+        // caching_sha2): request the server's RSA public key (0x02), then
+        // send RSA-OAEP(password||NUL XOR nonce).
+        sent_pubkey_request = true;
+        if (write_packet(std::string(1, '\x02'), (uint8_t)(seq + 1)) != 0) {
+          Close();
+          return -1;
+        }
+        continue;
+      }
+      if (sent_pubkey_request && fin.size() > 1) {
+        // AuthMoreData carrying the PEM public key.
+        std::string pem = fin.substr(1);
+        std::string plain = password;
+        plain.push_back('\0');
+        for (size_t i = 0; i < plain.size(); ++i)
+          plain[i] = (char)(plain[i] ^ current_nonce[i % current_nonce.size()]);
+        std::string enc;
+        {
+          BIO* bio = BIO_new_mem_buf(pem.data(), (int)pem.size());
+          EVP_PKEY* pkey = bio != nullptr ? PEM_read_bio_PUBKEY(bio, nullptr, nullptr, nullptr)
+                                          : nullptr;
+          if (bio != nullptr) BIO_free(bio);
+          EVP_PKEY_CTX* ctx = pkey != nullptr ? EVP_PKEY_CTX_new(pkey, nullptr) : nullptr;
+          bool ok = ctx != nullptr && EVP_PKEY_encrypt_init(ctx) > 0 &&
+                    EVP_PKEY_CTX_set_rsa_padding(ctx, RSA_PKCS1_OAEP_PADDING) > 0;
+          size_t outlen = 0;
+          if (ok)
+            ok = EVP_PKEY_encrypt(ctx, nullptr, &outlen, (const uint8_t*)plain.data(),
+                                  plain.size()) > 0;
+          if (ok) {
+            enc.resize(outlen);
+            ok = EVP_PKEY_encrypt(ctx, (uint8_t*)&enc[0], &outlen,
+                                  (const uint8_t*)plain.data(), plain.size()) > 0;
+            enc.resize(outlen);
+          }
+          if (ctx != nullptr) EVP_PKEY_CTX_free(ctx);
+          if (pkey != nullptr) EVP_PKEY_free(pkey);
+          if (!ok) {
+            Close();
+            return -1;
+          }
+        }
+        if (write_packet(enc, (uint8_t)(seq + 1)) != 0) {
+          Close();
+          return -1;
+        }
+        continue;
+      }
       Close();
       return -1;
     }

@@ -266,3 +266,78 @@ def test_mysql_caching_sha2_fast_auth():
     assert result["auth_ok"], "caching_sha2 scramble mismatch"
     c.close()
     srv.close()
+
+
+def test_mysql_caching_sha2_full_auth_rsa():
+    """caching_sha2 FULL auth over plain TCP (round-1 gap): server demands
+    full auth (0x01 0x04), client requests the RSA public key (0x02),
+    server sends PEM, client sends RSA-OAEP(password||NUL XOR nonce).
+    The mock decrypts with openssl pkeyutl and verifies."""
+    import socket
+    import subprocess
+    import tempfile
+    import threading
+    import os
+
+    tmp = tempfile.mkdtemp()
+    priv = os.path.join(tmp, "k.pem")
+    pub = os.path.join(tmp, "k.pub")
+    subprocess.run(["openssl", "genrsa", "-out", priv, "2048"], check=True,
+                   capture_output=True)
+    subprocess.run(["openssl", "rsa", "-in", priv, "-pubout", "-out", pub],
+                   check=True, capture_output=True)
+    pem_pub = open(pub, "rb").read()
+
+    srv = socket.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+    result = {}
+
+    def read_pkt(c):
+        head = c.recv(4)
+        ln = struct.unpack("<I", head[:3] + b"\x00")[0]
+        data = b""
+        while len(data) < ln:
+            data += c.recv(ln - len(data))
+        return data
+
+    def run():
+        c, _ = srv.accept()
+        hs = bytes([10]) + b"8.4.0-fake\0" + struct.pack("<I", 5)
+        hs += SALT[:8] + b"\x00"
+        hs += struct.pack("<H", 0xFFFF)
+        hs += bytes([33]) + struct.pack("<H", 2) + struct.pack("<H", 0xFFFF >> 16)
+        hs += bytes([21]) + b"\x00" * 10
+        hs += SALT[8:20] + b"\x00"
+        hs += b"caching_sha2_password\x00"
+        c.sendall(_packet(hs, 0))
+        read_pkt(c)                      # HandshakeResponse41
+        c.sendall(_packet(b"\x01\x04", 2))  # full auth required
+        req = read_pkt(c)
+        result["asked_key"] = req == b"\x02"
+        c.sendall(_packet(b"\x01" + pem_pub, 4))
+        enc = read_pkt(c)
+        with open(os.path.join(tmp, "enc.bin"), "wb") as f:
+            f.write(enc)
+        out = subprocess.run(
+            ["openssl", "pkeyutl", "-decrypt", "-inkey", priv,
+             "-in", os.path.join(tmp, "enc.bin"),
+             "-pkeyopt", "rsa_padding_mode:oaep"],
+            capture_output=True)
+        plain = out.stdout
+        expect = bytes((PASSWORD.encode() + b"\x00")[i] ^ SALT[i % len(SALT)]
+                       for i in range(len(PASSWORD) + 1))
+        result["decrypted_ok"] = plain == expect
+        c.sendall(_packet(b"\x00\x00\x00\x02\x00\x00\x00", 6))  # OK
+        c.recv(64)
+        c.close()
+
+    threading.Thread(target=run, daemon=True).start()
+    port = srv.getsockname()[1]
+    c = b.core.rpc.MysqlClient()
+    rc = c.connect("127.0.0.1", port, USER, PASSWORD)
+    assert rc == 0
+    assert result["asked_key"]
+    assert result["decrypted_ok"], "RSA-OAEP password blob mismatch"
+    c.close()
+    srv.close()
