@@ -1,5 +1,6 @@
 #include "rpc/load_balancer.h"
 #include "rpc/mysql_client.h"
+#include "rpc/flv.h"
 #include "rpc/rtmp_client.h"
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -50,6 +51,53 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
 void bind_rpc(py::module_& m) {
   auto r = m.def_submodule("rpc");
   r.def("rdma_live_recv_blocks", &bam::rdma::live_recv_blocks);
+  // RTMP play session remuxed into a standard FLV document (parity:
+  // reference rtmp.cpp FLV writer): collects up to ntags media messages.
+  r.def("rtmp_play_to_flv",
+        [](const std::string& host, int port, const std::string& app,
+           const std::string& stream, int ntags, int timeout_ms) {
+          std::string flv_doc;
+          {
+            py::gil_scoped_release rel;
+            bam::RtmpClient c;
+            if (c.Connect(host, port, app, timeout_ms) != 0)
+              throw std::runtime_error("rtmp connect failed");
+            if (c.Play(stream) != 0) throw std::runtime_error("rtmp play failed");
+            bam::flv::AppendHeader(&flv_doc);
+            for (int i = 0; i < ntags; ++i) {
+              bam::rtmp::Message m;
+              if (c.PollFrame(&m, timeout_ms) != 0) break;
+              if (m.type == 8 || m.type == 9 || m.type == 18)
+                bam::flv::AppendTag(&flv_doc, m.type, m.timestamp, m.payload);
+              else
+                --i;  // control message: not an FLV tag
+            }
+            c.Close();
+          }
+          return py::bytes(flv_doc);
+        },
+        py::arg("host"), py::arg("port"), py::arg("app"), py::arg("stream"),
+        py::arg("ntags"), py::arg("timeout_ms") = 5000);
+  r.def("flv_parse", [](py::bytes doc) {
+    std::vector<bam::flv::Tag> tags;
+    bool ha = false, hv = false;
+    if (!bam::flv::Parse(std::string(doc), &tags, &ha, &hv))
+      throw std::runtime_error("malformed FLV");
+    py::list out;
+    for (auto& t : tags)
+      out.append(py::make_tuple((int)t.type, t.timestamp, py::bytes(t.payload)));
+    return py::make_tuple(ha, hv, out);
+  });
+  r.def("flv_build", [](py::list tags) {
+    std::string doc;
+    bam::flv::AppendHeader(&doc);
+    for (auto t : tags) {
+      py::tuple tt = t.cast<py::tuple>();
+      bam::flv::AppendTag(&doc, (uint8_t)tt[0].cast<int>(), tt[1].cast<uint32_t>(),
+                          tt[2].cast<std::string>());
+    }
+    return py::bytes(doc);
+  });
   r.def("pb_stub_test", []() {
     std::string err;
     bool ok;

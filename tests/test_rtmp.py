@@ -71,3 +71,44 @@ def test_std_rpc_shares_port(port):
                                     "EchoService.Echo", b"rtmp-port")
     assert rc == 0, err
     assert resp == b"rtmp-port"
+
+
+def test_flv_build_and_parse():
+    """FLV container (rpc/flv.*, ≙ reference rtmp.cpp FLV writer): exact
+    header layout + PreviousTagSize chain + extended timestamps."""
+    tags = [(9, 100, b"\x17video"), (8, 120, b"\xafaudio"),
+            (9, (1 << 24) + 5, b"\x27late")]  # needs the extended ts byte
+    doc = r.flv_build(tags)
+    assert doc[:5] == b"FLV\x01\x05"
+    ha, hv, back = r.flv_parse(doc)
+    assert ha and hv
+    assert back == tags
+
+
+def test_rtmp_play_remuxed_to_flv(port):
+    """End-to-end remux: publisher pushes media through the relay; a play
+    session is remuxed into a standards-layout FLV document."""
+    pub = r.RtmpClient()
+    assert pub.connect("127.0.0.1", port, "live") == 0
+    assert pub.publish("flvcam") == 0
+
+    import threading
+    result = {}
+
+    def play():
+        result["flv"] = r.rtmp_play_to_flv("127.0.0.1", port, "live", "flvcam",
+                                           3, 8000)
+
+    t = threading.Thread(target=play)
+    t.start()
+    import time
+    time.sleep(0.3)  # let the player subscribe
+    frames = [(VIDEO, 0, b"\x17" + b"k" * 400),
+              (AUDIO, 21, b"\xaf" + b"s" * 80),
+              (VIDEO, 42, b"\x27" + b"d" * 900)]
+    for ty, ts, payload in frames:
+        assert pub.push_frame(ty, ts, payload) == 0
+    t.join(timeout=15)
+    pub.close()
+    _, _, tags = r.flv_parse(result["flv"])
+    assert [(ty, ts, p) for ty, ts, p in tags] == frames
