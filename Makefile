@@ -13,9 +13,12 @@ CXXFLAGS  := -O2 -g -std=c++17 -fPIC -pthread -Wall -Wno-unused-function \
              -Isrc -I. -I/usr/include -I/opt/conda/include -msse4.2 -fno-omit-frame-pointer -MMD -MP
 LDFLAGS   := -shared -pthread -ldl -lz -lssl -lcrypto
 
+# testsupport/ = C++ test scenario drivers compiled into the library and
+# driven from pytest (fibers cannot run Python code); kept separate from
+# product source for honest accounting.
 CORE_SRCS := $(wildcard src/base/*.cc) $(wildcard src/fiber/*.cc) $(wildcard src/rpc/*.cc) \
              $(wildcard src/rpc/policy/*.cc) $(wildcard src/rpc/builtin/*.cc) \
-             $(wildcard src/var/*.cc)
+             $(wildcard src/var/*.cc) $(wildcard src/testsupport/*.cc)
 CORE_ASM  := $(wildcard src/fiber/*.S)
 BIND_SRCS := $(wildcard src/bindings/*.cc)
 CORE_OBJS := $(CORE_SRCS:%.cc=build/%.o) $(CORE_ASM:%.S=build/%.o)
