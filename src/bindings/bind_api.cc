@@ -205,6 +205,39 @@ void bind_api(py::module_& m) {
            py::arg("attachment") = std::string(), py::arg("timeout_ms") = 0,
            py::arg("log_id") = 0);
 
+  m.def("http_call_progressive",
+        [](const std::string& addr, const std::string& method, py::bytes request,
+           int timeout_ms) {
+          bam::ChannelOptions opts;
+          opts.protocol = "http";
+          opts.timeout_ms = timeout_ms;
+          opts.max_retry = 0;
+          bam::Channel ch;
+          if (ch.Init(addr.c_str(), &opts) != 0)
+            throw std::runtime_error("channel init failed");
+          bam::Controller cntl;
+          cntl.set_timeout_ms(timeout_ms);
+          std::vector<std::string> chunks;
+          bool saw_done = false;
+          cntl.response_read_progressively(
+              [&chunks, &saw_done](const bam::IOBuf& c, bool done) {
+                if (!c.empty()) chunks.push_back(c.to_string());
+                if (done) saw_done = true;
+              });
+          bam::IOBuf req, resp;
+          req.append(std::string(request));
+          {
+            py::gil_scoped_release rel;
+            ch.CallMethod(method, &cntl, &req, &resp, nullptr);
+          }
+          if (cntl.Failed()) throw PyRpcError(cntl.ErrorCode(), cntl.ErrorText());
+          py::list out;
+          for (auto& c : chunks) out.append(py::bytes(c));
+          return py::make_tuple(out, saw_done, py::bytes(resp.to_string()));
+        },
+        py::arg("addr"), py::arg("method"), py::arg("request") = py::bytes(""),
+        py::arg("timeout_ms") = 5000);
+
   m.def("gen_self_signed_cert", [](const std::string& cn) {
     std::string cert, key;
     if (ssl::GenerateSelfSignedCert(&cert, &key, cn) != 0) {

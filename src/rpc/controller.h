@@ -61,6 +61,19 @@ class Controller {
   }
   void SetFailed(const std::string& reason) { SetFailed(EINTERNAL, reason); }
 
+  // ---- progressive response reading (parity: reference
+  // response_read_progressively / ProgressiveReader, socket.h:662) ----
+  // When set before CallMethod on an "http" channel, response BODY bytes
+  // are delivered to the reader incrementally as they arrive from the
+  // socket (chunk-by-chunk for chunked coding, read-by-read otherwise)
+  // instead of buffering the whole body; done=true marks the last chunk.
+  // The final response IOBuf stays empty.
+  typedef std::function<void(const IOBuf& chunk, bool done)> ProgressiveReader;
+  void response_read_progressively(ProgressiveReader r) {
+    progressive_reader_ = std::move(r);
+  }
+  const ProgressiveReader& progressive_reader() const { return progressive_reader_; }
+
   SessionId call_id() const { return cid_; }
   // Client-side cancellation (parity: reference StartCancel/IsCanceled):
   // conducts ECANCELED through the call's session — done runs (or the
@@ -133,6 +146,7 @@ class Controller {
   CompressType request_compress_ = COMPRESS_TYPE_NONE;
   IOBuf request_attachment_;
   IOBuf response_attachment_;
+  ProgressiveReader progressive_reader_;
   EndPoint remote_side_;
   EndPoint local_side_;
   SessionId cid_ = 0;

@@ -122,7 +122,123 @@ struct HttpClientMessage : public InputMessageBase {
   IOBuf body;
 };
 
-ParseResult ParseHttpResponse(IOBuf* source, Socket* /*sock*/, bool /*eof*/) {
+// ---- progressive response state (Controller::response_read_progressively,
+// parity: reference ProgressiveReader / socket.h:662) ----
+// Lives as the client socket's protocol_ctx while a progressive body is
+// being streamed; body bytes are delivered to the reader as they arrive
+// instead of buffering (the 1 MB+ payload path).
+const std::string CRLF("\r\n");
+
+struct HttpProgressiveCtx {
+  bool active = false;
+  bool chunked = false;
+  size_t remaining = 0;    // content-length mode
+  int64_t chunk_left = -1; // chunked mode: -1 = expect a chunk-size line
+  uint64_t cid = 0;
+};
+
+int g_http_protocol_index = -1;
+
+// Feeds available body bytes to the progressive reader. Returns true when
+// the body completed (RPC ended), false when more bytes are needed.
+bool progressive_pump(IOBuf* source, Socket* sock, HttpProgressiveCtx* ctx) {
+  for (;;) {
+    IOBuf chunk;
+    bool done = false;
+    if (!ctx->chunked) {
+      size_t take = std::min(ctx->remaining, source->size());
+      if (take == 0 && ctx->remaining > 0) return false;
+      source->cutn(&chunk, take);
+      ctx->remaining -= take;
+      done = ctx->remaining == 0;
+    } else {
+      if (ctx->chunk_left < 0) {
+        // need a "<hex>CRLF" size line
+        IOBuf line;
+        if (source->cut_until(&line, CRLF) != 0) return false;
+        std::string l = line.to_string();
+        char* endp = nullptr;
+        unsigned long long v = strtoull(l.c_str(), &endp, 16);
+        if (endp == l.c_str() || v > (unsigned long long)FLAG_http_max_body_size) {
+          // malformed: fail the call below via a zero-length done
+          ctx->chunk_left = 0;
+          v = 0;
+        }
+        ctx->chunk_left = (int64_t)v;
+        if (ctx->chunk_left == 0) {
+          // consume trailing CRLF (trailer-less)
+          IOBuf fin;
+          if (source->cut_until(&fin, CRLF) != 0) return false;
+          done = true;
+        }
+      }
+      if (!done && ctx->chunk_left > 0) {
+        size_t take = std::min((size_t)ctx->chunk_left, source->size());
+        if (take == 0) return false;
+        source->cutn(&chunk, take);
+        ctx->chunk_left -= (int64_t)take;
+        if (ctx->chunk_left == 0) {
+          IOBuf crlf;
+          if (source->size() >= 2) {
+            source->cutn(&crlf, 2);
+            ctx->chunk_left = -1;
+          } else {
+            // deliver this chunk now; CRLF on the next read
+            ctx->chunk_left = 0;
+            void* data = nullptr;
+            if (session_lock(ctx->cid, &data) == 0) {
+              Controller* cntl = (Controller*)data;
+              if (cntl->progressive_reader()) cntl->progressive_reader()(chunk, false);
+              session_unlock(ctx->cid);
+            }
+            return false;
+          }
+        }
+      } else if (!done && ctx->chunk_left == 0) {
+        // pending CRLF from a split chunk boundary
+        if (source->size() < 2) return false;
+        IOBuf crlf;
+        source->cutn(&crlf, 2);
+        ctx->chunk_left = -1;
+        continue;
+      }
+    }
+    // deliver
+    void* data = nullptr;
+    if (session_lock(ctx->cid, &data) == 0) {
+      Controller* cntl = (Controller*)data;
+      if (cntl->progressive_reader()) cntl->progressive_reader()(chunk, done);
+      if (done) {
+        sock->remove_pending_session(ctx->cid);
+        sock->pop_pipeline();
+        ctx->active = false;
+        EndRPC(cntl, ctx->cid);
+        return true;
+      }
+      session_unlock(ctx->cid);
+    } else if (done) {
+      // call timed out mid-body: drop the rest quietly
+      sock->pop_pipeline();
+      ctx->active = false;
+      return true;
+    }
+    if (done) return true;
+    if (!ctx->chunked && ctx->remaining == 0) return true;
+    if (source->empty()) return false;
+  }
+}
+
+ParseResult ParseHttpResponse(IOBuf* source, Socket* sock, bool eof) {
+  // Resume an in-flight progressive body first.
+  if (sock->protocol_ctx != nullptr && sock->protocol_ctx_owner == g_http_protocol_index) {
+    HttpProgressiveCtx* pctx = (HttpProgressiveCtx*)sock->protocol_ctx;
+    if (pctx->active) {
+      if (!progressive_pump(source, sock, pctx))
+        return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+      if (source->empty()) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+      // fall through: the next pipelined response is already buffered
+    }
+  }
   char probe[8];
   size_t n = std::min<size_t>(source->size(), 8);
   if (n < 5) return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
@@ -156,6 +272,51 @@ ParseResult ParseHttpResponse(IOBuf* source, Socket* /*sock*/, bool /*eof*/) {
     for (char& ch : k) ch = (char)tolower((unsigned char)ch);
     size_t v0 = l.find_first_not_of(' ', c + 1);
     headers[k] = v0 == std::string::npos ? "" : l.substr(v0);
+  }
+  // Progressive delivery: if the head-of-pipeline call asked for it,
+  // stream body bytes to its reader instead of buffering.
+  auto te_probe = headers.find("transfer-encoding");
+  const bool is_chunked =
+      te_probe != headers.end() && te_probe->second.find("chunked") != std::string::npos;
+  if (status == 200) {
+    uint64_t cid = sock->peek_pipeline();
+    void* data = nullptr;
+    if (cid != 0 && session_lock(cid, &data) == 0) {
+      Controller* cntl = (Controller*)data;
+      const bool want = (bool)cntl->progressive_reader();
+      session_unlock(cid);
+      if (want) {
+        HttpProgressiveCtx* pctx = nullptr;
+        if (sock->protocol_ctx != nullptr &&
+            sock->protocol_ctx_owner == g_http_protocol_index) {
+          pctx = (HttpProgressiveCtx*)sock->protocol_ctx;
+        } else if (sock->protocol_ctx == nullptr) {
+          pctx = new HttpProgressiveCtx;
+          sock->protocol_ctx = pctx;
+          sock->protocol_ctx_deleter = [](void* q) { delete (HttpProgressiveCtx*)q; };
+          sock->protocol_ctx_owner = g_http_protocol_index;
+        }
+        if (pctx != nullptr) {
+          source->pop_front(hend);  // headers consumed; body streams out
+          pctx->active = true;
+          pctx->cid = cid;
+          pctx->chunked = is_chunked;
+          pctx->chunk_left = -1;
+          pctx->remaining = 0;
+          if (!is_chunked) {
+            auto cl = headers.find("content-length");
+            if (cl != headers.end() &&
+                !parse_body_size(cl->second.c_str(), &pctx->remaining)) {
+              pctx->active = false;
+              return ParseResult::make_error(PARSE_ERROR_ABSOLUTELY_WRONG);
+            }
+          }
+          if (!progressive_pump(source, sock, pctx) || source->empty())
+            return ParseResult::make_error(PARSE_ERROR_NOT_ENOUGH_DATA);
+          return ParseHttpResponse(source, sock, eof);  // next pipelined response
+        }
+      }
+    }
   }
   IOBuf body;
   size_t consumed = hend;
@@ -419,7 +580,7 @@ void RegisterHttpProtocol() {
     p.support_server = true;
     p.support_client = true;
     p.name = "http";
-    RegisterProtocol(p);
+    g_http_protocol_index = RegisterProtocol(p);
   });
 }
 

@@ -253,6 +253,10 @@ class Socket {
     pipeline_q_.pop_front();
     return sid;
   }
+  uint64_t peek_pipeline() {
+    std::lock_guard<std::mutex> lk(pending_mu_);
+    return pipeline_q_.empty() ? 0 : pipeline_q_.front();
+  }
 
  private:
   std::deque<uint64_t> pipeline_q_;

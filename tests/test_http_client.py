@@ -90,3 +90,49 @@ def test_normal_content_length_still_works(port):
     assert b"200" in data.split(b"\r\n", 1)[0]
     assert data.endswith(body)
     s.close()
+
+
+def test_progressive_response_reading(port):
+    """Controller::response_read_progressively (≙ reference
+    ProgressiveReader / response_read_progressively): body bytes stream to
+    the reader as they arrive; the buffered response stays empty."""
+    big = bytes(range(256)) * 4096  # 1 MB
+    srv = b.Server()
+    srv.add_method("Big", "Blob", lambda req, att: (big, b""))
+    p = srv.start(0)
+    chunks, saw_done, resp = b.core.http_call_progressive(
+        "127.0.0.1:%d" % p, "/Big/Blob", b"x", 10000)
+    assert saw_done
+    assert resp == b""          # nothing buffered
+    assert b"".join(chunks) == big
+    srv.stop()
+
+
+def test_progressive_chunked_response():
+    """Progressive delivery of a CHUNKED body from a raw scripted server,
+    including a chunk split across TCP segments."""
+    import socket as pysock
+    import threading
+    srv = pysock.socket()
+    srv.bind(("127.0.0.1", 0))
+    srv.listen(1)
+
+    def run():
+        c, _ = srv.accept()
+        c.recv(4096)
+        c.sendall(b"HTTP/1.1 200 OK\r\nTransfer-Encoding: chunked\r\n\r\n")
+        c.sendall(b"5\r\nhello\r\n")
+        c.sendall(b"8\r\nwor")   # chunk split mid-payload
+        import time; time.sleep(0.05)
+        c.sendall(b"ld!!!\r\n")
+        c.sendall(b"0\r\n\r\n")
+        import time as t2; t2.sleep(0.2)
+        c.close()
+
+    threading.Thread(target=run, daemon=True).start()
+    sport = srv.getsockname()[1]
+    chunks, saw_done, resp = b.core.http_call_progressive(
+        "127.0.0.1:%d" % sport, "/stream/x", b"y", 8000)
+    assert saw_done
+    assert b"".join(chunks) == b"helloworld!!!"
+    srv.close()
