@@ -62,7 +62,13 @@ int gpu_fiber_wait_u64(const volatile unsigned long long* flag, uint64_t want,
   WaitSlot& s = g_slots[dev][kind];
   std::atomic<int>* b = slot_butex(s);
   bool parked = false;
+  // Hard deadline: if the ticket NEVER lands (faulted kernel, wedged
+  // stream), hand control back to the HIP lib's bounded-spin +
+  // hipStreamSynchronize fallback, which surfaces the device error
+  // instead of waiting forever.
+  const int64_t give_up_at = monotonic_time_us() + 10 * 1000000;
   while (*flag < want) {
+    if (monotonic_time_us() >= give_up_at) return -1;
     // Capture the butex value BEFORE the final flag check: if the wake
     // fires in between, butex_wait returns EWOULDBLOCK instead of parking
     // past the wake.
