@@ -14,6 +14,7 @@
 
 #include <atomic>
 #include <mutex>
+#include <vector>
 
 #include "gpu_api.h"
 #include "internal.h"
@@ -161,11 +162,15 @@ int run_spans(int nspans, NextFn next, int dev) {
       if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
       return -1;
     }
-    // Count ≤kSubSpan pieces first to pick the path.
+    // Materialize the spans ONCE (next() is stateful: it accumulates the
+    // running offset — calling it twice produced out-of-range dst
+    // pointers and a GPU memory fault), then count ≤kSubSpan pieces.
+    std::vector<SpanIn> spans;
+    spans.reserve((size_t)nspans);
     size_t npieces = 0;
     for (int i = 0; i < nspans; ++i) {
-      SpanIn s = next(i);
-      npieces += (s.len + kSubSpan - 1) / kSubSpan;
+      spans.push_back(next(i));
+      npieces += (spans.back().len + kSubSpan - 1) / kSubSpan;
     }
     if (npieces == 0) {
       if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
@@ -176,7 +181,7 @@ int run_spans(int nspans, NextFn next, int dev) {
       SpanBatch a;
       int batch = 0;
       for (int i = 0; i < nspans; ++i) {
-        SpanIn s = next(i);
+        const SpanIn& s = spans[(size_t)i];
         size_t off = 0;
         while (off < s.len) {
           size_t piece = s.len - off < kSubSpan ? s.len - off : kSubSpan;
@@ -220,7 +225,7 @@ int run_spans(int nspans, NextFn next, int dev) {
       if (rc == 0) {
         size_t w = 0;
         for (int i = 0; i < nspans; ++i) {
-          SpanIn s = next(i);
+          const SpanIn& s = spans[(size_t)i];
           size_t off = 0;
           while (off < s.len) {
             size_t piece = s.len - off < kSubSpan ? s.len - off : kSubSpan;
