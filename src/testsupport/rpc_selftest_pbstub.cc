@@ -99,6 +99,24 @@ bool pb_stub_test(std::string* err) {
       return false;
     }
   }
+  // mcpack converters on the generated structs (protoc-gen-mcpack role)
+  {
+    example::EchoRequest req;
+    req.message = "mc";
+    req.seq = -5;
+    req.tags = {7, 8};
+    std::string mc;
+    if (!req.SerializeAsMcpack(&mc)) {
+      *err = "mcpack serialize failed";
+      return false;
+    }
+    example::EchoRequest back;
+    if (!back.ParseFromMcpack(mc) || back.message != "mc" || back.seq != -5 ||
+        back.tags != std::vector<int32_t>({7, 8})) {
+      *err = "mcpack round trip mismatch";
+      return false;
+    }
+  }
   return true;
 }
 

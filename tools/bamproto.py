@@ -147,6 +147,7 @@ def generate(proto_path):
     o("#include <vector>")
     o("")
     o('#include "base/iobuf.h"')
+    o('#include "base/mcpack.h"')
     o('#include "base/pbgen.h"')
     o('#include "rpc/channel.h"')
     o('#include "rpc/closure.h"')
@@ -237,8 +238,134 @@ def emit_message(o, pool, msg, pkg):
     o("    out->append(_unknown);")
     o("  }")
     o("  std::string SerializeAsString() const { std::string s; SerializeToString(&s); return s; }")
+    o("")
+    emit_mcpack(o, pool, fields, pkg)
     o("};")
     o("")
+
+
+def mcpack_value_expr(f, expr):
+    """C++ expression building a bam::mcpack::Value from a scalar field."""
+    t = f["type"]
+    if t == T_BOOL:
+        return "bam::mcpack::Value::Bool(%s)" % expr
+    if t in (T_UINT32, T_UINT64, T_FIXED32, T_FIXED64):
+        return "bam::mcpack::Value::Uint((uint64_t)%s)" % expr
+    if t in (T_DOUBLE, T_FLOAT):
+        return "bam::mcpack::Value::Double((double)%s)" % expr
+    if t == T_STRING:
+        return "bam::mcpack::Value::Str(%s)" % expr
+    if t == T_BYTES:
+        return "bam::mcpack::Value::Bin(%s)" % expr
+    return "bam::mcpack::Value::Int((int64_t)%s)" % expr
+
+
+def mcpack_read_stmt(o, f, src, dst, indent):
+    """Emit assignment of mcpack Value `src` into scalar field `dst`."""
+    t = f["type"]
+    i = " " * indent
+    if t == T_BOOL:
+        o(f"{i}{dst} = {src}.type == bam::mcpack::Value::BOOL ? {src}.b : ({src}.i != 0);")
+    elif t in (T_UINT32, T_UINT64, T_FIXED32, T_FIXED64):
+        o(f"{i}{dst} = ({CPP_TYPE[t]})({src}.type == bam::mcpack::Value::UINT ? {src}.u : (uint64_t){src}.i);")
+    elif t in (T_DOUBLE, T_FLOAT):
+        o(f"{i}{dst} = ({CPP_TYPE[t]})({src}.type == bam::mcpack::Value::DOUBLE ? {src}.d : (double){src}.i);")
+    elif t in (T_STRING, T_BYTES):
+        o(f"{i}{dst} = {src}.str;")
+    else:
+        o(f"{i}{dst} = ({CPP_TYPE[t]})({src}.type == bam::mcpack::Value::UINT ? (int64_t){src}.u : {src}.i);")
+
+
+def emit_mcpack(o, pool, fields, pkg):
+    """mcpack (de)serializers on generated structs — the reference's
+    protoc-gen-mcpack role (mcpack2pb/generator.cpp)."""
+    o("  // ---- mcpack v2 (parity: protoc-gen-mcpack generated converters) ----")
+    o("  void ToMcpackValue(bam::mcpack::Value* out) const {")
+    o("    *out = bam::mcpack::Value::Object();")
+    for f in fields:
+        name = f["name"]
+        if f["is_map"]:
+            entry = pool.describe_message(f["type_name"])
+            vf = [x for x in entry if x["number"] == 2][0]
+            o("    { bam::mcpack::Value m = bam::mcpack::Value::Object();")
+            o("      for (const auto& kv : %s) {" % name)
+            if vf["type"] == T_MESSAGE:
+                o("        bam::mcpack::Value mv; kv.second.ToMcpackValue(&mv);")
+            else:
+                o("        bam::mcpack::Value mv = %s;" % mcpack_value_expr(vf, "kv.second"))
+            kf = [x for x in entry if x["number"] == 1][0]
+            if kf["type"] == T_STRING:
+                o("        m.obj[kv.first] = mv;")
+            else:
+                o("        m.obj[std::to_string(kv.first)] = mv;")
+            o("      }")
+            o("      out->obj[\"%s\"] = m; }" % name)
+        elif f["repeated"]:
+            o("    { bam::mcpack::Value a = bam::mcpack::Value::Array();")
+            o("      for (const auto& x : %s) {" % name)
+            if f["type"] == T_MESSAGE:
+                o("        bam::mcpack::Value e; x.ToMcpackValue(&e); a.arr.push_back(e);")
+            else:
+                o("        a.arr.push_back(%s);" % mcpack_value_expr(f, "x"))
+            o("      }")
+            o("      out->obj[\"%s\"] = a; }" % name)
+        elif f["type"] == T_MESSAGE:
+            o("    { bam::mcpack::Value m; %s.ToMcpackValue(&m); out->obj[\"%s\"] = m; }"
+              % (name, name))
+        else:
+            o("    out->obj[\"%s\"] = %s;" % (name, mcpack_value_expr(f, name)))
+    o("  }")
+    o("  bool FromMcpackValue(const bam::mcpack::Value& v) {")
+    o("    if (v.type != bam::mcpack::Value::OBJECT) return false;")
+    for f in fields:
+        name = f["name"]
+        o("    { auto it = v.obj.find(\"%s\");" % name)
+        o("      if (it != v.obj.end()) {")
+        if f["is_map"]:
+            entry = pool.describe_message(f["type_name"])
+            kf = [x for x in entry if x["number"] == 1][0]
+            vf = [x for x in entry if x["number"] == 2][0]
+            o("        for (const auto& kv : it->second.obj) {")
+            if kf["type"] == T_STRING:
+                key_expr = "kv.first"
+            else:
+                key_expr = "(%s)strtoll(kv.first.c_str(), nullptr, 10)" % CPP_TYPE[kf["type"]]
+            if vf["type"] == T_MESSAGE:
+                vt = cpp_name(vf["type_name"], pkg)
+                o("          %s mv; mv.FromMcpackValue(kv.second); %s[%s] = mv;"
+                  % (vt, name, key_expr))
+            else:
+                o("          %s mv{};" % CPP_TYPE[vf["type"]])
+                mcpack_read_stmt(o, vf, "kv.second", "mv", 10)
+                o("          %s[%s] = mv;" % (name, key_expr))
+            o("        }")
+        elif f["repeated"]:
+            o("        for (const auto& e : it->second.arr) {")
+            if f["type"] == T_MESSAGE:
+                o("          %s.emplace_back();" % name)
+                o("          if (!%s.back().FromMcpackValue(e)) return false;" % name)
+            else:
+                o("          %s x{};" % CPP_TYPE[f["type"]])
+                mcpack_read_stmt(o, f, "e", "x", 10)
+                o("          %s.push_back(x);" % name)
+            o("        }")
+        elif f["type"] == T_MESSAGE:
+            o("        if (!%s.FromMcpackValue(it->second)) return false;" % name)
+        else:
+            mcpack_read_stmt(o, f, "it->second", name, 8)
+        o("      } }")
+    o("    return true;")
+    o("  }")
+    o("  bool SerializeAsMcpack(std::string* out) const {")
+    o("    bam::mcpack::Value v;")
+    o("    ToMcpackValue(&v);")
+    o("    return bam::mcpack::Serialize(v, out);")
+    o("  }")
+    o("  bool ParseFromMcpack(const std::string& data) {")
+    o("    bam::mcpack::Value v;")
+    o("    if (!bam::mcpack::Parse(data.data(), data.size(), &v)) return false;")
+    o("    return FromMcpackValue(v);")
+    o("  }")
 
 
 def emit_field_parse(o, pool, f, pkg):
