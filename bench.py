@@ -180,7 +180,9 @@ def main():
         assert world_size >= 2, "--mode fanout needs >= 2 ranks"
         payload = bytes(range(256)) * 64  # 16 KB
         comp = b.core.snappy.compress(payload)
-        resp_cap = len(comp) + 256
+        # worst case of the PLAINTEXT: the GPU chunked compressor's output
+        # can exceed the host stream it echoes
+        resp_cap = len(payload) + len(payload) // 3 + 256
         rounds_per_step = max(1, args.calls_per_step // 100)
         if rank == 0:
             addrs = [""]

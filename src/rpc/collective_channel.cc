@@ -84,6 +84,9 @@ int echo_fn(const void* req, size_t len, void* resp, size_t resp_cap, size_t* re
 // The 16KB+snappy server work: decompress the broadcast payload, then
 // recompress the response — both on the GPU for device groups
 // (hip/snappy.hip kernels), host codec otherwise.
+// Stage-coded errors so GPU failures are diagnosable from test output.
+thread_local std::string g_snappy_echo_err;
+
 int snappy_echo_fn(const void* req, size_t len, void* resp, size_t resp_cap,
                    size_t* resp_len, int dev) {
   if (dev >= 0) {
@@ -104,8 +107,12 @@ int snappy_echo_fn(const void* req, size_t len, void* resp, size_t resp_cap,
     void* pp = plain.ensure(true, plain_cap);
     if (pp == nullptr) return ENOMEM;
     size_t plain_len = 0;
-    if (api->snappy_decompress(req, len, pp, plain_cap, &plain_len, dev) != 0)
+    if (api->snappy_decompress(req, len, pp, plain_cap, &plain_len, dev) != 0) {
+      g_snappy_echo_err = std::string("gpu decompress failed (len=") +
+                          std::to_string(len) + ", cap=" + std::to_string(plain_cap) +
+                          "): " + api->last_error();
       return EINVAL;
+    }
     // The device compressor needs worst-case output room (like host
     // MaxCompressedLength); compress into scratch, then move the actual
     // bytes into the (tightly sized) response slot.
@@ -114,9 +121,17 @@ int snappy_echo_fn(const void* req, size_t len, void* resp, size_t resp_cap,
     void* cp = comp.ensure(true, comp_cap);
     if (cp == nullptr) return ENOMEM;
     size_t comp_len = 0;
-    if (api->snappy_compress(pp, plain_len, cp, comp_cap, &comp_len, dev) != 0)
+    if (api->snappy_compress(pp, plain_len, cp, comp_cap, &comp_len, dev) != 0) {
+      g_snappy_echo_err = std::string("gpu compress failed (plain=") +
+                          std::to_string(plain_len) + ", cap=" + std::to_string(comp_cap) +
+                          "): " + api->last_error();
       return EINVAL;
-    if (comp_len > resp_cap) return EINVAL;
+    }
+    if (comp_len > resp_cap) {
+      g_snappy_echo_err = "compressed " + std::to_string(comp_len) + " > resp_cap " +
+                          std::to_string(resp_cap);
+      return EINVAL;
+    }
     api->memcpy_res(resp, 2, 0, cp, 2, 0, comp_len);
     *resp_len = comp_len;
     return 0;
@@ -177,6 +192,10 @@ int run_round(CommGroup* group, const std::string& method, const void* req_ext,
   int rc = fn(req, req_len, (char*)myslot + 8, resp_cap, &resp_len, dev);
   if (rc != 0) {
     *err = "method " + method + " failed rc=" + std::to_string(rc);
+    if (!g_snappy_echo_err.empty()) {
+      *err += ": " + g_snappy_echo_err;
+      g_snappy_echo_err.clear();
+    }
     return rc;
   }
   uint64_t hdr = resp_len;
