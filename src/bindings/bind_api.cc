@@ -8,6 +8,7 @@
 
 #include <memory>
 
+#include "rpc/couchbase.h"
 #include "bindings/bind.h"
 #include "rpc/authenticator.h"
 #include "rpc/ssl_util.h"
@@ -237,6 +238,45 @@ void bind_api(py::module_& m) {
         },
         py::arg("addr"), py::arg("method"), py::arg("request") = py::bytes(""),
         py::arg("timeout_ms") = 5000);
+
+  py::class_<bam::CouchbaseClient>(m, "CouchbaseClient")
+      .def(py::init<>())
+      .def("init",
+           [](bam::CouchbaseClient& c, const std::string& addr, const std::string& bucket,
+              const std::string& user, const std::string& password) {
+             py::gil_scoped_release rel;
+             return c.Init(addr, bucket, user, password);
+           },
+           py::arg("config_addr"), py::arg("bucket"), py::arg("user") = "",
+           py::arg("password") = "")
+      .def("set",
+           [](bam::CouchbaseClient& c, const std::string& k, py::bytes v) {
+             std::string val(v);
+             py::gil_scoped_release rel;
+             return c.Set(k, val);
+           })
+      .def("get",
+           [](bam::CouchbaseClient& c, const std::string& k) -> py::object {
+             std::string v;
+             int rc;
+             {
+               py::gil_scoped_release rel;
+               rc = c.Get(k, &v);
+             }
+             if (rc != 0) return py::none();
+             return py::bytes(v);
+           })
+      .def("delete_key",
+           [](bam::CouchbaseClient& c, const std::string& k) {
+             py::gil_scoped_release rel;
+             return c.Delete(k);
+           })
+      .def("nvbuckets", &bam::CouchbaseClient::nvbuckets)
+      .def("nservers", &bam::CouchbaseClient::nservers)
+      .def("last_error", &bam::CouchbaseClient::last_error)
+      .def_static("vbucket_of", [](py::bytes key, size_t nvb) {
+        return bam::CouchbaseClient::VBucketOf(std::string(key), nvb);
+      });
 
   m.def("gen_self_signed_cert", [](const std::string& cn) {
     std::string cert, key;

@@ -18,7 +18,8 @@ struct MemcacheResponse {
 };
 
 void PackMemcacheRequest(IOBuf* out, uint8_t opcode, const std::string& key,
-                         const std::string& value, const std::string& extras, uint64_t cas);
+                         const std::string& value, const std::string& extras, uint64_t cas,
+                         uint16_t vbucket = 0);
 
 class MemcacheClient {
  public:
@@ -34,7 +35,7 @@ class MemcacheClient {
   int Delete(const std::string& key);
   int Version(std::string* version);
   int RawCall(uint8_t opcode, const std::string& key, const std::string& value,
-              const std::string& extras, MemcacheResponse* out);
+              const std::string& extras, MemcacheResponse* out, uint16_t vbucket = 0);
 
   // Couchbase parity (reference policy/couchbase_authenticator.cpp +
   // couchbase_protocol.cpp): SASL PLAIN authentication over the memcache

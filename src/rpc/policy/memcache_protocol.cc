@@ -43,13 +43,15 @@ uint64_t get_u64_be(const char* p) {
 }  // namespace
 
 void PackMemcacheRequest(IOBuf* out, uint8_t opcode, const std::string& key,
-                         const std::string& value, const std::string& extras, uint64_t cas) {
+                         const std::string& value, const std::string& extras, uint64_t cas,
+                         uint16_t vbucket) {
   char h[kHeaderLen];
   memset(h, 0, sizeof(h));
   h[0] = (char)0x80;
   h[1] = (char)opcode;
   put_u16_be(h + 2, (uint16_t)key.size());
   h[4] = (char)extras.size();
+  put_u16_be(h + 6, vbucket);  // request header: vbucket id (couchbase)
   uint32_t body = (uint32_t)(extras.size() + key.size() + value.size());
   wire::put_u32_be(h + 8, body);
   put_u64_be(h + 16, cas);
@@ -175,11 +177,12 @@ int MemcacheClient::SaslAuthPlain(const std::string& user, const std::string& pa
 }
 
 int MemcacheClient::RawCall(uint8_t opcode, const std::string& key, const std::string& value,
-                            const std::string& extras, MemcacheResponse* out) {
+                            const std::string& extras, MemcacheResponse* out,
+                            uint16_t vbucket) {
   if (!init_ok_) return -1;
   Controller cntl;
   IOBuf request, response;
-  PackMemcacheRequest(&request, opcode, key, value, extras, 0);
+  PackMemcacheRequest(&request, opcode, key, value, extras, 0, vbucket);
   channel_.CallMethod("memcache.op", &cntl, &request, &response, nullptr);
   if (cntl.Failed()) return cntl.ErrorCode();
   std::string raw = response.to_string();

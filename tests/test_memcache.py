@@ -111,3 +111,127 @@ def test_couchbase_sasl_auth(mc_port):
     assert c.sasl_auth_plain("cb_user", "cb_pass") == 0
     rc = c.sasl_auth_plain("cb_user", "wrong")
     assert rc == 10000 + 0x20
+
+
+def start_vbucket_memcached(expected_owner, owner_id, redirect):
+    """memcached-binary mock that CHECKS the request's vbucket id: serves
+    keys whose vbucket it owns (per `expected_owner` map), answers
+    NOT_MY_VBUCKET (0x0007) when `redirect[0]` is set or it is not the
+    owner. Returns (port, store, seen_vbuckets)."""
+    store = {}
+    seen = []
+    lsock = socket.socket()
+    lsock.bind(("127.0.0.1", 0))
+    lsock.listen(8)
+    port = lsock.getsockname()[1]
+
+    def handle(conn):
+        try:
+            buf = b""
+            while True:
+                while len(buf) < 24:
+                    chunk = conn.recv(65536)
+                    if not chunk:
+                        return
+                    buf += chunk
+                magic, op, klen = struct.unpack(">BBH", buf[:4])
+                xlen = buf[4]
+                vbucket = struct.unpack(">H", buf[6:8])[0]
+                blen = struct.unpack(">I", buf[8:12])[0]
+                while len(buf) < 24 + blen:
+                    chunk = conn.recv(65536)
+                    if not chunk:
+                        return
+                    buf += chunk
+                body = buf[24:24 + blen]
+                buf = buf[24 + blen:]
+                key = body[xlen:xlen + klen].decode()
+                val = body[xlen + klen:]
+                seen.append((op, key, vbucket))
+                status = 0
+                out_val = b""
+                if redirect[0] or expected_owner.get(vbucket) != owner_id:
+                    status = 0x0007  # NOT_MY_VBUCKET
+                elif op == 0x01:
+                    store[key] = val
+                elif op == 0x00:
+                    if key in store:
+                        out_val = store[key]
+                    else:
+                        status = 1
+                elif op == 0x04:
+                    store.pop(key, None)
+                extras = b"\x00" * 4 if op == 0x00 and status == 0 else b""
+                resp = struct.pack(">BBHBBHIIQ", 0x81, op, 0, len(extras), 0,
+                                   status, len(extras) + len(out_val), 0, 0)
+                conn.sendall(resp + extras + out_val)
+        except OSError:
+            pass
+
+    def acceptor():
+        while True:
+            try:
+                conn, _ = lsock.accept()
+            except OSError:
+                return
+            threading.Thread(target=handle, args=(conn,), daemon=True).start()
+
+    threading.Thread(target=acceptor, daemon=True).start()
+    return port, store, seen
+
+
+def test_couchbase_vbucket_routing():
+    """CouchbaseClient (rpc/couchbase.*): REST vBucketServerMap config,
+    CRC32-based key->vbucket mapping, per-request vbucket ids, and
+    NOT_MY_VBUCKET -> config refetch + retry (rebalance)."""
+    NVB = 64
+    # phase 1: node A owns even vbuckets, node B odd.
+    ownerA = {vb: 0 for vb in range(0, NVB, 2)}
+    ownerA.update({vb: 1 for vb in range(1, NVB, 2)})
+    redirA = [False]
+    redirB = [False]
+    ownerB = dict(ownerA)
+    portA, storeA, seenA = start_vbucket_memcached(ownerA, 0, redirA)
+    portB, storeB, seenB = start_vbucket_memcached(ownerB, 1, redirB)
+
+    phase = [1]
+    cfg_srv = b.Server()
+
+    def config(req, att):
+        import json as pyjson
+        if phase[0] == 1:
+            vmap = [[vb % 2] for vb in range(NVB)]
+        else:
+            vmap = [[1] for _ in range(NVB)]  # everything moved to B
+        body = pyjson.dumps({"vBucketServerMap": {
+            "serverList": ["127.0.0.1:%d" % portA, "127.0.0.1:%d" % portB],
+            "vBucketMap": vmap,
+        }}).encode()
+        return body, b""
+
+    cfg_srv.add_method("pools", "cfg", config)
+    cfg_srv.add_restful_mapping("pools", "/pools/default/b/app => cfg")
+    cfg_port = cfg_srv.start(0)
+
+    c = b.core.CouchbaseClient()
+    assert c.init("127.0.0.1:%d" % cfg_port, "app") == 0, c.last_error()
+    assert c.nvbuckets() == NVB and c.nservers() == 2
+
+    keys = ["user:%d" % i for i in range(40)]
+    for k in keys:
+        assert c.set(k, ("v-" + k).encode()) == 0
+    for k in keys:
+        assert c.get(k) == ("v-" + k).encode()
+    # requests carried the right vbucket ids and split across both nodes
+    for op, key, vb in seenA + seenB:
+        assert vb == b.core.CouchbaseClient.vbucket_of(key.encode(), NVB)
+    assert storeA and storeB
+
+    # phase 2: rebalance — node A starts refusing; map moves all to B.
+    phase[0] = 2
+    redirA[0] = True
+    ownerB.update({vb: 1 for vb in range(NVB)})  # B owns everything now
+    k = next(k for k in keys if k in storeA)
+    storeB[k] = b"moved"  # B owns it after rebalance
+    assert c.get(k) == b"moved"  # NOT_MY_VBUCKET -> refetch -> retry on B
+    cfg_srv.stop()
