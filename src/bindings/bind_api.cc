@@ -8,6 +8,7 @@
 
 #include <memory>
 
+#include "rpc/redis_cluster.h"
 #include "rpc/couchbase.h"
 #include "bindings/bind.h"
 #include "rpc/authenticator.h"
@@ -320,7 +321,12 @@ bam::RedisReply py_to_reply(py::handle obj) {
   if (py::isinstance<py::bool_>(obj)) return RedisReply::Integer(obj.cast<bool>() ? 1 : 0);
   if (py::isinstance<py::int_>(obj)) return RedisReply::Integer(obj.cast<int64_t>());
   if (py::isinstance<py::bytes>(obj)) return RedisReply::Bulk(obj.cast<std::string>());
-  if (py::isinstance<py::str>(obj)) return RedisReply::Status(obj.cast<std::string>());
+  if (py::isinstance<py::str>(obj)) {
+    std::string v = obj.cast<std::string>();
+    // redis convention: a leading '-' marks an error reply ("-ERR ...")
+    if (!v.empty() && v[0] == '-') return RedisReply::Error(v.substr(1));
+    return RedisReply::Status(v);
+  }
   if (py::isinstance<py::list>(obj) || py::isinstance<py::tuple>(obj)) {
     RedisReply arr;
     arr.type = RedisReply::ARRAY;
@@ -378,6 +384,30 @@ void bind_redis(py::module_& m) {
       .def(py::init<>())
       .def("add_handler", &PyRedisServer::add_handler)
       .def("start", &PyRedisServer::start, py::arg("port") = 0);
+
+  py::class_<bam::RedisClusterClient>(m, "RedisClusterClient")
+      .def(py::init<>())
+      .def("init",
+           [](bam::RedisClusterClient& c, const std::string& seed, int timeout_ms) {
+             py::gil_scoped_release rel;
+             return c.Init(seed, timeout_ms);
+           },
+           py::arg("seed"), py::arg("timeout_ms") = 1000)
+      .def("command",
+           [](bam::RedisClusterClient& c, const std::vector<std::string>& args) {
+             bam::RedisReply r;
+             int rc;
+             {
+               py::gil_scoped_release rel;
+               rc = c.Command(args, &r);
+             }
+             if (rc != 0) throw std::runtime_error("cluster command: " + c.last_error());
+             return reply_to_py(r);
+           })
+      .def("nslots_mapped", &bam::RedisClusterClient::nslots_mapped)
+      .def_static("slot_of", [](py::bytes key) {
+        return bam::RedisClusterClient::SlotOf(std::string(key));
+      });
 
   m.def("redis_call", [](const std::string& addr, const std::vector<std::string>& args,
                          int timeout_ms) {

@@ -106,3 +106,84 @@ def test_pipelined_commands(redis_server):
         data += s.recv(4096)
     s.close()
     assert data == b"+OK\r\n" * 5
+
+
+def test_redis_cluster_slot_of():
+    """CRC16-CCITT slot mapping incl. hash tags (redis cluster spec
+    values: 'foo'->12182, '123456789'->12739, '{user1000}.following' ==
+    '{user1000}.followers')."""
+    C = b.core.RedisClusterClient
+    assert C.slot_of(b"foo") == 12182
+    assert C.slot_of(b"123456789") == 12739
+    assert C.slot_of(b"{user1000}.following") == C.slot_of(b"{user1000}.followers")
+    assert C.slot_of(b"{user1000}.following") == C.slot_of(b"user1000")
+
+
+def test_redis_cluster_routing_and_moved():
+    """Slot routing from CLUSTER SLOTS + -MOVED redirect with remap +
+    -ASK one-shot redirect, against two scripted RESP nodes."""
+    storeA, storeB = {}, {}
+    moved_mode = [False]
+
+    def make_node(store, my_half, ports):
+        srv = b.RedisServer()
+
+        def cluster(args):
+            lo = args[1].decode().lower() if len(args) > 1 else ""
+            if lo != "slots":
+                raise RuntimeError("unsupported")
+            if not moved_mode[0]:
+                return [[0, 8191, ["127.0.0.1", ports[0]]],
+                        [8192, 16383, ["127.0.0.1", ports[1]]]]
+            return [[0, 16383, ["127.0.0.1", ports[1]]]]
+
+        def set_(args):
+            k = args[1].decode()
+            slot = b.core.RedisClusterClient.slot_of(args[1])
+            owned = (slot < 8192) == (my_half == 0)
+            if moved_mode[0] and my_half == 0:
+                return "-MOVED %d 127.0.0.1:%d" % (slot, ports[1])
+            if not owned and not moved_mode[0]:
+                return "-MOVED %d 127.0.0.1:%d" % (slot, ports[1 - my_half])
+            store[k] = args[2]
+            return "OK"
+
+        def get(args):
+            k = args[1].decode()
+            slot = b.core.RedisClusterClient.slot_of(args[1])
+            if moved_mode[0] and my_half == 0:
+                return "-MOVED %d 127.0.0.1:%d" % (slot, ports[1])
+            return store.get(k)
+
+        def asking(args):
+            return "OK"
+
+        srv.add_handler("cluster", cluster)
+        srv.add_handler("set", set_)
+        srv.add_handler("get", get)
+        srv.add_handler("asking", asking)
+        return srv
+
+    ports = [0, 0]
+    srvA = make_node(storeA, 0, ports)
+    srvB = make_node(storeB, 1, ports)
+    ports[0] = srvA.start(0)
+    ports[1] = srvB.start(0)
+
+    c = b.core.RedisClusterClient()
+    assert c.init("127.0.0.1:%d" % ports[0]) == 0
+    assert c.nslots_mapped() == 16384
+
+    # keys landing on both halves route correctly
+    keys = [b"foo", b"bar", b"{user1000}.x", b"k%d" % 7]
+    for k in keys:
+        assert c.command(["SET", k.decode(), "v-" + k.decode()]) == "OK"
+        assert c.command(["GET", k.decode()]) == b"v-" + k
+    assert storeA and storeB
+
+    # rebalance: node A answers MOVED for everything; client follows and
+    # remaps (subsequent calls go straight to B)
+    moved_mode[0] = True
+    kA = next(iter(storeA))
+    storeB[kA] = b"after-move"
+    assert c.command(["GET", kA]) == b"after-move"
