@@ -44,7 +44,8 @@ int main() {
 }
 """)
     exe = tmp_path / "asan_fibers"
-    srcs = [f"src/fiber/{f}" for f in os.listdir(f"{REPO}/src/fiber")
+    srcs = ["src/testsupport/fiber_selftest.cc"] + \
+           [f"src/fiber/{f}" for f in os.listdir(f"{REPO}/src/fiber")
             if f.endswith((".cc", ".S"))]
     cmd = ["g++", "-O1", "-g", "-std=c++17", "-fsanitize=address",
            "-fno-omit-frame-pointer", "-pthread", f"-I{REPO}/src", str(main),
