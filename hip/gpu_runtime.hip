@@ -33,17 +33,20 @@ std::atomic<uint64_t> g_stat_uploads{0};
 std::atomic<uint64_t> g_stat_upload_bytes{0};
 std::atomic<uint64_t> g_stat_hbm_blocks{0};
 std::atomic<uint64_t> g_stat_hbm_bytes{0};
+std::atomic<uint64_t> g_stat_pstage_gathers{0};
 }  // namespace
 
 extern "C" const char* bam_gpu_stats_text(void) {
   static char buf[512];
   snprintf(buf, sizeof(buf),
            "gather_batches: %llu\ngather_bytes: %llu\ngather_fallbacks: %llu\n"
+           "pstage_gathers: %llu\n"
            "uploads_async: %llu\nupload_bytes: %llu\n"
            "hbm_blocks_live: %llu\nhbm_bytes_live: %llu\n",
            (unsigned long long)g_stat_gathers.load(),
            (unsigned long long)g_stat_gather_bytes.load(),
            (unsigned long long)g_stat_gather_fallbacks.load(),
+           (unsigned long long)g_stat_pstage_gathers.load(),
            (unsigned long long)g_stat_uploads.load(),
            (unsigned long long)g_stat_upload_bytes.load(),
            (unsigned long long)g_stat_hbm_blocks.load(),
@@ -702,6 +705,9 @@ extern "C" int bam_gpu_upload_async(void* dst_dev, const void* src, size_t n, in
   return rc;
 }
 
+extern "C" int bam_gpu_pstage_gather(void* host_dst, const void* const* srcs,
+                                     const size_t* lens, int nspans, size_t total, int dev);
+
 extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
                                       const size_t* lens, int nspans, int dev) {
   size_t total = 0;
@@ -710,6 +716,19 @@ extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
   g_stat_gathers.fetch_add(1, std::memory_order_relaxed);
   g_stat_gather_bytes.fetch_add(total, std::memory_order_relaxed);
   ScopedDevice sd(dev);
+  // Small batches: the persistent staging kernel (hip/pstage.hip) skips
+  // the per-batch launch entirely. It bypasses stream ordering, so drain
+  // pending async uploads first (same requirement as the direct path).
+  if (nspans <= 8 && total <= (32u << 10)) {
+    DirectState& st = g_direct[dev >= 0 && dev < kMaxDev ? dev : 0];
+    unsigned long long want = st.up_ticket.load(std::memory_order_acquire);
+    if (want == 0 || *st.up_flag >= want || upload_wait(st, want, dev)) {
+      if (bam_gpu_pstage_gather(host_dst, srcs, lens, nspans, total, dev) == 0) {
+        g_stat_pstage_gathers.fetch_add(1, std::memory_order_relaxed);
+        return 0;
+      }
+    }
+  }
   if (total <= direct_max_bytes() && nspans <= kDirectMaxSpans &&
       gather_direct(host_dst, srcs, lens, nspans, dev) == 0) {
     return 0;
