@@ -9,8 +9,10 @@ On GPUs (one process per device) the payload broadcasts over xGMI via
 RCCL and responses return in one all-gather; with backend "tcp" the same
 machinery runs on host buffers (works anywhere).
 """
+import os
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import brpc_amd as b
 
 rank, nranks, base_port = int(sys.argv[1]), int(sys.argv[2]), int(sys.argv[3])

@@ -8,8 +8,10 @@ otherwise) while the stream's socket carries descriptors + credit.
   python examples/stream_xgmi.py 0 36000 &
   python examples/stream_xgmi.py 1 36000
 """
+import os
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import brpc_amd as b
 
 rank, base_port = int(sys.argv[1]), int(sys.argv[2])

@@ -719,7 +719,7 @@ extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
   // Small batches: the persistent staging kernel (hip/pstage.hip) skips
   // the per-batch launch entirely. It bypasses stream ordering, so drain
   // pending async uploads first (same requirement as the direct path).
-  if (nspans <= 8 && total <= (32u << 10)) {
+  if (nspans <= 8 && total <= (4u << 10)) {
     DirectState& st = g_direct[dev >= 0 && dev < kMaxDev ? dev : 0];
     unsigned long long want = st.up_ticket.load(std::memory_order_acquire);
     if (want == 0 || *st.up_flag >= want || upload_wait(st, want, dev)) {
