@@ -34,19 +34,21 @@ std::atomic<uint64_t> g_stat_upload_bytes{0};
 std::atomic<uint64_t> g_stat_hbm_blocks{0};
 std::atomic<uint64_t> g_stat_hbm_bytes{0};
 std::atomic<uint64_t> g_stat_pstage_gathers{0};
+std::atomic<uint64_t> g_stat_pstage_uploads{0};
 }  // namespace
 
 extern "C" const char* bam_gpu_stats_text(void) {
   static char buf[512];
   snprintf(buf, sizeof(buf),
            "gather_batches: %llu\ngather_bytes: %llu\ngather_fallbacks: %llu\n"
-           "pstage_gathers: %llu\n"
+           "pstage_gathers: %llu\npstage_uploads: %llu\n"
            "uploads_async: %llu\nupload_bytes: %llu\n"
            "hbm_blocks_live: %llu\nhbm_bytes_live: %llu\n",
            (unsigned long long)g_stat_gathers.load(),
            (unsigned long long)g_stat_gather_bytes.load(),
            (unsigned long long)g_stat_gather_fallbacks.load(),
            (unsigned long long)g_stat_pstage_gathers.load(),
+           (unsigned long long)g_stat_pstage_uploads.load(),
            (unsigned long long)g_stat_uploads.load(),
            (unsigned long long)g_stat_upload_bytes.load(),
            (unsigned long long)g_stat_hbm_blocks.load(),
@@ -695,8 +697,39 @@ extern "C" void bam_gpu_quiesce(int dev) {
   }
 }
 
+extern "C" int bam_gpu_pstage_copy(void* dst, const void* src, size_t n, int dev);
+
+namespace {
+// Per-thread pinned bounce for small pstage uploads (allocated once).
+char* tls_upload_bounce(int dev) {
+  static thread_local char* bounce = nullptr;
+  if (bounce == nullptr) {
+    void* p = nullptr;
+    if (hipHostMalloc(&p, 4096, hipHostMallocDefault) != hipSuccess) return nullptr;
+    bounce = (char*)p;
+  }
+  (void)dev;
+  return bounce;
+}
+}  // namespace
+
 extern "C" int bam_gpu_upload_async(void* dst_dev, const void* src, size_t n, int dev) {
   ScopedDevice sd(dev);
+  // Small uploads: persistent-kernel copy, COMPLETE on return (no launch,
+  // and no ordering protocol needed for these bytes). TLS bounce keeps
+  // the path lock-free.
+  if (n <= 4096) {
+    char* bounce = tls_upload_bounce(dev);
+    if (bounce != nullptr) {
+      ::memcpy(bounce, src, n);
+      if (bam_gpu_pstage_copy(dst_dev, bounce, n, dev) == 0) {
+        g_stat_uploads.fetch_add(1, std::memory_order_relaxed);
+        g_stat_upload_bytes.fetch_add(n, std::memory_order_relaxed);
+        g_stat_pstage_uploads.fetch_add(1, std::memory_order_relaxed);
+        return 0;
+      }
+    }
+  }
   int rc = upload_direct(dst_dev, src, n, dev);
   if (rc == 0) {
     g_stat_uploads.fetch_add(1, std::memory_order_relaxed);

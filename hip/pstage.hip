@@ -161,15 +161,13 @@ bool pstage_launch(PState& st, int dev) {
 // Returns 0 on success (bytes staged into host_dst), nonzero = caller
 // falls back to the launch-based path. Requires: nspans<=8, total<=32K,
 // host_dst pinned, spans device-resident.
+extern "C" int bam_gpu_pstage_gather_nocheck(void* host_dst, const void* const* srcs,
+                                             const size_t* lens, int nspans, size_t total,
+                                             int dev);
+
 extern "C" int bam_gpu_pstage_gather(void* host_dst, const void* const* srcs,
                                      const size_t* lens, int nspans, size_t total,
                                      int dev) {
-  if (nspans <= 0 || nspans > kPSpans || total > kPMaxBytes) return 1;
-  static const bool disabled = [] {
-    const char* e = getenv("BAM_PSTAGE");
-    return e != nullptr && e[0] == '0';
-  }();
-  if (disabled) return 1;
   // host_dst must be pinned (device-visible VA); the IOBuf bounce base is
   // thread-stable, so cache the attribute lookup.
   {
@@ -186,6 +184,18 @@ extern "C" int bam_gpu_pstage_gather(void* host_dst, const void* const* srcs,
     }
     if (!tls_chk.ok) return 1;
   }
+  return bam_gpu_pstage_gather_nocheck(host_dst, srcs, lens, nspans, total, dev);
+}
+
+extern "C" int bam_gpu_pstage_gather_nocheck(void* host_dst, const void* const* srcs,
+                                             const size_t* lens, int nspans, size_t total,
+                                             int dev) {
+  if (nspans <= 0 || nspans > kPSpans || total > kPMaxBytes) return 1;
+  static const bool disabled = [] {
+    const char* e = getenv("BAM_PSTAGE");
+    return e != nullptr && e[0] == '0';
+  }();
+  if (disabled) return 1;
   PState& st = g_pstate[dev >= 0 && dev < kMaxDev ? dev : 0];
   if (!st.available.load(std::memory_order_acquire)) {
     std::lock_guard<std::mutex> lk(st.mu);
@@ -267,6 +277,18 @@ extern "C" int bam_gpu_pstage_gather(void* host_dst, const void* const* srcs,
   }
   __atomic_store_n(&s.state, 0u, __ATOMIC_RELEASE);
   return 0;
+}
+
+// Synchronous single-span copy through the persistent kernel (either
+// direction; both pointers must be device-dereferenceable — pinned host
+// or device memory). Used for small H2D uploads: completing before
+// return removes the whole async-ordering protocol for these bytes.
+extern "C" int bam_gpu_pstage_copy(void* dst, const void* src, size_t n, int dev) {
+  const void* srcs[1] = {src};
+  size_t lens[1] = {n};
+  // Reuse the gather entry but skip its pinned-dst check by calling the
+  // internal machinery: state/claim/wait logic is identical.
+  return bam_gpu_pstage_gather_nocheck(dst, srcs, lens, 1, n, dev);
 }
 
 extern "C" void bam_gpu_pstage_quit(int dev) {
