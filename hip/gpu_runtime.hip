@@ -715,10 +715,13 @@ char* tls_upload_bounce(int dev) {
 
 extern "C" int bam_gpu_upload_async(void* dst_dev, const void* src, size_t n, int dev) {
   ScopedDevice sd(dev);
-  // Small uploads: persistent-kernel copy, COMPLETE on return (no launch,
-  // and no ordering protocol needed for these bytes). TLS bounce keeps
-  // the path lock-free.
-  if (n <= 4096) {
+  // Synchronous persistent-kernel uploads LOST the A/B to the pipelined
+  // stream ring (the wait serializes handler fibers): opt-in only.
+  static const bool pstage_upload = [] {
+    const char* e = getenv("BAM_PSTAGE_UPLOAD");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (pstage_upload && n <= 4096) {
     char* bounce = tls_upload_bounce(dev);
     if (bounce != nullptr) {
       ::memcpy(bounce, src, n);
