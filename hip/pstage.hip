@@ -3,7 +3,7 @@
 // staging batch, which caps the HBM path at roughly half the host path.
 // This persistent kernel removes the launch: one resident workgroup per
 // GPU polls a pinned ring of gather descriptors (device->pinned copies,
-// ≤8 spans / ≤32 KiB — exactly the small-response shape) and publishes
+// ≤8 spans — the small-response shape) and publishes
 // completion straight into pinned flags. Submission is a few host stores;
 // completion arrives in ~1-2 µs with no runtime API call on the hot path.
 //
@@ -189,11 +189,14 @@ extern "C" int bam_gpu_pstage_gather_nocheck(void* host_dst, const void* const* 
                                              const size_t* lens, int nspans, size_t total,
                                              int dev) {
   if (nspans <= 0 || nspans > kPSpans || total > kPMaxBytes) return 1;
-  static const bool disabled = [] {
+  // Default OFF: the measured A/B (profiles/INDEX.md round 2) showed the
+  // resident poller's PCIe read pressure costs more under mixed staging
+  // load than the saved launches gain. BAM_PSTAGE=1 opts in.
+  static const bool enabled = [] {
     const char* e = getenv("BAM_PSTAGE");
-    return e != nullptr && e[0] == '0';
+    return e != nullptr && e[0] == '1';
   }();
-  if (disabled) return 1;
+  if (!enabled) return 1;
   PState& st = g_pstate[dev >= 0 && dev < kMaxDev ? dev : 0];
   if (!st.available.load(std::memory_order_acquire)) {
     std::lock_guard<std::mutex> lk(st.mu);
