@@ -262,7 +262,12 @@ bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse
   } else if (p == "/brpc_metrics" || p == "/metrics") {
     page_metrics(resp);
   } else if (p == "/rpcz") {
-    rpcz::DumpRecentSpans(&resp->body, req.query.count("verbose") != 0);
+    auto dbit = req.query.find("db");
+    if (dbit != req.query.end()) {
+      rpcz::DumpPersistedSpans(&resp->body, atoi(dbit->second.c_str()));
+    } else {
+      rpcz::DumpRecentSpans(&resp->body, req.query.count("verbose") != 0);
+    }
   } else {
     return false;
   }

@@ -6,6 +6,7 @@
 #include <vector>
 
 #include "base/flags.h"
+#include "base/recordio.h"
 #include "base/time.h"
 
 namespace bam {
@@ -15,10 +16,51 @@ BAM_DEFINE_int64(rpcz_sample_mod, 16,
                  "record 1 of every N spans per thread (1 = all; parity: the "
                  "reference samples spans through a budgeted bvar collector)");
 BAM_DEFINE_int64(rpcz_max_spans, 2048, "max spans kept in the rpcz ring");
+BAM_DEFINE_string(rpcz_db_path, "",
+                  "when set, sampled spans also persist to this recordio "
+                  "file (parity: the reference's leveldb-backed SpanDB; "
+                  "queried back via /rpcz?db=N)");
 
 namespace rpcz {
 
 namespace {
+
+// ---- persistent span store (recordio-backed SpanDB) ----
+// Spans serialize as one line per record: tab-separated fields. Writes
+// are buffered under a mutex and flushed opportunistically.
+struct SpanDb {
+  std::mutex mu;
+  RecordWriter* writer = nullptr;
+  std::string path;
+  int64_t written = 0;
+
+  void maybe_open() {
+    if (writer != nullptr && path == FLAG_rpcz_db_path) return;
+    delete writer;
+    writer = nullptr;
+    path = FLAG_rpcz_db_path;
+    if (!path.empty()) {
+      writer = new RecordWriter(path);
+      if (!writer->ok()) {
+        delete writer;
+        writer = nullptr;
+      }
+    }
+  }
+};
+SpanDb& span_db() {
+  static SpanDb* db = new SpanDb;
+  return *db;
+}
+
+std::string span_to_record(const Span& s) {
+  std::ostringstream os;
+  os << s.start_us << '\t' << s.end_us << '\t' << (s.server_side ? 'S' : 'C') << '\t'
+     << s.full_method << '\t' << endpoint2str(s.remote) << '\t' << s.error_code << '\t'
+     << s.log_id << '\t' << s.request_size << '\t' << s.response_size;
+  return os.str();
+}
+
 struct Ring {
   std::mutex mu;
   std::vector<Span> spans;
@@ -52,9 +94,51 @@ void RecordSpan(const Span& span) {
   }
   ++r.next;
   r.total.fetch_add(1, std::memory_order_relaxed);
+  if (!FLAG_rpcz_db_path.empty()) {
+    SpanDb& db = span_db();
+    std::lock_guard<std::mutex> dlk(db.mu);
+    db.maybe_open();
+    if (db.writer != nullptr) {
+      db.writer->Write(span_to_record(span));
+      if ((++db.written & 63) == 0) db.writer->Flush();
+    }
+  }
+}
+
+// Reads back up to `max` most recent persisted spans (whole-file scan —
+// recordio is append-only; /rpcz?db=N is a diagnostics query, not a hot
+// path). Returns lines.
+std::vector<std::string> ReadPersistedSpans(int max) {
+  std::vector<std::string> out;
+  std::string path;
+  {
+    SpanDb& db = span_db();
+    std::lock_guard<std::mutex> lk(db.mu);
+    if (db.writer != nullptr) db.writer->Flush();
+    path = db.path.empty() ? FLAG_rpcz_db_path : db.path;
+  }
+  if (path.empty()) return out;
+  RecordReader reader(path);
+  if (!reader.ok()) return out;
+  std::string rec;
+  while (reader.Next(&rec)) {
+    out.push_back(rec);
+    if ((int)out.size() > max * 4) out.erase(out.begin(), out.begin() + max);
+  }
+  if ((int)out.size() > max) out.erase(out.begin(), out.end() - max);
+  return out;
 }
 
 int64_t span_count() { return ring().total.load(std::memory_order_relaxed); }
+
+void DumpPersistedSpans(IOBuf* out, int max) {
+  std::vector<std::string> recs = ReadPersistedSpans(max > 0 ? max : 100);
+  std::ostringstream os;
+  os << "persisted_spans (last " << recs.size() << " from " << FLAG_rpcz_db_path << ")\n";
+  os << "start_us\tend_us\tside\tmethod\tremote\terror\tlog_id\treq\tresp\n";
+  for (const std::string& r : recs) os << r << "\n";
+  out->append(os.str());
+}
 
 void DumpRecentSpans(IOBuf* out, bool verbose) {
   Ring& r = ring();

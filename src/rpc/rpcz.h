@@ -31,6 +31,9 @@ bool enabled();
 void set_enabled(bool on);
 void RecordSpan(const Span& span);
 void DumpRecentSpans(IOBuf* out, bool verbose);
+// Reads back the newest `max` spans from the recordio SpanDB
+// (-rpcz_db_path); header-only output when no db is configured.
+void DumpPersistedSpans(IOBuf* out, int max);
 int64_t span_count();
 
 }  // namespace rpcz

@@ -185,3 +185,28 @@ def test_memory_page_has_malloc_stats():
         "http://127.0.0.1:%d/memory" % port, timeout=5).read().decode()
     assert "malloc_in_use_bytes" in body
     assert "iobuf_block_count" in body
+
+
+def test_rpcz_persistent_spandb(tmp_path):
+    """rpcz SpanDB (≙ reference leveldb-backed span store, brpc/span.cpp):
+    sampled spans persist to a recordio file and query back at /rpcz?db=N
+    across what would be a process restart (file survives)."""
+    import urllib.request
+    db = str(tmp_path / "spans.rio")
+    assert b.core.util.set_flag("rpcz_db_path", db) == 0
+    assert b.core.util.set_flag("rpcz_sample_mod", "1") == 0
+    try:
+        port = r.start_echo_server(0)
+        addr = "127.0.0.1:%d" % port
+        for i in range(50):
+            r.echo_once(addr, b"z" * 32, 2000)
+        body = urllib.request.urlopen(
+            "http://127.0.0.1:%d/rpcz?db=20" % port, timeout=5).read().decode()
+        assert "persisted_spans" in body
+        assert "EchoService.Echo" in body
+        # the file itself is recordio: readable independently
+        lines = [l for l in body.splitlines() if "EchoService.Echo" in l]
+        assert len(lines) >= 10
+    finally:
+        b.core.util.set_flag("rpcz_db_path", "")
+        b.core.util.set_flag("rpcz_sample_mod", "16")
