@@ -44,7 +44,10 @@ bool attachment_test(const std::string& addr);
 bool pb_stub_test(std::string* err);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
-                       bool hbm_request, bool pooled);
+                       bool hbm_request, bool pooled, int nchannels);
+BenchResult async_echo_bench(const std::string& addr, int payload_size, int pipeline,
+                             int64_t total_calls, int timeout_ms, const std::string& method,
+                             bool pooled);
 }  // namespace rpctest
 }  // namespace bam
 
@@ -284,12 +287,13 @@ void bind_rpc(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   r.def("echo_bench",
         [](const std::string& addr, int payload_size, int concurrency, int64_t total,
-           int timeout_ms, const std::string& method, bool hbm_request, bool pooled) {
+           int timeout_ms, const std::string& method, bool hbm_request, bool pooled,
+           int nchannels) {
           bam::rpctest::BenchResult b;
           {
             py::gil_scoped_release rel;
             b = bam::rpctest::echo_bench(addr, payload_size, concurrency, total, timeout_ms,
-                                         method, hbm_request, pooled);
+                                         method, hbm_request, pooled, nchannels);
           }
           py::dict d;
           d["qps"] = b.qps;
@@ -308,7 +312,31 @@ void bind_rpc(py::module_& m) {
         py::arg("addr"), py::arg("payload_size") = 64, py::arg("concurrency") = 8,
         py::arg("total") = 10000, py::arg("timeout_ms") = 5000,
         py::arg("method") = "EchoService.Echo", py::arg("hbm_request") = false,
-        py::arg("pooled") = false);
+        py::arg("pooled") = false, py::arg("nchannels") = 1);
+  r.def("async_echo_bench",
+        [](const std::string& addr, int payload_size, int pipeline, int64_t total,
+           int timeout_ms, const std::string& method, bool pooled) {
+          bam::rpctest::BenchResult b;
+          {
+            py::gil_scoped_release rel;
+            b = bam::rpctest::async_echo_bench(addr, payload_size, pipeline, total,
+                                               timeout_ms, method, pooled);
+          }
+          py::dict d;
+          d["qps"] = b.qps;
+          d["mbps"] = b.mbps;
+          d["p50_us"] = b.p50_us;
+          d["p99_us"] = b.p99_us;
+          d["p999_us"] = b.p999_us;
+          d["avg_us"] = b.avg_us;
+          d["errors"] = b.errors;
+          d["first_error"] = b.first_error;
+          d["total"] = b.total;
+          return d;
+        },
+        py::arg("addr"), py::arg("payload_size") = 64, py::arg("pipeline") = 64,
+        py::arg("total") = 10000, py::arg("timeout_ms") = 5000,
+        py::arg("method") = "EchoService.Echo", py::arg("pooled") = false);
 }
 
 // ---- combo channels & LBs ----

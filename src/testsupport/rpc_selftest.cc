@@ -13,6 +13,7 @@
 #include "fiber/fiber.h"
 #include "fiber/sync.h"
 #include "rpc/channel.h"
+#include "rpc/closure.h"
 #include "rpc/controller.h"
 #include "rpc/server.h"
 #include "rpc/shm_ring.h"
@@ -369,6 +370,8 @@ namespace {
 
 struct BenchWorkerArg {
   Channel* channel;
+  std::vector<std::unique_ptr<Channel>>* channels = nullptr;  // multi-channel mode
+  std::atomic<int>* next_worker = nullptr;
   std::atomic<int64_t>* remaining;
   std::vector<int64_t>* latencies;  // pre-sized; indexed by call #
   std::atomic<int64_t>* errors;
@@ -382,6 +385,11 @@ struct BenchWorkerArg {
 
 void bench_worker(void* raw) {
   BenchWorkerArg* a = (BenchWorkerArg*)raw;
+  Channel* ch = a->channel;
+  if (a->channels != nullptr && a->next_worker != nullptr) {
+    int wid = a->next_worker->fetch_add(1, std::memory_order_relaxed);
+    ch = (*a->channels)[wid % (int)a->channels->size()].get();
+  }
   for (;;) {
     int64_t idx = a->remaining->fetch_sub(1, std::memory_order_relaxed);
     if (idx <= 0) break;
@@ -392,7 +400,7 @@ void bench_worker(void* raw) {
     } else {
       request.append(a->payload);
     }
-    a->channel->CallMethod(a->method, &cntl, &request, &response, nullptr);
+    ch->CallMethod(a->method, &cntl, &request, &response, nullptr);
     if (cntl.Failed() || response.size() != a->payload.size()) {
       a->errors->fetch_add(1, std::memory_order_relaxed);
       std::lock_guard<std::mutex> lk(*a->err_mu);
@@ -411,13 +419,19 @@ void bench_worker(void* raw) {
 
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
-                       bool hbm_request, bool pooled) {
+                       bool hbm_request, bool pooled, int nchannels) {
   BenchResult res;
-  Channel channel;
-  ChannelOptions opts;
-  opts.timeout_ms = timeout_ms;
-  if (pooled) opts.connection_type = "pooled";
-  if (channel.Init(addr.c_str(), &opts) != 0) return res;
+  if (nchannels < 1) nchannels = 1;
+  std::vector<std::unique_ptr<Channel>> channels(nchannels);
+  for (int i = 0; i < nchannels; ++i) {
+    channels[i] = std::make_unique<Channel>();
+    ChannelOptions opts;
+    opts.timeout_ms = timeout_ms;
+    if (pooled) opts.connection_type = "pooled";
+    if (channels[i]->Init(addr.c_str(), &opts) != 0) return res;
+  }
+  Channel& channel = *channels[0];
+  (void)channel;
   std::string payload(payload_size, 'x');
   for (int i = 0; i < payload_size; ++i) payload[i] = (char)fast_rand();
   std::atomic<int64_t> remaining{total_calls};
@@ -425,8 +439,20 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
   std::vector<int64_t> latencies(total_calls, 0);
   CountdownEvent done_event(concurrency);
   std::mutex err_mu;
-  BenchWorkerArg arg{&channel,  &remaining,  &latencies,  &errors,   payload,
-                     method,    hbm_request, &done_event, &err_mu,   &res.first_error};
+  std::atomic<int> next_worker{0};
+  BenchWorkerArg arg;
+  arg.channel = &channel;
+  arg.channels = &channels;
+  arg.next_worker = &next_worker;
+  arg.remaining = &remaining;
+  arg.latencies = &latencies;
+  arg.errors = &errors;
+  arg.payload = payload;
+  arg.method = method;
+  arg.hbm_request = hbm_request;
+  arg.done_event = &done_event;
+  arg.err_mu = &err_mu;
+  arg.first_error = &res.first_error;
   int64_t t0 = monotonic_time_us();
   for (int i = 0; i < concurrency; ++i) {
     fiber_t th;
@@ -447,6 +473,110 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
     return latencies[i];
   };
   if (!latencies.empty()) {
+    res.p50_us = pct(0.50);
+    res.p90_us = pct(0.90);
+    res.p99_us = pct(0.99);
+    res.p999_us = pct(0.999);
+    res.max_us = latencies.back();
+  }
+  return res;
+}
+
+namespace {
+
+// Pipelined async bench (≙ reference test/brpc_channel_unittest.cpp async
+// paths + docs/cn/benchmark.md single-connection pipelining): `pipeline`
+// calls stay in flight on the channel; each completion immediately issues
+// the next, so the wait-free write queue coalesces requests into large
+// writev batches and the reader parses many responses per wakeup.
+struct AsyncBenchCtx {
+  Channel* channel = nullptr;
+  std::atomic<int64_t> remaining{0};
+  std::atomic<int64_t> errors{0};
+  std::string payload;
+  std::string method;
+  CountdownEvent* done_event = nullptr;  // count == pipeline
+  std::vector<int64_t>* latencies = nullptr;
+};
+
+void async_issue(AsyncBenchCtx* c);
+
+struct AsyncCall {
+  Controller cntl;
+  IOBuf req, resp;
+};
+
+void async_reissue_fiber(void* raw) { async_issue((AsyncBenchCtx*)raw); }
+
+void async_issue(AsyncBenchCtx* c) {
+  int64_t idx = c->remaining.fetch_sub(1, std::memory_order_relaxed);
+  if (idx <= 0) {
+    c->done_event->signal();
+    return;
+  }
+  AsyncCall* call = new AsyncCall;
+  call->req.append(c->payload);
+  c->channel->CallMethod(c->method, &call->cntl, &call->req, &call->resp,
+                         NewCallback([c, call, idx] {
+    if (call->cntl.Failed() || call->resp.size() != c->payload.size()) {
+      c->errors.fetch_add(1, std::memory_order_relaxed);
+    }
+    (*c->latencies)[idx - 1] = call->cntl.latency_us();
+    delete call;
+    // Completions normally arrive from the reader; an inline-failed call
+    // would recurse here, so bound the depth with a fiber trampoline.
+    thread_local int depth = 0;
+    if (++depth > 64) {
+      depth = 0;
+      fiber_t th;
+      fiber_start_background(&th, async_reissue_fiber, c);
+    } else {
+      async_issue(c);
+      --depth;
+    }
+  }));
+}
+
+}  // namespace
+
+BenchResult async_echo_bench(const std::string& addr, int payload_size, int pipeline,
+                             int64_t total_calls, int timeout_ms, const std::string& method,
+                             bool pooled) {
+  BenchResult res;
+  Channel channel;
+  ChannelOptions opts;
+  opts.timeout_ms = timeout_ms;
+  opts.max_retry = 0;
+  if (pooled) opts.connection_type = "pooled";
+  if (channel.Init(addr.c_str(), &opts) != 0) return res;
+  std::string payload(payload_size, 'x');
+  for (int i = 0; i < payload_size; ++i) payload[i] = (char)fast_rand();
+  std::vector<int64_t> latencies(total_calls, 0);
+  CountdownEvent done_event(pipeline);
+  AsyncBenchCtx ctx;
+  ctx.channel = &channel;
+  ctx.remaining.store(total_calls);
+  ctx.payload = payload;
+  ctx.method = method;
+  ctx.done_event = &done_event;
+  ctx.latencies = &latencies;
+  int64_t t0 = monotonic_time_us();
+  for (int i = 0; i < pipeline; ++i) async_issue(&ctx);
+  done_event.wait();
+  int64_t elapsed = monotonic_time_us() - t0;
+  std::sort(latencies.begin(), latencies.end());
+  res.total = total_calls;
+  res.errors = ctx.errors.load();
+  res.qps = total_calls * 1e6 / (double)elapsed;
+  res.mbps = res.qps * payload_size / 1e6;
+  int64_t sum = 0;
+  for (int64_t v : latencies) sum += v;
+  res.avg_us = total_calls > 0 ? sum / total_calls : 0;
+  if (!latencies.empty()) {
+    auto pct = [&](double p) {
+      size_t i = (size_t)(p * (latencies.size() - 1));
+      return latencies[i];
+    };
     res.p50_us = pct(0.50);
     res.p90_us = pct(0.90);
     res.p99_us = pct(0.99);

@@ -105,3 +105,25 @@ def test_interceptor_admission():
     rc, err = b.core.combo.call_with_logid(addr, 123)
     assert rc == 1004  # EAUTH
     assert "credential" in err
+
+
+def test_async_pipelined_bench():
+    """Pipelined async client (completions reissue; ≙ reference async
+    CallMethod w/ done, test/brpc_channel_unittest.cpp async paths):
+    all calls complete, none lost, latencies recorded."""
+    port = r.start_echo_server(0)
+    addr = "127.0.0.1:%d" % port
+    res = r.async_echo_bench(addr, 64, 16, 2000, 10000, "EchoService.Echo", True)
+    assert res["errors"] == 0, res["first_error"]
+    assert res["total"] == 2000
+    assert res["qps"] > 0 and res["p99_us"] > 0
+
+
+def test_echo_bench_multi_channel():
+    """nchannels>1 spreads sync workers over independent Channels."""
+    port = r.start_echo_server(0)
+    addr = "127.0.0.1:%d" % port
+    res = r.echo_bench(addr, 64, 8, 1000, 10000, "EchoService.Echo",
+                       False, True, 4)
+    assert res["errors"] == 0, res["first_error"]
+    assert res["total"] == 1000
