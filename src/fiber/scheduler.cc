@@ -262,9 +262,13 @@ void TaskControl::set_concurrency_hint(int n) {
 
 TaskControl::TaskControl() {
   int n = g_concurrency_hint.load(std::memory_order_relaxed);
-  if (n <= 0) {
+  bool chosen = n > 0;
+  if (!chosen) {
     const char* env = getenv("BAM_FIBER_WORKERS");
-    if (env != nullptr) n = atoi(env);
+    if (env != nullptr) {
+      n = atoi(env);
+      chosen = n > 0;
+    }
   }
   if (n <= 0) n = (int)std::thread::hardware_concurrency();
   if (n <= 0) n = 4;
@@ -272,8 +276,9 @@ TaskControl::TaskControl() {
   // the spin-before-park burn multiplies. Same-box measurements (MI355X
   // host, echo bench): cap 8 vs 16 -> HBM p99 ~380 vs ~700-1850 us at
   // equal QPS, host path +12% QPS. The reference's default is 8+1
-  // workers too. BAM_FIBER_WORKERS overrides.
-  if (n > 8) n = 8;
+  // workers too. An explicit BAM_FIBER_WORKERS / concurrency hint is
+  // taken as-is (no cap).
+  if (!chosen && n > 8) n = 8;
   start_workers(n);
 }
 
