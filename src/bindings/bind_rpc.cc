@@ -1,6 +1,7 @@
 #include "rpc/load_balancer.h"
 #include "rpc/mysql_client.h"
 #include "rpc/flv.h"
+#include "rpc/ts.h"
 #include "rpc/rtmp_client.h"
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -81,6 +82,39 @@ void bind_rpc(py::module_& m) {
         },
         py::arg("host"), py::arg("port"), py::arg("app"), py::arg("stream"),
         py::arg("ntags"), py::arg("timeout_ms") = 5000);
+  // HLS leg (≙ reference brpc/ts.cpp): FLV tags -> MPEG-TS + m3u8.
+  r.def("flv_to_ts", [](py::bytes flv_doc) {
+    std::vector<bam::flv::Tag> tags;
+    if (!bam::flv::Parse(std::string(flv_doc), &tags))
+      throw std::runtime_error("bad FLV document");
+    std::string out;
+    if (!bam::ts::FlvToTs(tags, &out))
+      throw std::runtime_error("no muxable a/v tags");
+    return py::bytes(out);
+  });
+  r.def("ts_mux_tags", [](py::list tags) {
+    bam::ts::TsMuxer mux;
+    std::string out;
+    mux.WriteTables(&out);
+    for (auto item : tags) {
+      py::tuple t = item.cast<py::tuple>();
+      bam::flv::Tag tag;
+      tag.type = t[0].cast<int>();
+      tag.timestamp = t[1].cast<uint32_t>();
+      tag.payload = t[2].cast<std::string>();
+      mux.Write(tag, &out);
+    }
+    return py::bytes(out);
+  });
+  r.def("hls_playlist", [](py::list segs, int target_s, int seq, bool ended) {
+    std::vector<bam::ts::HlsSegment> v;
+    for (auto item : segs) {
+      py::tuple t = item.cast<py::tuple>();
+      v.push_back({t[0].cast<std::string>(), t[1].cast<double>()});
+    }
+    return bam::ts::MakeHlsPlaylist(v, target_s, seq, ended);
+  }, py::arg("segments"), py::arg("target_duration_s") = 10,
+     py::arg("media_sequence") = 0, py::arg("ended") = true);
   r.def("flv_parse", [](py::bytes doc) {
     std::vector<bam::flv::Tag> tags;
     bool ha = false, hv = false;
