@@ -4,6 +4,7 @@
 
 #include <string.h>
 
+#include "base/fast_rand.h"
 #include "base/logging.h"
 #include "base/time.h"
 #include "rpc/load_balancer.h"
@@ -91,6 +92,9 @@ void EndRPC(Controller* cntl, SessionId locked_id) {
     span.remote = cntl->remote_side_;
     span.error_code = cntl->error_code_;
     span.log_id = cntl->log_id_;
+    span.trace_id = cntl->trace_id_;
+    span.span_id = cntl->span_id_;
+    span.parent_span_id = cntl->parent_span_id_;
     span.server_side = false;
     span.request_size = cntl->call.request_buf.size();
     rpcz::RecordSpan(span);
@@ -230,6 +234,21 @@ void IssueRPC(Controller* cntl) {
 void Channel::CallMethod(const std::string& full_method, Controller* cntl,
                          const IOBuf* request, IOBuf* response, Closure* done) {
   cntl->start_us_ = monotonic_time_us();
+  if (rpcz::enabled()) {
+    // Inherit the ambient trace (server handler context) or start a new
+    // one; every client call gets a fresh span id (≙ reference
+    // channel.cpp:524 CreateClientSpan).
+    if (cntl->trace_id_ == 0) {
+      rpcz::TraceContext tc = rpcz::current_trace();
+      if (tc.trace_id != 0) {
+        cntl->trace_id_ = tc.trace_id;
+        cntl->parent_span_id_ = tc.span_id;
+      } else {
+        cntl->trace_id_ = fast_rand() | 1;
+      }
+    }
+    if (cntl->span_id_ == 0) cntl->span_id_ = fast_rand() | 1;
+  }
   if (cntl->timeout_ms_ == -1) cntl->timeout_ms_ = options_.timeout_ms;
   if (cntl->max_retry_ == 3 /*default*/) cntl->max_retry_ = options_.max_retry;
   split_full_method(full_method, &cntl->call.service_name, &cntl->call.method_name);

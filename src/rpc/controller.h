@@ -45,6 +45,15 @@ class Controller {
   int64_t backup_request_ms() const { return backup_request_ms_; }
   void set_log_id(uint64_t id) { log_id_ = id; }
   uint64_t log_id() const { return log_id_; }
+
+  // Trace ids (≙ reference brpc/span.h trace propagation): client calls
+  // inherit the ambient trace (rpcz::current_trace) unless set explicitly;
+  // servers read them off RpcRequestMeta and re-export while the handler
+  // runs, so nested client calls chain parent_span_id automatically.
+  void set_trace_id(uint64_t id) { trace_id_ = id; }
+  uint64_t trace_id() const { return trace_id_; }
+  uint64_t span_id() const { return span_id_; }
+  uint64_t parent_span_id() const { return parent_span_id_; }
   void set_request_compress_type(CompressType t) { request_compress_ = t; }
   CompressType request_compress_type() const { return request_compress_; }
 
@@ -143,6 +152,9 @@ class Controller {
   int max_retry_ = 3;
   int retry_count_ = 0;
   uint64_t log_id_ = 0;
+  uint64_t trace_id_ = 0;
+  uint64_t span_id_ = 0;
+  uint64_t parent_span_id_ = 0;
   CompressType request_compress_ = COMPRESS_TYPE_NONE;
   IOBuf request_attachment_;
   IOBuf response_attachment_;

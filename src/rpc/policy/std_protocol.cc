@@ -33,6 +33,10 @@ void SerializeRpcMeta(const RpcMeta& meta, std::string* out) {
     wire::put_str_field(&sub, 1, meta.service_name);
     wire::put_str_field(&sub, 2, meta.method_name);
     if (meta.log_id != 0) wire::put_int_field(&sub, 3, (int64_t)meta.log_id);
+    if (meta.trace_id != 0) wire::put_int_field(&sub, 4, (int64_t)meta.trace_id);
+    if (meta.span_id != 0) wire::put_int_field(&sub, 5, (int64_t)meta.span_id);
+    if (meta.parent_span_id != 0)
+      wire::put_int_field(&sub, 6, (int64_t)meta.parent_span_id);
     wire::put_msg_field(out, 1, sub);
   }
   if (meta.has_response) {
@@ -67,6 +71,9 @@ bool ParseRpcMeta(const char* data, size_t n, RpcMeta* out) {
           if (f2 == 1) out->service_name = rr.read_string();
           else if (f2 == 2) out->method_name = rr.read_string();
           else if (f2 == 3) out->log_id = rr.varint();
+          else if (f2 == 4) out->trace_id = rr.varint();
+          else if (f2 == 5) out->span_id = rr.varint();
+          else if (f2 == 6) out->parent_span_id = rr.varint();
           else rr.skip(wt2);
           if (!rr.ok()) return false;
         }
@@ -155,6 +162,9 @@ void PackStdRequest(IOBuf* out, Controller* cntl, SessionId correlation_id) {
   meta.service_name = cntl->call.service_name;
   meta.method_name = cntl->call.method_name;
   meta.log_id = cntl->log_id();
+  meta.trace_id = cntl->trace_id();
+  meta.span_id = cntl->span_id();
+  meta.parent_span_id = cntl->parent_span_id();
   meta.compress_type = (int)cntl->request_compress_type();
   meta.correlation_id = (int64_t)correlation_id;
   meta.attachment_size = (int32_t)cntl->request_attachment().size();
@@ -255,6 +265,9 @@ static void SendStdResponse(SocketId sid, int64_t cid, Controller* cntl, IOBuf* 
     span.remote = cntl->remote_side();
     span.error_code = cntl->ErrorCode();
     span.log_id = cntl->log_id();
+    span.trace_id = cntl->trace_id();
+    span.span_id = cntl->span_id();
+    span.parent_span_id = cntl->parent_span_id();
     span.server_side = true;
     span.response_size = resp->size();
     rpcz::RecordSpan(span);
@@ -276,6 +289,9 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   cntl->server_socket_ = sock->id();
   cntl->server_cid_ = msg->meta.correlation_id;
   cntl->log_id_ = msg->meta.log_id;
+  cntl->trace_id_ = msg->meta.trace_id;
+  cntl->span_id_ = msg->meta.span_id;
+  cntl->parent_span_id_ = msg->meta.parent_span_id;
   cntl->call.service_name = msg->meta.service_name;
   cntl->call.method_name = msg->meta.method_name;
   cntl->remote_stream_id_ = msg->meta.stream_id;
@@ -379,7 +395,14 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
   server->concurrency.fetch_add(1, std::memory_order_relaxed);
   cntl->concurrency_counted_ = true;
   rpc_dump::SampleRequest(msg->meta.service_name, msg->meta.method_name, req_data);
+  // Export the inbound trace while the handler runs: nested client calls
+  // made inside it chain parent_span_id from this server span. (Handlers
+  // that stash `done` and return before issuing sub-calls fall outside
+  // the ambient window — pass ids explicitly via set_trace_id there.)
+  const bool ambient_trace = cntl->trace_id_ != 0;
+  if (ambient_trace) rpcz::set_current_trace(cntl->trace_id_, cntl->span_id_);
   (*fn)(cntl, req_data, resp, done);
+  if (ambient_trace) rpcz::clear_current_trace();
   delete msg;
 }
 

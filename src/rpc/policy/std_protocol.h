@@ -20,6 +20,11 @@ struct RpcMeta {
   std::string service_name;
   std::string method_name;
   uint64_t log_id = 0;
+  // trace propagation (≙ reference baidu_rpc_meta.proto RpcRequestMeta
+  // fields 4-6: trace_id/span_id/parent_span_id)
+  uint64_t trace_id = 0;
+  uint64_t span_id = 0;
+  uint64_t parent_span_id = 0;
   // response
   int error_code = 0;
   std::string error_text;

@@ -1,5 +1,9 @@
 #include "rpc/rpcz.h"
 
+#include <mutex>
+
+#include "fiber/key.h"
+
 #include <atomic>
 #include <mutex>
 #include <sstream>
@@ -57,7 +61,8 @@ std::string span_to_record(const Span& s) {
   std::ostringstream os;
   os << s.start_us << '\t' << s.end_us << '\t' << (s.server_side ? 'S' : 'C') << '\t'
      << s.full_method << '\t' << endpoint2str(s.remote) << '\t' << s.error_code << '\t'
-     << s.log_id << '\t' << s.request_size << '\t' << s.response_size;
+     << s.log_id << '\t' << s.request_size << '\t' << s.response_size << '\t'
+     << s.trace_id << '\t' << s.span_id << '\t' << s.parent_span_id;
   return os.str();
 }
 
@@ -72,6 +77,38 @@ Ring& ring() {
   return *r;
 }
 }  // namespace
+
+namespace {
+fiber_key_t g_trace_key;
+std::once_flag g_trace_key_once;
+
+void trace_dtor(void* p) { delete (TraceContext*)p; }
+
+fiber_key_t trace_key() {
+  std::call_once(g_trace_key_once, [] { fiber_key_create(&g_trace_key, trace_dtor); });
+  return g_trace_key;
+}
+}  // namespace
+
+TraceContext current_trace() {
+  TraceContext* c = (TraceContext*)fiber_getspecific(trace_key());
+  return c != nullptr ? *c : TraceContext{};
+}
+
+void set_current_trace(uint64_t trace_id, uint64_t span_id) {
+  TraceContext* c = (TraceContext*)fiber_getspecific(trace_key());
+  if (c == nullptr) {
+    c = new TraceContext;
+    fiber_setspecific(trace_key(), c);
+  }
+  c->trace_id = trace_id;
+  c->span_id = span_id;
+}
+
+void clear_current_trace() {
+  TraceContext* c = (TraceContext*)fiber_getspecific(trace_key());
+  if (c != nullptr) c->trace_id = c->span_id = 0;
+}
 
 bool enabled() { return FLAG_enable_rpcz; }
 void set_enabled(bool on) { FLAG_enable_rpcz = on; }
@@ -135,7 +172,8 @@ void DumpPersistedSpans(IOBuf* out, int max) {
   std::vector<std::string> recs = ReadPersistedSpans(max > 0 ? max : 100);
   std::ostringstream os;
   os << "persisted_spans (last " << recs.size() << " from " << FLAG_rpcz_db_path << ")\n";
-  os << "start_us\tend_us\tside\tmethod\tremote\terror\tlog_id\treq\tresp\n";
+  os << "start_us\tend_us\tside\tmethod\tremote\terror\tlog_id\treq\tresp"
+        "\ttrace_id\tspan_id\tparent_span_id\n";
   for (const std::string& r : recs) os << r << "\n";
   out->append(os.str());
 }
@@ -151,6 +189,11 @@ void DumpRecentSpans(IOBuf* out, bool verbose) {
        << " | " << endpoint2str(s.remote) << " | " << (s.end_us - s.start_us) << " | "
        << s.error_code << " | " << s.request_size << " | " << s.response_size;
     if (verbose && s.log_id != 0) os << " | log_id=" << s.log_id;
+    if (verbose && s.trace_id != 0) {
+      os << " | trace=" << std::hex << s.trace_id << " span=" << s.span_id;
+      if (s.parent_span_id != 0) os << " parent=" << s.parent_span_id;
+      os << std::dec;
+    }
     os << "\n";
   }
   out->append(os.str());

@@ -17,6 +17,9 @@ namespace rpcz {
 
 struct Span {
   int64_t start_us = 0;
+  uint64_t trace_id = 0;
+  uint64_t span_id = 0;
+  uint64_t parent_span_id = 0;
   int64_t end_us = 0;
   std::string full_method;
   EndPoint remote;
@@ -26,6 +29,18 @@ struct Span {
   uint64_t request_size = 0;
   uint64_t response_size = 0;
 };
+
+// Ambient trace for the CURRENT execution context (fiber-local; plain
+// TLS off-fiber): servers export the inbound ids while the handler runs
+// so nested client calls chain parent_span_id (≙ reference span.h:153
+// TLS parent chaining via bthread_set_span_funcs).
+struct TraceContext {
+  uint64_t trace_id = 0;
+  uint64_t span_id = 0;
+};
+TraceContext current_trace();
+void set_current_trace(uint64_t trace_id, uint64_t span_id);
+void clear_current_trace();
 
 bool enabled();
 void set_enabled(bool on);

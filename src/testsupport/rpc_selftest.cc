@@ -48,6 +48,36 @@ Service* NewEchoService() {
     cntl->set_response_compress_type(COMPRESS_TYPE_SNAPPY);
     done->Run();
   });
+  // Nested-call relay: request = "host:port|payload". The handler calls
+  // EchoService.Echo on that address; trace ids must chain through the
+  // ambient rpcz context (server span -> nested client span).
+  svc->AddMethod("Relay", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    std::string s = req.to_string();
+    size_t bar = s.find('|');
+    if (bar == std::string::npos) {
+      cntl->SetFailed(EREQUEST, "Relay wants addr|payload");
+      done->Run();
+      return;
+    }
+    Channel ch;
+    ChannelOptions opts;
+    opts.timeout_ms = 3000;
+    if (ch.Init(s.substr(0, bar).c_str(), &opts) != 0) {
+      cntl->SetFailed(EINTERNAL, "relay channel init failed");
+      done->Run();
+      return;
+    }
+    Controller c2;
+    IOBuf r2, resp2;
+    r2.append(s.substr(bar + 1));
+    ch.CallMethod("EchoService.Echo", &c2, &r2, &resp2, nullptr);
+    if (c2.Failed()) {
+      cntl->SetFailed(c2.ErrorCode(), "relay: " + c2.ErrorText());
+    } else {
+      resp->append(resp2);
+    }
+    done->Run();
+  });
   svc->AddMethod("Port", [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
     resp->append(std::to_string(cntl->server_->listen_address().port));
     done->Run();

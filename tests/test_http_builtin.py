@@ -210,3 +210,44 @@ def test_rpcz_persistent_spandb(tmp_path):
     finally:
         b.core.util.set_flag("rpcz_db_path", "")
         b.core.util.set_flag("rpcz_sample_mod", "16")
+
+
+def test_trace_id_chains_through_nested_calls():
+    """Trace propagation (≙ reference brpc/span.h: trace_id/span_id/
+    parent_span_id ride RpcRequestMeta fields 4-6; TLS parent chaining
+    span.h:153): client -> A.Relay -> nested Echo on B. The nested client
+    span and B's server span must share A's trace_id, and the nested
+    call's parent_span_id must be A's inbound span id."""
+    import re
+    import urllib.request
+    assert b.core.util.set_flag("rpcz_sample_mod", "1") == 0
+    try:
+        port_b = r.start_echo_server(0)
+        port_a = r.start_echo_server(0)
+        payload = b"trace-me-7391"
+        rc, resp, _ = r.call_method_once("127.0.0.1:%d" % port_a,
+                                         "EchoService.Relay",
+                                         ("127.0.0.1:%d|" % port_b).encode() + payload,
+                                         3000, 0)
+        assert rc == 0 and resp == payload
+        body = urllib.request.urlopen(
+            "http://127.0.0.1:%d/rpcz?verbose" % port_a, timeout=5).read().decode()
+        # rows: "... | trace=<hex> span=<hex> [parent=<hex>]"
+        rows = [l for l in body.splitlines() if "trace=" in l]
+        relay_s = [l for l in rows if "EchoService.Relay" in l and "| S |" in l]
+        echo_c = [l for l in rows if "EchoService.Echo" in l and "| C |" in l]
+        echo_s = [l for l in rows if "EchoService.Echo" in l and "| S |" in l]
+        assert relay_s and echo_c and echo_s
+
+        def ids(line):
+            m = re.search(r"trace=([0-9a-f]+) span=([0-9a-f]+)(?: parent=([0-9a-f]+))?",
+                          line)
+            return m.group(1), m.group(2), m.group(3)
+        rt, rs, _ = ids(relay_s[-1])
+        ct, cs, cp = ids(echo_c[-1])
+        st, ss, sp = ids(echo_s[-1])
+        assert ct == rt == st, (rt, ct, st)
+        assert cp == rs, "nested client parent %s != relay server span %s" % (cp, rs)
+        assert ss == cs and sp == cp  # B's server sees the nested call's ids
+    finally:
+        b.core.util.set_flag("rpcz_sample_mod", "16")
