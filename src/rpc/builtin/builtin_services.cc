@@ -4,6 +4,11 @@
 // /vars, /flags, /health, /version, /connections, /protobufs, /fibers
 // (≙ /bthreads), /memory, /brpc_metrics (Prometheus), /rpcz (spans).
 #include <string.h>
+#include <dirent.h>
+#include <errno.h>
+
+#include <algorithm>
+#include <vector>
 
 #include <sstream>
 
@@ -36,7 +41,7 @@ void page_index(HttpResponse* resp) {
   std::ostringstream os;
   os << "<html><head><title>brpc_amd</title></head><body><h1>brpc_amd server</h1><ul>";
   const char* pages[] = {"status", "vars",   "flags",  "health",       "version",
-                         "connections", "protobufs", "fibers", "memory", "threads",
+                         "connections", "sockets", "list", "dir", "protobufs", "fibers", "memory", "threads",
                          "hotspots/cpu", "hotspots/contention", "hotspots/gpu",
                          "rpcz", "brpc_metrics"};
   for (const char* p : pages) os << "<li><a href=\"/" << p << "\">/" << p << "</a></li>";
@@ -121,6 +126,80 @@ void page_connections(HttpResponse* resp) {
        << " | " << s->out_bytes.load() << " | " << s->in_messages.load() << " | "
        << s->out_messages.load() << "\n";
   }
+  resp->body.append(os.str());
+}
+
+void page_list(Server* server, HttpResponse* resp) {
+  // Machine-readable service list (≙ reference builtin/list_service.cpp,
+  // which returns a proto; ours is JSON).
+  std::ostringstream os;
+  os << "[";
+  bool first_s = true;
+  if (server != nullptr) {
+    for (const auto& kv : server->services()) {
+      if (!first_s) os << ",";
+      first_s = false;
+      os << "{\"service\":\"" << kv.first << "\",\"methods\":[";
+      bool first_m = true;
+      for (const auto& m : kv.second->methods()) {
+        if (!first_m) os << ",";
+        first_m = false;
+        os << "\"" << m.first << "\"";
+      }
+      os << "]}";
+    }
+  }
+  os << "]\n";
+  resp->headers["Content-Type"] = "application/json";
+  resp->body.append(os.str());
+}
+
+void page_sockets(const HttpRequest& req, HttpResponse* resp) {
+  // Per-socket detail (≙ reference builtin/sockets_service.cpp: one
+  // Socket's DebugString by id; summary without one).
+  auto it = req.query.find("id");
+  if (it == req.query.end()) {
+    page_connections(resp);
+    return;
+  }
+  SocketId id = (SocketId)strtoull(it->second.c_str(), nullptr, 10);
+  SocketUniquePtr s;
+  if (Socket::Address(id, &s) != 0) {
+    resp->status = 404;
+    resp->body.append("no such socket (recycled or never existed)\n");
+    return;
+  }
+  std::ostringstream os;
+  os << "socket_id: " << id << "\n"
+     << "fd: " << s->fd() << "\n"
+     << "remote: " << endpoint2str(s->remote_side()) << "\n"
+     << "local: " << endpoint2str(s->local_side()) << "\n"
+     << "in_bytes: " << s->in_bytes.load() << "\n"
+     << "out_bytes: " << s->out_bytes.load() << "\n"
+     << "in_messages: " << s->in_messages.load() << "\n"
+     << "out_messages: " << s->out_messages.load() << "\n"
+     << "failed: " << (s->Failed() ? 1 : 0) << "\n";
+  resp->body.append(os.str());
+}
+
+void page_dir(const HttpRequest& req, HttpResponse* resp) {
+  // Directory listing (≙ reference builtin/dir_service.cpp).
+  std::string path = "/";
+  auto it = req.query.find("path");
+  if (it != req.query.end() && !it->second.empty()) path = it->second;
+  DIR* d = opendir(path.c_str());
+  if (d == nullptr) {
+    resp->status = 404;
+    resp->body.append("cannot open " + path + ": " + strerror(errno) + "\n");
+    return;
+  }
+  std::vector<std::string> names;
+  while (struct dirent* e = readdir(d)) names.push_back(e->d_name);
+  closedir(d);
+  std::sort(names.begin(), names.end());
+  std::ostringstream os;
+  os << path << ":\n";
+  for (const std::string& n : names) os << n << "\n";
   resp->body.append(os.str());
 }
 
@@ -225,6 +304,12 @@ bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse
     resp->body.append("brpc_amd/1.0 gfx950\n");
   } else if (p == "/connections") {
     page_connections(resp);
+  } else if (p == "/sockets") {
+    page_sockets(req, resp);
+  } else if (p == "/list") {
+    page_list(server, resp);
+  } else if (p == "/dir") {
+    page_dir(req, resp);
   } else if (p == "/protobufs") {
     page_protobufs(server, resp);
   } else if (p == "/fibers" || p == "/bthreads") {

@@ -352,6 +352,10 @@ static void ProcessStdRequest(InputMessageBase* msg_base) {
       server != nullptr
           ? server->FindMethod(msg->meta.service_name, msg->meta.method_name, &svc)
           : nullptr;
+  if (fn == nullptr && server != nullptr && server->options().master_handler) {
+    // Generic/proxy pass-through (≙ reference BaiduMasterService).
+    fn = &server->options().master_handler;
+  }
   if (fn == nullptr) {
     cntl->SetFailed(msg->meta.service_name.empty() || svc == nullptr ? ENOSERVICE : ENOMETHOD,
                     "unknown service/method " + msg->meta.service_name + "." +

@@ -104,6 +104,12 @@ struct ServerOptions {
   // connection recycles.
   std::function<void*()> session_local_data_factory;
   std::function<void(void*)> session_local_data_deleter;
+  // Generic/proxy catch-all (parity: reference BaiduMasterService,
+  // brpc/baidu_master_service.h:36-73): when a baidu_std request names a
+  // service/method this server does not register, the master handler —
+  // if set — receives the raw serialized request instead of ENOMETHOD.
+  // cntl->call.service_name/method_name carry the original names.
+  MethodFn master_handler;
 };
 
 class MethodStatusRecorder;  // var/latency recorder per method (var layer)

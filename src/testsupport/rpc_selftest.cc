@@ -109,6 +109,20 @@ int start_echo_server(int port) {
   return server->listen_address().port;
 }
 
+// Server with ONLY a master catch-all (≙ reference BaiduMasterService):
+// any service/method echoes back "master:<svc>.<method>:<body>".
+int start_master_echo_server() {
+  Server* server = new Server;
+  ServerOptions opts;
+  opts.master_handler = [](Controller* cntl, const IOBuf& req, IOBuf* resp, Closure* done) {
+    resp->append("master:" + cntl->call.service_name + "." + cntl->call.method_name + ":");
+    resp->append(req);
+    done->Run();
+  };
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
 // Starts an echo server that ALSO serves nshead raw-body echo (body is
 // echoed back with "N:" prefixed). Returns port.
 int start_nshead_server() {

@@ -251,3 +251,33 @@ def test_trace_id_chains_through_nested_calls():
         assert ss == cs and sp == cp  # B's server sees the nested call's ids
     finally:
         b.core.util.set_flag("rpcz_sample_mod", "16")
+
+
+def test_list_sockets_dir_pages(base):
+    import json as _json
+    st, body = get(base + "/list")
+    assert st == 200
+    services = _json.loads(body)
+    assert any(s["service"] == "EchoService" and "Echo" in s["methods"]
+               for s in services)
+    st, body = get(base + "/sockets")
+    assert st == 200 and "socket_count" in body
+    # per-id detail: take an id from /connections
+    conn_line = [l for l in body.splitlines() if " | " in l and l[0].isdigit()]
+    if conn_line:
+        sid = conn_line[0].split(" | ")[0]
+        st, det = get(base + "/sockets?id=%s" % sid)
+        assert st == 200 and "remote:" in det and "in_bytes:" in det
+    st, body = get(base + "/dir?path=/root/repo/tests")
+    assert st == 200 and "test_http_builtin.py" in body
+
+
+def test_master_handler_catch_all():
+    """Generic/proxy pass-through (≙ reference BaiduMasterService,
+    baidu_master_service.h:36): unknown service/method lands in the
+    catch-all with the original names and raw request."""
+    port = r.start_master_echo_server()
+    rc, resp, err = r.call_method_once("127.0.0.1:%d" % port,
+                                       "NoSuch.Service", b"raw-bytes", 3000, 0)
+    assert rc == 0, err
+    assert resp == b"master:NoSuch.Service:raw-bytes"
