@@ -254,6 +254,13 @@ def main():
         res16 = run_echo_step(addr, 16384, args.concurrency,
                               max(200, args.calls_per_step // 4), method, pooled)
         e16, q16 = reduce_max_sum(1.0, res16["qps"])  # sum of per-rank QPS
+        # host-path pipelined-async ceiling (config 1 shape: loopback echo
+        # without GPU staging; the pipelined client coalesces writes —
+        # measured 500k QPS on a quiet 8-worker box, profiles/INDEX.md)
+        resh = b.core.rpc.async_echo_bench(addr, 64, 256,
+                                           max(2000, args.calls_per_step),
+                                           30000, "EchoService.Echo", True)
+        _, qh = reduce_max_sum(1.0, resh["qps"] if not resh["errors"] else 0.0)
         # streaming GB/s between ranks 0<->1 through the framework
         stream_gbps = None
         if group is not None:
@@ -281,6 +288,8 @@ def main():
             "p99_us": max(p99s),
             "echo16k_qps": q16,
             "echo16k_p99_us": res16["p99_us"],
+            "host64_async_qps": qh,
+            "host64_async_p99_us": resh["p99_us"],
             "stream_gbps": stream_gbps,
             "connection_type": "pooled" if pooled else "single",
             "parallelism": "dp%d" % (world_size if world_size > 1 else 1),

@@ -47,7 +47,16 @@ inline bool valid_ver_locked(const SessionMeta* m, uint32_t ver) {
 
 }  // namespace
 
+static std::atomic<int64_t> g_sessions_created{0};
+static std::atomic<int64_t> g_sessions_destroyed{0};
+
+void session_stats(int64_t* created, int64_t* destroyed) {
+  *created = g_sessions_created.load(std::memory_order_relaxed);
+  *destroyed = g_sessions_destroyed.load(std::memory_order_relaxed);
+}
+
 int session_create(SessionId* id, void* data, SessionOnError on_error, int range) {
+  g_sessions_created.fetch_add(1, std::memory_order_relaxed);
   if (range < 1) range = 1;
   ResourceId rid;
   SessionMeta* m = get_resource<SessionMeta>(&rid);
@@ -147,7 +156,11 @@ int unlock_impl(SessionId id, bool destroy) {
 
 int session_unlock(SessionId id) { return unlock_impl(id, false); }
 
-int session_unlock_and_destroy(SessionId id) { return unlock_impl(id, true); }
+int session_unlock_and_destroy(SessionId id) {
+  int rc = unlock_impl(id, true);
+  if (rc == 0) g_sessions_destroyed.fetch_add(1, std::memory_order_relaxed);
+  return rc;
+}
 
 int session_join(SessionId id) {
   SessionMeta* m = meta_of(id);

@@ -20,6 +20,10 @@ typedef int (*SessionOnError)(SessionId id, void* data, int error_code);
 // version within the range (all versions join/destroy together).
 int session_create(SessionId* id, void* data, SessionOnError on_error, int range = 1);
 
+// Diagnostics for the /ids builtin page (≙ reference builtin/ids_service):
+// process-lifetime created/destroyed counts (active = difference).
+void session_stats(int64_t* created, int64_t* destroyed);
+
 // Locks the session. Returns 0 and fills *data; EINVAL if destroyed/stale.
 int session_lock(SessionId id, void** data);
 int session_unlock(SessionId id);

@@ -17,6 +17,7 @@
 #include "base/iobuf.h"
 #include "base/time.h"
 #include "fiber/fiber.h"
+#include "fiber/session.h"
 #include <malloc.h>
 
 #include "rpc/policy/http_protocol.h"
@@ -41,7 +42,7 @@ void page_index(HttpResponse* resp) {
   std::ostringstream os;
   os << "<html><head><title>brpc_amd</title></head><body><h1>brpc_amd server</h1><ul>";
   const char* pages[] = {"status", "vars",   "flags",  "health",       "version",
-                         "connections", "sockets", "list", "dir", "protobufs", "fibers", "memory", "threads",
+                         "connections", "sockets", "list", "dir", "ids", "protobufs", "fibers", "memory", "threads",
                          "hotspots/cpu", "hotspots/contention", "hotspots/gpu",
                          "rpcz", "brpc_metrics"};
   for (const char* p : pages) os << "<li><a href=\"/" << p << "\">/" << p << "</a></li>";
@@ -306,6 +307,14 @@ bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse
     page_connections(resp);
   } else if (p == "/sockets") {
     page_sockets(req, resp);
+  } else if (p == "/ids") {
+    int64_t created = 0, destroyed = 0;
+    session_stats(&created, &destroyed);
+    std::ostringstream os;
+    os << "sessions_created: " << created << "\n"
+       << "sessions_destroyed: " << destroyed << "\n"
+       << "sessions_active: " << (created - destroyed) << "\n";
+    resp->body.append(os.str());
   } else if (p == "/list") {
     page_list(server, resp);
   } else if (p == "/dir") {

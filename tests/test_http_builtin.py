@@ -281,3 +281,11 @@ def test_master_handler_catch_all():
                                        "NoSuch.Service", b"raw-bytes", 3000, 0)
     assert rc == 0, err
     assert resp == b"master:NoSuch.Service:raw-bytes"
+
+
+def test_ids_page(base):
+    st, body = get(base + "/ids")
+    assert st == 200 and "sessions_created:" in body
+    created = int([l for l in body.splitlines()
+                   if l.startswith("sessions_created:")][0].split(":")[1])
+    assert created > 0  # the priming RPC used a session
