@@ -30,6 +30,8 @@
 namespace bam {
 std::string dump_fiber_stacks(int max_fibers);  // fiber/tracer.cc
 std::string CpuProfile(int seconds, int hz);    // rpc/cpu_profiler.cc
+std::string CpuProfileBinary(int seconds, int hz);
+std::string SymbolizeAddresses(const std::string& body);
 std::string ContentionProfile();                // rpc/cpu_profiler.cc
 namespace policy {
 
@@ -42,7 +44,7 @@ void page_index(HttpResponse* resp) {
   std::ostringstream os;
   os << "<html><head><title>brpc_amd</title></head><body><h1>brpc_amd server</h1><ul>";
   const char* pages[] = {"status", "vars",   "flags",  "health",       "version",
-                         "connections", "sockets", "list", "dir", "ids", "protobufs", "fibers", "memory", "threads",
+                         "connections", "sockets", "list", "dir", "ids", "pprof/profile", "pprof/symbol", "protobufs", "fibers", "memory", "threads",
                          "hotspots/cpu", "hotspots/contention", "hotspots/gpu",
                          "rpcz", "brpc_metrics"};
   for (const char* p : pages) os << "<li><a href=\"/" << p << "\">/" << p << "</a></li>";
@@ -327,6 +329,46 @@ bool DispatchBuiltinService(Server* server, const HttpRequest& req, HttpResponse
     page_memory(resp);
   } else if (p == "/threads") {
     page_threads(resp);
+  } else if (p == "/pprof/profile") {
+    // Remote pprof attach (≙ reference builtin/pprof_service.cpp):
+    // legacy gperftools binary CPU profile; `pprof http://host:port` works.
+    int seconds = 10;
+    auto it = req.query.find("seconds");
+    if (it != req.query.end()) seconds = atoi(it->second.c_str());
+    resp->content_type = "application/octet-stream";
+    resp->body.append(CpuProfileBinary(seconds, 100));
+  } else if (p == "/pprof/symbol") {
+    if (req.method == "POST" && req.body.size() > 0) {
+      resp->body.append(SymbolizeAddresses(req.body.to_string()));
+    } else {
+      resp->body.append("num_symbols: 1\n");
+    }
+  } else if (p == "/pprof/cmdline") {
+    FILE* f = fopen("/proc/self/cmdline", "r");
+    if (f != nullptr) {
+      char buf[4096];
+      size_t n = fread(buf, 1, sizeof(buf), f);
+      fclose(f);
+      resp->body.append(buf, n);
+    }
+  } else if (p == "/pprof/heap" || p == "/pprof/growth") {
+    // Minimal gperftools-text heap profile: the arena totals from
+    // mallinfo2 (glibc malloc — no tcmalloc in this image); detailed
+    // per-site allocation profiling is at /memory + the ASan builds.
+    struct mallinfo2 mi = mallinfo2();
+    std::ostringstream os;
+    os << "heap profile:      1:  " << mi.uordblks << " [     1:  "
+       << mi.uordblks << "] @ heap_v2/1\n";
+    os << "      1:  " << mi.uordblks << " [     1:  " << mi.uordblks
+       << "] @ 0x0\n\nMAPPED_LIBRARIES:\n";
+    FILE* f = fopen("/proc/self/maps", "r");
+    if (f != nullptr) {
+      char buf[8192];
+      size_t n;
+      while ((n = fread(buf, 1, sizeof(buf), f)) > 0) os.write(buf, n);
+      fclose(f);
+    }
+    resp->body.append(os.str());
   } else if (p == "/hotspots/cpu" || p == "/hotspots") {
     // parity: builtin/hotspots_service.cpp, self-contained SIGPROF sampler
     int seconds = 1;

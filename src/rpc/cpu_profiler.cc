@@ -117,15 +117,14 @@ std::string ContentionProfile() {
   return os.str();
 }
 
-// Samples the process for `seconds` (clamped to [1,30]) at `hz` and
-// returns a text report. Serializes concurrent profile requests.
-std::string CpuProfile(int seconds, int hz) {
+// Shared sampling pass: arms ITIMER_PROF for `seconds` at `hz` and fills
+// g_samples. Returns the sample count. Callers hold profile_mutex().
+static std::mutex& profile_mutex() {
   static std::mutex mu;
-  std::lock_guard<std::mutex> lk(mu);
-  if (seconds < 1) seconds = 1;
-  if (seconds > 30) seconds = 30;
-  if (hz < 10) hz = 10;
-  if (hz > 1000) hz = 1000;
+  return mu;
+}
+
+static int collect_cpu_samples(int seconds, int hz) {
   if (g_samples == nullptr) g_samples = new Sample[kMaxSamples];
   {
     // prime backtrace()'s lazy libgcc initialization outside the handler
@@ -152,8 +151,18 @@ std::string CpuProfile(int seconds, int hz) {
   memset(&it, 0, sizeof(it));
   setitimer(ITIMER_PROF, &it, nullptr);
   sigaction(SIGPROF, &old_sa, nullptr);
+  return std::min(g_sample_idx.load(std::memory_order_relaxed), kMaxSamples);
+}
 
-  const int n = std::min(g_sample_idx.load(std::memory_order_relaxed), kMaxSamples);
+// Samples the process for `seconds` (clamped to [1,30]) at `hz` and
+// returns a text report. Serializes concurrent profile requests.
+std::string CpuProfile(int seconds, int hz) {
+  std::lock_guard<std::mutex> lk(profile_mutex());
+  if (seconds < 1) seconds = 1;
+  if (seconds > 30) seconds = 30;
+  if (hz < 10) hz = 10;
+  if (hz > 1000) hz = 1000;
+  const int n = collect_cpu_samples(seconds, hz);
   std::map<std::string, std::pair<int, int>> agg;  // name -> {self, cumulative}
   for (int i = 0; i < n; ++i) {
     const Sample& s = g_samples[i];
@@ -185,6 +194,57 @@ std::string CpuProfile(int seconds, int hz) {
     if (++emitted >= 100) break;
   }
   if (n == 0) os << "(no samples: process mostly idle or blocked — ITIMER_PROF counts CPU time)\n";
+  return os.str();
+}
+
+
+// Legacy gperftools CPU-profile binary (what the pprof tool downloads from
+// /pprof/profile; parity: reference builtin/pprof_service.cpp serving
+// gperftools ProfilerStart output). Format: 8-byte LE words —
+// header [0, 3, 0, period_us, 0], samples [count, num_pcs, pcs...],
+// trailer [0, 1, 0].
+std::string CpuProfileBinary(int seconds, int hz) {
+  std::lock_guard<std::mutex> lk(profile_mutex());
+  if (seconds < 1) seconds = 1;
+  if (seconds > 30) seconds = 30;
+  if (hz < 10) hz = 10;
+  if (hz > 1000) hz = 1000;
+  const int n = collect_cpu_samples(seconds, hz);
+  std::string out;
+  auto put = [&out](uint64_t w) { out.append((const char*)&w, 8); };
+  put(0);
+  put(3);
+  put(0);
+  put((uint64_t)(1000000 / hz));
+  put(0);
+  for (int i = 0; i < n; ++i) {
+    const Sample& s = g_samples[i];
+    int depth = s.depth > 2 ? s.depth - 2 : 0;  // frames[0..1] = handler
+    if (depth == 0) continue;
+    put(1);
+    put((uint64_t)depth);
+    for (int d = 2; d < s.depth; ++d) put((uint64_t)(uintptr_t)s.frames[d]);
+  }
+  put(0);
+  put(1);
+  put(0);
+  return out;
+}
+
+// /pprof/symbol POST body: "0xaddr+0xaddr+..." -> "0xaddr\tname" lines.
+std::string SymbolizeAddresses(const std::string& body) {
+  std::ostringstream os;
+  size_t pos = 0;
+  while (pos < body.size()) {
+    size_t plus = body.find('+', pos);
+    std::string tok = body.substr(pos, plus == std::string::npos
+                                           ? std::string::npos : plus - pos);
+    pos = plus == std::string::npos ? body.size() : plus + 1;
+    while (!tok.empty() && (tok.back() == '\n' || tok.back() == '\r')) tok.pop_back();
+    if (tok.empty()) continue;
+    uintptr_t addr = (uintptr_t)strtoull(tok.c_str(), nullptr, 16);
+    os << tok << "\t" << frame_name((void*)addr) << "\n";
+  }
   return os.str();
 }
 

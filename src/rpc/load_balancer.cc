@@ -218,6 +218,108 @@ int ResolveNamingUrl(const std::string& url, std::vector<EndPoint>* out) {
     }
     return 0;
   }
+  if (url.rfind("dlist://", 0) == 0) {
+    // dlist://name1:port1,name2:port2 — like list:// but entries may be
+    // DNS names, re-resolved on every refresh (parity: reference
+    // DomainListNamingService, policy/list_naming_service.cpp:106).
+    std::string rest = url.substr(8);
+    size_t pos = 0;
+    while (pos < rest.size()) {
+      size_t comma = rest.find(',', pos);
+      std::string item = rest.substr(pos, comma == std::string::npos
+                                              ? std::string::npos : comma - pos);
+      pos = comma == std::string::npos ? rest.size() : comma + 1;
+      if (item.empty()) continue;
+      EndPoint ep;
+      if (str2endpoint(item.c_str(), &ep) == 0) {
+        out->push_back(ep);
+        continue;
+      }
+      size_t colon = item.rfind(':');
+      if (colon == std::string::npos) continue;
+      std::string host = item.substr(0, colon);
+      int port = atoi(item.c_str() + colon + 1);
+      struct addrinfo hints, *res = nullptr;
+      memset(&hints, 0, sizeof(hints));
+      hints.ai_family = AF_INET;
+      hints.ai_socktype = SOCK_STREAM;
+      if (getaddrinfo(host.c_str(), nullptr, &hints, &res) != 0) continue;
+      for (struct addrinfo* pp = res; pp != nullptr; pp = pp->ai_next) {
+        EndPoint e2;
+        e2.ip = ((struct sockaddr_in*)pp->ai_addr)->sin_addr;
+        e2.port = port;
+        out->push_back(e2);
+      }
+      freeaddrinfo(res);
+    }
+    return out->empty() ? -1 : 0;
+  }
+  if (url.rfind("discovery://", 0) == 0) {
+    // discovery://host:port/appid[?env=..&status=..] — bilibili discovery:
+    // GET /discovery/fetchs?appid=<appid>&env=<env>&status=<status>; JSON
+    // data.<appid>.instances[].addrs carries "scheme://ip:port" strings
+    // (parity: reference policy/discovery_naming_service.cpp:362-430).
+    std::string rest = url.substr(12);
+    size_t slash = rest.find('/');
+    if (slash == std::string::npos) return -1;
+    std::string agent = rest.substr(0, slash);
+    std::string appid = rest.substr(slash + 1);
+    std::string env = "prod", status = "1";
+    size_t q = appid.find('?');
+    if (q != std::string::npos) {
+      std::string params = appid.substr(q + 1);
+      appid = appid.substr(0, q);
+      size_t p2 = 0;
+      while (p2 < params.size()) {
+        size_t amp = params.find('&', p2);
+        std::string kv = params.substr(p2, amp == std::string::npos
+                                               ? std::string::npos : amp - p2);
+        p2 = amp == std::string::npos ? params.size() : amp + 1;
+        size_t eq = kv.find('=');
+        if (eq == std::string::npos) continue;
+        std::string k = kv.substr(0, eq), v = kv.substr(eq + 1);
+        if (k == "env") env = v;
+        else if (k == "status") status = v;
+      }
+    }
+    Channel ch;
+    ChannelOptions copt;
+    copt.protocol = "http";
+    copt.timeout_ms = 3000;
+    copt.max_retry = 0;
+    if (ch.Init(agent.c_str(), &copt) != 0) return -1;
+    Controller cntl;
+    IOBuf req, resp;
+    ch.CallMethod("/discovery/fetchs?appid=" + appid + "&env=" + env +
+                      "&status=" + status,
+                  &cntl, &req, &resp, nullptr);
+    if (cntl.Failed()) return -1;
+    json::Value root;
+    if (!json::Parse(resp.to_string(), &root) || root.type != json::Value::OBJECT)
+      return -1;
+    auto dit = root.obj->find("data");
+    if (dit == root.obj->end() || dit->second.type != json::Value::OBJECT) return -1;
+    auto ait = dit->second.obj->find(appid);
+    if (ait == dit->second.obj->end() || ait->second.type != json::Value::OBJECT)
+      return -1;
+    auto iit = ait->second.obj->find("instances");
+    if (iit == ait->second.obj->end() || iit->second.type != json::Value::ARRAY)
+      return -1;
+    for (const json::Value& inst : *iit->second.arr) {
+      if (inst.type != json::Value::OBJECT) continue;
+      auto adit = inst.obj->find("addrs");
+      if (adit == inst.obj->end() || adit->second.type != json::Value::ARRAY) continue;
+      for (const json::Value& a : *adit->second.arr) {
+        if (a.type != json::Value::STRING) continue;
+        std::string addr = a.str;
+        size_t sep = addr.find("://");
+        if (sep != std::string::npos) addr = addr.substr(sep + 3);
+        EndPoint ep;
+        if (str2endpoint(addr.c_str(), &ep) == 0) out->push_back(ep);
+      }
+    }
+    return 0;
+  }
   if (url.rfind("remotefile://", 0) == 0) {
     // remotefile://host:port/path — fetch a server-list file over HTTP
     // (parity: reference policy/remote_file_naming_service.cpp); same

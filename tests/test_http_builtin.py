@@ -289,3 +289,41 @@ def test_ids_page(base):
     created = int([l for l in body.splitlines()
                    if l.startswith("sessions_created:")][0].split(":")[1])
     assert created > 0  # the priming RPC used a session
+
+
+def test_pprof_endpoints(base):
+    """Remote pprof attach endpoints (≙ reference builtin/pprof_service.cpp):
+    /pprof/profile returns the legacy gperftools binary format (8-byte LE
+    words: [0,3,0,period,0] ... [0,1,0]); /pprof/symbol symbolizes
+    POSTed addresses; /pprof/cmdline is raw /proc/self/cmdline."""
+    import struct as _struct
+    with urllib.request.urlopen(base + "/pprof/profile?seconds=1",
+                                timeout=30) as resp:
+        prof = resp.read()
+    assert len(prof) >= 64 and len(prof) % 8 == 0
+    words = _struct.unpack("<%dQ" % (len(prof) // 8), prof)
+    assert words[0:3] == (0, 3, 0) and words[4] == 0
+    assert words[-3:] == (0, 1, 0)
+    # symbol: GET advertises support; POST symbolizes an address
+    st, body = get(base + "/pprof/symbol")
+    assert "num_symbols: 1" in body
+    addr = None
+    i = 5
+    while i + 1 < len(words) - 3:
+        count, depth = words[i], words[i + 1]
+        if count == 1 and depth > 0:
+            addr = words[i + 2]
+            break
+        i += 2 + depth
+    if addr:
+        req = urllib.request.Request(base + "/pprof/symbol",
+                                     data=("0x%x" % addr).encode(),
+                                     method="POST")
+        with urllib.request.urlopen(req, timeout=10) as resp:
+            sym = resp.read().decode()
+        assert ("0x%x" % addr) in sym and "\t" in sym
+    st, body = get(base + "/pprof/cmdline")
+    assert st == 200 and "python" in body
+    st, body = get(base + "/pprof/heap")
+    assert st == 200 and body.startswith("heap profile:")
+    assert "MAPPED_LIBRARIES:" in body

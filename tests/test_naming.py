@@ -84,3 +84,37 @@ def test_remotefile_naming():
         "remotefile://127.0.0.1:%d/lists/servers.txt" % agent_port)
     assert eps == ["127.0.0.1:%d" % echo_port] * 2, eps
     agent.stop()
+
+
+def test_dlist_naming():
+    """dlist:// (parity: reference DomainListNamingService,
+    policy/list_naming_service.cpp:106): entries may be DNS names,
+    resolved to (possibly several) A records on every refresh."""
+    eps = b.core.rpc.resolve_naming("dlist://localhost:8000,127.0.0.2:9000")
+    assert "127.0.0.1:8000" in eps, eps
+    assert "127.0.0.2:9000" in eps, eps
+
+
+def test_discovery_naming():
+    """discovery:// (parity: reference policy/discovery_naming_service.cpp
+    fetchs API): JSON data.<appid>.instances[].addrs with scheme prefixes
+    stripped."""
+    echo_port = r.start_echo_server(0)
+    agent = b.Server()
+
+    def fetchs(req, att):
+        body = json.dumps({"code": 0, "data": {"my.app": {"instances": [
+            {"addrs": ["grpc://127.0.0.1:%d" % echo_port,
+                       "http://127.0.0.1:%d" % (echo_port + 1)]},
+            {"addrs": ["127.0.0.1:%d" % echo_port]},
+        ]}}}).encode()
+        return body, b""
+
+    agent.add_method("disc", "fetchs", fetchs)
+    agent.add_restful_mapping("disc", "/discovery/fetchs => fetchs")
+    agent_port = agent.start(0)
+    eps = b.core.rpc.resolve_naming(
+        "discovery://127.0.0.1:%d/my.app?env=prod&status=1" % agent_port)
+    assert eps.count("127.0.0.1:%d" % echo_port) == 2, eps
+    assert "127.0.0.1:%d" % (echo_port + 1) in eps, eps
+    agent.stop()
