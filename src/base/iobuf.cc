@@ -650,6 +650,12 @@ int IOBuf::cut_until(IOBuf* out, const std::string& delim) {
   return -1;
 }
 
+bool IOBuf::has_residency(Residency res) const {
+  for (uint32_t i = 0; i < count_; ++i)
+    if (ref_at(i).block->res == res) return true;
+  return false;
+}
+
 ssize_t IOBuf::cut_into_file_descriptor(int fd, size_t size_hint) {
   if (empty()) return 0;
   struct iovec iov[kMaxIov];

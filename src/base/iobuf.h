@@ -104,6 +104,10 @@ class IOBuf {
   // payloads straight into place: RCCL recv, GPUDirect). n ≤ 2 GiB.
   int append_writable_block(size_t n, Residency res, int dev, void** out_ptr);
   int append_device_block(size_t n, int dev, void** out_ptr);  // = RES_HBM
+  // True if any block of `res` residency backs this buffer (used by the
+  // socket write path to route HBM payloads through the KeepWrite fiber
+  // so consecutive responses stage in ONE device gather).
+  bool has_residency(Residency res) const;
 
   // ---- cutting (front) ----
   // Move up to n bytes from the front of *this to the back of *out.

@@ -540,7 +540,16 @@ int Socket::Write(IOBuf* data, const WriteOptions* opt) {
   req->next.store(nullptr, std::memory_order_release);
   // We own the queue. Inline attempt only when connected (never block the
   // caller); otherwise hand to a KeepWrite fiber.
-  if (!connecting_.load(std::memory_order_acquire)) {
+  // HBM-resident payloads optionally skip the inline attempt: staging is a
+  // per-batch device gather (µs-scale), so routing through KeepWrite lets
+  // responses that complete close together coalesce into ONE kernel
+  // (BAM_DEFER_HBM_WRITE=0 restores inline; default chosen by same-box A/B).
+  static const int defer_hbm = [] {
+    const char* e = getenv("BAM_DEFER_HBM_WRITE");
+    return e == nullptr ? 1 : atoi(e);
+  }();
+  const bool hbm_defer = defer_hbm != 0 && req->data.has_residency(RES_HBM);
+  if (!connecting_.load(std::memory_order_acquire) && !hbm_defer) {
     ssize_t nw = write_bytes(&req->data, /*may_block=*/false);
     if (nw < 0 && errno != EAGAIN && errno != EWOULDBLOCK && errno != EINTR) {
       int err = errno;
