@@ -540,13 +540,15 @@ int Socket::Write(IOBuf* data, const WriteOptions* opt) {
   req->next.store(nullptr, std::memory_order_release);
   // We own the queue. Inline attempt only when connected (never block the
   // caller); otherwise hand to a KeepWrite fiber.
-  // HBM-resident payloads optionally skip the inline attempt: staging is a
-  // per-batch device gather (µs-scale), so routing through KeepWrite lets
-  // responses that complete close together coalesce into ONE kernel
-  // (BAM_DEFER_HBM_WRITE=0 restores inline; default chosen by same-box A/B).
+  // HBM-resident payloads CAN skip the inline attempt so that responses
+  // completing close together coalesce into one staging gather — but the
+  // same-box A/B (profiles/INDEX.md, r02) measured the handoff LOSING
+  // 1-7% at every point (hbm64 c32/64/128, hbm16k): completions do not
+  // cluster tightly enough to repay the fiber wake. Default OFF;
+  // BAM_DEFER_HBM_WRITE=1 re-enables for future re-measurement.
   static const int defer_hbm = [] {
     const char* e = getenv("BAM_DEFER_HBM_WRITE");
-    return e == nullptr ? 1 : atoi(e);
+    return e == nullptr ? 0 : atoi(e);
   }();
   const bool hbm_defer = defer_hbm != 0 && req->data.has_residency(RES_HBM);
   if (!connecting_.load(std::memory_order_acquire) && !hbm_defer) {
