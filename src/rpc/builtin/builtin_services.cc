@@ -40,15 +40,53 @@ namespace {
 int64_t g_start_time_us = monotonic_time_us();
 
 void page_index(HttpResponse* resp) {
+  // Dashboard (≙ reference /index with embedded flot plots; ours is a
+  // dependency-free canvas sparkline polling /vars once a second).
   resp->content_type = "text/html";
   std::ostringstream os;
-  os << "<html><head><title>brpc_amd</title></head><body><h1>brpc_amd server</h1><ul>";
+  os << "<html><head><title>brpc_amd</title><style>"
+        "body{font-family:monospace;margin:24px}"
+        "ul{columns:3;max-width:720px}canvas{border:1px solid #ccc}"
+        "#live{margin:12px 0;padding:8px;background:#f6f6f6}"
+        "</style></head><body><h1>brpc_amd server</h1>"
+        "<div id=live>loading /vars…</div>"
+        "<canvas id=plot width=600 height=120></canvas>"
+        "<div>requests/s (60 s window, 1 Hz poll of /vars)</div><ul>";
   const char* pages[] = {"status", "vars",   "flags",  "health",       "version",
                          "connections", "sockets", "list", "dir", "ids", "pprof/profile", "pprof/symbol", "protobufs", "fibers", "memory", "threads",
                          "hotspots/cpu", "hotspots/contention", "hotspots/gpu",
                          "rpcz", "brpc_metrics"};
   for (const char* p : pages) os << "<li><a href=\"/" << p << "\">/" << p << "</a></li>";
-  os << "</ul></body></html>";
+  os << "</ul><script>\n"
+        "const hist=[];\n"
+        "async function tick(){\n"
+        "  try{\n"
+        "    const t=await (await fetch('/vars')).text();\n"
+        "    const kv={};\n"
+        "    for(const line of t.split('\\n')){\n"
+        "      const i=line.indexOf(':');\n"
+        "      if(i>0)kv[line.slice(0,i).trim()]=line.slice(i+1).trim();\n"
+        "    }\n"
+        "    const qps=[];\n"
+        "    for(const k in kv)if(k.endsWith('_qps'))qps.push(k+'='+kv[k]);\n"
+        "    document.getElementById('live').textContent=\n"
+        "      qps.length?qps.join('  '):Object.keys(kv).length+' vars';\n"
+        "    let total=0;\n"
+        "    for(const k in kv)if(k.endsWith('_qps'))total+=parseFloat(kv[k])||0;\n"
+        "    hist.push(total);if(hist.length>60)hist.shift();\n"
+        "    const c=document.getElementById('plot'),g=c.getContext('2d');\n"
+        "    g.clearRect(0,0,c.width,c.height);\n"
+        "    const max=Math.max(1,...hist);\n"
+        "    g.beginPath();\n"
+        "    hist.forEach((v,i)=>{const x=i*c.width/60,y=c.height-4-v/max*(c.height-8);\n"
+        "      i?g.lineTo(x,y):g.moveTo(x,y);});\n"
+        "    g.strokeStyle='#06c';g.stroke();\n"
+        "    g.fillText(max.toFixed(0)+' qps peak',6,12);\n"
+        "  }catch(e){}\n"
+        "  setTimeout(tick,1000);\n"
+        "}\n"
+        "tick();\n"
+        "</script></body></html>";
   resp->body.append(os.str());
 }
 
