@@ -45,7 +45,9 @@ bool Parse(const std::string& data, std::vector<Tag>* out, bool* has_audio,
   if (has_audio != nullptr) *has_audio = (p[4] & 0x04) != 0;
   if (has_video != nullptr) *has_video = (p[4] & 0x01) != 0;
   uint32_t off = rd_u32(p + 5);
-  if (off < 9 || (size_t)(end - p) < off + 4) return false;
+  // uint64 sum: off + 4 must not wrap (a 0xffffffff data_offset slipped
+  // past the old uint32 check — found by fuzz_ts_flv).
+  if (off < 9 || (uint64_t)off + 4 > (uint64_t)(end - p)) return false;
   p += off;
   if (rd_u32(p) != 0) return false;  // PreviousTagSize0
   p += 4;

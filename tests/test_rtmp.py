@@ -112,3 +112,13 @@ def test_rtmp_play_remuxed_to_flv(port):
     pub.close()
     _, _, tags = r.flv_parse(result["flv"])
     assert [(ty, ts, p) for ty, ts, p in tags] == frames
+
+
+def test_flv_parse_data_offset_overflow():
+    """Regression (fuzz_ts_flv finding): a 0xffffffff FLV data_offset
+    wrapped the uint32 bounds check and read out of bounds."""
+    evil = bytes.fromhex("464c56017cffffffff00260a16")
+    try:
+        b.core.rpc.flv_parse(evil)
+    except RuntimeError:
+        pass  # malformed: rejected is the expected outcome
