@@ -2,7 +2,8 @@
 parser has a libFuzzer target under tests/fuzz/ built by `make fuzz` with
 ASan; this test builds them and runs each for a few seconds. Regression
 memo: fuzz_snappy found the preamble malloc-bomb fixed in base/snappy.cc;
-fuzz_mcpack found a parse_primitive stack overflow (type 0x3a)."""
+fuzz_mcpack found a parse_primitive stack overflow (type 0x3a);
+fuzz_ts_flv found the FLV data_offset uint32-overflow OOB read."""
 import os
 import subprocess
 
@@ -10,7 +11,7 @@ import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 TARGETS = ["fuzz_rpc_meta", "fuzz_http", "fuzz_redis", "fuzz_json", "fuzz_snappy",
-           "fuzz_mcpack", "fuzz_cut_until"]
+           "fuzz_mcpack", "fuzz_cut_until", "fuzz_hpack", "fuzz_ts_flv"]
 
 
 @pytest.mark.slow
