@@ -11,7 +11,7 @@ import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 TARGETS = ["fuzz_rpc_meta", "fuzz_http", "fuzz_redis", "fuzz_json", "fuzz_snappy",
-           "fuzz_mcpack", "fuzz_cut_until", "fuzz_hpack", "fuzz_ts_flv"]
+           "fuzz_mcpack", "fuzz_cut_until", "fuzz_hpack", "fuzz_ts_flv", "fuzz_h2"]
 
 
 @pytest.mark.slow
