@@ -17,6 +17,7 @@ bool semaphore_test();
 int rwlock_test(int nreaders, int nwriters, int iters);
 bool timer_test();
 bool fiber_key_test();
+bool fiber_interrupt_test(std::string* err);
 bool gpu_wait_selftest();
 int64_t fd_wait_selftest();
 bool stack_class_selftest();
@@ -43,6 +44,15 @@ void bind_fiber(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   f.def("timer_test", &bam::selftest::timer_test, py::call_guard<py::gil_scoped_release>());
   f.def("key_test", &bam::selftest::fiber_key_test, py::call_guard<py::gil_scoped_release>());
+  f.def("interrupt_test", []() {
+    std::string err;
+    bool ok;
+    {
+      py::gil_scoped_release rel;
+      ok = bam::selftest::fiber_interrupt_test(&err);
+    }
+    return py::make_tuple(ok, err);
+  });
   f.def("gpu_wait_test", &bam::selftest::gpu_wait_selftest,
         py::call_guard<py::gil_scoped_release>());
   f.def("fd_wait_test", &bam::selftest::fd_wait_selftest,

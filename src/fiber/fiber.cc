@@ -93,24 +93,98 @@ void sleep_wake_cb(void* a, void* /*b*/) {
 }
 }  // namespace
 
+namespace {
+// Bumps the sequence word and wakes the sleeper. The word is the meta's
+// persistent sleep butex, so no destroy hand-off is needed.
+void seq_sleep_wake_cb(void* a, void*) {
+  std::atomic<int>* word = (std::atomic<int>*)a;
+  word->fetch_add(1, std::memory_order_release);
+  butex_wake_all(word);
+}
+}  // namespace
+
 int fiber_usleep(uint64_t us) {
   TaskGroup* g = current_task_group();
   if (g == nullptr || g->cur() == nullptr || g->cur()->is_main) {
     usleep(us);
     return 0;
   }
-  std::atomic<int>* word = butex_create();
-  word->store(0, std::memory_order_relaxed);
-  timer_add(monotonic_time_us() + (int64_t)us, sleep_wake_cb, word, nullptr);
-  int v;
-  while ((v = word->load(std::memory_order_acquire)) < 1) {
-    butex_wait(word, v, nullptr);
+  FiberMeta* m = g->cur();
+  if (m->stop_requested.load(std::memory_order_acquire)) {
+    errno = ESTOP;
+    return -1;
   }
-  // The callback may still be inside butex_wake_all; it stores 2 when it is
-  // completely done touching the butex — only then is destroy safe.
-  while (word->load(std::memory_order_acquire) != 2) sched_yield();
-  butex_destroy(word);
+  std::atomic<int>* word = m->sleep_butex.load(std::memory_order_acquire);
+  if (word == nullptr) {
+    word = butex_create();
+    m->sleep_butex.store(word, std::memory_order_release);
+  }
+  const int start = word->load(std::memory_order_acquire);
+  TimerId tid = timer_add(monotonic_time_us() + (int64_t)us, seq_sleep_wake_cb,
+                          word, nullptr);
+  while (word->load(std::memory_order_acquire) == start) {
+    if (m->interrupted.load(std::memory_order_acquire) ||
+        m->stop_requested.load(std::memory_order_acquire)) {
+      break;
+    }
+    butex_wait(word, start, nullptr);
+  }
+  const bool intr = m->interrupted.exchange(false, std::memory_order_acq_rel);
+  const bool stop = m->stop_requested.load(std::memory_order_acquire);
+  if (word->load(std::memory_order_acquire) == start) {
+    // Timer has not fired: cancel it; if it is already in flight, wait for
+    // the bump so it cannot shorten this slot's NEXT sleep.
+    if (timer_delete(tid) != 0) {
+      while (word->load(std::memory_order_acquire) == start) sched_yield();
+    }
+  }
+  if (stop) {
+    errno = ESTOP;
+    return -1;
+  }
+  if (intr) {
+    errno = EINTR;
+    return -1;
+  }
   return 0;
+}
+
+namespace {
+// Resolves a live meta for tid (version must match). nullptr when ended.
+FiberMeta* live_meta_of(fiber_t tid) {
+  if (tid == 0) return nullptr;
+  FiberMeta* m = fiber_meta_of(tid);
+  if (m == nullptr) return nullptr;
+  std::atomic<int>* vb = m->version_butex;
+  if (vb == nullptr) return nullptr;
+  if ((uint32_t)vb->load(std::memory_order_acquire) != (uint32_t)(tid >> 32))
+    return nullptr;
+  return m;
+}
+}  // namespace
+
+int fiber_interrupt(fiber_t tid) {
+  FiberMeta* m = live_meta_of(tid);
+  if (m == nullptr) return EINVAL;
+  m->interrupted.store(true, std::memory_order_release);
+  std::atomic<int>* word = m->sleep_butex.load(std::memory_order_acquire);
+  if (word != nullptr) butex_wake_all(word);
+  return 0;
+}
+
+int fiber_stop(fiber_t tid) {
+  FiberMeta* m = live_meta_of(tid);
+  if (m == nullptr) return EINVAL;
+  m->stop_requested.store(true, std::memory_order_release);
+  std::atomic<int>* word = m->sleep_butex.load(std::memory_order_acquire);
+  if (word != nullptr) butex_wake_all(word);
+  return 0;
+}
+
+bool fiber_stop_requested() {
+  TaskGroup* g = current_task_group();
+  if (g == nullptr || g->cur() == nullptr || g->cur()->is_main) return false;
+  return g->cur()->stop_requested.load(std::memory_order_acquire);
 }
 
 fiber_t fiber_self() {

@@ -32,7 +32,19 @@ int fiber_start_background(fiber_t* tid, void (*fn)(void*), void* arg,
 
 // Waits for fiber termination. Returns 0; joining an ended/invalid id
 // returns 0 immediately.
+// errno value a stopped fiber's sleeps return (≙ reference bthread ESTOP).
+enum { ESTOP = 2017 };
+
 int fiber_join(fiber_t tid);
+// Wakes `tid` out of a fiber_usleep early: the sleep returns -1 with
+// errno=EINTR (≙ reference bthread_interrupt). No-op on finished ids.
+int fiber_interrupt(fiber_t tid);
+// Marks `tid` stopped and interrupts it: current and FUTURE fiber_usleep
+// calls return -1/ESTOP immediately (≙ bthread_stop). The fiber observes
+// it via the sleep result (or fiber_stop_requested()) and exits itself.
+int fiber_stop(fiber_t tid);
+// True if fiber_stop was called on the CALLING fiber.
+bool fiber_stop_requested();
 // True if the fiber still runs (diagnostics).
 bool fiber_exists(fiber_t tid);
 

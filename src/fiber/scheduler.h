@@ -54,6 +54,13 @@ struct FiberMeta {
   std::atomic<uint32_t> version{1};    // bumped at exit; fiber_t carries it
   std::atomic<int>* version_butex = nullptr;  // mirrors version; joiners wait here
   void* keytable = nullptr;                   // fiber-local storage (fiber/key.h)
+  // Sleep/interrupt state (≙ reference bthread_interrupt/bthread_stop,
+  // bthread/task_group.cpp interrupt paths): sleep_butex is lazily
+  // created once per meta SLOT and never destroyed (metas are pooled),
+  // so wakers can touch it without lifetime games.
+  std::atomic<std::atomic<int>*> sleep_butex{nullptr};
+  std::atomic<bool> interrupted{false};
+  std::atomic<bool> stop_requested{false};
   bool is_main = false;
 };
 

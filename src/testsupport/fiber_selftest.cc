@@ -4,6 +4,7 @@
 #include <errno.h>
 
 #include <atomic>
+#include <string>
 #include <vector>
 
 #include "base/fast_rand.h"
@@ -495,6 +496,82 @@ bool stack_class_selftest() {
   }
   for (fiber_t t : tids) fiber_join(t);
   return acc.load() == (int)(4 * 8 * 3);
+}
+
+
+// fiber_interrupt / fiber_stop semantics (≙ reference
+// bthread_interrupt/bthread_stop unittests in test/bthread_unittest.cpp):
+// (1) interrupt cuts a long sleep short with EINTR; (2) stop makes the
+// current AND subsequent sleeps return ESTOP immediately; (3) a full
+// sleep after an uninterrupted one still times out normally (the
+// persistent per-slot sleep butex must not leak wakes across sleeps).
+bool fiber_interrupt_test(std::string* err) {
+  struct State {
+    std::atomic<int> phase{0};
+    std::atomic<int64_t> t_interrupted{0};
+    std::atomic<int> rc1{0}, e1{0}, rc2{0}, e2{0}, rc3{0}, e3{0};
+  } st;
+  auto body = [](void* raw) {
+    State* s = (State*)raw;
+    int64_t t0 = monotonic_time_us();
+    int rc = fiber_usleep(5 * 1000 * 1000);  // interrupted early
+    s->rc1.store(rc);
+    s->e1.store(rc == -1 ? errno : 0);
+    s->t_interrupted.store(monotonic_time_us() - t0);
+    s->phase.store(1);
+    rc = fiber_usleep(20 * 1000);  // normal full sleep, must NOT be cut
+    s->rc2.store(rc);
+    s->e2.store(rc == -1 ? errno : 0);
+    s->phase.store(2);
+    while (s->phase.load() != 3) fiber_yield();  // wait for stop request
+    rc = fiber_usleep(5 * 1000 * 1000);  // stopped: immediate ESTOP
+    s->rc3.store(rc);
+    s->e3.store(rc == -1 ? errno : 0);
+  };
+  fiber_t th;
+  if (fiber_start_background(&th, body, &st) != 0) {
+    *err = "start failed";
+    return false;
+  }
+  usleep(50 * 1000);  // let it enter the long sleep
+  if (fiber_interrupt(th) != 0) {
+    *err = "interrupt failed";
+    return false;
+  }
+  int64_t give_up = monotonic_time_us() + 3 * 1000 * 1000;
+  while (st.phase.load() < 2 && monotonic_time_us() < give_up) usleep(1000);
+  if (st.phase.load() < 2) {
+    *err = "interrupted sleep did not finish";
+    return false;
+  }
+  if (st.rc1.load() != -1 || st.e1.load() != EINTR) {
+    *err = "sleep 1: want -1/EINTR, got " + std::to_string(st.rc1.load()) + "/" +
+           std::to_string(st.e1.load());
+    return false;
+  }
+  if (st.t_interrupted.load() > 2 * 1000 * 1000) {
+    *err = "interrupt took too long";
+    return false;
+  }
+  if (st.rc2.load() != 0) {
+    *err = "sleep 2 (uninterrupted) failed: errno " + std::to_string(st.e2.load());
+    return false;
+  }
+  if (fiber_stop(th) != 0) {
+    *err = "stop failed";
+    return false;
+  }
+  st.phase.store(3);
+  if (fiber_join(th) != 0) {
+    *err = "join failed";
+    return false;
+  }
+  if (st.rc3.load() != -1 || st.e3.load() != ESTOP) {
+    *err = "sleep 3: want -1/ESTOP, got " + std::to_string(st.rc3.load()) + "/" +
+           std::to_string(st.e3.load());
+    return false;
+  }
+  return true;
 }
 
 }  // namespace selftest

@@ -113,3 +113,11 @@ def test_stack_size_classes():
     """FIBER_ATTR_SMALL/NORMAL/LARGE pooled stack classes (≙ reference
     BTHREAD_ATTR_* stack sizes, bthread/stack_inl.h)."""
     assert f.stack_class_test()
+
+
+def test_fiber_interrupt_and_stop():
+    """fiber_interrupt cuts a sleep short with EINTR; fiber_stop makes
+    sleeps return ESTOP immediately (≙ reference bthread_interrupt /
+    bthread_stop, bthread/bthread.h)."""
+    ok, err = f.interrupt_test()
+    assert ok, err
