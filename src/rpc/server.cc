@@ -1,5 +1,10 @@
 #include "rpc/server.h"
 
+#include <signal.h>
+#include <string.h>
+
+#include <mutex>
+
 #include "rpc/ssl_util.h"
 #include "fiber/butex.h"
 #include "fiber/gpu_wait.h"
@@ -275,6 +280,25 @@ var::LatencyRecorder* Server::method_status(const std::string& service,
 int Server::Join() {
   while (concurrency.load(std::memory_order_acquire) > 0) usleep(1000);
   return 0;
+}
+
+static std::atomic<bool> g_asked_to_quit{false};
+static void quit_handler(int) { g_asked_to_quit.store(true, std::memory_order_release); }
+
+bool Server::IsAskedToQuit() { return g_asked_to_quit.load(std::memory_order_acquire); }
+
+void Server::RunUntilAskedToQuit() {
+  static std::once_flag once;
+  std::call_once(once, [] {
+    struct sigaction sa;
+    memset(&sa, 0, sizeof(sa));
+    sa.sa_handler = quit_handler;
+    sigaction(SIGINT, &sa, nullptr);
+    sigaction(SIGTERM, &sa, nullptr);
+  });
+  while (!IsAskedToQuit()) usleep(100 * 1000);
+  Stop(0);
+  Join();
 }
 
 }  // namespace bam

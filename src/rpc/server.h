@@ -142,6 +142,11 @@ class Server {
   int Start(const EndPoint& ep, const ServerOptions* opt);
   int Stop(int wait_ms = 0);
   int Join();
+  // Blocks until SIGINT/SIGTERM (≙ reference server.cpp:1895
+  // RunUntilAskedToQuit + IsAskedToQuit): installs the quit handler on
+  // first use, sleeps in 100 ms ticks, then Stop()+Join().
+  void RunUntilAskedToQuit();
+  static bool IsAskedToQuit();
 
   bool IsRunning() const { return running_.load(std::memory_order_acquire); }
   EndPoint listen_address() const { return listen_ep_; }
