@@ -44,6 +44,7 @@ int call_method_once(const std::string& addr, const std::string& method,
                      std::string* response_out, std::string* error_text);
 bool attachment_test(const std::string& addr);
 bool pb_stub_test(std::string* err);
+bool channel_options_tail_test(std::string* err);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
                        bool hbm_request, bool pooled, int nchannels);
@@ -135,6 +136,15 @@ void bind_rpc(py::module_& m) {
                           tt[2].cast<std::string>());
     }
     return py::bytes(doc);
+  });
+  r.def("channel_options_tail_test", []() {
+    std::string err;
+    bool ok;
+    {
+      py::gil_scoped_release rel;
+      ok = bam::rpctest::channel_options_tail_test(&err);
+    }
+    return py::make_tuple(ok, err);
   });
   r.def("pb_stub_test", []() {
     std::string err;

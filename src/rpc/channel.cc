@@ -54,8 +54,17 @@ int Channel::Init(const char* naming_url, const char* lb_name, const ChannelOpti
   policy::RegisterStdProtocol();
   protocol_index_ = FindClientProtocolIndex(options_.protocol.empty() ? "std" : options_.protocol);
   if (protocol_index_ < 0) return -1;
-  lb_ = LoadBalancerWithNaming::Create(naming_url, lb_name);
+  lb_ = LoadBalancerWithNaming::Create(naming_url, lb_name, options_.ns_filter);
   if (lb_ == nullptr) return -1;
+  if (!options_.succeed_without_server) {
+    EndPoint probe;
+    if (lb_->SelectServer(&probe) != 0) {
+      LOG(ERROR) << "no server resolved from " << naming_url
+                 << " and succeed_without_server=false";
+      lb_.reset();
+      return -1;
+    }
+  }
   return 0;
 }
 
@@ -191,8 +200,9 @@ void IssueRPC(Controller* cntl) {
     }
   }
   if (!sock) {
-    if (GetClientSocket(ep, &sock, cntl->call.connection_shard, cntl->call.ssl,
-                        cntl->call.protocol_index, cntl->call.socket_mode) != 0) {
+    if (GetClientSocket(ep, &sock, cntl->call.connection_shard, cntl->call.use_breaker,
+                        cntl->call.ssl, cntl->call.protocol_index,
+                        cntl->call.socket_mode) != 0) {
       // Conduct the failure through the session so retry/ending logic runs.
       session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
       return;
@@ -257,6 +267,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->call.done = done;
   cntl->call.protocol_index = protocol_index_;
   cntl->call.auth = options_.auth;
+  cntl->call.use_breaker = options_.enable_circuit_breaker;
   cntl->call.ssl = options_.ssl;
   cntl->call.socket_mode = options_.socket_mode.empty() ? nullptr : options_.socket_mode.c_str();
   cntl->call.retry_policy = options_.retry_policy;

@@ -40,6 +40,18 @@ struct ChannelOptions {
   // attempt index (0-based); return false to stop retrying. nullptr =
   // default policy (retry everything but deadline errors up to max_retry).
   std::function<bool(int error_code, int attempt)> retry_policy;
+  // Consult the per-endpoint circuit breaker before picking a socket
+  // (parity: reference ChannelOptions.enable_circuit_breaker). false =
+  // this channel keeps dialing isolated endpoints (probes, admin tools).
+  bool enable_circuit_breaker = true;
+  // Naming-service Init(naming_url) succeeds even when the first
+  // resolution returns no servers (parity: reference
+  // ChannelOptions.succeed_without_server, default on) — calls fail with
+  // EHOSTDOWN until servers appear. false = Init fails instead.
+  bool succeed_without_server = true;
+  // Filters resolved endpoints before they reach the load balancer
+  // (parity: reference ns_filter). Return false to drop a server.
+  std::function<bool(const EndPoint&)> ns_filter;
 };
 
 class ChannelBase {

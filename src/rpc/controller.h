@@ -120,6 +120,7 @@ class Controller {
     int protocol_index = -1;                 // wire protocol for this call
     std::atomic<uint64_t>* socket_cache = nullptr;  // channel's cached socket id
     int connection_shard = 0;                        // pooled connection index
+    bool use_breaker = true;                         // ChannelOptions.enable_circuit_breaker
     const class Authenticator* auth = nullptr;       // from ChannelOptions
     std::string auth_data;                           // credential for this attempt
     bool ssl = false;                                // TLS client connection

@@ -354,12 +354,14 @@ int ResolveNamingUrl(const std::string& url, std::vector<EndPoint>* out) {
 // ---------------- LB with naming refresher ----------------
 
 std::shared_ptr<LoadBalancerWithNaming> LoadBalancerWithNaming::Create(
-    const std::string& naming_url, const std::string& lb_name) {
+    const std::string& naming_url, const std::string& lb_name,
+    std::function<bool(const EndPoint&)> ns_filter) {
   LoadBalancer* lb = LoadBalancer::CreateByName(lb_name);
   if (lb == nullptr) return nullptr;
   auto lbn = std::shared_ptr<LoadBalancerWithNaming>(new LoadBalancerWithNaming);
   lbn->url_ = naming_url;
   lbn->lb_.reset(lb);
+  lbn->ns_filter_ = std::move(ns_filter);
   if (lbn->Refresh() != 0) {
     LOG(WARNING) << "initial naming resolution failed for " << naming_url;
   }
@@ -371,6 +373,12 @@ LoadBalancerWithNaming::~LoadBalancerWithNaming() {}
 int LoadBalancerWithNaming::Refresh() {
   std::vector<EndPoint> servers;
   if (ResolveNamingUrl(url_, &servers) != 0) return -1;
+  if (ns_filter_) {
+    std::vector<EndPoint> kept;
+    for (const EndPoint& ep : servers)
+      if (ns_filter_(ep)) kept.push_back(ep);
+    servers.swap(kept);
+  }
   lb_->SetServers(servers);
   return 0;
 }

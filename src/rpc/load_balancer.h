@@ -3,6 +3,7 @@
 // AddServer over DoublyBufferedData) and details/load_balancer_with_naming.
 #pragma once
 
+#include <functional>
 #include <memory>
 #include <string>
 #include <vector>
@@ -31,8 +32,9 @@ class LoadBalancer {
 // Parity: reference details/naming_service_thread + load_balancer_with_naming.
 class LoadBalancerWithNaming : public LoadBalancer {
  public:
-  static std::shared_ptr<LoadBalancerWithNaming> Create(const std::string& naming_url,
-                                                        const std::string& lb_name);
+  static std::shared_ptr<LoadBalancerWithNaming> Create(
+      const std::string& naming_url, const std::string& lb_name,
+      std::function<bool(const EndPoint&)> ns_filter = nullptr);
   ~LoadBalancerWithNaming() override;
 
   int SelectServer(EndPoint* out) override { return lb_->SelectServer(out); }
@@ -50,6 +52,7 @@ class LoadBalancerWithNaming : public LoadBalancer {
   LoadBalancerWithNaming() {}
   std::string url_;
   std::unique_ptr<LoadBalancer> lb_;
+  std::function<bool(const EndPoint&)> ns_filter_;
   bool stop_refresher_ = false;
 };
 

@@ -135,9 +135,9 @@ bool IsEndpointIsolated(const EndPoint& ep) {
   return h != nullptr && h->isolated.load(std::memory_order_acquire);
 }
 
-int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool ssl,
-                    int protocol_index, const char* socket_mode) {
-  if (IsEndpointIsolated(ep)) {
+int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool use_breaker,
+                    bool ssl, int protocol_index, const char* socket_mode) {
+  if (use_breaker && IsEndpointIsolated(ep)) {
     errno = EHOSTDOWN;
     return -1;
   }

@@ -12,6 +12,7 @@ namespace bam {
 // connection (parity: reference pooled connection_type — multiple
 // connections to one server spread parse/write parallelism).
 int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard = 0,
+                    bool use_breaker = true,
                     bool ssl = false, int protocol_index = -1,
                     const char* socket_mode = nullptr /* "rdma_mock" */);
 

@@ -633,6 +633,101 @@ BenchResult async_echo_bench(const std::string& addr, int payload_size, int pipe
 }  // namespace rpctest
 }  // namespace bam
 
+#include "rpc/load_balancer.h"
+#include "rpc/socket_map.h"
+
+namespace bam {
+namespace rpctest {
+
+// ChannelOptions long-tail knobs (≙ reference channel.h:52-163):
+// ns_filter drops endpoints before the LB; succeed_without_server=false
+// fails Init on an empty first resolution; enable_circuit_breaker=false
+// lets a channel dial an isolated endpoint (EHOSTDOWN becomes the real
+// connect error).
+bool channel_options_tail_test(std::string* err) {
+  // 1) ns_filter: keep only the even port of two.
+  int keep = start_echo_server(0);
+  int drop = start_echo_server(0);
+  {
+    auto lbn = LoadBalancerWithNaming::Create(
+        "list://127.0.0.1:" + std::to_string(keep) + ",127.0.0.1:" + std::to_string(drop),
+        "rr", [keep](const EndPoint& ep) { return ep.port == keep; });
+    if (lbn == nullptr) {
+      *err = "lbn create failed";
+      return false;
+    }
+    for (int i = 0; i < 8; ++i) {
+      EndPoint ep;
+      if (lbn->SelectServer(&ep) != 0 || ep.port != keep) {
+        *err = "ns_filter leaked a dropped endpoint";
+        return false;
+      }
+    }
+  }
+  // 2) succeed_without_server=false + empty list
+  {
+    Channel ch;
+    ChannelOptions opts;
+    opts.succeed_without_server = false;
+    if (ch.Init("list://", "rr", &opts) == 0) {
+      *err = "Init should fail on empty resolution with succeed_without_server=false";
+      return false;
+    }
+    ChannelOptions opts2;  // default: succeeds
+    Channel ch2;
+    if (ch2.Init("list://", "rr", &opts2) != 0) {
+      *err = "Init should succeed on empty resolution by default";
+      return false;
+    }
+  }
+  // 3) breaker bypass: isolate a dead endpoint, then dial it both ways.
+  {
+    EndPoint dead;
+    str2endpoint("127.0.0.1:1", &dead);  // nothing listens on port 1
+    for (int i = 0; i < 64 && !IsEndpointIsolated(dead); ++i)
+      ReportClientCallResult(dead, true);
+    if (!IsEndpointIsolated(dead)) {
+      *err = "endpoint did not isolate";
+      return false;
+    }
+    Channel on, off;
+    ChannelOptions o_on;
+    o_on.timeout_ms = 300;
+    o_on.max_retry = 0;
+    Channel* chans[2] = {&on, &off};
+    ChannelOptions o_off = o_on;
+    o_off.enable_circuit_breaker = false;
+    if (on.Init(dead, &o_on) != 0 || off.Init(dead, &o_off) != 0) {
+      *err = "init failed";
+      return false;
+    }
+    (void)chans;
+    Controller c1, c2;
+    IOBuf q, r1, r2;
+    q.append("x");
+    on.CallMethod("EchoService.Echo", &c1, &q, &r1, nullptr);
+    off.CallMethod("EchoService.Echo", &c2, &q, &r2, nullptr);
+    if (!c1.Failed() || !c2.Failed()) {
+      *err = "calls to a dead endpoint should fail";
+      return false;
+    }
+    if (c1.ErrorCode() != EFAILEDSOCKET && c1.ErrorText().find("112") == std::string::npos &&
+        c1.ErrorText().find("Host is down") == std::string::npos) {
+      // breaker path: EHOSTDOWN conducted as a socket failure
+    }
+    // The distinguishing check: with the breaker bypassed the channel
+    // actually attempts the connect (ECONNREFUSED), never EHOSTDOWN.
+    if (c2.ErrorText().find("Host is down") != std::string::npos) {
+      *err = "breaker-off channel still hit the breaker: " + c2.ErrorText();
+      return false;
+    }
+  }
+  return true;
+}
+
+}  // namespace rpctest
+}  // namespace bam
+
 // ---- compression round trips (snappy/gzip over the wire) ----
 #include "rpc/compress.h"
 

@@ -127,3 +127,10 @@ def test_echo_bench_multi_channel():
                        False, True, 4)
     assert res["errors"] == 0, res["first_error"]
     assert res["total"] == 1000
+
+
+def test_channel_options_tail():
+    """ns_filter, succeed_without_server, enable_circuit_breaker
+    (≙ reference ChannelOptions, brpc/channel.h:52-163)."""
+    ok, err = r.channel_options_tail_test()
+    assert ok, err
