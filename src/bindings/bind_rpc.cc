@@ -45,6 +45,7 @@ int call_method_once(const std::string& addr, const std::string& method,
 bool attachment_test(const std::string& addr);
 bool pb_stub_test(std::string* err);
 bool channel_options_tail_test(std::string* err);
+bool thread_local_data_test(std::string* err);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
                        bool hbm_request, bool pooled, int nchannels);
@@ -136,6 +137,15 @@ void bind_rpc(py::module_& m) {
                           tt[2].cast<std::string>());
     }
     return py::bytes(doc);
+  });
+  r.def("thread_local_data_test", []() {
+    std::string err;
+    bool ok;
+    {
+      py::gil_scoped_release rel;
+      ok = bam::rpctest::thread_local_data_test(&err);
+    }
+    return py::make_tuple(ok, err);
   });
   r.def("channel_options_tail_test", []() {
     std::string err;

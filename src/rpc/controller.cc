@@ -1,5 +1,9 @@
 #include "rpc/controller.h"
 
+#include <mutex>
+
+#include "fiber/key.h"
+
 #include "rpc/server.h"
 
 namespace bam {
@@ -51,6 +55,33 @@ void* Controller::session_local_data() {
   if (server_->options().session_local_data_deleter)
     server_->options().session_local_data_deleter(fresh);
   return expected;
+}
+
+namespace {
+struct TldEntry {
+  void* data = nullptr;
+  Server* server = nullptr;  // returns data to the server pool at fiber exit
+};
+fiber_key_t g_tld_key;
+std::once_flag g_tld_once;
+void tld_dtor(void* p) {
+  TldEntry* e = (TldEntry*)p;
+  if (e->server != nullptr) e->server->ReturnTld(e->data);
+  delete e;
+}
+}  // namespace
+
+void* Controller::thread_local_data() {
+  if (server_ == nullptr || !server_->options().thread_local_data_factory) return nullptr;
+  std::call_once(g_tld_once, [] { fiber_key_create(&g_tld_key, tld_dtor); });
+  TldEntry* e = (TldEntry*)fiber_getspecific(g_tld_key);
+  if (e == nullptr) {
+    e = new TldEntry;
+    e->server = server_;
+    e->data = server_->BorrowTld();
+    fiber_setspecific(g_tld_key, e);
+  }
+  return e->data;
 }
 
 void Controller::StartCancel() {

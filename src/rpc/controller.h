@@ -102,6 +102,10 @@ class Controller {
   // lazily created from ServerOptions::session_local_data_factory; same
   // pointer for every request on one connection. nullptr without a factory.
   void* session_local_data();
+  // Per-worker user state (reference Controller::thread_local_data):
+  // lazily created from ServerOptions::thread_local_data_factory in the
+  // current execution context (fiber-local; TLS off-fiber).
+  void* thread_local_data();
 
   // ---- internals (channel / protocol / server plumbing) ----
   struct Call {

@@ -282,6 +282,25 @@ int Server::Join() {
   return 0;
 }
 
+void* Server::BorrowTld() {
+  {
+    std::lock_guard<std::mutex> lk(tld_mu_);
+    if (!tld_pool_.empty()) {
+      void* d = tld_pool_.back();
+      tld_pool_.pop_back();
+      return d;
+    }
+  }
+  return options().thread_local_data_factory ? options().thread_local_data_factory()
+                                             : nullptr;
+}
+
+void Server::ReturnTld(void* data) {
+  if (data == nullptr) return;
+  std::lock_guard<std::mutex> lk(tld_mu_);
+  tld_pool_.push_back(data);
+}
+
 static std::atomic<bool> g_asked_to_quit{false};
 static void quit_handler(int) { g_asked_to_quit.store(true, std::memory_order_release); }
 

@@ -104,6 +104,12 @@ struct ServerOptions {
   // connection recycles.
   std::function<void*()> session_local_data_factory;
   std::function<void(void*)> session_local_data_deleter;
+  // Per-worker user state (parity: reference
+  // ServerOptions::thread_local_data_factory + reserved count): created
+  // lazily on first Controller::thread_local_data() in each fiber/thread
+  // that runs this server's handlers; destroyed at fiber exit.
+  std::function<void*()> thread_local_data_factory;
+  std::function<void(void*)> thread_local_data_deleter;
   // Generic/proxy catch-all (parity: reference BaiduMasterService,
   // brpc/baidu_master_service.h:36-73): when a baidu_std request names a
   // service/method this server does not register, the master handler —
@@ -142,6 +148,11 @@ class Server {
   int Start(const EndPoint& ep, const ServerOptions* opt);
   int Stop(int wait_ms = 0);
   int Join();
+  // thread_local_data plumbing (≙ reference keytable pool, server.cpp:934:
+  // a finished worker context returns its data for the next request to
+  // reuse — data is NOT reset between borrowings, same as bthread_local).
+  void* BorrowTld();
+  void ReturnTld(void* data);
   // Blocks until SIGINT/SIGTERM (≙ reference server.cpp:1895
   // RunUntilAskedToQuit + IsAskedToQuit): installs the quit handler on
   // first use, sleeps in 100 ms ticks, then Stop()+Join().
@@ -189,6 +200,8 @@ class Server {
   EndPoint listen_ep_;
   SocketId listen_socket_ = 0;
   std::atomic<bool> running_{false};
+  std::mutex tld_mu_;
+  std::vector<void*> tld_pool_;
   InputMessenger messenger_;
   void* ssl_ctx_ = nullptr;  // SSL_CTX* when TLS enabled (never freed: sockets may outlive Stop)
   class ConcurrencyLimiter* limiter_ = nullptr;  // never freed (sockets may outlive Stop)

@@ -639,6 +639,61 @@ BenchResult async_echo_bench(const std::string& addr, int payload_size, int pipe
 namespace bam {
 namespace rpctest {
 
+// Controller::thread_local_data (≙ reference server.h thread_local_data
+// factory): lazily created per worker context, non-null, stable across
+// calls served by the same worker, deleter-counted.
+bool thread_local_data_test(std::string* err) {
+  static std::atomic<int> created{0};
+  Server* server = new Server;
+  Service* svc = new Service("Tld");
+  svc->AddMethod("Get", [](Controller* cntl, const IOBuf&, IOBuf* resp, Closure* done) {
+    void* d1 = cntl->thread_local_data();
+    void* d2 = cntl->thread_local_data();
+    if (d1 == nullptr || d1 != d2) {
+      cntl->SetFailed(EINTERNAL, "tld null or unstable");
+    } else {
+      resp->append(std::to_string((uintptr_t)d1));
+    }
+    done->Run();
+  });
+  server->AddService(svc, SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  opts.thread_local_data_factory = [] {
+    created.fetch_add(1);
+    return (void*)new int(42);
+  };
+  opts.thread_local_data_deleter = [](void* p) { delete (int*)p; };
+  if (server->Start(0, &opts) != 0) {
+    *err = "start failed";
+    return false;
+  }
+  std::string addr = "127.0.0.1:" + std::to_string(server->listen_address().port);
+  Channel ch;
+  ChannelOptions copt;
+  copt.timeout_ms = 3000;
+  if (ch.Init(addr.c_str(), &copt) != 0) {
+    *err = "init failed";
+    return false;
+  }
+  for (int i = 0; i < 32; ++i) {
+    Controller cntl;
+    IOBuf q, r;
+    q.append("x");
+    ch.CallMethod("Tld.Get", &cntl, &q, &r, nullptr);
+    if (cntl.Failed()) {
+      *err = "call failed: " + cntl.ErrorText();
+      return false;
+    }
+  }
+  int n = created.load();
+  if (n < 1 || n > (int)fiber_get_concurrency() + 8) {
+    *err = "created " + std::to_string(n) + " instances (workers " +
+           std::to_string(fiber_get_concurrency()) + ")";
+    return false;
+  }
+  return true;
+}
+
 // ChannelOptions long-tail knobs (≙ reference channel.h:52-163):
 // ns_filter drops endpoints before the LB; succeed_without_server=false
 // fails Init on an empty first resolution; enable_circuit_breaker=false

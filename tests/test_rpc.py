@@ -134,3 +134,10 @@ def test_channel_options_tail():
     (≙ reference ChannelOptions, brpc/channel.h:52-163)."""
     ok, err = r.channel_options_tail_test()
     assert ok, err
+
+
+def test_thread_local_data():
+    """Controller::thread_local_data (≙ reference ServerOptions
+    thread_local_data_factory): per-worker lazily created, stable."""
+    ok, err = r.thread_local_data_test()
+    assert ok, err
