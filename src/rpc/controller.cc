@@ -1,5 +1,7 @@
 #include "rpc/controller.h"
 
+#include <ctype.h>
+
 #include <mutex>
 
 #include "fiber/key.h"
@@ -82,6 +84,31 @@ void* Controller::thread_local_data() {
     fiber_setspecific(g_tld_key, e);
   }
   return e->data;
+}
+
+static std::string lower(const std::string& s) {
+  std::string o = s;
+  for (char& c : o) c = (char)tolower((unsigned char)c);
+  return o;
+}
+
+void HttpHeaderExt::SetHeader(const std::string& k, const std::string& v) {
+  headers[lower(k)] = v;
+}
+
+const std::string* HttpHeaderExt::GetHeader(const std::string& k) const {
+  auto it = headers.find(lower(k));
+  return it == headers.end() ? nullptr : &it->second;
+}
+
+HttpHeaderExt& Controller::http_request() {
+  if (http_request_ == nullptr) http_request_.reset(new HttpHeaderExt);
+  return *http_request_;
+}
+
+HttpHeaderExt& Controller::http_response() {
+  if (http_response_ == nullptr) http_response_.reset(new HttpHeaderExt);
+  return *http_response_;
 }
 
 void Controller::StartCancel() {

@@ -8,6 +8,8 @@
 
 #include <stdint.h>
 
+#include <map>
+#include <memory>
 #include <string>
 
 #include "base/endpoint.h"
@@ -28,6 +30,21 @@ enum CompressType {
 
 class Channel;
 class Server;
+
+// Minimal HTTP header view for http/h2 channels (≙ reference brpc
+// HttpHeader used by Controller::http_request()/http_response(),
+// controller.h:400-430): verb/content-type/status plus a lower-cased
+// header map.
+struct HttpHeaderExt {
+  std::string method;        // client request: verb override (PUT, DELETE…)
+  std::string content_type;  // both directions
+  int status_code = 0;       // response side
+  std::map<std::string, std::string> headers;  // lower-cased names
+
+  void SetHeader(const std::string& k, const std::string& v);
+  // nullptr when absent; name lookup is case-insensitive.
+  const std::string* GetHeader(const std::string& k) const;
+};
 
 class Controller {
  public:
@@ -107,6 +124,14 @@ class Controller {
   // current execution context (fiber-local; TLS off-fiber).
   void* thread_local_data();
 
+  // HTTP-specific request/response views (http/h2 protocol channels and
+  // server handlers): lazily allocated; has_* avoids the allocation when
+  // only probing.
+  HttpHeaderExt& http_request();
+  HttpHeaderExt& http_response();
+  bool has_http_request() const { return http_request_ != nullptr; }
+  bool has_http_response() const { return http_response_ != nullptr; }
+
   // ---- internals (channel / protocol / server plumbing) ----
   struct Call {
     SessionId cid = 0;              // whole-call session
@@ -157,6 +182,8 @@ class Controller {
   int max_retry_ = 3;
   int retry_count_ = 0;
   uint64_t log_id_ = 0;
+  std::unique_ptr<HttpHeaderExt> http_request_;
+  std::unique_ptr<HttpHeaderExt> http_response_;
   uint64_t trace_id_ = 0;
   uint64_t span_id_ = 0;
   uint64_t parent_span_id_ = 0;

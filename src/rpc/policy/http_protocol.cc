@@ -380,7 +380,12 @@ void ProcessHttpResponse(InputMessageBase* mb) {
   }
   Controller* cntl = (Controller*)data;
   sock->remove_pending_session(cid);
-  if (msg->status != 200) {
+  HttpHeaderExt& hr = cntl->http_response();
+  hr.status_code = msg->status;
+  for (const auto& kv : msg->headers) hr.headers[kv.first] = kv.second;
+  auto ctit = msg->headers.find("content-type");
+  if (ctit != msg->headers.end()) hr.content_type = ctit->second;
+  if (msg->status / 100 != 2) {  // any 2xx is success (204, 202…)
     cntl->SetFailed(EHTTP, "HTTP status " + std::to_string(msg->status) + ": " +
                                msg->body.to_string().substr(0, 200));
   } else if (cntl->call.response != nullptr) {
@@ -400,14 +405,26 @@ void PackHttp1Request(IOBuf* out, Controller* cntl, uint64_t /*fifo-correlated*/
                      : svc.empty()                   ? "/" + m
                                                      : "/" + svc + "/" + m;
   const IOBuf& body = cntl->call.request_buf;
+  // Controller::http_request() overrides (≙ reference HttpHeader on the
+  // client side): custom verb, content-type and extra headers.
+  const HttpHeaderExt* hx = cntl->has_http_request() ? &cntl->http_request() : nullptr;
+  std::string verb = hx != nullptr && !hx->method.empty()
+                         ? hx->method
+                         : (body.empty() ? "GET" : "POST");
   std::string head;
   head.reserve(256);
-  head += (body.empty() ? "GET " : "POST ") + path + " HTTP/1.1\r\n";
+  head += verb + " " + path + " HTTP/1.1\r\n";
   head += "Host: " + endpoint2str(cntl->remote_side()) + "\r\n";
   head += "User-Agent: brpc-amd/1.0\r\n";
   head += "Accept: */*\r\n";
-  if (!body.empty()) {
-    head += "Content-Type: application/octet-stream\r\n";
+  if (hx != nullptr) {
+    for (const auto& kv : hx->headers) head += kv.first + ": " + kv.second + "\r\n";
+  }
+  if (!body.empty() || (hx != nullptr && !hx->content_type.empty())) {
+    head += "Content-Type: " +
+            (hx != nullptr && !hx->content_type.empty() ? hx->content_type
+                                                        : "application/octet-stream") +
+            "\r\n";
     head += "Content-Length: " + std::to_string(body.size()) + "\r\n";
   }
   head += "\r\n";
@@ -537,6 +554,11 @@ void ProcessHttpRequest(InputMessageBase* msg_base) {
       ctx->cntl.server_ = server;
       ctx->cntl.server_socket_ = sock->id();
       ctx->cntl.remote_side_ = sock->remote_side();
+      HttpHeaderExt& hreq = ctx->cntl.http_request();
+      hreq.method = msg->req.method;
+      for (const auto& kv : msg->req.headers) hreq.headers[kv.first] = kv.second;
+      auto ct = msg->req.headers.find("content-type");
+      if (ct != msg->req.headers.end()) hreq.content_type = ct->second;
       ctx->sid = sock->id();
       ctx->keep_alive = msg->req.keep_alive;
       Closure* done = NewCallback([ctx] {
@@ -548,6 +570,13 @@ void ProcessHttpRequest(InputMessageBase* msg_base) {
           r.body.append(ctx->cntl.ErrorText());
         } else {
           r.content_type = "application/octet-stream";
+          // Handler-set response view (status/content-type/headers).
+          if (ctx->cntl.has_http_response()) {
+            const HttpHeaderExt& hx = ctx->cntl.http_response();
+            if (hx.status_code != 0) r.status = hx.status_code;
+            if (!hx.content_type.empty()) r.content_type = hx.content_type;
+            for (const auto& kv : hx.headers) r.headers[kv.first] = kv.second;
+          }
           r.body.append(std::move(ctx->resp_body));
         }
         SendHttpResponseToSocket(ctx->sid, &r, ctx->keep_alive);

@@ -639,6 +639,74 @@ BenchResult async_echo_bench(const std::string& addr, int payload_size, int pipe
 namespace bam {
 namespace rpctest {
 
+// Controller::http_request()/http_response() (≙ reference HttpHeader
+// accessors, controller.h:400-430): custom verb + headers reach the
+// server handler; handler-set status/content-type/headers reach the
+// client's http_response view.
+bool http_header_ext_test(std::string* err) {
+  Server* server = new Server;
+  Service* svc = new Service("Hx");
+  svc->AddMethod("Probe", [](Controller* cntl, const IOBuf&, IOBuf* resp, Closure* done) {
+    const HttpHeaderExt& req = cntl->http_request();
+    const std::string* custom = req.GetHeader("X-Custom");
+    if (req.method != "PUT" || custom == nullptr || *custom != "v1") {
+      cntl->SetFailed(EREQUEST, "missing verb/header: method=" + req.method);
+      done->Run();
+      return;
+    }
+    HttpHeaderExt& r = cntl->http_response();
+    r.status_code = 202;
+    r.content_type = "application/json";
+    r.SetHeader("X-Served-By", "hx-test");
+    resp->append("{\"ok\":true}");
+    done->Run();
+  });
+  server->AddService(svc, SERVER_OWNS_SERVICE);
+  if (server->Start(0, nullptr) != 0) {
+    *err = "start failed";
+    return false;
+  }
+  std::string addr = "127.0.0.1:" + std::to_string(server->listen_address().port);
+  Channel ch;
+  ChannelOptions opts;
+  opts.protocol = "http";
+  opts.timeout_ms = 3000;
+  opts.max_retry = 0;
+  if (ch.Init(addr.c_str(), &opts) != 0) {
+    *err = "init failed";
+    return false;
+  }
+  Controller cntl;
+  cntl.http_request().method = "PUT";
+  cntl.http_request().SetHeader("X-Custom", "v1");
+  IOBuf q, r;
+  q.append("body");
+  ch.CallMethod("Hx.Probe", &cntl, &q, &r, nullptr);
+  if (cntl.Failed()) {
+    *err = "call failed: " + cntl.ErrorText();
+    return false;
+  }
+  const HttpHeaderExt& hr = cntl.http_response();
+  if (hr.status_code != 202) {
+    *err = "status " + std::to_string(hr.status_code);
+    return false;
+  }
+  if (hr.content_type != "application/json") {
+    *err = "content_type " + hr.content_type;
+    return false;
+  }
+  const std::string* sb = hr.GetHeader("x-served-by");
+  if (sb == nullptr || *sb != "hx-test") {
+    *err = "x-served-by missing";
+    return false;
+  }
+  if (r.to_string() != "{\"ok\":true}") {
+    *err = "body " + r.to_string();
+    return false;
+  }
+  return true;
+}
+
 // Controller::thread_local_data (≙ reference server.h thread_local_data
 // factory): lazily created per worker context, non-null, stable across
 // calls served by the same worker, deleter-counted.

@@ -136,3 +136,10 @@ def test_progressive_chunked_response():
     assert saw_done
     assert b"".join(chunks) == b"helloworld!!!"
     srv.close()
+
+
+def test_http_header_ext():
+    """Controller::http_request()/http_response() custom verb, headers,
+    status and content-type (≙ reference HttpHeader accessors)."""
+    ok, err = r.http_header_ext_test()
+    assert ok, err
