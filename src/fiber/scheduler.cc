@@ -118,6 +118,10 @@ FiberMeta* create_fiber_meta(void (*fn)(void*), void* arg, uint32_t stack_size) 
   }
   m->fn = fn;
   m->arg = arg;
+  // Meta slots are pooled: a previous fiber's interrupt/stop marks must
+  // not leak into this one (they poisoned later sleeps before this reset).
+  m->interrupted.store(false, std::memory_order_relaxed);
+  m->stop_requested.store(false, std::memory_order_relaxed);
   m->stack_size = stack_size;
   m->stack_base = alloc_stack(stack_size);
   CHECK(m->stack_base != nullptr) << "fiber stack allocation failed";
