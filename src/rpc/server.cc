@@ -221,6 +221,25 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   if (limiter_ == nullptr) {
     limiter_ = ConcurrencyLimiter::Create(options_.adaptive_max_concurrency);
   }
+  // gRPC health checking (≙ reference grpc_health_check, brpc/grpc.cpp +
+  // builtin registration server.cpp:501): grpc clients probe
+  // /grpc.health.v1.Health/Check; reply HealthCheckResponse{status:
+  // SERVING} (field 1 varint 1).
+  if (options_.has_builtin_services &&
+      services_.find("grpc.health.v1.Health") == services_.end()) {
+    Service* health = new Service("grpc.health.v1.Health");
+    health->AddMethod("Check",
+                      [](Controller*, const IOBuf&, IOBuf* resp, Closure* done) {
+      resp->append("\x08\x01", 2);
+      done->Run();
+    });
+    health->AddMethod("Watch",
+                      [](Controller*, const IOBuf&, IOBuf* resp, Closure* done) {
+      resp->append("\x08\x01", 2);
+      done->Run();
+    });
+    AddService(health, SERVER_OWNS_SERVICE);
+  }
   if (!options_.ssl_cert.empty() || !options_.ssl_key.empty()) {
     ssl_ctx_ = ssl::NewServerCtx(options_.ssl_cert, options_.ssl_key);
     if (ssl_ctx_ == nullptr) {

@@ -121,3 +121,12 @@ def test_grpc_flow_control_large_response(addr):
     resp = fut(big, timeout=20)
     assert resp == big
     channel.close()
+
+
+def test_grpc_health_check(addr):
+    """grpc.health.v1.Health/Check (≙ reference grpc_health_check): the
+    official grpc client probes health and gets SERVING (field 1 = 1)."""
+    with grpc.insecure_channel(addr) as ch:
+        call = ch.unary_unary("/grpc.health.v1.Health/Check")
+        resp = call(b"", timeout=10)
+        assert resp == b"\x08\x01"
