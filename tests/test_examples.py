@@ -10,7 +10,8 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 EXAMPLES = ["multi_threaded_echo.py", "parallel_echo.py", "streaming_echo.py",
             "tls_auth_echo.py", "rtmp_relay.py", "shm_ring_echo.py",
-            "redis_server.py", "grpc_interop.py", "backup_request.py"]
+            "redis_server.py", "grpc_interop.py", "backup_request.py",
+            "proxy_master.py"]
 
 
 @pytest.mark.parametrize("name", EXAMPLES)
