@@ -32,6 +32,7 @@ struct H2ClientStream {
   std::string grpc_message;
   int http_status = 0;
   bool is_grpc_response = false;
+  std::vector<std::pair<std::string, std::string>> resp_headers;
 };
 
 struct H2ClientCtx {
@@ -65,6 +66,17 @@ void flush_client_locked(H2ClientCtx* ctx) {
 }
 
 // Runs with ctx->mu RELEASED (see struct comment).
+namespace {
+void fill_http_response(Controller* cntl, const H2ClientStream& st) {
+  HttpHeaderExt& hr = cntl->http_response();
+  hr.status_code = st.http_status;
+  for (const auto& kv : st.resp_headers) {
+    if (!kv.first.empty() && kv.first[0] != ':') hr.headers[kv.first] = kv.second;
+    if (kv.first == "content-type") hr.content_type = kv.second;
+  }
+}
+}  // namespace
+
 void complete_stream(H2ClientCtx* ctx, H2ClientStream&& st, uint32_t error_code) {
   void* data = nullptr;
   if (session_lock(st.cid, &data) != 0) return;  // timed out already
@@ -78,10 +90,12 @@ void complete_stream(H2ClientCtx* ctx, H2ClientStream&& st, uint32_t error_code)
   } else if (st.grpc_status > 0) {
     cntl->SetFailed(st.grpc_status == 12 ? ENOMETHOD : EINTERNAL,
                     "grpc-status " + std::to_string(st.grpc_status) + ": " + st.grpc_message);
-  } else if (st.http_status != 0 && st.http_status != 200) {
+  } else if (st.http_status != 0 && st.http_status / 100 != 2) {
+    fill_http_response(cntl, st);
     cntl->SetFailed(EHTTP, "h2 status " + std::to_string(st.http_status));
   } else {
     if (st.is_grpc_response && st.body.size() >= 5) st.body.pop_front(5);
+    fill_http_response(cntl, st);
     if (cntl->call.response != nullptr) {
       cntl->call.response->clear();
       cntl->call.response->append(std::move(st.body));
@@ -110,6 +124,7 @@ int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
     cbs.on_header = [ctx](int32_t sid, const std::string& key, const std::string& val) {
       auto it = ctx->streams.find(sid);
       if (it == ctx->streams.end()) return;
+      it->second.resp_headers.emplace_back(key, val);
       if (key == "grpc-status") it->second.grpc_status = atoi(val.c_str());
       else if (key == "grpc-message") it->second.grpc_message = val;
       else if (key == ":status") it->second.http_status = atoi(val.c_str());
@@ -160,15 +175,27 @@ int IssueH2Request(Socket* sock, Controller* cntl, uint64_t cid, bool grpc) {
                      : svc.empty() && !grpc          ? "/" + mn
                                                      : "/" + svc + "/" + mn;
   std::string authority = endpoint2str(cntl->remote_side());
+  // Controller::http_request() overrides (same contract as the h1 client).
+  const HttpHeaderExt* hx = cntl->has_http_request() ? &cntl->http_request() : nullptr;
+  std::string verb = hx != nullptr && !hx->method.empty()
+                         ? hx->method
+                         : std::string(grpc || !payload.empty() ? "POST" : "GET");
+  std::string ctype = grpc ? "application/grpc"
+                     : hx != nullptr && !hx->content_type.empty()
+                         ? hx->content_type
+                         : "application/octet-stream";
   std::vector<hpack::Header> headers = {
-      {":method", grpc || !payload.empty() ? "POST" : "GET"},
+      {":method", verb},
       {":scheme", "http"},
       {":path", path},
       {":authority", authority},
-      {"content-type", grpc ? "application/grpc" : "application/octet-stream"},
+      {"content-type", ctype},
   };
   if (grpc) headers.push_back({"te", "trailers"});
   headers.push_back({"user-agent", grpc ? "brpc-amd-grpc/1.0" : "brpc-amd-h2/1.0"});
+  if (hx != nullptr) {
+    for (const auto& kv : hx->headers) headers.push_back({kv.first, kv.second});
+  }
   std::lock_guard<std::mutex> lk(ctx->mu);
   int32_t stream_id = ctx->session->SubmitRequest(headers, body, /*end_stream=*/true);
   if (stream_id < 0) return -1;
