@@ -1,5 +1,6 @@
 #include "rpc/event_dispatcher.h"
 
+#include <stdlib.h>
 #include <sys/epoll.h>
 #include <unistd.h>
 
@@ -15,9 +16,35 @@ EventDispatcher::EventDispatcher() {
   std::thread([this] { run(); }).detach();
 }
 
-EventDispatcher* EventDispatcher::singleton() {
-  static EventDispatcher* d = new EventDispatcher;
-  return d;
+struct DispatcherSetAccess {
+  static EventDispatcher* make() { return new EventDispatcher; }
+};
+
+namespace {
+struct DispatcherSet {
+  int n;
+  EventDispatcher** d;
+  DispatcherSet() {
+    const char* e = getenv("BAM_EVENT_DISPATCHERS");
+    n = e != nullptr ? atoi(e) : 1;
+    if (n < 1) n = 1;
+    if (n > 16) n = 16;
+    d = new EventDispatcher*[n];
+    for (int i = 0; i < n; ++i) d[i] = DispatcherSetAccess::make();
+  }
+};
+DispatcherSet& dispatchers() {
+  static DispatcherSet* s = new DispatcherSet;
+  return *s;
+}
+}  // namespace
+
+EventDispatcher* EventDispatcher::singleton() { return dispatchers().d[0]; }
+
+EventDispatcher* EventDispatcher::dispatcher_for(SocketId sid) {
+  DispatcherSet& s = dispatchers();
+  // low id bits are the pool slot: spreads adjacent sockets round-robin
+  return s.d[(uint32_t)sid % (uint32_t)s.n];
 }
 
 int EventDispatcher::add_consumer(SocketId sid, int fd) {

@@ -1,6 +1,8 @@
-// brpc_amd: EventDispatcher — edge-triggered epoll loop on a dedicated
-// pthread; consumers are Sockets addressed by versioned id.
-// Parity: reference brpc/event_dispatcher.h (epoll variant).
+// brpc_amd: EventDispatcher — edge-triggered epoll loops on dedicated
+// pthreads; consumers are Sockets addressed by versioned id. N loops
+// (BAM_EVENT_DISPATCHERS, default 1; ≙ reference -event_dispatcher_num,
+// brpc/event_dispatcher.h:197) shard sockets by id so one epoll thread
+// is not the ceiling at several hundred kQPS.
 #pragma once
 
 #include <atomic>
@@ -11,13 +13,15 @@ namespace bam {
 
 class EventDispatcher {
  public:
-  static EventDispatcher* singleton();
+  static EventDispatcher* singleton();          // shard 0 (legacy callers)
+  static EventDispatcher* dispatcher_for(SocketId sid);
 
   // Registers fd with EPOLLIN|EPOLLOUT|EPOLLET, data = socket id.
   int add_consumer(SocketId sid, int fd);
   int remove_consumer(int fd);
 
  private:
+  friend struct DispatcherSetAccess;
   EventDispatcher();
   void run();
 

@@ -136,7 +136,7 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   *id = s->id_;
   track_socket(s->id_);
   if (fd >= 0) {
-    if (EventDispatcher::singleton()->add_consumer(s->id_, fd) != 0) {
+    if (EventDispatcher::dispatcher_for(s->id_)->add_consumer(s->id_, fd) != 0) {
       s->SetFailed(errno, "epoll add failed");
       return -1;
     }
@@ -175,7 +175,7 @@ int Socket::SetFailed(int error_code, const char* error_text) {
   error_text_ = error_text != nullptr ? error_text : "";
   untrack_socket(id_);
   int fd = fd_.load(std::memory_order_acquire);
-  if (fd >= 0) EventDispatcher::singleton()->remove_consumer(fd);
+  if (fd >= 0) EventDispatcher::dispatcher_for(id_)->remove_consumer(fd);
   // Wake writers parked on epollout.
   epollout_butex_->fetch_add(1, std::memory_order_release);
   butex_wake_all(epollout_butex_);

@@ -2,9 +2,10 @@
 """GPU kernel perf probe: run on the MI355X box, summary goes to stdout
 (redirect into gpurun_out/, then commit the summary under profiles/)."""
 import json
+import os
 import sys
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import brpc_amd as b
 
 g = b.core.gpu
