@@ -46,6 +46,7 @@ bool attachment_test(const std::string& addr);
 bool pb_stub_test(std::string* err);
 bool channel_options_tail_test(std::string* err);
 bool thread_local_data_test(std::string* err);
+bool request_code_test(std::string* err);
 bool http_header_ext_test(std::string* err);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
@@ -145,6 +146,15 @@ void bind_rpc(py::module_& m) {
     {
       py::gil_scoped_release rel;
       ok = bam::rpctest::http_header_ext_test(&err);
+    }
+    return py::make_tuple(ok, err);
+  });
+  r.def("request_code_test", []() {
+    std::string err;
+    bool ok;
+    {
+      py::gil_scoped_release rel;
+      ok = bam::rpctest::request_code_test(&err);
     }
     return py::make_tuple(ok, err);
   });

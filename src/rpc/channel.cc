@@ -184,7 +184,10 @@ void IssueRPC(Controller* cntl) {
   // Select a server.
   EndPoint ep = cntl->call.server_ep;
   if (cntl->call.lb != nullptr) {
-    if (cntl->call.lb->SelectServer(&ep) != 0) {
+    int sel_rc = cntl->has_request_code()
+                     ? cntl->call.lb->SelectServerByCode(cntl->request_code(), &ep)
+                     : cntl->call.lb->SelectServer(&ep);
+    if (sel_rc != 0) {
       session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
       return;
     }

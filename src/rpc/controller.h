@@ -63,6 +63,15 @@ class Controller {
   void set_log_id(uint64_t id) { log_id_ = id; }
   uint64_t log_id() const { return log_id_; }
 
+  // Consistent-hash routing key (≙ reference Controller::set_request_code,
+  // controller.h: c_hash LBs route by this instead of a random pick).
+  void set_request_code(uint64_t code) {
+    request_code_ = code;
+    has_request_code_ = true;
+  }
+  uint64_t request_code() const { return request_code_; }
+  bool has_request_code() const { return has_request_code_; }
+
   // Trace ids (≙ reference brpc/span.h trace propagation): client calls
   // inherit the ambient trace (rpcz::current_trace) unless set explicitly;
   // servers read them off RpcRequestMeta and re-export while the handler
@@ -182,6 +191,8 @@ class Controller {
   int max_retry_ = 3;
   int retry_count_ = 0;
   uint64_t log_id_ = 0;
+  uint64_t request_code_ = 0;
+  bool has_request_code_ = false;
   std::unique_ptr<HttpHeaderExt> http_request_;
   std::unique_ptr<HttpHeaderExt> http_response_;
   uint64_t trace_id_ = 0;

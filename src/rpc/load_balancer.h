@@ -18,6 +18,11 @@ class LoadBalancer {
   virtual ~LoadBalancer() {}
   // Returns 0 and a server on success, nonzero if none available.
   virtual int SelectServer(EndPoint* out) = 0;
+  // Keyed selection (consistent hashing; ≙ reference request_code
+  // routing). Default: ignore the code.
+  virtual int SelectServerByCode(uint64_t /*code*/, EndPoint* out) {
+    return SelectServer(out);
+  }
   // Latency/error feedback after a call (la / p2c use it).
   virtual void Feedback(const EndPoint& server, int error_code, int64_t latency_us) {}
   virtual void SetServers(const std::vector<EndPoint>& servers) = 0;
@@ -38,6 +43,9 @@ class LoadBalancerWithNaming : public LoadBalancer {
   ~LoadBalancerWithNaming() override;
 
   int SelectServer(EndPoint* out) override { return lb_->SelectServer(out); }
+  int SelectServerByCode(uint64_t code, EndPoint* out) override {
+    return lb_->SelectServerByCode(code, out);
+  }
   void Feedback(const EndPoint& server, int error_code, int64_t latency_us) override {
     lb_->Feedback(server, error_code, latency_us);
   }

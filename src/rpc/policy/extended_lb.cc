@@ -69,6 +69,10 @@ class ConsistentHashLB : public LoadBalancer {
     return SelectByKey(fast_rand(), out);
   }
 
+  int SelectServerByCode(uint64_t code, EndPoint* out) override {
+    return SelectByKey(code, out);
+  }
+
   int SelectByKey(uint64_t key, EndPoint* out) {
     std::lock_guard<std::mutex> lk(mu_);
     if (ring_.empty()) return ENODATA;

@@ -4,6 +4,7 @@
 // process over 127.0.0.1.
 #include <algorithm>
 #include <atomic>
+#include <set>
 #include <memory>
 #include <mutex>
 #include <vector>
@@ -702,6 +703,57 @@ bool http_header_ext_test(std::string* err) {
   }
   if (r.to_string() != "{\"ok\":true}") {
     *err = "body " + r.to_string();
+    return false;
+  }
+  return true;
+}
+
+// Controller::set_request_code + c_hash (≙ reference request_code
+// consistent-hash routing): same code -> same backend on every call;
+// different codes spread across backends.
+bool request_code_test(std::string* err) {
+  std::string urls;
+  for (int i = 0; i < 3; ++i) {
+    int port = start_echo_server(0);
+    if (!urls.empty()) urls += ",";
+    urls += "127.0.0.1:" + std::to_string(port);
+  }
+  Channel ch;
+  ChannelOptions opts;
+  opts.timeout_ms = 3000;
+  if (ch.Init(("list://" + urls).c_str(), "c_hash", &opts) != 0) {
+    *err = "init failed";
+    return false;
+  }
+  auto port_for = [&](uint64_t code, std::string* got) -> bool {
+    Controller cntl;
+    cntl.set_request_code(code);
+    IOBuf q, r;
+    q.append("x");
+    ch.CallMethod("EchoService.Port", &cntl, &q, &r, nullptr);
+    if (cntl.Failed()) {
+      *err = "call failed: " + cntl.ErrorText();
+      return false;
+    }
+    *got = r.to_string();
+    return true;
+  };
+  std::set<std::string> spread;
+  for (uint64_t code : {7ull, 91ull, 1234567ull, 42424242ull, 777777777ull}) {
+    std::string first;
+    if (!port_for(code, &first)) return false;
+    for (int rep = 0; rep < 4; ++rep) {
+      std::string again;
+      if (!port_for(code, &again)) return false;
+      if (again != first) {
+        *err = "code " + std::to_string(code) + " moved " + first + "->" + again;
+        return false;
+      }
+    }
+    spread.insert(first);
+  }
+  if (spread.size() < 2) {
+    *err = "5 distinct codes all mapped to one backend";
     return false;
   }
   return true;

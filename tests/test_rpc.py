@@ -141,3 +141,10 @@ def test_thread_local_data():
     thread_local_data_factory): per-worker lazily created, stable."""
     ok, err = r.thread_local_data_test()
     assert ok, err
+
+
+def test_request_code_consistent_hash():
+    """set_request_code routes c_hash deterministically (≙ reference
+    Controller::set_request_code + consistent_hashing LB)."""
+    ok, err = r.request_code_test()
+    assert ok, err
