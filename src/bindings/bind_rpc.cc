@@ -22,6 +22,7 @@ struct BenchResult {
 int start_echo_server(int port);
 int start_nshead_server();
 int start_master_echo_server();
+int start_rdma_mock_echo_server();
 int64_t cancel_test(int port);
 int retry_policy_test(int max_retry);
 // naming resolution (rpc/load_balancer.h)
@@ -50,7 +51,8 @@ bool request_code_test(std::string* err);
 bool http_header_ext_test(std::string* err);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
-                       bool hbm_request, bool pooled, int nchannels);
+                       bool hbm_request, bool pooled, int nchannels,
+                       const std::string& socket_mode = "");
 BenchResult async_echo_bench(const std::string& addr, int payload_size, int pipeline,
                              int64_t total_calls, int timeout_ms, const std::string& method,
                              bool pooled);
@@ -260,6 +262,8 @@ void bind_rpc(py::module_& m) {
       .def("close", &bam::RtmpClient::Close);
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
         py::call_guard<py::gil_scoped_release>());
+  r.def("start_rdma_mock_server", &bam::rpctest::start_rdma_mock_echo_server,
+        py::call_guard<py::gil_scoped_release>());
   r.def("start_master_echo_server", &bam::rpctest::start_master_echo_server,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_nshead_server", &bam::rpctest::start_nshead_server,
@@ -390,6 +394,27 @@ void bind_rpc(py::module_& m) {
         py::arg("total") = 10000, py::arg("timeout_ms") = 5000,
         py::arg("method") = "EchoService.Echo", py::arg("hbm_request") = false,
         py::arg("pooled") = false, py::arg("nchannels") = 1);
+  r.def("echo_bench_mode",
+        [](const std::string& addr, int payload_size, int concurrency, int64_t total,
+           int timeout_ms, const std::string& method, const std::string& socket_mode) {
+          bam::rpctest::BenchResult b;
+          {
+            py::gil_scoped_release rel;
+            b = bam::rpctest::echo_bench(addr, payload_size, concurrency, total,
+                                         timeout_ms, method, false, false, 1,
+                                         socket_mode);
+          }
+          py::dict d;
+          d["qps"] = b.qps;
+          d["p99_us"] = b.p99_us;
+          d["errors"] = b.errors;
+          d["first_error"] = b.first_error;
+          d["total"] = b.total;
+          return d;
+        },
+        py::arg("addr"), py::arg("payload_size"), py::arg("concurrency"),
+        py::arg("total"), py::arg("timeout_ms"), py::arg("method"),
+        py::arg("socket_mode"));
   r.def("async_echo_bench",
         [](const std::string& addr, int payload_size, int pipeline, int64_t total,
            int timeout_ms, const std::string& method, bool pooled) {

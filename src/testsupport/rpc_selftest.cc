@@ -124,6 +124,17 @@ int start_master_echo_server() {
   return server->listen_address().port;
 }
 
+// Echo server whose accepted connections upgrade to the RDMA mock
+// endpoint (ServerOptions.socket_mode="rdma_mock"; rpc/rdma_transport.h).
+int start_rdma_mock_echo_server() {
+  Server* server = new Server;
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  opts.socket_mode = "rdma_mock";
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
 // Starts an echo server that ALSO serves nshead raw-body echo (body is
 // echoed back with "N:" prefixed). Returns port.
 int start_nshead_server() {
@@ -464,7 +475,8 @@ void bench_worker(void* raw) {
 
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
-                       bool hbm_request, bool pooled, int nchannels) {
+                       bool hbm_request, bool pooled, int nchannels,
+                       const std::string& socket_mode) {
   BenchResult res;
   if (nchannels < 1) nchannels = 1;
   std::vector<std::unique_ptr<Channel>> channels(nchannels);
@@ -473,6 +485,7 @@ BenchResult echo_bench(const std::string& addr, int payload_size, int concurrenc
     ChannelOptions opts;
     opts.timeout_ms = timeout_ms;
     if (pooled) opts.connection_type = "pooled";
+    opts.socket_mode = socket_mode;
     if (channels[i]->Init(addr.c_str(), &opts) != 0) return res;
   }
   Channel& channel = *channels[0];
