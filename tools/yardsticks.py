@@ -50,6 +50,7 @@ def main():
     if use_gpu:
         runs += [
             ("hbm64_single_c32", lambda: r.echo_bench(addr, 64, 32, n, 30000, H, False, False, 1)),
+            ("hbm64_single_c64", lambda: r.echo_bench(addr, 64, 64, n, 30000, H, False, False, 1)),
             ("hbm16k_single_c32", lambda: r.echo_bench(addr, 16384, 32, n // 4, 30000, H, False, False, 1)),
         ]
     for tag, fn in runs:
