@@ -126,10 +126,12 @@ class PyChannel {
  public:
   PyChannel(const std::string& addr, const std::string& lb, int timeout_ms, int max_retry,
             int backup_request_ms, int compress, const std::string& auth_user,
-            const std::string& auth_password, bool ssl, const std::string& socket_mode) {
+            const std::string& auth_password, bool ssl, const std::string& socket_mode,
+            const std::string& protocol) {
     ChannelOptions opts;
     opts.ssl = ssl;
     opts.socket_mode = socket_mode;
+    if (!protocol.empty()) opts.protocol = protocol;
     opts.timeout_ms = timeout_ms;
     opts.max_retry = max_retry;
     opts.backup_request_ms = backup_request_ms;
@@ -144,10 +146,12 @@ class PyChannel {
   }
 
   py::tuple call(const std::string& full_method, const std::string& request,
-                 const std::string& attachment, int timeout_ms, uint64_t log_id) {
+                 const std::string& attachment, int timeout_ms, uint64_t log_id,
+                 uint64_t request_code, bool has_code) {
     Controller cntl;
     if (timeout_ms > 0) cntl.set_timeout_ms(timeout_ms);
     if (log_id != 0) cntl.set_log_id(log_id);
+    if (has_code) cntl.set_request_code(request_code);
     cntl.set_request_compress_type(compress_);
     IOBuf req, resp;
     req.append(request);
@@ -160,6 +164,40 @@ class PyChannel {
     return py::make_tuple(py::bytes(resp.to_string()),
                           py::bytes(cntl.response_attachment().to_string()),
                           cntl.latency_us());
+  }
+
+  // HTTP-flavored call: custom verb + headers in, (status, headers, body)
+  // out — the python face of Controller::http_request()/http_response().
+  py::tuple http_call(const std::string& path, py::bytes body,
+                      const std::string& verb, py::dict headers, int timeout_ms) {
+    Controller cntl;
+    if (timeout_ms > 0) cntl.set_timeout_ms(timeout_ms);
+    if (!verb.empty()) cntl.http_request().method = verb;
+    for (auto item : headers) {
+      cntl.http_request().SetHeader(py::cast<std::string>(item.first),
+                                    py::cast<std::string>(item.second));
+    }
+    IOBuf req, resp;
+    req.append(std::string(body));
+    int ec = 0;
+    std::string etext;
+    {
+      py::gil_scoped_release rel;
+      channel_.CallMethod(path, &cntl, &req, &resp, nullptr);
+      if (cntl.Failed()) {
+        ec = cntl.ErrorCode();
+        etext = cntl.ErrorText();
+      }
+    }
+    py::dict rh;
+    int status = 0;
+    if (cntl.has_http_response()) {
+      const HttpHeaderExt& hr = cntl.http_response();
+      status = hr.status_code;
+      for (const auto& kv : hr.headers) rh[py::str(kv.first)] = kv.second;
+    }
+    if (ec != 0 && status == 0) throw PyRpcError(ec, etext);
+    return py::make_tuple(status, rh, py::bytes(resp.to_string()));
   }
 
  private:
@@ -198,14 +236,20 @@ void bind_api(py::module_& m) {
 
   py::class_<PyChannel>(m, "Channel")
       .def(py::init<const std::string&, const std::string&, int, int, int, int,
-                    const std::string&, const std::string&, bool, const std::string&>(),
+                    const std::string&, const std::string&, bool, const std::string&,
+                    const std::string&>(),
            py::arg("addr"), py::arg("lb") = "", py::arg("timeout_ms") = 500,
            py::arg("max_retry") = 3, py::arg("backup_request_ms") = -1,
            py::arg("compress") = 0, py::arg("auth_user") = "", py::arg("auth_password") = "",
-           py::arg("ssl") = false, py::arg("socket_mode") = "")
+           py::arg("ssl") = false, py::arg("socket_mode") = "",
+           py::arg("protocol") = "")
       .def("call", &PyChannel::call, py::arg("method"), py::arg("request"),
            py::arg("attachment") = std::string(), py::arg("timeout_ms") = 0,
-           py::arg("log_id") = 0);
+           py::arg("log_id") = 0, py::arg("request_code") = 0,
+           py::arg("has_request_code") = false)
+      .def("http_call", &PyChannel::http_call, py::arg("path"), py::arg("body") = py::bytes(),
+           py::arg("method") = "", py::arg("headers") = py::dict(),
+           py::arg("timeout_ms") = 0);
 
   m.def("http_call_progressive",
         [](const std::string& addr, const std::string& method, py::bytes request,

@@ -126,3 +126,26 @@ def test_start_cancel():
     lat_us = b.core.rpc.cancel_test(port)
     assert lat_us > 0, "expected ECANCELED, got errno %d" % -lat_us
     assert lat_us < 700000, lat_us  # canceled long before the sleep finished
+
+
+def test_channel_http_call_headers():
+    """Channel(protocol='http').http_call: custom verb + headers out,
+    (status, headers, body) back (python face of the HttpHeaderExt views)."""
+    port = b.core.rpc.start_echo_server(0)
+    ch = b.Channel("127.0.0.1:%d" % port, protocol="http", timeout_ms=3000)
+    status, headers, body = ch.http_call("/health")
+    assert status == 200 and body == b"OK\n"
+    assert "content-type" in headers
+    # RPC over http with a custom header (server echoes the body)
+    status, headers, body = ch.http_call("/EchoService/Echo", b"ping",
+                                         method="POST",
+                                         headers={"X-Probe": "1"})
+    assert status == 200 and body == b"ping"
+
+
+def test_channel_request_code_kwarg():
+    port = b.core.rpc.start_echo_server(0)
+    ch = b.Channel("list://127.0.0.1:%d" % port, lb="c_hash", timeout_ms=3000)
+    resp, att, lat = ch.call("EchoService.Echo", b"x", request_code=42,
+                             has_request_code=True)
+    assert resp == b"x"
