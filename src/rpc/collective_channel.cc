@@ -150,6 +150,14 @@ std::map<std::string, CollectiveMethodFn>& registry() {
     g_registry = new std::map<std::string, CollectiveMethodFn>;
     (*g_registry)["echo"] = echo_fn;
     (*g_registry)["snappy_echo"] = snappy_echo_fn;
+    // Test method: echoes unless BAM_COLL_FAIL=1 in THIS process — drives
+    // the error-marker slot path without desyncing the group.
+    (*g_registry)["maybe_fail_echo"] = [](const void* req, size_t req_len, void* resp,
+                                          size_t resp_cap, size_t* resp_len, int dev) {
+      const char* e = getenv("BAM_COLL_FAIL");
+      if (e != nullptr && e[0] == '1') return 77;
+      return echo_fn(req, req_len, resp, resp_cap, resp_len, dev);
+    };
   }
   return *g_registry;
 }
@@ -196,9 +204,12 @@ int run_round(CommGroup* group, const std::string& method, const void* req_ext,
       *err += ": " + g_snappy_echo_err;
       g_snappy_echo_err.clear();
     }
-    return rc;
+    // DO NOT bail before the all-gather: the peers are already committed
+    // to the collective and a missing participant hangs the group. Ship
+    // an error marker (len = kErrorSlot) so every rank sees a per-rank
+    // failure instead, then report the local error after the gather.
   }
-  uint64_t hdr = resp_len;
+  uint64_t hdr = rc == 0 ? resp_len : CollectiveChannel::kErrorSlot;
   if (device) {
     gpu::api()->memcpy_res(myslot, 2, 0, &hdr, 0, 0, 8);
   } else {
@@ -213,7 +224,7 @@ int run_round(CommGroup* group, const std::string& method, const void* req_ext,
     *err = "allgather failed";
     return EINTERNAL;
   }
-  return 0;
+  return rc;
 }
 
 struct ServerGroupState {

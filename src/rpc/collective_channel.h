@@ -72,6 +72,10 @@ class CollectiveChannel {
   int Call(const std::string& method, const void* req, size_t req_len, void* gathered,
            size_t resp_cap);
 
+  // Slot-length marker a failed participant publishes instead of a
+  // response (the round still completes on every rank).
+  static constexpr uint64_t kErrorSlot = ~0ull;
+
   static size_t slot_size(size_t resp_cap) { return 8 + ((resp_cap + 7) & ~(size_t)7); }
 
   int nranks() const { return group_ != nullptr ? group_->nranks() : 0; }

@@ -182,3 +182,54 @@ def test_stream_over_comm_two_processes():
         outs.append(out)
     assert "RECV_OK" in outs[0]
     assert "SEND_OK" in outs[1]
+
+
+FAIL_WORKER = r"""
+import os, sys
+sys.path.insert(0, %r)
+import brpc_amd as b
+c = b.core.comm
+rank, nranks, port = int(sys.argv[1]), int(sys.argv[2]), int(sys.argv[3])
+if rank == 1:
+    os.environ["BAM_COLL_FAIL"] = "1"
+h = c.create(nranks, rank, "tcp", "127.0.0.1", port)
+if rank == 0:
+    addrs = [""]
+    for r in range(1, nranks):
+        addrs.append("127.0.0.1:%%d" %% int(c.recv(h, r, 5).decode()))
+    payload = b"e" * 512
+    # rank 1 fails its method: the round must still COMPLETE (no hang —
+    # the failing rank publishes kErrorSlot and joins the all-gather) and
+    # the caller must SEE the per-rank failure through the control plane.
+    res = c.fanout_call(h, addrs, "maybe_fail_echo", payload, len(payload), 2, True)
+    assert res["rc"] == 77, res
+    assert "rank 1" in res["error"], res
+    c.host_broadcast(h, b"done", 0)
+    print("CALLER_OK")
+else:
+    p = c.fanout_serve(h, 0)
+    c.send(h, 0, ("%%05d" %% p).encode())
+    out = c.host_broadcast(h, b"", 0)
+    assert out == b"done"
+    print("SERVER_OK", rank)
+"""
+
+
+def test_collective_round_survives_rank_failure():
+    """A participant whose device method fails publishes kErrorSlot and
+    STILL joins the all-gather: the round completes on every rank instead
+    of hanging the group (failure model in docs/multi_gpu.md)."""
+    import subprocess
+    import sys as _sys
+    port = free_port_block()
+    script = FAIL_WORKER % (REPO,)
+    procs = [subprocess.Popen([_sys.executable, "-c", script, str(rk), "2", str(port)],
+                              stdout=subprocess.PIPE, stderr=subprocess.STDOUT)
+             for rk in range(2)]
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=120)
+        outs.append(out.decode())
+    assert procs[0].returncode == 0, outs[0][-1500:]
+    assert procs[1].returncode == 0, outs[1][-1500:]
+    assert "CALLER_OK" in outs[0]
