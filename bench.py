@@ -309,6 +309,12 @@ def main():
     if dist is not None:
         dist.barrier()
         dist.destroy_process_group()
+    sys.stdout.flush()
+    sys.stderr.flush()
+    # torch's HIP context and this process's own kernels double-free in
+    # the ROCm teardown race at interpreter exit (bisected on MI355X).
+    # Results are flushed; skip finalizers for a clean exit code.
+    os._exit(0)
 
 
 if __name__ == "__main__":
