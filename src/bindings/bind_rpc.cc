@@ -48,6 +48,7 @@ bool pb_stub_test(std::string* err);
 bool channel_options_tail_test(std::string* err);
 bool thread_local_data_test(std::string* err);
 bool request_code_test(std::string* err);
+bool short_connection_test(std::string* err);
 bool http_header_ext_test(std::string* err);
 BenchResult echo_bench(const std::string& addr, int payload_size, int concurrency,
                        int64_t total_calls, int timeout_ms, const std::string& method,
@@ -148,6 +149,15 @@ void bind_rpc(py::module_& m) {
     {
       py::gil_scoped_release rel;
       ok = bam::rpctest::http_header_ext_test(&err);
+    }
+    return py::make_tuple(ok, err);
+  });
+  r.def("short_connection_test", []() {
+    std::string err;
+    bool ok;
+    {
+      py::gil_scoped_release rel;
+      ok = bam::rpctest::short_connection_test(&err);
     }
     return py::make_tuple(ok, err);
   });

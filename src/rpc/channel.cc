@@ -93,6 +93,13 @@ static void split_full_method(const std::string& full, std::string* service,
 // session and runs done.
 void EndRPC(Controller* cntl, SessionId locked_id) {
   cntl->end_us_ = monotonic_time_us();
+  if (cntl->call.short_socket != 0) {
+    SocketUniquePtr s;
+    if (Socket::Address(cntl->call.short_socket, &s) == 0) {
+      s->SetFailed(0, "short connection done");
+    }
+    cntl->call.short_socket = 0;
+  }
   if (rpcz::enabled()) {
     rpcz::Span span;
     span.start_us = cntl->start_us_;
@@ -195,7 +202,7 @@ void IssueRPC(Controller* cntl) {
   }
   SocketUniquePtr sock;
   // Fast path: the channel's cached socket (skips the global map mutex).
-  std::atomic<uint64_t>* cache = cntl->call.socket_cache;
+  std::atomic<uint64_t>* cache = cntl->call.short_conn ? nullptr : cntl->call.socket_cache;
   if (cache != nullptr) {
     uint64_t sid = cache->load(std::memory_order_acquire);
     if (sid != 0 && (Socket::Address(sid, &sock) != 0 || sock->Failed())) {
@@ -205,14 +212,19 @@ void IssueRPC(Controller* cntl) {
   if (!sock) {
     if (GetClientSocket(ep, &sock, cntl->call.connection_shard, cntl->call.use_breaker,
                         cntl->call.ssl, cntl->call.protocol_index,
-                        cntl->call.socket_mode) != 0) {
+                        cntl->call.socket_mode, cntl->call.short_conn) != 0) {
       // Conduct the failure through the session so retry/ending logic runs.
       session_error(session_current_id(cntl->call.cid), EFAILEDSOCKET);
       return;
     }
-    if (cache != nullptr) cache->store(sock->id(), std::memory_order_release);
+    if (cntl->call.short_conn) {
+      cntl->call.short_socket = sock->id();
+    } else if (cache != nullptr) {
+      cache->store(sock->id(), std::memory_order_release);
+    }
   }
   cntl->remote_side_ = ep;
+  cntl->local_side_ = sock->local_side();
   cntl->call.auth_data.clear();
   if (cntl->call.auth != nullptr) {
     if (cntl->call.auth->GenerateCredential(&cntl->call.auth_data) != 0) {
@@ -274,6 +286,7 @@ void Channel::CallMethod(const std::string& full_method, Controller* cntl,
   cntl->call.ssl = options_.ssl;
   cntl->call.socket_mode = options_.socket_mode.empty() ? nullptr : options_.socket_mode.c_str();
   cntl->call.retry_policy = options_.retry_policy;
+  cntl->call.short_conn = options_.connection_type == "short";
   if (options_.connection_type == "pooled") {
     static std::atomic<uint32_t> rr{0};
     cntl->call.connection_shard =

@@ -159,6 +159,8 @@ class Controller {
     std::atomic<uint64_t>* socket_cache = nullptr;  // channel's cached socket id
     int connection_shard = 0;                        // pooled connection index
     bool use_breaker = true;                         // ChannelOptions.enable_circuit_breaker
+    bool short_conn = false;                         // connection_type == "short"
+    uint64_t short_socket = 0;                       // to close at EndRPC
     const class Authenticator* auth = nullptr;       // from ChannelOptions
     std::string auth_data;                           // credential for this attempt
     bool ssl = false;                                // TLS client connection

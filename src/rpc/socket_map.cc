@@ -136,10 +136,26 @@ bool IsEndpointIsolated(const EndPoint& ep) {
 }
 
 int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard, bool use_breaker,
-                    bool ssl, int protocol_index, const char* socket_mode) {
+                    bool ssl, int protocol_index, const char* socket_mode,
+                    bool short_conn) {
   if (use_breaker && IsEndpointIsolated(ep)) {
     errno = EHOSTDOWN;
     return -1;
+  }
+  if (short_conn) {
+    // "short" connection type (≙ reference CONNECTION_TYPE_SHORT): a
+    // fresh socket per call, never entered into the map; the caller
+    // closes it when the call completes.
+    InputMessenger* messenger = client_messenger();
+    SocketOptions opts;
+    opts.remote_side = ep;
+    opts.connect_on_create = true;
+    opts.on_edge_triggered_events = [messenger](Socket* s) { messenger->OnNewMessages(s); };
+    SocketId sid;
+    if (Socket::Create(opts, &sid) != 0) return -1;
+    if (Socket::Address(sid, out) != 0) return -1;
+    (*out)->client_protocol_hint = protocol_index;
+    return 0;
   }
   ClientSocketMap& m = the_map();
   // Key = (endpoint, shard, ssl, protocol) folded into one int (parity:

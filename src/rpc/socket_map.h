@@ -14,7 +14,8 @@ namespace bam {
 int GetClientSocket(const EndPoint& ep, SocketUniquePtr* out, int shard = 0,
                     bool use_breaker = true,
                     bool ssl = false, int protocol_index = -1,
-                    const char* socket_mode = nullptr /* "rdma_mock" */);
+                    const char* socket_mode = nullptr /* "rdma_mock" */,
+                    bool short_conn = false /* fresh socket, never pooled */);
 
 // Drops the cached socket for ep (e.g. after failure).
 void RemoveClientSocket(const EndPoint& ep, SocketId expected);

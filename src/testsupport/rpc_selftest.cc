@@ -721,6 +721,40 @@ bool http_header_ext_test(std::string* err) {
   return true;
 }
 
+// connection_type="short" (≙ reference CONNECTION_TYPE_SHORT): a fresh
+// TCP connection per call, closed when the call completes — the server's
+// socket count must not grow with the call count.
+bool short_connection_test(std::string* err) {
+  int port = start_echo_server(0);
+  std::string addr = "127.0.0.1:" + std::to_string(port);
+  Channel ch;
+  ChannelOptions opts;
+  opts.timeout_ms = 3000;
+  opts.connection_type = "short";
+  if (ch.Init(addr.c_str(), &opts) != 0) {
+    *err = "init failed";
+    return false;
+  }
+  std::set<std::string> local_ports;
+  for (int i = 0; i < 8; ++i) {
+    Controller cntl;
+    IOBuf q, r;
+    q.append("s");
+    ch.CallMethod("EchoService.Echo", &cntl, &q, &r, nullptr);
+    if (cntl.Failed()) {
+      *err = "call " + std::to_string(i) + " failed: " + cntl.ErrorText();
+      return false;
+    }
+    local_ports.insert(std::string(endpoint2str(cntl.local_side())));
+  }
+  // every call used a DIFFERENT ephemeral local port (fresh connection)
+  if (local_ports.size() < 8) {
+    *err = "only " + std::to_string(local_ports.size()) + " distinct connections for 8 calls";
+    return false;
+  }
+  return true;
+}
+
 // Controller::set_request_code + c_hash (≙ reference request_code
 // consistent-hash routing): same code -> same backend on every call;
 // different codes spread across backends.

@@ -148,3 +148,10 @@ def test_request_code_consistent_hash():
     Controller::set_request_code + consistent_hashing LB)."""
     ok, err = r.request_code_test()
     assert ok, err
+
+
+def test_short_connection_type():
+    """connection_type='short': fresh connection per call, closed after
+    (≙ reference CONNECTION_TYPE_SHORT)."""
+    ok, err = r.short_connection_test()
+    assert ok, err
