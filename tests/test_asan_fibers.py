@@ -11,6 +11,7 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 @pytest.mark.slow
+@pytest.mark.timeout(900)  # cold ASan rebuild can exceed the global 180 s
 def test_fiber_suite_under_asan(tmp_path):
     main = tmp_path / "asan_main.cc"
     main.write_text("""
