@@ -15,6 +15,7 @@ TARGETS = ["fuzz_rpc_meta", "fuzz_http", "fuzz_redis", "fuzz_json", "fuzz_snappy
 
 
 @pytest.mark.slow
+@pytest.mark.timeout(900)  # make fuzz cold build + 10 targets
 def test_build_and_run_fuzzers():
     build = subprocess.run(["make", "-j16", "fuzz"], cwd=REPO, capture_output=True,
                            text=True, timeout=900)
