@@ -218,6 +218,13 @@ void IssueRPC(Controller* cntl) {
       return;
     }
     if (cntl->call.short_conn) {
+      // A retry replaces the socket: close the previous attempt's now.
+      if (cntl->call.short_socket != 0) {
+        SocketUniquePtr prev;
+        if (Socket::Address(cntl->call.short_socket, &prev) == 0) {
+          prev->SetFailed(0, "short connection superseded by retry");
+        }
+      }
       cntl->call.short_socket = sock->id();
     } else if (cache != nullptr) {
       cache->store(sock->id(), std::memory_order_release);
