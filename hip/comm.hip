@@ -70,12 +70,12 @@ int finish_op(BamComm* c, ncclResult_t rc, const char* what) {
 struct ScopedDev {
   int old = -1;
   explicit ScopedDev(int dev) {
-    hipGetDevice(&old);
-    if (dev != old) hipSetDevice(dev);
+    (void)hipGetDevice(&old);
+    if (dev != old) (void)hipSetDevice(dev);
     else old = -1;
   }
   ~ScopedDev() {
-    if (old >= 0) hipSetDevice(old);
+    if (old >= 0) (void)hipSetDevice(old);
   }
 };
 

@@ -102,7 +102,8 @@ std::once_flag g_init_flag;
 void ensure_init() {
   std::call_once(g_init_flag, [] {
     build_tables();
-    hipMemcpyToSymbol(HIP_SYMBOL(d_tab), h_tab, sizeof(h_tab));
+    if (hipMemcpyToSymbol(HIP_SYMBOL(d_tab), h_tab, sizeof(h_tab)) != hipSuccess)
+      abort();  // tables half-uploaded => silently wrong CRCs; die loudly
     // base-256 digit tables: pows[k][b] = shift by b*256^k bytes
     static uint32_t pows[kPowLevels][256][32];
     for (int n = 0; n < 32; ++n) pows[0][0][n] = 1u << n;  // identity
@@ -114,7 +115,8 @@ void ensure_init() {
       gf2_matmul(pows[k][1], pows[k - 1][255], pows[k - 1][1]);
       for (int b = 2; b < 256; ++b) gf2_matmul(pows[k][b], pows[k][b - 1], pows[k][1]);
     }
-    hipMemcpyToSymbol(HIP_SYMBOL(d_pow256), pows, sizeof(pows));
+    if (hipMemcpyToSymbol(HIP_SYMBOL(d_pow256), pows, sizeof(pows)) != hipSuccess)
+      abort();
   });
 }
 
@@ -193,8 +195,8 @@ extern "C" uint32_t bam_gpu_crc32c(const void* dev_ptr, size_t n, uint32_t init,
   ensure_init();
   if (n == 0) return init;
   int old_dev = -1;
-  hipGetDevice(&old_dev);
-  if (dev != old_dev) hipSetDevice(dev);
+  (void)hipGetDevice(&old_dev);
+  if (dev != old_dev) (void)hipSetDevice(dev);
   const int nchunks = (int)((n + kChunkBytes - 1) / kChunkBytes);
   const int nblocks = (nchunks + 63) / 64;
 
@@ -202,19 +204,27 @@ extern "C" uint32_t bam_gpu_crc32c(const void* dev_ptr, size_t n, uint32_t init,
   {
     std::lock_guard<std::mutex> lk(g_scratch_mu);
     sc = &g_scratch[dev < 16 ? dev : 0];
-    if (sc->result == nullptr) hipMalloc(&sc->result, sizeof(uint32_t));
+    if (sc->result == nullptr &&
+        hipMalloc(&sc->result, sizeof(uint32_t)) != hipSuccess) {
+      sc->result = nullptr;
+      if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+      return init;  // no scratch: report "no progress" rather than garbage
+    }
   }
-  hipMemsetAsync(sc->result, 0, sizeof(uint32_t), 0);
+  (void)hipMemsetAsync(sc->result, 0, sizeof(uint32_t), 0);
   hipLaunchKernelGGL(crc_chunks_kernel, dim3(nblocks), dim3(64), 0, 0,
                      (const uint8_t*)dev_ptr, n, sc->result, nchunks);
   uint32_t crc = 0;
-  hipMemcpy(&crc, sc->result, sizeof(crc), hipMemcpyDeviceToHost);
+  if (hipMemcpy(&crc, sc->result, sizeof(crc), hipMemcpyDeviceToHost) != hipSuccess) {
+    if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+    return init;  // copy failed: do not fabricate a checksum
+  }
 
   if (init != 0) {
     uint32_t op[32];
     build_shift_operator(op, n);
     crc = combine_with_op(op, init, crc);
   }
-  if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
+  if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
   return crc;
 }

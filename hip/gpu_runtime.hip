@@ -92,12 +92,12 @@ int class_of(uint32_t cap) {
 struct ScopedDevice {
   int old = -1;
   explicit ScopedDevice(int dev) {
-    hipGetDevice(&old);
-    if (dev != old) hipSetDevice(dev);
+    (void)hipGetDevice(&old);
+    if (dev != old) (void)hipSetDevice(dev);
     else old = -1;
   }
   ~ScopedDevice() {
-    if (old >= 0) hipSetDevice(old);
+    if (old >= 0) (void)hipSetDevice(old);
   }
 };
 
@@ -224,7 +224,7 @@ extern "C" void bam_gpu_free_hbm(void* p, uint32_t cap, int dev) {
   int cls = class_of(cap);
   if (cls < 0) {
     ScopedDevice sd(dev);
-    hipFree(p);
+    (void)hipFree(p);
     return;
   }
   DevPool& pool = g_pools[dev];
@@ -272,7 +272,7 @@ extern "C" void bam_gpu_free_pinned(void* p, uint32_t cap, int /*dev*/) {
     g_pinned.freelist.emplace_back(p, cap);
     return;
   }
-  hipHostFree(p);
+  (void)hipHostFree(p);
 }
 
 // ---------------- memcpy ----------------
@@ -330,7 +330,10 @@ extern "C" int bam_gpu_fill(void* dst_dev, size_t n, uint64_t pattern, int dev) 
   if (tail != 0) {
     char tail_bytes[8];
     ::memcpy(tail_bytes, &pattern, 8);
-    hipMemcpy((char*)dst_dev + n - tail, tail_bytes, tail, hipMemcpyHostToDevice);
+    if (hipMemcpy((char*)dst_dev + n - tail, tail_bytes, tail,
+                  hipMemcpyHostToDevice) != hipSuccess) {
+      return -1;
+    }
   }
   hipError_t e = hipDeviceSynchronize();
   if (e != hipSuccess) {
@@ -547,10 +550,13 @@ bool direct_init(DirectState& st, int dev) {
   if (hipMalloc(&src_dev, 64) != hipSuccess) return false;
   char pattern[64];
   for (int i = 0; i < 64; ++i) pattern[i] = (char)(i * 7 + 3);
-  hipMemcpy(src_dev, pattern, 64, hipMemcpyHostToDevice);
+  if (hipMemcpy(src_dev, pattern, 64, hipMemcpyHostToDevice) != hipSuccess) {
+    (void)hipFree(src_dev);
+    return false;
+  }
   char* probe = nullptr;
   if (hipHostMalloc((void**)&probe, 64, hipHostMallocDefault) != hipSuccess) {
-    hipFree(src_dev);
+    (void)hipFree(src_dev);
     return false;
   }
   memset(probe, 0, 64);
@@ -569,8 +575,8 @@ bool direct_init(DirectState& st, int dev) {
             memcmp(probe, pattern, 64) == 0;
   st.launched_blocks = 1;
   st.ticket = 1;
-  hipFree(src_dev);
-  hipHostFree(probe);
+  (void)hipFree(src_dev);
+  (void)hipHostFree(probe);
   return ok;
 }
 
@@ -781,7 +787,8 @@ extern "C" int bam_gpu_gather_to_host(void* host_dst, const void* const* srcs,
     std::lock_guard<std::mutex> lk(g_stage_mu);
     StageScratch& s = g_stage[dev < kMaxDev ? dev : 0];
     if (s.cap < total) {
-      if (s.dev != nullptr) hipFree(s.dev);
+      if (s.dev != nullptr) (void)hipFree(s.dev);
+      s.dev = nullptr;
       s.cap = total * 2;
       if (hipMalloc(&s.dev, s.cap) != hipSuccess) {
         s.dev = nullptr;

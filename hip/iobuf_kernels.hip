@@ -148,8 +148,8 @@ template <typename NextFn>
 int run_spans(int nspans, NextFn next, int dev) {
   bam_gpu_quiesce(dev);  // order after any in-flight async HBM uploads
   int old_dev = -1;
-  hipGetDevice(&old_dev);
-  if (dev != old_dev) hipSetDevice(dev);
+  (void)hipGetDevice(&old_dev);
+  if (dev != old_dev) (void)hipSetDevice(dev);
   const int d = dev >= 0 && dev < kMaxDev ? dev : 0;
   const int idx = (int)(g_rr[d].fetch_add(1, std::memory_order_relaxed) % kCtxPerDev);
   SpanCtx& c = g_ctx[d][idx];
@@ -159,7 +159,7 @@ int run_spans(int nspans, NextFn next, int dev) {
     std::lock_guard<std::mutex> lk(c.mu);
     if (c.status == 0) c.status = ctx_init(c, d, idx) ? 1 : -1;
     if (c.status < 0) {
-      if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
+      if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
       return -1;
     }
     // Materialize the spans ONCE (next() is stateful: it accumulates the
@@ -173,7 +173,7 @@ int run_spans(int nspans, NextFn next, int dev) {
       npieces += (spans.back().len + kSubSpan - 1) / kSubSpan;
     }
     if (npieces == 0) {
-      if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
+      if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
       return 0;
     }
     if (npieces <= (size_t)kSpanArgMax) {
@@ -208,8 +208,8 @@ int run_spans(int nspans, NextFn next, int dev) {
       // the device table, ONE launch for every piece.
       if (c.table_cap < npieces) {
         size_t want = npieces * 2;
-        if (c.table_dev != nullptr) hipFree(c.table_dev);
-        if (c.table_pinned != nullptr) hipHostFree(c.table_pinned);
+        if (c.table_dev != nullptr) (void)hipFree(c.table_dev);
+        if (c.table_pinned != nullptr) (void)hipHostFree(c.table_pinned);
         c.table_dev = nullptr;
         c.table_pinned = nullptr;
         c.table_cap = 0;
@@ -261,7 +261,7 @@ int run_spans(int nspans, NextFn next, int dev) {
     c.status = -1;
     rc = -1;
   }
-  if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
+  if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
   return rc;
 }
 

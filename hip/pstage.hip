@@ -202,10 +202,10 @@ extern "C" int bam_gpu_pstage_gather_nocheck(void* host_dst, const void* const* 
     std::lock_guard<std::mutex> lk(st.mu);
     if (!st.available.load(std::memory_order_relaxed)) {
       int old = -1;
-      hipGetDevice(&old);
-      if (dev != old) hipSetDevice(dev);
+      (void)hipGetDevice(&old);
+      if (dev != old) (void)hipSetDevice(dev);
       bool ok = pstage_launch(st, dev);
-      if (dev != old && old >= 0) hipSetDevice(old);
+      if (dev != old && old >= 0) (void)hipSetDevice(old);
       if (!ok) return 1;
     }
   }
@@ -214,11 +214,11 @@ extern "C" int bam_gpu_pstage_gather_nocheck(void* host_dst, const void* const* 
     std::lock_guard<std::mutex> lk(st.mu);
     if (st.ctl->heartbeat == 0) {
       int old = -1;
-      hipGetDevice(&old);
-      if (dev != old) hipSetDevice(dev);
+      (void)hipGetDevice(&old);
+      if (dev != old) (void)hipSetDevice(dev);
       // The previous instance exited cleanly; its stream is idle.
       bool ok = pstage_launch(st, dev);
-      if (dev != old && old >= 0) hipSetDevice(old);
+      if (dev != old && old >= 0) (void)hipSetDevice(old);
       if (!ok) {
         st.available.store(false, std::memory_order_release);
         return 1;
@@ -256,10 +256,10 @@ extern "C" int bam_gpu_pstage_gather_nocheck(void* host_dst, const void* const* 
       std::lock_guard<std::mutex> lk(st.mu);
       if (st.ctl->heartbeat == 0) {
         int old = -1;
-        hipGetDevice(&old);
-        if (dev != old) hipSetDevice(dev);
+        (void)hipGetDevice(&old);
+        if (dev != old) (void)hipSetDevice(dev);
         bool ok = pstage_launch(st, dev);
-        if (dev != old && old >= 0) hipSetDevice(old);
+        if (dev != old && old >= 0) (void)hipSetDevice(old);
         if (!ok) {
           __atomic_store_n(&s.state, 0u, __ATOMIC_RELEASE);
           st.available.store(false, std::memory_order_release);

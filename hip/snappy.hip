@@ -296,8 +296,8 @@ extern "C" int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_
   bam_gpu_quiesce(dev);  // order after any in-flight async HBM uploads
   if (n == 0) return -1;
   int old_dev = -1;
-  hipGetDevice(&old_dev);
-  if (dev != old_dev) hipSetDevice(dev);
+  (void)hipGetDevice(&old_dev);
+  if (dev != old_dev) (void)hipSetDevice(dev);
   const int nchunks = (int)((n + kLaneChunk - 1) / kLaneChunk);
   const int nblocks = (nchunks + 63) / 64;
   SnappyScratch* sc;
@@ -305,15 +305,27 @@ extern "C" int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_
     std::lock_guard<std::mutex> lk(g_snappy_mu);
     sc = &g_snappy[dev < 16 ? dev : 0];
     if (sc->chunk_cap < (size_t)nchunks) {
-      if (sc->per_chunk) hipFree(sc->per_chunk);
-      if (sc->sizes) hipFree(sc->sizes);
-      if (sc->bsums) hipFree(sc->bsums);
+      if (sc->per_chunk) (void)hipFree(sc->per_chunk);
+      if (sc->sizes) (void)hipFree(sc->sizes);
+      if (sc->bsums) (void)hipFree(sc->bsums);
+      sc->per_chunk = nullptr;
+      sc->sizes = nullptr;
+      sc->bsums = nullptr;
       sc->chunk_cap = (size_t)nchunks * 2;
-      hipMalloc(&sc->per_chunk, sc->chunk_cap * kMaxPerChunk);
-      hipMalloc(&sc->sizes, sc->chunk_cap * sizeof(uint32_t));
-      hipMalloc(&sc->bsums, ((sc->chunk_cap + 255) / 256 + 1) * sizeof(uint64_t));
+      if (hipMalloc(&sc->per_chunk, sc->chunk_cap * kMaxPerChunk) != hipSuccess ||
+          hipMalloc(&sc->sizes, sc->chunk_cap * sizeof(uint32_t)) != hipSuccess ||
+          hipMalloc(&sc->bsums, ((sc->chunk_cap + 255) / 256 + 1) * sizeof(uint64_t)) !=
+              hipSuccess) {
+        sc->chunk_cap = 0;  // a failed alloc must not look usable next call
+        if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+        return -1;
+      }
     }
-    if (sc->total == nullptr) hipMalloc(&sc->total, sizeof(uint64_t));
+    if (sc->total == nullptr &&
+        hipMalloc(&sc->total, sizeof(uint64_t)) != hipSuccess) {
+      if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+      return -1;
+    }
   }
   hipLaunchKernelGGL(snappy_compress_kernel, dim3(nblocks), dim3(64), 0, 0,
                      (const uint8_t*)src_dev, n, sc->per_chunk, sc->sizes, nchunks);
@@ -323,17 +335,27 @@ extern "C" int bam_gpu_snappy_compress(const void* src_dev, size_t n, void* dst_
                      sc->bsums);
   hipLaunchKernelGGL(snappy_scan_bsums, dim3(1), dim3(64), 0, 0, sc->bsums, nb, sc->total);
   uint64_t payload_total = 0;
-  hipMemcpy(&payload_total, sc->total, sizeof(payload_total), hipMemcpyDeviceToHost);
+  if (hipMemcpy(&payload_total, sc->total, sizeof(payload_total),
+                hipMemcpyDeviceToHost) != hipSuccess) {
+    if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+    return -1;
+  }
   uint8_t pre[8];
   int pre_n;
   emit_varint_host(pre, n, &pre_n);
-  if (payload_total + pre_n > dst_cap) return -1;
-  hipMemcpy(dst_dev, pre, pre_n, hipMemcpyHostToDevice);
+  if (payload_total + pre_n > dst_cap) {
+    if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+    return -1;
+  }
+  if (hipMemcpy(dst_dev, pre, pre_n, hipMemcpyHostToDevice) != hipSuccess) {
+    if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+    return -1;
+  }
   hipLaunchKernelGGL(snappy_compact, dim3(nchunks), dim3(256), 0, 0, sc->per_chunk,
                      sc->sizes, sc->bsums, (uint8_t*)dst_dev + pre_n, nchunks, kMaxPerChunk);
   hipError_t e = hipDeviceSynchronize();
   *out_len = (size_t)payload_total + pre_n;
-  if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
+  if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
   return e == hipSuccess ? 0 : -1;
 }
 
@@ -341,19 +363,26 @@ extern "C" int bam_gpu_snappy_decompress(const void* src_dev, size_t n, void* ds
                                          size_t dst_cap, size_t* out_len, int dev) {
   bam_gpu_quiesce(dev);  // order after any in-flight async HBM uploads
   int old_dev = -1;
-  hipGetDevice(&old_dev);
-  if (dev != old_dev) hipSetDevice(dev);
+  (void)hipGetDevice(&old_dev);
+  if (dev != old_dev) (void)hipSetDevice(dev);
   SnappyScratch* sc;
   {
     std::lock_guard<std::mutex> lk(g_snappy_mu);
     sc = &g_snappy[dev < 16 ? dev : 0];
-    if (sc->result == nullptr) hipMalloc(&sc->result, sizeof(uint64_t));
+    if (sc->result == nullptr &&
+        hipMalloc(&sc->result, sizeof(uint64_t)) != hipSuccess) {
+      if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+      return -1;
+    }
   }
   hipLaunchKernelGGL(snappy_decompress_kernel, dim3(1), dim3(64), 0, 0,
                      (const uint8_t*)src_dev, n, (uint8_t*)dst_dev, dst_cap, sc->result);
   uint64_t res = 0;
-  hipMemcpy(&res, sc->result, sizeof(res), hipMemcpyDeviceToHost);
-  if (dev != old_dev && old_dev >= 0) hipSetDevice(old_dev);
+  if (hipMemcpy(&res, sc->result, sizeof(res), hipMemcpyDeviceToHost) != hipSuccess) {
+    if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
+    return -1;
+  }
+  if (dev != old_dev && old_dev >= 0) (void)hipSetDevice(old_dev);
   if (res == ~0ULL) return -1;
   *out_len = (size_t)res;
   return 0;
