@@ -18,6 +18,7 @@ int rwlock_test(int nreaders, int nwriters, int iters);
 bool timer_test();
 bool fiber_key_test();
 bool fiber_interrupt_test(std::string* err);
+bool execution_queue_urgent_test(std::string* err);
 bool gpu_wait_selftest();
 int64_t fd_wait_selftest();
 bool stack_class_selftest();
@@ -44,6 +45,15 @@ void bind_fiber(py::module_& m) {
         py::call_guard<py::gil_scoped_release>());
   f.def("timer_test", &bam::selftest::timer_test, py::call_guard<py::gil_scoped_release>());
   f.def("key_test", &bam::selftest::fiber_key_test, py::call_guard<py::gil_scoped_release>());
+  f.def("execution_queue_urgent_test", []() {
+    std::string err;
+    bool ok;
+    {
+      py::gil_scoped_release rel;
+      ok = bam::selftest::execution_queue_urgent_test(&err);
+    }
+    return py::make_tuple(ok, err);
+  });
   f.def("interrupt_test", []() {
     std::string err;
     bool ok;

@@ -34,20 +34,12 @@ class ExecutionQueue {
   }
 
   // Thread/fiber-safe. Returns 0, or EINVAL after stop().
-  int execute(T task) {
-    if (!started_ || stopped_.load(std::memory_order_acquire)) return EINVAL;
-    Node* node = new Node{nullptr, std::move(task)};
-    Node* old = head_.load(std::memory_order_relaxed);
-    do {
-      node->next = old;
-    } while (!head_.compare_exchange_weak(old, node, std::memory_order_release));
-    if (events_.fetch_add(1, std::memory_order_acq_rel) == 0) {
-      inflight_.fetch_add(1, std::memory_order_acq_rel);
-      fiber_t th;
-      if (fiber_start_background(&th, consumer_entry, this) != 0) consumer_entry(this);
-    }
-    return 0;
-  }
+  int execute(T task) { return push(&head_, std::move(task)); }
+
+  // High-priority lane (≙ reference TASK_OPTIONS_URGENT,
+  // bthread/execution_queue.h:78): urgent tasks run before anything
+  // still waiting in the normal lane, preserving order within each lane.
+  int execute_urgent(T task) { return push(&urgent_head_, std::move(task)); }
 
   void stop() { stopped_.store(true, std::memory_order_release); }
 
@@ -56,7 +48,8 @@ class ExecutionQueue {
     for (;;) {
       int v = idle_butex_->load(std::memory_order_acquire);
       if (inflight_.load(std::memory_order_acquire) == 0 &&
-          head_.load(std::memory_order_acquire) == nullptr)
+          head_.load(std::memory_order_acquire) == nullptr &&
+          urgent_head_.load(std::memory_order_acquire) == nullptr)
         return;
       butex_wait(idle_butex_, v, nullptr);
     }
@@ -68,33 +61,52 @@ class ExecutionQueue {
     T value;
   };
 
+  int push(std::atomic<Node*>* lane, T task) {
+    if (!started_ || stopped_.load(std::memory_order_acquire)) return EINVAL;
+    Node* node = new Node{nullptr, std::move(task)};
+    Node* old = lane->load(std::memory_order_relaxed);
+    do {
+      node->next = old;
+    } while (!lane->compare_exchange_weak(old, node, std::memory_order_release));
+    if (events_.fetch_add(1, std::memory_order_acq_rel) == 0) {
+      inflight_.fetch_add(1, std::memory_order_acq_rel);
+      fiber_t th;
+      if (fiber_start_background(&th, consumer_entry, this) != 0) consumer_entry(this);
+    }
+    return 0;
+  }
+
+  // Pops a lane into submission order and appends to *batch.
+  void drain_lane(std::atomic<Node*>* lane, std::vector<T>* batch) {
+    Node* h = lane->exchange(nullptr, std::memory_order_acq_rel);
+    Node* fifo = nullptr;
+    while (h != nullptr) {
+      Node* nx = h->next;
+      h->next = fifo;
+      fifo = h;
+      h = nx;
+    }
+    while (fifo != nullptr) {
+      batch->push_back(std::move(fifo->value));
+      Node* nx = fifo->next;
+      delete fifo;
+      fifo = nx;
+    }
+  }
+
   static void consumer_entry(void* raw) {
     ((ExecutionQueue*)raw)->consume();
   }
 
   void consume() {
     for (;;) {
-      Node* h = head_.exchange(nullptr, std::memory_order_acq_rel);
-      if (h != nullptr) {
-        // reverse into submission order
-        Node* fifo = nullptr;
-        while (h != nullptr) {
-          Node* nx = h->next;
-          h->next = fifo;
-          fifo = h;
-          h = nx;
-        }
-        std::vector<T> batch;
-        while (fifo != nullptr) {
-          batch.push_back(std::move(fifo->value));
-          Node* nx = fifo->next;
-          delete fifo;
-          fifo = nx;
-        }
-        if (handler_) handler_(batch);
-      }
+      std::vector<T> batch;
+      drain_lane(&urgent_head_, &batch);  // urgent lane jumps the queue
+      drain_lane(&head_, &batch);
+      if (!batch.empty() && handler_) handler_(batch);
       int v = events_.load(std::memory_order_acquire);
       if (head_.load(std::memory_order_acquire) == nullptr &&
+          urgent_head_.load(std::memory_order_acquire) == nullptr &&
           events_.compare_exchange_strong(v, 0, std::memory_order_acq_rel)) {
         break;
       }
@@ -105,6 +117,7 @@ class ExecutionQueue {
   }
 
   std::atomic<Node*> head_{nullptr};
+  std::atomic<Node*> urgent_head_{nullptr};
   std::atomic<int> events_{0};
   std::atomic<int> inflight_{0};
   std::atomic<bool> stopped_{false};

@@ -4,12 +4,14 @@
 #include <errno.h>
 
 #include <atomic>
+#include <mutex>
 #include <string>
 #include <vector>
 
 #include "base/fast_rand.h"
 #include "base/time.h"
 #include "fiber/butex.h"
+#include "fiber/execution_queue.h"
 #include "fiber/fiber.h"
 #include "fiber/sync.h"
 #include "fiber/timer_thread.h"
@@ -569,6 +571,57 @@ bool fiber_interrupt_test(std::string* err) {
   if (st.rc3.load() != -1 || st.e3.load() != ESTOP) {
     *err = "sleep 3: want -1/ESTOP, got " + std::to_string(st.rc3.load()) + "/" +
            std::to_string(st.e3.load());
+    return false;
+  }
+  return true;
+}
+
+
+// ExecutionQueue urgent lane (≙ reference TASK_OPTIONS_URGENT): urgent
+// tasks submitted while the consumer is busy run before earlier normal
+// tasks; order within each lane is preserved.
+bool execution_queue_urgent_test(std::string* err) {
+  ExecutionQueue<int> q;
+  std::vector<int> order;
+  std::mutex mu;
+  std::atomic<bool> gate{false};
+  q.start([&](std::vector<int>& batch) {
+    // first batch blocks until everything is queued so later submissions
+    // land in the lanes, not in this batch
+    if (!gate.exchange(true)) {
+      while (gate.load()) {
+        if (gate.load(std::memory_order_acquire) && order.empty()) {
+          // wait for release marker (order gets -1 pushed by the test)
+          std::lock_guard<std::mutex> lk(mu);
+          if (!order.empty()) break;
+        }
+        fiber_yield();
+        std::lock_guard<std::mutex> lk(mu);
+        if (!order.empty()) break;
+      }
+    }
+    std::lock_guard<std::mutex> lk(mu);
+    for (int v : batch) order.push_back(v);
+  });
+  q.execute(100);  // wakes the consumer; blocks in the gate
+  usleep(20 * 1000);
+  q.execute(1);
+  q.execute(2);
+  q.execute_urgent(91);
+  q.execute_urgent(92);
+  {
+    std::lock_guard<std::mutex> lk(mu);
+    order.push_back(-1);  // release the gate
+  }
+  q.stop();
+  q.join();
+  std::lock_guard<std::mutex> lk(mu);
+  // expected: -1 (marker), 100, then urgent 91,92 BEFORE normal 1,2
+  std::vector<int> want = {-1, 100, 91, 92, 1, 2};
+  if (order != want) {
+    std::string got;
+    for (int v : order) got += std::to_string(v) + ",";
+    *err = "order " + got;
     return false;
   }
   return true;

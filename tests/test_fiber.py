@@ -121,3 +121,10 @@ def test_fiber_interrupt_and_stop():
     bthread_stop, bthread/bthread.h)."""
     ok, err = f.interrupt_test()
     assert ok, err
+
+
+def test_execution_queue_urgent_lane():
+    """execute_urgent jumps queued normal tasks (≙ reference
+    TASK_OPTIONS_URGENT, bthread/execution_queue.h:78)."""
+    ok, err = f.execution_queue_urgent_test()
+    assert ok, err
