@@ -207,6 +207,26 @@ def test_rpcz_persistent_spandb(tmp_path):
         # the file itself is recordio: readable independently
         lines = [l for l in body.splitlines() if "EchoService.Echo" in l]
         assert len(lines) >= 10
+        # a FRESH process (≙ restart) can read the SpanDB file directly
+        import subprocess, sys as _sys
+        out = subprocess.run(
+            [_sys.executable, "-c", (
+                "import sys; sys.path.insert(0, %r)\n"
+                "import brpc_amd as b\n"
+                "rr = b.core.util.RecordReader(%r)\n"
+                "n = 0\n"
+                "while True:\n"
+                "    rec = rr.next()\n"
+                "    if rec is None: break\n"
+                "    assert b'EchoService.Echo' in rec\n"
+                "    n += 1\n"
+                "print('records', n)") % (
+                    __import__('os').path.dirname(
+                        __import__('os').path.dirname(
+                            __import__('os').path.abspath(__file__))), db)],
+            capture_output=True, text=True, timeout=120)
+        assert out.returncode == 0, out.stderr[-800:]
+        assert int(out.stdout.split()[-1]) >= 10, out.stdout
     finally:
         b.core.util.set_flag("rpcz_db_path", "")
         b.core.util.set_flag("rpcz_sample_mod", "16")
