@@ -23,6 +23,7 @@ int start_echo_server(int port);
 int start_nshead_server();
 int start_master_echo_server();
 int start_rdma_mock_echo_server();
+int start_idle_timeout_server(int idle_sec);
 int64_t cancel_test(int port);
 int retry_policy_test(int max_retry);
 // naming resolution (rpc/load_balancer.h)
@@ -271,6 +272,8 @@ void bind_rpc(py::module_& m) {
            py::arg("timeout_ms") = 3000)
       .def("close", &bam::RtmpClient::Close);
   r.def("start_echo_server", &bam::rpctest::start_echo_server, py::arg("port") = 0,
+        py::call_guard<py::gil_scoped_release>());
+  r.def("start_idle_timeout_server", &bam::rpctest::start_idle_timeout_server,
         py::call_guard<py::gil_scoped_release>());
   r.def("start_rdma_mock_server", &bam::rpctest::start_rdma_mock_echo_server,
         py::call_guard<py::gil_scoped_release>());

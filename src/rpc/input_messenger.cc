@@ -1,5 +1,7 @@
 #include "rpc/input_messenger.h"
 
+#include "base/time.h"
+
 #include <errno.h>
 
 #include "base/flags.h"
@@ -82,6 +84,7 @@ void InputMessenger::OnNewMessages(Socket* s) {
       eof = true;
     } else {
       s->in_bytes.fetch_add(nr, std::memory_order_relaxed);
+      s->last_active_us.store(monotonic_time_us(), std::memory_order_relaxed);
     }
 
     // Parse as many complete messages as possible.

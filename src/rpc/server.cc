@@ -3,8 +3,11 @@
 #include <signal.h>
 #include <string.h>
 
+#include <algorithm>
 #include <mutex>
 
+#include "base/time.h"
+#include "fiber/fiber.h"
 #include "rpc/ssl_util.h"
 #include "fiber/butex.h"
 #include "fiber/gpu_wait.h"
@@ -220,6 +223,42 @@ int Server::Start(const EndPoint& ep, const ServerOptions* opt) {
   (void)g_gpu_wakes;
   if (limiter_ == nullptr) {
     limiter_ = ConcurrencyLimiter::Create(options_.adaptive_max_concurrency);
+  }
+  // Idle-connection reaper (≙ reference ServerOptions.idle_timeout_sec,
+  // server.h:62): a background fiber closes server-side connections with
+  // no reads/writes for idle_timeout_sec. The listening socket never has
+  // payload traffic but also never matches (user() != this is skipped;
+  // the listen socket is reaped only by Stop()).
+  if (options_.idle_timeout_sec > 0) {
+    struct ReaperArg {
+      Server* server;
+      int idle_sec;
+    };
+    auto* ra = new ReaperArg{this, options_.idle_timeout_sec};
+    fiber_t th;
+    fiber_start_background(&th, [](void* raw) {
+      ReaperArg* a = (ReaperArg*)raw;
+      const int64_t idle_us = (int64_t)a->idle_sec * 1000000;
+      // sleep FIRST: this fiber starts while Start() is still working and
+      // running_ flips true only at its end.
+      do {
+        fiber_usleep(std::min<int64_t>(idle_us / 2 + 1000, 1000000));
+        std::vector<SocketId> ids;
+        ListSockets(&ids);
+        const int64_t now = monotonic_time_us();
+        for (SocketId id : ids) {
+          SocketUniquePtr sk;
+          if (Socket::Address(id, &sk) != 0) continue;
+          if (sk->user() != (void*)a->server) continue;  // not ours / listener path
+          if (id == a->server->listen_socket_id()) continue;
+          int64_t last = sk->last_active_us.load(std::memory_order_relaxed);
+          if (last != 0 && now - last > idle_us) {
+            sk->SetFailed(ETIMEDOUT, "idle connection reaped");
+          }
+        }
+      } while (a->server->IsRunning());
+      delete a;
+    }, ra);
   }
   // gRPC health checking (≙ reference grpc_health_check, brpc/grpc.cpp +
   // builtin registration server.cpp:501): grpc clients probe

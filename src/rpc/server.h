@@ -161,6 +161,7 @@ class Server {
 
   bool IsRunning() const { return running_.load(std::memory_order_acquire); }
   EndPoint listen_address() const { return listen_ep_; }
+  SocketId listen_socket_id() const { return listen_socket_; }
 
   // Request routing (called by protocol ProcessRequest).
   const MethodFn* FindMethod(const std::string& service, const std::string& method,

@@ -108,6 +108,7 @@ int Socket::Create(const SocketOptions& options, SocketId* id) {
   s->session_local_deleter = nullptr;
   s->transport_ = options.transport;
   s->write_head_.store(nullptr, std::memory_order_relaxed);
+  s->last_active_us.store(monotonic_time_us(), std::memory_order_relaxed);
   s->in_bytes = 0;
   s->out_bytes = 0;
   s->in_messages = 0;
@@ -483,6 +484,7 @@ int Socket::DoWrite(WriteRequest* req) {
       continue;  // loop top runs ReleaseAllWriteRequests
     }
     out_bytes.fetch_add(nw, std::memory_order_relaxed);
+    last_active_us.store(monotonic_time_us(), std::memory_order_relaxed);
     unwritten_bytes.fetch_sub(nw, std::memory_order_relaxed);
     if (!cur->data.empty()) continue;  // partial write; try again
     out_messages.fetch_add(1, std::memory_order_relaxed);

@@ -174,6 +174,8 @@ class Socket {
 
   // Per-connection stats (builtin /connections page).
   std::atomic<int64_t> in_bytes{0};
+  // monotonic µs of the last read or write (idle-timeout reaping).
+  std::atomic<int64_t> last_active_us{0};
   std::atomic<int64_t> out_bytes{0};
   std::atomic<int64_t> in_messages{0};
   std::atomic<int64_t> out_messages{0};

@@ -135,6 +135,16 @@ int start_rdma_mock_echo_server() {
   return server->listen_address().port;
 }
 
+// Echo server with a short idle timeout (reaper test).
+int start_idle_timeout_server(int idle_sec) {
+  Server* server = new Server;
+  server->AddService(NewEchoService(), SERVER_OWNS_SERVICE);
+  ServerOptions opts;
+  opts.idle_timeout_sec = idle_sec;
+  if (server->Start(0, &opts) != 0) return -1;
+  return server->listen_address().port;
+}
+
 // Starts an echo server that ALSO serves nshead raw-body echo (body is
 // echoed back with "N:" prefixed). Returns port.
 int start_nshead_server() {

@@ -155,3 +155,24 @@ def test_short_connection_type():
     (≙ reference CONNECTION_TYPE_SHORT)."""
     ok, err = r.short_connection_test()
     assert ok, err
+
+
+@pytest.mark.timeout(60)
+def test_idle_timeout_reaps_connections():
+    """ServerOptions.idle_timeout_sec closes idle server-side connections
+    (≙ reference idle_timeout_sec, brpc/server.h:62): an open socket with
+    no traffic dies; an active channel keeps working."""
+    import socket as pysock
+    import time
+    port = r.start_idle_timeout_server(1)  # 1-second idle timeout
+    s = pysock.create_connection(("127.0.0.1", port), timeout=5)
+    # connection should be closed by the reaper within ~2.5s
+    s.settimeout(4)
+    t0 = time.time()
+    data = s.recv(1)  # blocks until the server closes (returns b"")
+    assert data == b"", "expected server-side close"
+    assert time.time() - t0 < 3.5
+    s.close()
+    # server still serves fresh connections
+    rc, resp, _ = r.echo_once("127.0.0.1:%d" % port, b"alive", 3000)
+    assert rc == 0 and resp == b"alive"
