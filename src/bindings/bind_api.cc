@@ -197,7 +197,11 @@ class PyChannel {
       for (const auto& kv : hr.headers) rh[py::str(kv.first)] = kv.second;
     }
     if (ec != 0 && status == 0) throw PyRpcError(ec, etext);
-    return py::make_tuple(status, rh, py::bytes(resp.to_string()));
+    // on HTTP error statuses the page body rides the attachment
+    std::string body_out = resp.empty() && status / 100 != 2
+                               ? cntl.response_attachment().to_string()
+                               : resp.to_string();
+    return py::make_tuple(status, rh, py::bytes(body_out));
   }
 
  private:

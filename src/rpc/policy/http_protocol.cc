@@ -388,6 +388,10 @@ void ProcessHttpResponse(InputMessageBase* mb) {
   if (msg->status / 100 != 2) {  // any 2xx is success (204, 202…)
     cntl->SetFailed(EHTTP, "HTTP status " + std::to_string(msg->status) + ": " +
                                msg->body.to_string().substr(0, 200));
+    // keep the error page readable (≙ reference: body stays accessible
+    // via response_attachment on HTTP errors)
+    cntl->response_attachment().clear();
+    cntl->response_attachment().swap(msg->body);
   } else if (cntl->call.response != nullptr) {
     cntl->call.response->clear();
     cntl->call.response->swap(msg->body);

@@ -149,3 +149,31 @@ def test_channel_request_code_kwarg():
     resp, att, lat = ch.call("EchoService.Echo", b"x", request_code=42,
                              has_request_code=True)
     assert resp == b"x"
+
+
+def test_channel_http_call_404_and_error_paths():
+    """http_call surfaces HTTP error statuses as data (status, headers,
+    body), and transport failures as RpcError."""
+    port = b.core.rpc.start_echo_server(0)
+    ch = b.Channel("127.0.0.1:%d" % port, protocol="http", timeout_ms=3000)
+    status, headers, body = ch.http_call("/no/such/page")
+    assert status == 404 and b"no such page" in body
+    bad = b.Channel("127.0.0.1:1", protocol="http", timeout_ms=300)
+    try:
+        bad.http_call("/x")
+        assert False, "expected RpcError"
+    except b.RpcError:
+        pass
+
+
+def test_short_and_pooled_coexist():
+    """A short-conn channel and a pooled channel to the same server work
+    side by side (no registry cross-talk)."""
+    port = b.core.rpc.start_echo_server(0)
+    addr = "127.0.0.1:%d" % port
+    ok, err = b.core.rpc.short_connection_test()  # uses its own server
+    assert ok, err
+    pooled = b.Channel(addr, timeout_ms=2000)
+    for _ in range(4):
+        resp, _, _ = pooled.call("EchoService.Echo", b"both")
+        assert resp == b"both"
