@@ -56,7 +56,7 @@ struct ParallelDone {
   };
   std::vector<SubResult> subs;
 
-  explicit ParallelDone(int n) : pending(n), destroy_refs(n), subs(n), nsubs(n) {}
+  explicit ParallelDone(int n) : pending(n), destroy_refs(n), nsubs(n), subs(n) {}
 
   void OnSubDone(int idx) {
     subs[idx].completed.store(true, std::memory_order_release);
