@@ -143,3 +143,16 @@ def test_http_header_ext():
     status and content-type (≙ reference HttpHeader accessors)."""
     ok, err = r.http_header_ext_test()
     assert ok, err
+
+
+def test_h2_http_call_headers():
+    """Channel(protocol='h2').http_call: the HttpHeaderExt views work on
+    the in-tree h2 client too (custom header out, status/headers back)."""
+    import brpc_amd
+    port = r.start_echo_server(0)
+    ch = brpc_amd.Channel("127.0.0.1:%d" % port, protocol="h2", timeout_ms=3000)
+    status, headers, body = ch.http_call("/health")
+    assert status == 200 and body == b"OK\n"
+    status, headers, body = ch.http_call("/EchoService/Echo", b"h2ping",
+                                         headers={"X-H2-Probe": "yes"})
+    assert status == 200 and body == b"h2ping"

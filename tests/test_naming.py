@@ -118,3 +118,13 @@ def test_discovery_naming():
     assert eps.count("127.0.0.1:%d" % echo_port) == 2, eps
     assert "127.0.0.1:%d" % (echo_port + 1) in eps, eps
     agent.stop()
+
+
+def test_dlist_mixed_and_empty():
+    """dlist:// resolves mixed literal+hostname entries; an unresolvable
+    list fails instead of returning an empty server set."""
+    eps = b.core.rpc.resolve_naming("dlist://127.0.0.3:7,localhost:9")
+    assert "127.0.0.3:7" in eps and "127.0.0.1:9" in eps, eps
+    # wholly-unresolvable list -> empty (the refresher keeps the last
+    # good set; ResolveNamingUrl returns nonzero, binding maps to [])
+    assert b.core.rpc.resolve_naming("dlist://no.such.host.invalid:1") == []
